@@ -1,0 +1,122 @@
+// kernels.h — host-side launch API for the CDNA4 hot-path kernels.
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+
+namespace auron {
+
+// ---- Spark murmur3 (mur.rs:19-87) on device -------------------------------
+// hashes[i] must be pre-initialized with the seed (k_hash_init); each fold
+// call applies one column (spark_hash.rs:28-57 create_murmur3_hashes).
+void launch_hash_init(int32_t* hashes, int32_t seed, int64_t n, hipStream_t s);
+void launch_hash_fold_i64(const int64_t* vals, const uint8_t* valid, int64_t n,
+                          int32_t* hashes, hipStream_t s);
+// part_id = pmod(hash, P) (shuffle/mod.rs:178-188)
+void launch_pmod(const int32_t* hashes, int64_t n, int32_t num_partitions,
+                 uint32_t* part_ids, hipStream_t s);
+
+// ---- hash aggregation ------------------------------------------------------
+// GPU replacement for AggHashMap::upsert_records + per-agg partial_update
+// (agg_hash_map.rs:77-168, sum.rs:90-115, count.rs:90-149).
+// Open-addressing table over i64 keys; slot i64::MIN is the empty sentinel,
+// the two trailing acc rows (index cap, cap+1) hold the i64::MIN-key group
+// and the null-key group.
+struct AggTable {
+  int64_t cap = 0;           // power of two
+  int64_t* slot_key = nullptr;        // [cap], EMPTY = INT64_MIN
+  uint32_t* special_used = nullptr;   // [2]: {min-key group, null-key group}
+  // accumulators indexed by slot (cap + 2 entries):
+  double* sum = nullptr;
+  unsigned long long* cnt = nullptr;
+  uint32_t* sum_valid = nullptr;
+  unsigned long long* first_row = nullptr;  // global arrival index, for order
+  unsigned long long* num_groups = nullptr; // [1] device counter
+};
+
+// grid-stride row update: keys/vals length n, rows globally numbered starting
+// at row_offset (for cross-batch first-occurrence order).
+void launch_agg_update(const AggTable& t, const int64_t* keys,
+                       const uint8_t* key_valid, const double* vals,
+                       const uint8_t* val_valid, int64_t n, uint64_t row_offset,
+                       hipStream_t s);
+
+// merge rows of frozen partial state: per row [u8 valid][8B LE sum]?[varint cnt]
+// (acc.rs:335-347 + count.rs:193-211 + io/mod.rs:60-79)
+void launch_agg_merge_frozen(const AggTable& t, const int64_t* keys,
+                             const uint8_t* key_valid, const uint8_t* acc_data,
+                             const int32_t* acc_offsets, int64_t n,
+                             uint64_t row_offset, hipStream_t s);
+
+// compact occupied slots to dense arrays (unordered); returns count via
+// num_out (device). out_slot holds the source slot index per group.
+void launch_agg_compact(const AggTable& t, uint32_t* out_slot,
+                        unsigned long long* out_first_row,
+                        unsigned long long* num_out, hipStream_t s);
+
+// gather ordered group outputs given order[] (group -> slot):
+// keys + key validity bits, sums + validity bits, counts
+void launch_agg_gather_out(const AggTable& t, const uint32_t* order_slots,
+                           int64_t num_groups, int64_t* out_keys,
+                           uint8_t* out_key_validity, double* out_sums,
+                           uint8_t* out_sum_validity,
+                           long long* out_counts, hipStream_t s);
+
+// freeze ordered groups into the Binary agg-buf wire format (a8):
+// pass 1 computes per-group byte length, host scans, pass 2 writes bytes.
+void launch_agg_freeze_len(const AggTable& t, const uint32_t* order_slots,
+                           int64_t num_groups, int32_t* lens, hipStream_t s);
+void launch_agg_freeze_write(const AggTable& t, const uint32_t* order_slots,
+                             int64_t num_groups, const int32_t* offsets,
+                             uint8_t* out, hipStream_t s);
+
+// partial-skipping pass-through freeze (agg_ctx.rs:428-462)
+void launch_skip_freeze_len(const uint8_t* val_valid, int64_t n, int32_t* lens,
+                            hipStream_t s);
+void launch_skip_freeze_write(const double* vals, const uint8_t* val_valid,
+                              int64_t n, const int32_t* offsets, uint8_t* out,
+                              hipStream_t s);
+
+// fill an i64 array with a constant (table init: slot_key = INT64_MIN)
+void launch_fill_i64(int64_t* dst, int64_t value, int64_t n, hipStream_t s);
+
+// rebuild `src` table into the (larger, initialized) `dst` table; each source
+// slot holds a distinct key so plain stores after the CAS claim are race-free.
+void launch_agg_rebuild(const AggTable& dst, const AggTable& src, hipStream_t s);
+
+// ---- shuffle partition + gather -------------------------------------------
+// per-partition histogram of part_ids (LDS-staged, one global atomic per
+// (block, partition))
+void launch_histogram(const uint32_t* part_ids, int64_t n, uint32_t P,
+                      uint32_t* counts /* zeroed [P] */, hipStream_t s);
+// stable partition permutation = rocprim stable radix sort of
+// (part_id, row_idx) pairs over ceil(log2 P) bits
+void sort_pairs_u32_u32(const uint32_t* keys_in, const uint32_t* vals_in,
+                        uint32_t* keys_out, uint32_t* vals_out, int64_t n,
+                        int end_bit, void* temp, size_t* temp_bytes,
+                        hipStream_t s);
+void launch_iota_u32(uint32_t* dst, int64_t n, hipStream_t s);
+
+void launch_gather_8(const uint8_t* src, const uint32_t* perm, int64_t n,
+                     uint8_t* dst, hipStream_t s);  // 8-byte elements
+void launch_gather_bits(const uint8_t* src_bits, const uint32_t* perm, int64_t n,
+                        uint8_t* dst_bits, hipStream_t s);
+// binary column gather: lens then bytes (offsets host-scanned)
+void launch_gather_lens(const int32_t* src_offsets, const uint32_t* perm,
+                        int64_t n, int32_t* dst_lens, hipStream_t s);
+void launch_gather_bytes(const uint8_t* src_data, const int32_t* src_offsets,
+                         const uint32_t* perm, const int32_t* dst_offsets,
+                         int64_t n, uint8_t* dst_data, hipStream_t s);
+
+// sort groups by first_row: rocprim radix sort pairs wrapper
+void sort_pairs_u64_u32(const unsigned long long* keys_in, const uint32_t* vals_in,
+                        unsigned long long* keys_out, uint32_t* vals_out,
+                        int64_t n, void* temp, size_t* temp_bytes, hipStream_t s);
+
+// byte-plane transpose for batch_serde (batch_serde.rs:271-306): out[b*n+i] =
+// in[i*w+b]
+void launch_byte_transpose(const uint8_t* in, uint8_t* out, int64_t n, int w,
+                           hipStream_t s);
+
+}  // namespace auron

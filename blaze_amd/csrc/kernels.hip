@@ -1,0 +1,579 @@
+// kernels.hip — hand-written CDNA4 (gfx950) kernels for the Auron hot path.
+//
+// This path is hash/scatter work bound by HBM streaming + cache-side atomics —
+// MFMA is irrelevant (BASELINE.json north_star). Design notes:
+//  * wave64: block size 256 (4 waves), grid-stride loops capped at
+//    256 CU × 8 blocks (guide §6 G11), so one launch fills all 8 XCDs.
+//  * input reads are coalesced 8 B/lane streams; the hash table is sized so
+//    the hot accumulator lines live in L2/L3 (1M groups ≈ 50 MB < 256 MB L3),
+//    keeping HBM traffic near the 16 B/row algorithmic minimum.
+//  * f64 scatter-add uses the native global_atomic_add_f64 (unsafeAtomicAdd;
+//    compiled with -munsafe-fp-atomics); i64/u64 counts use integer atomics
+//    (bit-exact, order-free). Float SUM order is therefore arrival-order-free;
+//    parity bar per BASELINE.json: 1e-6 relative for SUM/AVG(f64), bit-exact
+//    integers.
+//  * record order: first-occurrence (atomicMin on a per-group arrival index,
+//    deterministic) matching AggHashMap's dense-index insertion order
+//    (agg_hash_map.rs:77-168).
+#include <hip/hip_runtime.h>
+
+#include <rocprim/device/device_radix_sort.hpp>
+
+#include "kernels.h"
+
+namespace auron {
+
+static constexpr int BLOCK = 256;
+static constexpr int64_t MAX_BLOCKS = 256 * 8;
+
+static inline int grid_for(int64_t n) {
+  int64_t b = (n + BLOCK - 1) / BLOCK;
+  if (b > MAX_BLOCKS) b = MAX_BLOCKS;
+  if (b < 1) b = 1;
+  return (int)b;
+}
+
+// ---- murmur3 (bit-exact restatement of mur.rs:19-87) ----------------------
+__device__ __forceinline__ uint32_t rotl32(uint32_t x, int r) {
+  return (x << r) | (x >> (32 - r));
+}
+
+__device__ __forceinline__ int32_t mur_mix_k1(int32_t k1) {
+  uint32_t k = (uint32_t)k1;
+  k *= 0xcc9e2d51u;
+  k = rotl32(k, 15);
+  k *= 0x1b873593u;
+  return (int32_t)k;
+}
+
+__device__ __forceinline__ int32_t mur_mix_h1(int32_t h1, int32_t k1) {
+  uint32_t h = (uint32_t)h1 ^ (uint32_t)k1;
+  h = rotl32(h, 13);
+  h = h * 5u + 0xe6546b64u;
+  return (int32_t)h;
+}
+
+__device__ __forceinline__ int32_t mur_fmix(int32_t h1, int32_t len) {
+  uint32_t h = (uint32_t)h1 ^ (uint32_t)len;
+  h ^= h >> 16;
+  h *= 0x85ebca6bu;
+  h ^= h >> 13;
+  h *= 0xc2b2ae35u;
+  h ^= h >> 16;
+  return (int32_t)h;
+}
+
+__device__ __forceinline__ int32_t murmur3_long(int64_t value, int32_t seed) {
+  int32_t low = (int32_t)value;
+  int32_t high = (int32_t)((uint64_t)value >> 32);
+  int32_t h1 = mur_mix_h1(seed, mur_mix_k1(low));
+  h1 = mur_mix_h1(h1, mur_mix_k1(high));
+  return mur_fmix(h1, 8);
+}
+
+__device__ __forceinline__ bool bit_get_dev(const uint8_t* bm, int64_t i) {
+  return (bm[i >> 3] >> (i & 7)) & 1;
+}
+
+__global__ void k_hash_init(int32_t* hashes, int32_t seed, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    hashes[i] = seed;
+}
+
+__global__ void k_hash_fold_i64(const int64_t* __restrict__ vals,
+                                const uint8_t* __restrict__ valid, int64_t n,
+                                int32_t* __restrict__ hashes) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    if (!valid || bit_get_dev(valid, i))
+      hashes[i] = murmur3_long(vals[i], hashes[i]);
+  }
+}
+
+__global__ void k_pmod(const int32_t* __restrict__ hashes, int64_t n, int32_t P,
+                       uint32_t* __restrict__ part_ids) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int32_t r = hashes[i] % P;           // rem_euclid, shuffle/mod.rs:180-183
+    part_ids[i] = (uint32_t)(r < 0 ? r + P : r);
+  }
+}
+
+void launch_hash_init(int32_t* hashes, int32_t seed, int64_t n, hipStream_t s) {
+  hipLaunchKernelGGL(k_hash_init, dim3(grid_for(n)), dim3(BLOCK), 0, s, hashes,
+                     seed, n);
+}
+void launch_hash_fold_i64(const int64_t* vals, const uint8_t* valid, int64_t n,
+                          int32_t* hashes, hipStream_t s) {
+  hipLaunchKernelGGL(k_hash_fold_i64, dim3(grid_for(n)), dim3(BLOCK), 0, s,
+                     vals, valid, n, hashes);
+}
+void launch_pmod(const int32_t* hashes, int64_t n, int32_t P, uint32_t* part_ids,
+                 hipStream_t s) {
+  hipLaunchKernelGGL(k_pmod, dim3(grid_for(n)), dim3(BLOCK), 0, s, hashes, n, P,
+                     part_ids);
+}
+
+// ---- hash aggregation ------------------------------------------------------
+static constexpr int64_t KEY_EMPTY = INT64_MIN;
+
+// internal slot hash — NOT part of the parity contract (agg_hash_map.rs's
+// foldhash only shapes its private layout); a strong 64-bit mix keeps probes
+// short at any key distribution.
+__device__ __forceinline__ uint64_t mix64(uint64_t x) {
+  x += 0x9E3779B97F4A7C15ull;
+  x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+  x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+  return x ^ (x >> 31);
+}
+
+// probe-or-insert; returns acc index (slot, or cap/cap+1 for the special
+// groups). Claims bump *num_groups.
+__device__ __forceinline__ int64_t agg_upsert_slot(const AggTable t, int64_t key,
+                                                   bool key_null) {
+  if (key_null || key == KEY_EMPTY) {
+    int which = key_null ? 1 : 0;
+    if (atomicCAS(&t.special_used[which], 0u, 1u) == 0u)
+      atomicAdd(t.num_groups, 1ull);
+    return t.cap + which;
+  }
+  uint64_t h = mix64((uint64_t)key);
+  int64_t mask = t.cap - 1;
+  int64_t i = (int64_t)(h & (uint64_t)mask);
+  while (true) {
+    int64_t cur = t.slot_key[i];
+    if (cur == key) return i;
+    if (cur == KEY_EMPTY) {
+      int64_t prev = (int64_t)atomicCAS((unsigned long long*)&t.slot_key[i],
+                                        (unsigned long long)KEY_EMPTY,
+                                        (unsigned long long)key);
+      if (prev == KEY_EMPTY) {
+        atomicAdd(t.num_groups, 1ull);
+        return i;
+      }
+      if (prev == key) return i;
+    }
+    i = (i + 1) & mask;
+  }
+}
+
+__global__ void k_agg_update(const AggTable t, const int64_t* __restrict__ keys,
+                             const uint8_t* __restrict__ key_valid,
+                             const double* __restrict__ vals,
+                             const uint8_t* __restrict__ val_valid, int64_t n,
+                             uint64_t row_offset) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    bool knull = key_valid && !bit_get_dev(key_valid, i);
+    int64_t a = agg_upsert_slot(t, knull ? 0 : keys[i], knull);
+    atomicMin(&t.first_row[a], row_offset + (uint64_t)i);
+    bool vvalid = !val_valid || bit_get_dev(val_valid, i);
+    if (vvalid) {
+      // sum.rs:90-115: SUM adds non-null args; valid-ness latches on
+      unsafeAtomicAdd(&t.sum[a], vals[i]);
+      atomicAdd(&t.cnt[a], 1ull);  // count.rs:90-149: COUNT(arg) non-null
+      if (!t.sum_valid[a]) atomicOr(&t.sum_valid[a], 1u);
+    }
+  }
+}
+
+// varint read (io/mod.rs:69-79)
+__device__ __forceinline__ uint64_t read_varint_dev(const uint8_t* p, int* used) {
+  uint64_t len = 0, factor = 1;
+  int k = 0;
+  while (true) {
+    uint8_t v = p[k++];
+    if (v < 128) {
+      len += (uint64_t)v * factor;
+      break;
+    }
+    len += (uint64_t)(v - 128) * factor;
+    factor *= 128;
+  }
+  *used = k;
+  return len;
+}
+
+__global__ void k_agg_merge_frozen(const AggTable t,
+                                   const int64_t* __restrict__ keys,
+                                   const uint8_t* __restrict__ key_valid,
+                                   const uint8_t* __restrict__ acc_data,
+                                   const int32_t* __restrict__ acc_offsets,
+                                   int64_t n, uint64_t row_offset) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    bool knull = key_valid && !bit_get_dev(key_valid, i);
+    int64_t a = agg_upsert_slot(t, knull ? 0 : keys[i], knull);
+    atomicMin(&t.first_row[a], row_offset + (uint64_t)i);
+    const uint8_t* p = acc_data + acc_offsets[i];
+    uint8_t valid = *p++;
+    if (valid) {  // sum.rs:117-145 partial_merge
+      double v;
+      memcpy(&v, p, 8);
+      p += 8;
+      unsafeAtomicAdd(&t.sum[a], v);
+      if (!t.sum_valid[a]) atomicOr(&t.sum_valid[a], 1u);
+    }
+    int used;
+    uint64_t c = read_varint_dev(p, &used);
+    if (c) atomicAdd(&t.cnt[a], c);
+  }
+}
+
+__global__ void k_agg_compact(const AggTable t, uint32_t* __restrict__ out_slot,
+                              unsigned long long* __restrict__ out_first_row,
+                              unsigned long long* __restrict__ num_out) {
+  int64_t total = t.cap + 2;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    bool used = (i < t.cap) ? (t.slot_key[i] != KEY_EMPTY)
+                            : (t.special_used[i - t.cap] != 0);
+    if (used) {
+      unsigned long long idx = atomicAdd(num_out, 1ull);
+      out_slot[idx] = (uint32_t)i;
+      out_first_row[idx] = t.first_row[i];
+    }
+  }
+}
+
+__global__ void k_agg_gather_out(const AggTable t,
+                                 const uint32_t* __restrict__ order_slots,
+                                 int64_t num_groups, int64_t* __restrict__ keys,
+                                 uint8_t* __restrict__ key_validity,
+                                 double* __restrict__ sums,
+                                 uint8_t* __restrict__ sum_validity,
+                                 long long* __restrict__ counts) {
+  // one thread per output byte-group of 8 rows for validity bitmaps
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < num_groups;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    uint32_t s = order_slots[i];
+    if (keys) keys[i] = (s < t.cap) ? t.slot_key[s]
+                                    : (s == t.cap ? KEY_EMPTY : 0);
+    if (sums) sums[i] = t.sum[s];
+    if (counts) counts[i] = (long long)t.cnt[s];
+    if ((i & 7) == 0) {
+      uint8_t kb = 0, sb = 0;
+      for (int j = 0; j < 8 && i + j < num_groups; j++) {
+        uint32_t sj = order_slots[i + j];
+        bool knull = (sj == t.cap + 1);
+        if (!knull) kb |= (uint8_t)(1u << j);
+        if (t.sum_valid[sj]) sb |= (uint8_t)(1u << j);
+      }
+      if (key_validity) key_validity[i >> 3] = kb;
+      if (sum_validity) sum_validity[i >> 3] = sb;
+    }
+  }
+}
+
+__device__ __forceinline__ int varint_len_dev(uint64_t v) {
+  int k = 1;
+  while (v >= 128) {
+    v /= 128;
+    k++;
+  }
+  return k;
+}
+
+__global__ void k_agg_freeze_len(const AggTable t,
+                                 const uint32_t* __restrict__ order_slots,
+                                 int64_t num_groups, int32_t* __restrict__ lens) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < num_groups;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    uint32_t s = order_slots[i];
+    lens[i] = 1 + (t.sum_valid[s] ? 8 : 0) + varint_len_dev(t.cnt[s]);
+  }
+}
+
+__global__ void k_agg_freeze_write(const AggTable t,
+                                   const uint32_t* __restrict__ order_slots,
+                                   int64_t num_groups,
+                                   const int32_t* __restrict__ offsets,
+                                   uint8_t* __restrict__ out) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < num_groups;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    uint32_t s = order_slots[i];
+    uint8_t* p = out + offsets[i];
+    // acc.rs:335-347: [u8 valid][8B LE value]? then count.rs:193-203 varint
+    if (t.sum_valid[s]) {
+      *p++ = 1;
+      double v = t.sum[s];
+      memcpy(p, &v, 8);
+      p += 8;
+    } else {
+      *p++ = 0;
+    }
+    uint64_t c = t.cnt[s];
+    while (c >= 128) {
+      *p++ = (uint8_t)(128 + c % 128);
+      c /= 128;
+    }
+    *p++ = (uint8_t)c;
+  }
+}
+
+void launch_agg_update(const AggTable& t, const int64_t* keys,
+                       const uint8_t* key_valid, const double* vals,
+                       const uint8_t* val_valid, int64_t n, uint64_t row_offset,
+                       hipStream_t s) {
+  hipLaunchKernelGGL(k_agg_update, dim3(grid_for(n)), dim3(BLOCK), 0, s, t, keys,
+                     key_valid, vals, val_valid, n, row_offset);
+}
+void launch_agg_merge_frozen(const AggTable& t, const int64_t* keys,
+                             const uint8_t* key_valid, const uint8_t* acc_data,
+                             const int32_t* acc_offsets, int64_t n,
+                             uint64_t row_offset, hipStream_t s) {
+  hipLaunchKernelGGL(k_agg_merge_frozen, dim3(grid_for(n)), dim3(BLOCK), 0, s, t,
+                     keys, key_valid, acc_data, acc_offsets, n, row_offset);
+}
+void launch_agg_compact(const AggTable& t, uint32_t* out_slot,
+                        unsigned long long* out_first_row,
+                        unsigned long long* num_out, hipStream_t s) {
+  hipLaunchKernelGGL(k_agg_compact, dim3(grid_for(t.cap + 2)), dim3(BLOCK), 0, s,
+                     t, out_slot, out_first_row, num_out);
+}
+void launch_agg_gather_out(const AggTable& t, const uint32_t* order_slots,
+                           int64_t num_groups, int64_t* out_keys,
+                           uint8_t* out_key_validity, double* out_sums,
+                           uint8_t* out_sum_validity, long long* out_counts,
+                           hipStream_t s) {
+  hipLaunchKernelGGL(k_agg_gather_out, dim3(grid_for(num_groups)), dim3(BLOCK),
+                     0, s, t, order_slots, num_groups, out_keys,
+                     out_key_validity, out_sums, out_sum_validity, out_counts);
+}
+void launch_agg_freeze_len(const AggTable& t, const uint32_t* order_slots,
+                           int64_t num_groups, int32_t* lens, hipStream_t s) {
+  hipLaunchKernelGGL(k_agg_freeze_len, dim3(grid_for(num_groups)), dim3(BLOCK),
+                     0, s, t, order_slots, num_groups, lens);
+}
+void launch_agg_freeze_write(const AggTable& t, const uint32_t* order_slots,
+                             int64_t num_groups, const int32_t* offsets,
+                             uint8_t* out, hipStream_t s) {
+  hipLaunchKernelGGL(k_agg_freeze_write, dim3(grid_for(num_groups)), dim3(BLOCK),
+                     0, s, t, order_slots, num_groups, offsets, out);
+}
+
+// partial-skipping pass-through freeze (agg_ctx.rs:428-462
+// process_partial_skipped): each input row becomes one record; SUM acc =
+// value when valid, COUNT = valid ? 1 : 0.
+__global__ void k_skip_freeze_len(const uint8_t* __restrict__ val_valid,
+                                  int64_t n, int32_t* __restrict__ lens) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    bool v = !val_valid || bit_get_dev(val_valid, i);
+    lens[i] = v ? 10 : 2;  // [1][8B sum][varint 1] vs [0][varint 0]
+  }
+}
+
+__global__ void k_skip_freeze_write(const double* __restrict__ vals,
+                                    const uint8_t* __restrict__ val_valid,
+                                    int64_t n, const int32_t* __restrict__ offsets,
+                                    uint8_t* __restrict__ out) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    uint8_t* p = out + offsets[i];
+    bool v = !val_valid || bit_get_dev(val_valid, i);
+    if (v) {
+      p[0] = 1;
+      double x = vals[i];
+      memcpy(p + 1, &x, 8);
+      p[9] = 1;
+    } else {
+      p[0] = 0;
+      p[1] = 0;
+    }
+  }
+}
+
+void launch_skip_freeze_len(const uint8_t* val_valid, int64_t n, int32_t* lens,
+                            hipStream_t s) {
+  hipLaunchKernelGGL(k_skip_freeze_len, dim3(grid_for(n)), dim3(BLOCK), 0, s,
+                     val_valid, n, lens);
+}
+void launch_skip_freeze_write(const double* vals, const uint8_t* val_valid,
+                              int64_t n, const int32_t* offsets, uint8_t* out,
+                              hipStream_t s) {
+  hipLaunchKernelGGL(k_skip_freeze_write, dim3(grid_for(n)), dim3(BLOCK), 0, s,
+                     vals, val_valid, n, offsets, out);
+}
+
+// ---- shuffle partition -----------------------------------------------------
+__global__ void k_histogram(const uint32_t* __restrict__ part_ids, int64_t n,
+                            uint32_t P, uint32_t* __restrict__ counts) {
+  extern __shared__ uint32_t lds_counts[];
+  for (uint32_t p = threadIdx.x; p < P; p += blockDim.x) lds_counts[p] = 0;
+  __syncthreads();
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    atomicAdd(&lds_counts[part_ids[i]], 1u);
+  __syncthreads();
+  for (uint32_t p = threadIdx.x; p < P; p += blockDim.x)
+    if (lds_counts[p]) atomicAdd(&counts[p], lds_counts[p]);
+}
+
+void launch_histogram(const uint32_t* part_ids, int64_t n, uint32_t P,
+                      uint32_t* counts, hipStream_t s) {
+  size_t lds = P * sizeof(uint32_t);
+  hipLaunchKernelGGL(k_histogram, dim3(grid_for(n)), dim3(BLOCK), lds, s,
+                     part_ids, n, P, counts);
+}
+
+__global__ void k_fill_i64(int64_t* __restrict__ dst, int64_t value, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    dst[i] = value;
+}
+
+void launch_fill_i64(int64_t* dst, int64_t value, int64_t n, hipStream_t s) {
+  hipLaunchKernelGGL(k_fill_i64, dim3(grid_for(n)), dim3(BLOCK), 0, s, dst,
+                     value, n);
+}
+
+__global__ void k_iota_u32(uint32_t* __restrict__ dst, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    dst[i] = (uint32_t)i;
+}
+
+void launch_iota_u32(uint32_t* dst, int64_t n, hipStream_t s) {
+  hipLaunchKernelGGL(k_iota_u32, dim3(grid_for(n)), dim3(BLOCK), 0, s, dst, n);
+}
+
+__global__ void k_agg_rebuild(const AggTable dst, const AggTable src) {
+  int64_t total = src.cap + 2;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    bool used = (i < src.cap) ? (src.slot_key[i] != KEY_EMPTY)
+                              : (src.special_used[i - src.cap] != 0);
+    if (!used) continue;
+    int64_t a;
+    if (i >= src.cap) {
+      int which = (int)(i - src.cap);
+      if (atomicCAS(&dst.special_used[which], 0u, 1u) == 0u)
+        atomicAdd(dst.num_groups, 1ull);
+      a = dst.cap + which;
+    } else {
+      a = agg_upsert_slot(dst, src.slot_key[i], false);
+    }
+    // distinct keys per source slot: single writer, plain stores
+    dst.sum[a] = src.sum[i];
+    dst.cnt[a] = src.cnt[i];
+    dst.sum_valid[a] = src.sum_valid[i];
+    dst.first_row[a] = src.first_row[i];
+  }
+}
+
+void launch_agg_rebuild(const AggTable& dst, const AggTable& src, hipStream_t s) {
+  hipLaunchKernelGGL(k_agg_rebuild, dim3(grid_for(src.cap + 2)), dim3(BLOCK), 0,
+                     s, dst, src);
+}
+
+void sort_pairs_u64_u32(const unsigned long long* keys_in, const uint32_t* vals_in,
+                        unsigned long long* keys_out, uint32_t* vals_out,
+                        int64_t n, void* temp, size_t* temp_bytes, hipStream_t s) {
+  hipError_t e = rocprim::radix_sort_pairs(temp, *temp_bytes, keys_in, keys_out,
+                                           vals_in, vals_out, (size_t)n, 0, 64, s);
+  if (e != hipSuccess) abort();
+}
+
+void sort_pairs_u32_u32(const uint32_t* keys_in, const uint32_t* vals_in,
+                        uint32_t* keys_out, uint32_t* vals_out, int64_t n,
+                        int end_bit, void* temp, size_t* temp_bytes,
+                        hipStream_t s) {
+  hipError_t e = rocprim::radix_sort_pairs(temp, *temp_bytes, keys_in, keys_out,
+                                           vals_in, vals_out, (size_t)n, 0,
+                                           end_bit, s);
+  if (e != hipSuccess) abort();
+}
+
+__global__ void k_gather_8(const uint8_t* __restrict__ src,
+                           const uint32_t* __restrict__ perm, int64_t n,
+                           uint8_t* __restrict__ dst) {
+  const uint64_t* s64 = (const uint64_t*)src;
+  uint64_t* d64 = (uint64_t*)dst;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    d64[i] = s64[perm[i]];
+}
+
+__global__ void k_gather_bits(const uint8_t* __restrict__ src_bits,
+                              const uint32_t* __restrict__ perm, int64_t n,
+                              uint8_t* __restrict__ dst_bits) {
+  int64_t nbytes = (n + 7) / 8;
+  for (int64_t b = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; b < nbytes;
+       b += (int64_t)gridDim.x * blockDim.x) {
+    uint8_t out = 0;
+    for (int j = 0; j < 8; j++) {
+      int64_t i = b * 8 + j;
+      if (i < n && bit_get_dev(src_bits, perm[i])) out |= (uint8_t)(1u << j);
+    }
+    dst_bits[b] = out;
+  }
+}
+
+__global__ void k_gather_lens(const int32_t* __restrict__ src_offsets,
+                              const uint32_t* __restrict__ perm, int64_t n,
+                              int32_t* __restrict__ dst_lens) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    uint32_t p = perm[i];
+    dst_lens[i] = src_offsets[p + 1] - src_offsets[p];
+  }
+}
+
+__global__ void k_gather_bytes(const uint8_t* __restrict__ src_data,
+                               const int32_t* __restrict__ src_offsets,
+                               const uint32_t* __restrict__ perm,
+                               const int32_t* __restrict__ dst_offsets,
+                               int64_t n, uint8_t* __restrict__ dst_data) {
+  // one wave per row run; byte-copy with lane stride
+  for (int64_t i = (int64_t)blockIdx.x * (blockDim.x / 64) + (threadIdx.x / 64);
+       i < n; i += (int64_t)gridDim.x * (blockDim.x / 64)) {
+    int lane = threadIdx.x & 63;
+    uint32_t p = perm[i];
+    int32_t len = src_offsets[p + 1] - src_offsets[p];
+    const uint8_t* s = src_data + src_offsets[p];
+    uint8_t* d = dst_data + dst_offsets[i];
+    for (int32_t b = lane; b < len; b += 64) d[b] = s[b];
+  }
+}
+
+void launch_gather_8(const uint8_t* src, const uint32_t* perm, int64_t n,
+                     uint8_t* dst, hipStream_t s) {
+  hipLaunchKernelGGL(k_gather_8, dim3(grid_for(n)), dim3(BLOCK), 0, s, src, perm,
+                     n, dst);
+}
+void launch_gather_bits(const uint8_t* src_bits, const uint32_t* perm, int64_t n,
+                        uint8_t* dst_bits, hipStream_t s) {
+  hipLaunchKernelGGL(k_gather_bits, dim3(grid_for((n + 7) / 8)), dim3(BLOCK), 0,
+                     s, src_bits, perm, n, dst_bits);
+}
+void launch_gather_lens(const int32_t* src_offsets, const uint32_t* perm,
+                        int64_t n, int32_t* dst_lens, hipStream_t s) {
+  hipLaunchKernelGGL(k_gather_lens, dim3(grid_for(n)), dim3(BLOCK), 0, s,
+                     src_offsets, perm, n, dst_lens);
+}
+void launch_gather_bytes(const uint8_t* src_data, const int32_t* src_offsets,
+                         const uint32_t* perm, const int32_t* dst_offsets,
+                         int64_t n, uint8_t* dst_data, hipStream_t s) {
+  hipLaunchKernelGGL(k_gather_bytes, dim3(grid_for(n * 64)), dim3(BLOCK), 0, s,
+                     src_data, src_offsets, perm, dst_offsets, n, dst_data);
+}
+
+__global__ void k_byte_transpose(const uint8_t* __restrict__ in,
+                                 uint8_t* __restrict__ out, int64_t n, int w) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n * w;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t row = i / w;
+    int b = (int)(i % w);
+    out[(int64_t)b * n + row] = in[i];
+  }
+}
+
+void launch_byte_transpose(const uint8_t* in, uint8_t* out, int64_t n, int w,
+                           hipStream_t s) {
+  hipLaunchKernelGGL(k_byte_transpose, dim3(grid_for(n * w)), dim3(BLOCK), 0, s,
+                     in, out, n, w);
+}
+
+}  // namespace auron

@@ -1,0 +1,195 @@
+"""plan.py — builds serialized protobuf TaskDefinition plans.
+
+Pure-python proto3 wire-format encoder for the operator surface the engine
+accepts (auron-serde/proto/auron.proto; field numbers cited per message).
+There is no protoc in this image; the encoding is verified against the
+engine's independent C++ decoder (auron_debug_decode_plan) in tests.
+"""
+import struct
+
+# AggFunction enum (auron.proto:128-141)
+AGG_MIN, AGG_MAX, AGG_SUM, AGG_AVG, AGG_COUNT = 0, 1, 2, 3, 4
+# AggMode enum (auron.proto:692-696)
+MODE_PARTIAL, MODE_PARTIAL_MERGE, MODE_FINAL = 0, 1, 2
+
+# ArrowType oneof tags (auron.proto:860-896)
+DT_INT32, DT_INT64, DT_FLOAT64, DT_UTF8, DT_BINARY = 8, 10, 13, 14, 15
+
+
+def _varint(v):
+    out = bytearray()
+    while True:
+        b = v & 0x7F
+        v >>= 7
+        if v:
+            out.append(b | 0x80)
+        else:
+            out.append(b)
+            return bytes(out)
+
+
+def _tag(field, wire):
+    return _varint((field << 3) | wire)
+
+
+def _len_field(field, payload):
+    return _tag(field, 2) + _varint(len(payload)) + payload
+
+
+def _varint_field(field, v):
+    return _tag(field, 0) + _varint(v)
+
+
+def arrow_type(dt_tag):
+    return _len_field(dt_tag, b"")  # EmptyMessage payload
+
+
+def field(name, dt_tag, nullable):
+    # Field (auron.proto:750-757)
+    out = _len_field(1, name.encode())
+    out += _len_field(2, arrow_type(dt_tag))
+    if nullable:
+        out += _varint_field(3, 1)
+    return out
+
+
+def schema(fields):
+    # Schema (auron.proto:746-748)
+    return b"".join(_len_field(1, f) for f in fields)
+
+
+def column(name, index):
+    # PhysicalExprNode{column = 1} -> PhysicalColumn (auron.proto:489-492)
+    col = _len_field(1, name.encode()) + (_varint_field(2, index) if index else b"")
+    return _len_field(1, col)
+
+
+def agg_expr(fn, children, return_type_tag):
+    # PhysicalExprNode{agg_expr = 5} -> PhysicalAggExprNode (auron.proto:143-148)
+    node = _varint_field(1, fn) if fn else b""
+    for c in children:
+        node += _len_field(3, c)
+    node += _len_field(4, arrow_type(return_type_tag))
+    return _len_field(5, node)
+
+
+def ffi_reader(fields, resource_id, num_partitions=1):
+    # PhysicalPlanNode{ffi_reader = 18} (auron.proto:703-707)
+    node = _varint_field(1, num_partitions)
+    node += _len_field(2, schema(fields))
+    node += _len_field(3, resource_id.encode())
+    return _len_field(18, node)
+
+
+def agg(input_plan, grouping, aggs, modes, grouping_names, agg_names,
+        supports_partial_skipping=False, exec_mode=0):
+    # PhysicalPlanNode{agg = 16} -> AggExecNode (auron.proto:675-685)
+    node = _len_field(1, input_plan)
+    node += _varint_field(2, exec_mode)
+    for g in grouping:
+        node += _len_field(3, g)
+    for a in aggs:
+        node += _len_field(4, a)
+    node += _len_field(5, b"".join(_varint(m) for m in modes))  # packed enums
+    for n in grouping_names:
+        node += _len_field(6, n.encode())
+    for n in agg_names:
+        node += _len_field(7, n.encode())
+    if supports_partial_skipping:
+        node += _varint_field(9, 1)
+    return _len_field(16, node)
+
+
+def hash_repartition(hash_exprs, partition_count):
+    # PhysicalRepartition{hash_repartition = 2} (auron.proto:629-645)
+    inner = b"".join(_len_field(1, e) for e in hash_exprs)
+    inner += _varint_field(2, partition_count)
+    return _len_field(2, inner)
+
+
+def single_repartition():
+    return _len_field(1, _varint_field(1, 1))
+
+
+def shuffle_writer(input_plan, repartition, data_file, index_file):
+    # PhysicalPlanNode{shuffle_writer = 2} (auron.proto:524-529)
+    node = _len_field(1, input_plan)
+    node += _len_field(2, repartition)
+    node += _len_field(3, data_file.encode())
+    node += _len_field(4, index_file.encode())
+    return _len_field(2, node)
+
+
+def task_definition(plan, stage_id=0, partition_id=0, task_id=0):
+    # TaskDefinition (auron.proto:735-740) + PartitionId (:729-733)
+    pid = _varint_field(2, stage_id) + _varint_field(4, partition_id) + \
+        _varint_field(5, task_id)
+    return _len_field(1, pid) + _len_field(2, plan)
+
+
+# ---- canned plans for the north-star configs -------------------------------
+
+def northstar_input_fields(key_nullable=False):
+    return [field("key", DT_INT64, key_nullable), field("val", DT_FLOAT64, True)]
+
+
+def sum_count_aggs(val_index=1):
+    return [
+        agg_expr(AGG_SUM, [column("val", val_index)], DT_FLOAT64),
+        agg_expr(AGG_COUNT, [column("val", val_index)], DT_INT64),
+    ]
+
+
+def plan_partial_final(resource_id="input0", skipping=False):
+    """FFIReader -> Agg(Partial) -> Agg(Final): config 1/2 shape
+    (mirrors agg_exec.rs fuzztest:714-843 topology)."""
+    reader = ffi_reader(northstar_input_fields(), resource_id)
+    partial = agg(reader, [column("key", 0)], sum_count_aggs(1),
+                  [MODE_PARTIAL, MODE_PARTIAL], ["key"], ["sum", "cnt"],
+                  supports_partial_skipping=skipping)
+    final = agg(partial, [column("key", 0)], sum_count_aggs(1),
+                [MODE_FINAL, MODE_FINAL], ["key"], ["sum", "cnt"])
+    return task_definition(final)
+
+
+def plan_partial_only(resource_id="input0", skipping=False):
+    reader = ffi_reader(northstar_input_fields(), resource_id)
+    partial = agg(reader, [column("key", 0)], sum_count_aggs(1),
+                  [MODE_PARTIAL, MODE_PARTIAL], ["key"], ["sum", "cnt"],
+                  supports_partial_skipping=skipping)
+    return task_definition(partial)
+
+
+def plan_final_only(resource_id="input0"):
+    """FFIReader(partial output schema) -> Agg(Final): stage-2 shape."""
+    fields = [field("key", DT_INT64, True),
+              field("#9223372036854775807", DT_BINARY, False)]
+    reader = ffi_reader(fields, resource_id)
+    final = agg(reader, [column("key", 0)], sum_count_aggs(1),
+                [MODE_FINAL, MODE_FINAL], ["key"], ["sum", "cnt"])
+    return task_definition(final)
+
+
+def plan_agg_shuffle(data_file, index_file, num_partitions=200,
+                     resource_id="input0", partition_id=0, skipping=False):
+    """FFIReader -> Agg(Partial) -> ShuffleWriter(hash(key), P): config 4
+    stage-1 shape (NativeShuffleExchangeBase.scala:246-293)."""
+    reader = ffi_reader(northstar_input_fields(), resource_id)
+    partial = agg(reader, [column("key", 0)], sum_count_aggs(1),
+                  [MODE_PARTIAL, MODE_PARTIAL], ["key"], ["sum", "cnt"],
+                  supports_partial_skipping=skipping)
+    rep = hash_repartition([column("key", 0)], num_partitions)
+    sw = shuffle_writer(partial, rep, data_file, index_file)
+    return task_definition(sw, partition_id=partition_id)
+
+
+def plan_shuffle_only(data_file, index_file, num_partitions=200,
+                      resource_id="input0", fields=None, hash_col=("key", 0),
+                      partition_id=0):
+    if fields is None:
+        fields = [field("key", DT_INT64, True),
+                  field("#9223372036854775807", DT_BINARY, False)]
+    reader = ffi_reader(fields, resource_id)
+    rep = hash_repartition([column(*hash_col)], num_partitions)
+    sw = shuffle_writer(reader, rep, data_file, index_file)
+    return task_definition(sw, partition_id=partition_id)
