@@ -1,0 +1,240 @@
+"""GPU parity tests: the HIP engine (through the C ABI) vs the CPU oracle.
+
+Parity bar (BASELINE.json north_star): bit-exact for integer keys/counts and
+for f64 sums of integer-valued inputs (exact in f64 regardless of addition
+order while < 2^53); 1e-6 relative for fractional f64 sums. Record ORDER is
+also checked: the engine reproduces the reference's first-occurrence
+insertion order (agg_hash_map.rs:77-168).
+"""
+import os
+
+import numpy as np
+import pytest
+
+import blaze_amd
+from blaze_amd import plan
+from oracle import pywrap as oracle
+
+pytestmark = pytest.mark.gpu
+
+
+def gen_northstar(n, seed=42, nkeys=1_000_000, null_frac=0.001,
+                  fractional=False):
+    rng = np.random.default_rng(seed)
+    keys = rng.integers(0, nkeys, n).astype(np.int64)
+    vals = rng.integers(0, 1_000_000, n).astype(np.float64)
+    if fractional:
+        vals += rng.random(n)
+    val_valid = rng.random(n) >= null_frac
+    return keys, vals, val_valid
+
+
+def batches_of(keys, vals, val_valid, batch=10_000):
+    out = []
+    for i in range(0, len(keys), batch):
+        out.append([(keys[i:i + batch], None),
+                    (vals[i:i + batch], val_valid[i:i + batch])])
+    return out
+
+
+def run_oracle(keys, vals, val_valid):
+    a = oracle.Agg()
+    for i in range(0, len(keys), 10_000):
+        a.update(keys[i:i + 10_000], vals[i:i + 10_000],
+                 val_valid=val_valid[i:i + 10_000] if val_valid is not None else None)
+    return a
+
+
+def engine_final(task_outputs):
+    """Concatenate final-agg output batches into flat arrays."""
+    keys, sums, sum_valid, cnts = [], [], [], []
+    for ob in task_outputs:
+        keys.append(ob[0]["values"])
+        sums.append(ob[1]["values"])
+        v = ob[1]["valid"]
+        sum_valid.append(v if v is not None else np.ones(len(ob[1]["values"]), bool))
+        cnts.append(ob[2]["values"])
+    return (np.concatenate(keys), np.concatenate(sums),
+            np.concatenate(sum_valid), np.concatenate(cnts))
+
+
+def assert_agg_parity(outputs, orc, fractional):
+    k, s, sv, c = engine_final(outputs)
+    ref = orc.output()
+    assert len(k) == orc.num_groups
+    # ORDER parity: first-occurrence insertion order
+    np.testing.assert_array_equal(k, ref["keys"])
+    np.testing.assert_array_equal(c, ref["counts"])
+    np.testing.assert_array_equal(sv, ref["sum_valid"])
+    if fractional:
+        np.testing.assert_allclose(s[sv], ref["sums"][ref["sum_valid"]],
+                                   rtol=1e-6)
+    else:
+        np.testing.assert_array_equal(s[sv], ref["sums"][ref["sum_valid"]])
+
+
+@pytest.mark.parametrize("n,nkeys", [(100_000, 5000), (2_000_000, 1_000_000)])
+def test_partial_final_parity(n, nkeys):
+    keys, vals, vv = gen_northstar(n, nkeys=nkeys)
+    t = blaze_amd.Task(plan.plan_partial_final(),
+                       batches=batches_of(keys, vals, vv))
+    outputs = t.run()
+    assert_agg_parity(outputs, run_oracle(keys, vals, vv), fractional=False)
+    t.finalize()
+
+
+def test_partial_final_parity_10m_config2():
+    # BASELINE.json config 2: 10M rows, key in [0,1e6), val uniform [0,1e6)
+    # 0.1% nulls, seed 42
+    keys, vals, vv = gen_northstar(10_000_000)
+    t = blaze_amd.Task(plan.plan_partial_final(),
+                       batches=batches_of(keys, vals, vv, batch=100_000))
+    outputs = t.run()
+    assert_agg_parity(outputs, run_oracle(keys, vals, vv), fractional=False)
+    t.finalize()
+
+
+def test_fractional_sums_tolerance():
+    keys, vals, vv = gen_northstar(500_000, nkeys=1000, fractional=True)
+    t = blaze_amd.Task(plan.plan_partial_final(),
+                       batches=batches_of(keys, vals, vv))
+    assert_agg_parity(t.run(), run_oracle(keys, vals, vv), fractional=True)
+    t.finalize()
+
+
+def test_partial_binary_aggbuf_format():
+    """a8 wire format: the engine's Binary agg-buf column must be
+    byte-identical to the oracle's freeze output (integer-valued sums)."""
+    keys, vals, vv = gen_northstar(50_000, nkeys=300)
+    t = blaze_amd.Task(plan.plan_partial_only(),
+                       batches=batches_of(keys, vals, vv))
+    outputs = t.run()
+    orc = run_oracle(keys, vals, vv)
+    exp_data, exp_offsets = orc.freeze()
+    exp_keys = orc.output()["keys"]
+    got_keys = np.concatenate([ob[0]["values"] for ob in outputs])
+    got_data = np.concatenate([ob[1]["data"] for ob in outputs])
+    # rebase chunked offsets
+    parts = []
+    base = 0
+    for ob in outputs:
+        parts.append(ob[1]["offsets"][:-1] + base)
+        base += ob[1]["offsets"][-1]
+    got_offsets = np.concatenate(parts + [np.array([base])])
+    np.testing.assert_array_equal(got_keys, exp_keys)
+    np.testing.assert_array_equal(got_offsets, exp_offsets)
+    np.testing.assert_array_equal(got_data, exp_data)
+    t.finalize()
+
+
+def test_final_only_merge():
+    """Final-stage agg consuming frozen partial rows (stage-2 topology)."""
+    keys, vals, vv = gen_northstar(100_000, nkeys=777)
+    # two oracle partials = two 'map tasks'
+    final_in = []
+    merged = oracle.Agg()
+    for half in (slice(0, 50_000), slice(50_000, None)):
+        p = oracle.Agg()
+        p.update(keys[half], vals[half], val_valid=vv[half])
+        data, offs = p.freeze()
+        pk = p.output()["keys"]
+        final_in.append([(pk, None), ("binary", data, offs.astype(np.int32), None)])
+        merged.merge_frozen(pk, data, offs)
+    t = blaze_amd.Task(plan.plan_final_only(), batches=final_in)
+    assert_agg_parity(t.run(), merged, fractional=False)
+    t.finalize()
+
+
+def test_edge_cases():
+    # empty input
+    t = blaze_amd.Task(plan.plan_partial_final(), batches=[])
+    assert t.run() == []
+    t.finalize()
+    # single group, extreme keys, null-key group
+    keys = np.array([2**63 - 1, -2**63, 0, -1, 2**63 - 1, -2**63], np.int64)
+    vals = np.arange(6, dtype=np.float64)
+    kv = np.array([True, True, True, True, False, True])
+    t = blaze_amd.Task(
+        plan.plan_partial_final(),
+        batches=[[(keys, kv), (vals, None)]])
+    outputs = t.run()
+    k, s, sv, c = engine_final(outputs)
+    orc = oracle.Agg()
+    orc.update(keys, vals, key_valid=kv)
+    ref = orc.output()
+    np.testing.assert_array_equal(k[ref["key_valid"]], ref["keys"][ref["key_valid"]])
+    np.testing.assert_array_equal(c, ref["counts"])
+    np.testing.assert_array_equal(s, ref["sums"])
+    t.finalize()
+
+
+def test_partial_skipping_end_to_end():
+    """supports_partial_skipping: high-cardinality input flips the partial
+    stage to pass-through after 20k rows (agg_table.rs:109-120); the final
+    stage must still produce the exact aggregate."""
+    n = 60_000
+    keys = np.arange(n, dtype=np.int64)  # all distinct -> ratio 1.0
+    vals = np.ones(n)
+    vv = np.ones(n, bool)
+    t = blaze_amd.Task(plan.plan_partial_final(resource_id="input0"),
+                       batches=batches_of(keys, vals, vv))
+    outputs = t.run()
+    k, s, sv, c = engine_final(outputs)
+    assert len(k) == n
+    np.testing.assert_array_equal(np.sort(k), keys)
+    np.testing.assert_array_equal(c, np.ones(n, np.int64))
+    np.testing.assert_array_equal(s, np.ones(n))
+    t.finalize()
+
+
+def test_shuffle_write_files(tmp_path):
+    """Config-4 stage 1 on one GPU: partial agg + 200-way hash shuffle write.
+    The data/index files must byte-match the oracle-constructed expectation
+    (same murmur3 ids, same stable row order, same serde + lz4 framing)."""
+    P = 200
+    keys, vals, vv = gen_northstar(200_000, nkeys=3000)
+    data_file = str(tmp_path / "shuffle.data")
+    index_file = str(tmp_path / "shuffle.index")
+    t = blaze_amd.Task(
+        plan.plan_agg_shuffle(data_file, index_file, num_partitions=P),
+        batches=batches_of(keys, vals, vv))
+    out = t.run()
+    assert out == []  # shuffle writer emits no batches
+    t.finalize()
+
+    # oracle expectation
+    orc = run_oracle(keys, vals, vv)
+    g = orc.output()
+    fz_data, fz_offs = orc.freeze()
+    hashes = oracle.hash_cols([(g["keys"], None)])
+    pids = oracle.partition_ids(hashes, P)
+    order = np.argsort(pids, kind="stable")
+    exp_index = []
+    exp_bytes = bytearray()
+    pos = 0
+    for p in range(P):
+        exp_index.append(pos)
+        rows = order[pids[order] == p]
+        if len(rows) == 0:
+            continue
+        w = oracle.IpcWriter()
+        for beg in range(0, len(rows), 10_000):
+            chunk = rows[beg:beg + 10_000]
+            kcol = g["keys"][chunk]
+            lens = (fz_offs[chunk + 1] - fz_offs[chunk])
+            offs = np.concatenate([[0], np.cumsum(lens)]).astype(np.int64)
+            bb = np.concatenate([fz_data[fz_offs[r]:fz_offs[r + 1]]
+                                 for r in chunk]) if len(chunk) else np.array([], np.uint8)
+            payload = oracle.serde_batch(len(chunk), [
+                ("prim", kcol), ("bytes", bb, offs)])
+            w.write_payload(payload)
+        w.finish_block()
+        blob = w.bytes()
+        exp_bytes += blob
+        pos += len(blob)
+    exp_index.append(pos)
+
+    got_data = open(data_file, "rb").read()
+    got_index = np.frombuffer(open(index_file, "rb").read(), dtype="<u8")
+    np.testing.assert_array_equal(got_index, np.array(exp_index, np.uint64))
+    assert got_data == bytes(exp_bytes)
