@@ -147,7 +147,7 @@ class _HostBatch:
         n = None
         for col in cols:
             ch = ArrowArray()
-            if isinstance(col, tuple) and col and col[0] == "binary":
+            if len(col) > 0 and isinstance(col[0], str) and col[0] == "binary":
                 _, data, offsets, valid = col
                 data = np.ascontiguousarray(data, dtype=np.uint8)
                 offsets = np.ascontiguousarray(offsets, dtype=np.int32)

@@ -347,6 +347,13 @@ class AggOp {
     skip_min_rows_ = conf.get_i("PARTIAL_AGG_SKIPPING_MIN_ROWS", 20000);
     int64_t slots = conf.get_i("AURON_HIP_AGG_TABLE_SLOTS", 1 << 22);
     init_table(slots);
+    AURON_HIP(hipEventCreate(&ev_start_));
+    AURON_HIP(hipEventCreate(&ev_stop_));
+  }
+
+  ~AggOp() {
+    (void)hipEventDestroy(ev_start_);
+    (void)hipEventDestroy(ev_stop_);
   }
 
   void consume(DevBatch&& b) {
@@ -366,9 +373,18 @@ class AggOp {
     } else {
       const DevColumn& val = b.cols.at(val_col_);
       if (val.dt != DType::Float64) FAIL("SUM arg must be Float64");
+      // HIP-event timing on the launch stream (roofline evidence for the
+      // dominant kernel; read back via auron_get_metric)
+      AURON_HIP(hipEventRecord(ev_start_, stream_));
       launch_agg_update(t_, (const int64_t*)key.values, key.validity,
                         (const double*)val.values, val.validity, b.num_rows,
                         row_cursor_, stream_);
+      AURON_HIP(hipEventRecord(ev_stop_, stream_));
+      AURON_HIP(hipEventSynchronize(ev_stop_));
+      float ms = 0.f;
+      AURON_HIP(hipEventElapsedTime(&ms, ev_start_, ev_stop_));
+      update_ns_ += (int64_t)(ms * 1e6);
+      update_rows_ += b.num_rows;
     }
     row_cursor_ += (uint64_t)b.num_rows;
     held_.push_back(std::move(b));  // keep borrowed buffers alive
@@ -612,6 +628,8 @@ class AggOp {
   double skip_ratio_ = 0.999;
   int64_t skip_min_rows_ = 20000;
   uint64_t row_cursor_ = 0;
+  int64_t update_ns_ = 0, update_rows_ = 0;
+  hipEvent_t ev_start_ = nullptr, ev_stop_ = nullptr;
   AggTable t_;
   DevBuf d_slot_key_, d_special_, d_sum_, d_cnt_, d_sum_valid_, d_first_, d_ng_;
   std::vector<DevBatch> held_, skipped_;
@@ -971,8 +989,11 @@ struct Runtime {
     } else {
       FAIL("plan without Agg/ShuffleWriter unsupported");
     }
-    if (!aggs.empty())
+    if (!aggs.empty()) {
       metrics["num_groups"] = (int64_t)aggs.back()->num_groups_host();
+      metrics["agg_update_ns"] = aggs.front()->update_ns_;
+      metrics["agg_update_rows"] = aggs.front()->update_rows_;
+    }
     metrics["input_rows"] = input_rows;
     int64_t out_rows = 0;
     for (auto& o : outputs) out_rows += o.first;
