@@ -1,0 +1,284 @@
+#!/usr/bin/env python3
+"""bench.py — the north-star benchmark: rows/sec hash-agg + shuffle over
+1B-row int64-key batches (BASELINE.json), on 1..8 MI355X GPUs.
+
+A step = one pass of the hot path over one synthetic batch set already
+resident in HBM: partial hash-agg (GPU) -> murmur3 200-way partition ->
+RCCL all-to-all exchange of partial records (owner = partition % N) ->
+final merge agg (GPU). At N=1 the exchange is a local pass-through.
+
+Workload (config 4 of BASELINE.json, largest single-GPU config): 1B rows
+total, key:int64 uniform [0,1e6), val:float64 integer-valued uniform [0,1e6)
+with 0.1% nulls, seed 42, synthetic (no network). Total work is FIXED as N
+grows (scaling: strong).
+
+Run:  python bench.py [--gpus N] [--steps K] [--warmup W] [--rows R]
+N>1 is launched by the driver via torch.distributed.run (one rank per GPU,
+RCCL backend); rank/world read from the env.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+TOTAL_ROWS_DEFAULT = 1_000_000_000
+NUM_PARTITIONS = 200
+NUM_KEYS = 1_000_000
+NULL_FRAC = 0.001
+SEED = 42
+
+METRIC = "rows/sec hash-agg+shuffle, 1B-row int64-key Arrow batches, 1/2/4/8 GPU"
+
+
+def make_device_input(torch, rows, rank, device):
+    g = torch.Generator(device=device)
+    g.manual_seed(SEED + 1000 * rank)
+    keys = torch.randint(0, NUM_KEYS, (rows,), dtype=torch.int64, device=device,
+                         generator=g)
+    vals = torch.randint(0, 1_000_000, (rows,), dtype=torch.int64, device=device,
+                         generator=g).to(torch.float64)
+    nulls = torch.rand(rows, device=device, generator=g) < NULL_FRAC
+    valid = ~nulls
+    pad = (-rows) % 8
+    if pad:
+        valid = torch.cat([valid, torch.ones(pad, dtype=torch.bool,
+                                             device=device)])
+    weights = torch.tensor([1, 2, 4, 8, 16, 32, 64, 128], dtype=torch.uint8,
+                           device=device)
+    bitmap = (valid.view(-1, 8).to(torch.uint8) * weights).sum(
+        dim=1).to(torch.uint8).contiguous()
+    return keys, vals, bitmap, int(nulls.sum().item())
+
+
+def collect_partial(outputs):
+    """Flatten partial-agg output batches -> (keys, acc_lens, acc_data)."""
+    keys, lens, datas = [], [], []
+    for ob in outputs:
+        keys.append(ob[0]["values"])
+        off = ob[1]["offsets"]
+        lens.append((off[1:] - off[:-1]).astype(np.int32))
+        datas.append(ob[1]["data"])
+    if not keys:
+        return (np.empty(0, np.int64), np.empty(0, np.int32),
+                np.empty(0, np.uint8))
+    return np.concatenate(keys), np.concatenate(lens), np.concatenate(datas)
+
+
+def exchange(torch, dist, device, keys, lens, data, dest, world):
+    """RCCL all-to-all of (keys, acc lens, acc bytes) by destination rank."""
+    order = np.argsort(dest, kind="stable")
+    keys_s, lens_s = keys[order], lens[order]
+    row_splits = np.bincount(dest, minlength=world).astype(np.int64)
+    # bytes per row, permuted: build permuted acc data
+    offs = np.concatenate([[0], np.cumsum(lens)]).astype(np.int64)
+    data_s = np.concatenate([data[offs[r]:offs[r] + lens[r]] for r in order]) \
+        if len(order) else np.empty(0, np.uint8)
+    byte_splits = np.zeros(world, dtype=np.int64)
+    np.add.at(byte_splits, dest, lens.astype(np.int64))
+
+    t_rows = torch.tensor(row_splits, device=device)
+    t_bytes = torch.tensor(byte_splits, device=device)
+    r_rows = torch.empty_like(t_rows)
+    r_bytes = torch.empty_like(t_bytes)
+    dist.all_to_all_single(r_rows, t_rows)
+    dist.all_to_all_single(r_bytes, t_bytes)
+    in_rows, in_bytes = r_rows.cpu().numpy(), r_bytes.cpu().numpy()
+
+    def a2a(src_np, splits_out, splits_in, dtype):
+        src = torch.from_numpy(np.ascontiguousarray(src_np)).to(device)
+        dst = torch.empty(int(splits_in.sum()), dtype=src.dtype, device=device)
+        dist.all_to_all_single(dst, src, splits_in.tolist(),
+                               splits_out.tolist())
+        return dst.cpu().numpy()
+
+    rk = a2a(keys_s, row_splits, in_rows, np.int64)
+    rl = a2a(lens_s, row_splits, in_rows, np.int32)
+    rd = a2a(data_s, byte_splits, in_bytes, np.uint8)
+    return rk, rl, rd
+
+
+def run_step(ba, plan, dev_input, world, rank, torch, dist, device, stats):
+    # stage 1: partial hash-agg on the HBM-resident slice
+    t = ba.Task(plan.plan_partial_only(), device_batches=[dev_input])
+    outs = t.run()
+    stats["agg_update_ns"] += t.metric("agg_update_ns")
+    stats["agg_update_rows"] += t.metric("agg_update_rows")
+    stats["num_groups"] = t.metric("num_groups")
+    t.finalize()
+    keys, lens, data = collect_partial(outs)
+
+    # stage 2: 200-way murmur3 partition, owner rank = partition % N
+    pids = ba.partition_ids(keys, NUM_PARTITIONS)
+    if world > 1:
+        dest = (pids % world).astype(np.int64)
+        keys, lens, data = exchange(torch, dist, device, keys, lens, data,
+                                    dest, world)
+
+    # stage 3: final merge agg of (local + received) partial records
+    offs = np.concatenate([[0], np.cumsum(lens)]).astype(np.int32)
+    t2 = ba.Task(plan.plan_final_only(),
+                 batches=[[(keys, None), ("binary", data, offs, None)]])
+    out2 = t2.run()
+    nfinal = sum(ob[0]["values"].shape[0] for ob in out2)
+    t2.finalize()
+    return nfinal
+
+
+def cpu_baseline_leg(total_rows):
+    """Time the oracle (CPU restatement, single thread) on a bounded sample of
+    the same workload; reported beside the GPU number, never as `value`."""
+    from oracle import pywrap as oracle
+
+    rng = np.random.default_rng(SEED)
+    cal_n = 2_000_000
+    keys = rng.integers(0, NUM_KEYS, cal_n).astype(np.int64)
+    vals = rng.integers(0, 1_000_000, cal_n).astype(np.float64)
+    vv = rng.random(cal_n) >= NULL_FRAC
+
+    def one_pass(k, v, m):
+        a = oracle.Agg()
+        a.update(k, v, val_valid=m)
+        data, offsets = a.freeze()
+        g = a.output()
+        h = oracle.hash_cols([(g["keys"], None)])
+        oracle.partition_ids(h, NUM_PARTITIONS)
+        b = oracle.Agg()
+        b.merge_frozen(g["keys"], data, offsets)
+        return len(k)
+
+    t0 = time.perf_counter()
+    one_pass(keys, vals, vv)
+    cal_rate = cal_n / (time.perf_counter() - t0)
+    # size the sample for ~10 s of CPU work, capped at 100M rows
+    sample = int(min(1e8, max(cal_n, cal_rate * 10)))
+    keys = rng.integers(0, NUM_KEYS, sample).astype(np.int64)
+    vals = rng.integers(0, 1_000_000, sample).astype(np.float64)
+    vv = rng.random(sample) >= NULL_FRAC
+    t0 = time.perf_counter()
+    one_pass(keys, vals, vv)
+    dt = time.perf_counter() - t0
+    return {
+        "value": sample / dt,
+        "unit": "rows/s",
+        "cores": 1,
+        "kind": "port",
+        "sample": f"{sample} rows of the {total_rows}-row workload "
+                  f"(partial agg + freeze + partition ids + final merge), "
+                  f"single thread",
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=3)
+    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--rows", type=int, default=TOTAL_ROWS_DEFAULT,
+                    help="TOTAL rows across all ranks (strong scaling)")
+    ap.add_argument("--skip-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    import torch
+
+    import blaze_amd as ba
+    from blaze_amd import plan
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    if world == 1 and args.gpus > 1:
+        print("N>1 must be launched via torch.distributed.run", file=sys.stderr)
+        sys.exit(2)
+
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+
+        dist = dist_mod
+        dist.init_process_group("nccl")
+    torch.cuda.set_device(local_rank)
+    device = f"cuda:{local_rank}"
+
+    my_rows = args.rows // world
+    keys, vals, bitmap, null_count = make_device_input(torch, my_rows, rank,
+                                                       device)
+    dev_input = ba.DeviceBatch([
+        {"ptr": keys.data_ptr(), "len": my_rows},
+        {"ptr": vals.data_ptr(), "len": my_rows,
+         "validity_ptr": bitmap.data_ptr(), "null_count": null_count},
+    ], device_id=local_rank).as_input()
+
+    stats = {"agg_update_ns": 0, "agg_update_rows": 0, "num_groups": 0}
+    for _ in range(args.warmup):
+        run_step(ba, plan, dev_input, world, rank, torch, dist, device, stats)
+
+    stats = {"agg_update_ns": 0, "agg_update_rows": 0, "num_groups": 0}
+    if dist:
+        dist.barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        run_step(ba, plan, dev_input, world, rank, torch, dist, device, stats)
+    torch.cuda.synchronize()
+    if dist:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+    # max over ranks
+    if dist:
+        t = torch.tensor([elapsed], device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    if rank == 0:
+        value = args.rows * args.steps / elapsed
+        upd_ns = max(stats["agg_update_ns"], 1)
+        upd_rows = stats["agg_update_rows"]
+        achieved_gbs = upd_rows * 16.0 / upd_ns  # bytes/ns == GB/s
+        roofline = {
+            "bound": "hbm",
+            "achieved": round(achieved_gbs, 1),
+            "peak": 8000.0,
+            "unit": "GB/s",
+            "frac": round(achieved_gbs / 8000.0, 4),
+            "traffic": None,  # PMC traffic in profiles/ (rocprofv3), see DESIGN.md
+        }
+        cpu = None if args.skip_cpu_baseline or world > 1 else \
+            cpu_baseline_leg(args.rows)
+        out = {
+            "metric": METRIC,
+            "value": round(value, 1),
+            "unit": "rows/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 2),
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "f64",
+            "data": "synthetic",
+            "config": {
+                "workload": "hashagg_shuffle_1b",
+                "rows_total": args.rows,
+                "distinct_keys": NUM_KEYS,
+                "null_frac": NULL_FRAC,
+                "num_partitions": NUM_PARTITIONS,
+                "num_groups": stats["num_groups"],
+                "exchange": "rccl_all_to_all" if world > 1 else "local",
+            },
+            "roofline": roofline,
+            "cpu_baseline": cpu,
+        }
+        print(json.dumps(out))
+    if dist:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -194,6 +194,64 @@ class _HostBatch:
         self.top.release = None
 
 
+class DeviceBatch:
+    """ArrowDeviceArray over raw HBM pointers (e.g. torch cuda tensors) —
+    zero-copy input for the engine. cols: list of dicts
+    {"ptr": data_ptr, "len": n, "validity_ptr": p-or-None, "null_count": k}.
+    The caller must keep the backing tensors alive for the task lifetime."""
+
+    def __init__(self, cols, device_id=0):
+        self.keep = []
+        self.children = []
+        n = None
+        for col in cols:
+            ch = ArrowArray()
+            rows = col["len"]
+            n = rows if n is None else n
+            assert rows == n
+            bufs = (c.c_void_p * 2)(
+                c.c_void_p(col.get("validity_ptr") or None),
+                c.c_void_p(col["ptr"]))
+            ch.length = rows
+            ch.null_count = col.get("null_count", 0)
+            ch.offset = 0
+            ch.n_buffers = 2
+            ch.buffers = bufs
+            ch.n_children = 0
+            ch.release = None
+            self.keep.append(bufs)
+            self.children.append(ch)
+        self.child_ptrs = (c.POINTER(ArrowArray) * len(self.children))(
+            *[c.pointer(ch) for ch in self.children])
+        self.struct = ArrowDeviceArray()
+        top = self.struct.array
+        top.length = n or 0
+        top.n_children = len(self.children)
+        top.children = self.child_ptrs
+        top.n_buffers = 0
+        top.release = None
+        self.struct.device_id = device_id
+        self.struct.device_type = 10  # ARROW_DEVICE_ROCM
+
+    def as_input(self):
+        return {"struct": self.struct, "keep": self}
+
+
+def partition_ids(keys, num_partitions):
+    """murmur3(seed 42) + pmod on the GPU (engine kernels) for host i64 keys."""
+    keys = np.ascontiguousarray(keys, dtype=np.int64)
+    out = np.empty(len(keys), dtype=np.uint32)
+    L = lib()
+    L.auron_partition_ids.restype = c.c_int32
+    L.auron_partition_ids.argtypes = [c.c_void_p, c.c_int64, c.c_int32,
+                                      c.c_void_p]
+    rc = L.auron_partition_ids(keys.ctypes.data, len(keys), num_partitions,
+                               out.ctypes.data)
+    if rc != 0:
+        raise RuntimeError("auron_partition_ids failed (GPU required)")
+    return out
+
+
 def _read_bitmap(ptr, n):
     if not ptr:
         return None

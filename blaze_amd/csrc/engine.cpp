@@ -1138,6 +1138,29 @@ void auron_on_exit(void) {
 
 const char* auron_version(void) { return "auron-hip 0.1 gfx950"; }
 
+// Compute murmur3(seed 42) + pmod partition ids for host-resident i64 keys on
+// the GPU (shuffle/mod.rs:163-188). Used by bench.py's RCCL exchange leg to
+// route partial-agg records to their owner rank. Returns 0 on success.
+int32_t auron_partition_ids(const int64_t* keys, int64_t n, int32_t P,
+                            uint32_t* out) {
+  try {
+    hipStream_t s;
+    AURON_HIP(hipStreamCreate(&s));
+    DevBuf d_keys(n * 8), d_hash(n * 4), d_out(n * 4);
+    AURON_HIP(hipMemcpyAsync(d_keys.get(), keys, n * 8, hipMemcpyHostToDevice, s));
+    launch_hash_init(d_hash.get<int32_t>(), 42, n, s);
+    launch_hash_fold_i64(d_keys.get<int64_t>(), nullptr, n,
+                         d_hash.get<int32_t>(), s);
+    launch_pmod(d_hash.get<int32_t>(), n, P, d_out.get<uint32_t>(), s);
+    AURON_HIP(hipMemcpyAsync(out, d_out.get(), n * 4, hipMemcpyDeviceToHost, s));
+    AURON_HIP(hipStreamSynchronize(s));
+    AURON_HIP(hipStreamDestroy(s));
+    return 0;
+  } catch (const std::exception&) {
+    return -1;
+  }
+}
+
 // test-only introspection: decode a TaskDefinition and render a one-line
 // summary (verifies the hand-rolled proto reader against encoders)
 int32_t auron_debug_decode_plan(const uint8_t* data, size_t len, char* out,
