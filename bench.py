@@ -104,8 +104,11 @@ def exchange(torch, dist, device, keys, lens, data, dest, world):
 
 
 def run_step(ba, plan, dev_input, world, rank, torch, dist, device, stats):
-    # stage 1: partial hash-agg on the HBM-resident slice
-    t = ba.Task(plan.plan_partial_only(), device_batches=[dev_input])
+    # stage 1: partial hash-agg on the HBM-resident slice.
+    # BATCH_SIZE is the reference's own conf knob (conf.rs:32); raised here so
+    # the ~1M-group output is emitted in few chunks instead of 100 × 10k.
+    conf = {"BATCH_SIZE": 1 << 20}
+    t = ba.Task(plan.plan_partial_only(), device_batches=[dev_input], conf=conf)
     outs = t.run()
     stats["agg_update_ns"] += t.metric("agg_update_ns")
     stats["agg_update_rows"] += t.metric("agg_update_rows")
@@ -122,7 +125,7 @@ def run_step(ba, plan, dev_input, world, rank, torch, dist, device, stats):
 
     # stage 3: final merge agg of (local + received) partial records
     offs = np.concatenate([[0], np.cumsum(lens)]).astype(np.int32)
-    t2 = ba.Task(plan.plan_final_only(),
+    t2 = ba.Task(plan.plan_final_only(), conf={"BATCH_SIZE": 1 << 20},
                  batches=[[(keys, None), ("binary", data, offs, None)]])
     out2 = t2.run()
     nfinal = sum(ob[0]["values"].shape[0] for ob in out2)
