@@ -103,13 +103,24 @@ def exchange(torch, dist, device, keys, lens, data, dest, world):
     return rk, rl, rd
 
 
+VERBOSE = os.environ.get("BENCH_VERBOSE", "0") == "1"
+
+
+def _log(msg):
+    if VERBOSE:
+        print(f"[bench {time.strftime('%H:%M:%S')}] {msg}", file=sys.stderr,
+              flush=True)
+
+
 def run_step(ba, plan, dev_input, world, rank, torch, dist, device, stats):
     # stage 1: partial hash-agg on the HBM-resident slice.
     # BATCH_SIZE is the reference's own conf knob (conf.rs:32); raised here so
     # the ~1M-group output is emitted in few chunks instead of 100 × 10k.
     conf = {"BATCH_SIZE": 1 << 20}
+    _log("stage1: partial agg start")
     t = ba.Task(plan.plan_partial_only(), device_batches=[dev_input], conf=conf)
     outs = t.run()
+    _log("stage1 done")
     stats["agg_update_ns"] += t.metric("agg_update_ns")
     stats["agg_update_rows"] += t.metric("agg_update_rows")
     stats["num_groups"] = t.metric("num_groups")
@@ -117,6 +128,7 @@ def run_step(ba, plan, dev_input, world, rank, torch, dist, device, stats):
     keys, lens, data = collect_partial(outs)
 
     # stage 2: 200-way murmur3 partition, owner rank = partition % N
+    _log("stage2: partition ids")
     pids = ba.partition_ids(keys, NUM_PARTITIONS)
     if world > 1:
         dest = (pids % world).astype(np.int64)
@@ -127,9 +139,11 @@ def run_step(ba, plan, dev_input, world, rank, torch, dist, device, stats):
     offs = np.concatenate([[0], np.cumsum(lens)]).astype(np.int32)
     t2 = ba.Task(plan.plan_final_only(), conf={"BATCH_SIZE": 1 << 20},
                  batches=[[(keys, None), ("binary", data, offs, None)]])
+    _log("stage3: final merge")
     out2 = t2.run()
     nfinal = sum(ob[0]["values"].shape[0] for ob in out2)
     t2.finalize()
+    _log(f"step done: {nfinal} groups")
     return nfinal
 
 
@@ -209,8 +223,11 @@ def main():
     device = f"cuda:{local_rank}"
 
     my_rows = args.rows // world
+    _log(f"generating {my_rows} rows on {device}")
     keys, vals, bitmap, null_count = make_device_input(torch, my_rows, rank,
                                                        device)
+    torch.cuda.synchronize()
+    _log("input ready")
     dev_input = ba.DeviceBatch([
         {"ptr": keys.data_ptr(), "len": my_rows},
         {"ptr": vals.data_ptr(), "len": my_rows,
