@@ -443,9 +443,13 @@ class AggOp {
 
   uint64_t num_groups_host() {
     uint64_t ng = 0;
+    uint32_t err = 0;
     AURON_HIP(hipMemcpyAsync(&ng, t_.num_groups, 8, hipMemcpyDeviceToHost,
                              stream_));
+    AURON_HIP(hipMemcpyAsync(&err, t_.error_flag, 4, hipMemcpyDeviceToHost,
+                             stream_));
     AURON_HIP(hipStreamSynchronize(stream_));
+    if (err) FAIL("agg hash table probe exhausted (table full/corrupt)");
     return ng;
   }
 
@@ -463,6 +467,7 @@ class AggOp {
     d_sum_valid_.alloc((cap + 2) * 4);
     d_first_.alloc((cap + 2) * 8);
     d_ng_.alloc(8);
+    d_err_.alloc(4);
     t_.slot_key = d_slot_key_.get<int64_t>();
     t_.special_used = d_special_.get<uint32_t>();
     t_.sum = d_sum_.get<double>();
@@ -470,6 +475,8 @@ class AggOp {
     t_.sum_valid = d_sum_valid_.get<uint32_t>();
     t_.first_row = d_first_.get<unsigned long long>();
     t_.num_groups = d_ng_.get<unsigned long long>();
+    t_.error_flag = d_err_.get<uint32_t>();
+    AURON_HIP(hipMemsetAsync(d_err_.get(), 0, 4, stream_));
     launch_fill_i64(t_.slot_key, INT64_MIN, cap, stream_);
     AURON_HIP(hipMemsetAsync(d_special_.get(), 0, 2 * 4, stream_));
     AURON_HIP(hipMemsetAsync(d_sum_.get(), 0, (cap + 2) * 8, stream_));
@@ -496,7 +503,7 @@ class AggOp {
     DevBuf ok = std::move(d_slot_key_), os = std::move(d_special_),
            osum = std::move(d_sum_), ocnt = std::move(d_cnt_),
            ov = std::move(d_sum_valid_), of = std::move(d_first_),
-           ong = std::move(d_ng_);
+           ong = std::move(d_ng_), oerr = std::move(d_err_);
     init_table(new_cap);
     launch_agg_rebuild(t_, old, stream_);
     AURON_HIP(hipStreamSynchronize(stream_));
@@ -631,7 +638,8 @@ class AggOp {
   int64_t update_ns_ = 0, update_rows_ = 0;
   hipEvent_t ev_start_ = nullptr, ev_stop_ = nullptr;
   AggTable t_;
-  DevBuf d_slot_key_, d_special_, d_sum_, d_cnt_, d_sum_valid_, d_first_, d_ng_;
+  DevBuf d_slot_key_, d_special_, d_sum_, d_cnt_, d_sum_valid_, d_first_, d_ng_,
+      d_err_;
   std::vector<DevBatch> held_, skipped_;
 };
 

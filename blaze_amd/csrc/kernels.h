@@ -33,6 +33,7 @@ struct AggTable {
   uint32_t* sum_valid = nullptr;
   unsigned long long* first_row = nullptr;  // global arrival index, for order
   unsigned long long* num_groups = nullptr; // [1] device counter
+  uint32_t* error_flag = nullptr;           // [1] raised on probe exhaustion
 };
 
 // grid-stride row update: keys/vals length n, rows globally numbered starting

@@ -19,9 +19,21 @@
 
 #include <rocprim/device/device_radix_sort.hpp>
 
+#include <stdexcept>
+#include <string>
+
 #include "kernels.h"
 
 namespace auron {
+
+// raise launch-configuration errors loudly (hipLaunchKernelGGL itself
+// reports nothing)
+static inline void check_launch(const char* name) {
+  hipError_t e = hipGetLastError();
+  if (e != hipSuccess)
+    throw std::runtime_error(std::string("kernel launch failed: ") + name +
+                             ": " + hipGetErrorString(e));
+}
 
 static constexpr int BLOCK = 256;
 static constexpr int64_t MAX_BLOCKS = 256 * 8;
@@ -103,16 +115,19 @@ __global__ void k_pmod(const int32_t* __restrict__ hashes, int64_t n, int32_t P,
 void launch_hash_init(int32_t* hashes, int32_t seed, int64_t n, hipStream_t s) {
   hipLaunchKernelGGL(k_hash_init, dim3(grid_for(n)), dim3(BLOCK), 0, s, hashes,
                      seed, n);
+  check_launch("k_hash_init");
 }
 void launch_hash_fold_i64(const int64_t* vals, const uint8_t* valid, int64_t n,
                           int32_t* hashes, hipStream_t s) {
   hipLaunchKernelGGL(k_hash_fold_i64, dim3(grid_for(n)), dim3(BLOCK), 0, s,
                      vals, valid, n, hashes);
+  check_launch("k_hash_fold_i64");
 }
 void launch_pmod(const int32_t* hashes, int64_t n, int32_t P, uint32_t* part_ids,
                  hipStream_t s) {
   hipLaunchKernelGGL(k_pmod, dim3(grid_for(n)), dim3(BLOCK), 0, s, hashes, n, P,
                      part_ids);
+  check_launch("k_pmod");
 }
 
 // ---- hash aggregation ------------------------------------------------------
@@ -129,7 +144,10 @@ __device__ __forceinline__ uint64_t mix64(uint64_t x) {
 }
 
 // probe-or-insert; returns acc index (slot, or cap/cap+1 for the special
-// groups). Claims bump *num_groups.
+// groups), or -1 when the probe exhausts the table (table full / corrupt) —
+// the caller skips the row and the host fails the task loudly via the error
+// flag. The probe is a BOUNDED loop: an unbounded no-side-effect loop
+// iteration is UB under C++ forward-progress rules and may be miscompiled.
 __device__ __forceinline__ int64_t agg_upsert_slot(const AggTable t, int64_t key,
                                                    bool key_null) {
   if (key_null || key == KEY_EMPTY) {
@@ -141,7 +159,7 @@ __device__ __forceinline__ int64_t agg_upsert_slot(const AggTable t, int64_t key
   uint64_t h = mix64((uint64_t)key);
   int64_t mask = t.cap - 1;
   int64_t i = (int64_t)(h & (uint64_t)mask);
-  while (true) {
+  for (int64_t probe = 0; probe <= mask; probe++) {
     int64_t cur = t.slot_key[i];
     if (cur == key) return i;
     if (cur == KEY_EMPTY) {
@@ -156,6 +174,8 @@ __device__ __forceinline__ int64_t agg_upsert_slot(const AggTable t, int64_t key
     }
     i = (i + 1) & mask;
   }
+  atomicOr(t.error_flag, 1u);
+  return -1;
 }
 
 __global__ void k_agg_update(const AggTable t, const int64_t* __restrict__ keys,
@@ -167,6 +187,7 @@ __global__ void k_agg_update(const AggTable t, const int64_t* __restrict__ keys,
        i += (int64_t)gridDim.x * blockDim.x) {
     bool knull = key_valid && !bit_get_dev(key_valid, i);
     int64_t a = agg_upsert_slot(t, knull ? 0 : keys[i], knull);
+    if (a < 0) continue;  // table full: error_flag raised, host aborts
     atomicMin(&t.first_row[a], row_offset + (uint64_t)i);
     bool vvalid = !val_valid || bit_get_dev(val_valid, i);
     if (vvalid) {
@@ -205,6 +226,7 @@ __global__ void k_agg_merge_frozen(const AggTable t,
        i += (int64_t)gridDim.x * blockDim.x) {
     bool knull = key_valid && !bit_get_dev(key_valid, i);
     int64_t a = agg_upsert_slot(t, knull ? 0 : keys[i], knull);
+    if (a < 0) continue;
     atomicMin(&t.first_row[a], row_offset + (uint64_t)i);
     const uint8_t* p = acc_data + acc_offsets[i];
     uint8_t valid = *p++;
@@ -318,6 +340,7 @@ void launch_agg_update(const AggTable& t, const int64_t* keys,
                        hipStream_t s) {
   hipLaunchKernelGGL(k_agg_update, dim3(grid_for(n)), dim3(BLOCK), 0, s, t, keys,
                      key_valid, vals, val_valid, n, row_offset);
+  check_launch("k_agg_update");
 }
 void launch_agg_merge_frozen(const AggTable& t, const int64_t* keys,
                              const uint8_t* key_valid, const uint8_t* acc_data,
@@ -325,12 +348,14 @@ void launch_agg_merge_frozen(const AggTable& t, const int64_t* keys,
                              uint64_t row_offset, hipStream_t s) {
   hipLaunchKernelGGL(k_agg_merge_frozen, dim3(grid_for(n)), dim3(BLOCK), 0, s, t,
                      keys, key_valid, acc_data, acc_offsets, n, row_offset);
+  check_launch("k_agg_merge_frozen");
 }
 void launch_agg_compact(const AggTable& t, uint32_t* out_slot,
                         unsigned long long* out_first_row,
                         unsigned long long* num_out, hipStream_t s) {
   hipLaunchKernelGGL(k_agg_compact, dim3(grid_for(t.cap + 2)), dim3(BLOCK), 0, s,
                      t, out_slot, out_first_row, num_out);
+  check_launch("k_agg_compact");
 }
 void launch_agg_gather_out(const AggTable& t, const uint32_t* order_slots,
                            int64_t num_groups, int64_t* out_keys,
@@ -340,17 +365,20 @@ void launch_agg_gather_out(const AggTable& t, const uint32_t* order_slots,
   hipLaunchKernelGGL(k_agg_gather_out, dim3(grid_for(num_groups)), dim3(BLOCK),
                      0, s, t, order_slots, num_groups, out_keys,
                      out_key_validity, out_sums, out_sum_validity, out_counts);
+  check_launch("k_agg_gather_out");
 }
 void launch_agg_freeze_len(const AggTable& t, const uint32_t* order_slots,
                            int64_t num_groups, int32_t* lens, hipStream_t s) {
   hipLaunchKernelGGL(k_agg_freeze_len, dim3(grid_for(num_groups)), dim3(BLOCK),
                      0, s, t, order_slots, num_groups, lens);
+  check_launch("k_agg_freeze_len");
 }
 void launch_agg_freeze_write(const AggTable& t, const uint32_t* order_slots,
                              int64_t num_groups, const int32_t* offsets,
                              uint8_t* out, hipStream_t s) {
   hipLaunchKernelGGL(k_agg_freeze_write, dim3(grid_for(num_groups)), dim3(BLOCK),
                      0, s, t, order_slots, num_groups, offsets, out);
+  check_launch("k_agg_freeze_write");
 }
 
 // partial-skipping pass-through freeze (agg_ctx.rs:428-462
@@ -389,12 +417,14 @@ void launch_skip_freeze_len(const uint8_t* val_valid, int64_t n, int32_t* lens,
                             hipStream_t s) {
   hipLaunchKernelGGL(k_skip_freeze_len, dim3(grid_for(n)), dim3(BLOCK), 0, s,
                      val_valid, n, lens);
+  check_launch("k_skip_freeze_len");
 }
 void launch_skip_freeze_write(const double* vals, const uint8_t* val_valid,
                               int64_t n, const int32_t* offsets, uint8_t* out,
                               hipStream_t s) {
   hipLaunchKernelGGL(k_skip_freeze_write, dim3(grid_for(n)), dim3(BLOCK), 0, s,
                      vals, val_valid, n, offsets, out);
+  check_launch("k_skip_freeze_write");
 }
 
 // ---- shuffle partition -----------------------------------------------------
@@ -416,6 +446,7 @@ void launch_histogram(const uint32_t* part_ids, int64_t n, uint32_t P,
   size_t lds = P * sizeof(uint32_t);
   hipLaunchKernelGGL(k_histogram, dim3(grid_for(n)), dim3(BLOCK), lds, s,
                      part_ids, n, P, counts);
+  check_launch("k_histogram");
 }
 
 __global__ void k_fill_i64(int64_t* __restrict__ dst, int64_t value, int64_t n) {
@@ -427,6 +458,7 @@ __global__ void k_fill_i64(int64_t* __restrict__ dst, int64_t value, int64_t n) 
 void launch_fill_i64(int64_t* dst, int64_t value, int64_t n, hipStream_t s) {
   hipLaunchKernelGGL(k_fill_i64, dim3(grid_for(n)), dim3(BLOCK), 0, s, dst,
                      value, n);
+  check_launch("k_fill_i64");
 }
 
 __global__ void k_iota_u32(uint32_t* __restrict__ dst, int64_t n) {
@@ -437,6 +469,7 @@ __global__ void k_iota_u32(uint32_t* __restrict__ dst, int64_t n) {
 
 void launch_iota_u32(uint32_t* dst, int64_t n, hipStream_t s) {
   hipLaunchKernelGGL(k_iota_u32, dim3(grid_for(n)), dim3(BLOCK), 0, s, dst, n);
+  check_launch("k_iota_u32");
 }
 
 __global__ void k_agg_rebuild(const AggTable dst, const AggTable src) {
@@ -454,6 +487,7 @@ __global__ void k_agg_rebuild(const AggTable dst, const AggTable src) {
       a = dst.cap + which;
     } else {
       a = agg_upsert_slot(dst, src.slot_key[i], false);
+      if (a < 0) continue;
     }
     // distinct keys per source slot: single writer, plain stores
     dst.sum[a] = src.sum[i];
@@ -466,6 +500,7 @@ __global__ void k_agg_rebuild(const AggTable dst, const AggTable src) {
 void launch_agg_rebuild(const AggTable& dst, const AggTable& src, hipStream_t s) {
   hipLaunchKernelGGL(k_agg_rebuild, dim3(grid_for(src.cap + 2)), dim3(BLOCK), 0,
                      s, dst, src);
+  check_launch("k_agg_rebuild");
 }
 
 void sort_pairs_u64_u32(const unsigned long long* keys_in, const uint32_t* vals_in,
@@ -542,22 +577,26 @@ void launch_gather_8(const uint8_t* src, const uint32_t* perm, int64_t n,
                      uint8_t* dst, hipStream_t s) {
   hipLaunchKernelGGL(k_gather_8, dim3(grid_for(n)), dim3(BLOCK), 0, s, src, perm,
                      n, dst);
+  check_launch("k_gather_8");
 }
 void launch_gather_bits(const uint8_t* src_bits, const uint32_t* perm, int64_t n,
                         uint8_t* dst_bits, hipStream_t s) {
   hipLaunchKernelGGL(k_gather_bits, dim3(grid_for((n + 7) / 8)), dim3(BLOCK), 0,
                      s, src_bits, perm, n, dst_bits);
+  check_launch("k_gather_bits");
 }
 void launch_gather_lens(const int32_t* src_offsets, const uint32_t* perm,
                         int64_t n, int32_t* dst_lens, hipStream_t s) {
   hipLaunchKernelGGL(k_gather_lens, dim3(grid_for(n)), dim3(BLOCK), 0, s,
                      src_offsets, perm, n, dst_lens);
+  check_launch("k_gather_lens");
 }
 void launch_gather_bytes(const uint8_t* src_data, const int32_t* src_offsets,
                          const uint32_t* perm, const int32_t* dst_offsets,
                          int64_t n, uint8_t* dst_data, hipStream_t s) {
   hipLaunchKernelGGL(k_gather_bytes, dim3(grid_for(n * 64)), dim3(BLOCK), 0, s,
                      src_data, src_offsets, perm, dst_offsets, n, dst_data);
+  check_launch("k_gather_bytes");
 }
 
 __global__ void k_byte_transpose(const uint8_t* __restrict__ in,
@@ -574,6 +613,7 @@ void launch_byte_transpose(const uint8_t* in, uint8_t* out, int64_t n, int w,
                            hipStream_t s) {
   hipLaunchKernelGGL(k_byte_transpose, dim3(grid_for(n * w)), dim3(BLOCK), 0, s,
                      in, out, n, w);
+  check_launch("k_byte_transpose");
 }
 
 }  // namespace auron
