@@ -364,31 +364,82 @@ class AggOp {
       skipped_.push_back(std::move(b));
       return;
     }
-    if (merge_mode_) {
-      const DevColumn& buf = b.cols.at(1);
-      if (buf.dt != DType::Binary) FAIL("agg-buf column must be Binary");
-      launch_agg_merge_frozen(t_, (const int64_t*)key.values, key.validity,
-                              (const uint8_t*)buf.values, buf.offsets,
-                              b.num_rows, row_cursor_, stream_);
-    } else {
-      const DevColumn& val = b.cols.at(val_col_);
-      if (val.dt != DType::Float64) FAIL("SUM arg must be Float64");
-      // HIP-event timing on the launch stream (roofline evidence for the
-      // dominant kernel; read back via auron_get_metric)
-      AURON_HIP(hipEventRecord(ev_start_, stream_));
-      launch_agg_update(t_, (const int64_t*)key.values, key.validity,
-                        (const double*)val.values, val.validity, b.num_rows,
-                        row_cursor_, stream_);
-      AURON_HIP(hipEventRecord(ev_stop_, stream_));
-      AURON_HIP(hipEventSynchronize(ev_stop_));
-      float ms = 0.f;
-      AURON_HIP(hipEventElapsedTime(&ms, ev_start_, ev_stop_));
-      update_ns_ += (int64_t)(ms * 1e6);
-      update_rows_ += b.num_rows;
+    // Process in chunks with a HARD no-overflow guarantee: before each chunk,
+    // capacity is ensured for the conservative bound ng_bound_ (last readback
+    // + every row since being a new key), so a probe can never exhaust the
+    // table mid-launch. Readbacks (stream syncs) happen only when the bound
+    // nears the 3/4 load threshold — rare once cardinality saturates.
+    auto slice_valid = [](const uint8_t* v, int64_t done) {
+      return v ? v + done / 8 : nullptr;  // done is a multiple of 8
+    };
+    int64_t done = 0;
+    while (done < b.num_rows) {
+      if (ng_bound_ * 4 >= (uint64_t)t_.cap * 3) refresh_ng();
+      int64_t free_slots = t_.cap * 3 / 4 - (int64_t)ng_bound_;
+      if (free_slots < (1 << 16)) {
+        ensure_capacity((int64_t)ng_bound_ + (1 << 20));
+        free_slots = t_.cap * 3 / 4 - (int64_t)ng_bound_;
+      }
+      int64_t chunk = std::min(b.num_rows - done, free_slots);
+      if (done + chunk < b.num_rows) chunk &= ~(int64_t)7;  // bitmap-sliceable
+      if (merge_mode_) {
+        const DevColumn& buf = b.cols.at(1);
+        if (buf.dt != DType::Binary) FAIL("agg-buf column must be Binary");
+        // acc_offsets are absolute into buf.values, so only the index shifts
+        launch_agg_merge_frozen(t_, (const int64_t*)key.values + done,
+                                slice_valid(key.validity, done),
+                                (const uint8_t*)buf.values, buf.offsets + done,
+                                chunk, row_cursor_, stream_);
+      } else {
+        const DevColumn& val = b.cols.at(val_col_);
+        if (val.dt != DType::Float64) FAIL("SUM arg must be Float64");
+        // HIP-event timing on the launch stream (roofline evidence for the
+        // dominant kernel; read back via auron_get_metric)
+        AURON_HIP(hipEventRecord(ev_start_, stream_));
+        launch_agg_update(t_, (const int64_t*)key.values + done,
+                          slice_valid(key.validity, done),
+                          (const double*)val.values + done,
+                          slice_valid(val.validity, done), chunk, row_cursor_,
+                          stream_);
+        AURON_HIP(hipEventRecord(ev_stop_, stream_));
+        AURON_HIP(hipEventSynchronize(ev_stop_));
+        float ms = 0.f;
+        AURON_HIP(hipEventElapsedTime(&ms, ev_start_, ev_stop_));
+        update_ns_ += (int64_t)(ms * 1e6);
+        update_rows_ += chunk;
+      }
+      done += chunk;
+      row_cursor_ += (uint64_t)chunk;
+      ng_bound_ += (uint64_t)chunk;
+      // partial skipping (agg_table.rs:109-120): needs the true cardinality
+      if (skip_enabled_ && !skipping_ &&
+          row_cursor_ >= (uint64_t)skip_min_rows_) {
+        refresh_ng();
+        if ((double)ng_true_ / (double)row_cursor_ >= skip_ratio_) {
+          skipping_ = true;
+          if (done < b.num_rows) {
+            // the un-aggregated tail of this batch passes through, like the
+            // reference flipping between batches
+            DevBatch tail;
+            tail.num_rows = b.num_rows - done;
+            for (const DevColumn& src : b.cols) {
+              DevColumn sc;
+              sc.dt = src.dt;
+              sc.len = tail.num_rows;
+              if (src.dt == DType::Binary || src.dt == DType::Utf8)
+                FAIL("binary columns unsupported in skip-tail slice");
+              sc.values =
+                  (const uint8_t*)src.values + done * dtype_width(src.dt);
+              sc.validity = slice_valid(src.validity, done);
+              tail.cols.push_back(std::move(sc));
+            }
+            skipped_.push_back(std::move(tail));
+          }
+          break;
+        }
+      }
     }
-    row_cursor_ += (uint64_t)b.num_rows;
     held_.push_back(std::move(b));  // keep borrowed buffers alive
-    maybe_grow_or_skip();
   }
 
   std::vector<OutField> output_fields() const {
@@ -486,16 +537,13 @@ class AggOp {
     AURON_HIP(hipMemsetAsync(d_ng_.get(), 0, 8, stream_));
   }
 
-  void maybe_grow_or_skip() {
-    uint64_t ng = num_groups_host();
-    // partial skipping (agg_table.rs:109-120, agg_ctx.rs:174-185)
-    if (skip_enabled_ && row_cursor_ >= (uint64_t)skip_min_rows_ &&
-        (double)ng / (double)row_cursor_ >= skip_ratio_) {
-      skipping_ = true;
-      return;
-    }
-    // grow at 3/4 load
-    if ((int64_t)ng * 4 >= t_.cap * 3) grow(t_.cap * 4);
+  void refresh_ng() {
+    ng_true_ = num_groups_host();
+    ng_bound_ = ng_true_;
+  }
+
+  void ensure_capacity(int64_t need) {
+    while (t_.cap * 3 / 4 < need) grow(t_.cap * 4);
   }
 
   void grow(int64_t new_cap) {
@@ -635,6 +683,7 @@ class AggOp {
   double skip_ratio_ = 0.999;
   int64_t skip_min_rows_ = 20000;
   uint64_t row_cursor_ = 0;
+  uint64_t ng_bound_ = 0, ng_true_ = 0;  // conservative bound / last readback
   int64_t update_ns_ = 0, update_rows_ = 0;
   hipEvent_t ev_start_ = nullptr, ev_stop_ = nullptr;
   AggTable t_;

@@ -187,6 +187,18 @@ def test_partial_skipping_end_to_end():
     t.finalize()
 
 
+def test_table_grow():
+    """Force the hash table to grow (4x rebuild at 3/4 load) several times via
+    a tiny AURON_HIP_AGG_TABLE_SLOTS conf; results must be unaffected."""
+    keys, vals, vv = gen_northstar(300_000, nkeys=100_000)
+    t = blaze_amd.Task(plan.plan_partial_final(),
+                       batches=batches_of(keys, vals, vv),
+                       conf={"AURON_HIP_AGG_TABLE_SLOTS": 1024})
+    outputs = t.run()
+    assert_agg_parity(outputs, run_oracle(keys, vals, vv), fractional=False)
+    t.finalize()
+
+
 def test_shuffle_write_files(tmp_path):
     """Config-4 stage 1 on one GPU: partial agg + 200-way hash shuffle write.
     The data/index files must byte-match the oracle-constructed expectation
