@@ -30,6 +30,20 @@ struct EngineError : std::runtime_error {
   using std::runtime_error::runtime_error;
 };
 
+// AURON_DEBUG=1 traces engine stages to stderr (diagnostics only)
+static bool debug_on() {
+  static bool v = getenv("AURON_DEBUG") != nullptr;
+  return v;
+}
+#define DBG(...)                               \
+  do {                                         \
+    if (debug_on()) {                          \
+      fprintf(stderr, "[auron] " __VA_ARGS__); \
+      fprintf(stderr, "\n");                   \
+      fflush(stderr);                          \
+    }                                          \
+  } while (0)
+
 #define FAIL(msg) throw EngineError(msg)
 
 // ---------------------------------------------------------------- columns --
@@ -373,6 +387,7 @@ class AggOp {
       return v ? v + done / 8 : nullptr;  // done is a multiple of 8
     };
     int64_t done = 0;
+    DBG("agg.consume n=%lld merge=%d", (long long)b.num_rows, (int)merge_mode_);
     while (done < b.num_rows) {
       if (ng_bound_ * 4 >= (uint64_t)t_.cap * 3) refresh_ng();
       int64_t free_slots = t_.cap * 3 / 4 - (int64_t)ng_bound_;
@@ -411,6 +426,9 @@ class AggOp {
       done += chunk;
       row_cursor_ += (uint64_t)chunk;
       ng_bound_ += (uint64_t)chunk;
+      DBG("agg.chunk done=%lld/%lld cap=%lld ng_bound=%llu", (long long)done,
+          (long long)b.num_rows, (long long)t_.cap,
+          (unsigned long long)ng_bound_);
       // partial skipping (agg_table.rs:109-120): needs the true cardinality
       if (skip_enabled_ && !skipping_ &&
           row_cursor_ >= (uint64_t)skip_min_rows_) {
@@ -458,6 +476,7 @@ class AggOp {
     std::vector<std::pair<int64_t, std::vector<HostOutCol>>> out;
     // 1) table contents
     uint64_t ng = num_groups_host();
+    DBG("agg.finish ng=%llu", (unsigned long long)ng);
     if (ng > 0) {
       DevBuf slots_u(ng * 4), first(ng * 8), slots_sorted(ng * 4),
           first_sorted(ng * 8), dcount(8);
@@ -559,6 +578,7 @@ class AggOp {
 
   std::pair<int64_t, std::vector<HostOutCol>> emit_groups(
       const uint32_t* order_slots, int64_t n) {
+    DBG("agg.emit n=%lld final=%d", (long long)n, (int)final_output_);
     std::vector<HostOutCol> cols;
     size_t bm = (n + 7) / 8;
     DevBuf keys(n * 8), kvalid(bm), sums(n * 8), svalid(bm), cnts(n * 8);
@@ -1012,6 +1032,7 @@ struct Runtime {
       if (!cb.next_input_batch) break;
       int rc = cb.next_input_batch(cb.user, reader.resource_id.c_str(), &arr,
                                    &sch, &dev);
+      DBG("reader rc=%d", rc);
       if (rc == 0) break;
       DevBatch b;
       if (rc == 2) {
@@ -1155,6 +1176,7 @@ int64_t auron_call_native(const uint8_t* task_definition, size_t len,
   auto rt = std::make_unique<Runtime>();
   if (callbacks) rt->cb = *callbacks;
   try {
+    DBG("call_native len=%zu", len);
     int ndev = 0;
     hip_check(hipGetDeviceCount(&ndev), "hipGetDeviceCount");
     if (ndev == 0) throw EngineError("no HIP device visible");
@@ -1162,6 +1184,7 @@ int64_t auron_call_native(const uint8_t* task_definition, size_t len,
     std::string err;
     rt->td = decode_task_definition(task_definition, len, &err);
     if (!rt->td) throw EngineError("plan decode failed: " + err);
+    DBG("call_native ready");
   } catch (const std::exception& e) {
     rt->set_error(e.what());
     return 0;
