@@ -64,7 +64,11 @@ class ArrowDeviceArray(c.Structure):
     ]
 
 
-GET_CONF = c.CFUNCTYPE(c.c_int, c.c_void_p, c.c_char_p, c.c_char_p, c.c_size_t)
+# NOTE: the out-param must be POINTER(c_char), not c_char_p — ctypes passes
+# c_char_p callback args as immutable python bytes (a copy), so writes would
+# never reach the caller's buffer.
+GET_CONF = c.CFUNCTYPE(c.c_int, c.c_void_p, c.c_char_p, c.POINTER(c.c_char),
+                       c.c_size_t)
 NEXT_INPUT = c.CFUNCTYPE(c.c_int, c.c_void_p, c.c_char_p, c.POINTER(ArrowArray),
                          c.POINTER(ArrowSchema), c.POINTER(ArrowDeviceArray))
 IMPORT_SCHEMA = c.CFUNCTYPE(None, c.c_void_p, c.POINTER(ArrowSchema))

@@ -312,7 +312,9 @@ struct Conf {
   AuronCallbacks* cb;
   std::string get(const char* key, const std::string& dflt) const {
     char buf[256];
-    if (cb && cb->get_conf && cb->get_conf(cb->user, key, buf, sizeof(buf)) == 0)
+    buf[0] = '\0';
+    if (cb && cb->get_conf &&
+        cb->get_conf(cb->user, key, buf, sizeof(buf)) == 0 && buf[0] != '\0')
       return std::string(buf);
     return dflt;
   }
@@ -356,6 +358,7 @@ class AggOp {
       val_col_ = sum_child.col_index;
     }
     batch_size_ = conf.get_i("BATCH_SIZE", 10000);
+    if (batch_size_ <= 0) FAIL("invalid BATCH_SIZE conf");
     skip_enabled_ = node.supports_partial_skipping && !merge_mode_;
     skip_ratio_ = conf.get_d("PARTIAL_AGG_SKIPPING_RATIO", 0.999);
     skip_min_rows_ = conf.get_i("PARTIAL_AGG_SKIPPING_MIN_ROWS", 20000);
@@ -729,6 +732,7 @@ class ShuffleOp {
     }
     P_ = (uint32_t)node.partitioning.partition_count;
     batch_size_ = conf.get_i("BATCH_SIZE", 10000);
+    if (batch_size_ <= 0) FAIL("invalid BATCH_SIZE conf");
     target_block_ = (size_t)conf.get_i("SHUFFLE_COMPRESSION_TARGET_BUF_SIZE",
                                        4194304);
     std::string codec = conf.get("SPARK_IO_COMPRESSION_CODEC", "lz4");
@@ -1217,6 +1221,17 @@ void auron_on_exit(void) {
 }
 
 const char* auron_version(void) { return "auron-hip 0.1 gfx950"; }
+
+// test-only: exercise the get_conf callback plumbing from the C side (the
+// ctypes out-param contract is easy to get wrong — see GET_CONF note)
+int32_t auron_debug_conf_roundtrip(AuronCallbacks* cb, const char* key,
+                                   char* out, size_t cap) {
+  Conf conf{cb};
+  std::string v = conf.get(key, "<default>");
+  if (v.size() + 1 > cap) return -1;
+  memcpy(out, v.c_str(), v.size() + 1);
+  return (int32_t)v.size();
+}
 
 // Compute murmur3(seed 42) + pmod partition ids for host-resident i64 keys on
 // the GPU (shuffle/mod.rs:163-188). Used by bench.py's RCCL exchange leg to

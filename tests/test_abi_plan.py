@@ -64,3 +64,34 @@ def test_call_native_without_gpu_fails_loudly():
     td = plan.plan_partial_final()
     with pytest.raises(RuntimeError, match="HIP|hip|device"):
         blaze_amd.Task(td, batches=[])
+
+
+def test_conf_callback_roundtrip():
+    """The get_conf out-param must actually reach the C side (regression for
+    the c_char_p-vs-POINTER(c_char) ctypes trap that silently fed garbage
+    BATCH_SIZE to the engine)."""
+    blaze_amd.build()
+    lib = blaze_amd.lib()
+    lib.auron_debug_conf_roundtrip.restype = ctypes.c_int32
+    lib.auron_debug_conf_roundtrip.argtypes = [
+        ctypes.POINTER(blaze_amd.AuronCallbacks), ctypes.c_char_p,
+        ctypes.c_char_p, ctypes.c_size_t]
+    conf = {"BATCH_SIZE": 1048576}
+
+    @blaze_amd.GET_CONF
+    def get_conf(user, key, value, cap):
+        k = key.decode()
+        if k in conf:
+            v = str(conf[k]).encode()[: cap - 1]
+            ctypes.memmove(value, v + b"\x00", len(v) + 1)
+            return 0
+        return 1
+
+    cb = blaze_amd.AuronCallbacks()
+    cb.get_conf = get_conf
+    out = ctypes.create_string_buffer(256)
+    rc = lib.auron_debug_conf_roundtrip(ctypes.byref(cb), b"BATCH_SIZE", out,
+                                        256)
+    assert rc == 7 and out.value == b"1048576"
+    lib.auron_debug_conf_roundtrip(ctypes.byref(cb), b"MISSING", out, 256)
+    assert out.value == b"<default>"
