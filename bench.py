@@ -117,33 +117,43 @@ def run_step(ba, plan, dev_input, world, rank, torch, dist, device, stats):
     # BATCH_SIZE is the reference's own conf knob (conf.rs:32); raised here so
     # the ~1M-group output is emitted in few chunks instead of 100 × 10k.
     conf = {"BATCH_SIZE": 1 << 20}
-    _log("stage1: partial agg start")
+
+    def mark(name, t0):
+        dt = time.perf_counter() - t0
+        stats.setdefault("stage_s", {}).setdefault(name, 0.0)
+        stats["stage_s"][name] += dt
+        _log(f"{name}: {dt*1000:.1f} ms")
+        return time.perf_counter()
+
+    tm = time.perf_counter()
     t = ba.Task(plan.plan_partial_only(), device_batches=[dev_input], conf=conf)
+    tm = mark("s1_create", tm)
     outs = t.run()
-    _log("stage1 done")
+    tm = mark("s1_run", tm)
     stats["agg_update_ns"] += t.metric("agg_update_ns")
     stats["agg_update_rows"] += t.metric("agg_update_rows")
     stats["num_groups"] = t.metric("num_groups")
     t.finalize()
     keys, lens, data = collect_partial(outs)
+    tm = mark("s1_collect", tm)
 
     # stage 2: 200-way murmur3 partition, owner rank = partition % N
-    _log("stage2: partition ids")
     pids = ba.partition_ids(keys, NUM_PARTITIONS)
+    tm = mark("s2_partition", tm)
     if world > 1:
         dest = (pids % world).astype(np.int64)
         keys, lens, data = exchange(torch, dist, device, keys, lens, data,
                                     dest, world)
+        tm = mark("s2_exchange", tm)
 
     # stage 3: final merge agg of (local + received) partial records
     offs = np.concatenate([[0], np.cumsum(lens)]).astype(np.int32)
     t2 = ba.Task(plan.plan_final_only(), conf={"BATCH_SIZE": 1 << 20},
                  batches=[[(keys, None), ("binary", data, offs, None)]])
-    _log("stage3: final merge")
     out2 = t2.run()
+    tm = mark("s3_final", tm)
     nfinal = sum(ob[0]["values"].shape[0] for ob in out2)
     t2.finalize()
-    _log(f"step done: {nfinal} groups")
     return nfinal
 
 
@@ -256,6 +266,8 @@ def main():
         elapsed = float(t.item())
 
     if rank == 0:
+        print(f"[bench] stage seconds over {args.steps} steps: "
+              f"{stats.get('stage_s', {})}", file=sys.stderr, flush=True)
         value = args.rows * args.steps / elapsed
         upd_ns = max(stats["agg_update_ns"], 1)
         upd_rows = stats["agg_update_rows"]

@@ -356,6 +356,11 @@ class AggOp {
       const Expr& sum_child = node.agg_exprs[0].children.at(0);
       if (sum_child.kind != Expr::Column) FAIL("SUM arg must be a Column");
       val_col_ = sum_child.col_index;
+      // the engine derives sum-validity from cnt>0, which requires SUM and
+      // COUNT to share the argument column (the north-star shape)
+      const Expr& cnt_child = node.agg_exprs[1].children.at(0);
+      if (cnt_child.kind != Expr::Column || cnt_child.col_index != val_col_)
+        FAIL("COUNT arg must be the same column as SUM arg");
     }
     batch_size_ = conf.get_i("BATCH_SIZE", 10000);
     if (batch_size_ <= 0) FAIL("invalid BATCH_SIZE conf");
@@ -537,7 +542,6 @@ class AggOp {
     d_special_.alloc(2 * 4);
     d_sum_.alloc((cap + 2) * 8);
     d_cnt_.alloc((cap + 2) * 8);
-    d_sum_valid_.alloc((cap + 2) * 4);
     d_first_.alloc((cap + 2) * 8);
     d_ng_.alloc(8);
     d_err_.alloc(4);
@@ -545,7 +549,6 @@ class AggOp {
     t_.special_used = d_special_.get<uint32_t>();
     t_.sum = d_sum_.get<double>();
     t_.cnt = d_cnt_.get<unsigned long long>();
-    t_.sum_valid = d_sum_valid_.get<uint32_t>();
     t_.first_row = d_first_.get<unsigned long long>();
     t_.num_groups = d_ng_.get<unsigned long long>();
     t_.error_flag = d_err_.get<uint32_t>();
@@ -554,7 +557,6 @@ class AggOp {
     AURON_HIP(hipMemsetAsync(d_special_.get(), 0, 2 * 4, stream_));
     AURON_HIP(hipMemsetAsync(d_sum_.get(), 0, (cap + 2) * 8, stream_));
     AURON_HIP(hipMemsetAsync(d_cnt_.get(), 0, (cap + 2) * 8, stream_));
-    AURON_HIP(hipMemsetAsync(d_sum_valid_.get(), 0, (cap + 2) * 4, stream_));
     AURON_HIP(hipMemsetAsync(d_first_.get(), 0xFF, (cap + 2) * 8, stream_));
     AURON_HIP(hipMemsetAsync(d_ng_.get(), 0, 8, stream_));
   }
@@ -572,7 +574,7 @@ class AggOp {
     AggTable old = t_;
     DevBuf ok = std::move(d_slot_key_), os = std::move(d_special_),
            osum = std::move(d_sum_), ocnt = std::move(d_cnt_),
-           ov = std::move(d_sum_valid_), of = std::move(d_first_),
+           of = std::move(d_first_),
            ong = std::move(d_ng_), oerr = std::move(d_err_);
     init_table(new_cap);
     launch_agg_rebuild(t_, old, stream_);
@@ -710,7 +712,7 @@ class AggOp {
   int64_t update_ns_ = 0, update_rows_ = 0;
   hipEvent_t ev_start_ = nullptr, ev_stop_ = nullptr;
   AggTable t_;
-  DevBuf d_slot_key_, d_special_, d_sum_, d_cnt_, d_sum_valid_, d_first_, d_ng_,
+  DevBuf d_slot_key_, d_special_, d_sum_, d_cnt_, d_first_, d_ng_,
       d_err_;
   std::vector<DevBatch> held_, skipped_;
 };

@@ -30,7 +30,6 @@ struct AggTable {
   // accumulators indexed by slot (cap + 2 entries):
   double* sum = nullptr;
   unsigned long long* cnt = nullptr;
-  uint32_t* sum_valid = nullptr;
   unsigned long long* first_row = nullptr;  // global arrival index, for order
   unsigned long long* num_groups = nullptr; // [1] device counter
   uint32_t* error_flag = nullptr;           // [1] raised on probe exhaustion

@@ -194,7 +194,8 @@ __global__ void k_agg_update(const AggTable t, const int64_t* __restrict__ keys,
       // sum.rs:90-115: SUM adds non-null args; valid-ness latches on
       unsafeAtomicAdd(&t.sum[a], vals[i]);
       atomicAdd(&t.cnt[a], 1ull);  // count.rs:90-149: COUNT(arg) non-null
-      if (!t.sum_valid[a]) atomicOr(&t.sum_valid[a], 1u);
+      // sum validity is implied by cnt>0 (same-column agg set; engine.cpp
+      // enforces SUM/COUNT share the argument column)
     }
   }
 }
@@ -235,7 +236,6 @@ __global__ void k_agg_merge_frozen(const AggTable t,
       memcpy(&v, p, 8);
       p += 8;
       unsafeAtomicAdd(&t.sum[a], v);
-      if (!t.sum_valid[a]) atomicOr(&t.sum_valid[a], 1u);
     }
     int used;
     uint64_t c = read_varint_dev(p, &used);
@@ -280,7 +280,7 @@ __global__ void k_agg_gather_out(const AggTable t,
         uint32_t sj = order_slots[i + j];
         bool knull = (sj == t.cap + 1);
         if (!knull) kb |= (uint8_t)(1u << j);
-        if (t.sum_valid[sj]) sb |= (uint8_t)(1u << j);
+        if (t.cnt[sj] != 0) sb |= (uint8_t)(1u << j);
       }
       if (key_validity) key_validity[i >> 3] = kb;
       if (sum_validity) sum_validity[i >> 3] = sb;
@@ -303,7 +303,7 @@ __global__ void k_agg_freeze_len(const AggTable t,
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < num_groups;
        i += (int64_t)gridDim.x * blockDim.x) {
     uint32_t s = order_slots[i];
-    lens[i] = 1 + (t.sum_valid[s] ? 8 : 0) + varint_len_dev(t.cnt[s]);
+    lens[i] = 1 + (t.cnt[s] != 0 ? 8 : 0) + varint_len_dev(t.cnt[s]);
   }
 }
 
@@ -317,7 +317,7 @@ __global__ void k_agg_freeze_write(const AggTable t,
     uint32_t s = order_slots[i];
     uint8_t* p = out + offsets[i];
     // acc.rs:335-347: [u8 valid][8B LE value]? then count.rs:193-203 varint
-    if (t.sum_valid[s]) {
+    if (t.cnt[s] != 0) {
       *p++ = 1;
       double v = t.sum[s];
       memcpy(p, &v, 8);
@@ -492,7 +492,6 @@ __global__ void k_agg_rebuild(const AggTable dst, const AggTable src) {
     // distinct keys per source slot: single writer, plain stores
     dst.sum[a] = src.sum[i];
     dst.cnt[a] = src.cnt[i];
-    dst.sum_valid[a] = src.sum_valid[i];
     dst.first_row[a] = src.first_row[i];
   }
 }
