@@ -73,4 +73,38 @@ class DevBuf {
   size_t size_ = 0;
 };
 
+// Pinned (page-locked) host buffer for small metadata readbacks — pageable
+// D2H copies cost milliseconds per call on this stack.
+class PinnedBuf {
+ public:
+  PinnedBuf() = default;
+  explicit PinnedBuf(size_t size) { alloc(size); }
+  PinnedBuf(const PinnedBuf&) = delete;
+  PinnedBuf& operator=(const PinnedBuf&) = delete;
+  ~PinnedBuf() { free(); }
+
+  void alloc(size_t size) {
+    free();
+    if (size == 0) return;
+    AURON_HIP(hipHostMalloc(&ptr_, size));
+    size_ = size;
+  }
+  void free() {
+    if (ptr_) {
+      (void)hipHostFree(ptr_);
+      ptr_ = nullptr;
+      size_ = 0;
+    }
+  }
+  template <typename T = void>
+  T* get() const {
+    return static_cast<T*>(ptr_);
+  }
+  size_t size() const { return size_; }
+
+ private:
+  void* ptr_ = nullptr;
+  size_t size_ = 0;
+};
+
 }  // namespace auron

@@ -374,6 +374,10 @@ class AggOp {
   }
 
   ~AggOp() {
+    for (auto& [e0, e1] : ev_pairs_) {
+      (void)hipEventDestroy(e0);
+      (void)hipEventDestroy(e1);
+    }
     (void)hipEventDestroy(ev_start_);
     (void)hipEventDestroy(ev_stop_);
   }
@@ -417,18 +421,19 @@ class AggOp {
         const DevColumn& val = b.cols.at(val_col_);
         if (val.dt != DType::Float64) FAIL("SUM arg must be Float64");
         // HIP-event timing on the launch stream (roofline evidence for the
-        // dominant kernel; read back via auron_get_metric)
-        AURON_HIP(hipEventRecord(ev_start_, stream_));
+        // dominant kernel; pairs are drained once at finish() so the hot
+        // loop never synchronizes for timing)
+        hipEvent_t e0, e1;
+        AURON_HIP(hipEventCreate(&e0));
+        AURON_HIP(hipEventCreate(&e1));
+        AURON_HIP(hipEventRecord(e0, stream_));
         launch_agg_update(t_, (const int64_t*)key.values + done,
                           slice_valid(key.validity, done),
                           (const double*)val.values + done,
                           slice_valid(val.validity, done), chunk, row_cursor_,
                           stream_);
-        AURON_HIP(hipEventRecord(ev_stop_, stream_));
-        AURON_HIP(hipEventSynchronize(ev_stop_));
-        float ms = 0.f;
-        AURON_HIP(hipEventElapsedTime(&ms, ev_start_, ev_stop_));
-        update_ns_ += (int64_t)(ms * 1e6);
+        AURON_HIP(hipEventRecord(e1, stream_));
+        ev_pairs_.push_back({e0, e1});
         update_rows_ += chunk;
       }
       done += chunk;
@@ -482,6 +487,7 @@ class AggOp {
   // drain: produce all output batches (host-staged)
   std::vector<std::pair<int64_t, std::vector<HostOutCol>>> finish() {
     std::vector<std::pair<int64_t, std::vector<HostOutCol>>> out;
+    drain_timing();
     // 1) table contents
     uint64_t ng = num_groups_host();
     DBG("agg.finish ng=%llu", (unsigned long long)ng);
@@ -520,15 +526,16 @@ class AggOp {
   }
 
   uint64_t num_groups_host() {
-    uint64_t ng = 0;
-    uint32_t err = 0;
-    AURON_HIP(hipMemcpyAsync(&ng, t_.num_groups, 8, hipMemcpyDeviceToHost,
+    if (!pinned_meta_.get()) pinned_meta_.alloc(16);
+    uint64_t* ng = pinned_meta_.get<uint64_t>();
+    uint32_t* err = (uint32_t*)(pinned_meta_.get<uint8_t>() + 8);
+    AURON_HIP(hipMemcpyAsync(ng, t_.num_groups, 8, hipMemcpyDeviceToHost,
                              stream_));
-    AURON_HIP(hipMemcpyAsync(&err, t_.error_flag, 4, hipMemcpyDeviceToHost,
+    AURON_HIP(hipMemcpyAsync(err, t_.error_flag, 4, hipMemcpyDeviceToHost,
                              stream_));
     AURON_HIP(hipStreamSynchronize(stream_));
-    if (err) FAIL("agg hash table probe exhausted (table full/corrupt)");
-    return ng;
+    if (*err) FAIL("agg hash table probe exhausted (table full/corrupt)");
+    return *ng;
   }
 
   int64_t batch_size() const { return batch_size_; }
@@ -538,32 +545,36 @@ class AggOp {
     int64_t cap = 1;
     while (cap < slots) cap <<= 1;
     t_.cap = cap;
-    d_slot_key_.alloc(cap * 8);
+    d_slots_.alloc((cap + 2) * sizeof(AggSlot));
     d_special_.alloc(2 * 4);
-    d_sum_.alloc((cap + 2) * 8);
-    d_cnt_.alloc((cap + 2) * 8);
-    d_first_.alloc((cap + 2) * 8);
     d_ng_.alloc(8);
     d_err_.alloc(4);
-    t_.slot_key = d_slot_key_.get<int64_t>();
+    t_.slots = d_slots_.get<AggSlot>();
     t_.special_used = d_special_.get<uint32_t>();
-    t_.sum = d_sum_.get<double>();
-    t_.cnt = d_cnt_.get<unsigned long long>();
-    t_.first_row = d_first_.get<unsigned long long>();
     t_.num_groups = d_ng_.get<unsigned long long>();
     t_.error_flag = d_err_.get<uint32_t>();
     AURON_HIP(hipMemsetAsync(d_err_.get(), 0, 4, stream_));
-    launch_fill_i64(t_.slot_key, INT64_MIN, cap, stream_);
     AURON_HIP(hipMemsetAsync(d_special_.get(), 0, 2 * 4, stream_));
-    AURON_HIP(hipMemsetAsync(d_sum_.get(), 0, (cap + 2) * 8, stream_));
-    AURON_HIP(hipMemsetAsync(d_cnt_.get(), 0, (cap + 2) * 8, stream_));
-    AURON_HIP(hipMemsetAsync(d_first_.get(), 0xFF, (cap + 2) * 8, stream_));
     AURON_HIP(hipMemsetAsync(d_ng_.get(), 0, 8, stream_));
+    launch_slots_init(t_.slots, cap + 2, stream_);
   }
 
   void refresh_ng() {
     ng_true_ = num_groups_host();
     ng_bound_ = ng_true_;
+  }
+
+  void drain_timing() {
+    if (ev_pairs_.empty()) return;
+    AURON_HIP(hipStreamSynchronize(stream_));
+    for (auto& [e0, e1] : ev_pairs_) {
+      float ms = 0.f;
+      AURON_HIP(hipEventElapsedTime(&ms, e0, e1));
+      update_ns_ += (int64_t)(ms * 1e6);
+      (void)hipEventDestroy(e0);
+      (void)hipEventDestroy(e1);
+    }
+    ev_pairs_.clear();
   }
 
   void ensure_capacity(int64_t need) {
@@ -572,9 +583,7 @@ class AggOp {
 
   void grow(int64_t new_cap) {
     AggTable old = t_;
-    DevBuf ok = std::move(d_slot_key_), os = std::move(d_special_),
-           osum = std::move(d_sum_), ocnt = std::move(d_cnt_),
-           of = std::move(d_first_),
+    DevBuf oslots = std::move(d_slots_), os = std::move(d_special_),
            ong = std::move(d_ng_), oerr = std::move(d_err_);
     init_table(new_cap);
     launch_agg_rebuild(t_, old, stream_);
@@ -710,10 +719,11 @@ class AggOp {
   uint64_t row_cursor_ = 0;
   uint64_t ng_bound_ = 0, ng_true_ = 0;  // conservative bound / last readback
   int64_t update_ns_ = 0, update_rows_ = 0;
+  std::vector<std::pair<hipEvent_t, hipEvent_t>> ev_pairs_;
   hipEvent_t ev_start_ = nullptr, ev_stop_ = nullptr;
   AggTable t_;
-  DevBuf d_slot_key_, d_special_, d_sum_, d_cnt_, d_first_, d_ng_,
-      d_err_;
+  DevBuf d_slots_, d_special_, d_ng_, d_err_;
+  PinnedBuf pinned_meta_;
   std::vector<DevBatch> held_, skipped_;
 };
 

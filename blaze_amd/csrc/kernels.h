@@ -20,17 +20,23 @@ void launch_pmod(const int32_t* hashes, int64_t n, int32_t num_partitions,
 // ---- hash aggregation ------------------------------------------------------
 // GPU replacement for AggHashMap::upsert_records + per-agg partial_update
 // (agg_hash_map.rs:77-168, sum.rs:90-115, count.rs:90-149).
-// Open-addressing table over i64 keys; slot i64::MIN is the empty sentinel,
-// the two trailing acc rows (index cap, cap+1) hold the i64::MIN-key group
-// and the null-key group.
+// Open-addressing table of 32-byte slots over i64 keys (key i64::MIN = empty
+// sentinel): key probe and the row's atomics all land in ONE 64-byte cache
+// line (2 slots/line), keeping the random traffic to one line per row. The
+// two trailing slots (index cap, cap+1) hold the i64::MIN-key group and the
+// null-key group.
+struct AggSlot {
+  long long key;                 // INT64_MIN = empty
+  unsigned long long cnt;        // COUNT accumulator (also: sum valid <=> >0)
+  double sum;                    // SUM(f64) accumulator
+  unsigned long long first_row;  // global arrival index, for record order
+};
+static_assert(sizeof(AggSlot) == 32, "AggSlot must be 32 bytes");
+
 struct AggTable {
-  int64_t cap = 0;           // power of two
-  int64_t* slot_key = nullptr;        // [cap], EMPTY = INT64_MIN
+  int64_t cap = 0;                    // power of two
+  AggSlot* slots = nullptr;           // [cap + 2]
   uint32_t* special_used = nullptr;   // [2]: {min-key group, null-key group}
-  // accumulators indexed by slot (cap + 2 entries):
-  double* sum = nullptr;
-  unsigned long long* cnt = nullptr;
-  unsigned long long* first_row = nullptr;  // global arrival index, for order
   unsigned long long* num_groups = nullptr; // [1] device counter
   uint32_t* error_flag = nullptr;           // [1] raised on probe exhaustion
 };
@@ -78,8 +84,10 @@ void launch_skip_freeze_write(const double* vals, const uint8_t* val_valid,
                               int64_t n, const int32_t* offsets, uint8_t* out,
                               hipStream_t s);
 
-// fill an i64 array with a constant (table init: slot_key = INT64_MIN)
+// fill an i64 array with a constant
 void launch_fill_i64(int64_t* dst, int64_t value, int64_t n, hipStream_t s);
+// initialize table slots (key = empty sentinel, accs zero, first_row = ~0)
+void launch_slots_init(AggSlot* slots, int64_t n, hipStream_t s);
 
 // rebuild `src` table into the (larger, initialized) `dst` table; each source
 // slot holds a distinct key so plain stores after the CAS claim are race-free.

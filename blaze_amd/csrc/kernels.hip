@@ -160,12 +160,12 @@ __device__ __forceinline__ int64_t agg_upsert_slot(const AggTable t, int64_t key
   int64_t mask = t.cap - 1;
   int64_t i = (int64_t)(h & (uint64_t)mask);
   for (int64_t probe = 0; probe <= mask; probe++) {
-    int64_t cur = t.slot_key[i];
+    long long cur = t.slots[i].key;
     if (cur == key) return i;
     if (cur == KEY_EMPTY) {
-      int64_t prev = (int64_t)atomicCAS((unsigned long long*)&t.slot_key[i],
-                                        (unsigned long long)KEY_EMPTY,
-                                        (unsigned long long)key);
+      long long prev = (long long)atomicCAS((unsigned long long*)&t.slots[i].key,
+                                            (unsigned long long)KEY_EMPTY,
+                                            (unsigned long long)key);
       if (prev == KEY_EMPTY) {
         atomicAdd(t.num_groups, 1ull);
         return i;
@@ -188,14 +188,19 @@ __global__ void k_agg_update(const AggTable t, const int64_t* __restrict__ keys,
     bool knull = key_valid && !bit_get_dev(key_valid, i);
     int64_t a = agg_upsert_slot(t, knull ? 0 : keys[i], knull);
     if (a < 0) continue;  // table full: error_flag raised, host aborts
-    atomicMin(&t.first_row[a], row_offset + (uint64_t)i);
+    AggSlot* sl = &t.slots[a];
+    uint64_t row = row_offset + (uint64_t)i;
+    // skip the atomic when first_row is already <= row: a stale (L1) read can
+    // only be HIGHER than the true value (first_row only decreases), so the
+    // skip is always safe
+    if (sl->first_row > row) atomicMin(&sl->first_row, row);
     bool vvalid = !val_valid || bit_get_dev(val_valid, i);
     if (vvalid) {
-      // sum.rs:90-115: SUM adds non-null args; valid-ness latches on
-      unsafeAtomicAdd(&t.sum[a], vals[i]);
-      atomicAdd(&t.cnt[a], 1ull);  // count.rs:90-149: COUNT(arg) non-null
+      // sum.rs:90-115: SUM adds non-null args; valid-ness latches on.
       // sum validity is implied by cnt>0 (same-column agg set; engine.cpp
       // enforces SUM/COUNT share the argument column)
+      unsafeAtomicAdd(&sl->sum, vals[i]);
+      atomicAdd(&sl->cnt, 1ull);  // count.rs:90-149: COUNT(arg) non-null
     }
   }
 }
@@ -228,18 +233,20 @@ __global__ void k_agg_merge_frozen(const AggTable t,
     bool knull = key_valid && !bit_get_dev(key_valid, i);
     int64_t a = agg_upsert_slot(t, knull ? 0 : keys[i], knull);
     if (a < 0) continue;
-    atomicMin(&t.first_row[a], row_offset + (uint64_t)i);
+    AggSlot* sl = &t.slots[a];
+    uint64_t row = row_offset + (uint64_t)i;
+    if (sl->first_row > row) atomicMin(&sl->first_row, row);
     const uint8_t* p = acc_data + acc_offsets[i];
     uint8_t valid = *p++;
     if (valid) {  // sum.rs:117-145 partial_merge
       double v;
       memcpy(&v, p, 8);
       p += 8;
-      unsafeAtomicAdd(&t.sum[a], v);
+      unsafeAtomicAdd(&sl->sum, v);
     }
     int used;
     uint64_t c = read_varint_dev(p, &used);
-    if (c) atomicAdd(&t.cnt[a], c);
+    if (c) atomicAdd(&sl->cnt, c);
   }
 }
 
@@ -249,12 +256,12 @@ __global__ void k_agg_compact(const AggTable t, uint32_t* __restrict__ out_slot,
   int64_t total = t.cap + 2;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (int64_t)gridDim.x * blockDim.x) {
-    bool used = (i < t.cap) ? (t.slot_key[i] != KEY_EMPTY)
+    bool used = (i < t.cap) ? (t.slots[i].key != KEY_EMPTY)
                             : (t.special_used[i - t.cap] != 0);
     if (used) {
       unsigned long long idx = atomicAdd(num_out, 1ull);
       out_slot[idx] = (uint32_t)i;
-      out_first_row[idx] = t.first_row[i];
+      out_first_row[idx] = t.slots[i].first_row;
     }
   }
 }
@@ -270,17 +277,17 @@ __global__ void k_agg_gather_out(const AggTable t,
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < num_groups;
        i += (int64_t)gridDim.x * blockDim.x) {
     uint32_t s = order_slots[i];
-    if (keys) keys[i] = (s < t.cap) ? t.slot_key[s]
+    if (keys) keys[i] = (s < t.cap) ? t.slots[s].key
                                     : (s == t.cap ? KEY_EMPTY : 0);
-    if (sums) sums[i] = t.sum[s];
-    if (counts) counts[i] = (long long)t.cnt[s];
+    if (sums) sums[i] = t.slots[s].sum;
+    if (counts) counts[i] = (long long)t.slots[s].cnt;
     if ((i & 7) == 0) {
       uint8_t kb = 0, sb = 0;
       for (int j = 0; j < 8 && i + j < num_groups; j++) {
         uint32_t sj = order_slots[i + j];
         bool knull = (sj == t.cap + 1);
         if (!knull) kb |= (uint8_t)(1u << j);
-        if (t.cnt[sj] != 0) sb |= (uint8_t)(1u << j);
+        if (t.slots[sj].cnt != 0) sb |= (uint8_t)(1u << j);
       }
       if (key_validity) key_validity[i >> 3] = kb;
       if (sum_validity) sum_validity[i >> 3] = sb;
@@ -303,7 +310,7 @@ __global__ void k_agg_freeze_len(const AggTable t,
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < num_groups;
        i += (int64_t)gridDim.x * blockDim.x) {
     uint32_t s = order_slots[i];
-    lens[i] = 1 + (t.cnt[s] != 0 ? 8 : 0) + varint_len_dev(t.cnt[s]);
+    lens[i] = 1 + (t.slots[s].cnt != 0 ? 8 : 0) + varint_len_dev(t.slots[s].cnt);
   }
 }
 
@@ -317,15 +324,15 @@ __global__ void k_agg_freeze_write(const AggTable t,
     uint32_t s = order_slots[i];
     uint8_t* p = out + offsets[i];
     // acc.rs:335-347: [u8 valid][8B LE value]? then count.rs:193-203 varint
-    if (t.cnt[s] != 0) {
+    if (t.slots[s].cnt != 0) {
       *p++ = 1;
-      double v = t.sum[s];
+      double v = t.slots[s].sum;
       memcpy(p, &v, 8);
       p += 8;
     } else {
       *p++ = 0;
     }
-    uint64_t c = t.cnt[s];
+    uint64_t c = t.slots[s].cnt;
     while (c >= 128) {
       *p++ = (uint8_t)(128 + c % 128);
       c /= 128;
@@ -461,6 +468,22 @@ void launch_fill_i64(int64_t* dst, int64_t value, int64_t n, hipStream_t s) {
   check_launch("k_fill_i64");
 }
 
+__global__ void k_slots_init(AggSlot* __restrict__ slots, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    slots[i].key = KEY_EMPTY;
+    slots[i].cnt = 0;
+    slots[i].sum = 0.0;
+    slots[i].first_row = ~0ull;
+  }
+}
+
+void launch_slots_init(AggSlot* slots, int64_t n, hipStream_t s) {
+  hipLaunchKernelGGL(k_slots_init, dim3(grid_for(n)), dim3(BLOCK), 0, s, slots,
+                     n);
+  check_launch("k_slots_init");
+}
+
 __global__ void k_iota_u32(uint32_t* __restrict__ dst, int64_t n) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x)
@@ -476,7 +499,7 @@ __global__ void k_agg_rebuild(const AggTable dst, const AggTable src) {
   int64_t total = src.cap + 2;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (int64_t)gridDim.x * blockDim.x) {
-    bool used = (i < src.cap) ? (src.slot_key[i] != KEY_EMPTY)
+    bool used = (i < src.cap) ? (src.slots[i].key != KEY_EMPTY)
                               : (src.special_used[i - src.cap] != 0);
     if (!used) continue;
     int64_t a;
@@ -486,13 +509,13 @@ __global__ void k_agg_rebuild(const AggTable dst, const AggTable src) {
         atomicAdd(dst.num_groups, 1ull);
       a = dst.cap + which;
     } else {
-      a = agg_upsert_slot(dst, src.slot_key[i], false);
+      a = agg_upsert_slot(dst, src.slots[i].key, false);
       if (a < 0) continue;
     }
     // distinct keys per source slot: single writer, plain stores
-    dst.sum[a] = src.sum[i];
-    dst.cnt[a] = src.cnt[i];
-    dst.first_row[a] = src.first_row[i];
+    dst.slots[a].sum = src.slots[i].sum;
+    dst.slots[a].cnt = src.slots[i].cnt;
+    dst.slots[a].first_row = src.slots[i].first_row;
   }
 }
 
