@@ -401,11 +401,16 @@ class AggOp {
     int64_t done = 0;
     DBG("agg.consume n=%lld merge=%d", (long long)b.num_rows, (int)merge_mode_);
     while (done < b.num_rows) {
-      if (ng_bound_ * 4 >= (uint64_t)t_.cap * 3) refresh_ng();
       int64_t free_slots = t_.cap * 3 / 4 - (int64_t)ng_bound_;
       if (free_slots < (1 << 16)) {
-        ensure_capacity((int64_t)ng_bound_ + (1 << 20));
-        free_slots = t_.cap * 3 / 4 - (int64_t)ng_bound_;
+        // the conservative bound is exhausted: read the true cardinality
+        // first; grow only if the table is genuinely near 3/4 load
+        refresh_ng();
+        free_slots = t_.cap * 3 / 4 - (int64_t)ng_true_;
+        if (free_slots < (1 << 16)) {
+          ensure_capacity((int64_t)ng_true_ + (1 << 20));
+          free_slots = t_.cap * 3 / 4 - (int64_t)ng_bound_;
+        }
       }
       int64_t chunk = std::min(b.num_rows - done, free_slots);
       if (done + chunk < b.num_rows) chunk &= ~(int64_t)7;  // bitmap-sliceable
