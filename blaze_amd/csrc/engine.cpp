@@ -367,7 +367,7 @@ class AggOp {
     skip_enabled_ = node.supports_partial_skipping && !merge_mode_;
     skip_ratio_ = conf.get_d("PARTIAL_AGG_SKIPPING_RATIO", 0.999);
     skip_min_rows_ = conf.get_i("PARTIAL_AGG_SKIPPING_MIN_ROWS", 20000);
-    int64_t slots = conf.get_i("AURON_HIP_AGG_TABLE_SLOTS", 1 << 22);
+    int64_t slots = conf.get_i("AURON_HIP_AGG_TABLE_SLOTS", 1 << 23);
     init_table(slots);
     AURON_HIP(hipEventCreate(&ev_start_));
     AURON_HIP(hipEventCreate(&ev_stop_));
