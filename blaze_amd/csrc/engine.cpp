@@ -401,6 +401,18 @@ class AggOp {
     int64_t done = 0;
     DBG("agg.consume n=%lld merge=%d", (long long)b.num_rows, (int)merge_mode_);
     while (done < b.num_rows) {
+      // two-phase path (update mode, large chunks): its table inserts are
+      // bounded by counted staged/leftover lists, not by chunk rows, so it
+      // chunks on partition-buffer size instead of table free slots
+      if (!merge_mode_ && b.num_rows - done >= AGG2_MIN_CHUNK) {
+        int64_t chunk2 = std::min(b.num_rows - done, AGG2_MAX_CHUNK);
+        if (done + chunk2 < b.num_rows) chunk2 &= ~(int64_t)7;
+        two_phase_chunk(b, done, chunk2);
+        done += chunk2;
+        row_cursor_ += (uint64_t)chunk2;
+        if (maybe_enter_skipping(b, done)) break;
+        continue;
+      }
       int64_t free_slots = t_.cap * 3 / 4 - (int64_t)ng_bound_;
       if (free_slots < (1 << 16)) {
         // the conservative bound is exhausted: read the true cardinality
@@ -447,35 +459,38 @@ class AggOp {
       DBG("agg.chunk done=%lld/%lld cap=%lld ng_bound=%llu", (long long)done,
           (long long)b.num_rows, (long long)t_.cap,
           (unsigned long long)ng_bound_);
-      // partial skipping (agg_table.rs:109-120): needs the true cardinality
-      if (skip_enabled_ && !skipping_ &&
-          row_cursor_ >= (uint64_t)skip_min_rows_) {
-        refresh_ng();
-        if ((double)ng_true_ / (double)row_cursor_ >= skip_ratio_) {
-          skipping_ = true;
-          if (done < b.num_rows) {
-            // the un-aggregated tail of this batch passes through, like the
-            // reference flipping between batches
-            DevBatch tail;
-            tail.num_rows = b.num_rows - done;
-            for (const DevColumn& src : b.cols) {
-              DevColumn sc;
-              sc.dt = src.dt;
-              sc.len = tail.num_rows;
-              if (src.dt == DType::Binary || src.dt == DType::Utf8)
-                FAIL("binary columns unsupported in skip-tail slice");
-              sc.values =
-                  (const uint8_t*)src.values + done * dtype_width(src.dt);
-              sc.validity = slice_valid(src.validity, done);
-              tail.cols.push_back(std::move(sc));
-            }
-            skipped_.push_back(std::move(tail));
-          }
-          break;
-        }
-      }
+      if (maybe_enter_skipping(b, done)) break;
     }
     held_.push_back(std::move(b));  // keep borrowed buffers alive
+  }
+
+  // partial skipping (agg_table.rs:109-120): needs the true cardinality.
+  // Returns true (and slices this batch's unprocessed tail into skipped_)
+  // when the policy flips to pass-through.
+  bool maybe_enter_skipping(const DevBatch& b, int64_t done) {
+    if (!skip_enabled_ || skipping_ || row_cursor_ < (uint64_t)skip_min_rows_)
+      return false;
+    refresh_ng();
+    if ((double)ng_true_ / (double)row_cursor_ < skip_ratio_) return false;
+    skipping_ = true;
+    if (done < b.num_rows) {
+      // the un-aggregated tail of this batch passes through, like the
+      // reference flipping between batches
+      DevBatch tail;
+      tail.num_rows = b.num_rows - done;
+      for (const DevColumn& src : b.cols) {
+        DevColumn sc;
+        sc.dt = src.dt;
+        sc.len = tail.num_rows;
+        if (src.dt == DType::Binary || src.dt == DType::Utf8)
+          FAIL("binary columns unsupported in skip-tail slice");
+        sc.values = (const uint8_t*)src.values + done * dtype_width(src.dt);
+        sc.validity = src.validity ? src.validity + done / 8 : nullptr;
+        tail.cols.push_back(std::move(sc));
+      }
+      skipped_.push_back(std::move(tail));
+    }
+    return true;
   }
 
   std::vector<OutField> output_fields() const {
@@ -562,6 +577,105 @@ class AggOp {
     AURON_HIP(hipMemsetAsync(d_special_.get(), 0, 2 * 4, stream_));
     AURON_HIP(hipMemsetAsync(d_ng_.get(), 0, 8, stream_));
     launch_slots_init(t_.slots, cap + 2, stream_);
+  }
+
+  static constexpr int64_t AGG2_MIN_CHUNK = 4 << 20;   // below: single-phase
+  static constexpr int64_t AGG2_MAX_CHUNK = 64 << 20;  // partition buffer size
+  static constexpr int AGG2_NBUCK_LOG2 = 11;           // 2048 buckets
+
+  // Two-phase aggregation of rows [done, done+chunk) of batch b:
+  // histogram -> scatter to bucket-major SoA -> per-bucket LDS aggregate ->
+  // merge counted staged groups into the slot table (kernels_agg2.hip).
+  void two_phase_chunk(const DevBatch& b, int64_t done, int64_t chunk) {
+    const int nbuck = 1 << AGG2_NBUCK_LOG2;
+    const DevColumn& key = b.cols.at(key_col_);
+    const DevColumn& val = b.cols.at(val_col_);
+    const int64_t* keys = (const int64_t*)key.values + done;
+    const double* vals = (const double*)val.values + done;
+    const uint8_t* kv = key.validity ? key.validity + done / 8 : nullptr;
+    const uint8_t* vv = val.validity ? val.validity + done / 8 : nullptr;
+
+    if (!d_bkey_) {
+      d_bkey_.alloc(AGG2_MAX_CHUNK * 8);
+      d_bval_.alloc(AGG2_MAX_CHUNK * 8);
+      d_browv_.alloc(AGG2_MAX_CHUNK * 4);
+      d_lokey_.alloc(AGG2_MAX_CHUNK * 8);
+      d_loval_.alloc(AGG2_MAX_CHUNK * 8);
+      d_lorowv_.alloc(AGG2_MAX_CHUNK * 4);
+      d_counts_.alloc((nbuck + 1) * 4);
+      d_cursors_.alloc(nbuck * 4);
+      d_offsets_.alloc((nbuck + 1) * 4);
+      d_staged_.alloc((int64_t)nbuck * AGG2_LSLOTS * sizeof(StagedGroup));
+      d_counters_.alloc(16);  // staged_n, lo_n
+      // layout: counts[nbuck+1] | offs[nbuck+1] | counters[2] (8-aligned)
+      pinned_agg2_.alloc(2 * (size_t)(nbuck + 1) * 4 + 64);
+    }
+    hipEvent_t e0, e1;
+    AURON_HIP(hipEventCreate(&e0));
+    AURON_HIP(hipEventCreate(&e1));
+    AURON_HIP(hipEventRecord(e0, stream_));
+    // P1: histogram (+ special-row count in counts[nbuck])
+    AURON_HIP(hipMemsetAsync(d_counts_.get(), 0, (nbuck + 1) * 4, stream_));
+    launch_agg2_hist(keys, kv, chunk, AGG2_NBUCK_LOG2,
+                     d_counts_.get<uint32_t>(),
+                     d_counts_.get<uint32_t>() + nbuck, stream_);
+    uint32_t* h_counts = pinned_agg2_.get<uint32_t>();
+    AURON_HIP(hipMemcpyAsync(h_counts, d_counts_.get(), (nbuck + 1) * 4,
+                             hipMemcpyDeviceToHost, stream_));
+    AURON_HIP(hipStreamSynchronize(stream_));
+    uint32_t special_rows = h_counts[nbuck];
+    // host exclusive scan -> offsets
+    uint32_t* h_offs = h_counts + nbuck + 1;
+    uint32_t acc = 0;
+    for (int i = 0; i < nbuck; i++) {
+      h_offs[i] = acc;
+      acc += h_counts[i];
+    }
+    h_offs[nbuck] = acc;
+    AURON_HIP(hipMemcpyAsync(d_offsets_.get(), h_offs, (nbuck + 1) * 4,
+                             hipMemcpyHostToDevice, stream_));
+    AURON_HIP(hipMemcpyAsync(d_cursors_.get(), h_offs, nbuck * 4,
+                             hipMemcpyHostToDevice, stream_));
+    // P2: scatter + specials
+    launch_agg2_scatter(keys, kv, vals, vv, chunk, AGG2_NBUCK_LOG2,
+                        d_cursors_.get<uint32_t>(), d_bkey_.get<int64_t>(),
+                        d_bval_.get<double>(), d_browv_.get<uint32_t>(),
+                        stream_);
+    if (special_rows)
+      launch_agg2_specials(t_, keys, kv, vals, vv, chunk, row_cursor_, stream_);
+    // A: per-bucket LDS aggregation
+    AURON_HIP(hipMemsetAsync(d_counters_.get(), 0, 16, stream_));
+    launch_agg2_bucket(d_bkey_.get<int64_t>(), d_bval_.get<double>(),
+                       d_browv_.get<uint32_t>(), d_offsets_.get<uint32_t>(),
+                       nbuck, d_staged_.get<StagedGroup>(),
+                       d_counters_.get<unsigned long long>(),
+                       (int64_t)nbuck * AGG2_LSLOTS,
+                       d_lokey_.get<int64_t>(), d_loval_.get<double>(),
+                       d_lorowv_.get<uint32_t>(),
+                       d_counters_.get<unsigned long long>() + 1,
+                       t_.error_flag, stream_);
+    unsigned long long* h_ctr =
+        (unsigned long long*)(pinned_agg2_.get<uint8_t>() +
+                              2 * (size_t)(nbuck + 1) * 4);
+    AURON_HIP(hipMemcpyAsync(h_ctr, d_counters_.get(), 16,
+                             hipMemcpyDeviceToHost, stream_));
+    AURON_HIP(hipStreamSynchronize(stream_));
+    int64_t staged_n = (int64_t)h_ctr[0], lo_n = (int64_t)h_ctr[1];
+    // merge: table inserts bounded by the COUNTED lists
+    ensure_capacity((int64_t)ng_bound_ + staged_n + lo_n);
+    if (staged_n)
+      launch_agg2_merge_groups(t_, d_staged_.get<StagedGroup>(), staged_n,
+                               row_cursor_, stream_);
+    if (lo_n)
+      launch_agg2_leftovers(t_, d_lokey_.get<int64_t>(),
+                            d_loval_.get<double>(), d_lorowv_.get<uint32_t>(),
+                            lo_n, row_cursor_, stream_);
+    AURON_HIP(hipEventRecord(e1, stream_));
+    ev_pairs_.push_back({e0, e1});
+    update_rows_ += chunk;
+    ng_bound_ += (uint64_t)(staged_n + lo_n);
+    DBG("agg.2phase chunk=%lld staged=%lld leftover=%lld special=%u",
+        (long long)chunk, (long long)staged_n, (long long)lo_n, special_rows);
   }
 
   void refresh_ng() {
@@ -729,6 +843,10 @@ class AggOp {
   AggTable t_;
   DevBuf d_slots_, d_special_, d_ng_, d_err_;
   PinnedBuf pinned_meta_;
+  // two-phase scratch (allocated on first large chunk)
+  DevBuf d_bkey_, d_bval_, d_browv_, d_lokey_, d_loval_, d_lorowv_, d_counts_,
+      d_cursors_, d_offsets_, d_staged_, d_counters_;
+  PinnedBuf pinned_agg2_;
   std::vector<DevBatch> held_, skipped_;
 };
 

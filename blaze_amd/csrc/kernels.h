@@ -93,6 +93,41 @@ void launch_slots_init(AggSlot* slots, int64_t n, hipStream_t s);
 // slot holds a distinct key so plain stores after the CAS claim are race-free.
 void launch_agg_rebuild(const AggTable& dst, const AggTable& src, hipStream_t s);
 
+// ---- two-phase aggregation (kernels_agg2.hip) ------------------------------
+// radix-partition rows into buckets whose groups fit in LDS, aggregate each
+// bucket in LDS, merge the counted per-bucket group lists into the table.
+struct StagedGroup {
+  int64_t key;
+  double sum;
+  unsigned long long cnt_first;  // cnt<<32 | chunk-local first_row
+};
+constexpr int AGG2_LSLOTS = 2048;  // LDS table entries per bucket
+
+void launch_agg2_hist(const int64_t* keys, const uint8_t* key_valid, int64_t n,
+                      int nbuck_log2, uint32_t* counts, uint32_t* special_rows,
+                      hipStream_t s);
+void launch_agg2_scatter(const int64_t* keys, const uint8_t* key_valid,
+                         const double* vals, const uint8_t* val_valid,
+                         int64_t n, int nbuck_log2, uint32_t* cursors,
+                         int64_t* out_key, double* out_val, uint32_t* out_rowv,
+                         hipStream_t s);
+void launch_agg2_specials(const AggTable& t, const int64_t* keys,
+                          const uint8_t* key_valid, const double* vals,
+                          const uint8_t* val_valid, int64_t n,
+                          uint64_t row_offset, hipStream_t s);
+void launch_agg2_bucket(const int64_t* bkey, const double* bval,
+                        const uint32_t* browv, const uint32_t* offsets,
+                        int nbuckets, StagedGroup* staged,
+                        unsigned long long* staged_n, int64_t staged_cap,
+                        int64_t* lo_key, double* lo_val, uint32_t* lo_rowv,
+                        unsigned long long* lo_n, uint32_t* error_flag,
+                        hipStream_t s);
+void launch_agg2_merge_groups(const AggTable& t, const StagedGroup* staged,
+                              int64_t n, uint64_t row_offset, hipStream_t s);
+void launch_agg2_leftovers(const AggTable& t, const int64_t* keys,
+                           const double* vals, const uint32_t* rowv, int64_t n,
+                           uint64_t row_offset, hipStream_t s);
+
 // ---- shuffle partition + gather -------------------------------------------
 // per-partition histogram of part_ids (LDS-staged, one global atomic per
 // (block, partition))
