@@ -602,9 +602,14 @@ class AggOp {
       d_lokey_.alloc(AGG2_MAX_CHUNK * 8);
       d_loval_.alloc(AGG2_MAX_CHUNK * 8);
       d_lorowv_.alloc(AGG2_MAX_CHUNK * 4);
-      d_counts_.alloc((nbuck + 1) * 4);
-      d_cursors_.alloc(nbuck * 4);
+      int64_t mat = (int64_t)nbuck << AGG2_GRID_LOG2;
+      d_counts_.alloc((mat + 1) * 4);   // +1: scan total slot
+      d_scanned_.alloc((mat + 1) * 4);
       d_offsets_.alloc((nbuck + 1) * 4);
+      size_t tb = 0;
+      scan_counts_matrix(d_counts_.get<uint32_t>(), d_scanned_.get<uint32_t>(),
+                         mat + 1, nullptr, &tb, stream_);
+      d_scan_tmp_.alloc(tb);
       d_staged_.alloc((int64_t)nbuck * AGG2_LSLOTS * sizeof(StagedGroup));
       d_counters_.alloc(16);  // staged_n, lo_n
       // layout: counts[nbuck+1] | offs[nbuck+1] | counters[2] (8-aligned)
@@ -614,37 +619,34 @@ class AggOp {
     AURON_HIP(hipEventCreate(&e0));
     AURON_HIP(hipEventCreate(&e1));
     AURON_HIP(hipEventRecord(e0, stream_));
-    // P1: histogram (+ special-row count in counts[nbuck])
-    AURON_HIP(hipMemsetAsync(d_counts_.get(), 0, (nbuck + 1) * 4, stream_));
+    // P1: per-block histogram matrix (+ special-row count)
+    int64_t mat = (int64_t)nbuck << AGG2_GRID_LOG2;
+    AURON_HIP(hipMemsetAsync(d_counters_.get(), 0, 16, stream_));
     launch_agg2_hist(keys, kv, chunk, AGG2_NBUCK_LOG2,
                      d_counts_.get<uint32_t>(),
-                     d_counts_.get<uint32_t>() + nbuck, stream_);
-    uint32_t* h_counts = pinned_agg2_.get<uint32_t>();
-    AURON_HIP(hipMemcpyAsync(h_counts, d_counts_.get(), (nbuck + 1) * 4,
+                     (uint32_t*)d_counters_.get<uint8_t>() + 2, stream_);
+    // device exclusive scan over the flat matrix -> per-(block,bucket) bases
+    size_t tb = d_scan_tmp_.size();
+    scan_counts_matrix(d_counts_.get<uint32_t>(), d_scanned_.get<uint32_t>(),
+                       mat + 1, d_scan_tmp_.get(), &tb, stream_);
+    uint32_t* h_meta = pinned_agg2_.get<uint32_t>();
+    AURON_HIP(hipMemcpyAsync(h_meta, (uint32_t*)d_counters_.get<uint8_t>() + 2,
+                             4, hipMemcpyDeviceToHost, stream_));
+    AURON_HIP(hipMemcpyAsync(h_meta + 1, d_scanned_.get<uint32_t>() + mat, 4,
                              hipMemcpyDeviceToHost, stream_));
     AURON_HIP(hipStreamSynchronize(stream_));
-    uint32_t special_rows = h_counts[nbuck];
-    // host exclusive scan -> offsets
-    uint32_t* h_offs = h_counts + nbuck + 1;
-    uint32_t acc = 0;
-    for (int i = 0; i < nbuck; i++) {
-      h_offs[i] = acc;
-      acc += h_counts[i];
-    }
-    h_offs[nbuck] = acc;
-    AURON_HIP(hipMemcpyAsync(d_offsets_.get(), h_offs, (nbuck + 1) * 4,
-                             hipMemcpyHostToDevice, stream_));
-    AURON_HIP(hipMemcpyAsync(d_cursors_.get(), h_offs, nbuck * 4,
-                             hipMemcpyHostToDevice, stream_));
+    uint32_t special_rows = h_meta[0];
+    uint32_t total_rows = h_meta[1];
+    launch_agg2_offsets(d_scanned_.get<uint32_t>(), AGG2_NBUCK_LOG2, total_rows,
+                        d_offsets_.get<uint32_t>(), stream_);
     // P2: scatter + specials
     launch_agg2_scatter(keys, kv, vals, vv, chunk, AGG2_NBUCK_LOG2,
-                        d_cursors_.get<uint32_t>(), d_bkey_.get<int64_t>(),
+                        d_scanned_.get<uint32_t>(), d_bkey_.get<int64_t>(),
                         d_bval_.get<double>(), d_browv_.get<uint32_t>(),
                         stream_);
     if (special_rows)
       launch_agg2_specials(t_, keys, kv, vals, vv, chunk, row_cursor_, stream_);
-    // A: per-bucket LDS aggregation
-    AURON_HIP(hipMemsetAsync(d_counters_.get(), 0, 16, stream_));
+    // A: per-bucket LDS aggregation (counters[0..1] already zeroed above)
     launch_agg2_bucket(d_bkey_.get<int64_t>(), d_bval_.get<double>(),
                        d_browv_.get<uint32_t>(), d_offsets_.get<uint32_t>(),
                        nbuck, d_staged_.get<StagedGroup>(),
@@ -845,7 +847,7 @@ class AggOp {
   PinnedBuf pinned_meta_;
   // two-phase scratch (allocated on first large chunk)
   DevBuf d_bkey_, d_bval_, d_browv_, d_lokey_, d_loval_, d_lorowv_, d_counts_,
-      d_cursors_, d_offsets_, d_staged_, d_counters_;
+      d_scanned_, d_scan_tmp_, d_offsets_, d_staged_, d_counters_;
   PinnedBuf pinned_agg2_;
   std::vector<DevBatch> held_, skipped_;
 };

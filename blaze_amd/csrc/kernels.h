@@ -5,6 +5,8 @@
 
 #include <cstdint>
 
+#include "plan.h"
+
 namespace auron {
 
 // ---- Spark murmur3 (mur.rs:19-87) on device -------------------------------
@@ -93,6 +95,24 @@ void launch_slots_init(AggSlot* slots, int64_t n, hipStream_t s);
 // slot holds a distinct key so plain stores after the CAS claim are race-free.
 void launch_agg_rebuild(const AggTable& dst, const AggTable& src, hipStream_t s);
 
+// ---- filter (filter_exec.rs:174-198, cached_exprs_evaluator.rs:82-95) ------
+// comparison ops (auron-serde/src/lib.rs:70-96 names)
+enum CmpOp : int32_t { CMP_EQ, CMP_NE, CMP_LT, CMP_LE, CMP_GT, CMP_GE };
+// mask[i] &= (col[i] op literal), null compares false; `first` initializes
+// the mask instead of ANDing
+void launch_cmp_lit(DType dt, const void* vals, const uint8_t* valid,
+                    int64_t n, CmpOp op, int64_t lit_i, double lit_f,
+                    uint8_t* mask, bool first, hipStream_t s);
+void launch_is_not_null(const uint8_t* valid, int64_t n, uint8_t* mask,
+                        bool first, bool negate, hipStream_t s);
+// positions of selected rows, stable: needs exclusive scan of mask
+void scan_mask_u8(const uint8_t* mask, uint32_t* positions /* n+1 */, int64_t n,
+                  void* temp, size_t* temp_bytes, hipStream_t s);
+void launch_sel_rows(const uint8_t* mask, const uint32_t* positions, int64_t n,
+                     uint32_t* sel, hipStream_t s);
+void launch_gather_4(const uint8_t* src, const uint32_t* perm, int64_t n,
+                     uint8_t* dst, hipStream_t s);  // 4-byte elements
+
 // ---- two-phase aggregation (kernels_agg2.hip) ------------------------------
 // radix-partition rows into buckets whose groups fit in LDS, aggregate each
 // bucket in LDS, merge the counted per-bucket group lists into the table.
@@ -103,12 +123,20 @@ struct StagedGroup {
 };
 constexpr int AGG2_LSLOTS = 2048;  // LDS table entries per bucket
 
+constexpr int AGG2_GRID_LOG2 = 11;  // hist/scatter grid: 2048 blocks
+// per-BLOCK bucket histogram into the bucket-major counts matrix
+// [nbuck << AGG2_GRID_LOG2]
 void launch_agg2_hist(const int64_t* keys, const uint8_t* key_valid, int64_t n,
-                      int nbuck_log2, uint32_t* counts, uint32_t* special_rows,
-                      hipStream_t s);
+                      int nbuck_log2, uint32_t* counts_matrix,
+                      uint32_t* special_rows, hipStream_t s);
+// exclusive scan over the flat counts matrix -> per-(block,bucket) bases
+void scan_counts_matrix(const uint32_t* counts, uint32_t* scanned, int64_t n,
+                        void* temp, size_t* temp_bytes, hipStream_t s);
+void launch_agg2_offsets(const uint32_t* scanned, int nbuck_log2,
+                         uint32_t total, uint32_t* offsets, hipStream_t s);
 void launch_agg2_scatter(const int64_t* keys, const uint8_t* key_valid,
                          const double* vals, const uint8_t* val_valid,
-                         int64_t n, int nbuck_log2, uint32_t* cursors,
+                         int64_t n, int nbuck_log2, const uint32_t* scanned,
                          int64_t* out_key, double* out_val, uint32_t* out_rowv,
                          hipStream_t s);
 void launch_agg2_specials(const AggTable& t, const int64_t* keys,

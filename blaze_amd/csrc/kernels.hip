@@ -18,6 +18,7 @@
 #include <hip/hip_runtime.h>
 
 #include <rocprim/device/device_radix_sort.hpp>
+#include <rocprim/device/device_scan.hpp>
 
 #include <stdexcept>
 #include <string>
@@ -675,6 +676,115 @@ void launch_gather_bytes(const uint8_t* src_data, const int32_t* src_offsets,
   hipLaunchKernelGGL(k_gather_bytes, dim3(grid_for(n * 64)), dim3(BLOCK), 0, s,
                      src_data, src_offsets, perm, dst_offsets, n, dst_data);
   check_launch("k_gather_bytes");
+}
+
+// ---- filter ----------------------------------------------------------------
+template <typename T>
+__global__ void k_cmp_lit(const T* __restrict__ vals,
+                          const uint8_t* __restrict__ valid, int64_t n,
+                          int32_t op, T lit, uint8_t* __restrict__ mask,
+                          int first) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    bool v = !valid || bit_get_dev(valid, i);
+    bool r = false;
+    if (v) {
+      T x = vals[i];
+      switch (op) {
+        case CMP_EQ: r = x == lit; break;
+        case CMP_NE: r = x != lit; break;
+        case CMP_LT: r = x < lit; break;
+        case CMP_LE: r = x <= lit; break;
+        case CMP_GT: r = x > lit; break;
+        case CMP_GE: r = x >= lit; break;
+      }
+    }
+    mask[i] = first ? (uint8_t)r : (uint8_t)(mask[i] & (uint8_t)r);
+  }
+}
+
+__global__ void k_is_not_null(const uint8_t* __restrict__ valid, int64_t n,
+                              uint8_t* __restrict__ mask, int first,
+                              int negate) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    bool v = !valid || bit_get_dev(valid, i);
+    bool r = negate ? !v : v;
+    mask[i] = first ? (uint8_t)r : (uint8_t)(mask[i] & (uint8_t)r);
+  }
+}
+
+__global__ void k_sel_rows(const uint8_t* __restrict__ mask,
+                           const uint32_t* __restrict__ positions, int64_t n,
+                           uint32_t* __restrict__ sel) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    if (mask[i]) sel[positions[i]] = (uint32_t)i;
+  }
+}
+
+__global__ void k_gather_4(const uint8_t* __restrict__ src,
+                           const uint32_t* __restrict__ perm, int64_t n,
+                           uint8_t* __restrict__ dst) {
+  const uint32_t* s32 = (const uint32_t*)src;
+  uint32_t* d32 = (uint32_t*)dst;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    d32[i] = s32[perm[i]];
+}
+
+void launch_cmp_lit(DType dt, const void* vals, const uint8_t* valid,
+                    int64_t n, CmpOp op, int64_t lit_i, double lit_f,
+                    uint8_t* mask, bool first, hipStream_t s) {
+  switch (dt) {
+    case DType::Int32:
+      hipLaunchKernelGGL(k_cmp_lit<int32_t>, dim3(grid_for(n)), dim3(BLOCK), 0,
+                         s, (const int32_t*)vals, valid, n, op, (int32_t)lit_i,
+                         mask, first);
+      break;
+    case DType::Int64:
+      hipLaunchKernelGGL(k_cmp_lit<int64_t>, dim3(grid_for(n)), dim3(BLOCK), 0,
+                         s, (const int64_t*)vals, valid, n, op, lit_i, mask,
+                         first);
+      break;
+    case DType::Float64:
+      hipLaunchKernelGGL(k_cmp_lit<double>, dim3(grid_for(n)), dim3(BLOCK), 0,
+                         s, (const double*)vals, valid, n, op, lit_f, mask,
+                         first);
+      break;
+    default:
+      throw std::runtime_error("cmp_lit: unsupported dtype");
+  }
+  check_launch("k_cmp_lit");
+}
+
+void launch_is_not_null(const uint8_t* valid, int64_t n, uint8_t* mask,
+                        bool first, bool negate, hipStream_t s) {
+  hipLaunchKernelGGL(k_is_not_null, dim3(grid_for(n)), dim3(BLOCK), 0, s, valid,
+                     n, mask, first, negate);
+  check_launch("k_is_not_null");
+}
+
+void scan_mask_u8(const uint8_t* mask, uint32_t* positions, int64_t n,
+                  void* temp, size_t* temp_bytes, hipStream_t s) {
+  hipError_t e = rocprim::exclusive_scan(
+      temp, *temp_bytes, mask, positions, 0u, (size_t)(n + 1),
+      rocprim::plus<uint32_t>(), s);
+  if (e != hipSuccess) abort();
+}
+
+void launch_sel_rows(const uint8_t* mask, const uint32_t* positions, int64_t n,
+                     uint32_t* sel, hipStream_t s) {
+  hipLaunchKernelGGL(k_sel_rows, dim3(grid_for(n)), dim3(BLOCK), 0, s, mask,
+                     positions, n, sel);
+  check_launch("k_sel_rows");
+}
+
+void launch_gather_4(const uint8_t* src, const uint32_t* perm, int64_t n,
+                     uint8_t* dst, hipStream_t s) {
+  hipLaunchKernelGGL(k_gather_4, dim3(grid_for(n)), dim3(BLOCK), 0, s, src,
+                     perm, n, dst);
+  check_launch("k_gather_4");
 }
 
 __global__ void k_byte_transpose(const uint8_t* __restrict__ in,

@@ -8,6 +8,8 @@
 // 20 B/row bucket read; group-table traffic becomes per-GROUP, not per-row.
 #include <hip/hip_runtime.h>
 
+#include <rocprim/device/device_scan.hpp>
+
 #include <stdexcept>
 #include <string>
 
@@ -51,10 +53,16 @@ __device__ __forceinline__ uint32_t bucket_of(int64_t key, int nbuck_log2) {
 
 }  // namespace
 
-// ---- phase P1: per-bucket histogram (reads keys only) ----------------------
+// ---- phase P1: per-BLOCK per-bucket histogram (reads keys only) ------------
+// counts_matrix is bucket-major: counts[(b << GRID_LOG2) | blockIdx]. An
+// exclusive scan over the flat matrix then gives every block a private,
+// contiguous output range per bucket — the scatter needs NO global atomics.
+static constexpr int GRID_LOG2 = 11;  // 2048 blocks, fixed for hist+scatter
+
 __global__ void k_agg2_hist(const int64_t* __restrict__ keys,
                             const uint8_t* __restrict__ key_valid, int64_t n,
-                            int nbuck_log2, uint32_t* __restrict__ counts,
+                            int nbuck_log2,
+                            uint32_t* __restrict__ counts_matrix,
                             uint32_t* __restrict__ special_rows) {
   extern __shared__ uint32_t lds_hist[];
   const uint32_t nbuck = 1u << nbuck_log2;
@@ -73,29 +81,44 @@ __global__ void k_agg2_hist(const int64_t* __restrict__ keys,
   }
   __syncthreads();
   for (uint32_t b = threadIdx.x; b < nbuck; b += blockDim.x)
-    if (lds_hist[b]) atomicAdd(&counts[b], lds_hist[b]);
+    counts_matrix[((size_t)b << GRID_LOG2) | blockIdx.x] = lds_hist[b];
   if (special) atomicAdd(special_rows, special);
 }
 
-// ---- phase P2: scatter rows into bucket-major SoA --------------------------
+// offsets[b] = scanned[(b << GRID_LOG2) | 0]; offsets[nbuck] = total
+__global__ void k_agg2_offsets(const uint32_t* __restrict__ scanned,
+                               int nbuck_log2, uint32_t total,
+                               uint32_t* __restrict__ offsets) {
+  int nbuck = 1 << nbuck_log2;
+  for (int b = (int)(blockIdx.x * blockDim.x + threadIdx.x); b <= nbuck;
+       b += (int)(gridDim.x * blockDim.x))
+    offsets[b] = (b == nbuck) ? total : scanned[(size_t)b << GRID_LOG2];
+}
+
+// ---- phase P2: scatter into per-(block,bucket) reserved ranges -------------
+// Must use the SAME grid geometry and row traversal as k_agg2_hist.
 // rowv packs the chunk-local row (bit 0..30) + value-validity (bit 31).
 __global__ void k_agg2_scatter(const int64_t* __restrict__ keys,
                                const uint8_t* __restrict__ key_valid,
                                const double* __restrict__ vals,
                                const uint8_t* __restrict__ val_valid, int64_t n,
                                int nbuck_log2,
-                               uint32_t* __restrict__ cursors /* [nbuck], init
-                                                                 to offsets */,
+                               const uint32_t* __restrict__ scanned,
                                int64_t* __restrict__ out_key,
                                double* __restrict__ out_val,
                                uint32_t* __restrict__ out_rowv) {
+  extern __shared__ uint32_t lds_cursor[];
+  const uint32_t nbuck = 1u << nbuck_log2;
+  for (uint32_t b = threadIdx.x; b < nbuck; b += blockDim.x)
+    lds_cursor[b] = scanned[((size_t)b << GRID_LOG2) | blockIdx.x];
+  __syncthreads();
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     bool knull = key_valid && !bit_get2(key_valid, i);
     int64_t k = keys[i];
     if (knull || k == KEY_EMPTY2) continue;  // specials handled separately
     uint32_t b = bucket_of(k, nbuck_log2);
-    uint32_t pos = atomicAdd(&cursors[b], 1u);
+    uint32_t pos = atomicAdd(&lds_cursor[b], 1u);  // block-local LDS cursor
     bool vvalid = !val_valid || bit_get2(val_valid, i);
     out_key[pos] = k;
     out_val[pos] = vals[i];
@@ -324,21 +347,38 @@ void launch_agg2_leftovers(const AggTable& t, const int64_t* keys,
 }
 
 void launch_agg2_hist(const int64_t* keys, const uint8_t* key_valid, int64_t n,
-                      int nbuck_log2, uint32_t* counts, uint32_t* special_rows,
-                      hipStream_t s) {
+                      int nbuck_log2, uint32_t* counts_matrix,
+                      uint32_t* special_rows, hipStream_t s) {
   size_t lds = (size_t)(1u << nbuck_log2) * 4;
-  hipLaunchKernelGGL(k_agg2_hist, dim3(grid2(n)), dim3(BLOCK), lds, s, keys,
-                     key_valid, n, nbuck_log2, counts, special_rows);
+  hipLaunchKernelGGL(k_agg2_hist, dim3(1 << GRID_LOG2), dim3(BLOCK), lds, s,
+                     keys, key_valid, n, nbuck_log2, counts_matrix,
+                     special_rows);
   check_launch2("k_agg2_hist");
+}
+
+void launch_agg2_offsets(const uint32_t* scanned, int nbuck_log2,
+                         uint32_t total, uint32_t* offsets, hipStream_t s) {
+  hipLaunchKernelGGL(k_agg2_offsets, dim3(8), dim3(BLOCK), 0, s, scanned,
+                     nbuck_log2, total, offsets);
+  check_launch2("k_agg2_offsets");
+}
+
+void scan_counts_matrix(const uint32_t* counts, uint32_t* scanned, int64_t n,
+                        void* temp, size_t* temp_bytes, hipStream_t s) {
+  hipError_t e = rocprim::exclusive_scan(temp, *temp_bytes, counts, scanned,
+                                         0u, (size_t)n,
+                                         rocprim::plus<uint32_t>(), s);
+  if (e != hipSuccess) abort();
 }
 
 void launch_agg2_scatter(const int64_t* keys, const uint8_t* key_valid,
                          const double* vals, const uint8_t* val_valid,
-                         int64_t n, int nbuck_log2, uint32_t* cursors,
+                         int64_t n, int nbuck_log2, const uint32_t* scanned,
                          int64_t* out_key, double* out_val, uint32_t* out_rowv,
                          hipStream_t s) {
-  hipLaunchKernelGGL(k_agg2_scatter, dim3(grid2(n)), dim3(BLOCK), 0, s, keys,
-                     key_valid, vals, val_valid, n, nbuck_log2, cursors,
+  size_t lds = (size_t)(1u << nbuck_log2) * 4;
+  hipLaunchKernelGGL(k_agg2_scatter, dim3(1 << GRID_LOG2), dim3(BLOCK), lds, s,
+                     keys, key_valid, vals, val_valid, n, nbuck_log2, scanned,
                      out_key, out_val, out_rowv);
   check_launch2("k_agg2_scatter");
 }

@@ -53,14 +53,21 @@ struct Schema {
 
 // PhysicalExprNode subset (auron.proto:58-121)
 struct Expr {
-  enum Kind { Column, AggExpr, Literal } kind = Column;
-  // Column (auron.proto:489-492)
+  enum Kind {
+    Column,     // PhysicalColumn (auron.proto:489-492)
+    AggExpr,    // PhysicalAggExprNode (auron.proto:143-148)
+    Literal,    // ScalarValue{ipc_bytes} (auron.proto:824-826)
+    BinaryExpr, // PhysicalBinaryExprNode (auron.proto:172-176)
+    IsNotNull,  // PhysicalIsNotNull (auron.proto:161-163)
+    IsNull,     // PhysicalIsNull (auron.proto:155-157)
+  } kind = Column;
   std::string col_name;
   uint32_t col_index = 0;
-  // AggExpr (PhysicalAggExprNode auron.proto:143-148)
   int32_t agg_function = -1;  // AggFunction enum: MIN=0 MAX=1 SUM=2 AVG=3 COUNT=4
-  std::vector<Expr> children;
+  std::vector<Expr> children;  // agg args / binary l,r / null-check operand
   DType return_type = DType::Unsupported;
+  std::string op;              // binary op name (auron-serde/src/lib.rs:70-96)
+  std::vector<uint8_t> literal_ipc;  // raw ScalarValue.ipc_bytes
 };
 
 // AggFunction enum values (auron.proto:128-141)
@@ -114,12 +121,33 @@ struct FFIReaderNode {
   std::string resource_id;       // = 3
 };
 
+// FilterExecNode (auron.proto:363-366): predicates are ANDed
+struct FilterNode {
+  std::unique_ptr<PlanNode> input;  // = 1
+  std::vector<Expr> predicates;     // = 2
+};
+
+// ProjectionExecNode (auron.proto:505-510)
+struct ProjectionNode {
+  std::unique_ptr<PlanNode> input;    // = 1
+  std::vector<Expr> exprs;            // = 2
+  std::vector<std::string> names;     // = 3
+};
+
 // PhysicalPlanNode oneof (auron.proto:27-55)
 struct PlanNode {
-  enum Kind { ShuffleWriter = 2, Agg = 16, FFIReader = 18 } kind;
+  enum Kind {
+    ShuffleWriter = 2,
+    Projection = 6,
+    Filter = 8,
+    Agg = 16,
+    FFIReader = 18,
+  } kind;
   std::unique_ptr<ShuffleWriterNode> shuffle_writer;
   std::unique_ptr<AggNode> agg;
   std::unique_ptr<FFIReaderNode> ffi_reader;
+  std::unique_ptr<FilterNode> filter;
+  std::unique_ptr<ProjectionNode> projection;
 };
 
 // TaskDefinition (auron.proto:735-740) + PartitionId (:729-733)

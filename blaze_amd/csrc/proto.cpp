@@ -144,13 +144,49 @@ Expr decode_expr(Reader r, std::string* err) {
     if (n == 0) break;
     switch (n) {
       case 1: return decode_column(r.sub());
-      case 2: {  // ScalarValue literal — accepted only as agg null-arg
+      case 2: {  // ScalarValue{ipc_bytes = 1} (auron.proto:824-826)
         Expr e;
         e.kind = Expr::Literal;
-        r.sub();
+        Reader s = r.sub();
+        while (true) {
+          auto [sf, sw] = s.tag();
+          if (sf == 0) break;
+          if (sf == 1 && sw == 2) {
+            Reader bytes = s.sub();
+            e.literal_ipc.assign(bytes.p, bytes.end);
+          } else {
+            s.skip(sw);
+          }
+        }
+        return e;
+      }
+      case 4: {  // PhysicalBinaryExprNode{l=1, r=2, op=3}
+        Expr e;
+        e.kind = Expr::BinaryExpr;
+        Reader s = r.sub();
+        while (true) {
+          auto [sf, sw] = s.tag();
+          if (sf == 0) break;
+          if (sf == 1) e.children.push_back(decode_expr(s.sub(), err));
+          else if (sf == 2) e.children.push_back(decode_expr(s.sub(), err));
+          else if (sf == 3) e.op = s.str();
+          else s.skip(sw);
+        }
         return e;
       }
       case 5: return decode_agg_expr(r.sub(), err);
+      case 6: case 7: {  // PhysicalIsNull / PhysicalIsNotNull {expr = 1}
+        Expr e;
+        e.kind = (n == 7) ? Expr::IsNotNull : Expr::IsNull;
+        Reader s = r.sub();
+        while (true) {
+          auto [sf, sw] = s.tag();
+          if (sf == 0) break;
+          if (sf == 1) e.children.push_back(decode_expr(s.sub(), err));
+          else s.skip(sw);
+        }
+        return e;
+      }
       default:
         if (err->empty())
           *err = "unsupported PhysicalExprNode field " + std::to_string(n);
@@ -271,6 +307,31 @@ std::unique_ptr<FFIReaderNode> decode_ffi_reader(Reader r) {
   return n;
 }
 
+std::unique_ptr<FilterNode> decode_filter(Reader r, std::string* err) {
+  auto n = std::make_unique<FilterNode>();
+  while (true) {
+    auto [f, w] = r.tag();
+    if (f == 0) break;
+    if (f == 1) n->input = decode_plan_node(r.sub(), err);
+    else if (f == 2) n->predicates.push_back(decode_expr(r.sub(), err));
+    else r.skip(w);
+  }
+  return n;
+}
+
+std::unique_ptr<ProjectionNode> decode_projection(Reader r, std::string* err) {
+  auto n = std::make_unique<ProjectionNode>();
+  while (true) {
+    auto [f, w] = r.tag();
+    if (f == 0) break;
+    if (f == 1) n->input = decode_plan_node(r.sub(), err);
+    else if (f == 2) n->exprs.push_back(decode_expr(r.sub(), err));
+    else if (f == 3) n->names.push_back(r.str());
+    else r.skip(w);
+  }
+  return n;
+}
+
 std::unique_ptr<PlanNode> decode_plan_node(Reader r, std::string* err) {
   auto node = std::make_unique<PlanNode>();
   while (true) {
@@ -280,6 +341,14 @@ std::unique_ptr<PlanNode> decode_plan_node(Reader r, std::string* err) {
       case 2:  // ShuffleWriterExecNode
         node->kind = PlanNode::ShuffleWriter;
         node->shuffle_writer = decode_shuffle_writer(r.sub(), err);
+        return node;
+      case 6:  // ProjectionExecNode
+        node->kind = PlanNode::Projection;
+        node->projection = decode_projection(r.sub(), err);
+        return node;
+      case 8:  // FilterExecNode
+        node->kind = PlanNode::Filter;
+        node->filter = decode_filter(r.sub(), err);
         return node;
       case 16:  // AggExecNode
         node->kind = PlanNode::Agg;
