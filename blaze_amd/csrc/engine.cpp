@@ -21,6 +21,7 @@
 #include "dev.h"
 #include "kernels.h"
 #include "plan.h"
+#include "ipc_scalar.h"
 #include "serde_host.h"
 
 namespace auron {
@@ -1102,6 +1103,225 @@ class ShuffleOp {
   std::vector<DevBatch> staged_;
 };
 
+// --------------------------------------------------------------- FilterOp --
+// FilterExec (filter_exec.rs:174-198): ANDed predicates -> selection mask ->
+// stable compaction. Predicate forms on the hot path: Column <op> Literal,
+// IsNotNull/IsNull(Column).
+class FilterOp {
+ public:
+  FilterOp(const FilterNode& node, hipStream_t stream) : stream_(stream) {
+    for (const Expr& e : node.predicates) compile(e);
+    if (preds_.empty()) FAIL("FilterExec without usable predicates");
+  }
+
+  DevBatch apply(DevBatch&& in) {
+    int64_t n = in.num_rows;
+    if (n == 0) return std::move(in);
+    DevBuf mask(n);
+    for (size_t pi = 0; pi < preds_.size(); pi++) {
+      const Pred& p = preds_[pi];
+      const DevColumn& c = in.cols.at(p.col);
+      bool first = pi == 0;
+      if (p.kind == Pred::NotNull || p.kind == Pred::Null) {
+        launch_is_not_null(c.validity, n, mask.get<uint8_t>(), first,
+                           p.kind == Pred::Null, stream_);
+      } else {
+        if (c.dt != p.dtype) FAIL("filter literal/column dtype mismatch");
+        launch_cmp_lit(c.dt, c.values, c.validity, n, p.op, p.lit_i, p.lit_f,
+                       mask.get<uint8_t>(), first, stream_);
+      }
+    }
+    // stable compaction: positions = exclusive scan of mask
+    DevBuf positions((n + 1) * 4);
+    size_t tb = 0;
+    scan_mask_u8(mask.get<uint8_t>(), positions.get<uint32_t>(), n, nullptr,
+                 &tb, stream_);
+    if (tb > scan_tmp_.size()) scan_tmp_.alloc(tb);
+    scan_mask_u8(mask.get<uint8_t>(), positions.get<uint32_t>(), n,
+                 scan_tmp_.get(), &tb, stream_);
+    if (!pinned_.get()) pinned_.alloc(8);
+    uint32_t* h_cnt = pinned_.get<uint32_t>();
+    AURON_HIP(hipMemcpyAsync(h_cnt, positions.get<uint32_t>() + n, 4,
+                             hipMemcpyDeviceToHost, stream_));
+    AURON_HIP(hipStreamSynchronize(stream_));
+    int64_t m = *h_cnt;
+    DevBatch out;
+    out.num_rows = m;
+    if (m == 0) {
+      for (auto& c : in.cols) {
+        DevColumn oc;
+        oc.dt = c.dt;
+        oc.len = 0;
+        out.cols.push_back(std::move(oc));
+      }
+      return out;
+    }
+    DevBuf sel(m * 4);
+    launch_sel_rows(mask.get<uint8_t>(), positions.get<uint32_t>(), n,
+                    sel.get<uint32_t>(), stream_);
+    for (auto& c : in.cols) {
+      out.cols.push_back(gather_col(c, sel.get<uint32_t>(), m));
+    }
+    held_.push_back(std::move(in));  // borrowed inputs stay alive
+    return out;
+  }
+
+ private:
+  struct Pred {
+    enum Kind { CmpLit, NotNull, Null } kind = CmpLit;
+    uint32_t col = 0;
+    CmpOp op = CMP_EQ;
+    DType dtype = DType::Unsupported;
+    int64_t lit_i = 0;
+    double lit_f = 0;
+  };
+
+  void compile(const Expr& e) {
+    if (e.kind == Expr::IsNotNull || e.kind == Expr::IsNull) {
+      const Expr& c = e.children.at(0);
+      if (c.kind != Expr::Column) FAIL("null-check operand must be a Column");
+      Pred p;
+      p.kind = e.kind == Expr::IsNull ? Pred::Null : Pred::NotNull;
+      p.col = c.col_index;
+      preds_.push_back(p);
+      return;
+    }
+    if (e.kind != Expr::BinaryExpr || e.children.size() != 2)
+      FAIL("unsupported filter predicate");
+    const Expr& l = e.children[0];
+    const Expr& r = e.children[1];
+    if (e.op == "And") {  // nested AND: flatten
+      compile(l);
+      compile(r);
+      return;
+    }
+    CmpOp op;
+    if (e.op == "Eq") op = CMP_EQ;
+    else if (e.op == "NotEq") op = CMP_NE;
+    else if (e.op == "Lt") op = CMP_LT;
+    else if (e.op == "LtEq") op = CMP_LE;
+    else if (e.op == "Gt") op = CMP_GT;
+    else if (e.op == "GtEq") op = CMP_GE;
+    else FAIL("unsupported filter operator: " + e.op);
+    const Expr* col = nullptr;
+    const Expr* lit = nullptr;
+    if (l.kind == Expr::Column && r.kind == Expr::Literal) {
+      col = &l;
+      lit = &r;
+    } else if (l.kind == Expr::Literal && r.kind == Expr::Column) {
+      col = &r;
+      lit = &l;
+      // flip: lit OP col == col FLIP(OP) lit
+      op = op == CMP_LT ? CMP_GT : op == CMP_LE ? CMP_GE
+           : op == CMP_GT ? CMP_LT : op == CMP_GE ? CMP_LE : op;
+    } else {
+      FAIL("filter predicate must be Column-vs-Literal on this path");
+    }
+    ScalarLit s;
+    std::string err;
+    if (!decode_ipc_scalar(lit->literal_ipc.data(), lit->literal_ipc.size(),
+                           &s, &err))
+      FAIL("literal decode: " + err);
+    Pred p;
+    p.kind = Pred::CmpLit;
+    p.col = col->col_index;
+    p.op = op;
+    p.dtype = s.dtype;
+    p.lit_i = s.i64;
+    p.lit_f = s.f64;
+    if (s.is_null) FAIL("null literal comparison unsupported");
+    preds_.push_back(p);
+  }
+
+  DevColumn gather_col(const DevColumn& c, const uint32_t* sel, int64_t m) {
+    DevColumn out;
+    out.dt = c.dt;
+    out.len = m;
+    size_t w = dtype_width(c.dt);
+    if (c.dt == DType::Binary || c.dt == DType::Utf8) {
+      DevBuf lens(m * 4);
+      launch_gather_lens(c.offsets, sel, m, lens.get<int32_t>(), stream_);
+      std::vector<int32_t> h_lens(m);
+      AURON_HIP(hipMemcpyAsync(h_lens.data(), lens.get(), m * 4,
+                               hipMemcpyDeviceToHost, stream_));
+      AURON_HIP(hipStreamSynchronize(stream_));
+      std::vector<int32_t> h_offs(m + 1, 0);
+      for (int64_t i = 0; i < m; i++) h_offs[i + 1] = h_offs[i] + h_lens[i];
+      out.own_offsets.alloc((m + 1) * 4);
+      AURON_HIP(hipMemcpyAsync(out.own_offsets.get(), h_offs.data(),
+                               (m + 1) * 4, hipMemcpyHostToDevice, stream_));
+      out.offsets = out.own_offsets.get<int32_t>();
+      out.data_len = h_offs[m];
+      out.own_values.alloc(out.data_len ? out.data_len : 1);
+      launch_gather_bytes((const uint8_t*)c.values, c.offsets, sel,
+                          out.own_offsets.get<int32_t>(), m,
+                          out.own_values.get<uint8_t>(), stream_);
+      out.values = out.own_values.get();
+    } else if (w == 8) {
+      out.own_values.alloc(m * 8);
+      launch_gather_8((const uint8_t*)c.values, sel, m,
+                      out.own_values.get<uint8_t>(), stream_);
+      out.values = out.own_values.get();
+    } else if (w == 4) {
+      out.own_values.alloc(m * 4);
+      launch_gather_4((const uint8_t*)c.values, sel, m,
+                      out.own_values.get<uint8_t>(), stream_);
+      out.values = out.own_values.get();
+    } else {
+      FAIL("filter gather: unsupported column width");
+    }
+    if (c.validity) {
+      out.own_validity.alloc((m + 7) / 8);
+      launch_gather_bits(c.validity, sel, m, out.own_validity.get<uint8_t>(),
+                         stream_);
+      out.validity = out.own_validity.get<uint8_t>();
+    }
+    return out;
+  }
+
+  hipStream_t stream_;
+  std::vector<Pred> preds_;
+  DevBuf scan_tmp_;
+  PinnedBuf pinned_;
+  std::vector<DevBatch> held_;
+};
+
+// -------------------------------------------------------------- ProjectOp --
+// ProjectionExec (project_exec.rs): Column-only projection on this path
+// (expression evaluation breadth is out of hot-path scope and fails loudly).
+class ProjectOp {
+ public:
+  explicit ProjectOp(const ProjectionNode& node) {
+    for (const Expr& e : node.exprs) {
+      if (e.kind != Expr::Column)
+        FAIL("ProjectionExec: only Column exprs on the hot path");
+      cols_.push_back(e.col_index);
+    }
+  }
+
+  DevBatch apply(DevBatch&& in) {
+    DevBatch out;
+    out.num_rows = in.num_rows;
+    for (uint32_t ci : cols_) {
+      DevColumn& src = in.cols.at(ci);
+      DevColumn view;
+      view.dt = src.dt;
+      view.len = src.len;
+      view.values = src.values;
+      view.validity = src.validity;
+      view.offsets = src.offsets;
+      view.data_len = src.data_len;
+      out.cols.push_back(std::move(view));
+    }
+    held_.push_back(std::move(in));  // owner of the viewed buffers
+    return out;
+  }
+
+ private:
+  std::vector<uint32_t> cols_;
+  std::vector<DevBatch> held_;
+};
+
 // ---------------------------------------------------------------- Runtime --
 struct Runtime {
   std::unique_ptr<TaskDefinition> td;
@@ -1110,6 +1330,14 @@ struct Runtime {
   bool started = false;
   bool schema_sent = false;
   std::string error;
+  struct Stage {
+    enum Kind { AggS, ShuffleS, FilterS, ProjectS } kind = AggS;
+    std::unique_ptr<AggOp> agg;
+    std::unique_ptr<ShuffleOp> shuffle;
+    std::unique_ptr<FilterOp> filter;
+    std::unique_ptr<ProjectOp> project;
+  };
+  std::vector<Stage> stages_;
   std::vector<OutField> out_fields;
   std::vector<std::pair<int64_t, std::vector<HostOutCol>>> outputs;
   size_t emit_idx = 0;
@@ -1136,6 +1364,8 @@ struct Runtime {
       switch (p->kind) {
         case PlanNode::ShuffleWriter: p = p->shuffle_writer->input.get(); break;
         case PlanNode::Agg: p = p->agg->input.get(); break;
+        case PlanNode::Filter: p = p->filter->input.get(); break;
+        case PlanNode::Projection: p = p->projection->input.get(); break;
         case PlanNode::FFIReader: p = nullptr; break;
       }
     }
@@ -1145,20 +1375,35 @@ struct Runtime {
       FAIL("plan leaf must be FFIReaderExec on this path");
     const FFIReaderNode& reader = *leaf->ffi_reader;
 
-    // middle ops (leaf-1 ... root)
-    std::vector<std::unique_ptr<AggOp>> aggs;
-    std::unique_ptr<ShuffleOp> shuffle;
+    // middle ops (leaf-1 ... root), in execution order
+    stages_.clear();
     for (auto it = chain.rbegin() + 1; it != chain.rend(); ++it) {
       const PlanNode* node = *it;
-      if (node->kind == PlanNode::Agg) {
-        if (shuffle) FAIL("operator above ShuffleWriter unsupported");
-        aggs.push_back(std::make_unique<AggOp>(*node->agg, conf, stream));
-      } else if (node->kind == PlanNode::ShuffleWriter) {
-        shuffle = std::make_unique<ShuffleOp>(*node->shuffle_writer, conf,
-                                              stream);
-      } else {
-        FAIL("unsupported operator in plan chain");
+      Stage st;
+      switch (node->kind) {
+        case PlanNode::Agg:
+          st.kind = Stage::AggS;
+          st.agg = std::make_unique<AggOp>(*node->agg, conf, stream);
+          break;
+        case PlanNode::ShuffleWriter:
+          st.kind = Stage::ShuffleS;
+          st.shuffle =
+              std::make_unique<ShuffleOp>(*node->shuffle_writer, conf, stream);
+          break;
+        case PlanNode::Filter:
+          st.kind = Stage::FilterS;
+          st.filter = std::make_unique<FilterOp>(*node->filter, stream);
+          break;
+        case PlanNode::Projection:
+          st.kind = Stage::ProjectS;
+          st.project = std::make_unique<ProjectOp>(*node->projection);
+          break;
+        default:
+          FAIL("unsupported operator in plan chain");
       }
+      if (!stages_.empty() && stages_.back().kind == Stage::ShuffleS)
+        FAIL("operator above ShuffleWriter unsupported");
+      stages_.push_back(std::move(st));
     }
 
     // pump input through agg chain
@@ -1186,32 +1431,40 @@ struct Runtime {
       }
       if (sch.release) sch.release(&sch);
       input_rows += b.num_rows;
-      feed(aggs, shuffle.get(), 0, std::move(b));
+      feed(0, std::move(b));
     }
-    // drain chain
-    for (size_t i = 0; i < aggs.size(); i++) {
-      auto outs = aggs[i]->finish();
-      for (auto& ob : outs) {
-        if (i + 1 < aggs.size() || shuffle) {
-          DevBatch b = host_out_to_dev(ob, aggs[i]->output_fields());
-          feed(aggs, shuffle.get(), i + 1, std::move(b));
-        } else {
-          outputs.push_back(std::move(ob));
+    // drain chain: pipeline breakers emit, transforms pass through
+    AggOp* last_agg = nullptr;
+    AggOp* first_agg = nullptr;
+    bool has_shuffle = false;
+    for (size_t i = 0; i < stages_.size(); i++) {
+      Stage& st = stages_[i];
+      if (st.kind == Stage::AggS) {
+        if (!first_agg) first_agg = st.agg.get();
+        last_agg = st.agg.get();
+        auto outs = st.agg->finish();
+        bool terminal = i + 1 == stages_.size();
+        for (auto& ob : outs) {
+          if (terminal) {
+            outputs.push_back(std::move(ob));
+          } else {
+            DevBatch b = host_out_to_dev(ob, st.agg->output_fields());
+            feed(i + 1, std::move(b));
+          }
         }
+        if (terminal) out_fields = st.agg->output_fields();
+      } else if (st.kind == Stage::ShuffleS) {
+        st.shuffle->finish();
+        has_shuffle = true;
+        out_fields.clear();
       }
     }
-    if (shuffle) {
-      shuffle->finish();
-      out_fields.clear();
-    } else if (!aggs.empty()) {
-      out_fields = aggs.back()->output_fields();
-    } else {
-      FAIL("plan without Agg/ShuffleWriter unsupported");
-    }
-    if (!aggs.empty()) {
-      metrics["num_groups"] = (int64_t)aggs.back()->num_groups_host();
-      metrics["agg_update_ns"] = aggs.front()->update_ns_;
-      metrics["agg_update_rows"] = aggs.front()->update_rows_;
+    if (!has_shuffle && !last_agg)
+      FAIL("plan without Agg/ShuffleWriter sink unsupported");
+    if (last_agg) {
+      metrics["num_groups"] = (int64_t)last_agg->num_groups_host();
+      metrics["agg_update_ns"] = first_agg->update_ns_;
+      metrics["agg_update_rows"] = first_agg->update_rows_;
     }
     metrics["input_rows"] = input_rows;
     int64_t out_rows = 0;
@@ -1223,15 +1476,27 @@ struct Runtime {
             .count();
   }
 
-  void feed(std::vector<std::unique_ptr<AggOp>>& aggs, ShuffleOp* shuffle,
-            size_t idx, DevBatch&& b) {
-    if (idx < aggs.size()) {
-      aggs[idx]->consume(std::move(b));
-    } else if (shuffle) {
-      shuffle->consume(std::move(b));
-    } else {
-      FAIL("batch fed past end of chain");
+  void feed(size_t idx, DevBatch&& b) {
+    while (idx < stages_.size()) {
+      Stage& st = stages_[idx];
+      switch (st.kind) {
+        case Stage::AggS:
+          st.agg->consume(std::move(b));
+          return;
+        case Stage::ShuffleS:
+          st.shuffle->consume(std::move(b));
+          return;
+        case Stage::FilterS:
+          b = st.filter->apply(std::move(b));
+          idx++;
+          break;
+        case Stage::ProjectS:
+          b = st.project->apply(std::move(b));
+          idx++;
+          break;
+      }
     }
+    FAIL("batch fed past end of chain");
   }
 
   DevBatch host_out_to_dev(const std::pair<int64_t, std::vector<HostOutCol>>& ob,
