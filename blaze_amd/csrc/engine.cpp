@@ -596,13 +596,9 @@ class AggOp {
     const uint8_t* kv = key.validity ? key.validity + done / 8 : nullptr;
     const uint8_t* vv = val.validity ? val.validity + done / 8 : nullptr;
 
-    if (!d_bkey_) {
-      d_bkey_.alloc(AGG2_MAX_CHUNK * 8);
-      d_bval_.alloc(AGG2_MAX_CHUNK * 8);
-      d_browv_.alloc(AGG2_MAX_CHUNK * 4);
-      d_lokey_.alloc(AGG2_MAX_CHUNK * 8);
-      d_loval_.alloc(AGG2_MAX_CHUNK * 8);
-      d_lorowv_.alloc(AGG2_MAX_CHUNK * 4);
+    if (!d_part_) {
+      d_part_.alloc(AGG2_MAX_CHUNK * sizeof(PartRow));
+      d_leftover_.alloc(AGG2_MAX_CHUNK * sizeof(PartRow));
       int64_t mat = (int64_t)nbuck << AGG2_GRID_LOG2;
       d_counts_.alloc((mat + 1) * 4);   // +1: scan total slot
       d_scanned_.alloc((mat + 1) * 4);
@@ -642,19 +638,16 @@ class AggOp {
                         d_offsets_.get<uint32_t>(), stream_);
     // P2: scatter + specials
     launch_agg2_scatter(keys, kv, vals, vv, chunk, AGG2_NBUCK_LOG2,
-                        d_scanned_.get<uint32_t>(), d_bkey_.get<int64_t>(),
-                        d_bval_.get<double>(), d_browv_.get<uint32_t>(),
+                        d_scanned_.get<uint32_t>(), d_part_.get<PartRow>(),
                         stream_);
     if (special_rows)
       launch_agg2_specials(t_, keys, kv, vals, vv, chunk, row_cursor_, stream_);
     // A: per-bucket LDS aggregation (counters[0..1] already zeroed above)
-    launch_agg2_bucket(d_bkey_.get<int64_t>(), d_bval_.get<double>(),
-                       d_browv_.get<uint32_t>(), d_offsets_.get<uint32_t>(),
+    launch_agg2_bucket(d_part_.get<PartRow>(), d_offsets_.get<uint32_t>(),
                        nbuck, d_staged_.get<StagedGroup>(),
                        d_counters_.get<unsigned long long>(),
                        (int64_t)nbuck * AGG2_LSLOTS,
-                       d_lokey_.get<int64_t>(), d_loval_.get<double>(),
-                       d_lorowv_.get<uint32_t>(),
+                       d_leftover_.get<PartRow>(),
                        d_counters_.get<unsigned long long>() + 1,
                        t_.error_flag, stream_);
     unsigned long long* h_ctr =
@@ -670,9 +663,8 @@ class AggOp {
       launch_agg2_merge_groups(t_, d_staged_.get<StagedGroup>(), staged_n,
                                row_cursor_, stream_);
     if (lo_n)
-      launch_agg2_leftovers(t_, d_lokey_.get<int64_t>(),
-                            d_loval_.get<double>(), d_lorowv_.get<uint32_t>(),
-                            lo_n, row_cursor_, stream_);
+      launch_agg2_leftovers(t_, d_leftover_.get<PartRow>(), lo_n, row_cursor_,
+                            stream_);
     AURON_HIP(hipEventRecord(e1, stream_));
     ev_pairs_.push_back({e0, e1});
     update_rows_ += chunk;
@@ -847,8 +839,8 @@ class AggOp {
   DevBuf d_slots_, d_special_, d_ng_, d_err_;
   PinnedBuf pinned_meta_;
   // two-phase scratch (allocated on first large chunk)
-  DevBuf d_bkey_, d_bval_, d_browv_, d_lokey_, d_loval_, d_lorowv_, d_counts_,
-      d_scanned_, d_scan_tmp_, d_offsets_, d_staged_, d_counters_;
+  DevBuf d_part_, d_leftover_, d_counts_, d_scanned_, d_scan_tmp_, d_offsets_,
+      d_staged_, d_counters_;
   PinnedBuf pinned_agg2_;
   std::vector<DevBatch> held_, skipped_;
 };
@@ -1624,6 +1616,30 @@ void auron_on_exit(void) {
 
 const char* auron_version(void) { return "auron-hip 0.1 gfx950"; }
 
+// test-only: decode a ScalarValue ipc_bytes literal and render it
+int32_t auron_debug_decode_scalar(const uint8_t* data, size_t len, char* out,
+                                  size_t cap) {
+  ScalarLit s;
+  std::string err;
+  std::string r;
+  if (!decode_ipc_scalar(data, len, &s, &err)) {
+    r = "ERROR: " + err;
+  } else if (s.is_null) {
+    r = "null dtype=" + std::to_string((int)s.dtype);
+  } else if (s.dtype == DType::Utf8 || s.dtype == DType::Binary) {
+    r = "str:" + s.utf8;
+  } else if (s.dtype == DType::Float64 || s.dtype == DType::Float32) {
+    char buf[64];
+    snprintf(buf, sizeof(buf), "f:%.17g", s.f64);
+    r = buf;
+  } else {
+    r = "i:" + std::to_string(s.i64) + " dtype=" + std::to_string((int)s.dtype);
+  }
+  if (r.size() + 1 > cap) return -1;
+  memcpy(out, r.c_str(), r.size() + 1);
+  return (int32_t)r.size();
+}
+
 // test-only: exercise the get_conf callback plumbing from the C side (the
 // ctypes out-param contract is easy to get wrong — see GET_CONF note)
 int32_t auron_debug_conf_roundtrip(AuronCallbacks* cb, const char* key,
@@ -1694,6 +1710,18 @@ int32_t auron_debug_decode_plan(const uint8_t* data, size_t len, char* out,
           s += ",skip=" + std::to_string((int)ag.supports_partial_skipping) +
                ")->";
           p = ag.input.get();
+          break;
+        }
+        case PlanNode::Filter: {
+          s += "Filter(npred=" +
+               std::to_string(p->filter->predicates.size()) + ")->";
+          p = p->filter->input.get();
+          break;
+        }
+        case PlanNode::Projection: {
+          s += "Project(ncols=" +
+               std::to_string(p->projection->exprs.size()) + ")->";
+          p = p->projection->input.get();
           break;
         }
         case PlanNode::FFIReader: {

@@ -121,6 +121,14 @@ struct StagedGroup {
   double sum;
   unsigned long long cnt_first;  // cnt<<32 | chunk-local first_row
 };
+// partitioned row record (AoS: a scatter touches ONE line run, not three)
+struct PartRow {
+  int64_t key;
+  double val;
+  uint32_t rowv;  // chunk-local row | valid<<31
+  uint32_t _pad;
+};
+static_assert(sizeof(PartRow) == 24, "PartRow must be 24 bytes");
 constexpr int AGG2_LSLOTS = 2048;  // LDS table entries per bucket
 
 constexpr int AGG2_GRID_LOG2 = 11;  // hist/scatter grid: 2048 blocks
@@ -137,23 +145,19 @@ void launch_agg2_offsets(const uint32_t* scanned, int nbuck_log2,
 void launch_agg2_scatter(const int64_t* keys, const uint8_t* key_valid,
                          const double* vals, const uint8_t* val_valid,
                          int64_t n, int nbuck_log2, const uint32_t* scanned,
-                         int64_t* out_key, double* out_val, uint32_t* out_rowv,
-                         hipStream_t s);
+                         PartRow* out, hipStream_t s);
 void launch_agg2_specials(const AggTable& t, const int64_t* keys,
                           const uint8_t* key_valid, const double* vals,
                           const uint8_t* val_valid, int64_t n,
                           uint64_t row_offset, hipStream_t s);
-void launch_agg2_bucket(const int64_t* bkey, const double* bval,
-                        const uint32_t* browv, const uint32_t* offsets,
+void launch_agg2_bucket(const PartRow* part, const uint32_t* offsets,
                         int nbuckets, StagedGroup* staged,
                         unsigned long long* staged_n, int64_t staged_cap,
-                        int64_t* lo_key, double* lo_val, uint32_t* lo_rowv,
-                        unsigned long long* lo_n, uint32_t* error_flag,
-                        hipStream_t s);
+                        PartRow* leftover, unsigned long long* lo_n,
+                        uint32_t* error_flag, hipStream_t s);
 void launch_agg2_merge_groups(const AggTable& t, const StagedGroup* staged,
                               int64_t n, uint64_t row_offset, hipStream_t s);
-void launch_agg2_leftovers(const AggTable& t, const int64_t* keys,
-                           const double* vals, const uint32_t* rowv, int64_t n,
+void launch_agg2_leftovers(const AggTable& t, const PartRow* rows, int64_t n,
                            uint64_t row_offset, hipStream_t s);
 
 // ---- shuffle partition + gather -------------------------------------------

@@ -104,9 +104,7 @@ __global__ void k_agg2_scatter(const int64_t* __restrict__ keys,
                                const uint8_t* __restrict__ val_valid, int64_t n,
                                int nbuck_log2,
                                const uint32_t* __restrict__ scanned,
-                               int64_t* __restrict__ out_key,
-                               double* __restrict__ out_val,
-                               uint32_t* __restrict__ out_rowv) {
+                               PartRow* __restrict__ out) {
   extern __shared__ uint32_t lds_cursor[];
   const uint32_t nbuck = 1u << nbuck_log2;
   for (uint32_t b = threadIdx.x; b < nbuck; b += blockDim.x)
@@ -120,9 +118,12 @@ __global__ void k_agg2_scatter(const int64_t* __restrict__ keys,
     uint32_t b = bucket_of(k, nbuck_log2);
     uint32_t pos = atomicAdd(&lds_cursor[b], 1u);  // block-local LDS cursor
     bool vvalid = !val_valid || bit_get2(val_valid, i);
-    out_key[pos] = k;
-    out_val[pos] = vals[i];
-    out_rowv[pos] = (uint32_t)i | (vvalid ? 0x80000000u : 0u);
+    PartRow r;
+    r.key = k;
+    r.val = vals[i];
+    r.rowv = (uint32_t)i | (vvalid ? 0x80000000u : 0u);
+    r._pad = 0;
+    out[pos] = r;
   }
 }
 
@@ -160,12 +161,10 @@ static constexpr int LSLOTS = AGG2_LSLOTS;  // 48 KB LDS -> 3 blocks/CU
 static constexpr int LPROBE = 64;
 
 __global__ void __launch_bounds__(256) k_agg2_bucket(
-    const int64_t* __restrict__ bkey, const double* __restrict__ bval,
-    const uint32_t* __restrict__ browv, const uint32_t* __restrict__ offsets,
+    const PartRow* __restrict__ part, const uint32_t* __restrict__ offsets,
     int nbuckets, StagedGroup* __restrict__ staged,
     unsigned long long* __restrict__ staged_n, int64_t staged_cap,
-    int64_t* __restrict__ lo_key, double* __restrict__ lo_val,
-    uint32_t* __restrict__ lo_rowv, unsigned long long* __restrict__ lo_n,
+    PartRow* __restrict__ leftover, unsigned long long* __restrict__ lo_n,
     uint32_t* __restrict__ error_flag) {
   __shared__ int64_t ls_key[LSLOTS];
   __shared__ double ls_sum[LSLOTS];
@@ -182,9 +181,9 @@ __global__ void __launch_bounds__(256) k_agg2_bucket(
     __syncthreads();
     uint32_t beg = offsets[b], end = offsets[b + 1];
     for (uint32_t i = beg + threadIdx.x; i < end; i += blockDim.x) {
-      int64_t k = bkey[i];
-      double v = bval[i];
-      uint32_t rowv = browv[i];
+      int64_t k = part[i].key;
+      double v = part[i].val;
+      uint32_t rowv = part[i].rowv;
       uint32_t row = rowv & 0x7FFFFFFFu;
       bool vvalid = (rowv & 0x80000000u) != 0;
       // probe LDS (low bits of the same mix; bucket used the high bits)
@@ -216,9 +215,7 @@ __global__ void __launch_bounds__(256) k_agg2_bucket(
       } else {
         // LDS window full: append raw row to the leftover list
         unsigned long long p = atomicAdd(lo_n, 1ull);
-        lo_key[p] = k;
-        lo_val[p] = v;
-        lo_rowv[p] = rowv;
+        leftover[p] = part[i];
       }
     }
     __syncthreads();
@@ -290,14 +287,12 @@ __global__ void k_agg2_merge_groups(const AggTable t,
 
 // leftover raw rows (LDS window overflow — rare): single-phase accumulate
 __global__ void k_agg2_leftovers(const AggTable t,
-                                 const int64_t* __restrict__ keys,
-                                 const double* __restrict__ vals,
-                                 const uint32_t* __restrict__ rowv, int64_t n,
+                                 const PartRow* __restrict__ rows, int64_t n,
                                  uint64_t row_offset) {
   const int64_t mask = t.cap - 1;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
-    int64_t key = keys[i];
+    int64_t key = rows[i].key;
     uint64_t h = mix64_2((uint64_t)key);
     int64_t s = (int64_t)(h & (uint64_t)mask);
     int64_t a = -1;
@@ -328,21 +323,20 @@ __global__ void k_agg2_leftovers(const AggTable t,
       continue;
     }
     AggSlot* sl = &t.slots[a];
-    uint32_t rv = rowv[i];
+    uint32_t rv = rows[i].rowv;
     uint64_t row = row_offset + (rv & 0x7FFFFFFFu);
     if (sl->first_row > row) atomicMin(&sl->first_row, row);
     if (rv & 0x80000000u) {
-      unsafeAtomicAdd(&sl->sum, vals[i]);
+      unsafeAtomicAdd(&sl->sum, rows[i].val);
       atomicAdd(&sl->cnt, 1ull);
     }
   }
 }
 
-void launch_agg2_leftovers(const AggTable& t, const int64_t* keys,
-                           const double* vals, const uint32_t* rowv, int64_t n,
+void launch_agg2_leftovers(const AggTable& t, const PartRow* rows, int64_t n,
                            uint64_t row_offset, hipStream_t s) {
   hipLaunchKernelGGL(k_agg2_leftovers, dim3(grid2(n)), dim3(BLOCK), 0, s, t,
-                     keys, vals, rowv, n, row_offset);
+                     rows, n, row_offset);
   check_launch2("k_agg2_leftovers");
 }
 
@@ -374,12 +368,11 @@ void scan_counts_matrix(const uint32_t* counts, uint32_t* scanned, int64_t n,
 void launch_agg2_scatter(const int64_t* keys, const uint8_t* key_valid,
                          const double* vals, const uint8_t* val_valid,
                          int64_t n, int nbuck_log2, const uint32_t* scanned,
-                         int64_t* out_key, double* out_val, uint32_t* out_rowv,
-                         hipStream_t s) {
+                         PartRow* out, hipStream_t s) {
   size_t lds = (size_t)(1u << nbuck_log2) * 4;
   hipLaunchKernelGGL(k_agg2_scatter, dim3(1 << GRID_LOG2), dim3(BLOCK), lds, s,
                      keys, key_valid, vals, val_valid, n, nbuck_log2, scanned,
-                     out_key, out_val, out_rowv);
+                     out);
   check_launch2("k_agg2_scatter");
 }
 
@@ -392,17 +385,15 @@ void launch_agg2_specials(const AggTable& t, const int64_t* keys,
   check_launch2("k_agg2_specials");
 }
 
-void launch_agg2_bucket(const int64_t* bkey, const double* bval,
-                        const uint32_t* browv, const uint32_t* offsets,
+void launch_agg2_bucket(const PartRow* part, const uint32_t* offsets,
                         int nbuckets, StagedGroup* staged,
                         unsigned long long* staged_n, int64_t staged_cap,
-                        int64_t* lo_key, double* lo_val, uint32_t* lo_rowv,
-                        unsigned long long* lo_n, uint32_t* error_flag,
-                        hipStream_t s) {
+                        PartRow* leftover, unsigned long long* lo_n,
+                        uint32_t* error_flag, hipStream_t s) {
   int blocks = nbuckets < (int)MAX_BLOCKS ? nbuckets : (int)MAX_BLOCKS;
-  hipLaunchKernelGGL(k_agg2_bucket, dim3(blocks), dim3(BLOCK), 0, s, bkey, bval,
-                     browv, offsets, nbuckets, staged, staged_n, staged_cap,
-                     lo_key, lo_val, lo_rowv, lo_n, error_flag);
+  hipLaunchKernelGGL(k_agg2_bucket, dim3(blocks), dim3(BLOCK), 0, s, part,
+                     offsets, nbuckets, staged, staged_n, staged_cap, leftover,
+                     lo_n, error_flag);
   check_launch2("k_agg2_bucket");
 }
 

@@ -193,3 +193,75 @@ def plan_shuffle_only(data_file, index_file, num_partitions=200,
     rep = hash_repartition([column(*hash_col)], num_partitions)
     sw = shuffle_writer(reader, rep, data_file, index_file)
     return task_definition(sw, partition_id=partition_id)
+
+
+# ---- filter / projection (a15 in-engine half) ------------------------------
+
+def literal_ipc(value, dtype):
+    """ScalarValue.ipc_bytes: an Arrow IPC stream of a 1-row single-column
+    batch (auron-serde/src/lib.rs:447-456). Built with pyarrow, exactly like
+    the JVM side serializes literals."""
+    import io
+
+    import pyarrow as pa
+    import pyarrow.ipc as ipc
+
+    arr = pa.array([value], type=dtype)
+    batch = pa.record_batch([arr], names=["lit"])
+    sink = io.BytesIO()
+    with ipc.new_stream(sink, batch.schema) as w:
+        w.write_batch(batch)
+    return sink.getvalue()
+
+
+def literal(value, dtype="int32"):
+    import pyarrow as pa
+
+    pa_type = {"int32": pa.int32(), "int64": pa.int64(),
+               "float64": pa.float64(), "utf8": pa.utf8()}[dtype]
+    # PhysicalExprNode{literal = 2} -> ScalarValue{ipc_bytes = 1}
+    return _len_field(2, _len_field(1, literal_ipc(value, pa_type)))
+
+
+def binary_expr(l, r, op):
+    # PhysicalExprNode{binary_expr = 4} -> PhysicalBinaryExprNode{l=1,r=2,op=3}
+    node = _len_field(1, l) + _len_field(2, r) + _len_field(3, op.encode())
+    return _len_field(4, node)
+
+
+def is_not_null(child):
+    return _len_field(7, _len_field(1, child))
+
+
+def filter_node(input_plan, predicates):
+    # PhysicalPlanNode{filter = 8} -> FilterExecNode (auron.proto:363-366)
+    node = _len_field(1, input_plan)
+    for p in predicates:
+        node += _len_field(2, p)
+    return _len_field(8, node)
+
+
+def projection(input_plan, exprs, names):
+    # PhysicalPlanNode{projection = 6} -> ProjectionExecNode (:505-510)
+    node = _len_field(1, input_plan)
+    for e in exprs:
+        node += _len_field(2, e)
+    for n in names:
+        node += _len_field(3, n.encode())
+    return _len_field(6, node)
+
+
+def plan_filter_project_agg(resource_id="input0", cutoff=500_000,
+                            cutoff_dtype="int64"):
+    """FFIReader(key,val) -> Filter(key < c) -> Project(key,val) ->
+    Agg(Partial) -> Agg(Final): the config-3 shape minus Parquet decode."""
+    reader = ffi_reader(northstar_input_fields(), resource_id)
+    filt = filter_node(reader, [
+        binary_expr(column("key", 0), literal(cutoff, cutoff_dtype), "Lt")])
+    proj = projection(filt, [column("key", 0), column("val", 1)],
+                      ["key", "val"])
+    partial = agg(proj, [column("key", 0)], sum_count_aggs(1),
+                  [MODE_PARTIAL, MODE_PARTIAL], ["key"], ["sum", "cnt"])
+    final = agg(partial, [column("key", 0)], sum_count_aggs(1),
+                [MODE_FINAL, MODE_FINAL], ["key"], ["sum", "cnt"])
+    return task_definition(final)

@@ -1,0 +1,56 @@
+"""Filter/Projection plan decode + the Arrow-IPC scalar literal reader
+(CPU-only): the C++ flatbuffer walker must agree with pyarrow-produced
+ScalarValue.ipc_bytes (the reference's literal encoding,
+auron-serde/src/lib.rs:447-456)."""
+import ctypes
+
+import pytest
+
+import blaze_amd
+from blaze_amd import plan
+
+pa = pytest.importorskip("pyarrow")
+
+
+def _decode_scalar(blob: bytes) -> str:
+    lib = blaze_amd.lib()
+    lib.auron_debug_decode_scalar.restype = ctypes.c_int32
+    lib.auron_debug_decode_scalar.argtypes = [ctypes.c_char_p, ctypes.c_size_t,
+                                              ctypes.c_char_p, ctypes.c_size_t]
+    out = ctypes.create_string_buffer(512)
+    rc = lib.auron_debug_decode_scalar(blob, len(blob), out, 512)
+    assert rc >= 0
+    return out.value.decode()
+
+
+def test_ipc_scalar_int32():
+    blob = plan.literal_ipc(123456, pa.int32())
+    assert _decode_scalar(blob) == "i:123456 dtype=8"
+
+
+def test_ipc_scalar_int64_negative():
+    blob = plan.literal_ipc(-(2**40), pa.int64())
+    assert _decode_scalar(blob) == f"i:{-(2**40)} dtype=10"
+
+
+def test_ipc_scalar_float64():
+    blob = plan.literal_ipc(3.5, pa.float64())
+    assert _decode_scalar(blob) == "f:3.5"
+
+
+def test_ipc_scalar_utf8():
+    blob = plan.literal_ipc("hello world", pa.utf8())
+    assert _decode_scalar(blob) == "str:hello world"
+
+
+def test_ipc_scalar_null():
+    blob = plan.literal_ipc(None, pa.int64())
+    assert _decode_scalar(blob).startswith("null")
+
+
+def test_filter_project_plan_decodes():
+    td = plan.plan_filter_project_agg(cutoff=777)
+    s = blaze_amd.debug_decode_plan(td)
+    # chain shape is intact down to the reader
+    assert "Agg(mode=2" in s and "Agg(mode=0" in s
+    assert "FFIReader(nfields=2,rid=input0)" in s
