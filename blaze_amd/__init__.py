@@ -76,6 +76,9 @@ IMPORT_BATCH = c.CFUNCTYPE(None, c.c_void_p, c.POINTER(ArrowArray))
 IMPORT_DEV = c.CFUNCTYPE(None, c.c_void_p, c.POINTER(ArrowDeviceArray),
                          c.POINTER(ArrowSchema))
 SET_ERROR = c.CFUNCTYPE(None, c.c_void_p, c.c_char_p)
+NEXT_IPC = c.CFUNCTYPE(c.c_int, c.c_void_p, c.c_char_p,
+                       c.POINTER(c.POINTER(c.c_uint8)),
+                       c.POINTER(c.c_size_t))
 
 
 class AuronCallbacks(c.Structure):
@@ -87,6 +90,7 @@ class AuronCallbacks(c.Structure):
         ("import_batch", IMPORT_BATCH),
         ("import_device_batch", IMPORT_DEV),
         ("set_error", SET_ERROR),
+        ("next_ipc_bytes", NEXT_IPC),
     ]
 
 
@@ -294,13 +298,19 @@ def _import_output(array_ptr, schema_fields):
 class Task:
     """One native task execution (mirrors JniBridge callNative/nextBatch)."""
 
-    def __init__(self, task_bytes, batches=None, conf=None, device_batches=None):
+    def __init__(self, task_bytes, batches=None, conf=None, device_batches=None,
+                 ipc_segments=None):
         """batches: list of _HostBatch col-spec lists (host numpy input).
         device_batches: list of prebuilt ArrowDeviceArray + keepalive (see
-        bench.py) for zero-copy HBM input."""
+        bench.py) for zero-copy HBM input.
+        ipc_segments: list of bytes (shuffle block streams) for IpcReaderExec
+        plans — plays the JVM fetch iterator's role."""
         self._conf = dict(conf or {})
         self._in_host = [_HostBatch(cols) for cols in (batches or [])]
         self._in_dev = list(device_batches or [])
+        self._ipc = [np.frombuffer(seg, dtype=np.uint8).copy()
+                     for seg in (ipc_segments or [])]
+        self._ipc_cursor = 0
         self._cursor = 0
         self.schema_fields = []   # (name, fmt)
         self.outputs = []
@@ -353,12 +363,25 @@ class Task:
         def set_error(user, msg):
             self.error = msg.decode()
 
+        @NEXT_IPC
+        def next_ipc(user, rid, data_out, len_out):
+            i = self._ipc_cursor
+            if i >= len(self._ipc):
+                return 0
+            self._ipc_cursor += 1
+            seg = self._ipc[i]
+            data_out[0] = c.cast(c.c_void_p(seg.ctypes.data),
+                                 c.POINTER(c.c_uint8))
+            len_out[0] = seg.size
+            return 1
+
         self._cb = AuronCallbacks(
             user=None, get_conf=get_conf, next_input_batch=next_input,
             import_schema=import_schema, import_batch=import_batch,
-            import_device_batch=c.cast(None, IMPORT_DEV), set_error=set_error)
+            import_device_batch=c.cast(None, IMPORT_DEV), set_error=set_error,
+            next_ipc_bytes=next_ipc)
         self._keep += [get_conf, next_input, import_schema, import_batch,
-                       set_error]
+                       set_error, next_ipc]
         self.handle = lib().auron_call_native(task_bytes, len(task_bytes),
                                               c.byref(self._cb))
         if self.handle == 0:

@@ -1361,14 +1361,14 @@ struct Runtime {
         case PlanNode::Agg: p = p->agg->input.get(); break;
         case PlanNode::Filter: p = p->filter->input.get(); break;
         case PlanNode::Projection: p = p->projection->input.get(); break;
+        case PlanNode::IpcReader:
         case PlanNode::FFIReader: p = nullptr; break;
       }
     }
-    // leaf must be FFIReader
+    // leaf must be a source: FFIReader (Arrow in) or IpcReader (shuffle bytes)
     const PlanNode* leaf = chain.back();
-    if (leaf->kind != PlanNode::FFIReader)
-      FAIL("plan leaf must be FFIReaderExec on this path");
-    const FFIReaderNode& reader = *leaf->ffi_reader;
+    if (leaf->kind != PlanNode::FFIReader && leaf->kind != PlanNode::IpcReader)
+      FAIL("plan leaf must be FFIReaderExec/IpcReaderExec on this path");
 
     // middle ops (leaf-1 ... root), in execution order
     stages_.clear();
@@ -1401,32 +1401,37 @@ struct Runtime {
       stages_.push_back(std::move(st));
     }
 
-    // pump input through agg chain
+    // pump input through the chain
     int64_t input_rows = 0;
-    while (true) {
-      ArrowArray arr;
-      ArrowSchema sch;
-      ArrowDeviceArray dev;
-      memset(&arr, 0, sizeof(arr));
-      memset(&sch, 0, sizeof(sch));
-      memset(&dev, 0, sizeof(dev));
-      if (!cb.next_input_batch) break;
-      int rc = cb.next_input_batch(cb.user, reader.resource_id.c_str(), &arr,
-                                   &sch, &dev);
-      DBG("reader rc=%d", rc);
-      if (rc == 0) break;
-      DevBatch b;
-      if (rc == 2) {
-        b = import_batch(&dev.array, sch.release ? &sch : nullptr,
-                         reader.schema, true, stream);
-      } else {
-        b = import_batch(&arr, sch.release ? &sch : nullptr, reader.schema,
-                         false, stream);
-        if (arr.release) arr.release(&arr);
+    if (leaf->kind == PlanNode::FFIReader) {
+      const FFIReaderNode& reader = *leaf->ffi_reader;
+      while (true) {
+        ArrowArray arr;
+        ArrowSchema sch;
+        ArrowDeviceArray dev;
+        memset(&arr, 0, sizeof(arr));
+        memset(&sch, 0, sizeof(sch));
+        memset(&dev, 0, sizeof(dev));
+        if (!cb.next_input_batch) break;
+        int rc = cb.next_input_batch(cb.user, reader.resource_id.c_str(), &arr,
+                                     &sch, &dev);
+        DBG("reader rc=%d", rc);
+        if (rc == 0) break;
+        DevBatch b;
+        if (rc == 2) {
+          b = import_batch(&dev.array, sch.release ? &sch : nullptr,
+                           reader.schema, true, stream);
+        } else {
+          b = import_batch(&arr, sch.release ? &sch : nullptr, reader.schema,
+                           false, stream);
+          if (arr.release) arr.release(&arr);
+        }
+        if (sch.release) sch.release(&sch);
+        input_rows += b.num_rows;
+        feed(0, std::move(b));
       }
-      if (sch.release) sch.release(&sch);
-      input_rows += b.num_rows;
-      feed(0, std::move(b));
+    } else {
+      input_rows = pump_ipc_reader(*leaf->ipc_reader);
     }
     // drain chain: pipeline breakers emit, transforms pass through
     AggOp* last_agg = nullptr;
@@ -1469,6 +1474,76 @@ struct Runtime {
         std::chrono::duration_cast<std::chrono::nanoseconds>(
             std::chrono::steady_clock::now() - t0)
             .count();
+  }
+
+  // IpcReaderExec source (ipc_reader_exec.rs:62-120): raw shuffle block
+  // streams in, decoded+deserialized on host, batches up to the device.
+  int64_t pump_ipc_reader(const IpcReaderNode& reader) {
+    if (!cb.next_ipc_bytes)
+      FAIL("plan holds IpcReaderExec but no next_ipc_bytes callback was given");
+    std::vector<int> widths;
+    for (const Field& f : reader.schema.fields) {
+      if (f.dtype == DType::Binary || f.dtype == DType::Utf8)
+        widths.push_back(0);
+      else if (dtype_width(f.dtype) > 0)
+        widths.push_back((int)dtype_width(f.dtype));
+      else
+        FAIL("IpcReader: unsupported column dtype");
+    }
+    int64_t input_rows = 0;
+    while (true) {
+      const uint8_t* data = nullptr;
+      size_t len = 0;
+      int rc = cb.next_ipc_bytes(cb.user, reader.resource_id.c_str(), &data,
+                                 &len);
+      DBG("ipc reader rc=%d len=%zu", rc, len);
+      if (rc == 0) break;
+      std::vector<uint8_t> payload;
+      std::string err;
+      if (!ipc_decode_blocks(data, len, &payload, &err)) FAIL(err);
+      size_t used = 0;
+      while (used < payload.size()) {
+        int64_t rows = 0;
+        std::vector<OwnedCol> cols;
+        if (!serde_read_batch(payload.data(), payload.size(), &used, widths,
+                              &rows, &cols, &err))
+          FAIL(err);
+        DevBatch b;
+        b.num_rows = rows;
+        for (size_t ci = 0; ci < cols.size(); ci++) {
+          OwnedCol& oc = cols[ci];
+          DevColumn c;
+          c.dt = reader.schema.fields[ci].dtype;
+          c.len = rows;
+          if (oc.byte_width == 0) {
+            c.own_offsets.alloc((rows + 1) * 4);
+            AURON_HIP(hipMemcpyAsync(c.own_offsets.get(), oc.offsets.data(),
+                                     (rows + 1) * 4, hipMemcpyHostToDevice,
+                                     stream));
+            c.offsets = c.own_offsets.get<int32_t>();
+            c.data_len = (int64_t)oc.values.size();
+          }
+          c.own_values.alloc(oc.values.empty() ? 1 : oc.values.size());
+          if (!oc.values.empty())
+            AURON_HIP(hipMemcpyAsync(c.own_values.get(), oc.values.data(),
+                                     oc.values.size(), hipMemcpyHostToDevice,
+                                     stream));
+          c.values = c.own_values.get();
+          if (!oc.validity.empty()) {
+            c.own_validity.alloc(oc.validity.size());
+            AURON_HIP(hipMemcpyAsync(c.own_validity.get(), oc.validity.data(),
+                                     oc.validity.size(), hipMemcpyHostToDevice,
+                                     stream));
+            c.validity = c.own_validity.get<uint8_t>();
+          }
+          AURON_HIP(hipStreamSynchronize(stream));  // host vectors die below
+          b.cols.push_back(std::move(c));
+        }
+        input_rows += rows;
+        feed(0, std::move(b));
+      }
+    }
+    return input_rows;
   }
 
   void feed(size_t idx, DevBatch&& b) {
@@ -1725,6 +1800,13 @@ int32_t auron_debug_decode_plan(const uint8_t* data, size_t len, char* out,
           s += "Project(ncols=" +
                std::to_string(p->projection->exprs.size()) + ")->";
           p = p->projection->input.get();
+          break;
+        }
+        case PlanNode::IpcReader: {
+          const auto& ir = *p->ipc_reader;
+          s += "IpcReader(nfields=" + std::to_string(ir.schema.fields.size()) +
+               ",rid=" + ir.resource_id + ")";
+          p = nullptr;
           break;
         }
         case PlanNode::FFIReader: {

@@ -121,6 +121,13 @@ struct FFIReaderNode {
   std::string resource_id;       // = 3
 };
 
+// IpcReaderExecNode (auron.proto:607-611)
+struct IpcReaderNode {
+  uint32_t num_partitions = 1;   // = 1
+  Schema schema;                 // = 2
+  std::string resource_id;       // = 3 (ipc_provider_resource_id)
+};
+
 // FilterExecNode (auron.proto:363-366): predicates are ANDed
 struct FilterNode {
   std::unique_ptr<PlanNode> input;  // = 1
@@ -138,6 +145,7 @@ struct ProjectionNode {
 struct PlanNode {
   enum Kind {
     ShuffleWriter = 2,
+    IpcReader = 3,
     Projection = 6,
     Filter = 8,
     Agg = 16,
@@ -148,6 +156,7 @@ struct PlanNode {
   std::unique_ptr<FFIReaderNode> ffi_reader;
   std::unique_ptr<FilterNode> filter;
   std::unique_ptr<ProjectionNode> projection;
+  std::unique_ptr<IpcReaderNode> ipc_reader;
 };
 
 // TaskDefinition (auron.proto:735-740) + PartitionId (:729-733)

@@ -292,6 +292,21 @@ std::unique_ptr<ShuffleWriterNode> decode_shuffle_writer(Reader r,
   return n;
 }
 
+std::unique_ptr<IpcReaderNode> decode_ipc_reader(Reader r) {
+  auto n = std::make_unique<IpcReaderNode>();
+  while (true) {
+    auto [f, w] = r.tag();
+    if (f == 0) break;
+    switch (f) {
+      case 1: n->num_partitions = (uint32_t)r.varint(); break;
+      case 2: n->schema = decode_schema(r.sub()); break;
+      case 3: n->resource_id = r.str(); break;
+      default: r.skip(w);
+    }
+  }
+  return n;
+}
+
 std::unique_ptr<FFIReaderNode> decode_ffi_reader(Reader r) {
   auto n = std::make_unique<FFIReaderNode>();
   while (true) {
@@ -341,6 +356,10 @@ std::unique_ptr<PlanNode> decode_plan_node(Reader r, std::string* err) {
       case 2:  // ShuffleWriterExecNode
         node->kind = PlanNode::ShuffleWriter;
         node->shuffle_writer = decode_shuffle_writer(r.sub(), err);
+        return node;
+      case 3:  // IpcReaderExecNode
+        node->kind = PlanNode::IpcReader;
+        node->ipc_reader = decode_ipc_reader(r.sub());
         return node;
       case 6:  // ProjectionExecNode
         node->kind = PlanNode::Projection;

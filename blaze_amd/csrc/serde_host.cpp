@@ -69,6 +69,116 @@ void serde_write_batch(const std::vector<HostCol>& cols, int64_t row_beg,
   }
 }
 
+namespace {
+
+bool read_len_at(const uint8_t* p, size_t len, size_t* pos, uint64_t* out) {
+  // io/mod.rs:69-79
+  uint64_t v = 0, factor = 1;
+  while (*pos < len) {
+    uint8_t b = p[(*pos)++];
+    if (b < 128) {
+      v += (uint64_t)b * factor;
+      *out = v;
+      return true;
+    }
+    v += (uint64_t)(b - 128) * factor;
+    factor *= 128;
+  }
+  return false;
+}
+
+void untranspose(const uint8_t* in, uint8_t* out, int64_t n, int w) {
+  // inverse of batch_serde.rs:271-306 byte-plane transpose
+  for (int64_t i = 0; i < n; i++)
+    for (int b = 0; b < w; b++) out[i * w + b] = in[(int64_t)b * n + i];
+}
+
+}  // namespace
+
+bool serde_read_batch(const uint8_t* p, size_t len, size_t* used,
+                      const std::vector<int>& dtype_widths, int64_t* rows,
+                      std::vector<OwnedCol>* cols, std::string* err) {
+  size_t pos = *used;
+  uint64_t n = 0;
+  if (!read_len_at(p, len, &pos, &n)) {
+    *err = "serde: truncated row count";
+    return false;
+  }
+  *rows = (int64_t)n;
+  cols->clear();
+  for (int w : dtype_widths) {
+    OwnedCol c;
+    c.byte_width = w;
+    uint64_t has_null = 0;
+    if (!read_len_at(p, len, &pos, &has_null)) {
+      *err = "serde: truncated null header";
+      return false;
+    }
+    size_t bm = (n + 7) / 8;
+    if (has_null) {
+      if (pos + bm > len) {
+        *err = "serde: truncated bitmap";
+        return false;
+      }
+      c.validity.assign(p + pos, p + pos + bm);
+      pos += bm;
+    }
+    if (w > 0) {
+      size_t nb = (size_t)n * w;
+      if (pos + nb > len) {
+        *err = "serde: truncated values";
+        return false;
+      }
+      c.values.resize(nb);
+      if (w > 1)
+        untranspose(p + pos, c.values.data(), (int64_t)n, w);
+      else
+        memcpy(c.values.data(), p + pos, nb);
+      pos += nb;
+    } else {
+      // binary: transposed i32 lens then raw data (batch_serde.rs:595-660)
+      if (pos + 4 * n > len) {
+        *err = "serde: truncated lens";
+        return false;
+      }
+      std::vector<int32_t> lens(n);
+      untranspose(p + pos, (uint8_t*)lens.data(), (int64_t)n, 4);
+      pos += 4 * n;
+      c.offsets.resize(n + 1);
+      c.offsets[0] = 0;
+      for (uint64_t i = 0; i < n; i++)
+        c.offsets[i + 1] = c.offsets[i] + lens[i];
+      size_t db = (size_t)c.offsets[n];
+      if (pos + db > len) {
+        *err = "serde: truncated binary data";
+        return false;
+      }
+      c.values.assign(p + pos, p + pos + db);
+      pos += db;
+    }
+    cols->push_back(std::move(c));
+  }
+  *used = pos;
+  return true;
+}
+
+bool ipc_decode_blocks(const uint8_t* p, size_t len,
+                       std::vector<uint8_t>* payload, std::string* err) {
+  size_t pos = 0;
+  while (pos + 4 <= len) {
+    uint32_t block_len;
+    memcpy(&block_len, p + pos, 4);
+    pos += 4;
+    if (pos + block_len > len) {
+      *err = "ipc: truncated block";
+      return false;
+    }
+    if (!lz4_decompress_frame(p + pos, block_len, payload, err)) return false;
+    pos += block_len;
+  }
+  return true;
+}
+
 bool IpcBlockWriter::write_payload(const uint8_t* p, size_t len,
                                    std::string* err) {
   staged_.insert(staged_.end(), p, p + len);

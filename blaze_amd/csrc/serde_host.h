@@ -38,6 +38,25 @@ class IpcBlockWriter {
   std::vector<uint8_t> out_;
 };
 
+// ---- read side (IpcReaderExec, ipc_reader_exec.rs:62-120) ------------------
+struct OwnedCol {
+  int byte_width = 0;              // >0 primitive, 0 binary
+  std::vector<uint8_t> values;     // prim values / binary data
+  std::vector<uint8_t> validity;   // empty = no nulls
+  std::vector<int32_t> offsets;    // binary: rows+1
+};
+
+// parse ONE batch_serde batch from p; *used advances past it. dtype_widths:
+// >0 for primitives, 0 for binary/utf8.
+bool serde_read_batch(const uint8_t* p, size_t len, size_t* used,
+                      const std::vector<int>& dtype_widths, int64_t* rows,
+                      std::vector<OwnedCol>* cols, std::string* err);
+
+// decode a [u32-LE len][lz4 frame] block stream into the concatenated
+// uncompressed payload (ipc_compression.rs:64-112 read side)
+bool ipc_decode_blocks(const uint8_t* p, size_t len,
+                       std::vector<uint8_t>* payload, std::string* err);
+
 // write the shuffle data + index files: per partition, rows
 // [part_offsets[p], part_offsets[p+1]) of the partition-sorted cols, framed
 // as IPC blocks finished at each partition boundary; index = (P+1) u64-LE.

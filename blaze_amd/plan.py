@@ -265,3 +265,24 @@ def plan_filter_project_agg(resource_id="input0", cutoff=500_000,
     final = agg(partial, [column("key", 0)], sum_count_aggs(1),
                 [MODE_FINAL, MODE_FINAL], ["key"], ["sum", "cnt"])
     return task_definition(final)
+
+
+# ---- IpcReaderExec (f.3: native shuffle read-back) -------------------------
+
+def ipc_reader(fields, resource_id, num_partitions=1):
+    # PhysicalPlanNode{ipc_reader = 3} -> IpcReaderExecNode (auron.proto:607-611)
+    node = _varint_field(1, num_partitions)
+    node += _len_field(2, schema(fields))
+    node += _len_field(3, resource_id.encode())
+    return _len_field(3, node)
+
+
+def plan_ipc_final(resource_id="ipc0"):
+    """IpcReader(partial output schema) -> Agg(Final): the reference's
+    stage-2 topology (NativeShuffleExchangeBase.scala:147-179) fully native."""
+    fields = [field("key", DT_INT64, True),
+              field("#9223372036854775807", DT_BINARY, False)]
+    reader = ipc_reader(fields, resource_id)
+    final = agg(reader, [column("key", 0)], sum_count_aggs(1),
+                [MODE_FINAL, MODE_FINAL], ["key"], ["sum", "cnt"])
+    return task_definition(final)

@@ -112,6 +112,14 @@ typedef struct AuronCallbacks {
 
   /* Error channel — mirrors wrapper.setError (rt.rs:284-293). */
   void (*set_error)(void* user, const char* message);
+
+  /* IpcReaderExecNode input (auron.proto:607-611; ipc_reader_exec.rs:62-120):
+   * pull the next raw shuffle block-stream segment ([u32-LE len][lz4 frame]*
+   * bytes, the format ShuffleWriter emits). Return 1 and set *data/*len
+   * (caller-owned, valid until the next call or finalize), 0 on exhaustion.
+   * May be NULL when the plan holds no IpcReaderExec. */
+  int (*next_ipc_bytes)(void* user, const char* resource_id,
+                        const uint8_t** data, size_t* len);
 } AuronCallbacks;
 
 /* Create a runtime for one task. Returns a handle (>0) or 0 on error (error
