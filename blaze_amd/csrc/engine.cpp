@@ -344,24 +344,47 @@ class AggOp {
     key_col_ = node.grouping_exprs[0].col_index;
     key_name_ = node.grouping_names.empty() ? std::string("key")
                                             : node.grouping_names[0];
-    if (node.agg_exprs.size() != 2 ||
-        node.agg_exprs[0].agg_function != AGG_SUM ||
-        node.agg_exprs[1].agg_function != AGG_COUNT)
-      FAIL("AggExec: agg set must be [SUM, COUNT] (north-star shape)");
-    if (node.modes.size() != 2 || node.modes[0] != node.modes[1])
-      FAIL("AggExec: mixed agg modes unsupported");
+    // agg set: any list of SUM/COUNT/AVG over ONE shared argument column —
+    // the shapes whose accumulator state is {sum, cnt} (sum.rs, count.rs,
+    // avg.rs; AVG freeze = sum ++ count, avg.rs:208-217). Anything else
+    // fails loudly at plan build.
+    if (node.agg_exprs.empty() || node.agg_exprs.size() > 8)
+      FAIL("AggExec: 1..8 aggregates supported");
+    if (node.modes.size() != node.agg_exprs.size())
+      FAIL("AggExec: modes/aggs length mismatch");
+    for (const AggMode m : node.modes)
+      if (m != node.modes[0]) FAIL("AggExec: mixed agg modes unsupported");
     mode_ = node.modes[0];
     merge_mode_ = (mode_ != AggMode::Partial);
     final_output_ = (mode_ == AggMode::Final);
-    if (!merge_mode_) {
-      const Expr& sum_child = node.agg_exprs[0].children.at(0);
-      if (sum_child.kind != Expr::Column) FAIL("SUM arg must be a Column");
-      val_col_ = sum_child.col_index;
-      // the engine derives sum-validity from cnt>0, which requires SUM and
-      // COUNT to share the argument column (the north-star shape)
-      const Expr& cnt_child = node.agg_exprs[1].children.at(0);
-      if (cnt_child.kind != Expr::Column || cnt_child.col_index != val_col_)
-        FAIL("COUNT arg must be the same column as SUM arg");
+    layout_ = 0;
+    bool have_col = false;
+    for (size_t i = 0; i < node.agg_exprs.size(); i++) {
+      const Expr& a = node.agg_exprs[i];
+      uint32_t k;
+      switch (a.agg_function) {
+        case AGG_SUM: k = AGGL_SUM; break;
+        case AGG_COUNT: k = AGGL_CNT; break;
+        case AGG_AVG: k = AGGL_AVG; break;
+        default:
+          FAIL("AggExec: only SUM/COUNT/AVG aggregates on this path");
+      }
+      layout_ |= k << (2 * i);
+      agg_kinds_.push_back(k);
+      agg_names_.push_back(i < node.agg_names.size() ? node.agg_names[i]
+                                                     : "agg" + std::to_string(i));
+      if (!merge_mode_) {
+        const Expr& child = a.children.at(0);
+        if (child.kind != Expr::Column)
+          FAIL("aggregate arg must be a Column");
+        if (!have_col) {
+          val_col_ = child.col_index;
+          have_col = true;
+        } else if (child.col_index != val_col_) {
+          // shared {sum,cnt} accumulators require one argument column
+          FAIL("all aggregates must share one argument column on this path");
+        }
+      }
     }
     batch_size_ = conf.get_i("BATCH_SIZE", 10000);
     if (batch_size_ <= 0) FAIL("invalid BATCH_SIZE conf");
@@ -434,7 +457,7 @@ class AggOp {
         launch_agg_merge_frozen(t_, (const int64_t*)key.values + done,
                                 slice_valid(key.validity, done),
                                 (const uint8_t*)buf.values, buf.offsets + done,
-                                chunk, row_cursor_, stream_);
+                                chunk, row_cursor_, layout_, stream_);
       } else {
         const DevColumn& val = b.cols.at(val_col_);
         if (val.dt != DType::Float64) FAIL("SUM arg must be Float64");
@@ -496,9 +519,14 @@ class AggOp {
 
   std::vector<OutField> output_fields() const {
     if (final_output_) {
-      return {{key_name_, DType::Int64, true},
-              {"sum", DType::Float64, true},
-              {"cnt", DType::Int64, false}};
+      std::vector<OutField> f = {{key_name_, DType::Int64, true}};
+      for (size_t i = 0; i < agg_kinds_.size(); i++) {
+        if (agg_kinds_[i] == AGGL_CNT)
+          f.push_back({agg_names_[i], DType::Int64, false});
+        else
+          f.push_back({agg_names_[i], DType::Float64, true});
+      }
+      return f;
     }
     // partial/partial-merge: grouping + AGG_BUF (agg/mod.rs:37)
     return {{key_name_, DType::Int64, true},
@@ -730,19 +758,42 @@ class AggOp {
     AURON_HIP(hipMemcpyAsync(sv.data(), svalid.get(), bm, hipMemcpyDeviceToHost,
                              stream_));
     if (final_output_) {
-      HostOutCol sum_col, cnt_col;
-      sum_col.dt = DType::Float64;
-      cnt_col.dt = DType::Int64;
-      d2h(sums, &sum_col.values, n * 8);
-      d2h(cnts, &cnt_col.values, n * 8);
+      std::vector<uint8_t> h_sums(n * 8), h_cnts(n * 8), h_avgs;
+      AURON_HIP(hipMemcpyAsync(h_sums.data(), sums.get(), n * 8,
+                               hipMemcpyDeviceToHost, stream_));
+      AURON_HIP(hipMemcpyAsync(h_cnts.data(), cnts.get(), n * 8,
+                               hipMemcpyDeviceToHost, stream_));
+      bool need_avg = false;
+      for (uint32_t k : agg_kinds_) need_avg |= (k == AGGL_AVG);
+      DevBuf avgs;
+      if (need_avg) {
+        avgs.alloc(n * 8);
+        launch_avg_div(sums.get<double>(), cnts.get<long long>(), n,
+                       avgs.get<double>(), stream_);
+        h_avgs.resize(n * 8);
+        AURON_HIP(hipMemcpyAsync(h_avgs.data(), avgs.get(), n * 8,
+                                 hipMemcpyDeviceToHost, stream_));
+      }
       AURON_HIP(hipStreamSynchronize(stream_));
       attach_validity(&key_col, kv, n);
-      attach_validity(&sum_col, sv, n);
-      cols = {std::move(key_col), std::move(sum_col), std::move(cnt_col)};
+      cols.push_back(std::move(key_col));
+      for (uint32_t k : agg_kinds_) {
+        HostOutCol ac;
+        if (k == AGGL_CNT) {
+          ac.dt = DType::Int64;
+          ac.values = h_cnts;
+        } else {
+          ac.dt = DType::Float64;
+          ac.values = (k == AGGL_AVG) ? h_avgs : h_sums;
+          attach_validity(&ac, sv, n);  // null iff cnt==0 (shared column)
+        }
+        cols.push_back(std::move(ac));
+      }
     } else {
       // freeze (a8): lens → host scan → write
       DevBuf lens(n * 4), offs((n + 1) * 4);
-      launch_agg_freeze_len(t_, order_slots, n, lens.get<int32_t>(), stream_);
+      launch_agg_freeze_len(t_, order_slots, n, lens.get<int32_t>(), layout_,
+                            stream_);
       std::vector<int32_t> h_lens(n);
       AURON_HIP(hipMemcpyAsync(h_lens.data(), lens.get(), n * 4,
                                hipMemcpyDeviceToHost, stream_));
@@ -753,7 +804,7 @@ class AggOp {
                                hipMemcpyHostToDevice, stream_));
       DevBuf data(h_offs[n] ? h_offs[n] : 1);
       launch_agg_freeze_write(t_, order_slots, n, offs.get<int32_t>(),
-                              data.get<uint8_t>(), stream_);
+                              data.get<uint8_t>(), layout_, stream_);
       HostOutCol buf_col;
       buf_col.dt = DType::Binary;
       buf_col.offsets = std::move(h_offs);
@@ -773,7 +824,8 @@ class AggOp {
     const DevColumn& key = b.cols.at(key_col_);
     const DevColumn& val = b.cols.at(val_col_);
     DevBuf lens(n * 4), offs((n + 1) * 4);
-    launch_skip_freeze_len(val.validity, n, lens.get<int32_t>(), stream_);
+    launch_skip_freeze_len(val.validity, n, lens.get<int32_t>(), layout_,
+                           stream_);
     std::vector<int32_t> h_lens(n);
     AURON_HIP(hipMemcpyAsync(h_lens.data(), lens.get(), n * 4,
                              hipMemcpyDeviceToHost, stream_));
@@ -784,7 +836,8 @@ class AggOp {
                              hipMemcpyHostToDevice, stream_));
     DevBuf data(h_offs[n] ? h_offs[n] : 1);
     launch_skip_freeze_write((const double*)val.values, val.validity, n,
-                             offs.get<int32_t>(), data.get<uint8_t>(), stream_);
+                             offs.get<int32_t>(), data.get<uint8_t>(), layout_,
+                             stream_);
     HostOutCol key_col, buf_col;
     key_col.dt = DType::Int64;
     key_col.values.resize(n * 8);
@@ -828,6 +881,9 @@ class AggOp {
   AggMode mode_ = AggMode::Partial;
   bool merge_mode_ = false, final_output_ = false;
   uint32_t key_col_ = 0, val_col_ = 0;
+  uint32_t layout_ = 0;
+  std::vector<uint32_t> agg_kinds_;
+  std::vector<std::string> agg_names_;
   std::string key_name_;
   int64_t batch_size_ = 10000;
   bool skip_enabled_ = false, skipping_ = false;

@@ -50,18 +50,29 @@ void launch_agg_update(const AggTable& t, const int64_t* keys,
                        const uint8_t* val_valid, int64_t n, uint64_t row_offset,
                        hipStream_t s);
 
-// merge rows of frozen partial state: per row [u8 valid][8B LE sum]?[varint cnt]
+// agg layout descriptor: 2 bits per agg LSB-first, 0-terminated
+// (AGGL_SUM=1, AGGL_CNT=2, AGGL_AVG=3); all aggs share one argument column.
+// AVG freeze = sum ++ count (avg.rs:208-217), so duplicated parts repeat the
+// shared accumulators.
+enum AggLayoutKind : uint32_t { AGGL_SUM = 1, AGGL_CNT = 2, AGGL_AVG = 3 };
+
+// merge rows of frozen partial state per the layout
 // (acc.rs:335-347 + count.rs:193-211 + io/mod.rs:60-79)
 void launch_agg_merge_frozen(const AggTable& t, const int64_t* keys,
                              const uint8_t* key_valid, const uint8_t* acc_data,
                              const int32_t* acc_offsets, int64_t n,
-                             uint64_t row_offset, hipStream_t s);
+                             uint64_t row_offset, uint32_t layout,
+                             hipStream_t s);
 
 // compact occupied slots to dense arrays (unordered); returns count via
 // num_out (device). out_slot holds the source slot index per group.
 void launch_agg_compact(const AggTable& t, uint32_t* out_slot,
                         unsigned long long* out_first_row,
                         unsigned long long* num_out, hipStream_t s);
+
+// divide sums by counts for AVG output columns (avg final = sum/count)
+void launch_avg_div(const double* sums, const long long* cnts, int64_t n,
+                    double* out, hipStream_t s);
 
 // gather ordered group outputs given order[] (group -> slot):
 // keys + key validity bits, sums + validity bits, counts
@@ -74,17 +85,18 @@ void launch_agg_gather_out(const AggTable& t, const uint32_t* order_slots,
 // freeze ordered groups into the Binary agg-buf wire format (a8):
 // pass 1 computes per-group byte length, host scans, pass 2 writes bytes.
 void launch_agg_freeze_len(const AggTable& t, const uint32_t* order_slots,
-                           int64_t num_groups, int32_t* lens, hipStream_t s);
+                           int64_t num_groups, int32_t* lens, uint32_t layout,
+                           hipStream_t s);
 void launch_agg_freeze_write(const AggTable& t, const uint32_t* order_slots,
                              int64_t num_groups, const int32_t* offsets,
-                             uint8_t* out, hipStream_t s);
+                             uint8_t* out, uint32_t layout, hipStream_t s);
 
 // partial-skipping pass-through freeze (agg_ctx.rs:428-462)
 void launch_skip_freeze_len(const uint8_t* val_valid, int64_t n, int32_t* lens,
-                            hipStream_t s);
+                            uint32_t layout, hipStream_t s);
 void launch_skip_freeze_write(const double* vals, const uint8_t* val_valid,
                               int64_t n, const int32_t* offsets, uint8_t* out,
-                              hipStream_t s);
+                              uint32_t layout, hipStream_t s);
 
 // fill an i64 array with a constant
 void launch_fill_i64(int64_t* dst, int64_t value, int64_t n, hipStream_t s);

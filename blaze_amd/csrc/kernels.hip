@@ -279,12 +279,100 @@ __device__ __forceinline__ uint64_t read_varint_dev(const uint8_t* p, int* used)
   return len;
 }
 
+__device__ __forceinline__ int varint_len_dev(uint64_t v) {
+  int k = 1;
+  while (v >= 128) {
+    v /= 128;
+    k++;
+  }
+  return k;
+}
+
+// agg layout: 2 bits per agg, LSB-first, 0-terminated. All aggs share one
+// argument column, so SUM/AVG parts duplicate the same sum and COUNT/AVG
+// parts the same count (avg.rs:208-217: AVG freeze = sum ++ count).
+// AGGL_SUM=1, AGGL_CNT=2, AGGL_AVG=3.
+__device__ __forceinline__ int agg_freeze_len(uint32_t layout, bool valid,
+                                              uint64_t cnt) {
+  int len = 0;
+  for (uint32_t l = layout; l & 3u; l >>= 2) {
+    uint32_t k = l & 3u;
+    if (k == 1 || k == 3) len += 1 + (valid ? 8 : 0);
+    if (k == 2 || k == 3) len += varint_len_dev(cnt);
+  }
+  return len;
+}
+
+__device__ __forceinline__ uint8_t* agg_freeze_write_rec(uint32_t layout,
+                                                         bool valid, double sum,
+                                                         uint64_t cnt,
+                                                         uint8_t* p) {
+  for (uint32_t l = layout; l & 3u; l >>= 2) {
+    uint32_t k = l & 3u;
+    if (k == 1 || k == 3) {  // acc.rs:335-347 prim freeze
+      if (valid) {
+        *p++ = 1;
+        memcpy(p, &sum, 8);
+        p += 8;
+      } else {
+        *p++ = 0;
+      }
+    }
+    if (k == 2 || k == 3) {  // count.rs:193-203 varint
+      uint64_t c = cnt;
+      while (c >= 128) {
+        *p++ = (uint8_t)(128 + c % 128);
+        c /= 128;
+      }
+      *p++ = (uint8_t)c;
+    }
+  }
+  return p;
+}
+
+// parse one frozen record; accumulates only the FIRST sum part and FIRST
+// count part (the rest are duplicates of the same shared-column accs)
+__device__ __forceinline__ void agg_parse_frozen(uint32_t layout,
+                                                 const uint8_t* p, bool* valid,
+                                                 double* sum, uint64_t* cnt) {
+  *valid = false;
+  *sum = 0.0;
+  *cnt = 0;
+  bool got_sum = false, got_cnt = false;
+  for (uint32_t l = layout; l & 3u; l >>= 2) {
+    uint32_t k = l & 3u;
+    if (k == 1 || k == 3) {
+      uint8_t v = *p++;
+      double x = 0;
+      if (v) {
+        memcpy(&x, p, 8);
+        p += 8;
+      }
+      if (!got_sum) {
+        *valid = v != 0;
+        *sum = x;
+        got_sum = true;
+      }
+    }
+    if (k == 2 || k == 3) {
+      int used;
+      uint64_t c = read_varint_dev(p, &used);
+      p += used;
+      if (!got_cnt) {
+        *cnt = c;
+        got_cnt = true;
+      }
+    }
+  }
+}
+
 __global__ void k_agg_merge_frozen(const AggTable t,
                                    const int64_t* __restrict__ keys,
                                    const uint8_t* __restrict__ key_valid,
                                    const uint8_t* __restrict__ acc_data,
                                    const int32_t* __restrict__ acc_offsets,
-                                   int64_t n, uint64_t row_offset) {
+                                   int64_t n, uint64_t row_offset,
+                                   uint32_t layout) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     bool knull = key_valid && !bit_get_dev(key_valid, i);
@@ -293,16 +381,11 @@ __global__ void k_agg_merge_frozen(const AggTable t,
     AggSlot* sl = &t.slots[a];
     uint64_t row = row_offset + (uint64_t)i;
     if (sl->first_row > row) atomicMin(&sl->first_row, row);
-    const uint8_t* p = acc_data + acc_offsets[i];
-    uint8_t valid = *p++;
-    if (valid) {  // sum.rs:117-145 partial_merge
-      double v;
-      memcpy(&v, p, 8);
-      p += 8;
-      unsafeAtomicAdd(&sl->sum, v);
-    }
-    int used;
-    uint64_t c = read_varint_dev(p, &used);
+    bool valid;
+    double v;
+    uint64_t c;
+    agg_parse_frozen(layout, acc_data + acc_offsets[i], &valid, &v, &c);
+    if (valid) unsafeAtomicAdd(&sl->sum, v);  // sum.rs:117-145 partial_merge
     if (c) atomicAdd(&sl->cnt, c);
   }
 }
@@ -352,22 +435,14 @@ __global__ void k_agg_gather_out(const AggTable t,
   }
 }
 
-__device__ __forceinline__ int varint_len_dev(uint64_t v) {
-  int k = 1;
-  while (v >= 128) {
-    v /= 128;
-    k++;
-  }
-  return k;
-}
-
 __global__ void k_agg_freeze_len(const AggTable t,
                                  const uint32_t* __restrict__ order_slots,
-                                 int64_t num_groups, int32_t* __restrict__ lens) {
+                                 int64_t num_groups, int32_t* __restrict__ lens,
+                                 uint32_t layout) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < num_groups;
        i += (int64_t)gridDim.x * blockDim.x) {
     uint32_t s = order_slots[i];
-    lens[i] = 1 + (t.slots[s].cnt != 0 ? 8 : 0) + varint_len_dev(t.slots[s].cnt);
+    lens[i] = agg_freeze_len(layout, t.slots[s].cnt != 0, t.slots[s].cnt);
   }
 }
 
@@ -375,26 +450,12 @@ __global__ void k_agg_freeze_write(const AggTable t,
                                    const uint32_t* __restrict__ order_slots,
                                    int64_t num_groups,
                                    const int32_t* __restrict__ offsets,
-                                   uint8_t* __restrict__ out) {
+                                   uint8_t* __restrict__ out, uint32_t layout) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < num_groups;
        i += (int64_t)gridDim.x * blockDim.x) {
     uint32_t s = order_slots[i];
-    uint8_t* p = out + offsets[i];
-    // acc.rs:335-347: [u8 valid][8B LE value]? then count.rs:193-203 varint
-    if (t.slots[s].cnt != 0) {
-      *p++ = 1;
-      double v = t.slots[s].sum;
-      memcpy(p, &v, 8);
-      p += 8;
-    } else {
-      *p++ = 0;
-    }
-    uint64_t c = t.slots[s].cnt;
-    while (c >= 128) {
-      *p++ = (uint8_t)(128 + c % 128);
-      c /= 128;
-    }
-    *p++ = (uint8_t)c;
+    agg_freeze_write_rec(layout, t.slots[s].cnt != 0, t.slots[s].sum,
+                         t.slots[s].cnt, out + offsets[i]);
   }
 }
 
@@ -409,9 +470,11 @@ void launch_agg_update(const AggTable& t, const int64_t* keys,
 void launch_agg_merge_frozen(const AggTable& t, const int64_t* keys,
                              const uint8_t* key_valid, const uint8_t* acc_data,
                              const int32_t* acc_offsets, int64_t n,
-                             uint64_t row_offset, hipStream_t s) {
+                             uint64_t row_offset, uint32_t layout,
+                             hipStream_t s) {
   hipLaunchKernelGGL(k_agg_merge_frozen, dim3(grid_for(n)), dim3(BLOCK), 0, s, t,
-                     keys, key_valid, acc_data, acc_offsets, n, row_offset);
+                     keys, key_valid, acc_data, acc_offsets, n, row_offset,
+                     layout);
   check_launch("k_agg_merge_frozen");
 }
 void launch_agg_compact(const AggTable& t, uint32_t* out_slot,
@@ -432,16 +495,17 @@ void launch_agg_gather_out(const AggTable& t, const uint32_t* order_slots,
   check_launch("k_agg_gather_out");
 }
 void launch_agg_freeze_len(const AggTable& t, const uint32_t* order_slots,
-                           int64_t num_groups, int32_t* lens, hipStream_t s) {
+                           int64_t num_groups, int32_t* lens, uint32_t layout,
+                           hipStream_t s) {
   hipLaunchKernelGGL(k_agg_freeze_len, dim3(grid_for(num_groups)), dim3(BLOCK),
-                     0, s, t, order_slots, num_groups, lens);
+                     0, s, t, order_slots, num_groups, lens, layout);
   check_launch("k_agg_freeze_len");
 }
 void launch_agg_freeze_write(const AggTable& t, const uint32_t* order_slots,
                              int64_t num_groups, const int32_t* offsets,
-                             uint8_t* out, hipStream_t s) {
+                             uint8_t* out, uint32_t layout, hipStream_t s) {
   hipLaunchKernelGGL(k_agg_freeze_write, dim3(grid_for(num_groups)), dim3(BLOCK),
-                     0, s, t, order_slots, num_groups, offsets, out);
+                     0, s, t, order_slots, num_groups, offsets, out, layout);
   check_launch("k_agg_freeze_write");
 }
 
@@ -449,45 +513,38 @@ void launch_agg_freeze_write(const AggTable& t, const uint32_t* order_slots,
 // process_partial_skipped): each input row becomes one record; SUM acc =
 // value when valid, COUNT = valid ? 1 : 0.
 __global__ void k_skip_freeze_len(const uint8_t* __restrict__ val_valid,
-                                  int64_t n, int32_t* __restrict__ lens) {
+                                  int64_t n, int32_t* __restrict__ lens,
+                                  uint32_t layout) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     bool v = !val_valid || bit_get_dev(val_valid, i);
-    lens[i] = v ? 10 : 2;  // [1][8B sum][varint 1] vs [0][varint 0]
+    lens[i] = agg_freeze_len(layout, v, v ? 1 : 0);
   }
 }
 
 __global__ void k_skip_freeze_write(const double* __restrict__ vals,
                                     const uint8_t* __restrict__ val_valid,
                                     int64_t n, const int32_t* __restrict__ offsets,
-                                    uint8_t* __restrict__ out) {
+                                    uint8_t* __restrict__ out, uint32_t layout) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
-    uint8_t* p = out + offsets[i];
     bool v = !val_valid || bit_get_dev(val_valid, i);
-    if (v) {
-      p[0] = 1;
-      double x = vals[i];
-      memcpy(p + 1, &x, 8);
-      p[9] = 1;
-    } else {
-      p[0] = 0;
-      p[1] = 0;
-    }
+    agg_freeze_write_rec(layout, v, v ? vals[i] : 0.0, v ? 1 : 0,
+                         out + offsets[i]);
   }
 }
 
 void launch_skip_freeze_len(const uint8_t* val_valid, int64_t n, int32_t* lens,
-                            hipStream_t s) {
+                            uint32_t layout, hipStream_t s) {
   hipLaunchKernelGGL(k_skip_freeze_len, dim3(grid_for(n)), dim3(BLOCK), 0, s,
-                     val_valid, n, lens);
+                     val_valid, n, lens, layout);
   check_launch("k_skip_freeze_len");
 }
 void launch_skip_freeze_write(const double* vals, const uint8_t* val_valid,
                               int64_t n, const int32_t* offsets, uint8_t* out,
-                              hipStream_t s) {
+                              uint32_t layout, hipStream_t s) {
   hipLaunchKernelGGL(k_skip_freeze_write, dim3(grid_for(n)), dim3(BLOCK), 0, s,
-                     vals, val_valid, n, offsets, out);
+                     vals, val_valid, n, offsets, out, layout);
   check_launch("k_skip_freeze_write");
 }
 
@@ -676,6 +733,21 @@ void launch_gather_bytes(const uint8_t* src_data, const int32_t* src_offsets,
   hipLaunchKernelGGL(k_gather_bytes, dim3(grid_for(n * 64)), dim3(BLOCK), 0, s,
                      src_data, src_offsets, perm, dst_offsets, n, dst_data);
   check_launch("k_gather_bytes");
+}
+
+__global__ void k_avg_div(const double* __restrict__ sums,
+                          const long long* __restrict__ cnts, int64_t n,
+                          double* __restrict__ out) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    out[i] = cnts[i] ? sums[i] / (double)cnts[i] : 0.0;
+}
+
+void launch_avg_div(const double* sums, const long long* cnts, int64_t n,
+                    double* out, hipStream_t s) {
+  hipLaunchKernelGGL(k_avg_div, dim3(grid_for(n)), dim3(BLOCK), 0, s, sums,
+                     cnts, n, out);
+  check_launch("k_avg_div");
 }
 
 // ---- filter ----------------------------------------------------------------

@@ -250,3 +250,46 @@ def test_shuffle_write_files(tmp_path):
     got_index = np.frombuffer(open(index_file, "rb").read(), dtype="<u8")
     np.testing.assert_array_equal(got_index, np.array(exp_index, np.uint64))
     assert got_data == bytes(exp_bytes)
+
+
+def test_avg_agg_layouts():
+    """Generalized agg lists over one shared column: [AVG], [SUM,COUNT,AVG]
+    (AVG freeze = sum ++ count, avg.rs:208-217; final = sum/count)."""
+    keys, vals, vv = gen_northstar(100_000, nkeys=500)
+    orc = run_oracle(keys, vals, vv)
+    ref = orc.output()
+    exp_avg = np.where(ref["counts"] > 0, ref["sums"] / np.maximum(ref["counts"], 1), 0)
+
+    def avg_aggs():
+        return [plan.agg_expr(plan.AGG_AVG, [plan.column("val", 1)],
+                              plan.DT_FLOAT64)]
+
+    def sca_aggs():
+        return [plan.agg_expr(plan.AGG_SUM, [plan.column("val", 1)],
+                              plan.DT_FLOAT64),
+                plan.agg_expr(plan.AGG_COUNT, [plan.column("val", 1)],
+                              plan.DT_INT64),
+                plan.agg_expr(plan.AGG_AVG, [plan.column("val", 1)],
+                              plan.DT_FLOAT64)]
+
+    for aggs_fn, names in ((avg_aggs, ["avg"]),
+                           (sca_aggs, ["sum", "cnt", "avg"])):
+        reader = plan.ffi_reader(plan.northstar_input_fields(), "input0")
+        partial = plan.agg(reader, [plan.column("key", 0)], aggs_fn(),
+                           [plan.MODE_PARTIAL] * len(names), ["key"], names)
+        final = plan.agg(partial, [plan.column("key", 0)], aggs_fn(),
+                         [plan.MODE_FINAL] * len(names), ["key"], names)
+        t = blaze_amd.Task(plan.task_definition(final),
+                           batches=batches_of(keys, vals, vv))
+        outs = t.run()
+        got_keys = np.concatenate([ob[0]["values"] for ob in outs])
+        np.testing.assert_array_equal(got_keys, ref["keys"])
+        got_avg = np.concatenate([ob[len(names)]["values"] for ob in outs])
+        np.testing.assert_allclose(got_avg[ref["counts"] > 0],
+                                   exp_avg[ref["counts"] > 0], rtol=1e-13)
+        if len(names) == 3:
+            got_sums = np.concatenate([ob[1]["values"] for ob in outs])
+            got_cnts = np.concatenate([ob[2]["values"] for ob in outs])
+            np.testing.assert_array_equal(got_sums, ref["sums"])
+            np.testing.assert_array_equal(got_cnts, ref["counts"])
+        t.finalize()
