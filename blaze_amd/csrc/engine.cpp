@@ -22,6 +22,7 @@
 #include "kernels.h"
 #include "plan.h"
 #include "ipc_scalar.h"
+#include "parquet.h"
 #include "serde_host.h"
 
 namespace auron {
@@ -1418,13 +1419,16 @@ struct Runtime {
         case PlanNode::Filter: p = p->filter->input.get(); break;
         case PlanNode::Projection: p = p->projection->input.get(); break;
         case PlanNode::IpcReader:
+        case PlanNode::ParquetScan:
         case PlanNode::FFIReader: p = nullptr; break;
       }
     }
     // leaf must be a source: FFIReader (Arrow in) or IpcReader (shuffle bytes)
     const PlanNode* leaf = chain.back();
-    if (leaf->kind != PlanNode::FFIReader && leaf->kind != PlanNode::IpcReader)
-      FAIL("plan leaf must be FFIReaderExec/IpcReaderExec on this path");
+    if (leaf->kind != PlanNode::FFIReader &&
+        leaf->kind != PlanNode::IpcReader &&
+        leaf->kind != PlanNode::ParquetScan)
+      FAIL("plan leaf must be FFIReader/IpcReader/ParquetScan on this path");
 
     // middle ops (leaf-1 ... root), in execution order
     stages_.clear();
@@ -1486,8 +1490,10 @@ struct Runtime {
         input_rows += b.num_rows;
         feed(0, std::move(b));
       }
-    } else {
+    } else if (leaf->kind == PlanNode::IpcReader) {
       input_rows = pump_ipc_reader(*leaf->ipc_reader);
+    } else {
+      input_rows = pump_parquet(*leaf->parquet);
     }
     // drain chain: pipeline breakers emit, transforms pass through
     AggOp* last_agg = nullptr;
@@ -1530,6 +1536,111 @@ struct Runtime {
         std::chrono::duration_cast<std::chrono::nanoseconds>(
             std::chrono::steady_clock::now() - t0)
             .count();
+  }
+
+  // ParquetScanExec source (parquet_exec.rs:150-203; decode scope per
+  // parquet.h): footer/pages on host, validity/dictionary expansion on GPU.
+  int64_t pump_parquet(const ParquetScanNode& node) {
+    int64_t input_rows = 0;
+    DevBuf scan_tmp;
+    for (const std::string& path : node.files) {
+      ParquetFile pf(path);
+      const auto& fcols = pf.columns();
+      std::vector<uint32_t> proj = node.projection;
+      if (proj.empty())
+        for (uint32_t i = 0; i < fcols.size(); i++) proj.push_back(i);
+      for (int rg = 0; rg < pf.num_row_groups(); rg++) {
+        int64_t rows = pf.row_group_rows(rg);
+        DevBatch b;
+        b.num_rows = rows;
+        for (uint32_t ci : proj) {
+          if (ci >= fcols.size()) FAIL("parquet: projection out of range");
+          PqColumnChunkData cd = pf.read_chunk(rg, (int)ci);
+          const int w = (int)dtype_width(fcols[ci].dtype());
+          DevColumn c;
+          c.dt = fcols[ci].dtype();
+          c.len = rows;
+          c.own_values.alloc((size_t)rows * w);
+          bool has_nulls = cd.null_count > 0;
+          DevBuf d_valid, d_mask, d_positions, d_packed, d_indices;
+          if (has_nulls) {
+            c.own_validity.alloc(cd.validity.size());
+            AURON_HIP(hipMemcpyAsync(c.own_validity.get(), cd.validity.data(),
+                                     cd.validity.size(), hipMemcpyHostToDevice,
+                                     stream));
+            c.validity = c.own_validity.get<uint8_t>();
+            d_mask.alloc(rows);
+            launch_bits_to_mask(c.validity, rows, d_mask.get<uint8_t>(),
+                                stream);
+            d_positions.alloc((rows + 1) * 4);
+            size_t tb = 0;
+            scan_mask_u8(d_mask.get<uint8_t>(), d_positions.get<uint32_t>(),
+                         rows, nullptr, &tb, stream);
+            if (tb > scan_tmp.size()) scan_tmp.alloc(tb);
+            scan_mask_u8(d_mask.get<uint8_t>(), d_positions.get<uint32_t>(),
+                         rows, scan_tmp.get(), &tb, stream);
+          }
+          if (!cd.uses_dict) {
+            // PLAIN: non-null values packed densely
+            if (!has_nulls) {
+              AURON_HIP(hipMemcpyAsync(c.own_values.get(), cd.plain.data(),
+                                       cd.plain.size(), hipMemcpyHostToDevice,
+                                       stream));
+            } else {
+              d_packed.alloc(cd.plain.empty() ? 1 : cd.plain.size());
+              if (!cd.plain.empty())
+                AURON_HIP(hipMemcpyAsync(d_packed.get(), cd.plain.data(),
+                                         cd.plain.size(),
+                                         hipMemcpyHostToDevice, stream));
+              launch_scatter_packed(w, d_packed.get<uint8_t>(),
+                                    d_positions.get<uint32_t>(),
+                                    d_mask.get<uint8_t>(), rows,
+                                    c.own_values.get<uint8_t>(), stream);
+            }
+          } else {
+            // dictionary: expand packed indices to rows, gather dict values
+            d_indices.alloc((size_t)rows * 4);
+            DevBuf d_idx_packed(cd.dict_indices.empty()
+                                    ? 4
+                                    : cd.dict_indices.size() * 4);
+            if (!cd.dict_indices.empty())
+              AURON_HIP(hipMemcpyAsync(d_idx_packed.get(),
+                                       cd.dict_indices.data(),
+                                       cd.dict_indices.size() * 4,
+                                       hipMemcpyHostToDevice, stream));
+            if (has_nulls) {
+              launch_scatter_packed(4, d_idx_packed.get<uint8_t>(),
+                                    d_positions.get<uint32_t>(),
+                                    d_mask.get<uint8_t>(), rows,
+                                    d_indices.get<uint8_t>(), stream);
+            } else {
+              AURON_HIP(hipMemcpyAsync(d_indices.get(), d_idx_packed.get(),
+                                       (size_t)rows * 4,
+                                       hipMemcpyDeviceToDevice, stream));
+            }
+            DevBuf d_dict(cd.dict_values.empty() ? 1 : cd.dict_values.size());
+            if (!cd.dict_values.empty())
+              AURON_HIP(hipMemcpyAsync(d_dict.get(), cd.dict_values.data(),
+                                       cd.dict_values.size(),
+                                       hipMemcpyHostToDevice, stream));
+            if (w == 8)
+              launch_gather_8(d_dict.get<uint8_t>(),
+                              d_indices.get<uint32_t>(), rows,
+                              c.own_values.get<uint8_t>(), stream);
+            else
+              launch_gather_4(d_dict.get<uint8_t>(),
+                              d_indices.get<uint32_t>(), rows,
+                              c.own_values.get<uint8_t>(), stream);
+          }
+          AURON_HIP(hipStreamSynchronize(stream));  // host cd dies here
+          c.values = c.own_values.get();
+          b.cols.push_back(std::move(c));
+        }
+        input_rows += rows;
+        feed(0, std::move(b));
+      }
+    }
+    return input_rows;
   }
 
   // IpcReaderExec source (ipc_reader_exec.rs:62-120): raw shuffle block
@@ -1856,6 +1967,14 @@ int32_t auron_debug_decode_plan(const uint8_t* data, size_t len, char* out,
           s += "Project(ncols=" +
                std::to_string(p->projection->exprs.size()) + ")->";
           p = p->projection->input.get();
+          break;
+        }
+        case PlanNode::ParquetScan: {
+          const auto& pq = *p->parquet;
+          s += "ParquetScan(nfiles=" + std::to_string(pq.files.size()) +
+               ",nfields=" + std::to_string(pq.schema.fields.size()) +
+               ",nproj=" + std::to_string(pq.projection.size()) + ")";
+          p = nullptr;
           break;
         }
         case PlanNode::IpcReader: {

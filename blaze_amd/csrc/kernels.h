@@ -125,6 +125,16 @@ void launch_sel_rows(const uint8_t* mask, const uint32_t* positions, int64_t n,
 void launch_gather_4(const uint8_t* src, const uint32_t* perm, int64_t n,
                      uint8_t* dst, hipStream_t s);  // 4-byte elements
 
+// ---- parquet decode helpers (parquet.cpp host side feeds these) ------------
+// expand an LSB validity bitmap into a per-row u8 mask
+void launch_bits_to_mask(const uint8_t* bits, int64_t n, uint8_t* mask,
+                         hipStream_t s);
+// scatter densely-packed non-null values into row slots:
+// out[i] = mask[i] ? packed[positions[i]] : 0  (positions = excl. scan of mask)
+void launch_scatter_packed(int width, const uint8_t* packed,
+                           const uint32_t* positions, const uint8_t* mask,
+                           int64_t n, uint8_t* out, hipStream_t s);
+
 // ---- two-phase aggregation (kernels_agg2.hip) ------------------------------
 // radix-partition rows into buckets whose groups fit in LDS, aggregate each
 // bucket in LDS, merge the counted per-bucket group lists into the table.

@@ -859,6 +859,47 @@ void launch_gather_4(const uint8_t* src, const uint32_t* perm, int64_t n,
   check_launch("k_gather_4");
 }
 
+__global__ void k_bits_to_mask(const uint8_t* __restrict__ bits, int64_t n,
+                               uint8_t* __restrict__ mask) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    mask[i] = bit_get_dev(bits, i);
+}
+
+template <typename T>
+__global__ void k_scatter_packed(const T* __restrict__ packed,
+                                 const uint32_t* __restrict__ positions,
+                                 const uint8_t* __restrict__ mask, int64_t n,
+                                 T* __restrict__ out) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    out[i] = mask[i] ? packed[positions[i]] : (T)0;
+}
+
+void launch_bits_to_mask(const uint8_t* bits, int64_t n, uint8_t* mask,
+                         hipStream_t s) {
+  hipLaunchKernelGGL(k_bits_to_mask, dim3(grid_for(n)), dim3(BLOCK), 0, s, bits,
+                     n, mask);
+  check_launch("k_bits_to_mask");
+}
+
+void launch_scatter_packed(int width, const uint8_t* packed,
+                           const uint32_t* positions, const uint8_t* mask,
+                           int64_t n, uint8_t* out, hipStream_t s) {
+  if (width == 8) {
+    hipLaunchKernelGGL(k_scatter_packed<uint64_t>, dim3(grid_for(n)),
+                       dim3(BLOCK), 0, s, (const uint64_t*)packed, positions,
+                       mask, n, (uint64_t*)out);
+  } else if (width == 4) {
+    hipLaunchKernelGGL(k_scatter_packed<uint32_t>, dim3(grid_for(n)),
+                       dim3(BLOCK), 0, s, (const uint32_t*)packed, positions,
+                       mask, n, (uint32_t*)out);
+  } else {
+    throw std::runtime_error("scatter_packed: unsupported width");
+  }
+  check_launch("k_scatter_packed");
+}
+
 __global__ void k_byte_transpose(const uint8_t* __restrict__ in,
                                  uint8_t* __restrict__ out, int64_t n, int w) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n * w;

@@ -1,0 +1,599 @@
+#include "parquet.h"
+
+#include <dlfcn.h>
+
+#include <cstdio>
+#include <cstring>
+#include <stdexcept>
+
+#include "lz4shim.h"
+
+namespace auron {
+namespace {
+
+[[noreturn]] void fail(const std::string& m) { throw std::runtime_error(m); }
+
+// ---- Thrift compact protocol (only what parquet.thrift needs) --------------
+struct TReader {
+  const uint8_t* p;
+  const uint8_t* end;
+
+  uint64_t uvarint() {
+    uint64_t v = 0;
+    int shift = 0;
+    while (p < end) {
+      uint8_t b = *p++;
+      v |= (uint64_t)(b & 0x7f) << shift;
+      if (!(b & 0x80)) return v;
+      shift += 7;
+    }
+    fail("thrift: truncated varint");
+  }
+  int64_t zigzag() {
+    uint64_t v = uvarint();
+    return (int64_t)(v >> 1) ^ -(int64_t)(v & 1);
+  }
+
+  // returns (field_id, type); type 0 = STOP
+  std::pair<int, int> field_header(int* last_id) {
+    if (p >= end) fail("thrift: truncated struct");
+    uint8_t b = *p++;
+    int type = b & 0x0f;
+    if (type == 0) return {0, 0};
+    int delta = (b >> 4) & 0x0f;
+    int id = delta ? *last_id + delta : (int)zigzag();
+    *last_id = id;
+    return {id, type};
+  }
+
+  void skip(int type) {
+    switch (type) {
+      case 1: case 2: break;                      // bool encoded in type
+      case 3: p++; break;                         // byte
+      case 4: case 5: case 6: zigzag(); break;    // i16/i32/i64
+      case 7: p += 8; break;                      // double
+      case 8: {                                   // binary
+        uint64_t n = uvarint();
+        p += n;
+        break;
+      }
+      case 9: case 10: {                          // list/set
+        uint8_t h = *p++;
+        uint64_t n = (h >> 4) == 15 ? uvarint() : (h >> 4);
+        int et = h & 0x0f;
+        for (uint64_t i = 0; i < n; i++) skip(et);
+        break;
+      }
+      case 11: {                                  // map
+        uint64_t n = uvarint();
+        if (n > 0) {
+          uint8_t kv = *p++;
+          for (uint64_t i = 0; i < n; i++) {
+            skip((kv >> 4) & 0x0f);
+            skip(kv & 0x0f);
+          }
+        }
+        break;
+      }
+      case 12: {                                  // struct
+        int last = 0;
+        while (true) {
+          auto [id, t] = field_header(&last);
+          if (t == 0) break;
+          skip(t);
+        }
+        break;
+      }
+      default: fail("thrift: unknown type " + std::to_string(type));
+    }
+    if (p > end) fail("thrift: overrun");
+  }
+
+  std::pair<uint64_t, int> list_header() {
+    uint8_t h = *p++;
+    uint64_t n = (h >> 4) == 15 ? uvarint() : (h >> 4);
+    return {n, h & 0x0f};
+  }
+
+  std::string binary() {
+    uint64_t n = uvarint();
+    if ((uint64_t)(end - p) < n) fail("thrift: truncated binary");
+    std::string s((const char*)p, n);
+    p += n;
+    return s;
+  }
+};
+
+struct SchemaElem {
+  int type = -1;
+  int repetition = -1;  // 0=REQUIRED 1=OPTIONAL 2=REPEATED
+  std::string name;
+  int num_children = 0;
+};
+
+SchemaElem read_schema_element(TReader& r) {
+  SchemaElem e;
+  int last = 0;
+  while (true) {
+    auto [id, t] = r.field_header(&last);
+    if (t == 0) break;
+    switch (id) {
+      case 1: e.type = (int)r.zigzag(); break;
+      case 3: e.repetition = (int)r.zigzag(); break;
+      case 4: e.name = r.binary(); break;
+      case 5: e.num_children = (int)r.zigzag(); break;
+      default: r.skip(t);
+    }
+  }
+  return e;
+}
+
+struct PageHeader {
+  int type = -1;  // 0 data, 2 dict, 3 data v2
+  int64_t uncompressed_size = 0, compressed_size = 0;
+  // data page v1
+  int num_values = 0, encoding = -1, def_encoding = -1;
+  // v2
+  int num_nulls = 0, num_rows = 0, def_len = 0, rep_len = 0;
+  bool v2_compressed = true;
+  // dict
+  int dict_num_values = 0;
+};
+
+PageHeader read_page_header(TReader& r) {
+  PageHeader h;
+  int last = 0;
+  while (true) {
+    auto [id, t] = r.field_header(&last);
+    if (t == 0) break;
+    switch (id) {
+      case 1: h.type = (int)r.zigzag(); break;
+      case 2: h.uncompressed_size = r.zigzag(); break;
+      case 3: h.compressed_size = r.zigzag(); break;
+      case 5: {  // DataPageHeader
+        int l2 = 0;
+        while (true) {
+          auto [id2, t2] = r.field_header(&l2);
+          if (t2 == 0) break;
+          switch (id2) {
+            case 1: h.num_values = (int)r.zigzag(); break;
+            case 2: h.encoding = (int)r.zigzag(); break;
+            case 3: h.def_encoding = (int)r.zigzag(); break;
+            default: r.skip(t2);
+          }
+        }
+        break;
+      }
+      case 7: {  // DictionaryPageHeader
+        int l2 = 0;
+        while (true) {
+          auto [id2, t2] = r.field_header(&l2);
+          if (t2 == 0) break;
+          if (id2 == 1) h.dict_num_values = (int)r.zigzag();
+          else r.skip(t2);
+        }
+        break;
+      }
+      case 8: {  // DataPageHeaderV2
+        int l2 = 0;
+        h.v2_compressed = true;
+        while (true) {
+          auto [id2, t2] = r.field_header(&l2);
+          if (t2 == 0) break;
+          switch (id2) {
+            case 1: h.num_values = (int)r.zigzag(); break;
+            case 2: h.num_nulls = (int)r.zigzag(); break;
+            case 3: h.num_rows = (int)r.zigzag(); break;
+            case 4: h.encoding = (int)r.zigzag(); break;
+            case 5: h.def_len = (int)r.zigzag(); break;
+            case 6: h.rep_len = (int)r.zigzag(); break;
+            case 7: h.v2_compressed = (t2 == 1); break;  // BOOL_TRUE
+            default: r.skip(t2);
+          }
+        }
+        h.type = 3;
+        break;
+      }
+      default: r.skip(t);
+    }
+  }
+  return h;
+}
+
+// ---- zstd (runtime binding; no headers in image) ---------------------------
+typedef size_t (*ZSTD_decompress_t)(void*, size_t, const void*, size_t);
+typedef unsigned (*ZSTD_isError_t)(size_t);
+
+bool zstd_uncompress(const uint8_t* src, size_t n, size_t out_size,
+                     std::vector<uint8_t>* out, std::string* err) {
+  static ZSTD_decompress_t dec = nullptr;
+  static ZSTD_isError_t iserr = nullptr;
+  if (!dec) {
+    void* h = dlopen("libzstd.so.1", RTLD_NOW | RTLD_GLOBAL);
+    if (!h) {
+      *err = "libzstd.so.1 not available";
+      return false;
+    }
+    dec = (ZSTD_decompress_t)dlsym(h, "ZSTD_decompress");
+    iserr = (ZSTD_isError_t)dlsym(h, "ZSTD_isError");
+  }
+  out->resize(out_size);
+  size_t rc = dec(out->data(), out_size, src, n);
+  if (iserr(rc) || rc != out_size) {
+    *err = "zstd decompress failed";
+    return false;
+  }
+  return true;
+}
+
+typedef int (*LZ4_decompress_safe_t)(const char*, char*, int, int);
+
+bool lz4raw_uncompress(const uint8_t* src, size_t n, size_t out_size,
+                       std::vector<uint8_t>* out, std::string* err) {
+  static LZ4_decompress_safe_t dec = nullptr;
+  if (!dec) {
+    void* h = dlopen("liblz4.so.1", RTLD_NOW | RTLD_GLOBAL);
+    if (!h) {
+      *err = "liblz4.so.1 not available";
+      return false;
+    }
+    dec = (LZ4_decompress_safe_t)dlsym(h, "LZ4_decompress_safe");
+  }
+  out->resize(out_size);
+  int rc = dec((const char*)src, (char*)out->data(), (int)n, (int)out_size);
+  if (rc < 0 || (size_t)rc != out_size) {
+    *err = "lz4 raw decompress failed";
+    return false;
+  }
+  return true;
+}
+
+// ---- RLE/bit-packed hybrid (parquet encoding spec) -------------------------
+void rle_bp_decode(const uint8_t* p, size_t len, int bit_width, int64_t count,
+                   std::vector<uint32_t>* out) {
+  out->clear();
+  out->reserve(count);
+  size_t pos = 0;
+  const int byte_w = (bit_width + 7) / 8;
+  while ((int64_t)out->size() < count) {
+    if (pos >= len) fail("rle: truncated stream");
+    uint64_t header = 0;
+    int shift = 0;
+    while (true) {
+      if (pos >= len) fail("rle: truncated varint");
+      uint8_t b = p[pos++];
+      header |= (uint64_t)(b & 0x7f) << shift;
+      if (!(b & 0x80)) break;
+      shift += 7;
+    }
+    if (header & 1) {
+      // bit-packed group: (header>>1) groups of 8 values
+      int64_t n = (int64_t)(header >> 1) * 8;
+      size_t bytes = (size_t)n * bit_width / 8;
+      if (pos + bytes > len) fail("rle: truncated bit-packed run");
+      uint64_t acc = 0;
+      int bits = 0;
+      size_t bp = pos;
+      for (int64_t i = 0; i < n && (int64_t)out->size() < count; i++) {
+        while (bits < bit_width) {
+          acc |= (uint64_t)p[bp++] << bits;
+          bits += 8;
+        }
+        out->push_back((uint32_t)(acc & ((1ull << bit_width) - 1)));
+        acc >>= bit_width;
+        bits -= bit_width;
+      }
+      pos += bytes;
+    } else {
+      int64_t n = (int64_t)(header >> 1);
+      uint32_t v = 0;
+      if (pos + byte_w > len) fail("rle: truncated repeated run");
+      for (int b = 0; b < byte_w; b++) v |= (uint32_t)p[pos + b] << (8 * b);
+      pos += byte_w;
+      for (int64_t i = 0; i < n && (int64_t)out->size() < count; i++)
+        out->push_back(v);
+    }
+  }
+}
+
+}  // namespace
+
+// ---- snappy ----------------------------------------------------------------
+bool snappy_uncompress(const uint8_t* src, size_t n, std::vector<uint8_t>* out,
+                       std::string* err) {
+  size_t pos = 0;
+  uint64_t out_len = 0;
+  int shift = 0;
+  while (pos < n) {
+    uint8_t b = src[pos++];
+    out_len |= (uint64_t)(b & 0x7f) << shift;
+    if (!(b & 0x80)) break;
+    shift += 7;
+  }
+  out->clear();
+  out->reserve(out_len);
+  while (pos < n) {
+    uint8_t tag = src[pos++];
+    int type = tag & 3;
+    if (type == 0) {  // literal
+      uint64_t len = (tag >> 2) + 1;
+      if (len > 60) {
+        int nb = (int)len - 60;
+        len = 0;
+        for (int i = 0; i < nb; i++) len |= (uint64_t)src[pos++] << (8 * i);
+        len += 1;
+      }
+      if (pos + len > n) {
+        *err = "snappy: truncated literal";
+        return false;
+      }
+      out->insert(out->end(), src + pos, src + pos + len);
+      pos += len;
+    } else {
+      uint64_t len, off;
+      if (type == 1) {  // copy with 1-byte offset
+        len = ((tag >> 2) & 7) + 4;
+        off = ((uint64_t)(tag >> 5) << 8) | src[pos++];
+      } else if (type == 2) {
+        len = (tag >> 2) + 1;
+        off = (uint64_t)src[pos] | ((uint64_t)src[pos + 1] << 8);
+        pos += 2;
+      } else {
+        len = (tag >> 2) + 1;
+        off = 0;
+        for (int i = 0; i < 4; i++) off |= (uint64_t)src[pos++] << (8 * i);
+      }
+      if (off == 0 || off > out->size()) {
+        *err = "snappy: bad copy offset";
+        return false;
+      }
+      size_t start = out->size() - off;
+      for (uint64_t i = 0; i < len; i++)
+        out->push_back((*out)[start + i]);  // may self-overlap
+    }
+  }
+  if (out->size() != out_len) {
+    *err = "snappy: length mismatch";
+    return false;
+  }
+  return true;
+}
+
+// ---- ParquetFile ------------------------------------------------------------
+ParquetFile::ParquetFile(const std::string& path) : path_(path) {
+  FILE* f = fopen(path.c_str(), "rb");
+  if (!f) fail("parquet: cannot open " + path);
+  fseek(f, 0, SEEK_END);
+  long sz = ftell(f);
+  fseek(f, 0, SEEK_SET);
+  file_.resize(sz);
+  if (fread(file_.data(), 1, sz, f) != (size_t)sz) {
+    fclose(f);
+    fail("parquet: short read " + path);
+  }
+  fclose(f);
+  if (sz < 12 || memcmp(file_.data() + sz - 4, "PAR1", 4) != 0)
+    fail("parquet: bad magic in " + path);
+  uint32_t meta_len;
+  memcpy(&meta_len, file_.data() + sz - 8, 4);
+  if ((long)meta_len + 12 > sz) fail("parquet: bad footer length");
+  const uint8_t* meta = file_.data() + sz - 8 - meta_len;
+
+  TReader r{meta, meta + meta_len};
+  int last = 0;
+  std::vector<SchemaElem> schema;
+  while (true) {
+    auto [id, t] = r.field_header(&last);
+    if (t == 0) break;
+    switch (id) {
+      case 2: {  // schema: list<SchemaElement>
+        auto [n, et] = r.list_header();
+        (void)et;
+        for (uint64_t i = 0; i < n; i++) schema.push_back(read_schema_element(r));
+        break;
+      }
+      case 4: {  // row_groups
+        auto [n, et] = r.list_header();
+        (void)et;
+        for (uint64_t i = 0; i < n; i++) {
+          RowGroupMeta rg;
+          int lrg = 0;
+          while (true) {
+            auto [idr, tr] = r.field_header(&lrg);
+            if (tr == 0) break;
+            if (idr == 1) {  // columns: list<ColumnChunk>
+              auto [nc, cet] = r.list_header();
+              (void)cet;
+              for (uint64_t c = 0; c < nc; c++) {
+                ChunkMeta cm;
+                int lcc = 0;
+                while (true) {
+                  auto [idc, tc] = r.field_header(&lcc);
+                  if (tc == 0) break;
+                  if (idc == 3) {  // ColumnMetaData
+                    int lmd = 0;
+                    while (true) {
+                      auto [idm, tm] = r.field_header(&lmd);
+                      if (tm == 0) break;
+                      switch (idm) {
+                        case 4: cm.codec = (int)r.zigzag(); break;
+                        case 5: cm.num_values = r.zigzag(); break;
+                        case 7: cm.total_compressed_size = r.zigzag(); break;
+                        case 9: cm.data_page_offset = r.zigzag(); break;
+                        case 11: cm.dict_page_offset = r.zigzag(); break;
+                        default: r.skip(tm);
+                      }
+                    }
+                  } else {
+                    r.skip(tc);
+                  }
+                }
+                rg.chunks.push_back(cm);
+              }
+            } else if (idr == 3) {
+              rg.num_rows = r.zigzag();
+            } else {
+              r.skip(tr);
+            }
+          }
+          row_groups_.push_back(std::move(rg));
+        }
+        break;
+      }
+      default: r.skip(t);
+    }
+  }
+
+  if (schema.empty()) fail("parquet: empty schema");
+  // flat schema: root + leaves only
+  for (size_t i = 1; i < schema.size(); i++) {
+    const SchemaElem& e = schema[i];
+    if (e.num_children > 0) fail("parquet: nested schema unsupported");
+    if (e.repetition == 2) fail("parquet: repeated fields unsupported");
+    PqColumnInfo ci;
+    ci.name = e.name;
+    ci.physical_type = e.type;
+    ci.nullable = e.repetition == 1;
+    if (ci.dtype() == DType::Unsupported)
+      fail("parquet: unsupported physical type for column " + e.name);
+    columns_.push_back(std::move(ci));
+  }
+  for (const auto& rg : row_groups_)
+    if (rg.chunks.size() != columns_.size())
+      fail("parquet: row group column count mismatch");
+}
+
+PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
+  const RowGroupMeta& g = row_groups_.at(rg);
+  const ChunkMeta& cm = g.chunks.at(col);
+  const PqColumnInfo& ci = columns_.at(col);
+  const int vw = (int)dtype_width(ci.dtype());
+
+  PqColumnChunkData out;
+  out.num_values = 0;
+  std::vector<bool> valid_bits;
+
+  int64_t pos = cm.dict_page_offset >= 0 ? cm.dict_page_offset
+                                         : cm.data_page_offset;
+  if (cm.dict_page_offset >= 0 && cm.data_page_offset < cm.dict_page_offset)
+    pos = cm.data_page_offset;  // defensive: some writers order differently
+  int64_t chunk_end = pos + cm.total_compressed_size;
+  if (chunk_end > (int64_t)file_.size()) fail("parquet: chunk overruns file");
+
+  auto decompress = [&](const uint8_t* src, size_t n, size_t out_size,
+                        std::vector<uint8_t>* buf) -> const uint8_t* {
+    std::string err;
+    switch (cm.codec) {
+      case 0: return src;  // UNCOMPRESSED
+      case 1:
+        if (!snappy_uncompress(src, n, buf, &err)) fail("parquet: " + err);
+        if (buf->size() != out_size) fail("parquet: snappy size mismatch");
+        return buf->data();
+      case 6:
+        if (!zstd_uncompress(src, n, out_size, buf, &err))
+          fail("parquet: " + err);
+        return buf->data();
+      case 7:
+        if (!lz4raw_uncompress(src, n, out_size, buf, &err))
+          fail("parquet: " + err);
+        return buf->data();
+      default:
+        fail("parquet: unsupported codec " + std::to_string(cm.codec));
+    }
+  };
+
+  while (pos < chunk_end && out.num_values < cm.num_values) {
+    TReader hr{file_.data() + pos, file_.data() + chunk_end};
+    PageHeader ph = read_page_header(hr);
+    const uint8_t* page = hr.p;
+    pos = (page - file_.data()) + ph.compressed_size;
+
+    if (ph.type == 2) {  // dictionary page
+      std::vector<uint8_t> buf;
+      const uint8_t* data =
+          decompress(page, ph.compressed_size, ph.uncompressed_size, &buf);
+      out.dict_values.assign(data, data + ph.uncompressed_size);
+      out.dict_count = ph.dict_num_values;
+      continue;
+    }
+    if (ph.type != 0 && ph.type != 3)
+      fail("parquet: unsupported page type " + std::to_string(ph.type));
+
+    std::vector<uint8_t> buf;
+    const uint8_t* data;
+    int64_t dlen;
+    std::vector<uint32_t> def_levels;
+    if (ph.type == 0) {  // V1: whole page compressed, def levels inside
+      data = decompress(page, ph.compressed_size, ph.uncompressed_size, &buf);
+      dlen = ph.uncompressed_size;
+      if (ci.nullable) {
+        if (ph.def_encoding != 3) fail("parquet: def levels must be RLE");
+        uint32_t ll;
+        memcpy(&ll, data, 4);
+        rle_bp_decode(data + 4, ll, 1, ph.num_values, &def_levels);
+        data += 4 + ll;
+        dlen -= 4 + ll;
+      }
+    } else {  // V2: def levels uncompressed before the (compressed) values
+      const uint8_t* dp = page;
+      if (ci.nullable) {
+        if (ph.def_len > 0)
+          rle_bp_decode(dp, ph.def_len, 1, ph.num_values, &def_levels);
+      }
+      dp += ph.def_len + ph.rep_len;
+      int64_t comp = ph.compressed_size - ph.def_len - ph.rep_len;
+      int64_t uncomp = ph.uncompressed_size - ph.def_len - ph.rep_len;
+      data = ph.v2_compressed ? decompress(dp, comp, uncomp, &buf) : dp;
+      dlen = uncomp;
+    }
+
+    int64_t nvals = ph.num_values;
+    int64_t non_null = nvals;
+    if (ci.nullable && !def_levels.empty()) {
+      non_null = 0;
+      for (int64_t i = 0; i < nvals; i++) {
+        bool v = def_levels[i] != 0;
+        valid_bits.push_back(v);
+        non_null += v;
+      }
+      out.null_count += nvals - non_null;
+    } else {
+      for (int64_t i = 0; i < nvals; i++) valid_bits.push_back(true);
+    }
+
+    switch (ph.encoding) {
+      case 0: {  // PLAIN
+        if (dlen < non_null * vw) fail("parquet: short PLAIN data");
+        out.plain.insert(out.plain.end(), data, data + non_null * vw);
+        break;
+      }
+      case 2:    // PLAIN_DICTIONARY
+      case 8: {  // RLE_DICTIONARY
+        if (out.dict_values.empty()) fail("parquet: dict page missing");
+        int bw = data[0];
+        std::vector<uint32_t> idx;
+        rle_bp_decode(data + 1, dlen - 1, bw, non_null, &idx);
+        out.dict_indices.insert(out.dict_indices.end(), idx.begin(), idx.end());
+        out.uses_dict = true;
+        break;
+      }
+      default:
+        fail("parquet: unsupported value encoding " +
+             std::to_string(ph.encoding));
+    }
+    out.num_values += nvals;
+  }
+
+  if (out.num_values != cm.num_values)
+    fail("parquet: value count mismatch in chunk");
+  if (out.uses_dict && !out.plain.empty())
+    fail("parquet: mixed PLAIN and dictionary pages unsupported");
+  if (out.null_count > 0) {
+    out.validity.assign((valid_bits.size() + 7) / 8, 0);
+    for (size_t i = 0; i < valid_bits.size(); i++)
+      if (valid_bits[i]) out.validity[i >> 3] |= (uint8_t)(1u << (i & 7));
+  }
+  return out;
+}
+
+}  // namespace auron

@@ -1,0 +1,96 @@
+// parquet.h — restricted Parquet reader for the config-3 scan path
+// (SURVEY.md §8 a15 / f.1): flat non-repeated columns of INT32/INT64/FLOAT/
+// DOUBLE, PLAIN + dictionary encodings, RLE/bit-packed definition levels,
+// UNCOMPRESSED/SNAPPY/ZSTD/LZ4_RAW page codecs. The reference delegates
+// decode to the third-party parquet crate (parquet_exec.rs:174-196, SURVEY.md
+// §8c ii — parity pinned against files written and read back by pyarrow).
+//
+// Host side: footer/page-header parsing (Thrift compact protocol, hand-rolled
+// — no thrift in this image), page decompression, RLE/bit-packed run
+// extraction. GPU side (engine): run expansion to validity bitmaps and
+// dictionary indices, dictionary gather, packed-value scatter.
+#pragma once
+
+#include <cstdint>
+#include <string>
+#include <vector>
+
+#include "plan.h"
+
+namespace auron {
+
+// decoded RLE/bit-packed hybrid stream as run list (host): either a repeated
+// run (count, value) or a bit-packed literal span (values materialized)
+struct RunList {
+  // flattened: literal values for every position, materialized on host.
+  // Levels and dict indices are small (u32); full materialization keeps the
+  // host side simple, the GPU does the wide work (gather/scatter).
+  std::vector<uint32_t> values;
+};
+
+struct PqColumnChunkData {
+  // per row-group column results, host-resident, ready for GPU upload
+  int64_t num_values = 0;              // value slots incl. nulls
+  std::vector<uint8_t> plain;          // PLAIN-encoded non-null values (LE)
+  std::vector<uint32_t> dict_indices;  // non-null dictionary indices
+  std::vector<uint8_t> dict_values;    // dictionary page PLAIN values
+  int64_t dict_count = 0;
+  std::vector<uint8_t> validity;       // LSB bitmap, empty = all valid
+  int64_t null_count = 0;
+  bool uses_dict = false;
+};
+
+struct PqColumnInfo {
+  std::string name;
+  int physical_type = -1;  // parquet: 1=INT32, 2=INT64, 4=FLOAT, 5=DOUBLE
+  bool nullable = false;   // max_def == 1
+  DType dtype() const {
+    switch (physical_type) {
+      case 1: return DType::Int32;
+      case 2: return DType::Int64;
+      case 4: return DType::Float32;
+      case 5: return DType::Float64;
+      default: return DType::Unsupported;
+    }
+  }
+};
+
+class ParquetFile {
+ public:
+  // reads and parses the footer; throws std::runtime_error on malformed or
+  // out-of-scope features
+  explicit ParquetFile(const std::string& path);
+
+  int num_row_groups() const { return (int)row_groups_.size(); }
+  int64_t row_group_rows(int rg) const { return row_groups_[rg].num_rows; }
+  const std::vector<PqColumnInfo>& columns() const { return columns_; }
+
+  // decode one column chunk of one row group
+  PqColumnChunkData read_chunk(int rg, int col) const;
+
+ private:
+  struct ChunkMeta {
+    int64_t data_page_offset = -1;
+    int64_t dict_page_offset = -1;
+    int64_t total_compressed_size = 0;
+    int64_t num_values = 0;
+    int codec = 0;
+  };
+  struct RowGroupMeta {
+    int64_t num_rows = 0;
+    std::vector<ChunkMeta> chunks;
+  };
+
+  std::string path_;
+  std::vector<uint8_t> file_;  // whole file (config sizes are test-scale;
+                               // streaming IO is a later row)
+  std::vector<PqColumnInfo> columns_;
+  std::vector<RowGroupMeta> row_groups_;
+};
+
+// snappy block decompression (format: varint uncompressed len, then
+// literal/copy tags) — no libsnappy in this image
+bool snappy_uncompress(const uint8_t* src, size_t n, std::vector<uint8_t>* out,
+                       std::string* err);
+
+}  // namespace auron

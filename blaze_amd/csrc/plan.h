@@ -121,6 +121,14 @@ struct FFIReaderNode {
   std::string resource_id;       // = 3
 };
 
+// ParquetScanExecNode (auron.proto:415-419) + FileScanExecConf (:404-413)
+struct ParquetScanNode {
+  std::vector<std::string> files;     // FileGroup.files[].path
+  Schema schema;                      // FileScanExecConf.schema = 4
+  std::vector<uint32_t> projection;   // FileScanExecConf.projection = 6
+  std::string fs_resource_id;         // = 3 (unused: POSIX paths read directly)
+};
+
 // IpcReaderExecNode (auron.proto:607-611)
 struct IpcReaderNode {
   uint32_t num_partitions = 1;   // = 1
@@ -146,6 +154,7 @@ struct PlanNode {
   enum Kind {
     ShuffleWriter = 2,
     IpcReader = 3,
+    ParquetScan = 5,
     Projection = 6,
     Filter = 8,
     Agg = 16,
@@ -157,6 +166,7 @@ struct PlanNode {
   std::unique_ptr<FilterNode> filter;
   std::unique_ptr<ProjectionNode> projection;
   std::unique_ptr<IpcReaderNode> ipc_reader;
+  std::unique_ptr<ParquetScanNode> parquet;
 };
 
 // TaskDefinition (auron.proto:735-740) + PartitionId (:729-733)

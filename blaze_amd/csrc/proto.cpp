@@ -292,6 +292,62 @@ std::unique_ptr<ShuffleWriterNode> decode_shuffle_writer(Reader r,
   return n;
 }
 
+std::unique_ptr<ParquetScanNode> decode_parquet_scan(Reader r,
+                                                     std::string* err) {
+  auto n = std::make_unique<ParquetScanNode>();
+  while (true) {
+    auto [f, w] = r.tag();
+    if (f == 0) break;
+    if (f == 1) {  // FileScanExecConf (auron.proto:404-413)
+      Reader s = r.sub();
+      while (true) {
+        auto [sf, sw] = s.tag();
+        if (sf == 0) break;
+        switch (sf) {
+          case 3: {  // FileGroup{files = 1: PartitionedFile{path = 1}}
+            Reader fg = s.sub();
+            while (true) {
+              auto [ff, fw] = fg.tag();
+              if (ff == 0) break;
+              if (ff == 1) {
+                Reader pf = fg.sub();
+                while (true) {
+                  auto [pff, pfw] = pf.tag();
+                  if (pff == 0) break;
+                  if (pff == 1) n->files.push_back(pf.str());
+                  else pf.skip(pfw);
+                }
+              } else {
+                fg.skip(fw);
+              }
+            }
+            break;
+          }
+          case 4: n->schema = decode_schema(s.sub()); break;
+          case 6:
+            if (sw == 2) {  // packed repeated uint32
+              Reader pr = s.sub();
+              while (pr.p < pr.end)
+                n->projection.push_back((uint32_t)pr.varint());
+            } else {
+              n->projection.push_back((uint32_t)s.varint());
+            }
+            break;
+          default: s.skip(sw);
+        }
+      }
+    } else if (f == 3) {
+      n->fs_resource_id = r.str();
+    } else if (f == 2) {
+      if (err->empty()) *err = "parquet pruning predicates unsupported";
+      r.skip(w);
+    } else {
+      r.skip(w);
+    }
+  }
+  return n;
+}
+
 std::unique_ptr<IpcReaderNode> decode_ipc_reader(Reader r) {
   auto n = std::make_unique<IpcReaderNode>();
   while (true) {
@@ -360,6 +416,10 @@ std::unique_ptr<PlanNode> decode_plan_node(Reader r, std::string* err) {
       case 3:  // IpcReaderExecNode
         node->kind = PlanNode::IpcReader;
         node->ipc_reader = decode_ipc_reader(r.sub());
+        return node;
+      case 5:  // ParquetScanExecNode
+        node->kind = PlanNode::ParquetScan;
+        node->parquet = decode_parquet_scan(r.sub(), err);
         return node;
       case 6:  // ProjectionExecNode
         node->kind = PlanNode::Projection;

@@ -286,3 +286,41 @@ def plan_ipc_final(resource_id="ipc0"):
     final = agg(reader, [column("key", 0)], sum_count_aggs(1),
                 [MODE_FINAL, MODE_FINAL], ["key"], ["sum", "cnt"])
     return task_definition(final)
+
+
+# ---- ParquetScanExec (a15 / config 3) --------------------------------------
+
+def partitioned_file(path, size):
+    # PartitionedFile (auron.proto:373-379)
+    return _len_field(1, path.encode()) + _varint_field(2, size)
+
+
+def parquet_scan(files, fields, projection=(), resource_id="fs0"):
+    """PhysicalPlanNode{parquet_scan = 5} -> ParquetScanExecNode
+    (auron.proto:415-419) with FileScanExecConf (:404-413).
+    files: list of (path, size)."""
+    fg = b"".join(_len_field(1, partitioned_file(p, sz)) for p, sz in files)
+    conf = _varint_field(1, 1) + _varint_field(2, 0)
+    conf += _len_field(3, fg)
+    conf += _len_field(4, schema(fields))
+    if projection:
+        conf += _len_field(6, b"".join(_varint(i) for i in projection))
+    node = _len_field(1, conf) + _len_field(3, resource_id.encode())
+    return _len_field(5, node)
+
+
+def plan_parquet_filter_agg(files, cutoff=None, key_field="key",
+                            val_field="val", key_dt=DT_INT64):
+    """Config-3 topology: ParquetScan -> [Filter(key < c)] -> Agg(Partial) ->
+    Agg(Final) over (key, val f64)."""
+    fields = [field(key_field, key_dt, True), field(val_field, DT_FLOAT64, True)]
+    node = parquet_scan(files, fields)
+    if cutoff is not None:
+        dt = {DT_INT64: "int64", DT_INT32: "int32"}[key_dt]
+        node = filter_node(node, [
+            binary_expr(column(key_field, 0), literal(cutoff, dt), "Lt")])
+    partial = agg(node, [column(key_field, 0)], sum_count_aggs(1),
+                  [MODE_PARTIAL, MODE_PARTIAL], [key_field], ["sum", "cnt"])
+    final = agg(partial, [column(key_field, 0)], sum_count_aggs(1),
+                [MODE_FINAL, MODE_FINAL], [key_field], ["sum", "cnt"])
+    return task_definition(final)
