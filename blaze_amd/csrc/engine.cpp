@@ -1861,6 +1861,36 @@ void auron_on_exit(void) {
 
 const char* auron_version(void) { return "auron-hip 0.1 gfx950"; }
 
+// test-only: parse a parquet file on the host and summarize chunk decode
+// results (CPU-checkable against pyarrow metadata — no GPU needed)
+int32_t auron_debug_parquet_summary(const char* path, char* out, size_t cap) {
+  std::string r;
+  try {
+    ParquetFile pf(path);
+    r = "cols=";
+    for (const auto& c : pf.columns())
+      r += c.name + ":" + std::to_string(c.physical_type) +
+           (c.nullable ? "?" : "") + ",";
+    r += " rgs=" + std::to_string(pf.num_row_groups());
+    for (int rg = 0; rg < pf.num_row_groups(); rg++) {
+      r += " [rg" + std::to_string(rg) + " rows=" +
+           std::to_string(pf.row_group_rows(rg));
+      for (size_t c = 0; c < pf.columns().size(); c++) {
+        PqColumnChunkData cd = pf.read_chunk(rg, (int)c);
+        r += " c" + std::to_string(c) + "{n=" + std::to_string(cd.num_values) +
+             ",nulls=" + std::to_string(cd.null_count) +
+             ",dict=" + std::to_string(cd.uses_dict ? cd.dict_count : 0) + "}";
+      }
+      r += "]";
+    }
+  } catch (const std::exception& ex) {
+    r = std::string("ERROR: ") + ex.what();
+  }
+  if (r.size() + 1 > cap) return -1;
+  memcpy(out, r.c_str(), r.size() + 1);
+  return (int32_t)r.size();
+}
+
 // test-only: decode a ScalarValue ipc_bytes literal and render it
 int32_t auron_debug_decode_scalar(const uint8_t* data, size_t len, char* out,
                                   size_t cap) {

@@ -54,3 +54,37 @@ def test_filter_project_plan_decodes():
     # chain shape is intact down to the reader
     assert "Agg(mode=2" in s and "Agg(mode=0" in s
     assert "FFIReader(nfields=2,rid=input0)" in s
+
+
+def test_parquet_host_decode_summary(tmp_path):
+    """Host-side parquet footer/page decode vs pyarrow metadata (no GPU)."""
+    import numpy as np
+    pq = pytest.importorskip("pyarrow.parquet")
+    rng = np.random.default_rng(17)
+    n = 30_000
+    keys = rng.integers(0, 100, n).astype(np.int64)
+    vals = rng.integers(0, 1000, n).astype(np.float64)
+    vv = rng.random(n) >= 0.1
+    t = pa.table({"key": pa.array(keys, pa.int64()),
+                  "val": pa.array(vals, pa.float64(), mask=~vv)})
+    path = str(tmp_path / "t.parquet")
+    pq.write_table(t, path, compression="snappy", row_group_size=10_000)
+
+    lib = blaze_amd.lib()
+    lib.auron_debug_parquet_summary.restype = ctypes.c_int32
+    lib.auron_debug_parquet_summary.argtypes = [ctypes.c_char_p,
+                                                ctypes.c_char_p,
+                                                ctypes.c_size_t]
+    out = ctypes.create_string_buffer(1 << 14)
+    rc = lib.auron_debug_parquet_summary(path.encode(), out, len(out))
+    s = out.value.decode()
+    assert rc > 0, s
+    assert "ERROR" not in s, s
+    assert "cols=key:2?,val:5?," in s
+    assert "rgs=3" in s
+    total_nulls = int((~vv).sum())
+    import re
+    nulls = sum(int(m) for m in re.findall(r"c1\{n=\d+,nulls=(\d+)", s))
+    assert nulls == total_nulls
+    rows = sum(int(m) for m in re.findall(r"rows=(\d+)", s))
+    assert rows == n
