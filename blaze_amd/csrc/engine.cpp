@@ -393,6 +393,16 @@ class AggOp {
     skip_ratio_ = conf.get_d("PARTIAL_AGG_SKIPPING_RATIO", 0.999);
     skip_min_rows_ = conf.get_i("PARTIAL_AGG_SKIPPING_MIN_ROWS", 20000);
     int64_t slots = conf.get_i("AURON_HIP_AGG_TABLE_SLOTS", 1 << 23);
+    // VRAM budget for the slot table (a9 analog of MemManager's budget,
+    // memmgr/mod.rs:36-105): exceeding it spills frozen records to host
+    // buckets instead of growing (agg_table.rs:540-588 semantics).
+    mem_budget_ = conf.get_i("AURON_HIP_MEM_BUDGET", 8LL << 30);
+    num_spill_buckets_ =
+        (int)std::max<int64_t>(16, conf.get_i("AURON_HIP_SPILL_BUCKETS", 16));
+    max_cap_ = 1;
+    while (max_cap_ * 2 * (int64_t)sizeof(AggSlot) <= mem_budget_)
+      max_cap_ *= 2;
+    if (slots > max_cap_) slots = max_cap_;
     init_table(slots);
     AURON_HIP(hipEventCreate(&ev_start_));
     AURON_HIP(hipEventCreate(&ev_stop_));
@@ -538,33 +548,55 @@ class AggOp {
   std::vector<std::pair<int64_t, std::vector<HostOutCol>>> finish() {
     std::vector<std::pair<int64_t, std::vector<HostOutCol>>> out;
     drain_timing();
-    // 1) table contents
-    uint64_t ng = num_groups_host();
-    DBG("agg.finish ng=%llu", (unsigned long long)ng);
-    if (ng > 0) {
-      DevBuf slots_u(ng * 4), first(ng * 8), slots_sorted(ng * 4),
-          first_sorted(ng * 8), dcount(8);
-      AURON_HIP(hipMemsetAsync(dcount.get(), 0, 8, stream_));
-      launch_agg_compact(t_, slots_u.get<uint32_t>(),
-                         first.get<unsigned long long>(),
-                         dcount.get<unsigned long long>(), stream_);
-      // order groups by first occurrence (insertion order,
-      // agg_hash_map.rs:77-168)
-      size_t tmp_bytes = 0;
-      sort_pairs_u64_u32(first.get<unsigned long long>(), slots_u.get<uint32_t>(),
-                         first_sorted.get<unsigned long long>(),
-                         slots_sorted.get<uint32_t>(), (int64_t)ng, nullptr,
-                         &tmp_bytes, stream_);
-      DevBuf tmp(tmp_bytes);
-      sort_pairs_u64_u32(first.get<unsigned long long>(), slots_u.get<uint32_t>(),
-                         first_sorted.get<unsigned long long>(),
-                         slots_sorted.get<uint32_t>(), (int64_t)ng, tmp.get(),
-                         &tmp_bytes, stream_);
-      for (uint64_t beg = 0; beg < ng; beg += (uint64_t)batch_size_) {
-        uint64_t len = std::min((uint64_t)batch_size_, ng - beg);
-        out.emplace_back(emit_groups(slots_sorted.get<uint32_t>() + beg,
-                                     (int64_t)len));
+    if (spill_.empty()) {
+      emit_table(&out, false);
+    } else {
+      // a9 analog (agg_table.rs:540-588 spill + RadixQueue merge): drain the
+      // live table into the buckets, then merge+emit bucket by bucket so
+      // peak table size is one bucket's groups. Output order is per-bucket
+      // (the reference's own spill path also gives up global insertion
+      // order), first-occurrence-ordered within each bucket.
+      spill_table();
+      for (size_t b = 0; b < spill_.size(); b++) {
+        SpillBucket& sb = spill_[b];
+        int64_t n = (int64_t)sb.keys.size();
+        if (n > 0) {
+          std::vector<int32_t> offs(n + 1, 0);
+          for (int64_t i = 0; i < n; i++) offs[i + 1] = offs[i] + sb.lens[i];
+          DevBuf d_keys(n * 8), d_first(n * 8), d_offs((n + 1) * 4),
+              d_data(offs[n] ? offs[n] : 1);
+          AURON_HIP(hipMemcpyAsync(d_keys.get(), sb.keys.data(), n * 8,
+                                   hipMemcpyHostToDevice, stream_));
+          AURON_HIP(hipMemcpyAsync(d_first.get(), sb.first_rows.data(), n * 8,
+                                   hipMemcpyHostToDevice, stream_));
+          AURON_HIP(hipMemcpyAsync(d_offs.get(), offs.data(), (n + 1) * 4,
+                                   hipMemcpyHostToDevice, stream_));
+          if (offs[n])
+            AURON_HIP(hipMemcpyAsync(d_data.get(), sb.data.data(), offs[n],
+                                     hipMemcpyHostToDevice, stream_));
+          for (int64_t done2 = 0; done2 < n;) {
+            ensure_capacity((int64_t)ng_bound_ +
+                            std::min<int64_t>(n - done2, 1 << 20));
+            int64_t free2 = t_.cap * 3 / 4 - (int64_t)ng_bound_;
+            if (free2 <= 0)
+              FAIL("spill bucket exceeds VRAM budget: raise "
+                   "AURON_HIP_SPILL_BUCKETS");
+            int64_t piece = std::min(n - done2, free2);
+            launch_agg_merge_spill(t_, d_keys.get<int64_t>() + done2,
+                                   d_data.get<uint8_t>(),
+                                   d_offs.get<int32_t>() + done2,
+                                   d_first.get<unsigned long long>() + done2,
+                                   piece, layout_, stream_);
+            done2 += piece;
+            ng_bound_ += (uint64_t)piece;
+          }
+          AURON_HIP(hipStreamSynchronize(stream_));
+        }
+        bool last = (b + 1 == spill_.size());
+        emit_table(&out, /*exclude_specials=*/!last);
+        if (!last) reset_main();
       }
+      spill_.clear();
     }
     // 2) partial-skipped pass-through batches (agg_ctx.rs:428-462)
     for (DevBatch& b : skipped_) {
@@ -686,14 +718,44 @@ class AggOp {
                              hipMemcpyDeviceToHost, stream_));
     AURON_HIP(hipStreamSynchronize(stream_));
     int64_t staged_n = (int64_t)h_ctr[0], lo_n = (int64_t)h_ctr[1];
-    // merge: table inserts bounded by the COUNTED lists
-    ensure_capacity((int64_t)ng_bound_ + staged_n + lo_n);
-    if (staged_n)
-      launch_agg2_merge_groups(t_, d_staged_.get<StagedGroup>(), staged_n,
+    // merge: table inserts bounded by the COUNTED lists; piecewise so a
+    // VRAM budget (spill) smaller than the staged list still works
+    for (int64_t done2 = 0; done2 < staged_n;) {
+      ensure_capacity((int64_t)ng_bound_ + std::min<int64_t>(
+                                               staged_n - done2, 1 << 20));
+      int64_t free2 = t_.cap * 3 / 4 - (int64_t)ng_bound_;
+      if (free2 < (1 << 14)) {
+        refresh_ng();
+        free2 = t_.cap * 3 / 4 - (int64_t)ng_true_;
+        if (free2 < (1 << 14)) {
+          spill_table();
+          continue;
+        }
+      }
+      int64_t piece = std::min(staged_n - done2, free2);
+      launch_agg2_merge_groups(t_, d_staged_.get<StagedGroup>() + done2, piece,
                                row_cursor_, stream_);
-    if (lo_n)
-      launch_agg2_leftovers(t_, d_leftover_.get<PartRow>(), lo_n, row_cursor_,
-                            stream_);
+      done2 += piece;
+      ng_bound_ += (uint64_t)piece;
+    }
+    for (int64_t done2 = 0; done2 < lo_n;) {
+      ensure_capacity((int64_t)ng_bound_ +
+                      std::min<int64_t>(lo_n - done2, 1 << 20));
+      int64_t free2 = t_.cap * 3 / 4 - (int64_t)ng_bound_;
+      if (free2 < (1 << 14)) {
+        refresh_ng();
+        free2 = t_.cap * 3 / 4 - (int64_t)ng_true_;
+        if (free2 < (1 << 14)) {
+          spill_table();
+          continue;
+        }
+      }
+      int64_t piece = std::min(lo_n - done2, free2);
+      launch_agg2_leftovers(t_, d_leftover_.get<PartRow>() + done2, piece,
+                            row_cursor_, stream_);
+      done2 += piece;
+      ng_bound_ += (uint64_t)piece;
+    }
     AURON_HIP(hipEventRecord(e1, stream_));
     ev_pairs_.push_back({e0, e1});
     update_rows_ += chunk;
@@ -703,6 +765,125 @@ class AggOp {
     refresh_ng();
     DBG("agg.2phase chunk=%lld staged=%lld leftover=%lld special=%u",
         (long long)chunk, (long long)staged_n, (long long)lo_n, special_rows);
+  }
+
+  static uint64_t host_mix64(uint64_t x) {
+    x += 0x9E3779B97F4A7C15ull;
+    x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+    x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+    return x ^ (x >> 31);
+  }
+
+  struct SpillBucket {
+    std::vector<int64_t> keys;
+    std::vector<unsigned long long> first_rows;
+    std::vector<int32_t> lens;
+    std::vector<uint8_t> data;
+  };
+
+  // compact the table and return (order_slots sorted by first_row, ng);
+  // device buffers owned by the out-params
+  int64_t table_order(DevBuf* slots_sorted, DevBuf* first_sorted) {
+    uint64_t ng = num_groups_host();
+    if (ng == 0) return 0;
+    DevBuf slots_u(ng * 4), first(ng * 8), dcount(8);
+    slots_sorted->alloc(ng * 4);
+    first_sorted->alloc(ng * 8);
+    AURON_HIP(hipMemsetAsync(dcount.get(), 0, 8, stream_));
+    launch_agg_compact(t_, slots_u.get<uint32_t>(),
+                       first.get<unsigned long long>(),
+                       dcount.get<unsigned long long>(), stream_);
+    size_t tmp_bytes = 0;
+    sort_pairs_u64_u32(first.get<unsigned long long>(), slots_u.get<uint32_t>(),
+                       first_sorted->get<unsigned long long>(),
+                       slots_sorted->get<uint32_t>(), (int64_t)ng, nullptr,
+                       &tmp_bytes, stream_);
+    DevBuf tmp(tmp_bytes);
+    sort_pairs_u64_u32(first.get<unsigned long long>(), slots_u.get<uint32_t>(),
+                       first_sorted->get<unsigned long long>(),
+                       slots_sorted->get<uint32_t>(), (int64_t)ng, tmp.get(),
+                       &tmp_bytes, stream_);
+    return (int64_t)ng;
+  }
+
+  // freeze the main-region groups to host spill buckets and reset the main
+  // region (the two special groups stay resident across spills)
+  void spill_table() {
+    drain_timing();
+    DevBuf order, first;
+    int64_t ng = table_order(&order, &first);
+    if (ng == 0) return;
+    std::vector<uint32_t> h_order(ng);
+    std::vector<unsigned long long> h_first(ng);
+    AURON_HIP(hipMemcpyAsync(h_order.data(), order.get(), ng * 4,
+                             hipMemcpyDeviceToHost, stream_));
+    AURON_HIP(hipMemcpyAsync(h_first.data(), first.get(), ng * 8,
+                             hipMemcpyDeviceToHost, stream_));
+    AURON_HIP(hipStreamSynchronize(stream_));
+    // filter out the special slots (they stay in the table)
+    std::vector<uint32_t> main_order;
+    std::vector<unsigned long long> main_first;
+    for (int64_t i = 0; i < ng; i++) {
+      if (h_order[i] < (uint32_t)t_.cap) {
+        main_order.push_back(h_order[i]);
+        main_first.push_back(h_first[i]);
+      }
+    }
+    int64_t n = (int64_t)main_order.size();
+    specials_count_ = ng - n;
+    if (n > 0) {
+      DevBuf d_order(n * 4), keys(n * 8), lens(n * 4), offs((n + 1) * 4);
+      AURON_HIP(hipMemcpyAsync(d_order.get(), main_order.data(), n * 4,
+                               hipMemcpyHostToDevice, stream_));
+      launch_agg_gather_out(t_, d_order.get<uint32_t>(), n,
+                            keys.get<int64_t>(), nullptr, nullptr, nullptr,
+                            nullptr, stream_);
+      launch_agg_freeze_len(t_, d_order.get<uint32_t>(), n, lens.get<int32_t>(),
+                            layout_, stream_);
+      std::vector<int32_t> h_lens(n);
+      std::vector<int64_t> h_keys(n);
+      AURON_HIP(hipMemcpyAsync(h_lens.data(), lens.get(), n * 4,
+                               hipMemcpyDeviceToHost, stream_));
+      AURON_HIP(hipMemcpyAsync(h_keys.data(), keys.get(), n * 8,
+                               hipMemcpyDeviceToHost, stream_));
+      AURON_HIP(hipStreamSynchronize(stream_));
+      std::vector<int32_t> h_offs(n + 1, 0);
+      for (int64_t i = 0; i < n; i++) h_offs[i + 1] = h_offs[i] + h_lens[i];
+      AURON_HIP(hipMemcpyAsync(offs.get(), h_offs.data(), (n + 1) * 4,
+                               hipMemcpyHostToDevice, stream_));
+      DevBuf data(h_offs[n] ? h_offs[n] : 1);
+      launch_agg_freeze_write(t_, d_order.get<uint32_t>(), n,
+                              offs.get<int32_t>(), data.get<uint8_t>(), layout_,
+                              stream_);
+      std::vector<uint8_t> h_data(h_offs[n]);
+      if (h_offs[n])
+        AURON_HIP(hipMemcpyAsync(h_data.data(), data.get(), h_offs[n],
+                                 hipMemcpyDeviceToHost, stream_));
+      AURON_HIP(hipStreamSynchronize(stream_));
+      if (spill_.empty()) spill_.resize(num_spill_buckets_);
+      for (int64_t i = 0; i < n; i++) {
+        SpillBucket& b =
+            spill_[host_mix64((uint64_t)h_keys[i]) % num_spill_buckets_];
+        b.keys.push_back(h_keys[i]);
+        b.first_rows.push_back(main_first[i]);
+        b.lens.push_back(h_lens[i]);
+        b.data.insert(b.data.end(), h_data.begin() + h_offs[i],
+                      h_data.begin() + h_offs[i + 1]);
+      }
+      spill_count_++;
+    }
+    reset_main();
+    DBG("agg.spill n=%lld buckets=%d", (long long)n, num_spill_buckets_);
+  }
+
+  void reset_main() {
+    launch_slots_init(t_.slots, t_.cap, stream_);  // specials untouched
+    uint64_t ng0 = (uint64_t)specials_count_;
+    AURON_HIP(hipMemcpyAsync(t_.num_groups, &ng0, 8, hipMemcpyHostToDevice,
+                             stream_));
+    AURON_HIP(hipStreamSynchronize(stream_));
+    ng_true_ = ng0;
+    ng_bound_ = ng0;
   }
 
   void refresh_ng() {
@@ -724,7 +905,10 @@ class AggOp {
   }
 
   void ensure_capacity(int64_t need) {
-    while (t_.cap * 3 / 4 < need) grow(t_.cap * 4);
+    while (t_.cap * 3 / 4 < need && t_.cap < max_cap_) grow(t_.cap * 4);
+    if (t_.cap * 3 / 4 < need && num_groups_host() > specials_count_) {
+      spill_table();  // frees the main region; caller re-checks free space
+    }
   }
 
   void grow(int64_t new_cap) {
@@ -734,6 +918,37 @@ class AggOp {
     init_table(new_cap);
     launch_agg_rebuild(t_, old, stream_);
     AURON_HIP(hipStreamSynchronize(stream_));
+  }
+
+  // emit every group in the table, first-occurrence-ordered, in
+  // batch_size_-row output chunks
+  void emit_table(std::vector<std::pair<int64_t, std::vector<HostOutCol>>>* out,
+                  bool exclude_specials) {
+    DevBuf order, first;
+    int64_t ng = table_order(&order, &first);
+    DBG("agg.finish ng=%lld excl=%d", (long long)ng, (int)exclude_specials);
+    if (ng == 0) return;
+    DevBuf filtered;
+    const uint32_t* order_ptr = order.get<uint32_t>();
+    if (exclude_specials) {
+      std::vector<uint32_t> h_order(ng);
+      AURON_HIP(hipMemcpyAsync(h_order.data(), order.get(), ng * 4,
+                               hipMemcpyDeviceToHost, stream_));
+      AURON_HIP(hipStreamSynchronize(stream_));
+      std::vector<uint32_t> keep;
+      for (int64_t i = 0; i < ng; i++)
+        if (h_order[i] < (uint32_t)t_.cap) keep.push_back(h_order[i]);
+      ng = (int64_t)keep.size();
+      if (ng == 0) return;
+      filtered.alloc(ng * 4);
+      AURON_HIP(hipMemcpyAsync(filtered.get(), keep.data(), ng * 4,
+                               hipMemcpyHostToDevice, stream_));
+      order_ptr = filtered.get<uint32_t>();
+    }
+    for (int64_t beg = 0; beg < ng; beg += batch_size_) {
+      int64_t len = std::min(batch_size_, ng - beg);
+      out->emplace_back(emit_groups(order_ptr + beg, len));
+    }
   }
 
   std::pair<int64_t, std::vector<HostOutCol>> emit_groups(
@@ -892,6 +1107,10 @@ class AggOp {
   int64_t skip_min_rows_ = 20000;
   uint64_t row_cursor_ = 0;
   uint64_t ng_bound_ = 0, ng_true_ = 0;  // conservative bound / last readback
+  int64_t mem_budget_ = 8LL << 30, max_cap_ = 1 << 23;
+  int num_spill_buckets_ = 16;
+  int64_t specials_count_ = 0, spill_count_ = 0;
+  std::vector<SpillBucket> spill_;
   int64_t update_ns_ = 0, update_rows_ = 0;
   std::vector<std::pair<hipEvent_t, hipEvent_t>> ev_pairs_;
   hipEvent_t ev_start_ = nullptr, ev_stop_ = nullptr;
@@ -1527,6 +1746,7 @@ struct Runtime {
       metrics["num_groups"] = (int64_t)last_agg->num_groups_host();
       metrics["agg_update_ns"] = first_agg->update_ns_;
       metrics["agg_update_rows"] = first_agg->update_rows_;
+      metrics["spill_count"] = first_agg->spill_count_ + last_agg->spill_count_;
     }
     metrics["input_rows"] = input_rows;
     int64_t out_rows = 0;

@@ -64,6 +64,13 @@ void launch_agg_merge_frozen(const AggTable& t, const int64_t* keys,
                              uint64_t row_offset, uint32_t layout,
                              hipStream_t s);
 
+// merge spilled records back (a9 analog): like merge_frozen but each record
+// carries its preserved global first_row explicitly
+void launch_agg_merge_spill(const AggTable& t, const int64_t* keys,
+                            const uint8_t* acc_data, const int32_t* acc_offsets,
+                            const unsigned long long* first_rows, int64_t n,
+                            uint32_t layout, hipStream_t s);
+
 // compact occupied slots to dense arrays (unordered); returns count via
 // num_out (device). out_slot holds the source slot index per group.
 void launch_agg_compact(const AggTable& t, uint32_t* out_slot,

@@ -390,6 +390,42 @@ __global__ void k_agg_merge_frozen(const AggTable t,
   }
 }
 
+__global__ void k_agg_merge_spill(const AggTable t,
+                                  const int64_t* __restrict__ keys,
+                                  const uint8_t* __restrict__ acc_data,
+                                  const int32_t* __restrict__ acc_offsets,
+                                  const unsigned long long* __restrict__ first_rows,
+                                  int64_t n, uint32_t layout) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    // special groups (null-key / i64::MIN-key) are never spilled — they stay
+    // resident in the table's trailing slots across spill resets — so every
+    // spilled key is a regular value here.
+    int64_t key = keys[i];
+    int32_t off = acc_offsets[i];
+    int64_t a = agg_upsert_slot(t, key, false);
+    if (a < 0) continue;
+    AggSlot* sl = &t.slots[a];
+    unsigned long long fr = first_rows[i];
+    if (sl->first_row > fr) atomicMin(&sl->first_row, fr);
+    bool valid;
+    double v;
+    uint64_t c;
+    agg_parse_frozen(layout, acc_data + off, &valid, &v, &c);
+    if (valid) unsafeAtomicAdd(&sl->sum, v);
+    if (c) atomicAdd(&sl->cnt, c);
+  }
+}
+
+void launch_agg_merge_spill(const AggTable& t, const int64_t* keys,
+                            const uint8_t* acc_data, const int32_t* acc_offsets,
+                            const unsigned long long* first_rows, int64_t n,
+                            uint32_t layout, hipStream_t s) {
+  hipLaunchKernelGGL(k_agg_merge_spill, dim3(grid_for(n)), dim3(BLOCK), 0, s, t,
+                     keys, acc_data, acc_offsets, first_rows, n, layout);
+  check_launch("k_agg_merge_spill");
+}
+
 __global__ void k_agg_compact(const AggTable t, uint32_t* __restrict__ out_slot,
                               unsigned long long* __restrict__ out_first_row,
                               unsigned long long* __restrict__ num_out) {
