@@ -293,3 +293,52 @@ def test_avg_agg_layouts():
             np.testing.assert_array_equal(got_sums, ref["sums"])
             np.testing.assert_array_equal(got_cnts, ref["counts"])
         t.finalize()
+
+
+def test_spill_to_host():
+    """a9 analog: a tiny AURON_HIP_MEM_BUDGET forces the table to spill
+    frozen records to host buckets mid-stream (agg_table.rs:540-588
+    semantics); the bucket-merged output must still be the exact aggregate
+    (order is per-bucket once spilling happens, like the reference)."""
+    keys, vals, vv = gen_northstar(400_000, nkeys=120_000)
+    conf = {"AURON_HIP_MEM_BUDGET": 1 << 20,       # ~16k-slot table cap
+            "AURON_HIP_AGG_TABLE_SLOTS": 1 << 14}
+    t = blaze_amd.Task(plan.plan_partial_final(),
+                       batches=batches_of(keys, vals, vv), conf=conf)
+    outputs = t.run()
+    assert t.metric("spill_count") > 0
+    k, s, sv, c = engine_final(outputs)
+    ref = run_oracle(keys, vals, vv).output()
+    assert len(k) == len(ref["keys"])
+    gi = np.argsort(k, kind="stable")
+    oi = np.argsort(ref["keys"], kind="stable")
+    np.testing.assert_array_equal(k[gi], ref["keys"][oi])
+    np.testing.assert_array_equal(c[gi], ref["counts"][oi])
+    np.testing.assert_array_equal(s[gi], ref["sums"][oi])
+    t.finalize()
+
+
+def test_spill_preserves_special_groups():
+    """Null-key and i64::MIN-key groups stay resident across spills and are
+    emitted exactly once."""
+    n = 120_000
+    rng = np.random.default_rng(3)
+    keys = rng.integers(0, 50_000, n).astype(np.int64)
+    keys[::1000] = -2**63
+    kv = np.ones(n, bool)
+    kv[::997] = False
+    vals = np.ones(n)
+    conf = {"AURON_HIP_MEM_BUDGET": 1 << 20,
+            "AURON_HIP_AGG_TABLE_SLOTS": 1 << 14}
+    t = blaze_amd.Task(plan.plan_partial_final(),
+                       batches=[[(keys, kv), (vals, None)]], conf=conf)
+    outputs = t.run()
+    assert t.metric("spill_count") > 0
+    k, s, sv, c = engine_final(outputs)
+    orc = oracle.Agg()
+    orc.update(keys, vals, key_valid=kv)
+    ref = orc.output()
+    assert len(k) == orc.num_groups
+    np.testing.assert_array_equal(np.sort(c), np.sort(ref["counts"]))
+    assert c.sum() == n
+    t.finalize()
