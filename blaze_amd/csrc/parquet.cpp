@@ -561,9 +561,28 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
       for (int64_t i = 0; i < nvals; i++) valid_bits.push_back(true);
     }
 
+    // writers may FALL BACK from dictionary to PLAIN mid-chunk (e.g. pyarrow
+    // once the dictionary page hits its size limit); flatten the accumulated
+    // dict-encoded prefix to PLAIN on host when a mix appears (the dict
+    // portion is small by construction in that case)
+    auto flatten_dict = [&]() {
+      if (!out.uses_dict) return;
+      std::vector<uint8_t> flat;
+      flat.reserve(out.dict_indices.size() * vw);
+      for (uint32_t ix : out.dict_indices) {
+        if ((int64_t)ix >= out.dict_count) fail("parquet: dict index range");
+        flat.insert(flat.end(), out.dict_values.begin() + (size_t)ix * vw,
+                    out.dict_values.begin() + (size_t)(ix + 1) * vw);
+      }
+      out.plain.insert(out.plain.end(), flat.begin(), flat.end());
+      out.dict_indices.clear();
+      out.uses_dict = false;
+    };
+
     switch (ph.encoding) {
       case 0: {  // PLAIN
         if (dlen < non_null * vw) fail("parquet: short PLAIN data");
+        flatten_dict();
         out.plain.insert(out.plain.end(), data, data + non_null * vw);
         break;
       }
@@ -573,8 +592,20 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
         int bw = data[0];
         std::vector<uint32_t> idx;
         rle_bp_decode(data + 1, dlen - 1, bw, non_null, &idx);
-        out.dict_indices.insert(out.dict_indices.end(), idx.begin(), idx.end());
-        out.uses_dict = true;
+        if (!out.plain.empty()) {
+          // already flattened: expand this page directly
+          for (uint32_t ix : idx) {
+            if ((int64_t)ix >= out.dict_count)
+              fail("parquet: dict index range");
+            out.plain.insert(out.plain.end(),
+                             out.dict_values.begin() + (size_t)ix * vw,
+                             out.dict_values.begin() + (size_t)(ix + 1) * vw);
+          }
+        } else {
+          out.dict_indices.insert(out.dict_indices.end(), idx.begin(),
+                                  idx.end());
+          out.uses_dict = true;
+        }
         break;
       }
       default:
@@ -586,8 +617,6 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
 
   if (out.num_values != cm.num_values)
     fail("parquet: value count mismatch in chunk");
-  if (out.uses_dict && !out.plain.empty())
-    fail("parquet: mixed PLAIN and dictionary pages unsupported");
   if (out.null_count > 0) {
     out.validity.assign((valid_bits.size() + 7) / 8, 0);
     for (size_t i = 0; i < valid_bits.size(); i++)
