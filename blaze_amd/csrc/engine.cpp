@@ -2152,17 +2152,27 @@ int32_t auron_debug_conf_roundtrip(AuronCallbacks* cb, const char* key,
 int32_t auron_partition_ids(const int64_t* keys, int64_t n, int32_t P,
                             uint32_t* out) {
   try {
-    hipStream_t s;
-    AURON_HIP(hipStreamCreate(&s));
+    // cached stream + pinned staging: this helper runs once per bench step
+    // and pageable H2D/D2H of the ~1M-group arrays costs milliseconds
+    static hipStream_t s = nullptr;
+    static PinnedBuf pin_in, pin_out;
+    if (!s) AURON_HIP(hipStreamCreate(&s));
+    if (pin_in.size() < (size_t)n * 8) {
+      pin_in.alloc(n * 8);
+      pin_out.alloc(n * 4);
+    }
+    memcpy(pin_in.get(), keys, n * 8);
     DevBuf d_keys(n * 8), d_hash(n * 4), d_out(n * 4);
-    AURON_HIP(hipMemcpyAsync(d_keys.get(), keys, n * 8, hipMemcpyHostToDevice, s));
+    AURON_HIP(hipMemcpyAsync(d_keys.get(), pin_in.get(), n * 8,
+                             hipMemcpyHostToDevice, s));
     launch_hash_init(d_hash.get<int32_t>(), 42, n, s);
     launch_hash_fold_i64(d_keys.get<int64_t>(), nullptr, n,
                          d_hash.get<int32_t>(), s);
     launch_pmod(d_hash.get<int32_t>(), n, P, d_out.get<uint32_t>(), s);
-    AURON_HIP(hipMemcpyAsync(out, d_out.get(), n * 4, hipMemcpyDeviceToHost, s));
+    AURON_HIP(hipMemcpyAsync(pin_out.get(), d_out.get(), n * 4,
+                             hipMemcpyDeviceToHost, s));
     AURON_HIP(hipStreamSynchronize(s));
-    AURON_HIP(hipStreamDestroy(s));
+    memcpy(out, pin_out.get(), n * 4);
     return 0;
   } catch (const std::exception&) {
     return -1;
