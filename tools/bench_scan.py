@@ -13,6 +13,9 @@ import sys
 import time
 
 import numpy as np
+# torch must be imported (and its bundled HIP runtime loaded) BEFORE
+# libauron_hip.so, or torch's own HIP init fails with "No HIP GPUs"
+import torch  # noqa: F401
 
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 sys.path.insert(0, REPO)
@@ -59,8 +62,6 @@ def main():
 
     for _ in range(args.warmup):
         step()
-    import torch
-
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     groups = upd_ns = 0
