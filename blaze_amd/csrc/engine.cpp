@@ -10,7 +10,9 @@
 // The GPU path is mandatory: any HIP failure aborts the task loudly; there is
 // no CPU fallback anywhere in this library.
 #include <chrono>
+#include <thread>
 #include <cstring>
+#include <atomic>
 #include <map>
 #include <memory>
 #include <mutex>
@@ -1769,13 +1771,47 @@ struct Runtime {
       std::vector<uint32_t> proj = node.projection;
       if (proj.empty())
         for (uint32_t i = 0; i < fcols.size(); i++) proj.push_back(i);
-      for (int rg = 0; rg < pf.num_row_groups(); rg++) {
+      for (uint32_t ci : proj)
+        if (ci >= fcols.size()) FAIL("parquet: projection out of range");
+      // decode every (row group, column) chunk in parallel on the host
+      // cores (page decompress + RLE are the CPU-heavy part of the scan);
+      // GPU expansion/upload below stays in row-group order
+      const int nrg = pf.num_row_groups();
+      const size_t width = proj.size();
+      std::vector<PqColumnChunkData> decoded((size_t)nrg * width);
+      std::string decode_err;
+      std::mutex err_mu;
+      {
+        unsigned nw = std::min<unsigned>(
+            std::max(1u, std::thread::hardware_concurrency()),
+            (unsigned)((size_t)nrg * width));
+        std::atomic<size_t> next{0};
+        std::vector<std::thread> ws;
+        for (unsigned w = 0; w < nw; w++) {
+          ws.emplace_back([&]() {
+            for (;;) {
+              size_t i = next.fetch_add(1);
+              if (i >= (size_t)nrg * width) break;
+              try {
+                decoded[i] =
+                    pf.read_chunk((int)(i / width), (int)proj[i % width]);
+              } catch (const std::exception& ex) {
+                std::lock_guard<std::mutex> lk(err_mu);
+                if (decode_err.empty()) decode_err = ex.what();
+              }
+            }
+          });
+        }
+        for (auto& w : ws) w.join();
+      }
+      if (!decode_err.empty()) FAIL(decode_err);
+      for (int rg = 0; rg < nrg; rg++) {
         int64_t rows = pf.row_group_rows(rg);
         DevBatch b;
         b.num_rows = rows;
-        for (uint32_t ci : proj) {
-          if (ci >= fcols.size()) FAIL("parquet: projection out of range");
-          PqColumnChunkData cd = pf.read_chunk(rg, (int)ci);
+        for (size_t pi = 0; pi < width; pi++) {
+          uint32_t ci = proj[pi];
+          PqColumnChunkData cd = std::move(decoded[(size_t)rg * width + pi]);
           const int w = (int)dtype_width(fcols[ci].dtype());
           DevColumn c;
           c.dt = fcols[ci].dtype();
