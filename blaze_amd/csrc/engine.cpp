@@ -953,6 +953,17 @@ class AggOp {
     }
   }
 
+  // D2H through a reusable pinned staging buffer (pageable D2H costs ms on
+  // this stack); dst vector is filled from the pinned copy
+  void d2h_pinned(void* dev, std::vector<uint8_t>* dst, size_t len) {
+    if (pinned_emit_.size() < len) pinned_emit_.alloc(len);
+    AURON_HIP(hipMemcpyAsync(pinned_emit_.get(), dev, len,
+                             hipMemcpyDeviceToHost, stream_));
+    AURON_HIP(hipStreamSynchronize(stream_));
+    dst->resize(len);
+    memcpy(dst->data(), pinned_emit_.get(), len);
+  }
+
   std::pair<int64_t, std::vector<HostOutCol>> emit_groups(
       const uint32_t* order_slots, int64_t n) {
     DBG("agg.emit n=%lld final=%d", (long long)n, (int)final_output_);
@@ -962,25 +973,18 @@ class AggOp {
     launch_agg_gather_out(t_, order_slots, n, keys.get<int64_t>(),
                           kvalid.get<uint8_t>(), sums.get<double>(),
                           svalid.get<uint8_t>(), cnts.get<long long>(), stream_);
-    auto d2h = [&](DevBuf& b, std::vector<uint8_t>* v, size_t len) {
-      v->resize(len);
-      AURON_HIP(hipMemcpyAsync(v->data(), b.get(), len, hipMemcpyDeviceToHost,
-                               stream_));
-    };
     HostOutCol key_col;
     key_col.dt = DType::Int64;
-    d2h(keys, &key_col.values, n * 8);
+    d2h_pinned(keys.get(), &key_col.values, n * 8);
     std::vector<uint8_t> kv(bm), sv(bm);
     AURON_HIP(hipMemcpyAsync(kv.data(), kvalid.get(), bm, hipMemcpyDeviceToHost,
                              stream_));
     AURON_HIP(hipMemcpyAsync(sv.data(), svalid.get(), bm, hipMemcpyDeviceToHost,
                              stream_));
     if (final_output_) {
-      std::vector<uint8_t> h_sums(n * 8), h_cnts(n * 8), h_avgs;
-      AURON_HIP(hipMemcpyAsync(h_sums.data(), sums.get(), n * 8,
-                               hipMemcpyDeviceToHost, stream_));
-      AURON_HIP(hipMemcpyAsync(h_cnts.data(), cnts.get(), n * 8,
-                               hipMemcpyDeviceToHost, stream_));
+      std::vector<uint8_t> h_sums, h_cnts, h_avgs;
+      d2h_pinned(sums.get(), &h_sums, n * 8);
+      d2h_pinned(cnts.get(), &h_cnts, n * 8);
       bool need_avg = false;
       for (uint32_t k : agg_kinds_) need_avg |= (k == AGGL_AVG);
       DevBuf avgs;
@@ -988,9 +992,7 @@ class AggOp {
         avgs.alloc(n * 8);
         launch_avg_div(sums.get<double>(), cnts.get<long long>(), n,
                        avgs.get<double>(), stream_);
-        h_avgs.resize(n * 8);
-        AURON_HIP(hipMemcpyAsync(h_avgs.data(), avgs.get(), n * 8,
-                                 hipMemcpyDeviceToHost, stream_));
+        d2h_pinned(avgs.get(), &h_avgs, n * 8);
       }
       AURON_HIP(hipStreamSynchronize(stream_));
       attach_validity(&key_col, kv, n);
@@ -1026,11 +1028,7 @@ class AggOp {
       HostOutCol buf_col;
       buf_col.dt = DType::Binary;
       buf_col.offsets = std::move(h_offs);
-      buf_col.values.resize(buf_col.offsets[n]);
-      AURON_HIP(hipMemcpyAsync(buf_col.values.data(), data.get(),
-                               buf_col.values.size(), hipMemcpyDeviceToHost,
-                               stream_));
-      AURON_HIP(hipStreamSynchronize(stream_));
+      d2h_pinned(data.get(), &buf_col.values, (size_t)buf_col.offsets[n]);
       attach_validity(&key_col, kv, n);
       cols = {std::move(key_col), std::move(buf_col)};
     }
@@ -1118,7 +1116,7 @@ class AggOp {
   hipEvent_t ev_start_ = nullptr, ev_stop_ = nullptr;
   AggTable t_;
   DevBuf d_slots_, d_special_, d_ng_, d_err_;
-  PinnedBuf pinned_meta_;
+  PinnedBuf pinned_meta_, pinned_emit_;
   // two-phase scratch (allocated on first large chunk)
   DevBuf d_part_, d_leftover_, d_counts_, d_scanned_, d_scan_tmp_, d_offsets_,
       d_staged_, d_counters_;

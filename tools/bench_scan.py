@@ -93,9 +93,9 @@ def main():
             "compressed_bytes_per_s": round(size / dt, 0),
             "groups_out": groups,
         },
-        "notes": "scan decode: footer/pages+decompress on host (single "
-                 "thread), validity/dictionary expansion + filter + agg on "
-                 "GPU; the host decompress bounds this config — see DESIGN.md",
+        "notes": "scan decode: footer/pages+decompress on host (parallel "
+                 "across cores), validity/dictionary expansion + filter + agg "
+                 "on GPU; host decode still bounds this config — DESIGN.md",
     }
     print(json.dumps(out))
 
