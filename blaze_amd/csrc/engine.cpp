@@ -671,7 +671,7 @@ class AggOp {
                          mat + 1, nullptr, &tb, stream_);
       d_scan_tmp_.alloc(tb);
       d_staged_.alloc((int64_t)nbuck * AGG2_LSLOTS * sizeof(StagedGroup));
-      d_counters_.alloc(16);  // staged_n, lo_n
+      d_counters_.alloc(24);  // staged_n, lo_n, special_rows
       // layout: counts[nbuck+1] | offs[nbuck+1] | counters[2] (8-aligned)
       pinned_agg2_.alloc(2 * (size_t)(nbuck + 1) * 4 + 64);
     }
@@ -679,33 +679,25 @@ class AggOp {
     AURON_HIP(hipEventCreate(&e0));
     AURON_HIP(hipEventCreate(&e1));
     AURON_HIP(hipEventRecord(e0, stream_));
-    // P1: per-block histogram matrix (+ special-row count)
+    // P1: per-block histogram matrix (+ special-row count). ONE host sync
+    // per chunk: the scan total feeds the offsets kernel on-device, and the
+    // special/staged/leftover counters are read back together after phase A.
     int64_t mat = (int64_t)nbuck << AGG2_GRID_LOG2;
-    AURON_HIP(hipMemsetAsync(d_counters_.get(), 0, 16, stream_));
+    AURON_HIP(hipMemsetAsync(d_counters_.get(), 0, 24, stream_));
     launch_agg2_hist(keys, kv, chunk, AGG2_NBUCK_LOG2,
                      d_counts_.get<uint32_t>(),
-                     (uint32_t*)d_counters_.get<uint8_t>() + 2, stream_);
-    // device exclusive scan over the flat matrix -> per-(block,bucket) bases
+                     (uint32_t*)(d_counters_.get<uint8_t>() + 16), stream_);
     size_t tb = d_scan_tmp_.size();
     scan_counts_matrix(d_counts_.get<uint32_t>(), d_scanned_.get<uint32_t>(),
                        mat + 1, d_scan_tmp_.get(), &tb, stream_);
-    uint32_t* h_meta = pinned_agg2_.get<uint32_t>();
-    AURON_HIP(hipMemcpyAsync(h_meta, (uint32_t*)d_counters_.get<uint8_t>() + 2,
-                             4, hipMemcpyDeviceToHost, stream_));
-    AURON_HIP(hipMemcpyAsync(h_meta + 1, d_scanned_.get<uint32_t>() + mat, 4,
-                             hipMemcpyDeviceToHost, stream_));
-    AURON_HIP(hipStreamSynchronize(stream_));
-    uint32_t special_rows = h_meta[0];
-    uint32_t total_rows = h_meta[1];
-    launch_agg2_offsets(d_scanned_.get<uint32_t>(), AGG2_NBUCK_LOG2, total_rows,
+    launch_agg2_offsets(d_scanned_.get<uint32_t>(), AGG2_NBUCK_LOG2,
                         d_offsets_.get<uint32_t>(), stream_);
-    // P2: scatter + specials
+    // P2: scatter
     launch_agg2_scatter(keys, kv, vals, vv, chunk, AGG2_NBUCK_LOG2,
                         d_scanned_.get<uint32_t>(), d_part_.get<PartRow>(),
                         stream_);
-    if (special_rows)
-      launch_agg2_specials(t_, keys, kv, vals, vv, chunk, row_cursor_, stream_);
-    // A: per-bucket LDS aggregation (counters[0..1] already zeroed above)
+    // A: per-bucket LDS aggregation (counters zeroed above; staged_n at +0,
+    // lo_n at +1, special count at byte offset 8 via the +2 uint32 slot)
     launch_agg2_bucket(d_part_.get<PartRow>(), d_offsets_.get<uint32_t>(),
                        nbuck, d_staged_.get<StagedGroup>(),
                        d_counters_.get<unsigned long long>(),
@@ -716,10 +708,21 @@ class AggOp {
     unsigned long long* h_ctr =
         (unsigned long long*)(pinned_agg2_.get<uint8_t>() +
                               2 * (size_t)(nbuck + 1) * 4);
-    AURON_HIP(hipMemcpyAsync(h_ctr, d_counters_.get(), 16,
+    AURON_HIP(hipMemcpyAsync(h_ctr, d_counters_.get(), 24,
                              hipMemcpyDeviceToHost, stream_));
+    uint64_t* h_ng = (uint64_t*)(h_ctr + 3);
+    AURON_HIP(hipMemcpyAsync(h_ng, t_.num_groups, 8, hipMemcpyDeviceToHost,
+                             stream_));
     AURON_HIP(hipStreamSynchronize(stream_));
-    int64_t staged_n = (int64_t)h_ctr[0], lo_n = (int64_t)h_ctr[1];
+    int64_t staged_n = (int64_t)h_ctr[0];
+    int64_t lo_n = (int64_t)h_ctr[1];
+    uint32_t special_rows = (uint32_t)h_ctr[2];
+    if (special_rows)
+      launch_agg2_specials(t_, keys, kv, vals, vv, chunk, row_cursor_, stream_);
+    // true cardinality from the piggybacked pre-merge read (no extra sync);
+    // the piecewise loops below grow the bound as they merge
+    ng_true_ = *h_ng;
+    ng_bound_ = *h_ng + (special_rows ? 2u : 0u);
     // merge: table inserts bounded by the COUNTED lists; piecewise so a
     // VRAM budget (spill) smaller than the staged list still works
     for (int64_t done2 = 0; done2 < staged_n;) {
@@ -761,10 +764,6 @@ class AggOp {
     AURON_HIP(hipEventRecord(e1, stream_));
     ev_pairs_.push_back({e0, e1});
     update_rows_ += chunk;
-    // refresh instead of accumulating the conservative bound: staged lists
-    // mostly merge into EXISTING groups, and a stale bound forces phantom
-    // table growth (measured: 8M->32M slots and 17 ms of compact/rebuild)
-    refresh_ng();
     DBG("agg.2phase chunk=%lld staged=%lld leftover=%lld special=%u",
         (long long)chunk, (long long)staged_n, (long long)lo_n, special_rows);
   }

@@ -170,7 +170,7 @@ void launch_agg2_hist(const int64_t* keys, const uint8_t* key_valid, int64_t n,
 void scan_counts_matrix(const uint32_t* counts, uint32_t* scanned, int64_t n,
                         void* temp, size_t* temp_bytes, hipStream_t s);
 void launch_agg2_offsets(const uint32_t* scanned, int nbuck_log2,
-                         uint32_t total, uint32_t* offsets, hipStream_t s);
+                         uint32_t* offsets, hipStream_t s);
 void launch_agg2_scatter(const int64_t* keys, const uint8_t* key_valid,
                          const double* vals, const uint8_t* val_valid,
                          int64_t n, int nbuck_log2, const uint32_t* scanned,

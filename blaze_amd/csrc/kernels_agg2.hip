@@ -85,14 +85,16 @@ __global__ void k_agg2_hist(const int64_t* __restrict__ keys,
   if (special) atomicAdd(special_rows, special);
 }
 
-// offsets[b] = scanned[(b << GRID_LOG2) | 0]; offsets[nbuck] = total
+// offsets[b] = scanned[(b << GRID_LOG2) | 0]; offsets[nbuck] = the scan
+// total (scanned[nbuck << GRID_LOG2], the extra slot) — read on device so the
+// host never synchronizes for it
 __global__ void k_agg2_offsets(const uint32_t* __restrict__ scanned,
-                               int nbuck_log2, uint32_t total,
+                               int nbuck_log2,
                                uint32_t* __restrict__ offsets) {
   int nbuck = 1 << nbuck_log2;
   for (int b = (int)(blockIdx.x * blockDim.x + threadIdx.x); b <= nbuck;
        b += (int)(gridDim.x * blockDim.x))
-    offsets[b] = (b == nbuck) ? total : scanned[(size_t)b << GRID_LOG2];
+    offsets[b] = scanned[(size_t)b << GRID_LOG2];
 }
 
 // ---- phase P2: scatter into per-(block,bucket) reserved ranges -------------
@@ -351,9 +353,9 @@ void launch_agg2_hist(const int64_t* keys, const uint8_t* key_valid, int64_t n,
 }
 
 void launch_agg2_offsets(const uint32_t* scanned, int nbuck_log2,
-                         uint32_t total, uint32_t* offsets, hipStream_t s) {
+                         uint32_t* offsets, hipStream_t s) {
   hipLaunchKernelGGL(k_agg2_offsets, dim3(8), dim3(BLOCK), 0, s, scanned,
-                     nbuck_log2, total, offsets);
+                     nbuck_log2, offsets);
   check_launch2("k_agg2_offsets");
 }
 

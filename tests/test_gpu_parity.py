@@ -342,3 +342,29 @@ def test_spill_preserves_special_groups():
     np.testing.assert_array_equal(np.sort(c), np.sort(ref["counts"]))
     assert c.sum() == n
     t.finalize()
+
+
+def test_two_phase_with_special_keys():
+    """Chunks >= 4M rows take the two-phase (radix+LDS) path; null keys and
+    i64::MIN keys must route through the special slots there too (regression:
+    the special counter once aliased the leftover counter)."""
+    n = 5_000_000
+    rng = np.random.default_rng(77)
+    keys = rng.integers(0, 200_000, n).astype(np.int64)
+    keys[::5000] = -2**63
+    kv = np.ones(n, bool)
+    kv[::4999] = False
+    vals = rng.integers(0, 1000, n).astype(np.float64)
+    t = blaze_amd.Task(plan.plan_partial_final(),
+                       batches=[[(keys, kv), (vals, None)]])
+    outputs = t.run()
+    k, s, sv, c = engine_final(outputs)
+    orc = oracle.Agg()
+    orc.update(keys, vals, key_valid=kv)
+    ref = orc.output()
+    assert len(k) == orc.num_groups
+    np.testing.assert_array_equal(c, ref["counts"])
+    np.testing.assert_array_equal(s, ref["sums"])
+    np.testing.assert_array_equal(k[ref["key_valid"]],
+                                  ref["keys"][ref["key_valid"]])
+    t.finalize()
