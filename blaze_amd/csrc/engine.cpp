@@ -34,18 +34,25 @@ struct EngineError : std::runtime_error {
   using std::runtime_error::runtime_error;
 };
 
-// AURON_DEBUG=1 traces engine stages to stderr (diagnostics only)
+// AURON_DEBUG=1 traces engine stages (with ms timestamps) to stderr
 static bool debug_on() {
   static bool v = getenv("AURON_DEBUG") != nullptr;
   return v;
 }
-#define DBG(...)                               \
-  do {                                         \
-    if (debug_on()) {                          \
-      fprintf(stderr, "[auron] " __VA_ARGS__); \
-      fprintf(stderr, "\n");                   \
-      fflush(stderr);                          \
-    }                                          \
+static double dbg_ms() {
+  static auto t0 = std::chrono::steady_clock::now();
+  return std::chrono::duration<double, std::milli>(
+             std::chrono::steady_clock::now() - t0)
+      .count();
+}
+#define DBG(...)                                          \
+  do {                                                    \
+    if (debug_on()) {                                     \
+      fprintf(stderr, "[auron %9.2f] ", dbg_ms());        \
+      fprintf(stderr, __VA_ARGS__);                       \
+      fprintf(stderr, "\n");                              \
+      fflush(stderr);                                     \
+    }                                                     \
   } while (0)
 
 #define FAIL(msg) throw EngineError(msg)
@@ -437,6 +444,7 @@ class AggOp {
     };
     int64_t done = 0;
     DBG("agg.consume n=%lld merge=%d", (long long)b.num_rows, (int)merge_mode_);
+
     while (done < b.num_rows) {
       // two-phase path (update mode, large chunks): its table inserts are
       // bounded by counted staged/leftover lists, not by chunk rows, so it
@@ -994,6 +1002,7 @@ class AggOp {
         d2h_pinned(avgs.get(), &h_avgs, n * 8);
       }
       AURON_HIP(hipStreamSynchronize(stream_));
+      DBG("agg.emit final d2h done");
       attach_validity(&key_col, kv, n);
       cols.push_back(std::move(key_col));
       for (uint32_t k : agg_kinds_) {
@@ -1028,6 +1037,7 @@ class AggOp {
       buf_col.dt = DType::Binary;
       buf_col.offsets = std::move(h_offs);
       d2h_pinned(data.get(), &buf_col.values, (size_t)buf_col.offsets[n]);
+      DBG("agg.emit partial freeze d2h done");
       attach_validity(&key_col, kv, n);
       cols = {std::move(key_col), std::move(buf_col)};
     }
