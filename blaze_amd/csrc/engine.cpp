@@ -2120,6 +2120,7 @@ void auron_finalize(int64_t handle) {
 void auron_on_exit(void) {
   std::lock_guard<std::mutex> lk(g_mu);
   g_runtimes.clear();
+  DevPool::inst().clear([](void* p) { (void)hipFree(p); });
 }
 
 const char* auron_version(void) { return "auron-hip 0.1 gfx950"; }
