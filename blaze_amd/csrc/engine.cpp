@@ -812,6 +812,9 @@ class AggOp {
                        first_sorted->get<unsigned long long>(),
                        slots_sorted->get<uint32_t>(), (int64_t)ng, tmp.get(),
                        &tmp_bytes, stream_);
+    // pooled buffers (slots_u/first/tmp) must not be recycled while the sort
+    // still reads them — the pool, unlike hipFree, does not synchronize
+    AURON_HIP(hipStreamSynchronize(stream_));
     return (int64_t)ng;
   }
 
@@ -1442,6 +1445,8 @@ class FilterOp {
     for (auto& c : in.cols) {
       out.cols.push_back(gather_col(c, sel.get<uint32_t>(), m));
     }
+    // sel/mask/positions are pooled: drain the gathers before recycling
+    AURON_HIP(hipStreamSynchronize(stream_));
     held_.push_back(std::move(in));  // borrowed inputs stay alive
     return out;
   }
