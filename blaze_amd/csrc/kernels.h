@@ -160,7 +160,7 @@ struct PartRow {
 static_assert(sizeof(PartRow) == 24, "PartRow must be 24 bytes");
 constexpr int AGG2_LSLOTS = 2048;  // LDS table entries per bucket
 
-constexpr int AGG2_GRID_LOG2 = 8;  // hist/scatter grid: 256 blocks
+constexpr int AGG2_GRID_LOG2 = 9;  // hist/scatter grid: 512 blocks
 // per-BLOCK bucket histogram into the bucket-major counts matrix
 // [nbuck << AGG2_GRID_LOG2]
 void launch_agg2_hist(const int64_t* keys, const uint8_t* key_valid, int64_t n,

@@ -57,7 +57,7 @@ __device__ __forceinline__ uint32_t bucket_of(int64_t key, int nbuck_log2) {
 // counts_matrix is bucket-major: counts[(b << GRID_LOG2) | blockIdx]. An
 // exclusive scan over the flat matrix then gives every block a private,
 // contiguous output range per bucket — the scatter needs NO global atomics.
-static constexpr int GRID_LOG2 = 8;  // 256 blocks: per-(block,bucket) runs ~128 rows (3 KB) and the scatter's open write-lines fit per-XCD L2
+static constexpr int GRID_LOG2 = 9;  // 512 blocks (measured best: 256 blocks lose streaming occupancy, 2048 thrash write-lines)
 
 __global__ void k_agg2_hist(const int64_t* __restrict__ keys,
                             const uint8_t* __restrict__ key_valid, int64_t n,
