@@ -428,8 +428,21 @@ class AggOp {
 
   void consume(DevBatch&& b) {
     if (b.num_rows == 0) return;
-    const DevColumn& key = b.cols.at(key_col_);
-    if (key.dt != DType::Int64) FAIL("grouping key must be Int64");
+    DevColumn& key = b.cols.at(key_col_);
+    if (key.dt != DType::Int64 && key.dt != DType::Int32)
+      FAIL("grouping key must be Int64/Int32");
+    if (key_dt_ == DType::Unsupported) key_dt_ = key.dt;
+    if (key.dt != key_dt_) FAIL("grouping key dtype changed mid-stream");
+    if (key.dt == DType::Int32) {
+      // widen to the i64 slot table; the narrow Int32 view is restored at
+      // output (values round-trip exactly)
+      DevBuf wide(b.num_rows * 8);
+      launch_widen_i32_i64((const int32_t*)key.values, b.num_rows,
+                           wide.get<int64_t>(), stream_);
+      key.own_values = std::move(wide);
+      key.values = key.own_values.get();
+      key.dt = DType::Int64;
+    }
     if (skipping_) {
       skipped_.push_back(std::move(b));
       return;
@@ -539,8 +552,9 @@ class AggOp {
   }
 
   std::vector<OutField> output_fields() const {
+    DType kd = key_dt_ == DType::Unsupported ? DType::Int64 : key_dt_;
     if (final_output_) {
-      std::vector<OutField> f = {{key_name_, DType::Int64, true}};
+      std::vector<OutField> f = {{key_name_, kd, true}};
       for (size_t i = 0; i < agg_kinds_.size(); i++) {
         if (agg_kinds_[i] == AGGL_CNT)
           f.push_back({agg_names_[i], DType::Int64, false});
@@ -550,7 +564,7 @@ class AggOp {
       return f;
     }
     // partial/partial-merge: grouping + AGG_BUF (agg/mod.rs:37)
-    return {{key_name_, DType::Int64, true},
+    return {{key_name_, kd, true},
             {"#9223372036854775807", DType::Binary, false}};
   }
 
@@ -984,8 +998,15 @@ class AggOp {
                           kvalid.get<uint8_t>(), sums.get<double>(),
                           svalid.get<uint8_t>(), cnts.get<long long>(), stream_);
     HostOutCol key_col;
-    key_col.dt = DType::Int64;
-    d2h_pinned(keys.get(), &key_col.values, n * 8);
+    key_col.dt = key_dt_ == DType::Unsupported ? DType::Int64 : key_dt_;
+    if (key_col.dt == DType::Int32) {
+      DevBuf k32(n * 4);
+      launch_narrow_i64_i32(keys.get<int64_t>(), n, k32.get<int32_t>(),
+                            stream_);
+      d2h_pinned(k32.get(), &key_col.values, n * 4);
+    } else {
+      d2h_pinned(keys.get(), &key_col.values, n * 8);
+    }
     std::vector<uint8_t> kv(bm), sv(bm);
     AURON_HIP(hipMemcpyAsync(kv.data(), kvalid.get(), bm, hipMemcpyDeviceToHost,
                              stream_));
@@ -1067,9 +1088,10 @@ class AggOp {
                              offs.get<int32_t>(), data.get<uint8_t>(), layout_,
                              stream_);
     HostOutCol key_col, buf_col;
-    key_col.dt = DType::Int64;
-    key_col.values.resize(n * 8);
-    AURON_HIP(hipMemcpyAsync(key_col.values.data(), key.values, n * 8,
+    key_col.dt = key.dt;
+    size_t kw = dtype_width(key.dt);
+    key_col.values.resize(n * kw);
+    AURON_HIP(hipMemcpyAsync(key_col.values.data(), key.values, n * kw,
                              hipMemcpyDeviceToHost, stream_));
     std::vector<uint8_t> kv;
     if (key.validity) {
@@ -1109,6 +1131,7 @@ class AggOp {
   AggMode mode_ = AggMode::Partial;
   bool merge_mode_ = false, final_output_ = false;
   uint32_t key_col_ = 0, val_col_ = 0;
+  DType key_dt_ = DType::Unsupported;
   uint32_t layout_ = 0;
   std::vector<uint32_t> agg_kinds_;
   std::vector<std::string> agg_names_;
@@ -1139,8 +1162,9 @@ class AggOp {
 // -------------------------------------------------------------- ShuffleOp --
 class ShuffleOp {
  public:
-  ShuffleOp(const ShuffleWriterNode& node, const Conf& conf, hipStream_t stream)
-      : stream_(stream), node_(node) {
+  ShuffleOp(const ShuffleWriterNode& node, const Conf& conf, hipStream_t stream,
+            uint32_t task_partition_id)
+      : stream_(stream), node_(node), task_partition_id_(task_partition_id) {
     if (node.partitioning.kind == Repartition::Hash) {
       if (node.partitioning.hash_exprs.empty())
         FAIL("hash partitioning without exprs");
@@ -1148,8 +1172,9 @@ class ShuffleOp {
         if (e.kind != Expr::Column) FAIL("hash expr must be a Column");
         hash_cols_.push_back(e.col_index);
       }
-    } else if (node.partitioning.kind != Repartition::Single) {
-      FAIL("only Hash/Single partitioning on the hot path");
+    } else if (node.partitioning.kind != Repartition::Single &&
+               node.partitioning.kind != Repartition::RoundRobin) {
+      FAIL("unsupported partitioning kind");
     }
     P_ = (uint32_t)node.partitioning.partition_count;
     batch_size_ = conf.get_i("BATCH_SIZE", 10000);
@@ -1186,18 +1211,33 @@ class ShuffleOp {
       emit(all, perm, part_offsets, ncols, n);
       return;
     }
-    // murmur3 seed 42 fold over hash cols (shuffle/mod.rs:163-176)
-    DevBuf hashes(n * 4), part_ids(n * 4);
-    launch_hash_init(hashes.get<int32_t>(), 42, n, stream_);
-    for (uint32_t ci : hash_cols_) {
-      const DevColumn& c = all.cols.at(ci);
-      if (c.dt != DType::Int64)
-        FAIL("hash expr column must be Int64 (hot-path scope)");
-      launch_hash_fold_i64((const int64_t*)c.values, c.validity, n,
-                           hashes.get<int32_t>(), stream_);
+    DevBuf part_ids(n * 4);
+    if (node_.partitioning.kind == Repartition::RoundRobin) {
+      // evaluate_robin_partition_ids (shuffle/mod.rs:190-202) with
+      // start_rows = partition_id * 1000193 (buffered_data.rs:291-293); one
+      // continuous counter over all rows == the reference's per-flush chain
+      uint32_t start = (uint32_t)(((uint64_t)task_partition_id_ * 1000193ull) %
+                                  (uint64_t)P_);
+      launch_robin_ids(n, start, P_, part_ids.get<uint32_t>(), stream_);
+    } else {
+      // murmur3 seed 42 fold over hash cols (shuffle/mod.rs:163-176)
+      DevBuf hashes(n * 4);
+      launch_hash_init(hashes.get<int32_t>(), 42, n, stream_);
+      for (uint32_t ci : hash_cols_) {
+        const DevColumn& c = all.cols.at(ci);
+        if (c.dt == DType::Int64)
+          launch_hash_fold_i64((const int64_t*)c.values, c.validity, n,
+                               hashes.get<int32_t>(), stream_);
+        else if (c.dt == DType::Int32)
+          launch_hash_fold_i32((const int32_t*)c.values, c.validity, n,
+                               hashes.get<int32_t>(), stream_);
+        else
+          FAIL("hash expr column must be Int64/Int32 (hot-path scope)");
+      }
+      launch_pmod(hashes.get<int32_t>(), n, (int32_t)P_,
+                  part_ids.get<uint32_t>(), stream_);
+      AURON_HIP(hipStreamSynchronize(stream_));  // hashes is pooled
     }
-    launch_pmod(hashes.get<int32_t>(), n, (int32_t)P_, part_ids.get<uint32_t>(),
-                stream_);
     // histogram → host scan
     DevBuf counts(P_ * 4);
     AURON_HIP(hipMemsetAsync(counts.get(), 0, P_ * 4, stream_));
@@ -1342,10 +1382,15 @@ class ShuffleOp {
         cols[c].offsets = h_offsets[c].data();
       } else {
         size_t w = dtype_width(src.dt);
-        if (w != 8) FAIL("shuffle gather supports 8-byte primitives (hot path)");
         DevBuf d_out(n * w);
-        launch_gather_8((const uint8_t*)src.values, perm.get<uint32_t>(), n,
-                        d_out.get<uint8_t>(), stream_);
+        if (w == 8)
+          launch_gather_8((const uint8_t*)src.values, perm.get<uint32_t>(), n,
+                          d_out.get<uint8_t>(), stream_);
+        else if (w == 4)
+          launch_gather_4((const uint8_t*)src.values, perm.get<uint32_t>(), n,
+                          d_out.get<uint8_t>(), stream_);
+        else
+          FAIL("shuffle gather supports 4/8-byte primitives (hot path)");
         h_values[c].resize(n * w);
         AURON_HIP(hipMemcpyAsync(h_values[c].data(), d_out.get(), n * w,
                                  hipMemcpyDeviceToHost, stream_));
@@ -1379,6 +1424,7 @@ class ShuffleOp {
 
   hipStream_t stream_;
   const ShuffleWriterNode& node_;
+  uint32_t task_partition_id_ = 0;
   std::vector<uint32_t> hash_cols_;
   uint32_t P_ = 1;
   int64_t batch_size_ = 10000;
@@ -1675,8 +1721,8 @@ struct Runtime {
           break;
         case PlanNode::ShuffleWriter:
           st.kind = Stage::ShuffleS;
-          st.shuffle =
-              std::make_unique<ShuffleOp>(*node->shuffle_writer, conf, stream);
+          st.shuffle = std::make_unique<ShuffleOp>(*node->shuffle_writer, conf,
+                                                   stream, td->partition_id);
           break;
         case PlanNode::Filter:
           st.kind = Stage::FilterS;

@@ -15,6 +15,18 @@ namespace auron {
 void launch_hash_init(int32_t* hashes, int32_t seed, int64_t n, hipStream_t s);
 void launch_hash_fold_i64(const int64_t* vals, const uint8_t* valid, int64_t n,
                           int32_t* hashes, hipStream_t s);
+// 4-byte int column fold (hash_array_primitive Int32, spark_hash.rs)
+void launch_hash_fold_i32(const int32_t* vals, const uint8_t* valid, int64_t n,
+                          int32_t* hashes, hipStream_t s);
+// dtype bridges for Int32 grouping keys (internally widened to the i64 table)
+void launch_widen_i32_i64(const int32_t* in, int64_t n, int64_t* out,
+                          hipStream_t s);
+void launch_narrow_i64_i32(const int64_t* in, int64_t n, int32_t* out,
+                           hipStream_t s);
+// round-robin partition ids (evaluate_robin_partition_ids,
+// shuffle/mod.rs:190-202): part[i] = (start + i) % P
+void launch_robin_ids(int64_t n, uint32_t start, uint32_t P, uint32_t* out,
+                      hipStream_t s);
 // part_id = pmod(hash, P) (shuffle/mod.rs:178-188)
 void launch_pmod(const int32_t* hashes, int64_t n, int32_t num_partitions,
                  uint32_t* part_ids, hipStream_t s);

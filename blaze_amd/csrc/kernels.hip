@@ -124,6 +124,65 @@ void launch_hash_fold_i64(const int64_t* vals, const uint8_t* valid, int64_t n,
                      vals, valid, n, hashes);
   check_launch("k_hash_fold_i64");
 }
+__global__ void k_hash_fold_i32(const int32_t* __restrict__ vals,
+                                const uint8_t* __restrict__ valid, int64_t n,
+                                int32_t* __restrict__ hashes) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    if (!valid || bit_get_dev(valid, i)) {
+      // 4-byte LE word: one mix + fmix(len=4) (mur.rs:19-30 aligned path)
+      int32_t h1 = mur_mix_h1(hashes[i], mur_mix_k1(vals[i]));
+      hashes[i] = mur_fmix(h1, 4);
+    }
+  }
+}
+
+__global__ void k_widen_i32_i64(const int32_t* __restrict__ in, int64_t n,
+                                int64_t* __restrict__ out) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    out[i] = (int64_t)in[i];
+}
+
+__global__ void k_narrow_i64_i32(const int64_t* __restrict__ in, int64_t n,
+                                 int32_t* __restrict__ out) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    out[i] = (int32_t)in[i];
+}
+
+__global__ void k_robin_ids(int64_t n, uint32_t start, uint32_t P,
+                            uint32_t* __restrict__ out) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    out[i] = (uint32_t)((start + (uint64_t)i) % P);
+}
+
+void launch_hash_fold_i32(const int32_t* vals, const uint8_t* valid, int64_t n,
+                          int32_t* hashes, hipStream_t s) {
+  hipLaunchKernelGGL(k_hash_fold_i32, dim3(grid_for(n)), dim3(BLOCK), 0, s,
+                     vals, valid, n, hashes);
+  check_launch("k_hash_fold_i32");
+}
+void launch_widen_i32_i64(const int32_t* in, int64_t n, int64_t* out,
+                          hipStream_t s) {
+  hipLaunchKernelGGL(k_widen_i32_i64, dim3(grid_for(n)), dim3(BLOCK), 0, s, in,
+                     n, out);
+  check_launch("k_widen_i32_i64");
+}
+void launch_narrow_i64_i32(const int64_t* in, int64_t n, int32_t* out,
+                           hipStream_t s) {
+  hipLaunchKernelGGL(k_narrow_i64_i32, dim3(grid_for(n)), dim3(BLOCK), 0, s, in,
+                     n, out);
+  check_launch("k_narrow_i64_i32");
+}
+void launch_robin_ids(int64_t n, uint32_t start, uint32_t P, uint32_t* out,
+                      hipStream_t s) {
+  hipLaunchKernelGGL(k_robin_ids, dim3(grid_for(n)), dim3(BLOCK), 0, s, n,
+                     start, P, out);
+  check_launch("k_robin_ids");
+}
+
 void launch_pmod(const int32_t* hashes, int64_t n, int32_t P, uint32_t* part_ids,
                  hipStream_t s) {
   hipLaunchKernelGGL(k_pmod, dim3(grid_for(n)), dim3(BLOCK), 0, s, hashes, n, P,

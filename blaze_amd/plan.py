@@ -140,10 +140,11 @@ def sum_count_aggs(val_index=1):
     ]
 
 
-def plan_partial_final(resource_id="input0", skipping=False):
+def plan_partial_final(resource_id="input0", skipping=False, key_dt=DT_INT64):
     """FFIReader -> Agg(Partial) -> Agg(Final): config 1/2 shape
     (mirrors agg_exec.rs fuzztest:714-843 topology)."""
-    reader = ffi_reader(northstar_input_fields(), resource_id)
+    reader = ffi_reader([field("key", key_dt, False),
+                         field("val", DT_FLOAT64, True)], resource_id)
     partial = agg(reader, [column("key", 0)], sum_count_aggs(1),
                   [MODE_PARTIAL, MODE_PARTIAL], ["key"], ["sum", "cnt"],
                   supports_partial_skipping=skipping)
@@ -324,3 +325,14 @@ def plan_parquet_filter_agg(files, cutoff=None, key_field="key",
     final = agg(partial, [column(key_field, 0)], sum_count_aggs(1),
                 [MODE_FINAL, MODE_FINAL], [key_field], ["sum", "cnt"])
     return task_definition(final)
+
+
+def plan_shuffle_robin(data_file, index_file, num_partitions, partition_id,
+                       fields=None):
+    """FFIReader -> ShuffleWriter(RoundRobin(P)) (shuffle/mod.rs:190-202)."""
+    if fields is None:
+        fields = northstar_input_fields()
+    reader = ffi_reader(fields, "input0")
+    rep = _len_field(3, _varint_field(1, num_partitions))  # round_robin = 3
+    sw = shuffle_writer(reader, rep, data_file, index_file)
+    return task_definition(sw, partition_id=partition_id)
