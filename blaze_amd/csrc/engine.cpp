@@ -1818,6 +1818,65 @@ struct Runtime {
             .count();
   }
 
+  // Evaluate a Column-vs-Literal pruning predicate against row-group stats:
+  // returns false only when the predicate PROVABLY excludes every row
+  // (parquet_exec.rs row-group stats pruning analog; conservative on any
+  // unsupported shape).
+  static bool rg_may_match(const Expr& e, const ParquetFile& pf, int rg) {
+    if (e.kind != Expr::BinaryExpr || e.children.size() != 2) return true;
+    if (e.op == "And")
+      return rg_may_match(e.children[0], pf, rg) &&
+             rg_may_match(e.children[1], pf, rg);
+    const Expr* col = nullptr;
+    const Expr* lit = nullptr;
+    bool flipped = false;
+    if (e.children[0].kind == Expr::Column &&
+        e.children[1].kind == Expr::Literal) {
+      col = &e.children[0];
+      lit = &e.children[1];
+    } else if (e.children[1].kind == Expr::Column &&
+               e.children[0].kind == Expr::Literal) {
+      col = &e.children[1];
+      lit = &e.children[0];
+      flipped = true;
+    } else {
+      return true;
+    }
+    if (col->col_index >= pf.columns().size()) return true;
+    PqColStats s = pf.column_stats(rg, (int)col->col_index);
+    if (!s.has_minmax) return true;
+    ScalarLit sl;
+    std::string err2;
+    if (!decode_ipc_scalar(lit->literal_ipc.data(), lit->literal_ipc.size(),
+                           &sl, &err2) ||
+        sl.is_null)
+      return true;
+    double lo, hi, v;
+    int pt = pf.columns()[col->col_index].physical_type;
+    if (pt == 1 || pt == 2) {
+      lo = (double)s.min_i;
+      hi = (double)s.max_i;
+      v = (double)sl.i64;
+    } else if (pt == 4 || pt == 5) {
+      lo = s.min_f;
+      hi = s.max_f;
+      v = sl.f64;
+    } else {
+      return true;
+    }
+    std::string op = e.op;
+    if (flipped) {
+      op = op == "Lt" ? "Gt" : op == "LtEq" ? "GtEq"
+           : op == "Gt" ? "Lt" : op == "GtEq" ? "LtEq" : op;
+    }
+    if (op == "Lt") return lo < v;
+    if (op == "LtEq") return lo <= v;
+    if (op == "Gt") return hi > v;
+    if (op == "GtEq") return hi >= v;
+    if (op == "Eq") return lo <= v && v <= hi;
+    return true;
+  }
+
   // ParquetScanExec source (parquet_exec.rs:150-203; decode scope per
   // parquet.h): footer/pages on host, validity/dictionary expansion on GPU.
   int64_t pump_parquet(const ParquetScanNode& node) {
@@ -1864,18 +1923,50 @@ struct Runtime {
       }
       if (!decode_err.empty()) FAIL(decode_err);
       for (int rg = 0; rg < nrg; rg++) {
+        bool keep = true;
+        for (const Expr& pr : node.pruning)
+          keep = keep && rg_may_match(pr, pf, rg);
+        if (!keep) {
+          DBG("parquet: pruned row group %d", rg);
+          continue;
+        }
         int64_t rows = pf.row_group_rows(rg);
         DevBatch b;
         b.num_rows = rows;
         for (size_t pi = 0; pi < width; pi++) {
           uint32_t ci = proj[pi];
           PqColumnChunkData cd = std::move(decoded[(size_t)rg * width + pi]);
-          const int w = (int)dtype_width(fcols[ci].dtype());
           DevColumn c;
           c.dt = fcols[ci].dtype();
           c.len = rows;
-          c.own_values.alloc((size_t)rows * w);
           bool has_nulls = cd.null_count > 0;
+          if (c.dt == DType::Utf8 || c.dt == DType::Binary) {
+            // row-aligned offsets/data assembled on host (parquet.cpp)
+            c.own_offsets.alloc((rows + 1) * 4);
+            AURON_HIP(hipMemcpyAsync(c.own_offsets.get(),
+                                     cd.bin_offsets.data(), (rows + 1) * 4,
+                                     hipMemcpyHostToDevice, stream));
+            c.offsets = c.own_offsets.get<int32_t>();
+            c.data_len = (int64_t)cd.bin_data.size();
+            c.own_values.alloc(cd.bin_data.empty() ? 1 : cd.bin_data.size());
+            if (!cd.bin_data.empty())
+              AURON_HIP(hipMemcpyAsync(c.own_values.get(), cd.bin_data.data(),
+                                       cd.bin_data.size(),
+                                       hipMemcpyHostToDevice, stream));
+            c.values = c.own_values.get();
+            if (has_nulls) {
+              c.own_validity.alloc(cd.validity.size());
+              AURON_HIP(hipMemcpyAsync(c.own_validity.get(),
+                                       cd.validity.data(), cd.validity.size(),
+                                       hipMemcpyHostToDevice, stream));
+              c.validity = c.own_validity.get<uint8_t>();
+            }
+            AURON_HIP(hipStreamSynchronize(stream));
+            b.cols.push_back(std::move(c));
+            continue;
+          }
+          const int w = (int)dtype_width(fcols[ci].dtype());
+          c.own_values.alloc((size_t)rows * w);
           DevBuf d_valid, d_mask, d_positions, d_packed, d_indices;
           if (has_nulls) {
             c.own_validity.alloc(cd.validity.size());

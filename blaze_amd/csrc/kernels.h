@@ -117,8 +117,6 @@ void launch_skip_freeze_write(const double* vals, const uint8_t* val_valid,
                               int64_t n, const int32_t* offsets, uint8_t* out,
                               uint32_t layout, hipStream_t s);
 
-// fill an i64 array with a constant
-void launch_fill_i64(int64_t* dst, int64_t value, int64_t n, hipStream_t s);
 // initialize table slots (key = empty sentinel, accs zero, first_row = ~0)
 void launch_slots_init(AggSlot* slots, int64_t n, hipStream_t s);
 
@@ -229,10 +227,5 @@ void launch_gather_bytes(const uint8_t* src_data, const int32_t* src_offsets,
 void sort_pairs_u64_u32(const unsigned long long* keys_in, const uint32_t* vals_in,
                         unsigned long long* keys_out, uint32_t* vals_out,
                         int64_t n, void* temp, size_t* temp_bytes, hipStream_t s);
-
-// byte-plane transpose for batch_serde (batch_serde.rs:271-306): out[b*n+i] =
-// in[i*w+b]
-void launch_byte_transpose(const uint8_t* in, uint8_t* out, int64_t n, int w,
-                           hipStream_t s);
 
 }  // namespace auron

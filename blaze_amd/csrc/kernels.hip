@@ -665,18 +665,6 @@ void launch_histogram(const uint32_t* part_ids, int64_t n, uint32_t P,
   check_launch("k_histogram");
 }
 
-__global__ void k_fill_i64(int64_t* __restrict__ dst, int64_t value, int64_t n) {
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += (int64_t)gridDim.x * blockDim.x)
-    dst[i] = value;
-}
-
-void launch_fill_i64(int64_t* dst, int64_t value, int64_t n, hipStream_t s) {
-  hipLaunchKernelGGL(k_fill_i64, dim3(grid_for(n)), dim3(BLOCK), 0, s, dst,
-                     value, n);
-  check_launch("k_fill_i64");
-}
-
 __global__ void k_slots_init(AggSlot* __restrict__ slots, int64_t n) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
@@ -993,23 +981,6 @@ void launch_scatter_packed(int width, const uint8_t* packed,
     throw std::runtime_error("scatter_packed: unsupported width");
   }
   check_launch("k_scatter_packed");
-}
-
-__global__ void k_byte_transpose(const uint8_t* __restrict__ in,
-                                 uint8_t* __restrict__ out, int64_t n, int w) {
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n * w;
-       i += (int64_t)gridDim.x * blockDim.x) {
-    int64_t row = i / w;
-    int b = (int)(i % w);
-    out[(int64_t)b * n + row] = in[i];
-  }
-}
-
-void launch_byte_transpose(const uint8_t* in, uint8_t* out, int64_t n, int w,
-                           hipStream_t s) {
-  hipLaunchKernelGGL(k_byte_transpose, dim3(grid_for(n * w)), dim3(BLOCK), 0, s,
-                     in, out, n, w);
-  check_launch("k_byte_transpose");
 }
 
 }  // namespace auron

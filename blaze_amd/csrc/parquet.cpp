@@ -109,6 +109,7 @@ struct SchemaElem {
   int repetition = -1;  // 0=REQUIRED 1=OPTIONAL 2=REPEATED
   std::string name;
   int num_children = 0;
+  int converted_type = -1;  // 0 = UTF8
 };
 
 SchemaElem read_schema_element(TReader& r) {
@@ -122,6 +123,7 @@ SchemaElem read_schema_element(TReader& r) {
       case 3: e.repetition = (int)r.zigzag(); break;
       case 4: e.name = r.binary(); break;
       case 5: e.num_children = (int)r.zigzag(); break;
+      case 6: e.converted_type = (int)r.zigzag(); break;
       default: r.skip(t);
     }
   }
@@ -421,6 +423,23 @@ ParquetFile::ParquetFile(const std::string& path) : path_(path) {
                         case 7: cm.total_compressed_size = r.zigzag(); break;
                         case 9: cm.data_page_offset = r.zigzag(); break;
                         case 11: cm.dict_page_offset = r.zigzag(); break;
+                        case 12: {  // Statistics{5:max_value, 6:min_value}
+                          int lst = 0;
+                          while (true) {
+                            auto [ids, ts] = r.field_header(&lst);
+                            if (ts == 0) break;
+                            if (ids == 5) {
+                              std::string v = r.binary();
+                              cm.stat_max.assign(v.begin(), v.end());
+                            } else if (ids == 6) {
+                              std::string v = r.binary();
+                              cm.stat_min.assign(v.begin(), v.end());
+                            } else {
+                              r.skip(ts);
+                            }
+                          }
+                          break;
+                        }
                         default: r.skip(tm);
                       }
                     }
@@ -453,6 +472,7 @@ ParquetFile::ParquetFile(const std::string& path) : path_(path) {
     PqColumnInfo ci;
     ci.name = e.name;
     ci.physical_type = e.type;
+    ci.utf8 = e.converted_type == 0;
     ci.nullable = e.repetition == 1;
     if (ci.dtype() == DType::Unsupported)
       fail("parquet: unsupported physical type for column " + e.name);
@@ -461,6 +481,25 @@ ParquetFile::ParquetFile(const std::string& path) : path_(path) {
   for (const auto& rg : row_groups_)
     if (rg.chunks.size() != columns_.size())
       fail("parquet: row group column count mismatch");
+}
+
+PqColStats ParquetFile::column_stats(int rg, int col) const {
+  const ChunkMeta& cm = row_groups_.at(rg).chunks.at(col);
+  const PqColumnInfo& ci = columns_.at(col);
+  PqColStats s;
+  if (cm.stat_min.empty() || cm.stat_max.empty()) return s;
+  auto rd = [&](const std::vector<uint8_t>& v, int64_t* i, double* f) -> bool {
+    switch (ci.physical_type) {
+      case 1: { int32_t x; if (v.size() < 4) return false; memcpy(&x, v.data(), 4); *i = x; return true; }
+      case 2: { int64_t x; if (v.size() < 8) return false; memcpy(&x, v.data(), 8); *i = x; return true; }
+      case 4: { float x; if (v.size() < 4) return false; memcpy(&x, v.data(), 4); *f = x; return true; }
+      case 5: { double x; if (v.size() < 8) return false; memcpy(&x, v.data(), 8); *f = x; return true; }
+      default: return false;
+    }
+  };
+  s.has_minmax = rd(cm.stat_min, &s.min_i, &s.min_f) &&
+                 rd(cm.stat_max, &s.max_i, &s.max_f);
+  return s;
 }
 
 PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
@@ -472,6 +511,12 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
   PqColumnChunkData out;
   out.num_values = 0;
   std::vector<bool> valid_bits;
+  const bool is_bytes = ci.physical_type == 6;
+  // BYTE_ARRAY values accumulate as non-null (len, bytes) pairs; row-aligned
+  // offsets are assembled at the end against the validity bits
+  std::vector<int32_t> nn_lens;
+  std::vector<uint8_t> nn_data;
+  std::vector<int32_t> dict_lens;  // dict entry lens for byte-array dicts
 
   int64_t pos = cm.dict_page_offset >= 0 ? cm.dict_page_offset
                                          : cm.data_page_offset;
@@ -514,6 +559,23 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
           decompress(page, ph.compressed_size, ph.uncompressed_size, &buf);
       out.dict_values.assign(data, data + ph.uncompressed_size);
       out.dict_count = ph.dict_num_values;
+      if (is_bytes) {
+        // index the dictionary's (u32 len, bytes) entries
+        size_t pos2 = 0;
+        dict_lens.clear();
+        std::vector<int32_t> offs;
+        offs.push_back(0);
+        for (int64_t i = 0; i < out.dict_count; i++) {
+          if (pos2 + 4 > out.dict_values.size())
+            fail("parquet: truncated byte-array dict");
+          uint32_t l;
+          memcpy(&l, out.dict_values.data() + pos2, 4);
+          pos2 += 4;
+          dict_lens.push_back((int32_t)pos2);  // start of bytes
+          dict_lens.push_back((int32_t)l);     // length
+          pos2 += l;
+        }
+      }
       continue;
     }
     if (ph.type != 0 && ph.type != 3)
@@ -581,6 +643,20 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
 
     switch (ph.encoding) {
       case 0: {  // PLAIN
+        if (is_bytes) {
+          int64_t pos2 = 0;
+          for (int64_t i = 0; i < non_null; i++) {
+            if (pos2 + 4 > dlen) fail("parquet: short BYTE_ARRAY data");
+            uint32_t l;
+            memcpy(&l, data + pos2, 4);
+            pos2 += 4;
+            if (pos2 + l > dlen) fail("parquet: short BYTE_ARRAY value");
+            nn_lens.push_back((int32_t)l);
+            nn_data.insert(nn_data.end(), data + pos2, data + pos2 + l);
+            pos2 += l;
+          }
+          break;
+        }
         if (dlen < non_null * vw) fail("parquet: short PLAIN data");
         flatten_dict();
         out.plain.insert(out.plain.end(), data, data + non_null * vw);
@@ -592,6 +668,18 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
         int bw = data[0];
         std::vector<uint32_t> idx;
         rle_bp_decode(data + 1, dlen - 1, bw, non_null, &idx);
+        if (is_bytes) {
+          for (uint32_t ix : idx) {
+            if ((int64_t)ix >= out.dict_count)
+              fail("parquet: dict index range");
+            int32_t start = dict_lens[(size_t)ix * 2];
+            int32_t l = dict_lens[(size_t)ix * 2 + 1];
+            nn_lens.push_back(l);
+            nn_data.insert(nn_data.end(), out.dict_values.begin() + start,
+                           out.dict_values.begin() + start + l);
+          }
+          break;
+        }
         if (!out.plain.empty()) {
           // already flattened: expand this page directly
           for (uint32_t ix : idx) {
@@ -617,6 +705,17 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
 
   if (out.num_values != cm.num_values)
     fail("parquet: value count mismatch in chunk");
+  if (is_bytes) {
+    // assemble row-aligned offsets (null rows zero-length)
+    out.bin_offsets.assign(out.num_values + 1, 0);
+    out.bin_data = std::move(nn_data);
+    size_t nn = 0;
+    for (int64_t i = 0; i < out.num_values; i++) {
+      int32_t l = valid_bits[i] ? nn_lens[nn++] : 0;
+      out.bin_offsets[i + 1] = out.bin_offsets[i] + l;
+    }
+    if (nn != nn_lens.size()) fail("parquet: byte-array count mismatch");
+  }
   if (out.null_count > 0) {
     out.validity.assign((valid_bits.size() + 7) / 8, 0);
     for (size_t i = 0; i < valid_bits.size(); i++)

@@ -38,11 +38,15 @@ struct PqColumnChunkData {
   std::vector<uint8_t> validity;       // LSB bitmap, empty = all valid
   int64_t null_count = 0;
   bool uses_dict = false;
+  // BYTE_ARRAY columns: row-aligned (null rows zero-length), host-assembled
+  std::vector<int32_t> bin_offsets;    // num_values + 1
+  std::vector<uint8_t> bin_data;
 };
 
 struct PqColumnInfo {
   std::string name;
-  int physical_type = -1;  // parquet: 1=INT32, 2=INT64, 4=FLOAT, 5=DOUBLE
+  int physical_type = -1;  // 1=INT32, 2=INT64, 4=FLOAT, 5=DOUBLE, 6=BYTE_ARRAY
+  bool utf8 = false;       // converted_type UTF8
   bool nullable = false;   // max_def == 1
   DType dtype() const {
     switch (physical_type) {
@@ -50,9 +54,18 @@ struct PqColumnInfo {
       case 2: return DType::Int64;
       case 4: return DType::Float32;
       case 5: return DType::Float64;
+      case 6: return utf8 ? DType::Utf8 : DType::Binary;
       default: return DType::Unsupported;
     }
   }
+};
+
+// row-group column statistics for pruning (parquet Statistics min_value=6 /
+// max_value=5, PLAIN-encoded; parquet_exec.rs row-group stats pruning analog)
+struct PqColStats {
+  bool has_minmax = false;
+  int64_t min_i = 0, max_i = 0;  // INT32/INT64 widened
+  double min_f = 0, max_f = 0;   // FLOAT/DOUBLE
 };
 
 class ParquetFile {
@@ -64,6 +77,7 @@ class ParquetFile {
   int num_row_groups() const { return (int)row_groups_.size(); }
   int64_t row_group_rows(int rg) const { return row_groups_[rg].num_rows; }
   const std::vector<PqColumnInfo>& columns() const { return columns_; }
+  PqColStats column_stats(int rg, int col) const;
 
   // decode one column chunk of one row group
   PqColumnChunkData read_chunk(int rg, int col) const;
@@ -75,6 +89,7 @@ class ParquetFile {
     int64_t total_compressed_size = 0;
     int64_t num_values = 0;
     int codec = 0;
+    std::vector<uint8_t> stat_min, stat_max;  // PLAIN-encoded min/max_value
   };
   struct RowGroupMeta {
     int64_t num_rows = 0;

@@ -126,6 +126,7 @@ struct ParquetScanNode {
   std::vector<std::string> files;     // FileGroup.files[].path
   Schema schema;                      // FileScanExecConf.schema = 4
   std::vector<uint32_t> projection;   // FileScanExecConf.projection = 6
+  std::vector<Expr> pruning;          // pruning_predicates = 2 (rg stats)
   std::string fs_resource_id;         // = 3 (unused: POSIX paths read directly)
 };
 

@@ -339,8 +339,7 @@ std::unique_ptr<ParquetScanNode> decode_parquet_scan(Reader r,
     } else if (f == 3) {
       n->fs_resource_id = r.str();
     } else if (f == 2) {
-      if (err->empty()) *err = "parquet pruning predicates unsupported";
-      r.skip(w);
+      n->pruning.push_back(decode_expr(r.sub(), err));
     } else {
       r.skip(w);
     }

@@ -296,9 +296,10 @@ def partitioned_file(path, size):
     return _len_field(1, path.encode()) + _varint_field(2, size)
 
 
-def parquet_scan(files, fields, projection=(), resource_id="fs0"):
+def parquet_scan(files, fields, projection=(), pruning=(), resource_id="fs0"):
     """PhysicalPlanNode{parquet_scan = 5} -> ParquetScanExecNode
-    (auron.proto:415-419) with FileScanExecConf (:404-413).
+    (auron.proto:415-419) with FileScanExecConf (:404-413) and optional
+    pruning_predicates (= 2, row-group stats pruning).
     files: list of (path, size)."""
     fg = b"".join(_len_field(1, partitioned_file(p, sz)) for p, sz in files)
     conf = _varint_field(1, 1) + _varint_field(2, 0)
@@ -306,7 +307,10 @@ def parquet_scan(files, fields, projection=(), resource_id="fs0"):
     conf += _len_field(4, schema(fields))
     if projection:
         conf += _len_field(6, b"".join(_varint(i) for i in projection))
-    node = _len_field(1, conf) + _len_field(3, resource_id.encode())
+    node = _len_field(1, conf)
+    for pr in pruning:
+        node += _len_field(2, pr)
+    node += _len_field(3, resource_id.encode())
     return _len_field(5, node)
 
 

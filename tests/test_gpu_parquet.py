@@ -97,3 +97,103 @@ def test_parquet_data_page_v2(tmp_path):
     ps = _write(tmp_path, "v2.parquet", keys, vals, vv, "snappy", True,
                 row_group_size=20_000, data_page_version="2.0")
     _run_and_check(ps, keys, vals, vv)
+
+
+def test_parquet_utf8_column_through_shuffle(tmp_path):
+    """BYTE_ARRAY/UTF8 columns decode (plain + dictionary) and round-trip
+    through a Single-partition shuffle write."""
+    import os
+    rng = np.random.default_rng(41)
+    n = 50_000
+    keys = rng.integers(0, 1000, n).astype(np.int64)
+    names = np.array([f"name_{k % 97}" for k in keys], dtype=object)
+    names[::113] = None
+    t = pa.table({"key": pa.array(keys, pa.int64()),
+                  "name": pa.array(names, pa.utf8())})
+    path = str(tmp_path / "s.parquet")
+    pq.write_table(t, path, compression="zstd", row_group_size=20_000)
+    size = os.path.getsize(path)
+
+    fields = [plan.field("key", plan.DT_INT64, True),
+              plan.field("name", plan.DT_UTF8, True)]
+    scan = plan.parquet_scan([(path, size)], fields)
+    rep = plan.single_repartition()
+    data_file = str(tmp_path / "s.data")
+    index_file = str(tmp_path / "s.index")
+    sw = plan.shuffle_writer(scan, rep, data_file, index_file)
+    tk = blaze_amd.Task(plan.task_definition(sw))
+    assert tk.run() == []
+    tk.finalize()
+
+    payload = oracle.ipc_decode(open(data_file, "rb").read())
+    # decode batches: [rows][key col][name col (bytes)]
+    got_keys, got_names = [], []
+    pos = 0
+    while pos < len(payload):
+        rows, k = oracle.read_len(payload[pos:])
+        pos += k
+        # key col: has_null header + transposed i64
+        hn, k2 = oracle.read_len(payload[pos:])
+        pos += k2
+        if hn:
+            pos += (rows + 7) // 8
+        planes = np.frombuffer(payload[pos:pos + 8 * rows],
+                               dtype=np.uint8).reshape(8, rows)
+        got_keys.append(np.ascontiguousarray(planes.T).reshape(-1).view(np.int64))
+        pos += 8 * rows
+        # name col: has_null + bitmap + transposed lens + data
+        hn, k2 = oracle.read_len(payload[pos:])
+        pos += k2
+        valid = None
+        if hn:
+            bm = np.frombuffer(payload[pos:pos + (rows + 7) // 8], np.uint8)
+            valid = np.unpackbits(bm, bitorder="little")[:rows].astype(bool)
+            pos += (rows + 7) // 8
+        lp = np.frombuffer(payload[pos:pos + 4 * rows],
+                           dtype=np.uint8).reshape(4, rows)
+        lens = np.ascontiguousarray(lp.T).reshape(-1).view(np.int32)
+        pos += 4 * rows
+        total = int(lens.sum())
+        data = payload[pos:pos + total]
+        pos += total
+        offs = np.concatenate([[0], np.cumsum(lens)])
+        for i in range(rows):
+            if valid is not None and not valid[i]:
+                got_names.append(None)
+            else:
+                got_names.append(data[offs[i]:offs[i + 1]].decode())
+    got_keys = np.concatenate(got_keys)
+    np.testing.assert_array_equal(got_keys, keys)
+    exp = [None if v is None else v for v in names.tolist()]
+    assert got_names == exp
+
+
+def test_parquet_row_group_pruning(tmp_path):
+    """pruning_predicates against row-group min/max stats: sorted keys give
+    disjoint per-group ranges; key < c must drop whole groups."""
+    import os
+    n = 100_000
+    keys = np.arange(n, dtype=np.int64)  # sorted -> rg k covers [k*25k,(k+1)*25k)
+    vals = np.ones(n)
+    t = pa.table({"key": pa.array(keys, pa.int64()),
+                  "val": pa.array(vals, pa.float64())})
+    path = str(tmp_path / "p.parquet")
+    pq.write_table(t, path, compression="snappy", row_group_size=25_000)
+    size = os.path.getsize(path)
+    fields = [plan.field("key", plan.DT_INT64, True),
+              plan.field("val", plan.DT_FLOAT64, True)]
+    cutoff = 30_000
+    pruning = [plan.binary_expr(plan.column("key", 0),
+                                plan.literal(cutoff, "int64"), "Lt")]
+    scan = plan.parquet_scan([(path, size)], fields, pruning=pruning)
+    partial = plan.agg(scan, [plan.column("key", 0)], plan.sum_count_aggs(1),
+                       [plan.MODE_PARTIAL] * 2, ["key"], ["sum", "cnt"])
+    final = plan.agg(partial, [plan.column("key", 0)], plan.sum_count_aggs(1),
+                     [plan.MODE_FINAL] * 2, ["key"], ["sum", "cnt"])
+    tk = blaze_amd.Task(plan.task_definition(final))
+    outs = tk.run()
+    got = np.concatenate([ob[0]["values"] for ob in outs])
+    tk.finalize()
+    # groups 0 (0..25k) and 1 (25k..50k) survive; 2,3 pruned
+    assert len(got) == 50_000
+    assert got.max() == 49_999
