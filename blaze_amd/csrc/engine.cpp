@@ -354,10 +354,11 @@ class AggOp {
     key_col_ = node.grouping_exprs[0].col_index;
     key_name_ = node.grouping_names.empty() ? std::string("key")
                                             : node.grouping_names[0];
-    // agg set: any list of SUM/COUNT/AVG over ONE shared argument column —
-    // the shapes whose accumulator state is {sum, cnt} (sum.rs, count.rs,
-    // avg.rs; AVG freeze = sum ++ count, avg.rs:208-217). Anything else
-    // fails loudly at plan build.
+    // agg set: any list of SUM/COUNT/AVG/MIN/MAX over ONE shared argument
+    // column — {sum, cnt} accumulators (sum.rs, count.rs, avg.rs; AVG freeze
+    // = sum ++ count, avg.rs:208-217) plus the optional side {min, max} pair
+    // (maxmin.rs:104-119; single-phase path only). Anything else fails
+    // loudly at plan build.
     if (node.agg_exprs.empty() || node.agg_exprs.size() > 8)
       FAIL("AggExec: 1..8 aggregates supported");
     if (node.modes.size() != node.agg_exprs.size())
@@ -376,10 +377,12 @@ class AggOp {
         case AGG_SUM: k = AGGL_SUM; break;
         case AGG_COUNT: k = AGGL_CNT; break;
         case AGG_AVG: k = AGGL_AVG; break;
+        case AGG_MIN: k = AGGL_MIN; has_mm_ = true; break;
+        case AGG_MAX: k = AGGL_MAX; has_mm_ = true; break;
         default:
-          FAIL("AggExec: only SUM/COUNT/AVG aggregates on this path");
+          FAIL("AggExec: only SUM/COUNT/AVG/MIN/MAX aggregates on this path");
       }
-      layout_ |= k << (2 * i);
+      layout_ |= k << (3 * i);
       agg_kinds_.push_back(k);
       agg_names_.push_back(i < node.agg_names.size() ? node.agg_names[i]
                                                      : "agg" + std::to_string(i));
@@ -461,8 +464,10 @@ class AggOp {
     while (done < b.num_rows) {
       // two-phase path (update mode, large chunks): its table inserts are
       // bounded by counted staged/leftover lists, not by chunk rows, so it
-      // chunks on partition-buffer size instead of table free slots
-      if (!merge_mode_ && b.num_rows - done >= AGG2_MIN_CHUNK) {
+      // chunks on partition-buffer size instead of table free slots.
+      // MIN/MAX agg sets stay single-phase: the LDS bucket kernel's slot
+      // holds {key,cnt,sum,first} only (perf note in DESIGN.md)
+      if (!merge_mode_ && !has_mm_ && b.num_rows - done >= AGG2_MIN_CHUNK) {
         int64_t chunk2 = std::min(b.num_rows - done, AGG2_MAX_CHUNK);
         if (done + chunk2 < b.num_rows) chunk2 &= ~(int64_t)7;
         two_phase_chunk(b, done, chunk2);
@@ -663,6 +668,11 @@ class AggOp {
     AURON_HIP(hipMemsetAsync(d_special_.get(), 0, 2 * 4, stream_));
     AURON_HIP(hipMemsetAsync(d_ng_.get(), 0, 8, stream_));
     launch_slots_init(t_.slots, cap + 2, stream_);
+    if (has_mm_) {
+      d_mm_.alloc((cap + 2) * 16);
+      t_.mm = d_mm_.get<unsigned long long>();
+      launch_mm_init(t_.mm, cap + 2, stream_);
+    }
   }
 
   static constexpr int64_t AGG2_MIN_CHUNK = 4 << 20;   // below: single-phase
@@ -863,7 +873,8 @@ class AggOp {
                                hipMemcpyHostToDevice, stream_));
       launch_agg_gather_out(t_, d_order.get<uint32_t>(), n,
                             keys.get<int64_t>(), nullptr, nullptr, nullptr,
-                            nullptr, stream_);
+                            nullptr, nullptr, nullptr, nullptr, nullptr,
+                            stream_);
       launch_agg_freeze_len(t_, d_order.get<uint32_t>(), n, lens.get<int32_t>(),
                             layout_, stream_);
       std::vector<int32_t> h_lens(n);
@@ -904,6 +915,7 @@ class AggOp {
 
   void reset_main() {
     launch_slots_init(t_.slots, t_.cap, stream_);  // specials untouched
+    if (has_mm_) launch_mm_init(t_.mm, t_.cap, stream_);
     uint64_t ng0 = (uint64_t)specials_count_;
     AURON_HIP(hipMemcpyAsync(t_.num_groups, &ng0, 8, hipMemcpyHostToDevice,
                              stream_));
@@ -940,7 +952,8 @@ class AggOp {
   void grow(int64_t new_cap) {
     AggTable old = t_;
     DevBuf oslots = std::move(d_slots_), os = std::move(d_special_),
-           ong = std::move(d_ng_), oerr = std::move(d_err_);
+           ong = std::move(d_ng_), oerr = std::move(d_err_),
+           omm = std::move(d_mm_);
     init_table(new_cap);
     launch_agg_rebuild(t_, old, stream_);
     AURON_HIP(hipStreamSynchronize(stream_));
@@ -994,9 +1007,18 @@ class AggOp {
     std::vector<HostOutCol> cols;
     size_t bm = (n + 7) / 8;
     DevBuf keys(n * 8), kvalid(bm), sums(n * 8), svalid(bm), cnts(n * 8);
+    DevBuf mins, mvalid, maxs, xvalid;
+    if (has_mm_ && final_output_) {
+      mins.alloc(n * 8);
+      mvalid.alloc(bm);
+      maxs.alloc(n * 8);
+      xvalid.alloc(bm);
+    }
     launch_agg_gather_out(t_, order_slots, n, keys.get<int64_t>(),
                           kvalid.get<uint8_t>(), sums.get<double>(),
-                          svalid.get<uint8_t>(), cnts.get<long long>(), stream_);
+                          svalid.get<uint8_t>(), cnts.get<long long>(),
+                          mins.get<double>(), mvalid.get<uint8_t>(),
+                          maxs.get<double>(), xvalid.get<uint8_t>(), stream_);
     HostOutCol key_col;
     key_col.dt = key_dt_ == DType::Unsupported ? DType::Int64 : key_dt_;
     if (key_col.dt == DType::Int32) {
@@ -1013,7 +1035,8 @@ class AggOp {
     AURON_HIP(hipMemcpyAsync(sv.data(), svalid.get(), bm, hipMemcpyDeviceToHost,
                              stream_));
     if (final_output_) {
-      std::vector<uint8_t> h_sums, h_cnts, h_avgs;
+      std::vector<uint8_t> h_sums, h_cnts, h_avgs, h_mins, h_maxs;
+      std::vector<uint8_t> mv(bm), xv(bm);
       d2h_pinned(sums.get(), &h_sums, n * 8);
       d2h_pinned(cnts.get(), &h_cnts, n * 8);
       bool need_avg = false;
@@ -1025,6 +1048,14 @@ class AggOp {
                        avgs.get<double>(), stream_);
         d2h_pinned(avgs.get(), &h_avgs, n * 8);
       }
+      if (has_mm_) {
+        d2h_pinned(mins.get(), &h_mins, n * 8);
+        d2h_pinned(maxs.get(), &h_maxs, n * 8);
+        AURON_HIP(hipMemcpyAsync(mv.data(), mvalid.get(), bm,
+                                 hipMemcpyDeviceToHost, stream_));
+        AURON_HIP(hipMemcpyAsync(xv.data(), xvalid.get(), bm,
+                                 hipMemcpyDeviceToHost, stream_));
+      }
       AURON_HIP(hipStreamSynchronize(stream_));
       DBG("agg.emit final d2h done");
       attach_validity(&key_col, kv, n);
@@ -1034,6 +1065,10 @@ class AggOp {
         if (k == AGGL_CNT) {
           ac.dt = DType::Int64;
           ac.values = h_cnts;
+        } else if (k == AGGL_MIN || k == AGGL_MAX) {
+          ac.dt = DType::Float64;
+          ac.values = (k == AGGL_MIN) ? h_mins : h_maxs;
+          attach_validity(&ac, (k == AGGL_MIN) ? mv : xv, n);
         } else {
           ac.dt = DType::Float64;
           ac.values = (k == AGGL_AVG) ? h_avgs : h_sums;
@@ -1133,6 +1168,7 @@ class AggOp {
   uint32_t key_col_ = 0, val_col_ = 0;
   DType key_dt_ = DType::Unsupported;
   uint32_t layout_ = 0;
+  bool has_mm_ = false;  // agg list contains MIN/MAX: side mm array active
   std::vector<uint32_t> agg_kinds_;
   std::vector<std::string> agg_names_;
   std::string key_name_;
@@ -1150,7 +1186,7 @@ class AggOp {
   std::vector<std::pair<hipEvent_t, hipEvent_t>> ev_pairs_;
   hipEvent_t ev_start_ = nullptr, ev_stop_ = nullptr;
   AggTable t_;
-  DevBuf d_slots_, d_special_, d_ng_, d_err_;
+  DevBuf d_slots_, d_special_, d_ng_, d_err_, d_mm_;
   PinnedBuf pinned_meta_, pinned_emit_;
   // two-phase scratch (allocated on first large chunk)
   DevBuf d_part_, d_leftover_, d_counts_, d_scanned_, d_scan_tmp_, d_offsets_,

@@ -47,12 +47,29 @@ struct AggSlot {
 };
 static_assert(sizeof(AggSlot) == 32, "AggSlot must be 32 bytes");
 
+// MIN/MAX accumulators (maxmin.rs:104-119 partial_update compare-and-keep)
+// live in an OPTIONAL side array `mm` so the 32-byte hot-path slot is
+// untouched when no MIN/MAX agg is present. Values are stored under the
+// order-preserving f64<->u64 map (sign ? ~bits : bits|MSB), so device
+// atomicMin/atomicMax on u64 implement f64 min/max directly and the init
+// sentinels (min=~0ull, max=0ull) make merging guard-free: merging an
+// absent/invalid part is a no-op atomic. Validity = "accumulator left its
+// sentinel"; the only values colliding with a sentinel are the two extreme
+// NaN payloads (0x7FFFFFFFFFFFFFFF / 0xFFFFFFFFFFFFFFFF), documented edge
+// (the reference's own NaN ordering is arrival-order-dependent:
+// maxmin.rs:110-116 partial_cmp==None takes the new value).
+constexpr unsigned long long MM_MIN_INIT = ~0ull;
+constexpr unsigned long long MM_MAX_INIT = 0ull;
+
 struct AggTable {
   int64_t cap = 0;                    // power of two
   AggSlot* slots = nullptr;           // [cap + 2]
   uint32_t* special_used = nullptr;   // [2]: {min-key group, null-key group}
   unsigned long long* num_groups = nullptr; // [1] device counter
   uint32_t* error_flag = nullptr;           // [1] raised on probe exhaustion
+  // interleaved {min_u64, max_u64} per slot, [2*(cap+2)]; null when the agg
+  // list has no MIN/MAX (the north-star path) — kernels skip it then
+  unsigned long long* mm = nullptr;
 };
 
 // grid-stride row update: keys/vals length n, rows globally numbered starting
@@ -62,11 +79,15 @@ void launch_agg_update(const AggTable& t, const int64_t* keys,
                        const uint8_t* val_valid, int64_t n, uint64_t row_offset,
                        hipStream_t s);
 
-// agg layout descriptor: 2 bits per agg LSB-first, 0-terminated
-// (AGGL_SUM=1, AGGL_CNT=2, AGGL_AVG=3); all aggs share one argument column.
-// AVG freeze = sum ++ count (avg.rs:208-217), so duplicated parts repeat the
-// shared accumulators.
-enum AggLayoutKind : uint32_t { AGGL_SUM = 1, AGGL_CNT = 2, AGGL_AVG = 3 };
+// agg layout descriptor: 3 bits per agg LSB-first, 0-terminated (8 aggs max
+// fits 24 bits); all aggs share one argument column. AVG freeze = sum ++
+// count (avg.rs:208-217); MIN/MAX freeze = the same prim [u8 valid][8B LE]
+// format as SUM (both are AccPrimColumn saves, acc.rs:335-347 — maxmin.rs
+// create_acc_column:91-93 uses the same generic prim column as sum.rs), so
+// duplicated parts repeat the shared accumulators.
+enum AggLayoutKind : uint32_t {
+  AGGL_SUM = 1, AGGL_CNT = 2, AGGL_AVG = 3, AGGL_MIN = 4, AGGL_MAX = 5
+};
 
 // merge rows of frozen partial state per the layout
 // (acc.rs:335-347 + count.rs:193-211 + io/mod.rs:60-79)
@@ -94,12 +115,15 @@ void launch_avg_div(const double* sums, const long long* cnts, int64_t n,
                     double* out, hipStream_t s);
 
 // gather ordered group outputs given order[] (group -> slot):
-// keys + key validity bits, sums + validity bits, counts
+// keys + key validity bits, sums + validity bits, counts, and (when t.mm is
+// set and the out pointers are non-null) min/max values + validity bits
 void launch_agg_gather_out(const AggTable& t, const uint32_t* order_slots,
                            int64_t num_groups, int64_t* out_keys,
                            uint8_t* out_key_validity, double* out_sums,
-                           uint8_t* out_sum_validity,
-                           long long* out_counts, hipStream_t s);
+                           uint8_t* out_sum_validity, long long* out_counts,
+                           double* out_mins, uint8_t* out_min_validity,
+                           double* out_maxs, uint8_t* out_max_validity,
+                           hipStream_t s);
 
 // freeze ordered groups into the Binary agg-buf wire format (a8):
 // pass 1 computes per-group byte length, host scans, pass 2 writes bytes.
@@ -119,6 +143,9 @@ void launch_skip_freeze_write(const double* vals, const uint8_t* val_valid,
 
 // initialize table slots (key = empty sentinel, accs zero, first_row = ~0)
 void launch_slots_init(AggSlot* slots, int64_t n, hipStream_t s);
+
+// initialize n interleaved {min,max} accumulator pairs to their sentinels
+void launch_mm_init(unsigned long long* mm, int64_t n, hipStream_t s);
 
 // rebuild `src` table into the (larger, initialized) `dst` table; each source
 // slot holds a distinct key so plain stores after the CAS claim are race-free.

@@ -238,6 +238,21 @@ __device__ __forceinline__ int64_t agg_upsert_slot(const AggTable t, int64_t key
   return -1;
 }
 
+// order-preserving f64 <-> u64 map: monotone, so u64 atomicMin/atomicMax
+// implement f64 min/max (maxmin.rs:104-119 compare-and-keep semantics for
+// all comparable values; see kernels.h on the NaN-sentinel edge)
+__device__ __forceinline__ uint64_t f64_omap(double x) {
+  uint64_t b;
+  memcpy(&b, &x, 8);
+  return (b >> 63) ? ~b : (b | 0x8000000000000000ull);
+}
+__device__ __forceinline__ double f64_omap_inv(uint64_t u) {
+  uint64_t b = (u >> 63) ? (u & 0x7FFFFFFFFFFFFFFFull) : ~u;
+  double x;
+  memcpy(&x, &b, 8);
+  return x;
+}
+
 // Per-row slot resolution + accumulate (the slow path of the batched kernel
 // below; also correct standalone).
 __device__ __forceinline__ void agg_accum_row(const AggTable t, int64_t key,
@@ -256,6 +271,11 @@ __device__ __forceinline__ void agg_accum_row(const AggTable t, int64_t key,
     // enforces SUM/COUNT share the argument column)
     unsafeAtomicAdd(&sl->sum, val);
     atomicAdd(&sl->cnt, 1ull);  // count.rs:90-149: COUNT(arg) non-null
+    if (t.mm) {  // maxmin.rs:104-119: MIN/MAX of non-null args
+      uint64_t u = f64_omap(val);
+      atomicMin(&t.mm[2 * a], u);
+      atomicMax(&t.mm[2 * a + 1], u);
+    }
   }
 }
 
@@ -306,6 +326,11 @@ __global__ void k_agg_update(const AggTable t, const int64_t* __restrict__ keys,
         if (vvalid[k]) {
           unsafeAtomicAdd(&sl->sum, val[k]);
           atomicAdd(&sl->cnt, 1ull);
+          if (t.mm) {
+            uint64_t u = f64_omap(val[k]);
+            atomicMin(&t.mm[2 * slot[k]], u);
+            atomicMax(&t.mm[2 * slot[k] + 1], u);
+          }
         }
       } else {
         agg_accum_row(t, knull[k] ? 0 : key[k], knull[k], val[k], vvalid[k],
@@ -347,36 +372,53 @@ __device__ __forceinline__ int varint_len_dev(uint64_t v) {
   return k;
 }
 
-// agg layout: 2 bits per agg, LSB-first, 0-terminated. All aggs share one
+// agg layout: 3 bits per agg, LSB-first, 0-terminated. All aggs share one
 // argument column, so SUM/AVG parts duplicate the same sum and COUNT/AVG
 // parts the same count (avg.rs:208-217: AVG freeze = sum ++ count).
-// AGGL_SUM=1, AGGL_CNT=2, AGGL_AVG=3.
+// AGGL_SUM=1, AGGL_CNT=2, AGGL_AVG=3, AGGL_MIN=4, AGGL_MAX=5; MIN/MAX parts
+// are prim freezes [u8 valid][8B LE f64]? like SUM's (acc.rs:335-347 — the
+// maxmin accumulator is the same generic prim column, maxmin.rs:91-93).
+// minu/maxu are order-mapped u64 accumulators (sentinel = invalid).
 __device__ __forceinline__ int agg_freeze_len(uint32_t layout, bool valid,
-                                              uint64_t cnt) {
+                                              uint64_t cnt, uint64_t minu,
+                                              uint64_t maxu) {
   int len = 0;
-  for (uint32_t l = layout; l & 3u; l >>= 2) {
-    uint32_t k = l & 3u;
+  for (uint32_t l = layout; l & 7u; l >>= 3) {
+    uint32_t k = l & 7u;
     if (k == 1 || k == 3) len += 1 + (valid ? 8 : 0);
     if (k == 2 || k == 3) len += varint_len_dev(cnt);
+    if (k == 4) len += 1 + (minu != MM_MIN_INIT ? 8 : 0);
+    if (k == 5) len += 1 + (maxu != MM_MAX_INIT ? 8 : 0);
   }
   return len;
+}
+
+__device__ __forceinline__ uint8_t* agg_prim_freeze_part(bool valid, double v,
+                                                         uint8_t* p) {
+  if (valid) {
+    *p++ = 1;
+    memcpy(p, &v, 8);
+    p += 8;
+  } else {
+    *p++ = 0;
+  }
+  return p;
 }
 
 __device__ __forceinline__ uint8_t* agg_freeze_write_rec(uint32_t layout,
                                                          bool valid, double sum,
                                                          uint64_t cnt,
+                                                         uint64_t minu,
+                                                         uint64_t maxu,
                                                          uint8_t* p) {
-  for (uint32_t l = layout; l & 3u; l >>= 2) {
-    uint32_t k = l & 3u;
-    if (k == 1 || k == 3) {  // acc.rs:335-347 prim freeze
-      if (valid) {
-        *p++ = 1;
-        memcpy(p, &sum, 8);
-        p += 8;
-      } else {
-        *p++ = 0;
-      }
-    }
+  for (uint32_t l = layout; l & 7u; l >>= 3) {
+    uint32_t k = l & 7u;
+    if (k == 1 || k == 3)  // acc.rs:335-347 prim freeze
+      p = agg_prim_freeze_part(valid, sum, p);
+    if (k == 4)
+      p = agg_prim_freeze_part(minu != MM_MIN_INIT, f64_omap_inv(minu), p);
+    if (k == 5)
+      p = agg_prim_freeze_part(maxu != MM_MAX_INIT, f64_omap_inv(maxu), p);
     if (k == 2 || k == 3) {  // count.rs:193-203 varint
       uint64_t c = cnt;
       while (c >= 128) {
@@ -389,28 +431,40 @@ __device__ __forceinline__ uint8_t* agg_freeze_write_rec(uint32_t layout,
   return p;
 }
 
-// parse one frozen record; accumulates only the FIRST sum part and FIRST
-// count part (the rest are duplicates of the same shared-column accs)
+// parse one frozen record; accumulates only the FIRST part of each family
+// (the rest are duplicates of the same shared-column accs). minu/maxu come
+// back order-mapped, at their sentinels when the part is absent or invalid —
+// so merging them with atomicMin/atomicMax needs no guard.
 __device__ __forceinline__ void agg_parse_frozen(uint32_t layout,
                                                  const uint8_t* p, bool* valid,
-                                                 double* sum, uint64_t* cnt) {
+                                                 double* sum, uint64_t* cnt,
+                                                 uint64_t* minu,
+                                                 uint64_t* maxu) {
   *valid = false;
   *sum = 0.0;
   *cnt = 0;
-  bool got_sum = false, got_cnt = false;
-  for (uint32_t l = layout; l & 3u; l >>= 2) {
-    uint32_t k = l & 3u;
-    if (k == 1 || k == 3) {
+  *minu = MM_MIN_INIT;
+  *maxu = MM_MAX_INIT;
+  bool got_sum = false, got_cnt = false, got_min = false, got_max = false;
+  for (uint32_t l = layout; l & 7u; l >>= 3) {
+    uint32_t k = l & 7u;
+    if (k == 1 || k == 3 || k == 4 || k == 5) {
       uint8_t v = *p++;
       double x = 0;
       if (v) {
         memcpy(&x, p, 8);
         p += 8;
       }
-      if (!got_sum) {
+      if ((k == 1 || k == 3) && !got_sum) {
         *valid = v != 0;
         *sum = x;
         got_sum = true;
+      } else if (k == 4 && !got_min) {
+        if (v) *minu = f64_omap(x);
+        got_min = true;
+      } else if (k == 5 && !got_max) {
+        if (v) *maxu = f64_omap(x);
+        got_max = true;
       }
     }
     if (k == 2 || k == 3) {
@@ -442,10 +496,15 @@ __global__ void k_agg_merge_frozen(const AggTable t,
     if (sl->first_row > row) atomicMin(&sl->first_row, row);
     bool valid;
     double v;
-    uint64_t c;
-    agg_parse_frozen(layout, acc_data + acc_offsets[i], &valid, &v, &c);
+    uint64_t c, mnu, mxu;
+    agg_parse_frozen(layout, acc_data + acc_offsets[i], &valid, &v, &c, &mnu,
+                     &mxu);
     if (valid) unsafeAtomicAdd(&sl->sum, v);  // sum.rs:117-145 partial_merge
     if (c) atomicAdd(&sl->cnt, c);
+    if (t.mm) {  // maxmin.rs:196-216 partial_merge; sentinels are no-ops
+      atomicMin(&t.mm[2 * a], mnu);
+      atomicMax(&t.mm[2 * a + 1], mxu);
+    }
   }
 }
 
@@ -469,10 +528,14 @@ __global__ void k_agg_merge_spill(const AggTable t,
     if (sl->first_row > fr) atomicMin(&sl->first_row, fr);
     bool valid;
     double v;
-    uint64_t c;
-    agg_parse_frozen(layout, acc_data + off, &valid, &v, &c);
+    uint64_t c, mnu, mxu;
+    agg_parse_frozen(layout, acc_data + off, &valid, &v, &c, &mnu, &mxu);
     if (valid) unsafeAtomicAdd(&sl->sum, v);
     if (c) atomicAdd(&sl->cnt, c);
+    if (t.mm) {
+      atomicMin(&t.mm[2 * a], mnu);
+      atomicMax(&t.mm[2 * a + 1], mxu);
+    }
   }
 }
 
@@ -507,7 +570,11 @@ __global__ void k_agg_gather_out(const AggTable t,
                                  uint8_t* __restrict__ key_validity,
                                  double* __restrict__ sums,
                                  uint8_t* __restrict__ sum_validity,
-                                 long long* __restrict__ counts) {
+                                 long long* __restrict__ counts,
+                                 double* __restrict__ mins,
+                                 uint8_t* __restrict__ min_validity,
+                                 double* __restrict__ maxs,
+                                 uint8_t* __restrict__ max_validity) {
   // one thread per output byte-group of 8 rows for validity bitmaps
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < num_groups;
        i += (int64_t)gridDim.x * blockDim.x) {
@@ -516,16 +583,24 @@ __global__ void k_agg_gather_out(const AggTable t,
                                     : (s == t.cap ? KEY_EMPTY : 0);
     if (sums) sums[i] = t.slots[s].sum;
     if (counts) counts[i] = (long long)t.slots[s].cnt;
+    if (mins) mins[i] = f64_omap_inv(t.mm[2 * s]);
+    if (maxs) maxs[i] = f64_omap_inv(t.mm[2 * s + 1]);
     if ((i & 7) == 0) {
-      uint8_t kb = 0, sb = 0;
+      uint8_t kb = 0, sb = 0, mb = 0, xb = 0;
       for (int j = 0; j < 8 && i + j < num_groups; j++) {
         uint32_t sj = order_slots[i + j];
         bool knull = (sj == t.cap + 1);
         if (!knull) kb |= (uint8_t)(1u << j);
         if (t.slots[sj].cnt != 0) sb |= (uint8_t)(1u << j);
+        if (t.mm) {
+          if (t.mm[2 * sj] != MM_MIN_INIT) mb |= (uint8_t)(1u << j);
+          if (t.mm[2 * sj + 1] != MM_MAX_INIT) xb |= (uint8_t)(1u << j);
+        }
       }
       if (key_validity) key_validity[i >> 3] = kb;
       if (sum_validity) sum_validity[i >> 3] = sb;
+      if (min_validity) min_validity[i >> 3] = mb;
+      if (max_validity) max_validity[i >> 3] = xb;
     }
   }
 }
@@ -537,7 +612,9 @@ __global__ void k_agg_freeze_len(const AggTable t,
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < num_groups;
        i += (int64_t)gridDim.x * blockDim.x) {
     uint32_t s = order_slots[i];
-    lens[i] = agg_freeze_len(layout, t.slots[s].cnt != 0, t.slots[s].cnt);
+    lens[i] = agg_freeze_len(layout, t.slots[s].cnt != 0, t.slots[s].cnt,
+                             t.mm ? t.mm[2 * s] : MM_MIN_INIT,
+                             t.mm ? t.mm[2 * s + 1] : MM_MAX_INIT);
   }
 }
 
@@ -550,7 +627,9 @@ __global__ void k_agg_freeze_write(const AggTable t,
        i += (int64_t)gridDim.x * blockDim.x) {
     uint32_t s = order_slots[i];
     agg_freeze_write_rec(layout, t.slots[s].cnt != 0, t.slots[s].sum,
-                         t.slots[s].cnt, out + offsets[i]);
+                         t.slots[s].cnt, t.mm ? t.mm[2 * s] : MM_MIN_INIT,
+                         t.mm ? t.mm[2 * s + 1] : MM_MAX_INIT,
+                         out + offsets[i]);
   }
 }
 
@@ -583,10 +662,13 @@ void launch_agg_gather_out(const AggTable& t, const uint32_t* order_slots,
                            int64_t num_groups, int64_t* out_keys,
                            uint8_t* out_key_validity, double* out_sums,
                            uint8_t* out_sum_validity, long long* out_counts,
+                           double* out_mins, uint8_t* out_min_validity,
+                           double* out_maxs, uint8_t* out_max_validity,
                            hipStream_t s) {
   hipLaunchKernelGGL(k_agg_gather_out, dim3(grid_for(num_groups)), dim3(BLOCK),
                      0, s, t, order_slots, num_groups, out_keys,
-                     out_key_validity, out_sums, out_sum_validity, out_counts);
+                     out_key_validity, out_sums, out_sum_validity, out_counts,
+                     out_mins, out_min_validity, out_maxs, out_max_validity);
   check_launch("k_agg_gather_out");
 }
 void launch_agg_freeze_len(const AggTable& t, const uint32_t* order_slots,
@@ -613,7 +695,10 @@ __global__ void k_skip_freeze_len(const uint8_t* __restrict__ val_valid,
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     bool v = !val_valid || bit_get_dev(val_valid, i);
-    lens[i] = agg_freeze_len(layout, v, v ? 1 : 0);
+    // a valid row's MIN/MAX parts are always present (8+1 bytes): pass any
+    // non-sentinel accumulator value when valid
+    lens[i] = agg_freeze_len(layout, v, v ? 1 : 0, v ? 0 : MM_MIN_INIT,
+                             v ? ~0ull : MM_MAX_INIT);
   }
 }
 
@@ -624,7 +709,9 @@ __global__ void k_skip_freeze_write(const double* __restrict__ vals,
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     bool v = !val_valid || bit_get_dev(val_valid, i);
+    uint64_t u = v ? f64_omap(vals[i]) : 0;  // row is its own group: min=max=v
     agg_freeze_write_rec(layout, v, v ? vals[i] : 0.0, v ? 1 : 0,
+                         v ? u : MM_MIN_INIT, v ? u : MM_MAX_INIT,
                          out + offsets[i]);
   }
 }
@@ -681,6 +768,19 @@ void launch_slots_init(AggSlot* slots, int64_t n, hipStream_t s) {
   check_launch("k_slots_init");
 }
 
+__global__ void k_mm_init(unsigned long long* __restrict__ mm, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    mm[2 * i] = MM_MIN_INIT;
+    mm[2 * i + 1] = MM_MAX_INIT;
+  }
+}
+
+void launch_mm_init(unsigned long long* mm, int64_t n, hipStream_t s) {
+  hipLaunchKernelGGL(k_mm_init, dim3(grid_for(n)), dim3(BLOCK), 0, s, mm, n);
+  check_launch("k_mm_init");
+}
+
 __global__ void k_iota_u32(uint32_t* __restrict__ dst, int64_t n) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x)
@@ -713,6 +813,10 @@ __global__ void k_agg_rebuild(const AggTable dst, const AggTable src) {
     dst.slots[a].sum = src.slots[i].sum;
     dst.slots[a].cnt = src.slots[i].cnt;
     dst.slots[a].first_row = src.slots[i].first_row;
+    if (src.mm) {
+      dst.mm[2 * a] = src.mm[2 * i];
+      dst.mm[2 * a + 1] = src.mm[2 * i + 1];
+    }
   }
 }
 

@@ -63,7 +63,9 @@ struct Expr {
   } kind = Column;
   std::string col_name;
   uint32_t col_index = 0;
-  int32_t agg_function = -1;  // AggFunction enum: MIN=0 MAX=1 SUM=2 AVG=3 COUNT=4
+  // AggFunction enum: MIN=0 MAX=1 SUM=2 AVG=3 COUNT=4. Default 0 (MIN):
+  // proto3 omits zero-valued enums on the wire, so an absent field IS MIN.
+  int32_t agg_function = 0;
   std::vector<Expr> children;  // agg args / binary l,r / null-check operand
   DType return_type = DType::Unsupported;
   std::string op;              // binary op name (auron-serde/src/lib.rs:70-96)

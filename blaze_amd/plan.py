@@ -140,6 +140,52 @@ def sum_count_aggs(val_index=1):
     ]
 
 
+_AGG_FN = {"min": AGG_MIN, "max": AGG_MAX, "sum": AGG_SUM,
+           "avg": AGG_AVG, "count": AGG_COUNT}
+
+
+def named_aggs(names, val_index=1):
+    """Agg exprs for a list of function names over the shared val column
+    (MIN/MAX return the input type — maxmin.rs:81-83 data_type())."""
+    out = []
+    for nm in names:
+        fn = _AGG_FN[nm]
+        dt = DT_INT64 if fn == AGG_COUNT else DT_FLOAT64
+        out.append(agg_expr(fn, [column("val", val_index)], dt))
+    return out
+
+
+def plan_partial_final_named(agg_fns, resource_id="input0", key_dt=DT_INT64):
+    """FFIReader -> Agg(Partial) -> Agg(Final) with an arbitrary agg list
+    (e.g. ["min", "max", "sum", "count", "avg"]) over the shared val column."""
+    reader = ffi_reader([field("key", key_dt, False),
+                         field("val", DT_FLOAT64, True)], resource_id)
+    names = list(agg_fns)
+    partial = agg(reader, [column("key", 0)], named_aggs(names),
+                  [MODE_PARTIAL] * len(names), ["key"], names)
+    final = agg(partial, [column("key", 0)], named_aggs(names),
+                [MODE_FINAL] * len(names), ["key"], names)
+    return task_definition(final)
+
+
+def plan_partial_only_named(agg_fns, resource_id="input0", skipping=False):
+    reader = ffi_reader(northstar_input_fields(), resource_id)
+    names = list(agg_fns)
+    partial = agg(reader, [column("key", 0)], named_aggs(names),
+                  [MODE_PARTIAL] * len(names), ["key"], names,
+                  supports_partial_skipping=skipping)
+    return task_definition(partial)
+
+
+def plan_final_only_named(agg_fns, resource_id="input0"):
+    names = list(agg_fns)
+    reader = ffi_reader([field("key", DT_INT64, True),
+                         field("agg_buf", DT_BINARY, False)], resource_id)
+    final = agg(reader, [column("key", 0)], named_aggs(names),
+                [MODE_FINAL] * len(names), ["key"], names)
+    return task_definition(final)
+
+
 def plan_partial_final(resource_id="input0", skipping=False, key_dt=DT_INT64):
     """FFIReader -> Agg(Partial) -> Agg(Final): config 1/2 shape
     (mirrors agg_exec.rs fuzztest:714-843 topology)."""

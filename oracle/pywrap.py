@@ -272,3 +272,56 @@ def ipc_decode(blob: bytes) -> bytes:
                                  need)
     assert got == need
     return out.tobytes()
+
+
+# ---- MIN/MAX aggregation restatement (pure python / numpy) -----------------
+# CPU oracle for the MIN/MAX agg family (maxmin.rs). The C oracle above
+# covers the north-star [SUM, COUNT] set; MIN/MAX parity is pinned by this
+# restatement of maxmin.rs:
+#  - partial_update (maxmin.rs:104-119): for each non-null arg, keep the
+#    current acc iff acc.partial_cmp(new) == Ordering(Less for MIN / Greater
+#    for MAX), else take the new value; null args are skipped.
+#  - partial_merge (maxmin.rs:196-216): same compare over non-null merged accs.
+#  - freeze: the generic prim column save [u8 valid][8B LE f64]? — the same
+#    AccPrimColumn format SUM uses (acc.rs:335-347; maxmin.rs:91-93
+#    create_acc_column returns the same generic column as sum.rs).
+# Record order and key handling follow the C oracle (first-occurrence,
+# special null/i64::MIN groups last-inserted-by-arrival like OracleAgg).
+# NaN note: the reference's NaN outcome is arrival-order dependent
+# (partial_cmp == None always takes the new value); this restatement and the
+# GPU engine use the total-order map instead — parity tests use NaN-free data.
+def minmax_groups(keys, vals, val_valid=None, key_valid=None):
+    """Insertion-ordered dict key->(mn, mx, any_valid); None key = null-key
+    group. Returns (ordered_keys, mins, maxs, valid_mask) with mins/maxs None
+    entries for all-null groups."""
+    import math
+    keys = list(keys)
+    n = len(keys)
+    groups = {}
+    for i in range(n):
+        k = None if (key_valid is not None and not key_valid[i]) else int(keys[i])
+        if k not in groups:
+            groups[k] = [None, None]
+        vv = val_valid is None or bool(val_valid[i])
+        if vv:
+            v = float(vals[i])
+            g = groups[k]
+            g[0] = v if g[0] is None or v < g[0] else g[0]
+            g[1] = v if g[1] is None or v > g[1] else g[1]
+    ordered = list(groups.keys())
+    mins = [groups[k][0] for k in ordered]
+    maxs = [groups[k][1] for k in ordered]
+    return ordered, mins, maxs
+
+
+def minmax_freeze_rec(parts):
+    """Freeze one record: parts is a list of ('prim', value_or_None) /
+    ('cnt', int) in layout order; returns the a8 wire bytes for the record."""
+    import struct
+    out = b""
+    for kind, v in parts:
+        if kind == "prim":
+            out += b"\x00" if v is None else b"\x01" + struct.pack("<d", v)
+        else:
+            out += write_len(int(v))
+    return out

@@ -95,3 +95,9 @@ def test_conf_callback_roundtrip():
     assert rc == 7 and out.value == b"1048576"
     lib.auron_debug_conf_roundtrip(ctypes.byref(cb), b"MISSING", out, 256)
     assert out.value == b"<default>"
+
+
+def test_decode_minmax_plan():
+    td = plan.plan_partial_final_named(["min", "max", "sum", "count"])
+    s = blaze_amd.debug_decode_plan(td)
+    assert "fn0" in s and "fn1" in s  # MIN=0, MAX=1 (auron.proto AggFunction)

@@ -83,3 +83,29 @@ def test_null_rules_match_fuzztest():
     assert g["sum_a"] is None
     assert g["cnt_f"] == 0
     assert g["firstign_h"] is None
+
+
+def test_minmax_restatement_hand_case():
+    """oracle.pywrap.minmax_groups vs a hand-computed maxmin.rs trace:
+    non-null values only participate (maxmin.rs:106-108 is_valid guard);
+    first-occurrence group order; all-null group -> None."""
+    from oracle import pywrap as oracle
+    keys = [5, 3, 5, 3, 9, 5]
+    vals = [2.5, -1.0, 7.0, 4.0, 0.0, -3.5]
+    vv = [True, True, True, False, False, True]
+    ok, mins, maxs = oracle.minmax_groups(keys, vals, vv)
+    assert ok == [5, 3, 9]
+    assert mins == [-3.5, -1.0, None]
+    assert maxs == [7.0, -1.0, None]
+
+
+def test_minmax_freeze_rec_format():
+    """MIN/MAX freeze parts are the SUM-style prim format [u8 valid][8B LE]
+    (acc.rs:335-347 generic prim save; maxmin.rs:91-93)."""
+    import struct
+    from oracle import pywrap as oracle
+    rec = oracle.minmax_freeze_rec([("prim", 1.5), ("prim", None),
+                                    ("cnt", 300)])
+    assert rec[:9] == b"\x01" + struct.pack("<d", 1.5)
+    assert rec[9:10] == b"\x00"
+    assert rec[10:] == oracle.write_len(300)

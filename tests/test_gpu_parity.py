@@ -425,3 +425,131 @@ def test_round_robin_shuffle_write(tmp_path):
                                dtype=np.uint8).reshape(8, rows)
         got_keys = np.ascontiguousarray(planes.T).reshape(-1).view(np.int64)
         np.testing.assert_array_equal(got_keys, keys[exp_rows])
+
+
+def _minmax_cols(outs, names):
+    """Concat output cols: returns {name: (values, valid)} keyed by agg name
+    (col 0 is the key)."""
+    res = {"key": (np.concatenate([ob[0]["values"] for ob in outs]), None)}
+    for j, nm in enumerate(names):
+        vals = np.concatenate([ob[1 + j]["values"] for ob in outs])
+        vv = np.concatenate(
+            [ob[1 + j]["valid"] if ob[1 + j]["valid"] is not None
+             else np.ones(len(ob[1 + j]["values"]), bool) for ob in outs])
+        res[nm] = (vals, vv)
+    return res
+
+
+def test_minmax_agg_family():
+    """MIN/MAX aggregates (maxmin.rs:104-119 partial_update, :196-216
+    partial_merge) in a mixed list with SUM/COUNT/AVG, partial->final.
+    MIN/MAX select actual input values, so parity is bit-exact even for
+    fractional f64."""
+    keys, vals, vv = gen_northstar(300_000, nkeys=2000, fractional=True)
+    names = ["min", "max", "sum", "count", "avg"]
+    t = blaze_amd.Task(plan.plan_partial_final_named(names),
+                       batches=batches_of(keys, vals, vv))
+    outs = t.run()
+    got = _minmax_cols(outs, names)
+    ref = run_oracle(keys, vals, vv).output()
+    ok, mins, maxs = oracle.minmax_groups(keys, vals, vv)
+    np.testing.assert_array_equal(got["key"][0], ref["keys"])
+    np.testing.assert_array_equal(np.array(ok), ref["keys"])
+    exp_valid = np.array([m is not None for m in mins])
+    np.testing.assert_array_equal(got["min"][1], exp_valid)
+    np.testing.assert_array_equal(got["max"][1], exp_valid)
+    np.testing.assert_array_equal(
+        got["min"][0][exp_valid],
+        np.array([m for m in mins if m is not None]))
+    np.testing.assert_array_equal(
+        got["max"][0][exp_valid],
+        np.array([m for m in maxs if m is not None]))
+    np.testing.assert_array_equal(got["count"][0], ref["counts"])
+    t.finalize()
+
+
+def test_minmax_only_with_all_null_groups():
+    """[MIN] alone: the frozen record carries only the min part (no count on
+    the wire) and all-null groups emit null (maxmin.rs nullable:85-87)."""
+    rng = np.random.default_rng(7)
+    n = 2000
+    keys = rng.integers(0, 500, n).astype(np.int64)
+    vals = rng.random(n) * 1000 - 500  # negative values exercise the omap
+    vv = rng.random(n) >= 0.5
+    names = ["min"]
+    t = blaze_amd.Task(plan.plan_partial_final_named(names),
+                       batches=batches_of(keys, vals, vv, batch=100))
+    outs = t.run()
+    got = _minmax_cols(outs, names)
+    ok, mins, maxs = oracle.minmax_groups(keys, vals, vv)
+    np.testing.assert_array_equal(got["key"][0], np.array(ok))
+    exp_valid = np.array([m is not None for m in mins])
+    assert (~exp_valid).sum() > 0, "test data must produce all-null groups"
+    np.testing.assert_array_equal(got["min"][1], exp_valid)
+    np.testing.assert_array_equal(
+        got["min"][0][exp_valid],
+        np.array([m for m in mins if m is not None]))
+    t.finalize()
+
+
+def test_minmax_spill_and_grow():
+    """MIN/MAX accumulators survive table grow (mm pairs carried by the
+    rebuild) and host spill (mm parts ride the frozen a8 records)."""
+    keys, vals, vv = gen_northstar(400_000, nkeys=120_000, fractional=True)
+    names = ["min", "max", "count"]
+    conf = {"AURON_HIP_MEM_BUDGET": 1 << 20,
+            "AURON_HIP_AGG_TABLE_SLOTS": 1 << 10}
+    t = blaze_amd.Task(plan.plan_partial_final_named(names),
+                       batches=batches_of(keys, vals, vv), conf=conf)
+    outs = t.run()
+    assert t.metric("spill_count") > 0
+    got = _minmax_cols(outs, names)
+    ok, mins, maxs = oracle.minmax_groups(keys, vals, vv)
+    k = got["key"][0]
+    assert len(k) == len(ok)
+    gi = np.argsort(k, kind="stable")
+    oi = np.argsort(np.array(ok), kind="stable")
+    np.testing.assert_array_equal(k[gi], np.array(ok)[oi])
+    exp_min = np.array([np.nan if m is None else m for m in mins])
+    exp_max = np.array([np.nan if m is None else m for m in maxs])
+    exp_valid = ~np.isnan(exp_min)
+    np.testing.assert_array_equal(got["min"][1][gi], exp_valid[oi])
+    np.testing.assert_array_equal(got["min"][0][gi][exp_valid[oi]],
+                                  exp_min[oi][exp_valid[oi]])
+    np.testing.assert_array_equal(got["max"][0][gi][exp_valid[oi]],
+                                  exp_max[oi][exp_valid[oi]])
+    t.finalize()
+
+
+def test_minmax_partial_skipping():
+    """Partial-skipping pass-through freezes each row as its own record with
+    min=max=value (agg_ctx.rs:428-462 analog); the final stage must still
+    produce the exact min/max."""
+    n = 60_000
+    rng = np.random.default_rng(11)
+    keys = np.arange(n, dtype=np.int64) % (n // 2)  # 2 rows/group
+    vals = rng.random(n) * 100
+    vv = np.ones(n, bool)
+
+    reader = plan.ffi_reader(plan.northstar_input_fields(), "input0")
+    names = ["min", "max"]
+    partial = plan.agg(reader, [plan.column("key", 0)],
+                       plan.named_aggs(names),
+                       [plan.MODE_PARTIAL] * 2, ["key"], names,
+                       supports_partial_skipping=True)
+    final = plan.agg(partial, [plan.column("key", 0)],
+                     plan.named_aggs(names),
+                     [plan.MODE_FINAL] * 2, ["key"], names)
+    t = blaze_amd.Task(plan.task_definition(final),
+                       batches=batches_of(keys, vals, vv))
+    outs = t.run()
+    got = _minmax_cols(outs, names)
+    ok, mins, maxs = oracle.minmax_groups(keys, vals, vv)
+    k = got["key"][0]
+    assert len(k) == len(ok)
+    gi = np.argsort(k, kind="stable")
+    oi = np.argsort(np.array(ok), kind="stable")
+    np.testing.assert_array_equal(k[gi], np.array(ok)[oi])
+    np.testing.assert_array_equal(got["min"][0][gi], np.array(mins)[oi])
+    np.testing.assert_array_equal(got["max"][0][gi], np.array(maxs)[oi])
+    t.finalize()
