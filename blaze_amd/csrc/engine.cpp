@@ -379,8 +379,11 @@ class AggOp {
         case AGG_AVG: k = AGGL_AVG; break;
         case AGG_MIN: k = AGGL_MIN; has_mm_ = true; break;
         case AGG_MAX: k = AGGL_MAX; has_mm_ = true; break;
+        case AGG_FIRST: k = AGGL_FIRST; has_first_ = true; break;
+        case AGG_FIRST_IGNORES_NULL: k = AGGL_FIRSTIN; has_first_ = true; break;
         default:
-          FAIL("AggExec: only SUM/COUNT/AVG/MIN/MAX aggregates on this path");
+          FAIL("AggExec: only SUM/COUNT/AVG/MIN/MAX/FIRST[_IGNORES_NULL] "
+               "aggregates on this path");
       }
       layout_ |= k << (3 * i);
       agg_kinds_.push_back(k);
@@ -467,7 +470,8 @@ class AggOp {
       // chunks on partition-buffer size instead of table free slots.
       // MIN/MAX agg sets stay single-phase: the LDS bucket kernel's slot
       // holds {key,cnt,sum,first} only (perf note in DESIGN.md)
-      if (!merge_mode_ && !has_mm_ && b.num_rows - done >= AGG2_MIN_CHUNK) {
+      if (!merge_mode_ && !has_mm_ && !has_first_ &&
+          b.num_rows - done >= AGG2_MIN_CHUNK) {
         int64_t chunk2 = std::min(b.num_rows - done, AGG2_MAX_CHUNK);
         if (done + chunk2 < b.num_rows) chunk2 &= ~(int64_t)7;
         two_phase_chunk(b, done, chunk2);
@@ -497,6 +501,12 @@ class AggOp {
                                 slice_valid(key.validity, done),
                                 (const uint8_t*)buf.values, buf.offsets + done,
                                 chunk, row_cursor_, layout_, stream_);
+        if (has_first_)  // pass B: capture the winning records' values
+          launch_first_capture_frozen(t_, (const int64_t*)key.values + done,
+                                      slice_valid(key.validity, done),
+                                      (const uint8_t*)buf.values,
+                                      buf.offsets + done, nullptr, chunk,
+                                      row_cursor_, layout_, stream_);
       } else {
         const DevColumn& val = b.cols.at(val_col_);
         if (val.dt != DType::Float64) FAIL("SUM arg must be Float64");
@@ -512,6 +522,12 @@ class AggOp {
                           (const double*)val.values + done,
                           slice_valid(val.validity, done), chunk, row_cursor_,
                           stream_);
+        if (has_first_)  // pass B: same stream, so it runs after pass A
+          launch_first_capture_update(t_, (const int64_t*)key.values + done,
+                                      slice_valid(key.validity, done),
+                                      (const double*)val.values + done,
+                                      slice_valid(val.validity, done), chunk,
+                                      row_cursor_, stream_);
         AURON_HIP(hipEventRecord(e1, stream_));
         ev_pairs_.push_back({e0, e1});
         update_rows_ += chunk;
@@ -616,6 +632,12 @@ class AggOp {
                                    d_offs.get<int32_t>() + done2,
                                    d_first.get<unsigned long long>() + done2,
                                    piece, layout_, stream_);
+            if (has_first_)  // pass B with the records' preserved priorities
+              launch_first_capture_frozen(
+                  t_, d_keys.get<int64_t>() + done2, nullptr,
+                  d_data.get<uint8_t>(), d_offs.get<int32_t>() + done2,
+                  d_first.get<unsigned long long>() + done2, piece, 0,
+                  layout_, stream_);
             done2 += piece;
             ng_bound_ += (uint64_t)piece;
           }
@@ -672,6 +694,15 @@ class AggOp {
       d_mm_.alloc((cap + 2) * 16);
       t_.mm = d_mm_.get<unsigned long long>();
       launch_mm_init(t_.mm, cap + 2, stream_);
+    }
+    if (has_first_) {
+      d_frow_.alloc((cap + 2) * 16);
+      d_fval_.alloc((cap + 2) * 16);
+      d_fst_.alloc((cap + 2) * 2);
+      t_.f_row = d_frow_.get<unsigned long long>();
+      t_.f_val = d_fval_.get<double>();
+      t_.f_st = d_fst_.get<uint8_t>();
+      launch_first_init(t_.f_row, t_.f_val, t_.f_st, cap + 2, stream_);
     }
   }
 
@@ -916,6 +947,8 @@ class AggOp {
   void reset_main() {
     launch_slots_init(t_.slots, t_.cap, stream_);  // specials untouched
     if (has_mm_) launch_mm_init(t_.mm, t_.cap, stream_);
+    if (has_first_) launch_first_init(t_.f_row, t_.f_val, t_.f_st, t_.cap,
+                                      stream_);
     uint64_t ng0 = (uint64_t)specials_count_;
     AURON_HIP(hipMemcpyAsync(t_.num_groups, &ng0, 8, hipMemcpyHostToDevice,
                              stream_));
@@ -953,7 +986,8 @@ class AggOp {
     AggTable old = t_;
     DevBuf oslots = std::move(d_slots_), os = std::move(d_special_),
            ong = std::move(d_ng_), oerr = std::move(d_err_),
-           omm = std::move(d_mm_);
+           omm = std::move(d_mm_), ofrow = std::move(d_frow_),
+           ofval = std::move(d_fval_), ofst = std::move(d_fst_);
     init_table(new_cap);
     launch_agg_rebuild(t_, old, stream_);
     AURON_HIP(hipStreamSynchronize(stream_));
@@ -1056,6 +1090,20 @@ class AggOp {
         AURON_HIP(hipMemcpyAsync(xv.data(), xvalid.get(), bm,
                                  hipMemcpyDeviceToHost, stream_));
       }
+      std::vector<uint8_t> h_firsts[2];
+      std::vector<uint8_t> fv[2] = {std::vector<uint8_t>(bm),
+                                    std::vector<uint8_t>(bm)};
+      if (has_first_) {
+        DevBuf fvals(n * 8), fvalid(bm);
+        for (int w = 0; w < 2; w++) {
+          launch_first_gather(t_, order_slots, n, w, fvals.get<double>(),
+                              fvalid.get<uint8_t>(), stream_);
+          d2h_pinned(fvals.get(), &h_firsts[w], n * 8);
+          AURON_HIP(hipMemcpyAsync(fv[w].data(), fvalid.get(), bm,
+                                   hipMemcpyDeviceToHost, stream_));
+          AURON_HIP(hipStreamSynchronize(stream_));
+        }
+      }
       AURON_HIP(hipStreamSynchronize(stream_));
       DBG("agg.emit final d2h done");
       attach_validity(&key_col, kv, n);
@@ -1069,6 +1117,11 @@ class AggOp {
           ac.dt = DType::Float64;
           ac.values = (k == AGGL_MIN) ? h_mins : h_maxs;
           attach_validity(&ac, (k == AGGL_MIN) ? mv : xv, n);
+        } else if (k == AGGL_FIRST || k == AGGL_FIRSTIN) {
+          int w = (k == AGGL_FIRSTIN) ? 1 : 0;
+          ac.dt = DType::Float64;
+          ac.values = h_firsts[w];
+          attach_validity(&ac, fv[w], n);
         } else {
           ac.dt = DType::Float64;
           ac.values = (k == AGGL_AVG) ? h_avgs : h_sums;
@@ -1169,6 +1222,7 @@ class AggOp {
   DType key_dt_ = DType::Unsupported;
   uint32_t layout_ = 0;
   bool has_mm_ = false;  // agg list contains MIN/MAX: side mm array active
+  bool has_first_ = false;  // FIRST family: f_row/f_val/f_st arrays active
   std::vector<uint32_t> agg_kinds_;
   std::vector<std::string> agg_names_;
   std::string key_name_;
@@ -1186,7 +1240,7 @@ class AggOp {
   std::vector<std::pair<hipEvent_t, hipEvent_t>> ev_pairs_;
   hipEvent_t ev_start_ = nullptr, ev_stop_ = nullptr;
   AggTable t_;
-  DevBuf d_slots_, d_special_, d_ng_, d_err_, d_mm_;
+  DevBuf d_slots_, d_special_, d_ng_, d_err_, d_mm_, d_frow_, d_fval_, d_fst_;
   PinnedBuf pinned_meta_, pinned_emit_;
   // two-phase scratch (allocated on first large chunk)
   DevBuf d_part_, d_leftover_, d_counts_, d_scanned_, d_scan_tmp_, d_offsets_,

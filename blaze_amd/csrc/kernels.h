@@ -70,6 +70,20 @@ struct AggTable {
   // interleaved {min_u64, max_u64} per slot, [2*(cap+2)]; null when the agg
   // list has no MIN/MAX (the north-star path) — kernels skip it then
   unsigned long long* mm = nullptr;
+  // FIRST / FIRST_IGNORES_NULL accumulators (first.rs / first_ignores_null.rs)
+  // — optional side arrays, [2*(cap+2)] each, even index = FIRST, odd =
+  // FIRST_IGNORES_NULL. A parallel reduction can't latch "first arrival"
+  // directly, so it runs in TWO passes per chunk on the same stream:
+  // pass A (inside update/merge kernels) takes atomicMin of each candidate
+  // row's priority into f_row; pass B (k_first_capture_*) re-probes each row
+  // and the unique row whose priority equals f_row[..] plain-stores the
+  // value and state. Priorities are global arrival rows (update/merge) or
+  // the record's preserved first_row (spill merge), matching the reference's
+  // sequential update/merge order.
+  unsigned long long* f_row = nullptr;  // candidate priority, init ~0ull
+  double* f_val = nullptr;              // captured value
+  uint8_t* f_st = nullptr;              // 0 = untouched, 1 = first-was-null
+                                        // (FIRST only), 2 = valid
 };
 
 // grid-stride row update: keys/vals length n, rows globally numbered starting
@@ -86,7 +100,8 @@ void launch_agg_update(const AggTable& t, const int64_t* keys,
 // create_acc_column:91-93 uses the same generic prim column as sum.rs), so
 // duplicated parts repeat the shared accumulators.
 enum AggLayoutKind : uint32_t {
-  AGGL_SUM = 1, AGGL_CNT = 2, AGGL_AVG = 3, AGGL_MIN = 4, AGGL_MAX = 5
+  AGGL_SUM = 1, AGGL_CNT = 2, AGGL_AVG = 3, AGGL_MIN = 4, AGGL_MAX = 5,
+  AGGL_FIRST = 6, AGGL_FIRSTIN = 7
 };
 
 // merge rows of frozen partial state per the layout
@@ -146,6 +161,33 @@ void launch_slots_init(AggSlot* slots, int64_t n, hipStream_t s);
 
 // initialize n interleaved {min,max} accumulator pairs to their sentinels
 void launch_mm_init(unsigned long long* mm, int64_t n, hipStream_t s);
+
+// initialize n {f_row, f_val, f_st} pair-slots (2 entries each)
+void launch_first_init(unsigned long long* f_row, double* f_val, uint8_t* f_st,
+                       int64_t n, hipStream_t s);
+
+// FIRST/FIRST_IGNORES_NULL pass B over update-mode rows: the unique row
+// whose global row index equals the slot's captured f_row priority stores
+// its value/state (first.rs:91-148 partial_update latch semantics)
+void launch_first_capture_update(const AggTable& t, const int64_t* keys,
+                                 const uint8_t* key_valid, const double* vals,
+                                 const uint8_t* val_valid, int64_t n,
+                                 uint64_t row_offset, hipStream_t s);
+
+// pass B over frozen records (merge mode / spill merge): prio = explicit
+// per-record priorities (spill), or null for row_offset + i (merge)
+void launch_first_capture_frozen(const AggTable& t, const int64_t* keys,
+                                 const uint8_t* key_valid,
+                                 const uint8_t* acc_data,
+                                 const int32_t* acc_offsets,
+                                 const unsigned long long* prio, int64_t n,
+                                 uint64_t row_offset, uint32_t layout,
+                                 hipStream_t s);
+
+// gather one FIRST-family output column (which: 0 = FIRST, 1 = IGNORES_NULL)
+void launch_first_gather(const AggTable& t, const uint32_t* order_slots,
+                         int64_t num_groups, int which, double* out_vals,
+                         uint8_t* out_validity, hipStream_t s);
 
 // rebuild `src` table into the (larger, initialized) `dst` table; each source
 // slot holds a distinct key so plain stores after the CAS claim are race-free.

@@ -277,6 +277,10 @@ __device__ __forceinline__ void agg_accum_row(const AggTable t, int64_t key,
       atomicMax(&t.mm[2 * a + 1], u);
     }
   }
+  if (t.f_row) {  // pass A: FIRST = any row, FIRST_IGNORES_NULL = valid rows
+    atomicMin(&t.f_row[2 * a], (unsigned long long)row);
+    if (vvalid) atomicMin(&t.f_row[2 * a + 1], (unsigned long long)row);
+  }
 }
 
 // The dominant kernel. PMC evidence (profiles/): one-row-at-a-time leaves
@@ -332,6 +336,11 @@ __global__ void k_agg_update(const AggTable t, const int64_t* __restrict__ keys,
             atomicMax(&t.mm[2 * slot[k] + 1], u);
           }
         }
+        if (t.f_row) {
+          atomicMin(&t.f_row[2 * slot[k]], (unsigned long long)row);
+          if (vvalid[k])
+            atomicMin(&t.f_row[2 * slot[k] + 1], (unsigned long long)row);
+        }
       } else {
         agg_accum_row(t, knull[k] ? 0 : key[k], knull[k], val[k], vvalid[k],
                       row);
@@ -379,16 +388,33 @@ __device__ __forceinline__ int varint_len_dev(uint64_t v) {
 // are prim freezes [u8 valid][8B LE f64]? like SUM's (acc.rs:335-347 — the
 // maxmin accumulator is the same generic prim column, maxmin.rs:91-93).
 // minu/maxu are order-mapped u64 accumulators (sentinel = invalid).
-__device__ __forceinline__ int agg_freeze_len(uint32_t layout, bool valid,
-                                              uint64_t cnt, uint64_t minu,
-                                              uint64_t maxu) {
+// per-record accumulator snapshot for freeze/parse. FIRST freeze = prim
+// value part ++ flag byte 0|2 (first.rs:315-319 freeze_to_rows = values ++
+// flags; acc.rs:156-166 AccBooleanColumn byte 0 = None / 1+v = Some(v), and
+// the FIRST flag is only ever Some(true) = 2). FIRST_IGNORES_NULL freeze =
+// prim value part only (first_ignores_null.rs:79-81 generic prim column).
+struct AccSnap {
+  bool valid = false;    // sum part valid
+  double sum = 0.0;
+  uint64_t cnt = 0;
+  uint64_t minu = MM_MIN_INIT, maxu = MM_MAX_INIT;
+  uint8_t f_st = 0;      // FIRST state: 0 untouched / 1 first-null / 2 valid
+  double f_val = 0.0;
+  uint8_t fn_st = 0;     // FIRST_IGNORES_NULL state: 0 untouched / 2 valid
+  double fn_val = 0.0;
+};
+
+__device__ __forceinline__ int agg_freeze_len(uint32_t layout,
+                                              const AccSnap& a) {
   int len = 0;
   for (uint32_t l = layout; l & 7u; l >>= 3) {
     uint32_t k = l & 7u;
-    if (k == 1 || k == 3) len += 1 + (valid ? 8 : 0);
-    if (k == 2 || k == 3) len += varint_len_dev(cnt);
-    if (k == 4) len += 1 + (minu != MM_MIN_INIT ? 8 : 0);
-    if (k == 5) len += 1 + (maxu != MM_MAX_INIT ? 8 : 0);
+    if (k == 1 || k == 3) len += 1 + (a.valid ? 8 : 0);
+    if (k == 2 || k == 3) len += varint_len_dev(a.cnt);
+    if (k == 4) len += 1 + (a.minu != MM_MIN_INIT ? 8 : 0);
+    if (k == 5) len += 1 + (a.maxu != MM_MAX_INIT ? 8 : 0);
+    if (k == 6) len += 2 + (a.f_st == 2 ? 8 : 0);  // prim part + flag byte
+    if (k == 7) len += 1 + (a.fn_st == 2 ? 8 : 0);
   }
   return len;
 }
@@ -406,19 +432,25 @@ __device__ __forceinline__ uint8_t* agg_prim_freeze_part(bool valid, double v,
 }
 
 __device__ __forceinline__ uint8_t* agg_freeze_write_rec(uint32_t layout,
-                                                         bool valid, double sum,
-                                                         uint64_t cnt,
-                                                         uint64_t minu,
-                                                         uint64_t maxu,
+                                                         const AccSnap& a,
                                                          uint8_t* p) {
+  const bool valid = a.valid;
+  const double sum = a.sum;
+  const uint64_t cnt = a.cnt;
   for (uint32_t l = layout; l & 7u; l >>= 3) {
     uint32_t k = l & 7u;
     if (k == 1 || k == 3)  // acc.rs:335-347 prim freeze
       p = agg_prim_freeze_part(valid, sum, p);
     if (k == 4)
-      p = agg_prim_freeze_part(minu != MM_MIN_INIT, f64_omap_inv(minu), p);
+      p = agg_prim_freeze_part(a.minu != MM_MIN_INIT, f64_omap_inv(a.minu), p);
     if (k == 5)
-      p = agg_prim_freeze_part(maxu != MM_MAX_INIT, f64_omap_inv(maxu), p);
+      p = agg_prim_freeze_part(a.maxu != MM_MAX_INIT, f64_omap_inv(a.maxu), p);
+    if (k == 6) {  // FIRST: prim value ++ flag byte (first.rs:315-319)
+      p = agg_prim_freeze_part(a.f_st == 2, a.f_val, p);
+      *p++ = a.f_st ? 2 : 0;
+    }
+    if (k == 7)  // FIRST_IGNORES_NULL: prim value only
+      p = agg_prim_freeze_part(a.fn_st == 2, a.fn_val, p);
     if (k == 2 || k == 3) {  // count.rs:193-203 varint
       uint64_t c = cnt;
       while (c >= 128) {
@@ -436,44 +468,49 @@ __device__ __forceinline__ uint8_t* agg_freeze_write_rec(uint32_t layout,
 // back order-mapped, at their sentinels when the part is absent or invalid —
 // so merging them with atomicMin/atomicMax needs no guard.
 __device__ __forceinline__ void agg_parse_frozen(uint32_t layout,
-                                                 const uint8_t* p, bool* valid,
-                                                 double* sum, uint64_t* cnt,
-                                                 uint64_t* minu,
-                                                 uint64_t* maxu) {
-  *valid = false;
-  *sum = 0.0;
-  *cnt = 0;
-  *minu = MM_MIN_INIT;
-  *maxu = MM_MAX_INIT;
-  bool got_sum = false, got_cnt = false, got_min = false, got_max = false;
+                                                 const uint8_t* p, AccSnap* a) {
+  *a = AccSnap{};
+  uint32_t got = 0;  // bit per family
   for (uint32_t l = layout; l & 7u; l >>= 3) {
     uint32_t k = l & 7u;
-    if (k == 1 || k == 3 || k == 4 || k == 5) {
+    if (k != 2) {  // every non-pure-COUNT part starts with a prim part
       uint8_t v = *p++;
       double x = 0;
       if (v) {
         memcpy(&x, p, 8);
         p += 8;
       }
-      if ((k == 1 || k == 3) && !got_sum) {
-        *valid = v != 0;
-        *sum = x;
-        got_sum = true;
-      } else if (k == 4 && !got_min) {
-        if (v) *minu = f64_omap(x);
-        got_min = true;
-      } else if (k == 5 && !got_max) {
-        if (v) *maxu = f64_omap(x);
-        got_max = true;
+      if ((k == 1 || k == 3) && !(got & 1u)) {
+        a->valid = v != 0;
+        a->sum = x;
+        got |= 1u;
+      } else if (k == 4 && !(got & 2u)) {
+        if (v) a->minu = f64_omap(x);
+        got |= 2u;
+      } else if (k == 5 && !(got & 4u)) {
+        if (v) a->maxu = f64_omap(x);
+        got |= 4u;
+      } else if (k == 6 && !(got & 8u)) {
+        a->f_val = x;
+        a->f_st = v ? 2 : 0;  // refined by the flag byte below
+        got |= 8u;
+      } else if (k == 7 && !(got & 16u)) {
+        a->fn_val = x;
+        a->fn_st = v ? 2 : 0;
+        got |= 16u;
       }
+    }
+    if (k == 6) {  // FIRST flag byte: 0 = untouched, 2 = touched
+      uint8_t fl = *p++;
+      if (a->f_st == 0 && fl) a->f_st = 1;  // touched but first value null
     }
     if (k == 2 || k == 3) {
       int used;
       uint64_t c = read_varint_dev(p, &used);
       p += used;
-      if (!got_cnt) {
-        *cnt = c;
-        got_cnt = true;
+      if (!(got & 32u)) {
+        a->cnt = c;
+        got |= 32u;
       }
     }
   }
@@ -494,16 +531,17 @@ __global__ void k_agg_merge_frozen(const AggTable t,
     AggSlot* sl = &t.slots[a];
     uint64_t row = row_offset + (uint64_t)i;
     if (sl->first_row > row) atomicMin(&sl->first_row, row);
-    bool valid;
-    double v;
-    uint64_t c, mnu, mxu;
-    agg_parse_frozen(layout, acc_data + acc_offsets[i], &valid, &v, &c, &mnu,
-                     &mxu);
-    if (valid) unsafeAtomicAdd(&sl->sum, v);  // sum.rs:117-145 partial_merge
-    if (c) atomicAdd(&sl->cnt, c);
+    AccSnap acc;
+    agg_parse_frozen(layout, acc_data + acc_offsets[i], &acc);
+    if (acc.valid) unsafeAtomicAdd(&sl->sum, acc.sum);  // sum.rs:117-145
+    if (acc.cnt) atomicAdd(&sl->cnt, acc.cnt);
     if (t.mm) {  // maxmin.rs:196-216 partial_merge; sentinels are no-ops
-      atomicMin(&t.mm[2 * a], mnu);
-      atomicMax(&t.mm[2 * a + 1], mxu);
+      atomicMin(&t.mm[2 * a], acc.minu);
+      atomicMax(&t.mm[2 * a + 1], acc.maxu);
+    }
+    if (t.f_row) {  // pass A: earliest TOUCHED record wins (first.rs:198-207)
+      if (acc.f_st) atomicMin(&t.f_row[2 * a], (unsigned long long)row);
+      if (acc.fn_st) atomicMin(&t.f_row[2 * a + 1], (unsigned long long)row);
     }
   }
 }
@@ -526,15 +564,17 @@ __global__ void k_agg_merge_spill(const AggTable t,
     AggSlot* sl = &t.slots[a];
     unsigned long long fr = first_rows[i];
     if (sl->first_row > fr) atomicMin(&sl->first_row, fr);
-    bool valid;
-    double v;
-    uint64_t c, mnu, mxu;
-    agg_parse_frozen(layout, acc_data + off, &valid, &v, &c, &mnu, &mxu);
-    if (valid) unsafeAtomicAdd(&sl->sum, v);
-    if (c) atomicAdd(&sl->cnt, c);
+    AccSnap acc;
+    agg_parse_frozen(layout, acc_data + off, &acc);
+    if (acc.valid) unsafeAtomicAdd(&sl->sum, acc.sum);
+    if (acc.cnt) atomicAdd(&sl->cnt, acc.cnt);
     if (t.mm) {
-      atomicMin(&t.mm[2 * a], mnu);
-      atomicMax(&t.mm[2 * a + 1], mxu);
+      atomicMin(&t.mm[2 * a], acc.minu);
+      atomicMax(&t.mm[2 * a + 1], acc.maxu);
+    }
+    if (t.f_row) {  // priority = the record's preserved group first_row
+      if (acc.f_st) atomicMin(&t.f_row[2 * a], fr);
+      if (acc.fn_st) atomicMin(&t.f_row[2 * a + 1], fr);
     }
   }
 }
@@ -605,6 +645,25 @@ __global__ void k_agg_gather_out(const AggTable t,
   }
 }
 
+// snapshot one slot's accumulators for freeze
+__device__ __forceinline__ AccSnap table_snap(const AggTable& t, uint32_t s) {
+  AccSnap a;
+  a.valid = t.slots[s].cnt != 0;
+  a.sum = t.slots[s].sum;
+  a.cnt = t.slots[s].cnt;
+  if (t.mm) {
+    a.minu = t.mm[2 * s];
+    a.maxu = t.mm[2 * s + 1];
+  }
+  if (t.f_row) {
+    a.f_st = t.f_st[2 * s];
+    a.f_val = t.f_val[2 * s];
+    a.fn_st = t.f_st[2 * s + 1];
+    a.fn_val = t.f_val[2 * s + 1];
+  }
+  return a;
+}
+
 __global__ void k_agg_freeze_len(const AggTable t,
                                  const uint32_t* __restrict__ order_slots,
                                  int64_t num_groups, int32_t* __restrict__ lens,
@@ -612,9 +671,7 @@ __global__ void k_agg_freeze_len(const AggTable t,
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < num_groups;
        i += (int64_t)gridDim.x * blockDim.x) {
     uint32_t s = order_slots[i];
-    lens[i] = agg_freeze_len(layout, t.slots[s].cnt != 0, t.slots[s].cnt,
-                             t.mm ? t.mm[2 * s] : MM_MIN_INIT,
-                             t.mm ? t.mm[2 * s + 1] : MM_MAX_INIT);
+    lens[i] = agg_freeze_len(layout, table_snap(t, s));
   }
 }
 
@@ -626,10 +683,7 @@ __global__ void k_agg_freeze_write(const AggTable t,
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < num_groups;
        i += (int64_t)gridDim.x * blockDim.x) {
     uint32_t s = order_slots[i];
-    agg_freeze_write_rec(layout, t.slots[s].cnt != 0, t.slots[s].sum,
-                         t.slots[s].cnt, t.mm ? t.mm[2 * s] : MM_MIN_INIT,
-                         t.mm ? t.mm[2 * s + 1] : MM_MAX_INIT,
-                         out + offsets[i]);
+    agg_freeze_write_rec(layout, table_snap(t, s), out + offsets[i]);
   }
 }
 
@@ -687,18 +741,31 @@ void launch_agg_freeze_write(const AggTable& t, const uint32_t* order_slots,
 }
 
 // partial-skipping pass-through freeze (agg_ctx.rs:428-462
-// process_partial_skipped): each input row becomes one record; SUM acc =
-// value when valid, COUNT = valid ? 1 : 0.
+// process_partial_skipped): each input row becomes its own single-row group;
+// SUM acc = value when valid, COUNT = valid ? 1 : 0, MIN = MAX = value,
+// FIRST = value-or-null (touched), FIRST_IGNORES_NULL = value when valid.
+__device__ __forceinline__ AccSnap row_snap(bool v, double val) {
+  AccSnap a;
+  a.valid = v;
+  a.sum = val;
+  a.cnt = v ? 1 : 0;
+  uint64_t u = f64_omap(val);
+  a.minu = v ? u : MM_MIN_INIT;
+  a.maxu = v ? u : MM_MAX_INIT;
+  a.f_st = v ? 2 : 1;  // a row always touches its own group's FIRST
+  a.f_val = val;
+  a.fn_st = v ? 2 : 0;
+  a.fn_val = val;
+  return a;
+}
+
 __global__ void k_skip_freeze_len(const uint8_t* __restrict__ val_valid,
                                   int64_t n, int32_t* __restrict__ lens,
                                   uint32_t layout) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     bool v = !val_valid || bit_get_dev(val_valid, i);
-    // a valid row's MIN/MAX parts are always present (8+1 bytes): pass any
-    // non-sentinel accumulator value when valid
-    lens[i] = agg_freeze_len(layout, v, v ? 1 : 0, v ? 0 : MM_MIN_INIT,
-                             v ? ~0ull : MM_MAX_INIT);
+    lens[i] = agg_freeze_len(layout, row_snap(v, 0.0));
   }
 }
 
@@ -709,9 +776,7 @@ __global__ void k_skip_freeze_write(const double* __restrict__ vals,
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     bool v = !val_valid || bit_get_dev(val_valid, i);
-    uint64_t u = v ? f64_omap(vals[i]) : 0;  // row is its own group: min=max=v
-    agg_freeze_write_rec(layout, v, v ? vals[i] : 0.0, v ? 1 : 0,
-                         v ? u : MM_MIN_INIT, v ? u : MM_MAX_INIT,
+    agg_freeze_write_rec(layout, row_snap(v, v ? vals[i] : 0.0),
                          out + offsets[i]);
   }
 }
@@ -781,6 +846,128 @@ void launch_mm_init(unsigned long long* mm, int64_t n, hipStream_t s) {
   check_launch("k_mm_init");
 }
 
+__global__ void k_first_init(unsigned long long* __restrict__ f_row,
+                             double* __restrict__ f_val,
+                             uint8_t* __restrict__ f_st, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < 2 * n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    f_row[i] = ~0ull;
+    f_val[i] = 0.0;
+    f_st[i] = 0;
+  }
+}
+
+void launch_first_init(unsigned long long* f_row, double* f_val, uint8_t* f_st,
+                       int64_t n, hipStream_t s) {
+  hipLaunchKernelGGL(k_first_init, dim3(grid_for(2 * n)), dim3(BLOCK), 0, s,
+                     f_row, f_val, f_st, n);
+  check_launch("k_first_init");
+}
+
+// FIRST-family pass B (see kernels.h): the unique row whose priority equals
+// the slot's captured f_row plain-stores its value and state. Runs after the
+// chunk's pass A on the same stream.
+__global__ void k_first_capture_update(const AggTable t,
+                                       const int64_t* __restrict__ keys,
+                                       const uint8_t* __restrict__ key_valid,
+                                       const double* __restrict__ vals,
+                                       const uint8_t* __restrict__ val_valid,
+                                       int64_t n, uint64_t row_offset) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    bool knull = key_valid && !bit_get_dev(key_valid, i);
+    int64_t a = agg_upsert_slot(t, knull ? 0 : keys[i], knull);
+    if (a < 0) continue;
+    unsigned long long row = row_offset + (uint64_t)i;
+    bool vv = !val_valid || bit_get_dev(val_valid, i);
+    if (t.f_row[2 * a] == row) {  // first.rs:91-148: first row latches,
+      t.f_val[2 * a] = vv ? vals[i] : 0.0;  // null first value => touched-null
+      t.f_st[2 * a] = vv ? 2 : 1;
+    }
+    if (vv && t.f_row[2 * a + 1] == row) {  // first_ignores_null.rs:83-117
+      t.f_val[2 * a + 1] = vals[i];
+      t.f_st[2 * a + 1] = 2;
+    }
+  }
+}
+
+void launch_first_capture_update(const AggTable& t, const int64_t* keys,
+                                 const uint8_t* key_valid, const double* vals,
+                                 const uint8_t* val_valid, int64_t n,
+                                 uint64_t row_offset, hipStream_t s) {
+  hipLaunchKernelGGL(k_first_capture_update, dim3(grid_for(n)), dim3(BLOCK), 0,
+                     s, t, keys, key_valid, vals, val_valid, n, row_offset);
+  check_launch("k_first_capture_update");
+}
+
+__global__ void k_first_capture_frozen(const AggTable t,
+                                       const int64_t* __restrict__ keys,
+                                       const uint8_t* __restrict__ key_valid,
+                                       const uint8_t* __restrict__ acc_data,
+                                       const int32_t* __restrict__ acc_offsets,
+                                       const unsigned long long* __restrict__ prio,
+                                       int64_t n, uint64_t row_offset,
+                                       uint32_t layout) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    bool knull = key_valid && !bit_get_dev(key_valid, i);
+    int64_t a = agg_upsert_slot(t, knull ? 0 : keys[i], knull);
+    if (a < 0) continue;
+    unsigned long long pr = prio ? prio[i] : row_offset + (uint64_t)i;
+    AccSnap acc;
+    agg_parse_frozen(layout, acc_data + acc_offsets[i], &acc);
+    if (acc.f_st && t.f_row[2 * a] == pr) {
+      t.f_val[2 * a] = acc.f_val;
+      t.f_st[2 * a] = acc.f_st;
+    }
+    if (acc.fn_st && t.f_row[2 * a + 1] == pr) {
+      t.f_val[2 * a + 1] = acc.fn_val;
+      t.f_st[2 * a + 1] = acc.fn_st;
+    }
+  }
+}
+
+void launch_first_capture_frozen(const AggTable& t, const int64_t* keys,
+                                 const uint8_t* key_valid,
+                                 const uint8_t* acc_data,
+                                 const int32_t* acc_offsets,
+                                 const unsigned long long* prio, int64_t n,
+                                 uint64_t row_offset, uint32_t layout,
+                                 hipStream_t s) {
+  hipLaunchKernelGGL(k_first_capture_frozen, dim3(grid_for(n)), dim3(BLOCK), 0,
+                     s, t, keys, key_valid, acc_data, acc_offsets, prio, n,
+                     row_offset, layout);
+  check_launch("k_first_capture_frozen");
+}
+
+__global__ void k_first_gather(const AggTable t,
+                               const uint32_t* __restrict__ order_slots,
+                               int64_t num_groups, int which,
+                               double* __restrict__ out_vals,
+                               uint8_t* __restrict__ out_validity) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < num_groups; i += (int64_t)gridDim.x * blockDim.x) {
+    uint32_t s = order_slots[i];
+    out_vals[i] = t.f_val[2 * s + which];
+    if ((i & 7) == 0) {
+      uint8_t b = 0;
+      for (int j = 0; j < 8 && i + j < num_groups; j++)
+        if (t.f_st[2 * order_slots[i + j] + which] == 2)
+          b |= (uint8_t)(1u << j);
+      out_validity[i >> 3] = b;
+    }
+  }
+}
+
+void launch_first_gather(const AggTable& t, const uint32_t* order_slots,
+                         int64_t num_groups, int which, double* out_vals,
+                         uint8_t* out_validity, hipStream_t s) {
+  hipLaunchKernelGGL(k_first_gather, dim3(grid_for(num_groups)), dim3(BLOCK),
+                     0, s, t, order_slots, num_groups, which, out_vals,
+                     out_validity);
+  check_launch("k_first_gather");
+}
+
 __global__ void k_iota_u32(uint32_t* __restrict__ dst, int64_t n) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x)
@@ -816,6 +1003,13 @@ __global__ void k_agg_rebuild(const AggTable dst, const AggTable src) {
     if (src.mm) {
       dst.mm[2 * a] = src.mm[2 * i];
       dst.mm[2 * a + 1] = src.mm[2 * i + 1];
+    }
+    if (src.f_row) {
+      for (int j = 0; j < 2; j++) {
+        dst.f_row[2 * a + j] = src.f_row[2 * i + j];
+        dst.f_val[2 * a + j] = src.f_val[2 * i + j];
+        dst.f_st[2 * a + j] = src.f_st[2 * i + j];
+      }
     }
   }
 }

@@ -9,6 +9,7 @@ import struct
 
 # AggFunction enum (auron.proto:128-141)
 AGG_MIN, AGG_MAX, AGG_SUM, AGG_AVG, AGG_COUNT = 0, 1, 2, 3, 4
+AGG_FIRST, AGG_FIRST_IGNORES_NULL = 7, 8
 # AggMode enum (auron.proto:692-696)
 MODE_PARTIAL, MODE_PARTIAL_MERGE, MODE_FINAL = 0, 1, 2
 
@@ -141,7 +142,8 @@ def sum_count_aggs(val_index=1):
 
 
 _AGG_FN = {"min": AGG_MIN, "max": AGG_MAX, "sum": AGG_SUM,
-           "avg": AGG_AVG, "count": AGG_COUNT}
+           "avg": AGG_AVG, "count": AGG_COUNT, "first": AGG_FIRST,
+           "first_ignores_null": AGG_FIRST_IGNORES_NULL}
 
 
 def named_aggs(names, val_index=1):

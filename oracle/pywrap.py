@@ -325,3 +325,29 @@ def minmax_freeze_rec(parts):
         else:
             out += write_len(int(v))
     return out
+
+
+# ---- FIRST / FIRST_IGNORES_NULL restatement (pure python) ------------------
+# first.rs:91-148 partial_update: the group's FIRST acc latches on the first
+# processed row — value = that row's arg (null included), flag = touched.
+# first_ignores_null.rs:83-117: the acc latches on the first NON-NULL arg.
+# Merge (first.rs:198-207 / first_ignores_null.rs:146-170): the earliest
+# touched (resp. valid) partial wins, in sequential merge order.
+def first_groups(keys, vals, val_valid=None, key_valid=None):
+    """Insertion-ordered: returns (ordered_keys, firsts, firsts_ignore_null)
+    where firsts[i] is (touched, value_or_None) and firsts_ignore_null[i] is
+    value_or_None."""
+    n = len(keys)
+    groups = {}
+    for i in range(n):
+        k = None if (key_valid is not None and not key_valid[i]) else int(keys[i])
+        vv = val_valid is None or bool(val_valid[i])
+        v = float(vals[i]) if vv else None
+        if k not in groups:
+            groups[k] = [True, v, v]           # touched, first, first_nonnull
+        elif groups[k][2] is None and v is not None:
+            groups[k][2] = v
+    ordered = list(groups.keys())
+    firsts = [(groups[k][0], groups[k][1]) for k in ordered]
+    firsts_nn = [groups[k][2] for k in ordered]
+    return ordered, firsts, firsts_nn

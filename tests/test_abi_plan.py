@@ -101,3 +101,9 @@ def test_decode_minmax_plan():
     td = plan.plan_partial_final_named(["min", "max", "sum", "count"])
     s = blaze_amd.debug_decode_plan(td)
     assert "fn0" in s and "fn1" in s  # MIN=0, MAX=1 (auron.proto AggFunction)
+
+
+def test_decode_first_plan():
+    td = plan.plan_partial_final_named(["first", "first_ignores_null"])
+    s = blaze_amd.debug_decode_plan(td)
+    assert "fn7" in s and "fn8" in s

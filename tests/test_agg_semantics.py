@@ -109,3 +109,17 @@ def test_minmax_freeze_rec_format():
     assert rec[:9] == b"\x01" + struct.pack("<d", 1.5)
     assert rec[9:10] == b"\x00"
     assert rec[10:] == oracle.write_len(300)
+
+
+def test_first_restatement_hand_case():
+    """oracle.pywrap.first_groups vs a hand trace of first.rs:91-148 /
+    first_ignores_null.rs:83-117: FIRST latches the first row's value even
+    when null; FIRST_IGNORES_NULL latches the first non-null."""
+    from oracle import pywrap as oracle
+    keys = [1, 1, 2, 2, 3]
+    vals = [9.0, 4.0, 5.0, 6.0, 7.0]
+    vv = [False, True, True, True, False]
+    ok, firsts, firsts_nn = oracle.first_groups(keys, vals, vv)
+    assert ok == [1, 2, 3]
+    assert firsts == [(True, None), (True, 5.0), (True, None)]
+    assert firsts_nn == [4.0, 5.0, None]

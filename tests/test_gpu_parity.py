@@ -553,3 +553,87 @@ def test_minmax_partial_skipping():
     np.testing.assert_array_equal(got["min"][0][gi], np.array(mins)[oi])
     np.testing.assert_array_equal(got["max"][0][gi], np.array(maxs)[oi])
     t.finalize()
+
+
+def test_first_agg_family():
+    """FIRST / FIRST_IGNORES_NULL (first.rs, first_ignores_null.rs) in a
+    mixed list, partial->final. The engine's two-pass capture must reproduce
+    the reference's sequential latch semantics exactly."""
+    keys, vals, vv = gen_northstar(200_000, nkeys=1500, null_frac=0.3,
+                                   fractional=True)
+    names = ["first", "first_ignores_null", "count"]
+    t = blaze_amd.Task(plan.plan_partial_final_named(names),
+                       batches=batches_of(keys, vals, vv))
+    outs = t.run()
+    got = _minmax_cols(outs, names)
+    ok, firsts, firsts_nn = oracle.first_groups(keys, vals, vv)
+    np.testing.assert_array_equal(got["key"][0], np.array(ok))
+    exp_f_valid = np.array([f[1] is not None for f in firsts])
+    exp_fn_valid = np.array([f is not None for f in firsts_nn])
+    assert (~exp_f_valid).sum() > 0      # some groups start with a null value
+    np.testing.assert_array_equal(got["first"][1], exp_f_valid)
+    np.testing.assert_array_equal(
+        got["first"][0][exp_f_valid],
+        np.array([f[1] for f in firsts if f[1] is not None]))
+    np.testing.assert_array_equal(got["first_ignores_null"][1], exp_fn_valid)
+    np.testing.assert_array_equal(
+        got["first_ignores_null"][0][exp_fn_valid],
+        np.array([f for f in firsts_nn if f is not None]))
+    t.finalize()
+
+
+def test_first_spill_grow_and_skipping():
+    """FIRST survives table grow + host spill (frozen parts + preserved
+    first_row priorities) and partial-skipping pass-through."""
+    keys, vals, vv = gen_northstar(300_000, nkeys=90_000, null_frac=0.2,
+                                   fractional=True)
+    names = ["first", "first_ignores_null"]
+    conf = {"AURON_HIP_MEM_BUDGET": 1 << 20,
+            "AURON_HIP_AGG_TABLE_SLOTS": 1 << 10}
+    t = blaze_amd.Task(plan.plan_partial_final_named(names),
+                       batches=batches_of(keys, vals, vv), conf=conf)
+    outs = t.run()
+    assert t.metric("spill_count") > 0
+    got = _minmax_cols(outs, names)
+    ok, firsts, firsts_nn = oracle.first_groups(keys, vals, vv)
+    k = got["key"][0]
+    assert len(k) == len(ok)
+    gi = np.argsort(k, kind="stable")
+    oi = np.argsort(np.array(ok), kind="stable")
+    np.testing.assert_array_equal(k[gi], np.array(ok)[oi])
+    exp_f = np.array([np.nan if f[1] is None else f[1] for f in firsts])
+    exp_fv = ~np.isnan(exp_f)
+    np.testing.assert_array_equal(got["first"][1][gi], exp_fv[oi])
+    np.testing.assert_array_equal(got["first"][0][gi][exp_fv[oi]],
+                                  exp_f[oi][exp_fv[oi]])
+    exp_n = np.array([np.nan if f is None else f for f in firsts_nn])
+    exp_nv = ~np.isnan(exp_n)
+    np.testing.assert_array_equal(got["first_ignores_null"][1][gi], exp_nv[oi])
+    np.testing.assert_array_equal(
+        got["first_ignores_null"][0][gi][exp_nv[oi]], exp_n[oi][exp_nv[oi]])
+
+    # skipping: distinct keys flip the partial stage to pass-through
+    n = 50_000
+    keys2 = np.arange(n, dtype=np.int64)
+    rng = np.random.default_rng(5)
+    vals2 = rng.random(n)
+    vv2 = rng.random(n) >= 0.3
+    reader = plan.ffi_reader(plan.northstar_input_fields(), "input0")
+    partial = plan.agg(reader, [plan.column("key", 0)],
+                       plan.named_aggs(names),
+                       [plan.MODE_PARTIAL] * 2, ["key"], names,
+                       supports_partial_skipping=True)
+    final = plan.agg(partial, [plan.column("key", 0)],
+                     plan.named_aggs(names),
+                     [plan.MODE_FINAL] * 2, ["key"], names)
+    t2 = blaze_amd.Task(plan.task_definition(final),
+                        batches=batches_of(keys2, vals2, vv2))
+    outs2 = t2.run()
+    got2 = _minmax_cols(outs2, names)
+    k2 = got2["key"][0]
+    gi2 = np.argsort(k2, kind="stable")
+    np.testing.assert_array_equal(k2[gi2], keys2)
+    np.testing.assert_array_equal(got2["first"][1][gi2], vv2)
+    np.testing.assert_array_equal(got2["first"][0][gi2][vv2], vals2[vv2])
+    np.testing.assert_array_equal(got2["first_ignores_null"][1][gi2], vv2)
+    t2.finalize()
