@@ -722,8 +722,20 @@ class AggOp {
     const uint8_t* kv = key.validity ? key.validity + done / 8 : nullptr;
     const uint8_t* vv = val.validity ? val.validity + done / 8 : nullptr;
 
-    if (!d_part_) {
-      d_part_.alloc(AGG2_MAX_CHUNK * sizeof(PartRow));
+    if (!d_partkv_) {
+      // A/B toggle for the partition-record layout: the 24B AoS record
+      // (default) vs 16B kv + 4B rowv split streams. Same-box A/B (4
+      // interleaved 1B-row runs) measured the split at 82.1/82.6 ms/step vs
+      // AoS at 78.4/88.9 — a wash within box noise, so the simpler AoS
+      // stays; AURON_AGG2_SPLIT=1 selects the split for experiments.
+      const char* e = getenv("AURON_AGG2_SPLIT");
+      agg2_split_ = (e && e[0] == '1');
+      if (agg2_split_) {
+        d_partkv_.alloc(AGG2_MAX_CHUNK * sizeof(PartKV));
+        d_rowv_.alloc(AGG2_MAX_CHUNK * 4);
+      } else {
+        d_partkv_.alloc(AGG2_MAX_CHUNK * sizeof(PartRow));
+      }
       d_leftover_.alloc(AGG2_MAX_CHUNK * sizeof(PartRow));
       int64_t mat = (int64_t)nbuck << AGG2_GRID_LOG2;
       d_counts_.alloc((mat + 1) * 4);   // +1: scan total slot
@@ -756,18 +768,34 @@ class AggOp {
     launch_agg2_offsets(d_scanned_.get<uint32_t>(), AGG2_NBUCK_LOG2,
                         d_offsets_.get<uint32_t>(), stream_);
     // P2: scatter
-    launch_agg2_scatter(keys, kv, vals, vv, chunk, AGG2_NBUCK_LOG2,
-                        d_scanned_.get<uint32_t>(), d_part_.get<PartRow>(),
-                        stream_);
+    if (agg2_split_)
+      launch_agg2_scatter(keys, kv, vals, vv, chunk, AGG2_NBUCK_LOG2,
+                          d_scanned_.get<uint32_t>(), d_partkv_.get<PartKV>(),
+                          d_rowv_.get<uint32_t>(), stream_);
+    else
+      launch_agg2_scatter24(keys, kv, vals, vv, chunk, AGG2_NBUCK_LOG2,
+                            d_scanned_.get<uint32_t>(),
+                            d_partkv_.get<PartRow>(), stream_);
     // A: per-bucket LDS aggregation (counters zeroed above; staged_n at +0,
     // lo_n at +1, special count at byte offset 8 via the +2 uint32 slot)
-    launch_agg2_bucket(d_part_.get<PartRow>(), d_offsets_.get<uint32_t>(),
-                       nbuck, d_staged_.get<StagedGroup>(),
-                       d_counters_.get<unsigned long long>(),
-                       (int64_t)nbuck * AGG2_LSLOTS,
-                       d_leftover_.get<PartRow>(),
-                       d_counters_.get<unsigned long long>() + 1,
-                       t_.error_flag, stream_);
+    if (agg2_split_)
+      launch_agg2_bucket(d_partkv_.get<PartKV>(), d_rowv_.get<uint32_t>(),
+                         d_offsets_.get<uint32_t>(),
+                         nbuck, d_staged_.get<StagedGroup>(),
+                         d_counters_.get<unsigned long long>(),
+                         (int64_t)nbuck * AGG2_LSLOTS,
+                         d_leftover_.get<PartRow>(),
+                         d_counters_.get<unsigned long long>() + 1,
+                         t_.error_flag, stream_);
+    else
+      launch_agg2_bucket24(d_partkv_.get<PartRow>(),
+                           d_offsets_.get<uint32_t>(),
+                           nbuck, d_staged_.get<StagedGroup>(),
+                           d_counters_.get<unsigned long long>(),
+                           (int64_t)nbuck * AGG2_LSLOTS,
+                           d_leftover_.get<PartRow>(),
+                           d_counters_.get<unsigned long long>() + 1,
+                           t_.error_flag, stream_);
     unsigned long long* h_ctr =
         (unsigned long long*)(pinned_agg2_.get<uint8_t>() +
                               2 * (size_t)(nbuck + 1) * 4);
@@ -1243,7 +1271,8 @@ class AggOp {
   DevBuf d_slots_, d_special_, d_ng_, d_err_, d_mm_, d_frow_, d_fval_, d_fst_;
   PinnedBuf pinned_meta_, pinned_emit_;
   // two-phase scratch (allocated on first large chunk)
-  DevBuf d_part_, d_leftover_, d_counts_, d_scanned_, d_scan_tmp_, d_offsets_,
+  bool agg2_split_ = false;
+  DevBuf d_partkv_, d_rowv_, d_leftover_, d_counts_, d_scanned_, d_scan_tmp_, d_offsets_,
       d_staged_, d_counters_;
   PinnedBuf pinned_agg2_;
   std::vector<DevBatch> held_, skipped_;

@@ -106,7 +106,8 @@ __global__ void k_agg2_scatter(const int64_t* __restrict__ keys,
                                const uint8_t* __restrict__ val_valid, int64_t n,
                                int nbuck_log2,
                                const uint32_t* __restrict__ scanned,
-                               PartRow* __restrict__ out) {
+                               PartKV* __restrict__ out_kv,
+                               uint32_t* __restrict__ out_rowv) {
   extern __shared__ uint32_t lds_cursor[];
   const uint32_t nbuck = 1u << nbuck_log2;
   for (uint32_t b = threadIdx.x; b < nbuck; b += blockDim.x)
@@ -120,12 +121,8 @@ __global__ void k_agg2_scatter(const int64_t* __restrict__ keys,
     uint32_t b = bucket_of(k, nbuck_log2);
     uint32_t pos = atomicAdd(&lds_cursor[b], 1u);  // block-local LDS cursor
     bool vvalid = !val_valid || bit_get2(val_valid, i);
-    PartRow r;
-    r.key = k;
-    r.val = vals[i];
-    r.rowv = (uint32_t)i | (vvalid ? 0x80000000u : 0u);
-    r._pad = 0;
-    out[pos] = r;
+    out_kv[pos] = PartKV{k, vals[i]};   // one aligned 16-byte store
+    out_rowv[pos] = (uint32_t)i | (vvalid ? 0x80000000u : 0u);
   }
 }
 
@@ -163,7 +160,8 @@ static constexpr int LSLOTS = AGG2_LSLOTS;  // 48 KB LDS -> 3 blocks/CU
 static constexpr int LPROBE = 64;
 
 __global__ void __launch_bounds__(256) k_agg2_bucket(
-    const PartRow* __restrict__ part, const uint32_t* __restrict__ offsets,
+    const PartKV* __restrict__ part, const uint32_t* __restrict__ part_rowv,
+    const uint32_t* __restrict__ offsets,
     int nbuckets, StagedGroup* __restrict__ staged,
     unsigned long long* __restrict__ staged_n, int64_t staged_cap,
     PartRow* __restrict__ leftover, unsigned long long* __restrict__ lo_n,
@@ -185,7 +183,7 @@ __global__ void __launch_bounds__(256) k_agg2_bucket(
     for (uint32_t i = beg + threadIdx.x; i < end; i += blockDim.x) {
       int64_t k = part[i].key;
       double v = part[i].val;
-      uint32_t rowv = part[i].rowv;
+      uint32_t rowv = part_rowv[i];
       uint32_t row = rowv & 0x7FFFFFFFu;
       bool vvalid = (rowv & 0x80000000u) != 0;
       // probe LDS (low bits of the same mix; bucket used the high bits)
@@ -217,7 +215,7 @@ __global__ void __launch_bounds__(256) k_agg2_bucket(
       } else {
         // LDS window full: append raw row to the leftover list
         unsigned long long p = atomicAdd(lo_n, 1ull);
-        leftover[p] = part[i];
+        leftover[p] = PartRow{k, v, rowv, 0};
       }
     }
     __syncthreads();
@@ -370,11 +368,11 @@ void scan_counts_matrix(const uint32_t* counts, uint32_t* scanned, int64_t n,
 void launch_agg2_scatter(const int64_t* keys, const uint8_t* key_valid,
                          const double* vals, const uint8_t* val_valid,
                          int64_t n, int nbuck_log2, const uint32_t* scanned,
-                         PartRow* out, hipStream_t s) {
+                         PartKV* out_kv, uint32_t* out_rowv, hipStream_t s) {
   size_t lds = (size_t)(1u << nbuck_log2) * 4;
   hipLaunchKernelGGL(k_agg2_scatter, dim3(1 << GRID_LOG2), dim3(BLOCK), lds, s,
                      keys, key_valid, vals, val_valid, n, nbuck_log2, scanned,
-                     out);
+                     out_kv, out_rowv);
   check_launch2("k_agg2_scatter");
 }
 
@@ -387,16 +385,143 @@ void launch_agg2_specials(const AggTable& t, const int64_t* keys,
   check_launch2("k_agg2_specials");
 }
 
-void launch_agg2_bucket(const PartRow* part, const uint32_t* offsets,
+void launch_agg2_bucket(const PartKV* part_kv, const uint32_t* part_rowv,
+                        const uint32_t* offsets,
                         int nbuckets, StagedGroup* staged,
                         unsigned long long* staged_n, int64_t staged_cap,
                         PartRow* leftover, unsigned long long* lo_n,
                         uint32_t* error_flag, hipStream_t s) {
   int blocks = nbuckets < (int)MAX_BLOCKS ? nbuckets : (int)MAX_BLOCKS;
-  hipLaunchKernelGGL(k_agg2_bucket, dim3(blocks), dim3(BLOCK), 0, s, part,
+  hipLaunchKernelGGL(k_agg2_bucket, dim3(blocks), dim3(BLOCK), 0, s, part_kv,
+                     part_rowv, offsets, nbuckets, staged, staged_n,
+                     staged_cap, leftover, lo_n, error_flag);
+  check_launch2("k_agg2_bucket");
+}
+
+
+// ---- 24B AoS variants (A/B comparison path, selected by the engine) --------
+__global__ void k_agg2_scatter24(const int64_t* __restrict__ keys,
+                                 const uint8_t* __restrict__ key_valid,
+                                 const double* __restrict__ vals,
+                                 const uint8_t* __restrict__ val_valid,
+                                 int64_t n, int nbuck_log2,
+                                 const uint32_t* __restrict__ scanned,
+                                 PartRow* __restrict__ out) {
+  extern __shared__ uint32_t lds_cursor[];
+  const uint32_t nbuck = 1u << nbuck_log2;
+  for (uint32_t b = threadIdx.x; b < nbuck; b += blockDim.x)
+    lds_cursor[b] = scanned[((size_t)b << GRID_LOG2) | blockIdx.x];
+  __syncthreads();
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    bool knull = key_valid && !bit_get2(key_valid, i);
+    int64_t k = keys[i];
+    if (knull || k == KEY_EMPTY2) continue;
+    uint32_t b = bucket_of(k, nbuck_log2);
+    uint32_t pos = atomicAdd(&lds_cursor[b], 1u);
+    bool vvalid = !val_valid || bit_get2(val_valid, i);
+    PartRow r;
+    r.key = k;
+    r.val = vals[i];
+    r.rowv = (uint32_t)i | (vvalid ? 0x80000000u : 0u);
+    r._pad = 0;
+    out[pos] = r;
+  }
+}
+
+void launch_agg2_scatter24(const int64_t* keys, const uint8_t* key_valid,
+                           const double* vals, const uint8_t* val_valid,
+                           int64_t n, int nbuck_log2, const uint32_t* scanned,
+                           PartRow* out, hipStream_t s) {
+  size_t lds = (size_t)(1u << nbuck_log2) * 4;
+  hipLaunchKernelGGL(k_agg2_scatter24, dim3(1 << GRID_LOG2), dim3(BLOCK), lds,
+                     s, keys, key_valid, vals, val_valid, n, nbuck_log2,
+                     scanned, out);
+  check_launch2("k_agg2_scatter24");
+}
+
+__global__ void __launch_bounds__(256) k_agg2_bucket24(
+    const PartRow* __restrict__ part, const uint32_t* __restrict__ offsets,
+    int nbuckets, StagedGroup* __restrict__ staged,
+    unsigned long long* __restrict__ staged_n, int64_t staged_cap,
+    PartRow* __restrict__ leftover, unsigned long long* __restrict__ lo_n,
+    uint32_t* __restrict__ error_flag) {
+  __shared__ int64_t ls_key[LSLOTS];
+  __shared__ double ls_sum[LSLOTS];
+  __shared__ uint32_t ls_cnt[LSLOTS];
+  __shared__ uint32_t ls_first[LSLOTS];
+  for (int b = blockIdx.x; b < nbuckets; b += gridDim.x) {
+    for (int s = threadIdx.x; s < LSLOTS; s += blockDim.x) {
+      ls_key[s] = KEY_EMPTY2;
+      ls_sum[s] = 0.0;
+      ls_cnt[s] = 0;
+      ls_first[s] = 0xFFFFFFFFu;
+    }
+    __syncthreads();
+    uint32_t beg = offsets[b], end = offsets[b + 1];
+    for (uint32_t i = beg + threadIdx.x; i < end; i += blockDim.x) {
+      int64_t k = part[i].key;
+      double v = part[i].val;
+      uint32_t rowv = part[i].rowv;
+      uint32_t row = rowv & 0x7FFFFFFFu;
+      bool vvalid = (rowv & 0x80000000u) != 0;
+      uint32_t h = (uint32_t)mix64_2((uint64_t)k) & (LSLOTS - 1);
+      int found = -1;
+      for (int p = 0; p < LPROBE; p++) {
+        int64_t cur = ls_key[h];
+        if (cur == k) {
+          found = (int)h;
+          break;
+        }
+        if (cur == KEY_EMPTY2) {
+          long long prev = atomicCAS((unsigned long long*)&ls_key[h],
+                                     (unsigned long long)KEY_EMPTY2,
+                                     (unsigned long long)k);
+          if (prev == (long long)KEY_EMPTY2 || prev == (long long)k) {
+            found = (int)h;
+            break;
+          }
+        }
+        h = (h + 1) & (LSLOTS - 1);
+      }
+      if (found >= 0) {
+        atomicMin(&ls_first[found], row);
+        if (vvalid) {
+          unsafeAtomicAdd(&ls_sum[found], v);
+          atomicAdd(&ls_cnt[found], 1u);
+        }
+      } else {
+        unsigned long long p = atomicAdd(lo_n, 1ull);
+        leftover[p] = part[i];
+      }
+    }
+    __syncthreads();
+    for (int s = threadIdx.x; s < LSLOTS; s += blockDim.x) {
+      if (ls_key[s] == KEY_EMPTY2) continue;
+      unsigned long long p = atomicAdd(staged_n, 1ull);
+      if ((int64_t)p >= staged_cap) {
+        atomicOr(error_flag, 2u);
+        continue;
+      }
+      staged[p].key = ls_key[s];
+      staged[p].sum = ls_sum[s];
+      staged[p].cnt_first =
+          ((unsigned long long)ls_cnt[s] << 32) | ls_first[s];
+    }
+    __syncthreads();
+  }
+}
+
+void launch_agg2_bucket24(const PartRow* part, const uint32_t* offsets,
+                          int nbuckets, StagedGroup* staged,
+                          unsigned long long* staged_n, int64_t staged_cap,
+                          PartRow* leftover, unsigned long long* lo_n,
+                          uint32_t* error_flag, hipStream_t s) {
+  int blocks = nbuckets < (int)MAX_BLOCKS ? nbuckets : (int)MAX_BLOCKS;
+  hipLaunchKernelGGL(k_agg2_bucket24, dim3(blocks), dim3(BLOCK), 0, s, part,
                      offsets, nbuckets, staged, staged_n, staged_cap, leftover,
                      lo_n, error_flag);
-  check_launch2("k_agg2_bucket");
+  check_launch2("k_agg2_bucket24");
 }
 
 void launch_agg2_merge_groups(const AggTable& t, const StagedGroup* staged,
