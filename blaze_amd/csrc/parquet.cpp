@@ -253,11 +253,12 @@ bool lz4raw_uncompress(const uint8_t* src, size_t n, size_t out_size,
 // ---- RLE/bit-packed hybrid (parquet encoding spec) -------------------------
 void rle_bp_decode(const uint8_t* p, size_t len, int bit_width, int64_t count,
                    std::vector<uint32_t>* out) {
-  out->clear();
-  out->reserve(count);
+  out->resize(count);
+  uint32_t* dst = out->data();
+  int64_t filled = 0;
   size_t pos = 0;
   const int byte_w = (bit_width + 7) / 8;
-  while ((int64_t)out->size() < count) {
+  while (filled < count) {
     if (pos >= len) fail("rle: truncated stream");
     uint64_t header = 0;
     int shift = 0;
@@ -273,18 +274,31 @@ void rle_bp_decode(const uint8_t* p, size_t len, int bit_width, int64_t count,
       int64_t n = (int64_t)(header >> 1) * 8;
       size_t bytes = (size_t)n * bit_width / 8;
       if (pos + bytes > len) fail("rle: truncated bit-packed run");
-      uint64_t acc = 0;
-      int bits = 0;
-      size_t bp = pos;
-      for (int64_t i = 0; i < n && (int64_t)out->size() < count; i++) {
-        while (bits < bit_width) {
-          acc |= (uint64_t)p[bp++] << bits;
-          bits += 8;
+      int64_t take = n < count - filled ? n : count - filled;
+      if (bit_width == 1) {  // def-level fast path: 1 byte -> 8 values
+        int64_t i = 0;
+        for (; i + 8 <= take; i += 8) {
+          uint8_t b = p[pos + (size_t)(i >> 3)];
+          for (int j = 0; j < 8; j++) dst[filled + i + j] = (b >> j) & 1;
         }
-        out->push_back((uint32_t)(acc & ((1ull << bit_width) - 1)));
-        acc >>= bit_width;
-        bits -= bit_width;
+        for (; i < take; i++)
+          dst[filled + i] = (p[pos + (size_t)(i >> 3)] >> (i & 7)) & 1;
+      } else {
+        uint64_t acc = 0;
+        int bits = 0;
+        size_t bp = pos;
+        const uint32_t mask = (uint32_t)((1ull << bit_width) - 1);
+        for (int64_t i = 0; i < take; i++) {
+          while (bits < bit_width) {
+            acc |= (uint64_t)p[bp++] << bits;
+            bits += 8;
+          }
+          dst[filled + i] = (uint32_t)acc & mask;
+          acc >>= bit_width;
+          bits -= bit_width;
+        }
       }
+      filled += take;
       pos += bytes;
     } else {
       int64_t n = (int64_t)(header >> 1);
@@ -292,8 +306,9 @@ void rle_bp_decode(const uint8_t* p, size_t len, int bit_width, int64_t count,
       if (pos + byte_w > len) fail("rle: truncated repeated run");
       for (int b = 0; b < byte_w; b++) v |= (uint32_t)p[pos + b] << (8 * b);
       pos += byte_w;
-      for (int64_t i = 0; i < n && (int64_t)out->size() < count; i++)
-        out->push_back(v);
+      int64_t take = n < count - filled ? n : count - filled;
+      for (int64_t i = 0; i < take; i++) dst[filled + i] = v;
+      filled += take;
     }
   }
 }
@@ -312,8 +327,9 @@ bool snappy_uncompress(const uint8_t* src, size_t n, std::vector<uint8_t>* out,
     if (!(b & 0x80)) break;
     shift += 7;
   }
-  out->clear();
-  out->reserve(out_len);
+  out->resize(out_len);  // exact output size is in the header
+  uint8_t* dst = out->data();
+  size_t op = 0;
   while (pos < n) {
     uint8_t tag = src[pos++];
     int type = tag & 3;
@@ -325,11 +341,12 @@ bool snappy_uncompress(const uint8_t* src, size_t n, std::vector<uint8_t>* out,
         for (int i = 0; i < nb; i++) len |= (uint64_t)src[pos++] << (8 * i);
         len += 1;
       }
-      if (pos + len > n) {
+      if (pos + len > n || op + len > out_len) {
         *err = "snappy: truncated literal";
         return false;
       }
-      out->insert(out->end(), src + pos, src + pos + len);
+      memcpy(dst + op, src + pos, len);
+      op += len;
       pos += len;
     } else {
       uint64_t len, off;
@@ -345,16 +362,31 @@ bool snappy_uncompress(const uint8_t* src, size_t n, std::vector<uint8_t>* out,
         off = 0;
         for (int i = 0; i < 4; i++) off |= (uint64_t)src[pos++] << (8 * i);
       }
-      if (off == 0 || off > out->size()) {
+      if (off == 0 || off > op || op + len > out_len) {
         *err = "snappy: bad copy offset";
         return false;
       }
-      size_t start = out->size() - off;
-      for (uint64_t i = 0; i < len; i++)
-        out->push_back((*out)[start + i]);  // may self-overlap
+      // self-overlapping copies repeat the trailing pattern: double the
+      // copied span until it covers len, then one bounded memcpy per step
+      size_t start = op - off;
+      if (off >= len) {
+        memcpy(dst + op, dst + start, len);
+        op += len;
+      } else {
+        // done stays a multiple of off, so the replicated region starting
+        // at `start` is a valid (non-overlapping) source for each memcpy
+        size_t done = 0;
+        while (done < len) {
+          size_t avail = (op + done) - start;
+          size_t take = len - done < avail ? len - done : avail;
+          memcpy(dst + op + done, dst + start, take);
+          done += take;
+        }
+        op += len;
+      }
     }
   }
-  if (out->size() != out_len) {
+  if (op != out_len) {
     *err = "snappy: length mismatch";
     return false;
   }
@@ -510,7 +542,10 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
 
   PqColumnChunkData out;
   out.num_values = 0;
-  std::vector<bool> valid_bits;
+  if (ci.physical_type != 6)
+    out.plain.reserve((size_t)cm.num_values * vw);
+  std::vector<uint8_t> valid_bits;  // byte per value (bit-packed at the end)
+  valid_bits.reserve((size_t)cm.num_values);
   const bool is_bytes = ci.physical_type == 6;
   // BYTE_ARRAY values accumulate as non-null (len, bytes) pairs; row-aligned
   // offsets are assembled at the end against the validity bits
@@ -613,32 +648,61 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
     int64_t non_null = nvals;
     if (ci.nullable && !def_levels.empty()) {
       non_null = 0;
+      size_t base = valid_bits.size();
+      valid_bits.resize(base + (size_t)nvals);
       for (int64_t i = 0; i < nvals; i++) {
-        bool v = def_levels[i] != 0;
-        valid_bits.push_back(v);
+        uint8_t v = def_levels[i] != 0;
+        valid_bits[base + (size_t)i] = v;
         non_null += v;
       }
       out.null_count += nvals - non_null;
     } else {
-      for (int64_t i = 0; i < nvals; i++) valid_bits.push_back(true);
+      valid_bits.resize(valid_bits.size() + (size_t)nvals, 1);
     }
 
     // writers may FALL BACK from dictionary to PLAIN mid-chunk (e.g. pyarrow
     // once the dictionary page hits its size limit); flatten the accumulated
     // dict-encoded prefix to PLAIN on host when a mix appears (the dict
     // portion is small by construction in that case)
+    // expand dict indices to PLAIN values appended to out.plain (typed
+    // copies: the per-value vector::insert this replaces dominated the
+    // whole chunk decode)
+    auto expand_dict = [&](const uint32_t* idx, size_t cnt) {
+      size_t base = out.plain.size();
+      out.plain.resize(base + cnt * (size_t)vw);
+      if (vw == 8) {
+        uint64_t* d = reinterpret_cast<uint64_t*>(out.plain.data() + base);
+        const uint64_t* dict =
+            reinterpret_cast<const uint64_t*>(out.dict_values.data());
+        for (size_t i = 0; i < cnt; i++) {
+          if ((int64_t)idx[i] >= out.dict_count)
+            fail("parquet: dict index range");
+          d[i] = dict[idx[i]];
+        }
+      } else if (vw == 4) {
+        uint32_t* d = reinterpret_cast<uint32_t*>(out.plain.data() + base);
+        const uint32_t* dict =
+            reinterpret_cast<const uint32_t*>(out.dict_values.data());
+        for (size_t i = 0; i < cnt; i++) {
+          if ((int64_t)idx[i] >= out.dict_count)
+            fail("parquet: dict index range");
+          d[i] = dict[idx[i]];
+        }
+      } else {
+        for (size_t i = 0; i < cnt; i++) {
+          if ((int64_t)idx[i] >= out.dict_count)
+            fail("parquet: dict index range");
+          memcpy(out.plain.data() + base + i * (size_t)vw,
+                 out.dict_values.data() + (size_t)idx[i] * vw, (size_t)vw);
+        }
+      }
+    };
     auto flatten_dict = [&]() {
       if (!out.uses_dict) return;
-      std::vector<uint8_t> flat;
-      flat.reserve(out.dict_indices.size() * vw);
-      for (uint32_t ix : out.dict_indices) {
-        if ((int64_t)ix >= out.dict_count) fail("parquet: dict index range");
-        flat.insert(flat.end(), out.dict_values.begin() + (size_t)ix * vw,
-                    out.dict_values.begin() + (size_t)(ix + 1) * vw);
-      }
-      out.plain.insert(out.plain.end(), flat.begin(), flat.end());
+      std::vector<uint32_t> idx = std::move(out.dict_indices);
       out.dict_indices.clear();
       out.uses_dict = false;
+      expand_dict(idx.data(), idx.size());
     };
 
     switch (ph.encoding) {
@@ -682,13 +746,7 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
         }
         if (!out.plain.empty()) {
           // already flattened: expand this page directly
-          for (uint32_t ix : idx) {
-            if ((int64_t)ix >= out.dict_count)
-              fail("parquet: dict index range");
-            out.plain.insert(out.plain.end(),
-                             out.dict_values.begin() + (size_t)ix * vw,
-                             out.dict_values.begin() + (size_t)(ix + 1) * vw);
-          }
+          expand_dict(idx.data(), idx.size());
         } else {
           out.dict_indices.insert(out.dict_indices.end(), idx.begin(),
                                   idx.end());
