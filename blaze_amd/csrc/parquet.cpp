@@ -327,7 +327,11 @@ bool snappy_uncompress(const uint8_t* src, size_t n, std::vector<uint8_t>* out,
     if (!(b & 0x80)) break;
     shift += 7;
   }
-  out->resize(out_len);  // exact output size is in the header
+  // 32 bytes of slack let the hot paths issue fixed-size 16/32-byte copies
+  // that overshoot the true length (the classic snappy fast path); the final
+  // resize trims back to out_len without reallocating.
+  constexpr size_t kPad = 32;
+  out->resize(out_len + kPad);
   uint8_t* dst = out->data();
   size_t op = 0;
   while (pos < n) {
@@ -335,6 +339,12 @@ bool snappy_uncompress(const uint8_t* src, size_t n, std::vector<uint8_t>* out,
     int type = tag & 3;
     if (type == 0) {  // literal
       uint64_t len = (tag >> 2) + 1;
+      if (len <= 16 && pos + 16 <= n && op + len <= out_len) {
+        memcpy(dst + op, src + pos, 16);  // overshoot into the slack
+        op += len;
+        pos += len;
+        continue;
+      }
       if (len > 60) {
         int nb = (int)len - 60;
         len = 0;
@@ -366,15 +376,29 @@ bool snappy_uncompress(const uint8_t* src, size_t n, std::vector<uint8_t>* out,
         *err = "snappy: bad copy offset";
         return false;
       }
-      // self-overlapping copies repeat the trailing pattern: double the
-      // copied span until it covers len, then one bounded memcpy per step
       size_t start = op - off;
-      if (off >= len) {
+      if (off >= 16 && len <= 32) {
+        // non-overlapping at 16-byte granularity: two fixed copies overshoot
+        // into the slack
+        memcpy(dst + op, dst + start, 16);
+        memcpy(dst + op + 16, dst + start + 16, 16);
+        op += len;
+      } else if (off >= len) {
         memcpy(dst + op, dst + start, len);
         op += len;
+      } else if (off >= 8) {
+        // overlapping but 8-byte stampable: each 8-byte chunk's source is
+        // fully written before it is read
+        size_t done = 0;
+        while (done < len) {
+          memcpy(dst + op + done, dst + start + done, 8);
+          done += 8;
+        }
+        op += len;
       } else {
-        // done stays a multiple of off, so the replicated region starting
-        // at `start` is a valid (non-overlapping) source for each memcpy
+        // tight overlap (off < 8): replicate the pattern by doubling; done
+        // stays a multiple of off, so `start` is a valid non-overlapping
+        // source for each memcpy
         size_t done = 0;
         while (done < len) {
           size_t avail = (op + done) - start;
@@ -386,6 +410,7 @@ bool snappy_uncompress(const uint8_t* src, size_t n, std::vector<uint8_t>* out,
       }
     }
   }
+  out->resize(out_len);  // trim the slack (no reallocation)
   if (op != out_len) {
     *err = "snappy: length mismatch";
     return false;
