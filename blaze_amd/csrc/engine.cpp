@@ -730,6 +730,14 @@ class AggOp {
       // stays; AURON_AGG2_SPLIT=1 selects the split for experiments.
       const char* e = getenv("AURON_AGG2_SPLIT");
       agg2_split_ = (e && e[0] == '1');
+      // hist/scatter grid (blocks = 1<<log2): 512 default; runtime-tunable
+      // for write-line-footprint experiments (PMC measured ~1.9x write
+      // amplification at 512 blocks x 1024 buckets = one open line per L2
+      // line per XCD)
+      const char* g = getenv("AURON_AGG2_GRID_LOG2");
+      if (g && g[0]) agg2_grid_log2_ = atoi(g);
+      if (agg2_grid_log2_ < 6 || agg2_grid_log2_ > AGG2_GRID_LOG2_MAX)
+        agg2_grid_log2_ = 9;
       if (agg2_split_) {
         d_partkv_.alloc(AGG2_MAX_CHUNK * sizeof(PartKV));
         d_rowv_.alloc(AGG2_MAX_CHUNK * 4);
@@ -737,7 +745,7 @@ class AggOp {
         d_partkv_.alloc(AGG2_MAX_CHUNK * sizeof(PartRow));
       }
       d_leftover_.alloc(AGG2_MAX_CHUNK * sizeof(PartRow));
-      int64_t mat = (int64_t)nbuck << AGG2_GRID_LOG2;
+      int64_t mat = (int64_t)nbuck << AGG2_GRID_LOG2_MAX;
       d_counts_.alloc((mat + 1) * 4);   // +1: scan total slot
       d_scanned_.alloc((mat + 1) * 4);
       d_offsets_.alloc((nbuck + 1) * 4);
@@ -757,24 +765,25 @@ class AggOp {
     // P1: per-block histogram matrix (+ special-row count). ONE host sync
     // per chunk: the scan total feeds the offsets kernel on-device, and the
     // special/staged/leftover counters are read back together after phase A.
-    int64_t mat = (int64_t)nbuck << AGG2_GRID_LOG2;
+    int64_t mat = (int64_t)nbuck << agg2_grid_log2_;
     AURON_HIP(hipMemsetAsync(d_counters_.get(), 0, 24, stream_));
-    launch_agg2_hist(keys, kv, chunk, AGG2_NBUCK_LOG2,
+    launch_agg2_hist(keys, kv, chunk, AGG2_NBUCK_LOG2, agg2_grid_log2_,
                      d_counts_.get<uint32_t>(),
                      (uint32_t*)(d_counters_.get<uint8_t>() + 16), stream_);
     size_t tb = d_scan_tmp_.size();
     scan_counts_matrix(d_counts_.get<uint32_t>(), d_scanned_.get<uint32_t>(),
                        mat + 1, d_scan_tmp_.get(), &tb, stream_);
     launch_agg2_offsets(d_scanned_.get<uint32_t>(), AGG2_NBUCK_LOG2,
-                        d_offsets_.get<uint32_t>(), stream_);
+                        agg2_grid_log2_, d_offsets_.get<uint32_t>(), stream_);
     // P2: scatter
     if (agg2_split_)
       launch_agg2_scatter(keys, kv, vals, vv, chunk, AGG2_NBUCK_LOG2,
-                          d_scanned_.get<uint32_t>(), d_partkv_.get<PartKV>(),
-                          d_rowv_.get<uint32_t>(), stream_);
+                          agg2_grid_log2_, d_scanned_.get<uint32_t>(),
+                          d_partkv_.get<PartKV>(), d_rowv_.get<uint32_t>(),
+                          stream_);
     else
       launch_agg2_scatter24(keys, kv, vals, vv, chunk, AGG2_NBUCK_LOG2,
-                            d_scanned_.get<uint32_t>(),
+                            agg2_grid_log2_, d_scanned_.get<uint32_t>(),
                             d_partkv_.get<PartRow>(), stream_);
     // A: per-bucket LDS aggregation (counters zeroed above; staged_n at +0,
     // lo_n at +1, special count at byte offset 8 via the +2 uint32 slot)
@@ -1272,6 +1281,7 @@ class AggOp {
   PinnedBuf pinned_meta_, pinned_emit_;
   // two-phase scratch (allocated on first large chunk)
   bool agg2_split_ = false;
+  int agg2_grid_log2_ = 9;
   DevBuf d_partkv_, d_rowv_, d_leftover_, d_counts_, d_scanned_, d_scan_tmp_, d_offsets_,
       d_staged_, d_counters_;
   PinnedBuf pinned_agg2_;

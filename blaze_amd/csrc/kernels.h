@@ -249,24 +249,27 @@ struct PartRow {
 static_assert(sizeof(PartRow) == 24, "PartRow must be 24 bytes");
 constexpr int AGG2_LSLOTS = 2048;  // LDS table entries per bucket
 
-constexpr int AGG2_GRID_LOG2 = 9;  // hist/scatter grid: 512 blocks
+constexpr int AGG2_GRID_LOG2_MAX = 10;  // matrix sizing bound for the
+                                        // runtime-tunable hist/scatter grid
 // per-BLOCK bucket histogram into the bucket-major counts matrix
 // [nbuck << AGG2_GRID_LOG2]
 void launch_agg2_hist(const int64_t* keys, const uint8_t* key_valid, int64_t n,
-                      int nbuck_log2, uint32_t* counts_matrix,
+                      int nbuck_log2, int grid_log2, uint32_t* counts_matrix,
                       uint32_t* special_rows, hipStream_t s);
 // exclusive scan over the flat counts matrix -> per-(block,bucket) bases
 void scan_counts_matrix(const uint32_t* counts, uint32_t* scanned, int64_t n,
                         void* temp, size_t* temp_bytes, hipStream_t s);
 void launch_agg2_offsets(const uint32_t* scanned, int nbuck_log2,
-                         uint32_t* offsets, hipStream_t s);
+                         int grid_log2, uint32_t* offsets, hipStream_t s);
 void launch_agg2_scatter(const int64_t* keys, const uint8_t* key_valid,
                          const double* vals, const uint8_t* val_valid,
-                         int64_t n, int nbuck_log2, const uint32_t* scanned,
+                         int64_t n, int nbuck_log2, int grid_log2,
+                         const uint32_t* scanned,
                          PartKV* out_kv, uint32_t* out_rowv, hipStream_t s);
 void launch_agg2_scatter24(const int64_t* keys, const uint8_t* key_valid,
                            const double* vals, const uint8_t* val_valid,
-                           int64_t n, int nbuck_log2, const uint32_t* scanned,
+                           int64_t n, int nbuck_log2, int grid_log2,
+                           const uint32_t* scanned,
                            PartRow* out, hipStream_t s);
 void launch_agg2_bucket24(const PartRow* part, const uint32_t* offsets,
                           int nbuckets, StagedGroup* staged,

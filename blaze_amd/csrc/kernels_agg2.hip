@@ -57,11 +57,10 @@ __device__ __forceinline__ uint32_t bucket_of(int64_t key, int nbuck_log2) {
 // counts_matrix is bucket-major: counts[(b << GRID_LOG2) | blockIdx]. An
 // exclusive scan over the flat matrix then gives every block a private,
 // contiguous output range per bucket — the scatter needs NO global atomics.
-static constexpr int GRID_LOG2 = 9;  // 512 blocks (measured best: 256 blocks lose streaming occupancy, 2048 thrash write-lines)
 
 __global__ void k_agg2_hist(const int64_t* __restrict__ keys,
                             const uint8_t* __restrict__ key_valid, int64_t n,
-                            int nbuck_log2,
+                            int nbuck_log2, int grid_log2,
                             uint32_t* __restrict__ counts_matrix,
                             uint32_t* __restrict__ special_rows) {
   extern __shared__ uint32_t lds_hist[];
@@ -81,7 +80,7 @@ __global__ void k_agg2_hist(const int64_t* __restrict__ keys,
   }
   __syncthreads();
   for (uint32_t b = threadIdx.x; b < nbuck; b += blockDim.x)
-    counts_matrix[((size_t)b << GRID_LOG2) | blockIdx.x] = lds_hist[b];
+    counts_matrix[((size_t)b << grid_log2) | blockIdx.x] = lds_hist[b];
   if (special) atomicAdd(special_rows, special);
 }
 
@@ -89,12 +88,12 @@ __global__ void k_agg2_hist(const int64_t* __restrict__ keys,
 // total (scanned[nbuck << GRID_LOG2], the extra slot) — read on device so the
 // host never synchronizes for it
 __global__ void k_agg2_offsets(const uint32_t* __restrict__ scanned,
-                               int nbuck_log2,
+                               int nbuck_log2, int grid_log2,
                                uint32_t* __restrict__ offsets) {
   int nbuck = 1 << nbuck_log2;
   for (int b = (int)(blockIdx.x * blockDim.x + threadIdx.x); b <= nbuck;
        b += (int)(gridDim.x * blockDim.x))
-    offsets[b] = scanned[(size_t)b << GRID_LOG2];
+    offsets[b] = scanned[(size_t)b << grid_log2];
 }
 
 // ---- phase P2: scatter into per-(block,bucket) reserved ranges -------------
@@ -104,14 +103,14 @@ __global__ void k_agg2_scatter(const int64_t* __restrict__ keys,
                                const uint8_t* __restrict__ key_valid,
                                const double* __restrict__ vals,
                                const uint8_t* __restrict__ val_valid, int64_t n,
-                               int nbuck_log2,
+                               int nbuck_log2, int grid_log2,
                                const uint32_t* __restrict__ scanned,
                                PartKV* __restrict__ out_kv,
                                uint32_t* __restrict__ out_rowv) {
   extern __shared__ uint32_t lds_cursor[];
   const uint32_t nbuck = 1u << nbuck_log2;
   for (uint32_t b = threadIdx.x; b < nbuck; b += blockDim.x)
-    lds_cursor[b] = scanned[((size_t)b << GRID_LOG2) | blockIdx.x];
+    lds_cursor[b] = scanned[((size_t)b << grid_log2) | blockIdx.x];
   __syncthreads();
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
@@ -341,19 +340,19 @@ void launch_agg2_leftovers(const AggTable& t, const PartRow* rows, int64_t n,
 }
 
 void launch_agg2_hist(const int64_t* keys, const uint8_t* key_valid, int64_t n,
-                      int nbuck_log2, uint32_t* counts_matrix,
+                      int nbuck_log2, int grid_log2, uint32_t* counts_matrix,
                       uint32_t* special_rows, hipStream_t s) {
   size_t lds = (size_t)(1u << nbuck_log2) * 4;
-  hipLaunchKernelGGL(k_agg2_hist, dim3(1 << GRID_LOG2), dim3(BLOCK), lds, s,
-                     keys, key_valid, n, nbuck_log2, counts_matrix,
+  hipLaunchKernelGGL(k_agg2_hist, dim3(1 << grid_log2), dim3(BLOCK), lds, s,
+                     keys, key_valid, n, nbuck_log2, grid_log2, counts_matrix,
                      special_rows);
   check_launch2("k_agg2_hist");
 }
 
 void launch_agg2_offsets(const uint32_t* scanned, int nbuck_log2,
-                         uint32_t* offsets, hipStream_t s) {
+                         int grid_log2, uint32_t* offsets, hipStream_t s) {
   hipLaunchKernelGGL(k_agg2_offsets, dim3(8), dim3(BLOCK), 0, s, scanned,
-                     nbuck_log2, offsets);
+                     nbuck_log2, grid_log2, offsets);
   check_launch2("k_agg2_offsets");
 }
 
@@ -367,12 +366,13 @@ void scan_counts_matrix(const uint32_t* counts, uint32_t* scanned, int64_t n,
 
 void launch_agg2_scatter(const int64_t* keys, const uint8_t* key_valid,
                          const double* vals, const uint8_t* val_valid,
-                         int64_t n, int nbuck_log2, const uint32_t* scanned,
+                         int64_t n, int nbuck_log2, int grid_log2,
+                         const uint32_t* scanned,
                          PartKV* out_kv, uint32_t* out_rowv, hipStream_t s) {
   size_t lds = (size_t)(1u << nbuck_log2) * 4;
-  hipLaunchKernelGGL(k_agg2_scatter, dim3(1 << GRID_LOG2), dim3(BLOCK), lds, s,
-                     keys, key_valid, vals, val_valid, n, nbuck_log2, scanned,
-                     out_kv, out_rowv);
+  hipLaunchKernelGGL(k_agg2_scatter, dim3(1 << grid_log2), dim3(BLOCK), lds, s,
+                     keys, key_valid, vals, val_valid, n, nbuck_log2,
+                     grid_log2, scanned, out_kv, out_rowv);
   check_launch2("k_agg2_scatter");
 }
 
@@ -404,13 +404,13 @@ __global__ void k_agg2_scatter24(const int64_t* __restrict__ keys,
                                  const uint8_t* __restrict__ key_valid,
                                  const double* __restrict__ vals,
                                  const uint8_t* __restrict__ val_valid,
-                                 int64_t n, int nbuck_log2,
+                                 int64_t n, int nbuck_log2, int grid_log2,
                                  const uint32_t* __restrict__ scanned,
                                  PartRow* __restrict__ out) {
   extern __shared__ uint32_t lds_cursor[];
   const uint32_t nbuck = 1u << nbuck_log2;
   for (uint32_t b = threadIdx.x; b < nbuck; b += blockDim.x)
-    lds_cursor[b] = scanned[((size_t)b << GRID_LOG2) | blockIdx.x];
+    lds_cursor[b] = scanned[((size_t)b << grid_log2) | blockIdx.x];
   __syncthreads();
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
@@ -431,12 +431,13 @@ __global__ void k_agg2_scatter24(const int64_t* __restrict__ keys,
 
 void launch_agg2_scatter24(const int64_t* keys, const uint8_t* key_valid,
                            const double* vals, const uint8_t* val_valid,
-                           int64_t n, int nbuck_log2, const uint32_t* scanned,
+                           int64_t n, int nbuck_log2, int grid_log2,
+                           const uint32_t* scanned,
                            PartRow* out, hipStream_t s) {
   size_t lds = (size_t)(1u << nbuck_log2) * 4;
-  hipLaunchKernelGGL(k_agg2_scatter24, dim3(1 << GRID_LOG2), dim3(BLOCK), lds,
+  hipLaunchKernelGGL(k_agg2_scatter24, dim3(1 << grid_log2), dim3(BLOCK), lds,
                      s, keys, key_valid, vals, val_valid, n, nbuck_log2,
-                     scanned, out);
+                     grid_log2, scanned, out);
   check_launch2("k_agg2_scatter24");
 }
 
