@@ -386,6 +386,21 @@ class AggOp {
                "aggregates on this path");
       }
       layout_ |= k << (3 * i);
+      // accumulator arithmetic type follows the agg's declared data type
+      // (sum.rs:78-88 casts inputs to it; maxmin.rs:81-83 preserves it)
+      if (k != AGGL_CNT) {
+        bool is_int = (a.return_type == DType::Int64);
+        if (!val_typed_seen_) {
+          val_is_int_ = is_int;
+          val_typed_seen_ = true;
+        } else if (val_is_int_ != is_int) {
+          FAIL("AggExec: mixed i64/f64 accumulator types unsupported "
+               "(shared-argument agg set)");
+        }
+        if (k == AGGL_AVG && is_int)
+          FAIL("AggExec: AVG with integer output unsupported (avg.rs "
+               "declares a floating result)");
+      }
       agg_kinds_.push_back(k);
       agg_names_.push_back(i < node.agg_names.size() ? node.agg_names[i]
                                                      : "agg" + std::to_string(i));
@@ -419,6 +434,7 @@ class AggOp {
       max_cap_ *= 2;
     if (slots > max_cap_) slots = max_cap_;
     init_table(slots);
+    t_.sum_int = val_is_int_ ? 1u : 0u;
     AURON_HIP(hipEventCreate(&ev_start_));
     AURON_HIP(hipEventCreate(&ev_stop_));
   }
@@ -509,7 +525,9 @@ class AggOp {
                                       row_cursor_, layout_, stream_);
       } else {
         const DevColumn& val = b.cols.at(val_col_);
-        if (val.dt != DType::Float64) FAIL("SUM arg must be Float64");
+        const DType want = val_is_int_ ? DType::Int64 : DType::Float64;
+        if (val.dt != want)
+          FAIL("agg arg dtype must match the declared accumulator type");
         // HIP-event timing on the launch stream (roofline evidence for the
         // dominant kernel; pairs are drained once at finish() so the hot
         // loop never synchronizes for timing)
@@ -789,7 +807,7 @@ class AggOp {
     // lo_n at +1, special count at byte offset 8 via the +2 uint32 slot)
     if (agg2_split_)
       launch_agg2_bucket(d_partkv_.get<PartKV>(), d_rowv_.get<uint32_t>(),
-                         d_offsets_.get<uint32_t>(),
+                         d_offsets_.get<uint32_t>(), val_is_int_ ? 1 : 0,
                          nbuck, d_staged_.get<StagedGroup>(),
                          d_counters_.get<unsigned long long>(),
                          (int64_t)nbuck * AGG2_LSLOTS,
@@ -798,7 +816,7 @@ class AggOp {
                          t_.error_flag, stream_);
     else
       launch_agg2_bucket24(d_partkv_.get<PartRow>(),
-                           d_offsets_.get<uint32_t>(),
+                           d_offsets_.get<uint32_t>(), val_is_int_ ? 1 : 0,
                            nbuck, d_staged_.get<StagedGroup>(),
                            d_counters_.get<unsigned long long>(),
                            (int64_t)nbuck * AGG2_LSLOTS,
@@ -1145,23 +1163,28 @@ class AggOp {
       DBG("agg.emit final d2h done");
       attach_validity(&key_col, kv, n);
       cols.push_back(std::move(key_col));
+      const DType vdt = val_is_int_ ? DType::Int64 : DType::Float64;
       for (uint32_t k : agg_kinds_) {
         HostOutCol ac;
         if (k == AGGL_CNT) {
           ac.dt = DType::Int64;
           ac.values = h_cnts;
         } else if (k == AGGL_MIN || k == AGGL_MAX) {
-          ac.dt = DType::Float64;
+          ac.dt = vdt;
           ac.values = (k == AGGL_MIN) ? h_mins : h_maxs;
           attach_validity(&ac, (k == AGGL_MIN) ? mv : xv, n);
         } else if (k == AGGL_FIRST || k == AGGL_FIRSTIN) {
           int w = (k == AGGL_FIRSTIN) ? 1 : 0;
-          ac.dt = DType::Float64;
+          ac.dt = vdt;
           ac.values = h_firsts[w];
           attach_validity(&ac, fv[w], n);
-        } else {
+        } else if (k == AGGL_AVG) {
           ac.dt = DType::Float64;
-          ac.values = (k == AGGL_AVG) ? h_avgs : h_sums;
+          ac.values = h_avgs;
+          attach_validity(&ac, sv, n);
+        } else {
+          ac.dt = vdt;  // SUM output carries the accumulator type's raw bits
+          ac.values = h_sums;
           attach_validity(&ac, sv, n);  // null iff cnt==0 (shared column)
         }
         cols.push_back(std::move(ac));
@@ -1211,7 +1234,7 @@ class AggOp {
     DevBuf data(h_offs[n] ? h_offs[n] : 1);
     launch_skip_freeze_write((const double*)val.values, val.validity, n,
                              offs.get<int32_t>(), data.get<uint8_t>(), layout_,
-                             stream_);
+                             val_is_int_ ? 1 : 0, stream_);
     HostOutCol key_col, buf_col;
     key_col.dt = key.dt;
     size_t kw = dtype_width(key.dt);
@@ -1259,6 +1282,7 @@ class AggOp {
   DType key_dt_ = DType::Unsupported;
   uint32_t layout_ = 0;
   bool has_mm_ = false;  // agg list contains MIN/MAX: side mm array active
+  bool val_is_int_ = false, val_typed_seen_ = false;  // i64 accumulator mode
   bool has_first_ = false;  // FIRST family: f_row/f_val/f_st arrays active
   std::vector<uint32_t> agg_kinds_;
   std::vector<std::string> agg_names_;

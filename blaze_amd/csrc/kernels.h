@@ -84,6 +84,13 @@ struct AggTable {
   double* f_val = nullptr;              // captured value
   uint8_t* f_st = nullptr;              // 0 = untouched, 1 = first-was-null
                                         // (FIRST only), 2 = valid
+  // accumulator arithmetic type (sum.rs:78-88: the acc column is the agg's
+  // declared data type and inputs are cast to it). 0 = f64 (north star);
+  // 1 = i64 — sums use wrapping integer atomicAdd on the same 8-byte slot
+  // field, MIN/MAX use the i64 order map (sign-bit flip), and all
+  // freeze/parse paths move the raw 8 bytes either way, so the wire format
+  // is the typed prim save in both modes.
+  uint32_t sum_int = 0;
 };
 
 // grid-stride row update: keys/vals length n, rows globally numbered starting
@@ -149,12 +156,13 @@ void launch_agg_freeze_write(const AggTable& t, const uint32_t* order_slots,
                              int64_t num_groups, const int32_t* offsets,
                              uint8_t* out, uint32_t layout, hipStream_t s);
 
-// partial-skipping pass-through freeze (agg_ctx.rs:428-462)
+// partial-skipping pass-through freeze (agg_ctx.rs:428-462); is_int selects
+// the i64 order map for MIN/MAX parts (values themselves move as raw bytes)
 void launch_skip_freeze_len(const uint8_t* val_valid, int64_t n, int32_t* lens,
                             uint32_t layout, hipStream_t s);
 void launch_skip_freeze_write(const double* vals, const uint8_t* val_valid,
                               int64_t n, const int32_t* offsets, uint8_t* out,
-                              uint32_t layout, hipStream_t s);
+                              uint32_t layout, int is_int, hipStream_t s);
 
 // initialize table slots (key = empty sentinel, accs zero, first_row = ~0)
 void launch_slots_init(AggSlot* slots, int64_t n, hipStream_t s);
@@ -272,7 +280,7 @@ void launch_agg2_scatter24(const int64_t* keys, const uint8_t* key_valid,
                            const uint32_t* scanned,
                            PartRow* out, hipStream_t s);
 void launch_agg2_bucket24(const PartRow* part, const uint32_t* offsets,
-                          int nbuckets, StagedGroup* staged,
+                          int is_int, int nbuckets, StagedGroup* staged,
                           unsigned long long* staged_n, int64_t staged_cap,
                           PartRow* leftover, unsigned long long* lo_n,
                           uint32_t* error_flag, hipStream_t s);
@@ -281,7 +289,7 @@ void launch_agg2_specials(const AggTable& t, const int64_t* keys,
                           const uint8_t* val_valid, int64_t n,
                           uint64_t row_offset, hipStream_t s);
 void launch_agg2_bucket(const PartKV* part_kv, const uint32_t* part_rowv,
-                        const uint32_t* offsets,
+                        const uint32_t* offsets, int is_int,
                         int nbuckets, StagedGroup* staged,
                         unsigned long long* staged_n, int64_t staged_cap,
                         PartRow* leftover, unsigned long long* lo_n,

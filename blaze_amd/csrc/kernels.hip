@@ -253,6 +253,40 @@ __device__ __forceinline__ double f64_omap_inv(uint64_t u) {
   return x;
 }
 
+// typed order map: i64 mode flips the sign bit (monotone over int64).
+// Sentinel collisions in i64 mode: min acc == i64::MAX maps to ~0 and
+// max acc == i64::MIN maps to 0 (the init sentinels) — a group whose every
+// value is that extreme reads as empty; documented edge like the f64 NaN
+// payloads. `v` carries the 8 raw value bytes through a double register
+// (loads/stores are bit-preserving; no FP arithmetic touches it).
+__device__ __forceinline__ uint64_t val_omap(double v, bool is_int) {
+  uint64_t b;
+  memcpy(&b, &v, 8);
+  if (is_int) return b ^ 0x8000000000000000ull;
+  return (b >> 63) ? ~b : (b | 0x8000000000000000ull);
+}
+__device__ __forceinline__ double val_omap_inv(uint64_t u, bool is_int) {
+  uint64_t b = is_int ? (u ^ 0x8000000000000000ull)
+                      : ((u >> 63) ? (u & 0x7FFFFFFFFFFFFFFFull) : ~u);
+  double x;
+  memcpy(&x, &b, 8);
+  return x;
+}
+
+// typed sum accumulate: i64 mode adds the raw bits with wrapping integer
+// arithmetic (sum.rs release-mode `v + x` wraps), f64 mode uses the native
+// global f64 atomic add
+__device__ __forceinline__ void sum_accum(double* acc, double v, bool is_int) {
+  if (is_int) {
+    uint64_t b;
+    memcpy(&b, &v, 8);
+    atomicAdd(reinterpret_cast<unsigned long long*>(acc),
+              (unsigned long long)b);
+  } else {
+    unsafeAtomicAdd(acc, v);
+  }
+}
+
 // Per-row slot resolution + accumulate (the slow path of the batched kernel
 // below; also correct standalone).
 __device__ __forceinline__ void agg_accum_row(const AggTable t, int64_t key,
@@ -269,10 +303,10 @@ __device__ __forceinline__ void agg_accum_row(const AggTable t, int64_t key,
     // sum.rs:90-115: SUM adds non-null args; valid-ness latches on.
     // sum validity is implied by cnt>0 (same-column agg set; engine.cpp
     // enforces SUM/COUNT share the argument column)
-    unsafeAtomicAdd(&sl->sum, val);
+    sum_accum(&sl->sum, val, t.sum_int);
     atomicAdd(&sl->cnt, 1ull);  // count.rs:90-149: COUNT(arg) non-null
     if (t.mm) {  // maxmin.rs:104-119: MIN/MAX of non-null args
-      uint64_t u = f64_omap(val);
+      uint64_t u = val_omap(val, t.sum_int);
       atomicMin(&t.mm[2 * a], u);
       atomicMax(&t.mm[2 * a + 1], u);
     }
@@ -328,10 +362,10 @@ __global__ void k_agg_update(const AggTable t, const int64_t* __restrict__ keys,
         AggSlot* sl = &t.slots[slot[k]];
         if (sl->first_row > row) atomicMin(&sl->first_row, row);
         if (vvalid[k]) {
-          unsafeAtomicAdd(&sl->sum, val[k]);
+          sum_accum(&sl->sum, val[k], t.sum_int);
           atomicAdd(&sl->cnt, 1ull);
           if (t.mm) {
-            uint64_t u = f64_omap(val[k]);
+            uint64_t u = val_omap(val[k], t.sum_int);
             atomicMin(&t.mm[2 * slot[k]], u);
             atomicMax(&t.mm[2 * slot[k] + 1], u);
           }
@@ -433,7 +467,8 @@ __device__ __forceinline__ uint8_t* agg_prim_freeze_part(bool valid, double v,
 
 __device__ __forceinline__ uint8_t* agg_freeze_write_rec(uint32_t layout,
                                                          const AccSnap& a,
-                                                         uint8_t* p) {
+                                                         uint8_t* p,
+                                                         bool is_int = false) {
   const bool valid = a.valid;
   const double sum = a.sum;
   const uint64_t cnt = a.cnt;
@@ -442,9 +477,11 @@ __device__ __forceinline__ uint8_t* agg_freeze_write_rec(uint32_t layout,
     if (k == 1 || k == 3)  // acc.rs:335-347 prim freeze
       p = agg_prim_freeze_part(valid, sum, p);
     if (k == 4)
-      p = agg_prim_freeze_part(a.minu != MM_MIN_INIT, f64_omap_inv(a.minu), p);
+      p = agg_prim_freeze_part(a.minu != MM_MIN_INIT,
+                               val_omap_inv(a.minu, is_int), p);
     if (k == 5)
-      p = agg_prim_freeze_part(a.maxu != MM_MAX_INIT, f64_omap_inv(a.maxu), p);
+      p = agg_prim_freeze_part(a.maxu != MM_MAX_INIT,
+                               val_omap_inv(a.maxu, is_int), p);
     if (k == 6) {  // FIRST: prim value ++ flag byte (first.rs:315-319)
       p = agg_prim_freeze_part(a.f_st == 2, a.f_val, p);
       *p++ = a.f_st ? 2 : 0;
@@ -468,7 +505,8 @@ __device__ __forceinline__ uint8_t* agg_freeze_write_rec(uint32_t layout,
 // back order-mapped, at their sentinels when the part is absent or invalid —
 // so merging them with atomicMin/atomicMax needs no guard.
 __device__ __forceinline__ void agg_parse_frozen(uint32_t layout,
-                                                 const uint8_t* p, AccSnap* a) {
+                                                 const uint8_t* p, AccSnap* a,
+                                                 bool is_int = false) {
   *a = AccSnap{};
   uint32_t got = 0;  // bit per family
   for (uint32_t l = layout; l & 7u; l >>= 3) {
@@ -485,10 +523,10 @@ __device__ __forceinline__ void agg_parse_frozen(uint32_t layout,
         a->sum = x;
         got |= 1u;
       } else if (k == 4 && !(got & 2u)) {
-        if (v) a->minu = f64_omap(x);
+        if (v) a->minu = val_omap(x, is_int);
         got |= 2u;
       } else if (k == 5 && !(got & 4u)) {
-        if (v) a->maxu = f64_omap(x);
+        if (v) a->maxu = val_omap(x, is_int);
         got |= 4u;
       } else if (k == 6 && !(got & 8u)) {
         a->f_val = x;
@@ -532,8 +570,8 @@ __global__ void k_agg_merge_frozen(const AggTable t,
     uint64_t row = row_offset + (uint64_t)i;
     if (sl->first_row > row) atomicMin(&sl->first_row, row);
     AccSnap acc;
-    agg_parse_frozen(layout, acc_data + acc_offsets[i], &acc);
-    if (acc.valid) unsafeAtomicAdd(&sl->sum, acc.sum);  // sum.rs:117-145
+    agg_parse_frozen(layout, acc_data + acc_offsets[i], &acc, t.sum_int);
+    if (acc.valid) sum_accum(&sl->sum, acc.sum, t.sum_int);  // sum.rs:117-145
     if (acc.cnt) atomicAdd(&sl->cnt, acc.cnt);
     if (t.mm) {  // maxmin.rs:196-216 partial_merge; sentinels are no-ops
       atomicMin(&t.mm[2 * a], acc.minu);
@@ -565,8 +603,8 @@ __global__ void k_agg_merge_spill(const AggTable t,
     unsigned long long fr = first_rows[i];
     if (sl->first_row > fr) atomicMin(&sl->first_row, fr);
     AccSnap acc;
-    agg_parse_frozen(layout, acc_data + off, &acc);
-    if (acc.valid) unsafeAtomicAdd(&sl->sum, acc.sum);
+    agg_parse_frozen(layout, acc_data + off, &acc, t.sum_int);
+    if (acc.valid) sum_accum(&sl->sum, acc.sum, t.sum_int);
     if (acc.cnt) atomicAdd(&sl->cnt, acc.cnt);
     if (t.mm) {
       atomicMin(&t.mm[2 * a], acc.minu);
@@ -623,8 +661,8 @@ __global__ void k_agg_gather_out(const AggTable t,
                                     : (s == t.cap ? KEY_EMPTY : 0);
     if (sums) sums[i] = t.slots[s].sum;
     if (counts) counts[i] = (long long)t.slots[s].cnt;
-    if (mins) mins[i] = f64_omap_inv(t.mm[2 * s]);
-    if (maxs) maxs[i] = f64_omap_inv(t.mm[2 * s + 1]);
+    if (mins) mins[i] = val_omap_inv(t.mm[2 * s], t.sum_int);
+    if (maxs) maxs[i] = val_omap_inv(t.mm[2 * s + 1], t.sum_int);
     if ((i & 7) == 0) {
       uint8_t kb = 0, sb = 0, mb = 0, xb = 0;
       for (int j = 0; j < 8 && i + j < num_groups; j++) {
@@ -671,7 +709,7 @@ __global__ void k_agg_freeze_len(const AggTable t,
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < num_groups;
        i += (int64_t)gridDim.x * blockDim.x) {
     uint32_t s = order_slots[i];
-    lens[i] = agg_freeze_len(layout, table_snap(t, s));
+    lens[i] = agg_freeze_len(layout, table_snap(t, s));  // len is type-free
   }
 }
 
@@ -683,7 +721,8 @@ __global__ void k_agg_freeze_write(const AggTable t,
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < num_groups;
        i += (int64_t)gridDim.x * blockDim.x) {
     uint32_t s = order_slots[i];
-    agg_freeze_write_rec(layout, table_snap(t, s), out + offsets[i]);
+    agg_freeze_write_rec(layout, table_snap(t, s), out + offsets[i],
+                         t.sum_int);
   }
 }
 
@@ -744,12 +783,13 @@ void launch_agg_freeze_write(const AggTable& t, const uint32_t* order_slots,
 // process_partial_skipped): each input row becomes its own single-row group;
 // SUM acc = value when valid, COUNT = valid ? 1 : 0, MIN = MAX = value,
 // FIRST = value-or-null (touched), FIRST_IGNORES_NULL = value when valid.
-__device__ __forceinline__ AccSnap row_snap(bool v, double val) {
+__device__ __forceinline__ AccSnap row_snap(bool v, double val,
+                                            bool is_int = false) {
   AccSnap a;
   a.valid = v;
   a.sum = val;
   a.cnt = v ? 1 : 0;
-  uint64_t u = f64_omap(val);
+  uint64_t u = val_omap(val, is_int);
   a.minu = v ? u : MM_MIN_INIT;
   a.maxu = v ? u : MM_MAX_INIT;
   a.f_st = v ? 2 : 1;  // a row always touches its own group's FIRST
@@ -772,12 +812,13 @@ __global__ void k_skip_freeze_len(const uint8_t* __restrict__ val_valid,
 __global__ void k_skip_freeze_write(const double* __restrict__ vals,
                                     const uint8_t* __restrict__ val_valid,
                                     int64_t n, const int32_t* __restrict__ offsets,
-                                    uint8_t* __restrict__ out, uint32_t layout) {
+                                    uint8_t* __restrict__ out, uint32_t layout,
+                                    int is_int) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     bool v = !val_valid || bit_get_dev(val_valid, i);
-    agg_freeze_write_rec(layout, row_snap(v, v ? vals[i] : 0.0),
-                         out + offsets[i]);
+    agg_freeze_write_rec(layout, row_snap(v, v ? vals[i] : 0.0, is_int),
+                         out + offsets[i], is_int);
   }
 }
 
@@ -789,9 +830,9 @@ void launch_skip_freeze_len(const uint8_t* val_valid, int64_t n, int32_t* lens,
 }
 void launch_skip_freeze_write(const double* vals, const uint8_t* val_valid,
                               int64_t n, const int32_t* offsets, uint8_t* out,
-                              uint32_t layout, hipStream_t s) {
+                              uint32_t layout, int is_int, hipStream_t s) {
   hipLaunchKernelGGL(k_skip_freeze_write, dim3(grid_for(n)), dim3(BLOCK), 0, s,
-                     vals, val_valid, n, offsets, out, layout);
+                     vals, val_valid, n, offsets, out, layout, is_int);
   check_launch("k_skip_freeze_write");
 }
 
@@ -915,7 +956,7 @@ __global__ void k_first_capture_frozen(const AggTable t,
     if (a < 0) continue;
     unsigned long long pr = prio ? prio[i] : row_offset + (uint64_t)i;
     AccSnap acc;
-    agg_parse_frozen(layout, acc_data + acc_offsets[i], &acc);
+    agg_parse_frozen(layout, acc_data + acc_offsets[i], &acc, t.sum_int);
     if (acc.f_st && t.f_row[2 * a] == pr) {
       t.f_val[2 * a] = acc.f_val;
       t.f_st[2 * a] = acc.f_st;

@@ -33,6 +33,20 @@ inline int grid2(int64_t n) {
   return (int)b;
 }
 constexpr int64_t KEY_EMPTY2 = INT64_MIN;
+// typed accumulate (see kernels.hip sum_accum): i64 mode adds raw bits with
+// wrapping integer atomics; value bytes ride through double registers
+__device__ __forceinline__ void sum_accum2(double* acc, double v, bool is_int) {
+  if (is_int) {
+    uint64_t b;
+    memcpy(&b, &v, 8);
+    atomicAdd(reinterpret_cast<unsigned long long*>(acc),
+              (unsigned long long)b);
+  } else {
+    unsafeAtomicAdd(acc, v);
+  }
+}
+
+
 
 __device__ __forceinline__ uint64_t mix64_2(uint64_t x) {
   x += 0x9E3779B97F4A7C15ull;
@@ -145,7 +159,7 @@ __global__ void k_agg2_specials(const AggTable t,
     if (sl->first_row > row) atomicMin(&sl->first_row, row);
     bool vvalid = !val_valid || bit_get2(val_valid, i);
     if (vvalid) {
-      unsafeAtomicAdd(&sl->sum, vals[i]);
+      sum_accum2(&sl->sum, vals[i], t.sum_int);
       atomicAdd(&sl->cnt, 1ull);
     }
   }
@@ -160,7 +174,7 @@ static constexpr int LPROBE = 64;
 
 __global__ void __launch_bounds__(256) k_agg2_bucket(
     const PartKV* __restrict__ part, const uint32_t* __restrict__ part_rowv,
-    const uint32_t* __restrict__ offsets,
+    const uint32_t* __restrict__ offsets, int is_int,
     int nbuckets, StagedGroup* __restrict__ staged,
     unsigned long long* __restrict__ staged_n, int64_t staged_cap,
     PartRow* __restrict__ leftover, unsigned long long* __restrict__ lo_n,
@@ -208,7 +222,7 @@ __global__ void __launch_bounds__(256) k_agg2_bucket(
       if (found >= 0) {
         atomicMin(&ls_first[found], row);
         if (vvalid) {
-          unsafeAtomicAdd(&ls_sum[found], v);
+          sum_accum2(&ls_sum[found], v, is_int);
           atomicAdd(&ls_cnt[found], 1u);
         }
       } else {
@@ -278,7 +292,7 @@ __global__ void k_agg2_merge_groups(const AggTable t,
     if (sl->first_row > row) atomicMin(&sl->first_row, row);
     uint32_t cnt = (uint32_t)(cf >> 32);
     if (cnt) {  // cnt==0: group seen only via null values — key+order only
-      unsafeAtomicAdd(&sl->sum, staged[i].sum);
+      sum_accum2(&sl->sum, staged[i].sum, t.sum_int);
       atomicAdd(&sl->cnt, (unsigned long long)cnt);
     }
   }
@@ -326,7 +340,7 @@ __global__ void k_agg2_leftovers(const AggTable t,
     uint64_t row = row_offset + (rv & 0x7FFFFFFFu);
     if (sl->first_row > row) atomicMin(&sl->first_row, row);
     if (rv & 0x80000000u) {
-      unsafeAtomicAdd(&sl->sum, rows[i].val);
+      sum_accum2(&sl->sum, rows[i].val, t.sum_int);
       atomicAdd(&sl->cnt, 1ull);
     }
   }
@@ -386,14 +400,14 @@ void launch_agg2_specials(const AggTable& t, const int64_t* keys,
 }
 
 void launch_agg2_bucket(const PartKV* part_kv, const uint32_t* part_rowv,
-                        const uint32_t* offsets,
+                        const uint32_t* offsets, int is_int,
                         int nbuckets, StagedGroup* staged,
                         unsigned long long* staged_n, int64_t staged_cap,
                         PartRow* leftover, unsigned long long* lo_n,
                         uint32_t* error_flag, hipStream_t s) {
   int blocks = nbuckets < (int)MAX_BLOCKS ? nbuckets : (int)MAX_BLOCKS;
   hipLaunchKernelGGL(k_agg2_bucket, dim3(blocks), dim3(BLOCK), 0, s, part_kv,
-                     part_rowv, offsets, nbuckets, staged, staged_n,
+                     part_rowv, offsets, is_int, nbuckets, staged, staged_n,
                      staged_cap, leftover, lo_n, error_flag);
   check_launch2("k_agg2_bucket");
 }
@@ -443,7 +457,7 @@ void launch_agg2_scatter24(const int64_t* keys, const uint8_t* key_valid,
 
 __global__ void __launch_bounds__(256) k_agg2_bucket24(
     const PartRow* __restrict__ part, const uint32_t* __restrict__ offsets,
-    int nbuckets, StagedGroup* __restrict__ staged,
+    int is_int, int nbuckets, StagedGroup* __restrict__ staged,
     unsigned long long* __restrict__ staged_n, int64_t staged_cap,
     PartRow* __restrict__ leftover, unsigned long long* __restrict__ lo_n,
     uint32_t* __restrict__ error_flag) {
@@ -488,7 +502,7 @@ __global__ void __launch_bounds__(256) k_agg2_bucket24(
       if (found >= 0) {
         atomicMin(&ls_first[found], row);
         if (vvalid) {
-          unsafeAtomicAdd(&ls_sum[found], v);
+          sum_accum2(&ls_sum[found], v, is_int);
           atomicAdd(&ls_cnt[found], 1u);
         }
       } else {
@@ -514,14 +528,14 @@ __global__ void __launch_bounds__(256) k_agg2_bucket24(
 }
 
 void launch_agg2_bucket24(const PartRow* part, const uint32_t* offsets,
-                          int nbuckets, StagedGroup* staged,
+                          int is_int, int nbuckets, StagedGroup* staged,
                           unsigned long long* staged_n, int64_t staged_cap,
                           PartRow* leftover, unsigned long long* lo_n,
                           uint32_t* error_flag, hipStream_t s) {
   int blocks = nbuckets < (int)MAX_BLOCKS ? nbuckets : (int)MAX_BLOCKS;
   hipLaunchKernelGGL(k_agg2_bucket24, dim3(blocks), dim3(BLOCK), 0, s, part,
-                     offsets, nbuckets, staged, staged_n, staged_cap, leftover,
-                     lo_n, error_flag);
+                     offsets, is_int, nbuckets, staged, staged_n, staged_cap,
+                     leftover, lo_n, error_flag);
   check_launch2("k_agg2_bucket24");
 }
 

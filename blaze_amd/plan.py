@@ -146,26 +146,30 @@ _AGG_FN = {"min": AGG_MIN, "max": AGG_MAX, "sum": AGG_SUM,
            "first_ignores_null": AGG_FIRST_IGNORES_NULL}
 
 
-def named_aggs(names, val_index=1):
-    """Agg exprs for a list of function names over the shared val column
-    (MIN/MAX return the input type — maxmin.rs:81-83 data_type())."""
+def named_aggs(names, val_index=1, val_dt=DT_FLOAT64):
+    """Agg exprs for a list of function names over the shared val column.
+    val_dt is the declared accumulator/output type (sum.rs:78-88 casts the
+    input to it; MIN/MAX preserve it — maxmin.rs:81-83)."""
     out = []
     for nm in names:
         fn = _AGG_FN[nm]
-        dt = DT_INT64 if fn == AGG_COUNT else DT_FLOAT64
+        dt = DT_INT64 if fn == AGG_COUNT else val_dt
         out.append(agg_expr(fn, [column("val", val_index)], dt))
     return out
 
 
-def plan_partial_final_named(agg_fns, resource_id="input0", key_dt=DT_INT64):
+def plan_partial_final_named(agg_fns, resource_id="input0", key_dt=DT_INT64,
+                             val_dt=DT_FLOAT64):
     """FFIReader -> Agg(Partial) -> Agg(Final) with an arbitrary agg list
     (e.g. ["min", "max", "sum", "count", "avg"]) over the shared val column."""
     reader = ffi_reader([field("key", key_dt, False),
-                         field("val", DT_FLOAT64, True)], resource_id)
+                         field("val", val_dt, True)], resource_id)
     names = list(agg_fns)
-    partial = agg(reader, [column("key", 0)], named_aggs(names),
+    partial = agg(reader, [column("key", 0)],
+                  named_aggs(names, val_dt=val_dt),
                   [MODE_PARTIAL] * len(names), ["key"], names)
-    final = agg(partial, [column("key", 0)], named_aggs(names),
+    final = agg(partial, [column("key", 0)],
+                named_aggs(names, val_dt=val_dt),
                 [MODE_FINAL] * len(names), ["key"], names)
     return task_definition(final)
 

@@ -304,7 +304,7 @@ def minmax_groups(keys, vals, val_valid=None, key_valid=None):
             groups[k] = [None, None]
         vv = val_valid is None or bool(val_valid[i])
         if vv:
-            v = float(vals[i])
+            v = vals[i]  # type-preserving: np.int64 stays exact (i64 mode)
             g = groups[k]
             g[0] = v if g[0] is None or v < g[0] else g[0]
             g[1] = v if g[1] is None or v > g[1] else g[1]
@@ -351,3 +351,25 @@ def first_groups(keys, vals, val_valid=None, key_valid=None):
     firsts = [(groups[k][0], groups[k][1]) for k in ordered]
     firsts_nn = [groups[k][2] for k in ordered]
     return ordered, firsts, firsts_nn
+
+
+# ---- typed (Int64) aggregation restatement ---------------------------------
+# sum.rs:78-88: SUM's accumulator column is the agg's declared data type and
+# inputs are cast to it; for bigint that is i64 with WRAPPING addition
+# (release-mode Rust `v + x`). numpy int64 adds wrap the same way.
+def int_sum_groups(keys, vals, val_valid=None):
+    """Insertion-ordered (keys, sums, counts) with np.int64 wrapping sums."""
+    import numpy as np
+    groups = {}
+    n = len(keys)
+    for i in range(n):
+        k = int(keys[i])
+        if k not in groups:
+            groups[k] = [np.int64(0), 0]
+        if val_valid is None or val_valid[i]:
+            with np.errstate(over="ignore"):
+                groups[k][0] = np.int64(groups[k][0] + np.int64(vals[i]))
+            groups[k][1] += 1
+    ordered = list(groups.keys())
+    return (ordered, [groups[k][0] for k in ordered],
+            [groups[k][1] for k in ordered])

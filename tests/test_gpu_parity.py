@@ -637,3 +637,68 @@ def test_first_spill_grow_and_skipping():
     np.testing.assert_array_equal(got2["first"][0][gi2][vv2], vals2[vv2])
     np.testing.assert_array_equal(got2["first_ignores_null"][1][gi2], vv2)
     t2.finalize()
+
+
+def test_int64_typed_aggregation():
+    """i64 accumulator mode (sum.rs:78-88: acc type = declared data type):
+    SUM over bigint is bit-exact WRAPPING i64 arithmetic, MIN/MAX use the
+    i64 order, outputs are Int64 columns. Includes values large enough that
+    f64 accumulation would lose precision (> 2^53)."""
+    rng = np.random.default_rng(17)
+    n = 200_000
+    keys = rng.integers(0, 3000, n).astype(np.int64)
+    # values near 2^60: three adds exceed 2^62, wrapping differences from
+    # f64 rounding would be visible immediately
+    vals = (rng.integers(0, 1 << 60, n) - (1 << 59)).astype(np.int64)
+    vv = rng.random(n) >= 0.05
+    names = ["sum", "min", "max", "count"]
+    t = blaze_amd.Task(
+        plan.plan_partial_final_named(names, val_dt=plan.DT_INT64),
+        batches=batches_of(keys, vals, vv))
+    outs = t.run()
+    got = _minmax_cols(outs, names)
+    ok, sums, cnts = oracle.int_sum_groups(keys, vals, vv)
+    np.testing.assert_array_equal(got["key"][0], np.array(ok))
+    assert got["sum"][0].dtype == np.int64
+    np.testing.assert_array_equal(got["sum"][0][got["sum"][1]],
+                                  np.array(sums, np.int64)[got["sum"][1]])
+    np.testing.assert_array_equal(got["count"][0], np.array(cnts))
+    omin, omax = oracle.minmax_groups(keys, vals, vv)[1:3]
+    exp_valid = np.array([m is not None for m in omin])
+    np.testing.assert_array_equal(got["min"][1], exp_valid)
+    assert got["min"][0].dtype == np.int64
+    np.testing.assert_array_equal(
+        got["min"][0][exp_valid],
+        np.array([m for m in omin if m is not None], np.int64))
+    np.testing.assert_array_equal(
+        got["max"][0][exp_valid],
+        np.array([m for m in omax if m is not None], np.int64))
+    t.finalize()
+
+
+def test_int64_sum_two_phase_1m_rows():
+    """The two-phase (LDS bucket) path in i64 mode: large single batch so
+    chunks exceed AGG2_MIN_CHUNK; wrapping sums must stay bit-exact."""
+    rng = np.random.default_rng(23)
+    n = 6_000_000
+    keys = rng.integers(0, 50_000, n).astype(np.int64)
+    vals = rng.integers(-(1 << 50), 1 << 50, n).astype(np.int64)
+    names = ["sum", "count"]
+    t = blaze_amd.Task(
+        plan.plan_partial_final_named(names, val_dt=plan.DT_INT64),
+        batches=[[(keys, None), (vals, None)]])
+    outs = t.run()
+    got = _minmax_cols(outs, names)
+    # ground truth via numpy bincount-style aggregation (wrapping)
+    order = {}
+    for k in keys:
+        if int(k) not in order:
+            order[int(k)] = len(order)
+    exp_keys = np.array(sorted(order, key=order.get), np.int64)
+    with np.errstate(over="ignore"):
+        exp_sums = np.zeros(len(order), np.int64)
+        np.add.at(exp_sums, np.array([order[int(k)] for k in keys]), vals)
+    np.testing.assert_array_equal(got["key"][0], exp_keys)
+    np.testing.assert_array_equal(got["sum"][0], exp_sums)
+    assert got["count"][0].sum() == n
+    t.finalize()
