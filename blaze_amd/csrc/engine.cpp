@@ -594,11 +594,14 @@ class AggOp {
     DType kd = key_dt_ == DType::Unsupported ? DType::Int64 : key_dt_;
     if (final_output_) {
       std::vector<OutField> f = {{key_name_, kd, true}};
+      const DType vdt = val_is_int_ ? DType::Int64 : DType::Float64;
       for (size_t i = 0; i < agg_kinds_.size(); i++) {
         if (agg_kinds_[i] == AGGL_CNT)
           f.push_back({agg_names_[i], DType::Int64, false});
-        else
+        else if (agg_kinds_[i] == AGGL_AVG)
           f.push_back({agg_names_[i], DType::Float64, true});
+        else
+          f.push_back({agg_names_[i], vdt, true});
       }
       return f;
     }
@@ -2550,7 +2553,8 @@ int32_t auron_debug_decode_plan(const uint8_t* data, size_t len, char* out,
                ",ngroup=" + std::to_string(ag.grouping_exprs.size()) +
                ",nagg=" + std::to_string(ag.agg_exprs.size());
           for (const auto& e : ag.agg_exprs)
-            s += ",fn" + std::to_string(e.agg_function);
+            s += ",fn" + std::to_string(e.agg_function) + "rt" +
+                 std::to_string((int)e.return_type);
           s += ",skip=" + std::to_string((int)ag.supports_partial_skipping) +
                ")->";
           p = ag.input.get();
