@@ -28,7 +28,7 @@ def test_decode_partial_final_plan():
     s = blaze_amd.debug_decode_plan(td)
     assert "Agg(mode=2" in s           # Final on top
     assert "Agg(mode=0" in s           # Partial below
-    assert ",fn2,fn4," in s or ",fn2,fn4)" in s  # SUM=2, COUNT=4
+    assert ",fn2rt13,fn4rt10" in s  # SUM=2 (f64 out), COUNT=4 (i64 out)
     assert "FFIReader(nfields=2,rid=input0)" in s
 
 
