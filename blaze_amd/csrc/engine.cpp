@@ -465,6 +465,30 @@ class AggOp {
       key.values = key.own_values.get();
       key.dt = DType::Int64;
     }
+    if (!merge_mode_) {
+      // sum.rs:78-88 prepare_partial_args: the argument is CAST to the
+      // accumulator type before update. Narrower numeric args widen on
+      // device; anything else fails loudly.
+      DevColumn& val = b.cols.at(val_col_);
+      const DType want = val_is_int_ ? DType::Int64 : DType::Float64;
+      if (val.dt != want) {
+        DevBuf wide(b.num_rows * 8);
+        if (val.dt == DType::Int32 && want == DType::Int64)
+          launch_widen_i32_i64((const int32_t*)val.values, b.num_rows,
+                               wide.get<int64_t>(), stream_);
+        else if (val.dt == DType::Int32 && want == DType::Float64)
+          launch_cast_i32_f64((const int32_t*)val.values, b.num_rows,
+                              wide.get<double>(), stream_);
+        else if (val.dt == DType::Int64 && want == DType::Float64)
+          launch_cast_i64_f64((const int64_t*)val.values, b.num_rows,
+                              wide.get<double>(), stream_);
+        else
+          FAIL("unsupported agg argument cast");
+        val.own_values = std::move(wide);
+        val.values = val.own_values.get();
+        val.dt = want;
+      }
+    }
     if (skipping_) {
       skipped_.push_back(std::move(b));
       return;
@@ -527,7 +551,8 @@ class AggOp {
         const DevColumn& val = b.cols.at(val_col_);
         const DType want = val_is_int_ ? DType::Int64 : DType::Float64;
         if (val.dt != want)
-          FAIL("agg arg dtype must match the declared accumulator type");
+          FAIL("agg arg dtype must match the declared accumulator type "
+               "(narrower args are widened at batch import)");
         // HIP-event timing on the launch stream (roofline evidence for the
         // dominant kernel; pairs are drained once at finish() so the hot
         // loop never synchronizes for timing)

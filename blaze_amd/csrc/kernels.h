@@ -23,6 +23,11 @@ void launch_widen_i32_i64(const int32_t* in, int64_t n, int64_t* out,
                           hipStream_t s);
 void launch_narrow_i64_i32(const int64_t* in, int64_t n, int32_t* out,
                            hipStream_t s);
+// numeric agg-argument casts (sum.rs:78-88 prepare_partial_args analog)
+void launch_cast_i32_f64(const int32_t* in, int64_t n, double* out,
+                         hipStream_t s);
+void launch_cast_i64_f64(const int64_t* in, int64_t n, double* out,
+                         hipStream_t s);
 // round-robin partition ids (evaluate_robin_partition_ids,
 // shuffle/mod.rs:190-202): part[i] = (start + i) % P
 void launch_robin_ids(int64_t n, uint32_t start, uint32_t P, uint32_t* out,

@@ -1322,4 +1322,32 @@ void launch_scatter_packed(int width, const uint8_t* packed,
   check_launch("k_scatter_packed");
 }
 
+__global__ void k_cast_i32_f64(const int32_t* __restrict__ in, int64_t n,
+                               double* __restrict__ out) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    out[i] = (double)in[i];
+}
+
+void launch_cast_i32_f64(const int32_t* in, int64_t n, double* out,
+                         hipStream_t s) {
+  hipLaunchKernelGGL(k_cast_i32_f64, dim3(grid_for(n)), dim3(BLOCK), 0, s, in,
+                     n, out);
+  check_launch("k_cast_i32_f64");
+}
+
+__global__ void k_cast_i64_f64(const int64_t* __restrict__ in, int64_t n,
+                               double* __restrict__ out) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    out[i] = (double)in[i];
+}
+
+void launch_cast_i64_f64(const int64_t* in, int64_t n, double* out,
+                         hipStream_t s) {
+  hipLaunchKernelGGL(k_cast_i64_f64, dim3(grid_for(n)), dim3(BLOCK), 0, s, in,
+                     n, out);
+  check_launch("k_cast_i64_f64");
+}
+
 }  // namespace auron

@@ -702,3 +702,51 @@ def test_int64_sum_two_phase_1m_rows():
     np.testing.assert_array_equal(got["sum"][0], exp_sums)
     assert got["count"][0].sum() == n
     t.finalize()
+
+
+def test_agg_argument_casts():
+    """sum.rs:78-88 prepare_partial_args: the argument column is cast to the
+    accumulator type. Int32 input with f64 aggs, and Int32 input with i64
+    aggs, both via device-side widening."""
+    rng = np.random.default_rng(31)
+    n = 100_000
+    keys = rng.integers(0, 1000, n).astype(np.int64)
+    vals32 = rng.integers(-(1 << 30), 1 << 30, n).astype(np.int32)
+    vv = rng.random(n) >= 0.1
+
+    # i32 arg -> f64 accs
+    names = ["sum", "count"]
+    reader = plan.ffi_reader([plan.field("key", plan.DT_INT64, False),
+                              plan.field("val", plan.DT_INT32, True)],
+                             "input0")
+    partial = plan.agg(reader, [plan.column("key", 0)], plan.named_aggs(names),
+                       [plan.MODE_PARTIAL] * 2, ["key"], names)
+    final = plan.agg(partial, [plan.column("key", 0)], plan.named_aggs(names),
+                     [plan.MODE_FINAL] * 2, ["key"], names)
+    t = blaze_amd.Task(plan.task_definition(final),
+                       batches=batches_of(keys, vals32, vv))
+    outs = t.run()
+    got = _minmax_cols(outs, names)
+    ref = run_oracle(keys, vals32.astype(np.float64), vv).output()
+    np.testing.assert_array_equal(got["key"][0], ref["keys"])
+    np.testing.assert_array_equal(got["sum"][0][got["sum"][1]],
+                                  ref["sums"][ref["sum_valid"]])
+    t.finalize()
+
+    # i32 arg -> i64 accs (SUM(int) returns bigint in Spark)
+    partial2 = plan.agg(reader, [plan.column("key", 0)],
+                        plan.named_aggs(names, val_dt=plan.DT_INT64),
+                        [plan.MODE_PARTIAL] * 2, ["key"], names)
+    final2 = plan.agg(partial2, [plan.column("key", 0)],
+                      plan.named_aggs(names, val_dt=plan.DT_INT64),
+                      [plan.MODE_FINAL] * 2, ["key"], names)
+    t2 = blaze_amd.Task(plan.task_definition(final2),
+                        batches=batches_of(keys, vals32, vv))
+    outs2 = t2.run()
+    got2 = _minmax_cols(outs2, names)
+    ok, sums, cnts = oracle.int_sum_groups(keys, vals32.astype(np.int64), vv)
+    np.testing.assert_array_equal(got2["key"][0], np.array(ok))
+    assert got2["sum"][0].dtype == np.int64
+    np.testing.assert_array_equal(got2["sum"][0][got2["sum"][1]],
+                                  np.array(sums, np.int64)[got2["sum"][1]])
+    t2.finalize()
