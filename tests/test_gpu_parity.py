@@ -750,3 +750,81 @@ def test_agg_argument_casts():
     np.testing.assert_array_equal(got2["sum"][0][got2["sum"][1]],
                                   np.array(sums, np.int64)[got2["sum"][1]])
     t2.finalize()
+
+
+@pytest.mark.parametrize("seed", [101, 202, 303, 404])
+def test_agg_fuzz_randomized(seed):
+    """Randomized end-to-end parity in the spirit of the reference's own
+    agg fuzz test (agg_exec.rs:714-843): random agg lists, accumulator
+    modes, cardinalities, null fractions, batch sizes and table confs, each
+    checked against the python/numpy oracle restatements."""
+    rng = np.random.default_rng(seed)
+    n = int(rng.integers(5_000, 120_000))
+    nkeys = int(rng.integers(10, max(11, n // 2)))
+    null_frac = float(rng.choice([0.0, 0.01, 0.3]))
+    batch = int(rng.choice([77, 1000, 8192]))
+    int_mode = bool(rng.integers(0, 2))
+    pool = ["sum", "count", "min", "max", "first", "first_ignores_null"] + \
+        ([] if int_mode else ["avg"])
+    names = list(rng.choice(pool, size=int(rng.integers(1, 5)), replace=False))
+    keys = rng.integers(0, nkeys, n).astype(np.int64)
+    if int_mode:
+        vals = rng.integers(-(1 << 40), 1 << 40, n).astype(np.int64)
+    else:
+        vals = (rng.random(n) * 1e6 - 5e5)
+    vv = rng.random(n) >= null_frac
+    conf = {"BATCH_SIZE": batch}
+    if rng.integers(0, 2):
+        conf["AURON_HIP_AGG_TABLE_SLOTS"] = 1 << 10  # force grows
+    t = blaze_amd.Task(
+        plan.plan_partial_final_named(
+            names, val_dt=plan.DT_INT64 if int_mode else plan.DT_FLOAT64),
+        batches=batches_of(keys, vals, vv, batch=max(512, batch)),
+        conf=conf)
+    outs = t.run()
+    got = _minmax_cols(outs, names)
+    ok, mins, maxs = oracle.minmax_groups(keys, vals, vv)
+    np.testing.assert_array_equal(got["key"][0], np.array(ok))
+    fok, firsts, firsts_nn = oracle.first_groups(keys, vals, vv)
+    for nm in names:
+        v, valid = got[nm]
+        if nm == "count":
+            _, _, cnts = oracle.int_sum_groups(keys, np.zeros(n), vv)
+            np.testing.assert_array_equal(v, np.array(cnts))
+        elif nm == "sum":
+            if int_mode:
+                _, sums, _ = oracle.int_sum_groups(keys, vals, vv)
+                np.testing.assert_array_equal(v[valid],
+                                              np.array(sums, np.int64)[valid])
+            else:
+                agg = np.zeros(len(ok))
+                idx = {k: i for i, k in enumerate(ok)}
+                np.add.at(agg, [idx[int(k)] for k in keys[vv]], vals[vv])
+                np.testing.assert_allclose(v[valid], agg[valid], rtol=1e-9)
+        elif nm == "avg":
+            _, _, cnts = oracle.int_sum_groups(keys, np.zeros(n), vv)
+            agg = np.zeros(len(ok))
+            idx = {k: i for i, k in enumerate(ok)}
+            np.add.at(agg, [idx[int(k)] for k in keys[vv]], vals[vv])
+            cnts = np.array(cnts, float)
+            exp = np.divide(agg, np.maximum(cnts, 1))
+            np.testing.assert_allclose(v[valid], exp[valid], rtol=1e-9)
+        elif nm in ("min", "max"):
+            src = mins if nm == "min" else maxs
+            exp_valid = np.array([m is not None for m in src])
+            np.testing.assert_array_equal(valid, exp_valid)
+            np.testing.assert_array_equal(
+                v[exp_valid], np.array([m for m in src if m is not None]))
+        elif nm == "first":
+            exp_valid = np.array([f[1] is not None for f in firsts])
+            np.testing.assert_array_equal(valid, exp_valid)
+            np.testing.assert_array_equal(
+                v[exp_valid],
+                np.array([f[1] for f in firsts if f[1] is not None]))
+        elif nm == "first_ignores_null":
+            exp_valid = np.array([f is not None for f in firsts_nn])
+            np.testing.assert_array_equal(valid, exp_valid)
+            np.testing.assert_array_equal(
+                v[exp_valid],
+                np.array([f for f in firsts_nn if f is not None]))
+    t.finalize()
