@@ -136,14 +136,12 @@ DevColumn import_child(const ArrowArray* a, DType dt, bool device,
       int32_t dl = off_src[a->length];
       c.data_len = dl;
       c.own_offsets.alloc((a->length + 1) * 4);
-      AURON_HIP(hipMemcpyAsync(c.own_offsets.get(), off_src,
-                               (a->length + 1) * 4, hipMemcpyHostToDevice,
-                               stream));
+      PinnedUploader::inst().copy(c.own_offsets.get(), off_src,
+                                  (a->length + 1) * 4, stream);
       c.offsets = c.own_offsets.get<int32_t>();
       if (dl > 0) {
         c.own_values.alloc(dl);
-        AURON_HIP(hipMemcpyAsync(c.own_values.get(), data_src, dl,
-                                 hipMemcpyHostToDevice, stream));
+        PinnedUploader::inst().copy(c.own_values.get(), data_src, dl, stream);
       }
       c.values = c.own_values.get();
     }
@@ -154,8 +152,8 @@ DevColumn import_child(const ArrowArray* a, DType dt, bool device,
       c.values = val_src;
     } else {
       c.own_values.alloc(a->length * w);
-      AURON_HIP(hipMemcpyAsync(c.own_values.get(), val_src, a->length * w,
-                               hipMemcpyHostToDevice, stream));
+      PinnedUploader::inst().copy(c.own_values.get(), val_src,
+                                  (size_t)a->length * w, stream);
       c.values = c.own_values.get();
     }
   }
@@ -2131,9 +2129,9 @@ struct Runtime {
             c.data_len = (int64_t)cd.bin_data.size();
             c.own_values.alloc(cd.bin_data.empty() ? 1 : cd.bin_data.size());
             if (!cd.bin_data.empty())
-              AURON_HIP(hipMemcpyAsync(c.own_values.get(), cd.bin_data.data(),
-                                       cd.bin_data.size(),
-                                       hipMemcpyHostToDevice, stream));
+              PinnedUploader::inst().copy(c.own_values.get(),
+                                          cd.bin_data.data(),
+                                          cd.bin_data.size(), stream);
             c.values = c.own_values.get();
             if (has_nulls) {
               c.own_validity.alloc(cd.validity.size());
@@ -2169,15 +2167,13 @@ struct Runtime {
           if (!cd.uses_dict) {
             // PLAIN: non-null values packed densely
             if (!has_nulls) {
-              AURON_HIP(hipMemcpyAsync(c.own_values.get(), cd.plain.data(),
-                                       cd.plain.size(), hipMemcpyHostToDevice,
-                                       stream));
+              PinnedUploader::inst().copy(c.own_values.get(), cd.plain.data(),
+                                          cd.plain.size(), stream);
             } else {
               d_packed.alloc(cd.plain.empty() ? 1 : cd.plain.size());
               if (!cd.plain.empty())
-                AURON_HIP(hipMemcpyAsync(d_packed.get(), cd.plain.data(),
-                                         cd.plain.size(),
-                                         hipMemcpyHostToDevice, stream));
+                PinnedUploader::inst().copy(d_packed.get(), cd.plain.data(),
+                                            cd.plain.size(), stream);
               launch_scatter_packed(w, d_packed.get<uint8_t>(),
                                     d_positions.get<uint32_t>(),
                                     d_mask.get<uint8_t>(), rows,
@@ -2190,10 +2186,9 @@ struct Runtime {
                                     ? 4
                                     : cd.dict_indices.size() * 4);
             if (!cd.dict_indices.empty())
-              AURON_HIP(hipMemcpyAsync(d_idx_packed.get(),
-                                       cd.dict_indices.data(),
-                                       cd.dict_indices.size() * 4,
-                                       hipMemcpyHostToDevice, stream));
+              PinnedUploader::inst().copy(d_idx_packed.get(),
+                                          cd.dict_indices.data(),
+                                          cd.dict_indices.size() * 4, stream);
             if (has_nulls) {
               launch_scatter_packed(4, d_idx_packed.get<uint8_t>(),
                                     d_positions.get<uint32_t>(),
@@ -2346,9 +2341,8 @@ struct Runtime {
         c.values = c.own_values.get();
       } else {
         c.own_values.alloc(h.values.size());
-        AURON_HIP(hipMemcpyAsync(c.own_values.get(), h.values.data(),
-                                 h.values.size(), hipMemcpyHostToDevice,
-                                 stream));
+        PinnedUploader::inst().copy(c.own_values.get(), h.values.data(),
+                                    h.values.size(), stream);
         c.values = c.own_values.get();
       }
       if (!h.validity.empty()) {
