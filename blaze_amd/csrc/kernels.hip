@@ -241,18 +241,6 @@ __device__ __forceinline__ int64_t agg_upsert_slot(const AggTable t, int64_t key
 // order-preserving f64 <-> u64 map: monotone, so u64 atomicMin/atomicMax
 // implement f64 min/max (maxmin.rs:104-119 compare-and-keep semantics for
 // all comparable values; see kernels.h on the NaN-sentinel edge)
-__device__ __forceinline__ uint64_t f64_omap(double x) {
-  uint64_t b;
-  memcpy(&b, &x, 8);
-  return (b >> 63) ? ~b : (b | 0x8000000000000000ull);
-}
-__device__ __forceinline__ double f64_omap_inv(uint64_t u) {
-  uint64_t b = (u >> 63) ? (u & 0x7FFFFFFFFFFFFFFFull) : ~u;
-  double x;
-  memcpy(&x, &b, 8);
-  return x;
-}
-
 // typed order map: i64 mode flips the sign bit (monotone over int64).
 // Sentinel collisions in i64 mode: min acc == i64::MAX maps to ~0 and
 // max acc == i64::MIN maps to 0 (the init sentinels) — a group whose every
