@@ -178,6 +178,12 @@ class PinnedBuf {
       size_ = 0;
     }
   }
+  // forget the allocation without freeing (static-destruction teardown:
+  // hipHostFree after HIP runtime exit crashes)
+  void* release() {
+    size_ = 0;
+    return std::exchange(ptr_, nullptr);
+  }
   template <typename T = void>
   T* get() const {
     return static_cast<T*>(ptr_);
@@ -199,6 +205,15 @@ class PinnedUploader {
   static PinnedUploader& inst() {
     static PinnedUploader u;
     return u;
+  }
+
+  // the singleton outlives every engine object and is destroyed during
+  // static teardown, possibly after the HIP runtime has shut down —
+  // hipHostFree/hipEventDestroy there segfault (observed as a core dump at
+  // smoke()'s exit). Leak the two 16 MB staging buffers instead; the
+  // process is exiting.
+  ~PinnedUploader() {
+    for (auto& b : bufs_) (void)b.pin.release();
   }
 
   void copy(void* dst, const void* src, size_t len, hipStream_t s) {
