@@ -116,7 +116,10 @@ def run_step(ba, plan, dev_input, world, rank, torch, dist, device, stats):
     # stage 1: partial hash-agg on the HBM-resident slice.
     # BATCH_SIZE is the reference's own conf knob (conf.rs:32); raised here so
     # the ~1M-group output is emitted in few chunks instead of 100 × 10k.
-    conf = {"BATCH_SIZE": 1 << 20}
+    # Table sized for the 1M-group workload (2^22 slots at <=3/4 load; the
+    # engine grows 4x on overflow, so this is a starting size, not a cap) —
+    # the default 2^23 doubles slot-scan and compact time.
+    conf = {"BATCH_SIZE": 1 << 20, "AURON_HIP_AGG_TABLE_SLOTS": 1 << 22}
 
     def mark(name, t0):
         dt = time.perf_counter() - t0
@@ -148,7 +151,9 @@ def run_step(ba, plan, dev_input, world, rank, torch, dist, device, stats):
 
     # stage 3: final merge agg of (local + received) partial records
     offs = np.concatenate([[0], np.cumsum(lens)]).astype(np.int32)
-    t2 = ba.Task(plan.plan_final_only(), conf={"BATCH_SIZE": 1 << 20},
+    t2 = ba.Task(plan.plan_final_only(),
+                 conf={"BATCH_SIZE": 1 << 20,
+                       "AURON_HIP_AGG_TABLE_SLOTS": 1 << 22},
                  batches=[[(keys, None), ("binary", data, offs, None)]])
     out2 = t2.run()
     tm = mark("s3_final", tm)
