@@ -1058,7 +1058,8 @@ class AggOp {
 
   void ensure_capacity(int64_t need) {
     while (t_.cap * 3 / 4 < need && t_.cap < max_cap_) grow(t_.cap * 4);
-    if (t_.cap * 3 / 4 < need && num_groups_host() > specials_count_) {
+    if (t_.cap * 3 / 4 < need &&
+        (int64_t)num_groups_host() > specials_count_) {
       spill_table();  // frees the main region; caller re-checks free space
     }
   }

@@ -114,8 +114,9 @@ typedef struct AuronCallbacks {
   void (*set_error)(void* user, const char* message);
 
   /* IpcReaderExecNode input (auron.proto:607-611; ipc_reader_exec.rs:62-120):
-   * pull the next raw shuffle block-stream segment ([u32-LE len][lz4 frame]*
-   * bytes, the format ShuffleWriter emits). Return 1 and set *data/*len
+   * pull the next raw shuffle block-stream segment (repeated [u32-LE
+   * len][lz4 frame] records, the format ShuffleWriter emits). Return 1 and
+   * set *data/*len
    * (caller-owned, valid until the next call or finalize), 0 on exhaustion.
    * May be NULL when the plan holds no IpcReaderExec. */
   int (*next_ipc_bytes)(void* user, const char* resource_id,
