@@ -330,6 +330,32 @@ def ipc_reader(fields, resource_id, num_partitions=1):
     return _len_field(3, node)
 
 
+def plan_agg_shuffle_named(data_file, index_file, agg_fns,
+                           num_partitions=200, resource_id="input0",
+                           val_dt=DT_FLOAT64, partition_id=0):
+    """FFIReader -> Agg(Partial, arbitrary agg list) -> ShuffleWriter."""
+    names = list(agg_fns)
+    reader = ffi_reader([field("key", DT_INT64, False),
+                         field("val", val_dt, True)], resource_id)
+    partial = agg(reader, [column("key", 0)],
+                  named_aggs(names, val_dt=val_dt),
+                  [MODE_PARTIAL] * len(names), ["key"], names)
+    rep = hash_repartition([column("key", 0)], num_partitions)
+    sw = shuffle_writer(partial, rep, data_file, index_file)
+    return task_definition(sw, partition_id=partition_id)
+
+
+def plan_ipc_final_named(agg_fns, resource_id="ipc0", val_dt=DT_FLOAT64):
+    names = list(agg_fns)
+    fields = [field("key", DT_INT64, True),
+              field("#9223372036854775807", DT_BINARY, False)]
+    reader = ipc_reader(fields, resource_id)
+    final = agg(reader, [column("key", 0)],
+                named_aggs(names, val_dt=val_dt),
+                [MODE_FINAL] * len(names), ["key"], names)
+    return task_definition(final)
+
+
 def plan_ipc_final(resource_id="ipc0"):
     """IpcReader(partial output schema) -> Agg(Final): the reference's
     stage-2 topology (NativeShuffleExchangeBase.scala:147-179) fully native."""
