@@ -373,3 +373,40 @@ def int_sum_groups(keys, vals, val_valid=None):
     ordered = list(groups.keys())
     return (ordered, [groups[k][0] for k in ordered],
             [groups[k][1] for k in ordered])
+
+
+# ---- COLLECT_LIST / COLLECT_SET restatement (pure python) ------------------
+# collect.rs: COLLECT_LIST accumulates every non-null arg in arrival order
+# (collect.rs:119-138 partial_update appends non-null scalars); COLLECT_SET
+# dedups, keeping FIRST-occurrence order (AccSet.append appends only novel
+# values). Freeze (collect.rs:237-241 save_raw) = write_len(raw_len) ++ the
+# concatenated non-nullable scalar encodings — for i64/f64 args that is the
+# raw LE 8-byte values (scalar_serde.rs:35-47 write_prim non-nullable).
+# Merge (collect.rs:139-158) concatenates in merge arrival order (set: with
+# dedup). CPU parity anchor for the round-2 device implementation.
+def collect_groups(keys, vals, val_valid=None, distinct=False):
+    """Insertion-ordered (keys, lists): lists hold non-null values in
+    arrival order; distinct=True dedups keeping first occurrence."""
+    groups = {}
+    n = len(keys)
+    for i in range(n):
+        k = int(keys[i])
+        if k not in groups:
+            groups[k] = ([], set())
+        if val_valid is None or val_valid[i]:
+            v = vals[i]
+            lst, seen = groups[k]
+            if distinct:
+                if v in seen:
+                    continue
+                seen.add(v)
+            lst.append(v)
+    ordered = list(groups.keys())
+    return ordered, [groups[k][0] for k in ordered]
+
+
+def collect_freeze_rec(values, fmt="<d"):
+    """Freeze one COLLECT record: varint(byte-len) ++ packed values."""
+    import struct
+    raw = b"".join(struct.pack(fmt, v) for v in values)
+    return write_len(len(raw)) + raw

@@ -123,3 +123,31 @@ def test_first_restatement_hand_case():
     assert ok == [1, 2, 3]
     assert firsts == [(True, None), (True, 5.0), (True, None)]
     assert firsts_nn == [4.0, 5.0, None]
+
+
+def test_collect_restatement_hand_case():
+    """collect.rs semantics: LIST keeps arrival order incl. duplicates; SET
+    dedups keeping first occurrence; null args skipped."""
+    from oracle import pywrap as oracle
+    keys = [1, 2, 1, 1, 2, 1]
+    vals = [5.0, 7.0, 5.0, 3.0, 7.0, 9.0]
+    vv = [True, True, True, True, True, False]
+    ok, lists = oracle.collect_groups(keys, vals, vv)
+    assert ok == [1, 2]
+    assert lists == [[5.0, 5.0, 3.0], [7.0, 7.0]]
+    ok2, sets = oracle.collect_groups(keys, vals, vv, distinct=True)
+    assert sets == [[5.0, 3.0], [7.0]]
+
+
+def test_collect_freeze_format():
+    """COLLECT freeze = varint(raw_len) ++ raw LE scalar bytes
+    (collect.rs:237-241 save_raw; scalar_serde.rs:35-47)."""
+    import struct
+    from oracle import pywrap as oracle
+    rec = oracle.collect_freeze_rec([1.5, -2.0])
+    assert rec == oracle.write_len(16) + struct.pack("<d", 1.5) + \
+        struct.pack("<d", -2.0)
+    assert oracle.collect_freeze_rec([]) == oracle.write_len(0)
+    reci = oracle.collect_freeze_rec([7, -9], fmt="<q")
+    assert reci == oracle.write_len(16) + struct.pack("<q", 7) + \
+        struct.pack("<q", -9)
