@@ -2443,6 +2443,78 @@ void auron_on_exit(void) {
 
 const char* auron_version(void) { return "auron-hip 0.1 gfx950"; }
 
+// test-only: host-side batch_serde + IPC block codec roundtrip over a
+// 3-column batch (i64 key, f64 val, binary) — CPU-checkable byte parity
+// against the oracle writer plus write->read self-consistency, covering the
+// engine's IpcReader decode path without a GPU. Returns the block-stream
+// length (copied into out up to cap), or -1 with the error text in out.
+int64_t auron_debug_serde_roundtrip(const int64_t* k, const uint8_t* kvalid,
+                                    const double* v, const uint8_t* vvalid,
+                                    const int32_t* boffs, const uint8_t* bdata,
+                                    int64_t n, int64_t batch_size, uint8_t* out,
+                                    size_t cap) {
+  auto fail = [&](const std::string& m) -> int64_t {
+    snprintf((char*)out, cap, "%s", m.c_str());
+    return -1;
+  };
+  std::string err;
+  std::vector<HostCol> cols(3);
+  cols[0] = {8, (const uint8_t*)k, kvalid, nullptr};
+  cols[1] = {8, (const uint8_t*)v, vvalid, nullptr};
+  cols[2] = {0, bdata, nullptr, boffs};
+  IpcBlockWriter w;
+  for (int64_t beg = 0; beg < n; beg += batch_size) {
+    int64_t end = std::min(n, beg + batch_size);
+    std::vector<uint8_t> payload;
+    serde_write_batch(cols, beg, end, &payload);
+    if (!w.write_payload(payload.data(), payload.size(), &err))
+      return fail(err);
+  }
+  if (!w.finish_block(&err)) return fail(err);
+  std::vector<uint8_t> stream = w.take();
+
+  // decode side: blocks -> payload -> batches; verify every row round-trips
+  std::vector<uint8_t> payload;
+  if (!ipc_decode_blocks(stream.data(), stream.size(), &payload, &err))
+    return fail(err);
+  size_t pos = 0;
+  int64_t row = 0;
+  while (pos < payload.size()) {
+    size_t used = 0;
+    int64_t rows = 0;
+    std::vector<OwnedCol> rc;
+    if (!serde_read_batch(payload.data() + pos, payload.size() - pos, &used,
+                          {8, 8, 0}, &rows, &rc, &err))
+      return fail(err);
+    pos += used;
+    for (int64_t i = 0; i < rows; i++, row++) {
+      bool kv2 = rc[0].validity.empty() ||
+                 ((rc[0].validity[i >> 3] >> (i & 7)) & 1);
+      bool kv1 = !kvalid || ((kvalid[row >> 3] >> (row & 7)) & 1);
+      if (kv1 != kv2) return fail("key validity mismatch");
+      if (kv1 && memcmp(rc[0].values.data() + i * 8, (const uint8_t*)&k[row],
+                        8) != 0)
+        return fail("key value mismatch");
+      bool vv2 = rc[1].validity.empty() ||
+                 ((rc[1].validity[i >> 3] >> (i & 7)) & 1);
+      bool vv1 = !vvalid || ((vvalid[row >> 3] >> (row & 7)) & 1);
+      if (vv1 != vv2) return fail("val validity mismatch");
+      if (vv1 && memcmp(rc[1].values.data() + i * 8, (const uint8_t*)&v[row],
+                        8) != 0)
+        return fail("val value mismatch");
+      int32_t l1 = boffs[row + 1] - boffs[row];
+      int32_t l2 = rc[2].offsets[i + 1] - rc[2].offsets[i];
+      if (l1 != l2) return fail("binary length mismatch");
+      if (l1 && memcmp(rc[2].values.data() + rc[2].offsets[i],
+                       bdata + boffs[row], l1) != 0)
+        return fail("binary bytes mismatch");
+    }
+  }
+  if (row != n) return fail("row count mismatch");
+  if (stream.size() <= cap) memcpy(out, stream.data(), stream.size());
+  return (int64_t)stream.size();
+}
+
 // test-only: parse a parquet file on the host and summarize chunk decode
 // results (CPU-checkable against pyarrow metadata — no GPU needed)
 int32_t auron_debug_parquet_summary(const char* path, char* out, size_t cap) {
