@@ -752,7 +752,7 @@ def test_agg_argument_casts():
     t2.finalize()
 
 
-@pytest.mark.parametrize("seed", [101, 202, 303, 404])
+@pytest.mark.parametrize("seed", [101, 202, 303, 404, 505, 616, 727, 838])
 def test_agg_fuzz_randomized(seed):
     """Randomized end-to-end parity in the spirit of the reference's own
     agg fuzz test (agg_exec.rs:714-843): random agg lists, accumulator
