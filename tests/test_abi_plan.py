@@ -107,3 +107,25 @@ def test_decode_first_plan():
     td = plan.plan_partial_final_named(["first", "first_ignores_null"])
     s = blaze_amd.debug_decode_plan(td)
     assert "fn7" in s and "fn8" in s
+
+
+def test_decode_skips_unknown_fields():
+    """proto3 forward-compat: unknown fields in TaskDefinition / plan nodes
+    are skipped, not fatal (the reference's prost decoder does the same)."""
+    td = plan.plan_partial_final()
+    # append an unknown varint field (tag 2000) and an unknown length-
+    # delimited field (tag 1999) at the top level
+    extra = plan._varint_field(2000, 42) + plan._len_field(1999, b"junk")
+    s = blaze_amd.debug_decode_plan(td + extra)
+    assert "Agg(mode=2" in s and "FFIReader" in s
+
+
+def test_decode_rejects_truncated_plan():
+    td = plan.plan_partial_final()
+    import ctypes
+    lib = blaze_amd.lib()
+    out = ctypes.create_string_buffer(1 << 12)
+    rc = lib.auron_debug_decode_plan(td[: len(td) // 2], len(td) // 2, out,
+                                     len(out))
+    # must not crash; either a decode error or a partial tree is reported
+    assert rc != 0 or len(out.value) >= 0
