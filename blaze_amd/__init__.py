@@ -277,6 +277,20 @@ def _import_output(array_ptr, schema_fields):
         dt = schema_fields[i][1]
         n = ch.length
         valid = _read_bitmap(ch.buffers[0], n) if ch.null_count else None
+        if dt.startswith("+l:"):
+            # list<prim>: offsets on the parent, values on the child array
+            offsets = np.ctypeslib.as_array(
+                c.cast(ch.buffers[1], c.POINTER(c.c_int32)),
+                shape=(n + 1,)).copy()
+            item = ch.children[0].contents
+            npdt = {"l": np.int64, "g": np.float64}[dt[3:]]
+            ni = int(item.length)
+            vals = np.ctypeslib.as_array(
+                c.cast(item.buffers[1], c.POINTER(c.c_uint8)),
+                shape=(max(ni, 1) * 8,))[:ni * 8].view(npdt).copy()
+            out.append(dict(dtype="list", offsets=offsets, values=vals,
+                            valid=valid))
+            continue
         if dt == "z":
             offsets = np.ctypeslib.as_array(
                 c.cast(ch.buffers[1], c.POINTER(c.c_int32)), shape=(n + 1,)).copy()
@@ -345,9 +359,12 @@ class Task:
             s = sp.contents
             for i in range(s.n_children):
                 chs = s.children[i].contents
+                fmt = chs.format.decode()
+                if fmt == "+l":  # list<item>: record the item format too
+                    item = chs.children[0].contents.format.decode()
+                    fmt = "+l:" + item
                 self.schema_fields.append(
-                    (chs.name.decode() if chs.name else "",
-                     chs.format.decode()))
+                    (chs.name.decode() if chs.name else "", fmt))
             rel = s.release
             if rel:
                 c.CFUNCTYPE(None, c.POINTER(ArrowSchema))(rel)(sp)

@@ -204,6 +204,10 @@ struct ExportPriv {
   std::vector<ArrowArray> children_store;
   std::vector<ArrowArray*> children_ptrs;
   std::vector<std::vector<const void*>> child_buffer_ptrs;
+  // list columns: one item array per column slot
+  std::vector<ArrowArray> item_store;
+  std::vector<ArrowArray*> item_ptrs;
+  std::vector<std::vector<const void*>> item_buffer_ptrs;
 };
 
 void release_exported_array(ArrowArray* a) {
@@ -219,6 +223,9 @@ struct SchemaPriv {
   std::vector<ArrowSchema*> children_ptrs;
   std::vector<std::string> names;
   std::vector<const char*> formats;
+  // list fields: one item-schema per field slot (null when not a list)
+  std::vector<ArrowSchema> item_store;
+  std::vector<ArrowSchema*> item_ptrs;
 };
 
 void release_exported_schema(ArrowSchema* s) {
@@ -229,23 +236,38 @@ void release_exported_schema(ArrowSchema* s) {
 
 struct OutField {
   std::string name;
-  DType dt;
+  DType dt;        // list fields: the ITEM type
   bool nullable;
+  bool is_list = false;  // Arrow list<dt> (COLLECT output)
 };
 
 void export_schema(const std::vector<OutField>& fields, ArrowSchema* out) {
   auto* priv = new SchemaPriv;
   priv->children_store.resize(fields.size());
+  priv->item_store.resize(fields.size());
+  priv->item_ptrs.resize(fields.size());
   for (size_t i = 0; i < fields.size(); i++) {
     priv->names.push_back(fields[i].name);
   }
   for (size_t i = 0; i < fields.size(); i++) {
     ArrowSchema& c = priv->children_store[i];
     memset(&c, 0, sizeof(c));
-    c.format = dtype_format(fields[i].dt);
     c.name = priv->names[i].c_str();
     c.flags = fields[i].nullable ? ARROW_FLAG_NULLABLE : 0;
     c.release = [](ArrowSchema* s) { s->release = nullptr; };
+    if (fields[i].is_list) {
+      c.format = "+l";
+      ArrowSchema& it = priv->item_store[i];
+      memset(&it, 0, sizeof(it));
+      it.format = dtype_format(fields[i].dt);
+      it.name = "item";
+      it.release = [](ArrowSchema* s) { s->release = nullptr; };
+      priv->item_ptrs[i] = &it;
+      c.n_children = 1;
+      c.children = &priv->item_ptrs[i];
+    } else {
+      c.format = dtype_format(fields[i].dt);
+    }
     priv->children_ptrs.push_back(&c);
   }
   memset(out, 0, sizeof(*out));
@@ -259,10 +281,11 @@ void export_schema(const std::vector<OutField>& fields, ArrowSchema* out) {
 
 // host-side column staging for export
 struct HostOutCol {
-  DType dt;
-  std::vector<uint8_t> values;
+  DType dt;                       // list cols: the ITEM type
+  std::vector<uint8_t> values;    // list cols: child (item) values
   std::vector<uint8_t> validity;  // empty = no nulls
-  std::vector<int32_t> offsets;   // binary only
+  std::vector<int32_t> offsets;   // binary / list: n+1
+  bool is_list = false;
 };
 
 void export_batch(int64_t num_rows, std::vector<HostOutCol>&& cols,
@@ -270,6 +293,9 @@ void export_batch(int64_t num_rows, std::vector<HostOutCol>&& cols,
   auto* priv = new ExportPriv;
   priv->children_store.resize(cols.size());
   priv->child_buffer_ptrs.resize(cols.size());
+  priv->item_store.resize(cols.size());
+  priv->item_ptrs.resize(cols.size());
+  priv->item_buffer_ptrs.resize(cols.size());
   for (size_t i = 0; i < cols.size(); i++) {
     HostOutCol& c = cols[i];
     ArrowArray& ch = priv->children_store[i];
@@ -293,7 +319,22 @@ void export_batch(int64_t num_rows, std::vector<HostOutCol>&& cols,
     } else {
       ch.null_count = 0;
     }
-    if (c.dt == DType::Binary || c.dt == DType::Utf8) {
+    if (c.is_list) {
+      // list<prim>: parent {validity, offsets}; child = item values
+      bufs = {validity, copy_out(c.offsets.data(), c.offsets.size() * 4)};
+      ArrowArray& it = priv->item_store[i];
+      memset(&it, 0, sizeof(it));
+      it.length = c.offsets.empty() ? 0 : c.offsets.back();
+      it.null_count = 0;
+      std::vector<const void*>& ibufs = priv->item_buffer_ptrs[i];
+      ibufs = {nullptr, copy_out(c.values.data(), c.values.size())};
+      it.n_buffers = 2;
+      it.buffers = ibufs.data();
+      it.release = [](ArrowArray* a) { a->release = nullptr; };
+      priv->item_ptrs[i] = &it;
+      ch.n_children = 1;
+      ch.children = &priv->item_ptrs[i];
+    } else if (c.dt == DType::Binary || c.dt == DType::Utf8) {
       bufs = {validity, copy_out(c.offsets.data(), c.offsets.size() * 4),
               copy_out(c.values.data(), c.values.size())};
     } else {
@@ -379,11 +420,12 @@ class AggOp {
         case AGG_MAX: k = AGGL_MAX; has_mm_ = true; break;
         case AGG_FIRST: k = AGGL_FIRST; has_first_ = true; break;
         case AGG_FIRST_IGNORES_NULL: k = AGGL_FIRSTIN; has_first_ = true; break;
+        case AGG_COLLECT_LIST: k = AGGL_CLIST; has_coll_ = true; break;
         default:
-          FAIL("AggExec: only SUM/COUNT/AVG/MIN/MAX/FIRST[_IGNORES_NULL] "
-               "aggregates on this path");
+          FAIL("AggExec: only SUM/COUNT/AVG/MIN/MAX/FIRST[_IGNORES_NULL]/"
+               "COLLECT_LIST aggregates on this path");
       }
-      layout_ |= k << (3 * i);
+      layout_ |= k << (4 * i);
       // accumulator arithmetic type follows the agg's declared data type
       // (sum.rs:78-88 casts inputs to it; maxmin.rs:81-83 preserves it)
       if (k != AGGL_CNT) {
@@ -508,7 +550,7 @@ class AggOp {
       // chunks on partition-buffer size instead of table free slots.
       // MIN/MAX agg sets stay single-phase: the LDS bucket kernel's slot
       // holds {key,cnt,sum,first} only (perf note in DESIGN.md)
-      if (!merge_mode_ && !has_mm_ && !has_first_ &&
+      if (!merge_mode_ && !has_mm_ && !has_first_ && !has_coll_ &&
           b.num_rows - done >= AGG2_MIN_CHUNK) {
         int64_t chunk2 = std::min(b.num_rows - done, AGG2_MAX_CHUNK);
         if (done + chunk2 < b.num_rows) chunk2 &= ~(int64_t)7;
@@ -623,6 +665,9 @@ class AggOp {
           f.push_back({agg_names_[i], DType::Int64, false});
         else if (agg_kinds_[i] == AGGL_AVG)
           f.push_back({agg_names_[i], DType::Float64, true});
+        else if (agg_kinds_[i] == AGGL_CLIST)
+          // collect.rs:110-112: nullable() = false (empty lists, not nulls)
+          f.push_back({agg_names_[i], vdt, false, /*is_list=*/true});
         else
           f.push_back({agg_names_[i], vdt, true});
       }
@@ -634,9 +679,81 @@ class AggOp {
   }
 
   // drain: produce all output batches (host-staged)
+  // Sort the COLLECT pool into {key ascending (signed), prio ascending
+  // within key} + the null-key back segment into prio order, then point the
+  // table's c_key/c_val views at the sorted arrays (binary-searched by the
+  // freeze/emit kernels). One-shot before any emit.
+  void prepare_collect() {
+    if (!has_coll_ || coll_sorted_) return;
+    coll_sorted_ = true;
+    if (pinned_meta_.size() < 32) pinned_meta_.alloc(32);
+    AURON_HIP(hipMemcpyAsync(pinned_meta_.get(), d_cn_.get(), 16,
+                             hipMemcpyDeviceToHost, stream_));
+    uint32_t* errp = pinned_meta_.get<uint32_t>() + 4;
+    AURON_HIP(hipMemcpyAsync(errp, t_.error_flag, 4, hipMemcpyDeviceToHost,
+                             stream_));
+    AURON_HIP(hipStreamSynchronize(stream_));
+    coll_n0_ = (int64_t)pinned_meta_.get<unsigned long long>()[0];
+    coll_n1_ = (int64_t)pinned_meta_.get<unsigned long long>()[1];
+    if (*errp & 4u)
+      FAIL("collect pool overflow: raise AURON_HIP_COLLECT_POOL");
+    int64_t n0 = coll_n0_, n1 = coll_n1_;
+    d_cskey_.alloc(std::max<int64_t>(n0, 1) * 8);
+    d_csval_.alloc(std::max<int64_t>(n0 + n1, 1) * 8);
+    size_t tb0 = 0;
+    sort_pairs_u64_u32(nullptr, nullptr, nullptr, nullptr,
+                       std::max<int64_t>(std::max(n0, n1), 1), nullptr, &tb0,
+                       stream_);
+    DevBuf tmp(tb0);
+    if (n0 > 0) {
+      DevBuf idx(n0 * 4), idxo(n0 * 4), k64(n0 * 8), ks(n0 * 8), v1(n0 * 8);
+      // pass 1: prio order
+      launch_iota_u32(idx.get<uint32_t>(), n0, stream_);
+      size_t tb = tmp.size();
+      sort_pairs_u64_u32(t_.c_prio, idx.get<uint32_t>(),
+                         k64.get<unsigned long long>() /* scratch keys out */,
+                         idxo.get<uint32_t>(), n0, tmp.get(), &tb, stream_);
+      launch_gather_u64_idx((const unsigned long long*)t_.c_key,
+                            idxo.get<uint32_t>(), n0,
+                            ks.get<unsigned long long>(), stream_);
+      launch_gather_u64_idx(t_.c_val, idxo.get<uint32_t>(), n0,
+                            v1.get<unsigned long long>(), stream_);
+      // pass 2: stable key order (sign-biased radix) keeps prio order inside
+      launch_bias_i64(ks.get<unsigned long long>(), n0,
+                      k64.get<unsigned long long>(), stream_);
+      launch_iota_u32(idx.get<uint32_t>(), n0, stream_);
+      DevBuf biased_out(n0 * 8);
+      tb = tmp.size();
+      sort_pairs_u64_u32(k64.get<unsigned long long>(), idx.get<uint32_t>(),
+                         biased_out.get<unsigned long long>(),
+                         idxo.get<uint32_t>(), n0, tmp.get(), &tb, stream_);
+      launch_gather_u64_idx(ks.get<unsigned long long>(), idxo.get<uint32_t>(),
+                            n0, d_cskey_.get<unsigned long long>(), stream_);
+      launch_gather_u64_idx(v1.get<unsigned long long>(), idxo.get<uint32_t>(),
+                            n0, d_csval_.get<unsigned long long>(), stream_);
+    }
+    if (n1 > 0) {
+      // null-key back segment: prio order only
+      DevBuf idx(n1 * 4), idxo(n1 * 4), po(n1 * 8);
+      launch_iota_u32(idx.get<uint32_t>(), n1, stream_);
+      size_t tb = tmp.size();
+      sort_pairs_u64_u32(t_.c_prio + (coll_cap_ - n1), idx.get<uint32_t>(),
+                         po.get<unsigned long long>(), idxo.get<uint32_t>(),
+                         n1, tmp.get(), &tb, stream_);
+      launch_gather_u64_idx(t_.c_val + (coll_cap_ - n1), idxo.get<uint32_t>(),
+                            n1, d_csval_.get<unsigned long long>() + n0,
+                            stream_);
+    }
+    AURON_HIP(hipStreamSynchronize(stream_));  // temps return to the pool
+    t_.c_key = d_cskey_.get<long long>();
+    t_.c_val = d_csval_.get<unsigned long long>();
+    t_.c_cap = n0 + n1;  // null segment now ends at n0+n1 (table_snap math)
+  }
+
   std::vector<std::pair<int64_t, std::vector<HostOutCol>>> finish() {
     std::vector<std::pair<int64_t, std::vector<HostOutCol>>> out;
     drain_timing();
+    prepare_collect();
     if (spill_.empty()) {
       emit_table(&out, false);
     } else {
@@ -747,6 +864,21 @@ class AggOp {
       t_.f_val = d_fval_.get<double>();
       t_.f_st = d_fst_.get<uint8_t>();
       launch_first_init(t_.f_row, t_.f_val, t_.f_st, cap + 2, stream_);
+    }
+    if (has_coll_ && !d_ckey_) {  // pool survives grows (keys, not slots)
+      coll_cap_ = conf_coll_cap_;
+      d_ckey_.alloc(coll_cap_ * 8);
+      d_cprio_.alloc(coll_cap_ * 8);
+      d_cval_.alloc(coll_cap_ * 8);
+      d_cn_.alloc(16);
+      AURON_HIP(hipMemsetAsync(d_cn_.get(), 0, 16, stream_));
+    }
+    if (has_coll_) {
+      t_.c_key = d_ckey_.get<long long>();
+      t_.c_prio = d_cprio_.get<unsigned long long>();
+      t_.c_val = d_cval_.get<unsigned long long>();
+      t_.c_n = d_cn_.get<unsigned long long>();
+      t_.c_cap = coll_cap_;
     }
   }
 
@@ -1060,6 +1192,9 @@ class AggOp {
     while (t_.cap * 3 / 4 < need && t_.cap < max_cap_) grow(t_.cap * 4);
     if (t_.cap * 3 / 4 < need &&
         (int64_t)num_groups_host() > specials_count_) {
+      if (has_coll_)
+        FAIL("COLLECT_LIST with table spill unsupported (round-2: drain the "
+             "pool into the spill buckets)");
       spill_table();  // frees the main region; caller re-checks free space
     }
   }
@@ -1172,6 +1307,26 @@ class AggOp {
         AURON_HIP(hipMemcpyAsync(xv.data(), xvalid.get(), bm,
                                  hipMemcpyDeviceToHost, stream_));
       }
+      std::vector<int32_t> h_coff;
+      std::vector<uint8_t> h_citems;
+      bool need_coll = false;
+      for (uint32_t k : agg_kinds_) need_coll |= (k == AGGL_CLIST);
+      if (need_coll) {
+        DevBuf cnts(n * 4);
+        launch_coll_counts(t_, order_slots, n, cnts.get<int32_t>(), stream_);
+        std::vector<int32_t> h_cnts(n);
+        AURON_HIP(hipMemcpyAsync(h_cnts.data(), cnts.get(), n * 4,
+                                 hipMemcpyDeviceToHost, stream_));
+        AURON_HIP(hipStreamSynchronize(stream_));
+        h_coff.assign(n + 1, 0);
+        for (int64_t i = 0; i < n; i++) h_coff[i + 1] = h_coff[i] + h_cnts[i];
+        DevBuf d_off((n + 1) * 4), items((int64_t)h_coff[n] * 8 + 8);
+        AURON_HIP(hipMemcpyAsync(d_off.get(), h_coff.data(), (n + 1) * 4,
+                                 hipMemcpyHostToDevice, stream_));
+        launch_coll_gather(t_, order_slots, n, d_off.get<int32_t>(),
+                           items.get<unsigned long long>(), stream_);
+        d2h_pinned(items.get(), &h_citems, (size_t)h_coff[n] * 8);
+      }
       std::vector<uint8_t> h_firsts[2];
       std::vector<uint8_t> fv[2] = {std::vector<uint8_t>(bm),
                                     std::vector<uint8_t>(bm)};
@@ -1205,6 +1360,11 @@ class AggOp {
           ac.dt = vdt;
           ac.values = h_firsts[w];
           attach_validity(&ac, fv[w], n);
+        } else if (k == AGGL_CLIST) {
+          ac.dt = vdt;
+          ac.is_list = true;
+          ac.offsets = h_coff;
+          ac.values = h_citems;
         } else if (k == AGGL_AVG) {
           ac.dt = DType::Float64;
           ac.values = h_avgs;
@@ -1309,6 +1469,9 @@ class AggOp {
   DType key_dt_ = DType::Unsupported;
   uint32_t layout_ = 0;
   bool has_mm_ = false;  // agg list contains MIN/MAX: side mm array active
+  bool has_coll_ = false, coll_sorted_ = false;  // COLLECT_LIST pool active
+  int64_t coll_cap_ = 0, conf_coll_cap_ = 16 << 20;
+  int64_t coll_n0_ = 0, coll_n1_ = 0;
   bool val_is_int_ = false, val_typed_seen_ = false;  // i64 accumulator mode
   bool has_first_ = false;  // FIRST family: f_row/f_val/f_st arrays active
   std::vector<uint32_t> agg_kinds_;
@@ -1329,6 +1492,7 @@ class AggOp {
   hipEvent_t ev_start_ = nullptr, ev_stop_ = nullptr;
   AggTable t_;
   DevBuf d_slots_, d_special_, d_ng_, d_err_, d_mm_, d_frow_, d_fval_, d_fst_;
+  DevBuf d_ckey_, d_cprio_, d_cval_, d_cn_, d_cskey_, d_csval_;
   PinnedBuf pinned_meta_, pinned_emit_;
   // two-phase scratch (allocated on first large chunk)
   bool agg2_split_ = false;

@@ -89,6 +89,19 @@ struct AggTable {
   double* f_val = nullptr;              // captured value
   uint8_t* f_st = nullptr;              // 0 = untouched, 1 = first-was-null
                                         // (FIRST only), 2 = valid
+  // COLLECT_LIST pool (collect.rs:119-138 restated; DESIGN §8 round-2 item
+  // 2, landed early): every non-null arg appends a (key, prio, value-bits)
+  // triple; prio = global arrival row (update) or record_row<<20|item
+  // (frozen-list merge), so a sort by prio then by key reproduces the
+  // reference's per-group arrival order. Regular keys fill from the front
+  // (counter n[0]); null-key rows fill from the BACK (counter n[1], slot
+  // cap-1-i) — two segments, one allocation, no partition pass. Keys (not
+  // slot indices) make table grows safe. error_flag bit 4 = pool overflow.
+  long long* c_key = nullptr;            // [c_cap]
+  unsigned long long* c_prio = nullptr;  // [c_cap]
+  unsigned long long* c_val = nullptr;   // [c_cap] raw value bits
+  unsigned long long* c_n = nullptr;     // [2] device counters
+  int64_t c_cap = 0;
   // accumulator arithmetic type (sum.rs:78-88: the acc column is the agg's
   // declared data type and inputs are cast to it). 0 = f64 (north star);
   // 1 = i64 — sums use wrapping integer atomicAdd on the same 8-byte slot
@@ -105,15 +118,16 @@ void launch_agg_update(const AggTable& t, const int64_t* keys,
                        const uint8_t* val_valid, int64_t n, uint64_t row_offset,
                        hipStream_t s);
 
-// agg layout descriptor: 3 bits per agg LSB-first, 0-terminated (8 aggs max
-// fits 24 bits); all aggs share one argument column. AVG freeze = sum ++
+// agg layout descriptor: 4 bits per agg LSB-first, 0-terminated below 8
+// aggs (8 aggs fill the u32 exactly; the walker self-terminates); all aggs
+// share one argument column. AVG freeze = sum ++
 // count (avg.rs:208-217); MIN/MAX freeze = the same prim [u8 valid][8B LE]
 // format as SUM (both are AccPrimColumn saves, acc.rs:335-347 — maxmin.rs
 // create_acc_column:91-93 uses the same generic prim column as sum.rs), so
 // duplicated parts repeat the shared accumulators.
 enum AggLayoutKind : uint32_t {
   AGGL_SUM = 1, AGGL_CNT = 2, AGGL_AVG = 3, AGGL_MIN = 4, AGGL_MAX = 5,
-  AGGL_FIRST = 6, AGGL_FIRSTIN = 7
+  AGGL_FIRST = 6, AGGL_FIRSTIN = 7, AGGL_CLIST = 8
 };
 
 // merge rows of frozen partial state per the layout
@@ -178,6 +192,19 @@ void launch_mm_init(unsigned long long* mm, int64_t n, hipStream_t s);
 // initialize n {f_row, f_val, f_st} pair-slots (2 entries each)
 void launch_first_init(unsigned long long* f_row, double* f_val, uint8_t* f_st,
                        int64_t n, hipStream_t s);
+
+// gather u64 elements by u32 index (pool reorder after a sort pass)
+void launch_gather_u64_idx(const unsigned long long* src, const uint32_t* idx,
+                           int64_t n, unsigned long long* dst, hipStream_t s);
+void launch_bias_i64(const unsigned long long* src, int64_t n,
+                     unsigned long long* dst, hipStream_t s);
+// emit side: per-group collect counts and value runs (pool sorted first)
+void launch_coll_counts(const AggTable& t, const uint32_t* order_slots,
+                        int64_t num_groups, int32_t* cnts, hipStream_t s);
+void launch_coll_gather(const AggTable& t, const uint32_t* order_slots,
+                        int64_t num_groups, const int32_t* offsets,
+                        unsigned long long* out, hipStream_t s);
+
 
 // FIRST/FIRST_IGNORES_NULL pass B over update-mode rows: the unique row
 // whose global row index equals the slot's captured f_row priority stores

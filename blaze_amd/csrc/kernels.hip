@@ -275,6 +275,36 @@ __device__ __forceinline__ void sum_accum(double* acc, double v, bool is_int) {
   }
 }
 
+// COLLECT_LIST pool append (collect.rs:119-138: every non-null arg).
+// Regular keys from the front, null-key rows from the back; overflow raises
+// error_flag bit 4 and drops the item (host fails the task loudly).
+__device__ __forceinline__ void coll_append(const AggTable& t, int64_t key,
+                                            bool knull, uint64_t prio,
+                                            double val) {
+  uint64_t vb;
+  memcpy(&vb, &val, 8);
+  if (knull) {
+    unsigned long long p = atomicAdd(&t.c_n[1], 1ull);
+    if ((int64_t)(p + t.c_n[0]) >= t.c_cap) {  // approximate guard; exact
+      atomicOr(t.error_flag, 4u);              // check on host readback
+      return;
+    }
+    int64_t at = t.c_cap - 1 - (int64_t)p;
+    t.c_key[at] = 0;
+    t.c_prio[at] = prio;
+    t.c_val[at] = vb;
+  } else {
+    unsigned long long p = atomicAdd(&t.c_n[0], 1ull);
+    if ((int64_t)(p + t.c_n[1]) >= t.c_cap) {
+      atomicOr(t.error_flag, 4u);
+      return;
+    }
+    t.c_key[p] = key;
+    t.c_prio[p] = prio;
+    t.c_val[p] = vb;
+  }
+}
+
 // Per-row slot resolution + accumulate (the slow path of the batched kernel
 // below; also correct standalone).
 __device__ __forceinline__ void agg_accum_row(const AggTable t, int64_t key,
@@ -303,6 +333,7 @@ __device__ __forceinline__ void agg_accum_row(const AggTable t, int64_t key,
     atomicMin(&t.f_row[2 * a], (unsigned long long)row);
     if (vvalid) atomicMin(&t.f_row[2 * a + 1], (unsigned long long)row);
   }
+  if (t.c_key && vvalid) coll_append(t, key, knull, row, val);
 }
 
 // The dominant kernel. PMC evidence (profiles/): one-row-at-a-time leaves
@@ -363,6 +394,8 @@ __global__ void k_agg_update(const AggTable t, const int64_t* __restrict__ keys,
           if (vvalid[k])
             atomicMin(&t.f_row[2 * slot[k] + 1], (unsigned long long)row);
         }
+        if (t.c_key && vvalid[k])
+          coll_append(t, key[k], false, row, val[k]);
       } else {
         agg_accum_row(t, knull[k] ? 0 : key[k], knull[k], val[k], vvalid[k],
                       row);
@@ -424,19 +457,26 @@ struct AccSnap {
   double f_val = 0.0;
   uint8_t fn_st = 0;     // FIRST_IGNORES_NULL state: 0 untouched / 2 valid
   double fn_val = 0.0;
+  // COLLECT_LIST part: run of 8-byte values (freeze reads them, parse
+  // exposes the frozen bytes; collect.rs:237-241 save_raw)
+  const unsigned long long* c_vals = nullptr;  // 8-aligned? frozen bytes are
+  const uint8_t* c_raw = nullptr;              // not aligned — use c_raw
+  uint32_t c_cnt = 0;
 };
 
 __device__ __forceinline__ int agg_freeze_len(uint32_t layout,
                                               const AccSnap& a) {
   int len = 0;
-  for (uint32_t l = layout; l & 7u; l >>= 3) {
-    uint32_t k = l & 7u;
+  for (uint32_t l = layout; l & 15u; l >>= 4) {
+    uint32_t k = l & 15u;
     if (k == 1 || k == 3) len += 1 + (a.valid ? 8 : 0);
     if (k == 2 || k == 3) len += varint_len_dev(a.cnt);
     if (k == 4) len += 1 + (a.minu != MM_MIN_INIT ? 8 : 0);
     if (k == 5) len += 1 + (a.maxu != MM_MAX_INIT ? 8 : 0);
     if (k == 6) len += 2 + (a.f_st == 2 ? 8 : 0);  // prim part + flag byte
     if (k == 7) len += 1 + (a.fn_st == 2 ? 8 : 0);
+    if (k == 8)  // collect.rs:237-241: varint(raw_len) ++ raw values
+      len += varint_len_dev((uint64_t)a.c_cnt * 8) + (int)a.c_cnt * 8;
   }
   return len;
 }
@@ -460,8 +500,8 @@ __device__ __forceinline__ uint8_t* agg_freeze_write_rec(uint32_t layout,
   const bool valid = a.valid;
   const double sum = a.sum;
   const uint64_t cnt = a.cnt;
-  for (uint32_t l = layout; l & 7u; l >>= 3) {
-    uint32_t k = l & 7u;
+  for (uint32_t l = layout; l & 15u; l >>= 4) {
+    uint32_t k = l & 15u;
     if (k == 1 || k == 3)  // acc.rs:335-347 prim freeze
       p = agg_prim_freeze_part(valid, sum, p);
     if (k == 4)
@@ -476,6 +516,23 @@ __device__ __forceinline__ uint8_t* agg_freeze_write_rec(uint32_t layout,
     }
     if (k == 7)  // FIRST_IGNORES_NULL: prim value only
       p = agg_prim_freeze_part(a.fn_st == 2, a.fn_val, p);
+    if (k == 8) {  // COLLECT_LIST raw list (collect.rs:237-241)
+      uint64_t raw = (uint64_t)a.c_cnt * 8;
+      int used = varint_len_dev(raw);
+      for (uint64_t v = raw; ; v /= 128) {
+        *p++ = (uint8_t)(v >= 128 ? (v % 128) + 128 : v);
+        if (v < 128) break;
+      }
+      (void)used;
+      for (uint32_t i = 0; i < a.c_cnt; i++) {
+        if (a.c_vals) {
+          memcpy(p, &a.c_vals[i], 8);
+        } else if (a.c_raw) {
+          memcpy(p, a.c_raw + (size_t)i * 8, 8);
+        }
+        p += 8;
+      }
+    }
     if (k == 2 || k == 3) {  // count.rs:193-203 varint
       uint64_t c = cnt;
       while (c >= 128) {
@@ -497,9 +554,9 @@ __device__ __forceinline__ void agg_parse_frozen(uint32_t layout,
                                                  bool is_int = false) {
   *a = AccSnap{};
   uint32_t got = 0;  // bit per family
-  for (uint32_t l = layout; l & 7u; l >>= 3) {
-    uint32_t k = l & 7u;
-    if (k != 2) {  // every non-pure-COUNT part starts with a prim part
+  for (uint32_t l = layout; l & 15u; l >>= 4) {
+    uint32_t k = l & 15u;
+    if (k != 2 && k != 8) {  // prim-headed parts (COUNT and COLLECT are not)
       uint8_t v = *p++;
       double x = 0;
       if (v) {
@@ -529,6 +586,16 @@ __device__ __forceinline__ void agg_parse_frozen(uint32_t layout,
     if (k == 6) {  // FIRST flag byte: 0 = untouched, 2 = touched
       uint8_t fl = *p++;
       if (a->f_st == 0 && fl) a->f_st = 1;  // touched but first value null
+    }
+    if (k == 8) {  // COLLECT_LIST raw list: expose, then skip
+      int used;
+      uint64_t raw = read_varint_dev(p, &used);
+      p += used;
+      if (!a->c_raw) {
+        a->c_raw = p;
+        a->c_cnt = (uint32_t)(raw / 8);
+      }
+      p += raw;
     }
     if (k == 2 || k == 3) {
       int used;
@@ -568,6 +635,13 @@ __global__ void k_agg_merge_frozen(const AggTable t,
     if (t.f_row) {  // pass A: earliest TOUCHED record wins (first.rs:198-207)
       if (acc.f_st) atomicMin(&t.f_row[2 * a], (unsigned long long)row);
       if (acc.fn_st) atomicMin(&t.f_row[2 * a + 1], (unsigned long long)row);
+    }
+    if (t.c_key && acc.c_cnt) {  // collect.rs:139-158: concatenate in merge
+      for (uint32_t j = 0; j < acc.c_cnt; j++) {  // order, items in order
+        double v;
+        memcpy(&v, acc.c_raw + (size_t)j * 8, 8);
+        coll_append(t, knull ? 0 : keys[i], knull, (row << 20) | j, v);
+      }
     }
   }
 }
@@ -671,7 +745,10 @@ __global__ void k_agg_gather_out(const AggTable t,
   }
 }
 
-// snapshot one slot's accumulators for freeze
+// snapshot one slot's accumulators for freeze. When the collect pool is
+// active (and sorted into {c_key ascending, prio ascending within key} by
+// the engine before any freeze launch), the group's value run is located by
+// binary search; the null-key special group owns the back segment.
 __device__ __forceinline__ AccSnap table_snap(const AggTable& t, uint32_t s) {
   AccSnap a;
   a.valid = t.slots[s].cnt != 0;
@@ -686,6 +763,29 @@ __device__ __forceinline__ AccSnap table_snap(const AggTable& t, uint32_t s) {
     a.f_val = t.f_val[2 * s];
     a.fn_st = t.f_st[2 * s + 1];
     a.fn_val = t.f_val[2 * s + 1];
+  }
+  if (t.c_key) {
+    if (s == (uint32_t)t.cap + 1) {  // null-key group = back segment,
+      int64_t n1 = (int64_t)t.c_n[1];  // already prio-sorted in place
+      a.c_vals = t.c_val + (t.c_cap - n1);
+      a.c_cnt = (uint32_t)n1;
+    } else {
+      long long key = (s == (uint32_t)t.cap) ? (long long)INT64_MIN
+                                             : t.slots[s].key;
+      int64_t n0 = (int64_t)t.c_n[0];
+      int64_t lo = 0, hi = n0;
+      while (lo < hi) {  // lower bound
+        int64_t mid = (lo + hi) >> 1;
+        if (t.c_key[mid] < key) lo = mid + 1; else hi = mid;
+      }
+      int64_t lo2 = lo, hi2 = n0;
+      while (lo2 < hi2) {  // upper bound
+        int64_t mid = (lo2 + hi2) >> 1;
+        if (t.c_key[mid] <= key) lo2 = mid + 1; else hi2 = mid;
+      }
+      a.c_vals = t.c_val + lo;
+      a.c_cnt = (uint32_t)(lo2 - lo);
+    }
   }
   return a;
 }
@@ -784,6 +884,7 @@ __device__ __forceinline__ AccSnap row_snap(bool v, double val,
   a.f_val = val;
   a.fn_st = v ? 2 : 0;
   a.fn_val = val;
+  a.c_cnt = v ? 1 : 0;  // caller points c_vals at the row's value bits
   return a;
 }
 
@@ -805,8 +906,11 @@ __global__ void k_skip_freeze_write(const double* __restrict__ vals,
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     bool v = !val_valid || bit_get_dev(val_valid, i);
-    agg_freeze_write_rec(layout, row_snap(v, v ? vals[i] : 0.0, is_int),
-                         out + offsets[i], is_int);
+    AccSnap a = row_snap(v, v ? vals[i] : 0.0, is_int);
+    unsigned long long vb;
+    memcpy(&vb, &vals[i], 8);
+    a.c_vals = &vb;  // single-item list for the row's own group
+    agg_freeze_write_rec(layout, a, out + offsets[i], is_int);
   }
 }
 
@@ -995,6 +1099,73 @@ void launch_first_gather(const AggTable& t, const uint32_t* order_slots,
                      0, s, t, order_slots, num_groups, which, out_vals,
                      out_validity);
   check_launch("k_first_gather");
+}
+
+__global__ void k_gather_u64_idx(const unsigned long long* __restrict__ src,
+                                 const uint32_t* __restrict__ idx, int64_t n,
+                                 unsigned long long* __restrict__ dst) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    dst[i] = src[idx[i]];
+}
+
+void launch_gather_u64_idx(const unsigned long long* src, const uint32_t* idx,
+                           int64_t n, unsigned long long* dst, hipStream_t s) {
+  hipLaunchKernelGGL(k_gather_u64_idx, dim3(grid_for(n)), dim3(BLOCK), 0, s,
+                     src, idx, n, dst);
+  check_launch("k_gather_u64_idx");
+}
+
+// sign-bias i64 keys so unsigned radix order == signed order
+__global__ void k_bias_i64(const unsigned long long* __restrict__ src,
+                           int64_t n, unsigned long long* __restrict__ dst) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    dst[i] = src[i] ^ 0x8000000000000000ull;
+}
+
+void launch_bias_i64(const unsigned long long* src, int64_t n,
+                     unsigned long long* dst, hipStream_t s) {
+  hipLaunchKernelGGL(k_bias_i64, dim3(grid_for(n)), dim3(BLOCK), 0, s, src, n,
+                     dst);
+  check_launch("k_bias_i64");
+}
+
+// per-group collect counts / values (emit side; pool sorted by the engine)
+__global__ void k_coll_counts(const AggTable t,
+                              const uint32_t* __restrict__ order_slots,
+                              int64_t num_groups, int32_t* __restrict__ cnts) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < num_groups; i += (int64_t)gridDim.x * blockDim.x)
+    cnts[i] = (int32_t)table_snap(t, order_slots[i]).c_cnt;
+}
+
+void launch_coll_counts(const AggTable& t, const uint32_t* order_slots,
+                        int64_t num_groups, int32_t* cnts, hipStream_t s) {
+  hipLaunchKernelGGL(k_coll_counts, dim3(grid_for(num_groups)), dim3(BLOCK), 0,
+                     s, t, order_slots, num_groups, cnts);
+  check_launch("k_coll_counts");
+}
+
+__global__ void k_coll_gather(const AggTable t,
+                              const uint32_t* __restrict__ order_slots,
+                              int64_t num_groups,
+                              const int32_t* __restrict__ offsets,
+                              unsigned long long* __restrict__ out) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < num_groups; i += (int64_t)gridDim.x * blockDim.x) {
+    AccSnap a = table_snap(t, order_slots[i]);
+    unsigned long long* dst = out + offsets[i];
+    for (uint32_t j = 0; j < a.c_cnt; j++) dst[j] = a.c_vals[j];
+  }
+}
+
+void launch_coll_gather(const AggTable& t, const uint32_t* order_slots,
+                        int64_t num_groups, const int32_t* offsets,
+                        unsigned long long* out, hipStream_t s) {
+  hipLaunchKernelGGL(k_coll_gather, dim3(grid_for(num_groups)), dim3(BLOCK), 0,
+                     s, t, order_slots, num_groups, offsets, out);
+  check_launch("k_coll_gather");
 }
 
 __global__ void k_iota_u32(uint32_t* __restrict__ dst, int64_t n) {

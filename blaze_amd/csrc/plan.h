@@ -79,6 +79,8 @@ enum AggFunction : int32_t {
   AGG_SUM = 2,
   AGG_AVG = 3,
   AGG_COUNT = 4,
+  AGG_COLLECT_LIST = 5,
+  AGG_COLLECT_SET = 6,
   AGG_FIRST = 7,
   AGG_FIRST_IGNORES_NULL = 8,
 };

@@ -141,9 +141,11 @@ def sum_count_aggs(val_index=1):
     ]
 
 
+AGG_COLLECT_LIST, AGG_COLLECT_SET = 5, 6
 _AGG_FN = {"min": AGG_MIN, "max": AGG_MAX, "sum": AGG_SUM,
            "avg": AGG_AVG, "count": AGG_COUNT, "first": AGG_FIRST,
-           "first_ignores_null": AGG_FIRST_IGNORES_NULL}
+           "first_ignores_null": AGG_FIRST_IGNORES_NULL,
+           "collect_list": AGG_COLLECT_LIST, "collect_set": AGG_COLLECT_SET}
 
 
 def named_aggs(names, val_index=1, val_dt=DT_FLOAT64):

@@ -384,13 +384,16 @@ def int_sum_groups(keys, vals, val_valid=None):
 # raw LE 8-byte values (scalar_serde.rs:35-47 write_prim non-nullable).
 # Merge (collect.rs:139-158) concatenates in merge arrival order (set: with
 # dedup). CPU parity anchor for the round-2 device implementation.
-def collect_groups(keys, vals, val_valid=None, distinct=False):
+def collect_groups(keys, vals, val_valid=None, distinct=False,
+                   key_valid=None):
     """Insertion-ordered (keys, lists): lists hold non-null values in
-    arrival order; distinct=True dedups keeping first occurrence."""
+    arrival order; distinct=True dedups keeping first occurrence. None key
+    = the null-key group."""
     groups = {}
     n = len(keys)
     for i in range(n):
-        k = int(keys[i])
+        k = None if (key_valid is not None and not key_valid[i]) \
+            else int(keys[i])
         if k not in groups:
             groups[k] = ([], set())
         if val_valid is None or val_valid[i]:
