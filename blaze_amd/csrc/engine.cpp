@@ -421,6 +421,11 @@ class AggOp {
         case AGG_FIRST: k = AGGL_FIRST; has_first_ = true; break;
         case AGG_FIRST_IGNORES_NULL: k = AGGL_FIRSTIN; has_first_ = true; break;
         case AGG_COLLECT_LIST: k = AGGL_CLIST; has_coll_ = true; break;
+        case AGG_COLLECT_SET:
+          k = AGGL_CSET;
+          has_coll_ = true;
+          coll_set_ = true;
+          break;
         default:
           FAIL("AggExec: only SUM/COUNT/AVG/MIN/MAX/FIRST[_IGNORES_NULL]/"
                "COLLECT_LIST aggregates on this path");
@@ -456,6 +461,16 @@ class AggOp {
           FAIL("all aggregates must share one argument column on this path");
         }
       }
+    }
+    {
+      bool has_l = false, has_s = false;
+      for (uint32_t k : agg_kinds_) {
+        has_l |= (k == AGGL_CLIST);
+        has_s |= (k == AGGL_CSET);
+      }
+      if (has_l && has_s)
+        FAIL("COLLECT_LIST and COLLECT_SET together share one argument pool "
+             "— unsupported on this path");
     }
     batch_size_ = conf.get_i("BATCH_SIZE", 10000);
     if (batch_size_ <= 0) FAIL("invalid BATCH_SIZE conf");
@@ -665,7 +680,7 @@ class AggOp {
           f.push_back({agg_names_[i], DType::Int64, false});
         else if (agg_kinds_[i] == AGGL_AVG)
           f.push_back({agg_names_[i], DType::Float64, true});
-        else if (agg_kinds_[i] == AGGL_CLIST)
+        else if (agg_kinds_[i] == AGGL_CLIST || agg_kinds_[i] == AGGL_CSET)
           // collect.rs:110-112: nullable() = false (empty lists, not nulls)
           f.push_back({agg_names_[i], vdt, false, /*is_list=*/true});
         else
@@ -693,61 +708,164 @@ class AggOp {
     AURON_HIP(hipMemcpyAsync(errp, t_.error_flag, 4, hipMemcpyDeviceToHost,
                              stream_));
     AURON_HIP(hipStreamSynchronize(stream_));
-    coll_n0_ = (int64_t)pinned_meta_.get<unsigned long long>()[0];
-    coll_n1_ = (int64_t)pinned_meta_.get<unsigned long long>()[1];
+    int64_t n0 = (int64_t)pinned_meta_.get<unsigned long long>()[0];
+    int64_t n1 = (int64_t)pinned_meta_.get<unsigned long long>()[1];
     if (*errp & 4u)
       FAIL("collect pool overflow: raise AURON_HIP_COLLECT_POOL");
-    int64_t n0 = coll_n0_, n1 = coll_n1_;
+    coll_n0_ = n0;
+    coll_n1_ = n1;
+    int64_t nmax = std::max<int64_t>(std::max(n0, n1), 1);
+    size_t tb0 = 0;
+    sort_pairs_u64_u32(nullptr, nullptr, nullptr, nullptr, nmax, nullptr,
+                       &tb0, stream_);
+    DevBuf tmp(tb0), idx(nmax * 4), idxo(nmax * 4), scr(nmax * 8);
+
+    // one stable radix pass over `by`, permuting key/prio/val in place
+    // (ping-pong through scratch buffers)
+    auto sort_pass = [&](unsigned long long* by, int64_t n,
+                         unsigned long long* key, unsigned long long* prio,
+                         unsigned long long* val, DevBuf& alt_key,
+                         DevBuf& alt_prio, DevBuf& alt_val) {
+      launch_iota_u32(idx.get<uint32_t>(), n, stream_);
+      size_t tb = tmp.size();
+      sort_pairs_u64_u32(by, idx.get<uint32_t>(),
+                         scr.get<unsigned long long>(), idxo.get<uint32_t>(),
+                         n, tmp.get(), &tb, stream_);
+      if (key)
+        launch_gather_u64_idx(key, idxo.get<uint32_t>(), n,
+                              alt_key.get<unsigned long long>(), stream_);
+      launch_gather_u64_idx(prio, idxo.get<uint32_t>(), n,
+                            alt_prio.get<unsigned long long>(), stream_);
+      launch_gather_u64_idx(val, idxo.get<uint32_t>(), n,
+                            alt_val.get<unsigned long long>(), stream_);
+    };
+
     d_cskey_.alloc(std::max<int64_t>(n0, 1) * 8);
     d_csval_.alloc(std::max<int64_t>(n0 + n1, 1) * 8);
-    size_t tb0 = 0;
-    sort_pairs_u64_u32(nullptr, nullptr, nullptr, nullptr,
-                       std::max<int64_t>(std::max(n0, n1), 1), nullptr, &tb0,
-                       stream_);
-    DevBuf tmp(tb0);
     if (n0 > 0) {
-      DevBuf idx(n0 * 4), idxo(n0 * 4), k64(n0 * 8), ks(n0 * 8), v1(n0 * 8);
+      // working copies (ping-pong pairs)
+      DevBuf ka(n0 * 8), kb(n0 * 8), pa(n0 * 8), pb(n0 * 8), va(n0 * 8),
+          vb(n0 * 8), biased(n0 * 8);
+      AURON_HIP(hipMemcpyAsync(ka.get(), t_.c_key, n0 * 8,
+                               hipMemcpyDeviceToDevice, stream_));
+      AURON_HIP(hipMemcpyAsync(pa.get(), t_.c_prio, n0 * 8,
+                               hipMemcpyDeviceToDevice, stream_));
+      AURON_HIP(hipMemcpyAsync(va.get(), t_.c_val, n0 * 8,
+                               hipMemcpyDeviceToDevice, stream_));
+      auto K = [&](DevBuf& b) { return b.get<unsigned long long>(); };
       // pass 1: prio order
-      launch_iota_u32(idx.get<uint32_t>(), n0, stream_);
-      size_t tb = tmp.size();
-      sort_pairs_u64_u32(t_.c_prio, idx.get<uint32_t>(),
-                         k64.get<unsigned long long>() /* scratch keys out */,
-                         idxo.get<uint32_t>(), n0, tmp.get(), &tb, stream_);
-      launch_gather_u64_idx((const unsigned long long*)t_.c_key,
-                            idxo.get<uint32_t>(), n0,
-                            ks.get<unsigned long long>(), stream_);
-      launch_gather_u64_idx(t_.c_val, idxo.get<uint32_t>(), n0,
-                            v1.get<unsigned long long>(), stream_);
-      // pass 2: stable key order (sign-biased radix) keeps prio order inside
-      launch_bias_i64(ks.get<unsigned long long>(), n0,
-                      k64.get<unsigned long long>(), stream_);
-      launch_iota_u32(idx.get<uint32_t>(), n0, stream_);
-      DevBuf biased_out(n0 * 8);
-      tb = tmp.size();
-      sort_pairs_u64_u32(k64.get<unsigned long long>(), idx.get<uint32_t>(),
-                         biased_out.get<unsigned long long>(),
-                         idxo.get<uint32_t>(), n0, tmp.get(), &tb, stream_);
-      launch_gather_u64_idx(ks.get<unsigned long long>(), idxo.get<uint32_t>(),
-                            n0, d_cskey_.get<unsigned long long>(), stream_);
-      launch_gather_u64_idx(v1.get<unsigned long long>(), idxo.get<uint32_t>(),
-                            n0, d_csval_.get<unsigned long long>(), stream_);
+      sort_pass(K(pa), n0, K(ka), K(pa), K(va), kb, pb, vb);
+      std::swap(ka, kb); std::swap(pa, pb); std::swap(va, vb);
+      if (coll_set_) {
+        // AccSet dedup (collect.rs AccSet.append: only novel values):
+        // sort by value then key (stable keeps prio asc inside runs),
+        // keep each (key,value) run head = first occurrence
+        sort_pass(K(va), n0, K(ka), K(pa), K(va), kb, pb, vb);
+        std::swap(ka, kb); std::swap(pa, pb); std::swap(va, vb);
+        launch_bias_i64(K(ka), n0, biased.get<unsigned long long>(), stream_);
+        sort_pass(biased.get<unsigned long long>(), n0, K(ka), K(pa), K(va),
+                  kb, pb, vb);
+        std::swap(ka, kb); std::swap(pa, pb); std::swap(va, vb);
+        DevBuf mark(n0), pos((n0 + 1) * 4);
+        launch_coll_mark_heads((const long long*)K(ka), K(va), n0, 1,
+                               mark.get<uint8_t>(), stream_);
+        size_t stb = 0;
+        scan_mask_u8(mark.get<uint8_t>(), pos.get<uint32_t>(), n0, nullptr,
+                     &stb, stream_);
+        DevBuf stmp(stb);
+        scan_mask_u8(mark.get<uint8_t>(), pos.get<uint32_t>(), n0, stmp.get(),
+                     &stb, stream_);
+        for (auto arr : {&ka, &pa, &va}) {
+          launch_compact_u64(K(*arr), mark.get<uint8_t>(),
+                             pos.get<uint32_t>(), n0,
+                             scr.get<unsigned long long>(), stream_);
+          AURON_HIP(hipMemcpyAsync(arr->get(), scr.get(), n0 * 8,
+                                   hipMemcpyDeviceToDevice, stream_));
+        }
+        // new count = scan total (slot n0 of pos)
+        AURON_HIP(hipMemcpyAsync(pinned_meta_.get(),
+                                 pos.get<uint32_t>() + n0, 4,
+                                 hipMemcpyDeviceToHost, stream_));
+        AURON_HIP(hipStreamSynchronize(stream_));
+        n0 = (int64_t)*pinned_meta_.get<uint32_t>();
+        // device counter must match the binary-search bound
+        unsigned long long nn = (unsigned long long)n0;
+        AURON_HIP(hipMemcpyAsync(d_cn_.get(), &nn, 8, hipMemcpyHostToDevice,
+                                 stream_));
+        if (n0 > 0) {
+          // restore prio order for the final key pass
+          sort_pass(K(pa), n0, K(ka), K(pa), K(va), kb, pb, vb);
+          std::swap(ka, kb); std::swap(pa, pb); std::swap(va, vb);
+        }
+      }
+      if (n0 > 0) {
+        // final pass: sign-biased key order, stable (prio asc within key)
+        launch_bias_i64(K(ka), n0, biased.get<unsigned long long>(), stream_);
+        launch_iota_u32(idx.get<uint32_t>(), n0, stream_);
+        size_t tb = tmp.size();
+        sort_pairs_u64_u32(biased.get<unsigned long long>(),
+                           idx.get<uint32_t>(),
+                           scr.get<unsigned long long>(),
+                           idxo.get<uint32_t>(), n0, tmp.get(), &tb, stream_);
+        launch_gather_u64_idx(K(ka), idxo.get<uint32_t>(), n0,
+                              d_cskey_.get<unsigned long long>(), stream_);
+        launch_gather_u64_idx(K(va), idxo.get<uint32_t>(), n0,
+                              d_csval_.get<unsigned long long>(), stream_);
+      }
+      coll_n0_ = n0;
     }
     if (n1 > 0) {
-      // null-key back segment: prio order only
-      DevBuf idx(n1 * 4), idxo(n1 * 4), po(n1 * 8);
-      launch_iota_u32(idx.get<uint32_t>(), n1, stream_);
-      size_t tb = tmp.size();
-      sort_pairs_u64_u32(t_.c_prio + (coll_cap_ - n1), idx.get<uint32_t>(),
-                         po.get<unsigned long long>(), idxo.get<uint32_t>(),
-                         n1, tmp.get(), &tb, stream_);
-      launch_gather_u64_idx(t_.c_val + (coll_cap_ - n1), idxo.get<uint32_t>(),
-                            n1, d_csval_.get<unsigned long long>() + n0,
-                            stream_);
+      // null-key back segment: prio order (+ value dedup for SET)
+      DevBuf pa(n1 * 8), pb(n1 * 8), va(n1 * 8), vb(n1 * 8);
+      AURON_HIP(hipMemcpyAsync(pa.get(), t_.c_prio + (coll_cap_ - n1), n1 * 8,
+                               hipMemcpyDeviceToDevice, stream_));
+      AURON_HIP(hipMemcpyAsync(va.get(), t_.c_val + (coll_cap_ - n1), n1 * 8,
+                               hipMemcpyDeviceToDevice, stream_));
+      auto K = [&](DevBuf& b) { return b.get<unsigned long long>(); };
+      sort_pass(K(pa), n1, nullptr, K(pa), K(va), pb, pb, vb);
+      std::swap(pa, pb); std::swap(va, vb);
+      if (coll_set_) {
+        sort_pass(K(va), n1, nullptr, K(pa), K(va), pb, pb, vb);
+        std::swap(pa, pb); std::swap(va, vb);
+        DevBuf mark(n1), pos((n1 + 1) * 4);
+        launch_coll_mark_heads(nullptr, K(va), n1, 0, mark.get<uint8_t>(),
+                               stream_);
+        size_t stb = 0;
+        scan_mask_u8(mark.get<uint8_t>(), pos.get<uint32_t>(), n1, nullptr,
+                     &stb, stream_);
+        DevBuf stmp(stb);
+        scan_mask_u8(mark.get<uint8_t>(), pos.get<uint32_t>(), n1, stmp.get(),
+                     &stb, stream_);
+        for (auto arr : {&pa, &va}) {
+          launch_compact_u64(K(*arr), mark.get<uint8_t>(),
+                             pos.get<uint32_t>(), n1,
+                             scr.get<unsigned long long>(), stream_);
+          AURON_HIP(hipMemcpyAsync(arr->get(), scr.get(), n1 * 8,
+                                   hipMemcpyDeviceToDevice, stream_));
+        }
+        AURON_HIP(hipMemcpyAsync(pinned_meta_.get(),
+                                 pos.get<uint32_t>() + n1, 4,
+                                 hipMemcpyDeviceToHost, stream_));
+        AURON_HIP(hipStreamSynchronize(stream_));
+        n1 = (int64_t)*pinned_meta_.get<uint32_t>();
+        unsigned long long nn = (unsigned long long)n1;
+        AURON_HIP(hipMemcpyAsync(d_cn_.get<uint8_t>() + 8, &nn, 8,
+                                 hipMemcpyHostToDevice, stream_));
+        if (n1 > 0) {
+          sort_pass(K(pa), n1, nullptr, K(pa), K(va), pb, pb, vb);
+          std::swap(pa, pb); std::swap(va, vb);
+        }
+      }
+      if (n1 > 0)
+        AURON_HIP(hipMemcpyAsync(
+            d_csval_.get<unsigned long long>() + coll_n0_, va.get(), n1 * 8,
+            hipMemcpyDeviceToDevice, stream_));
+      coll_n1_ = n1;
     }
     AURON_HIP(hipStreamSynchronize(stream_));  // temps return to the pool
     t_.c_key = d_cskey_.get<long long>();
     t_.c_val = d_csval_.get<unsigned long long>();
-    t_.c_cap = n0 + n1;  // null segment now ends at n0+n1 (table_snap math)
+    t_.c_cap = coll_n0_ + coll_n1_;  // null segment ends the sorted view
   }
 
   std::vector<std::pair<int64_t, std::vector<HostOutCol>>> finish() {
@@ -1310,7 +1428,8 @@ class AggOp {
       std::vector<int32_t> h_coff;
       std::vector<uint8_t> h_citems;
       bool need_coll = false;
-      for (uint32_t k : agg_kinds_) need_coll |= (k == AGGL_CLIST);
+      for (uint32_t k : agg_kinds_)
+        need_coll |= (k == AGGL_CLIST || k == AGGL_CSET);
       if (need_coll) {
         DevBuf cnts(n * 4);
         launch_coll_counts(t_, order_slots, n, cnts.get<int32_t>(), stream_);
@@ -1360,7 +1479,7 @@ class AggOp {
           ac.dt = vdt;
           ac.values = h_firsts[w];
           attach_validity(&ac, fv[w], n);
-        } else if (k == AGGL_CLIST) {
+        } else if (k == AGGL_CLIST || k == AGGL_CSET) {
           ac.dt = vdt;
           ac.is_list = true;
           ac.offsets = h_coff;
@@ -1469,7 +1588,8 @@ class AggOp {
   DType key_dt_ = DType::Unsupported;
   uint32_t layout_ = 0;
   bool has_mm_ = false;  // agg list contains MIN/MAX: side mm array active
-  bool has_coll_ = false, coll_sorted_ = false;  // COLLECT_LIST pool active
+  bool has_coll_ = false, coll_sorted_ = false;  // COLLECT pool active
+  bool coll_set_ = false;  // COLLECT_SET: dedup at prepare
   int64_t coll_cap_ = 0, conf_coll_cap_ = 16 << 20;
   int64_t coll_n0_ = 0, coll_n1_ = 0;
   bool val_is_int_ = false, val_typed_seen_ = false;  // i64 accumulator mode

@@ -127,7 +127,7 @@ void launch_agg_update(const AggTable& t, const int64_t* keys,
 // duplicated parts repeat the shared accumulators.
 enum AggLayoutKind : uint32_t {
   AGGL_SUM = 1, AGGL_CNT = 2, AGGL_AVG = 3, AGGL_MIN = 4, AGGL_MAX = 5,
-  AGGL_FIRST = 6, AGGL_FIRSTIN = 7, AGGL_CLIST = 8
+  AGGL_FIRST = 6, AGGL_FIRSTIN = 7, AGGL_CLIST = 8, AGGL_CSET = 9
 };
 
 // merge rows of frozen partial state per the layout
@@ -204,6 +204,13 @@ void launch_coll_counts(const AggTable& t, const uint32_t* order_slots,
 void launch_coll_gather(const AggTable& t, const uint32_t* order_slots,
                         int64_t num_groups, const int32_t* offsets,
                         unsigned long long* out, hipStream_t s);
+// COLLECT_SET dedup: mark (key,value)-run heads; compact by scanned marks
+void launch_coll_mark_heads(const long long* key,
+                            const unsigned long long* val, int64_t n,
+                            int use_key, uint8_t* mark, hipStream_t s);
+void launch_compact_u64(const unsigned long long* src, const uint8_t* mark,
+                        const uint32_t* pos, int64_t n,
+                        unsigned long long* dst, hipStream_t s);
 
 
 // FIRST/FIRST_IGNORES_NULL pass B over update-mode rows: the unique row

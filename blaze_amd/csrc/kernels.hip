@@ -475,7 +475,7 @@ __device__ __forceinline__ int agg_freeze_len(uint32_t layout,
     if (k == 5) len += 1 + (a.maxu != MM_MAX_INIT ? 8 : 0);
     if (k == 6) len += 2 + (a.f_st == 2 ? 8 : 0);  // prim part + flag byte
     if (k == 7) len += 1 + (a.fn_st == 2 ? 8 : 0);
-    if (k == 8)  // collect.rs:237-241: varint(raw_len) ++ raw values
+    if (k == 8 || k == 9)  // collect.rs:237-241: varint(raw_len) ++ values
       len += varint_len_dev((uint64_t)a.c_cnt * 8) + (int)a.c_cnt * 8;
   }
   return len;
@@ -516,7 +516,7 @@ __device__ __forceinline__ uint8_t* agg_freeze_write_rec(uint32_t layout,
     }
     if (k == 7)  // FIRST_IGNORES_NULL: prim value only
       p = agg_prim_freeze_part(a.fn_st == 2, a.fn_val, p);
-    if (k == 8) {  // COLLECT_LIST raw list (collect.rs:237-241)
+    if (k == 8 || k == 9) {  // COLLECT raw list (collect.rs:237-241)
       uint64_t raw = (uint64_t)a.c_cnt * 8;
       int used = varint_len_dev(raw);
       for (uint64_t v = raw; ; v /= 128) {
@@ -556,7 +556,7 @@ __device__ __forceinline__ void agg_parse_frozen(uint32_t layout,
   uint32_t got = 0;  // bit per family
   for (uint32_t l = layout; l & 15u; l >>= 4) {
     uint32_t k = l & 15u;
-    if (k != 2 && k != 8) {  // prim-headed parts (COUNT and COLLECT are not)
+    if (k != 2 && k != 8 && k != 9) {  // prim-headed parts
       uint8_t v = *p++;
       double x = 0;
       if (v) {
@@ -587,7 +587,7 @@ __device__ __forceinline__ void agg_parse_frozen(uint32_t layout,
       uint8_t fl = *p++;
       if (a->f_st == 0 && fl) a->f_st = 1;  // touched but first value null
     }
-    if (k == 8) {  // COLLECT_LIST raw list: expose, then skip
+    if (k == 8 || k == 9) {  // COLLECT raw list: expose, then skip
       int used;
       uint64_t raw = read_varint_dev(p, &used);
       p += used;
@@ -1166,6 +1166,47 @@ void launch_coll_gather(const AggTable& t, const uint32_t* order_slots,
   hipLaunchKernelGGL(k_coll_gather, dim3(grid_for(num_groups)), dim3(BLOCK), 0,
                      s, t, order_slots, num_groups, offsets, out);
   check_launch("k_coll_gather");
+}
+
+// COLLECT_SET dedup helpers: entries sorted {key, value, prio} — mark each
+// (key,value) run head (it carries the minimum prio = first occurrence,
+// AccSet.append semantics: only novel values append), then compact by the
+// scanned positions.
+__global__ void k_coll_mark_heads(const long long* __restrict__ key,
+                                  const unsigned long long* __restrict__ val,
+                                  int64_t n, int use_key,
+                                  uint8_t* __restrict__ mark) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    mark[i] = (i == 0 || val[i] != val[i - 1] ||
+               (use_key && key[i] != key[i - 1]))
+                  ? 1
+                  : 0;
+}
+
+void launch_coll_mark_heads(const long long* key,
+                            const unsigned long long* val, int64_t n,
+                            int use_key, uint8_t* mark, hipStream_t s) {
+  hipLaunchKernelGGL(k_coll_mark_heads, dim3(grid_for(n)), dim3(BLOCK), 0, s,
+                     key, val, n, use_key, mark);
+  check_launch("k_coll_mark_heads");
+}
+
+__global__ void k_compact_u64(const unsigned long long* __restrict__ src,
+                              const uint8_t* __restrict__ mark,
+                              const uint32_t* __restrict__ pos, int64_t n,
+                              unsigned long long* __restrict__ dst) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    if (mark[i]) dst[pos[i]] = src[i];
+}
+
+void launch_compact_u64(const unsigned long long* src, const uint8_t* mark,
+                        const uint32_t* pos, int64_t n,
+                        unsigned long long* dst, hipStream_t s) {
+  hipLaunchKernelGGL(k_compact_u64, dim3(grid_for(n)), dim3(BLOCK), 0, s, src,
+                     mark, pos, n, dst);
+  check_launch("k_compact_u64");
 }
 
 __global__ void k_iota_u32(uint32_t* __restrict__ dst, int64_t n) {

@@ -132,3 +132,61 @@ def test_collect_list_grow_and_skipping():
     np.testing.assert_array_equal(k2[gi], keys2)
     flat = np.concatenate([gl2[int(i)] for i in gi])
     np.testing.assert_array_equal(flat, vals)
+
+
+def test_collect_set_dedup():
+    """COLLECT_SET: distinct values per group, FIRST-occurrence order
+    (AccSet.append appends only novel values)."""
+    rng = np.random.default_rng(53)
+    n = 40_000
+    keys = rng.integers(0, 300, n).astype(np.int64)
+    vals = rng.integers(0, 25, n).astype(np.float64)  # heavy duplication
+    vv = rng.random(n) >= 0.1
+    got = _run(["collect_set", "count"], keys, vals, vv)
+    ok, sets = oracle.collect_groups(keys, vals, vv, distinct=True)
+    np.testing.assert_array_equal(got["key"], np.array(ok))
+    gl = _concat_lists(got["collect_set"])
+    assert len(gl) == len(sets)
+    for g, e in zip(gl, sets):
+        np.testing.assert_array_equal(g, np.array(e))
+
+
+def test_collect_set_i64_null_keys_multistage():
+    """SET over i64 values with null keys, chained partial->final (dedup
+    happens at both stages; the frozen set wire re-merges losslessly)."""
+    rng = np.random.default_rng(59)
+    n = 15_000
+    keys = rng.integers(0, 40, n).astype(np.int64)
+    kv = rng.random(n) >= 0.15
+    vals = rng.integers(-8, 8, n).astype(np.int64)
+    vv = rng.random(n) >= 0.25
+    names = ["collect_set"]
+    batches = [[(keys[i:i + 777], kv[i:i + 777]),
+                (vals[i:i + 777], vv[i:i + 777])]
+               for i in range(0, n, 777)]
+    reader = plan.ffi_reader([plan.field("key", plan.DT_INT64, True),
+                              plan.field("val", plan.DT_INT64, True)],
+                             "input0")
+    partial = plan.agg(reader, [plan.column("key", 0)],
+                       plan.named_aggs(names, val_dt=plan.DT_INT64),
+                       [plan.MODE_PARTIAL], ["key"], names)
+    final = plan.agg(partial, [plan.column("key", 0)],
+                     plan.named_aggs(names, val_dt=plan.DT_INT64),
+                     [plan.MODE_FINAL], ["key"], names)
+    t = blaze_amd.Task(plan.task_definition(final), batches=batches)
+    outs = t.run()
+    got_keys = np.concatenate([ob[0]["values"] for ob in outs])
+    got_kv = np.concatenate(
+        [ob[0]["valid"] if ob[0]["valid"] is not None
+         else np.ones(len(ob[0]["values"]), bool) for ob in outs])
+    gl = _concat_lists([ob[1] for ob in outs])
+    t.finalize()
+    ok, sets = oracle.collect_groups(keys, vals, vv, distinct=True,
+                                     key_valid=kv)
+    exp_kv = np.array([k is not None for k in ok])
+    np.testing.assert_array_equal(got_kv, exp_kv)
+    np.testing.assert_array_equal(
+        got_keys[exp_kv], np.array([k for k in ok if k is not None]))
+    assert len(gl) == len(sets)
+    for g, e in zip(gl, sets):
+        np.testing.assert_array_equal(g, np.array(e, np.int64))
