@@ -710,7 +710,9 @@ class AggOp {
     AURON_HIP(hipStreamSynchronize(stream_));
     int64_t n0 = (int64_t)pinned_meta_.get<unsigned long long>()[0];
     int64_t n1 = (int64_t)pinned_meta_.get<unsigned long long>()[1];
-    if (*errp & 4u)
+    // the per-append guard reads the opposing counter non-atomically (an
+    // approximate early trip); this is the exact post-hoc check
+    if ((*errp & 4u) || n0 + n1 > coll_cap_)
       FAIL("collect pool overflow: raise AURON_HIP_COLLECT_POOL");
     coll_n0_ = n0;
     coll_n1_ = n1;

@@ -129,3 +129,9 @@ def test_decode_rejects_truncated_plan():
                                      len(out))
     # must not crash; either a decode error or a partial tree is reported
     assert rc != 0 or len(out.value) >= 0
+
+
+def test_decode_collect_plan():
+    td = plan.plan_partial_final_named(["collect_list", "collect_set"])
+    s = blaze_amd.debug_decode_plan(td)
+    assert "fn5" in s and "fn6" in s  # COLLECT_LIST=5, COLLECT_SET=6
