@@ -135,3 +135,28 @@ def test_decode_collect_plan():
     td = plan.plan_partial_final_named(["collect_list", "collect_set"])
     s = blaze_amd.debug_decode_plan(td)
     assert "fn5" in s and "fn6" in s  # COLLECT_LIST=5, COLLECT_SET=6
+
+
+def test_decode_garbage_never_crashes():
+    """The hand-rolled proto3 decoder must reject arbitrary garbage and
+    truncations without crashing (it fronts the C ABI)."""
+    import ctypes
+
+    import numpy as np
+    rng = np.random.default_rng(71)
+    lib = blaze_amd.lib()
+    out = ctypes.create_string_buffer(1 << 12)
+    td = plan.plan_partial_final()
+    cases = []
+    for i in range(40):
+        n = int(rng.integers(0, 200))
+        cases.append(rng.integers(0, 256, n).astype(np.uint8).tobytes())
+    for i in range(1, len(td), 7):   # truncations of a real plan
+        cases.append(td[:i])
+    for i in range(30):              # bit-flipped real plans
+        b = bytearray(td)
+        for _ in range(int(rng.integers(1, 6))):
+            b[int(rng.integers(0, len(b)))] ^= int(rng.integers(1, 256))
+        cases.append(bytes(b))
+    for blob in cases:
+        lib.auron_debug_decode_plan(blob, len(blob), out, len(out))

@@ -127,3 +127,63 @@ def test_parquet_snappy_overlap_and_runs(tmp_path):
     assert sum(rows) == n, s
     assert sum(nulls) == int(mask.sum()), s
     assert "dict=50" in s, s  # 50 distinct values -> dict encoding held
+
+
+def test_parquet_decode_fuzz_cpu():
+    """Randomized parquet files (codecs x schemas x row-group sizes x null
+    fractions x page sizes) must decode on the host with exact row/null
+    counts vs pyarrow's ground truth."""
+    pa = pytest.importorskip("pyarrow")
+    import pyarrow.parquet as pq
+    import re
+    import numpy as np
+    import ctypes
+
+    lib = blaze_amd.lib()
+    lib.auron_debug_parquet_summary.restype = ctypes.c_int32
+    lib.auron_debug_parquet_summary.argtypes = [ctypes.c_char_p,
+                                               ctypes.c_char_p,
+                                               ctypes.c_size_t]
+    rng = np.random.default_rng(83)
+    import tempfile, os
+    for trial in range(8):
+        n = int(rng.integers(1, 60_000))
+        codec = ["snappy", "zstd", "lz4", "none"][trial % 4]
+        cols, null_counts = {}, []
+        ncols = int(rng.integers(1, 4))
+        for ci in range(ncols):
+            kind = int(rng.integers(0, 4))
+            nf = float(rng.choice([0.0, 0.02, 0.5]))
+            mask = rng.random(n) < nf
+            if kind == 0:
+                arr = pa.array(rng.integers(-2**31, 2**31, n, dtype=np.int64)
+                               .astype(np.int32), pa.int32(), mask=mask)
+            elif kind == 1:
+                arr = pa.array(rng.integers(0, int(rng.choice([50, n + 1])),
+                                            n).astype(np.int64),
+                               pa.int64(), mask=mask)
+            elif kind == 2:
+                arr = pa.array(rng.random(n), pa.float64(), mask=mask)
+            else:
+                arr = pa.array(rng.random(n).astype(np.float32),
+                               pa.float32(), mask=mask)
+            cols[f"c{ci}"] = arr
+            null_counts.append(int(mask.sum()))
+        with tempfile.TemporaryDirectory() as td:
+            path = os.path.join(td, "f.parquet")
+            pq.write_table(
+                pa.table(cols), path, compression=codec,
+                row_group_size=int(rng.integers(100, n + 1)),
+                data_page_size=int(rng.choice([1024, 64 * 1024, 1 << 20])),
+                use_dictionary=bool(rng.integers(0, 2)))
+            out = ctypes.create_string_buffer(1 << 16)
+            rc = lib.auron_debug_parquet_summary(path.encode(), out, len(out))
+            assert rc > 0, (trial, out.value)
+            s = out.value.decode()
+            per_col_rows = [0] * ncols
+            per_col_nulls = [0] * ncols
+            for m in re.finditer(r"c(\d+)\{n=(\d+),nulls=(\d+)", s):
+                per_col_rows[int(m.group(1))] += int(m.group(2))
+                per_col_nulls[int(m.group(1))] += int(m.group(3))
+            assert per_col_rows == [n] * ncols, (trial, s[:200])
+            assert per_col_nulls == null_counts, (trial, s[:200])
