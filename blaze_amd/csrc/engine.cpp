@@ -478,6 +478,8 @@ class AggOp {
     skip_ratio_ = conf.get_d("PARTIAL_AGG_SKIPPING_RATIO", 0.999);
     skip_min_rows_ = conf.get_i("PARTIAL_AGG_SKIPPING_MIN_ROWS", 20000);
     int64_t slots = conf.get_i("AURON_HIP_AGG_TABLE_SLOTS", 1 << 23);
+    conf_coll_cap_ =
+        std::max<int64_t>(1024, conf.get_i("AURON_HIP_COLLECT_POOL", 16 << 20));
     // VRAM budget for the slot table (a9 analog of MemManager's budget,
     // memmgr/mod.rs:36-105): exceeding it spills frozen records to host
     // buckets instead of growing (agg_table.rs:540-588 semantics).
