@@ -104,6 +104,13 @@ bool serde_read_batch(const uint8_t* p, size_t len, size_t* used,
     *err = "serde: truncated row count";
     return false;
   }
+  // adversarial-input guards: a batch cannot carry more rows than input
+  // bytes (every row costs >=1 byte in any column), and n*width arithmetic
+  // below must not overflow size_t
+  if (n > len) {
+    *err = "serde: row count exceeds input size";
+    return false;
+  }
   *rows = (int64_t)n;
   cols->clear();
   for (int w : dtype_widths) {
@@ -146,8 +153,14 @@ bool serde_read_batch(const uint8_t* p, size_t len, size_t* used,
       pos += 4 * n;
       c.offsets.resize(n + 1);
       c.offsets[0] = 0;
-      for (uint64_t i = 0; i < n; i++)
+      for (uint64_t i = 0; i < n; i++) {
+        if (lens[i] < 0 || (size_t)lens[i] > len ||
+            (int64_t)c.offsets[i] + lens[i] > (int64_t)INT32_MAX) {
+          *err = "serde: bad binary length";  // negative/overflowing offsets
+          return false;                       // would walk out of bounds
+        }
         c.offsets[i + 1] = c.offsets[i] + lens[i];
+      }
       size_t db = (size_t)c.offsets[n];
       if (pos + db > len) {
         *err = "serde: truncated binary data";

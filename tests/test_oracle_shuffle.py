@@ -191,3 +191,50 @@ def test_engine_serde_ipc_roundtrip_cpu():
         assert len(got) >= 8
         blen = int.from_bytes(got[:4], "little")
         assert 4 + blen <= len(got)  # first [u32-LE len][frame] block sane
+
+
+def test_engine_ipc_parse_rejects_garbage_cpu():
+    """The shuffle read side parses UNTRUSTED bytes (files on disk): random
+    garbage, truncations and bit flips must error cleanly, never crash.
+    Exercised through the roundtrip export's decode half by corrupting a
+    valid stream is not possible there, so feed the garbage as ipc_segments
+    would arrive: via oracle-writer streams mutated adversarially and
+    decoded by oracle.ipc_decode (same framing) plus the engine roundtrip
+    export run on valid data before/after to prove the library stays sane."""
+    import ctypes
+    import blaze_amd
+    rng = np.random.default_rng(91)
+    lib = blaze_amd.lib()
+    f = lib.auron_debug_serde_roundtrip
+    f.restype = ctypes.c_int64
+
+    def roundtrip_ok():
+        n = 64
+        keys = np.arange(n, dtype=np.int64)
+        vals = np.ones(n)
+        offs = np.zeros(n + 1, dtype=np.int32)
+        out = ctypes.create_string_buffer(1 << 16)
+        r = f(keys.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)), None,
+              vals.ctypes.data_as(ctypes.POINTER(ctypes.c_double)), None,
+              offs.ctypes.data_as(ctypes.POINTER(ctypes.c_int32)), b"",
+              ctypes.c_int64(n), ctypes.c_int64(16), out, len(out))
+        assert r > 0, out.value
+        return out.raw[:r]
+
+    stream = roundtrip_ok()
+    # garbage / truncated / flipped streams through the oracle decoder (the
+    # same [u32-LE len][lz4 frame] framing) -- errors, never crashes
+    cases = [bytes(rng.integers(0, 256, int(rng.integers(1, 100))).astype(
+        np.uint8)) for _ in range(20)]
+    cases += [stream[:i] for i in range(1, len(stream), 97)]
+    for _ in range(20):
+        b = bytearray(stream)
+        b[int(rng.integers(0, len(b)))] ^= int(rng.integers(1, 256))
+        cases.append(bytes(b))
+    for blob in cases:
+        try:
+            oracle.ipc_decode(blob)
+        except AssertionError:
+            pass
+    # library still healthy
+    roundtrip_ok()
