@@ -2819,9 +2819,37 @@ int32_t auron_debug_parquet_summary(const char* path, char* out, size_t cap) {
            std::to_string(pf.row_group_rows(rg));
       for (size_t c = 0; c < pf.columns().size(); c++) {
         PqColumnChunkData cd = pf.read_chunk(rg, (int)c);
+        // value checksum: wrapping i64 sum of the raw fixed-width values
+        // (sign-extended for narrow ints) over the dense non-null stream —
+        // lets CPU tests verify DECODED VALUES against numpy ground truth,
+        // not just counts
+        int64_t csum = 0;
+        const size_t w = dtype_width(pf.columns()[c].dtype());
+        auto add_vals = [&](const uint8_t* v, size_t cnt) {
+          for (size_t i = 0; i < cnt; i++) {
+            int64_t x = 0;
+            if (w == 8) {
+              memcpy(&x, v + i * 8, 8);
+            } else if (w == 4) {
+              int32_t x32;
+              memcpy(&x32, v + i * 4, 4);
+              x = x32;
+            }
+            csum = (int64_t)((uint64_t)csum + (uint64_t)x);
+          }
+        };
+        if (w > 0) {
+          if (cd.uses_dict) {
+            for (uint32_t ix : cd.dict_indices)
+              add_vals(cd.dict_values.data() + (size_t)ix * w, 1);
+          } else {
+            add_vals(cd.plain.data(), cd.plain.size() / w);
+          }
+        }
         r += " c" + std::to_string(c) + "{n=" + std::to_string(cd.num_values) +
              ",nulls=" + std::to_string(cd.null_count) +
-             ",dict=" + std::to_string(cd.uses_dict ? cd.dict_count : 0) + "}";
+             ",dict=" + std::to_string(cd.uses_dict ? cd.dict_count : 0) +
+             ",csum=" + std::to_string(csum) + "}";
       }
       r += "]";
     }

@@ -187,3 +187,53 @@ def test_parquet_decode_fuzz_cpu():
                 per_col_nulls[int(m.group(1))] += int(m.group(3))
             assert per_col_rows == [n] * ncols, (trial, s[:200])
             assert per_col_nulls == null_counts, (trial, s[:200])
+
+
+def test_parquet_delta_binary_packed_cpu():
+    """DELTA_BINARY_PACKED (encoding 5, parquet delta spec) decode for
+    INT32/INT64 incl. negatives, nulls, multi-block streams — verified
+    against pyarrow-written files via the host decode summary."""
+    pa = pytest.importorskip("pyarrow")
+    import pyarrow.parquet as pq
+    import numpy as np
+    import re
+    import ctypes
+    import tempfile, os
+
+    lib = blaze_amd.lib()
+    lib.auron_debug_parquet_summary.restype = ctypes.c_int32
+    lib.auron_debug_parquet_summary.argtypes = [ctypes.c_char_p,
+                                               ctypes.c_char_p,
+                                               ctypes.c_size_t]
+    rng = np.random.default_rng(101)
+    for n in (1, 100, 5000, 70_000):
+        a64 = rng.integers(-(1 << 40), 1 << 40, n).astype(np.int64)
+        a32 = rng.integers(-(1 << 30), 1 << 30, n).astype(np.int32)
+        mask = rng.random(n) < 0.1
+        with tempfile.TemporaryDirectory() as td:
+            path = os.path.join(td, "d.parquet")
+            pq.write_table(
+                pa.table({"a": pa.array(a64, pa.int64(), mask=mask),
+                          "b": pa.array(a32, pa.int32())}),
+                path, compression="snappy", use_dictionary=False,
+                column_encoding={"a": "DELTA_BINARY_PACKED",
+                                 "b": "DELTA_BINARY_PACKED"},
+                version="2.6", data_page_version="2.0")
+            encs = pq.ParquetFile(path).metadata.row_group(0).column(0).encodings
+            assert "DELTA_BINARY_PACKED" in encs
+            out = ctypes.create_string_buffer(1 << 16)
+            rc = lib.auron_debug_parquet_summary(path.encode(), out, len(out))
+            assert rc > 0, (n, out.value)
+            s = out.value.decode()
+            rows = sum(int(m) for m in re.findall(r"c0\{n=(\d+)", s))
+            nulls = sum(int(m) for m in re.findall(r"c0\{n=\d+,nulls=(\d+)", s))
+            assert rows == n and nulls == int(mask.sum()), (n, s[:200])
+            # VALUE verification: wrapping i64 checksums of the decoded
+            # non-null value streams vs numpy ground truth
+            with np.errstate(over="ignore"):
+                cs0 = np.int64(a64[~mask].sum()) if (~mask).any() else np.int64(0)
+                cs1 = np.int64(a32.astype(np.int64).sum()) if n else np.int64(0)
+            got0 = sum(int(m) for m in re.findall(r"c0\{[^}]*csum=(-?\d+)", s))
+            got1 = sum(int(m) for m in re.findall(r"c1\{[^}]*csum=(-?\d+)", s))
+            assert np.int64(got0) == cs0, (n, got0, int(cs0))
+            assert np.int64(got1) == cs1, (n, got1, int(cs1))
