@@ -2845,6 +2845,17 @@ int32_t auron_debug_parquet_summary(const char* path, char* out, size_t cap) {
           } else {
             add_vals(cd.plain.data(), cd.plain.size() / w);
           }
+        } else if (!cd.bin_offsets.empty()) {
+          // byte-array columns: position-weighted rolling checksum over the
+          // row-aligned lengths and data bytes
+          for (size_t i = 0; i + 1 < cd.bin_offsets.size(); i++) {
+            int64_t l = cd.bin_offsets[i + 1] - cd.bin_offsets[i];
+            csum = (int64_t)((uint64_t)csum + (uint64_t)(l * (int64_t)(i + 1)));
+          }
+          for (size_t j = 0; j < cd.bin_data.size(); j++)
+            csum = (int64_t)((uint64_t)csum +
+                             (uint64_t)((int64_t)cd.bin_data[j] *
+                                        (int64_t)(j + 1)));
         }
         r += " c" + std::to_string(c) + "{n=" + std::to_string(cd.num_values) +
              ",nulls=" + std::to_string(cd.null_count) +

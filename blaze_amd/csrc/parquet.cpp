@@ -321,9 +321,13 @@ void rle_bp_decode(const uint8_t* p, size_t len, int bit_width, int64_t count,
 // v[i+1] = v[i] + min_delta + packed[i]; arithmetic wraps in i64 (arrow-cpp
 // semantics; INT32 columns truncate per value). Data bytes exist only for
 // miniblocks that contain remaining values (arrow-cpp writer behavior).
-void delta_bp_decode(const uint8_t* p, size_t len, int64_t count, int vw,
-                     std::vector<uint8_t>* out) {
-  size_t pos = 0;
+// core reader: ONE delta stream starting at *pos; advances *pos exactly past
+// the stream (arrow-cpp writes full padded miniblocks only while values
+// remain, so the cursor lands on the next stream for the composed byte-array
+// encodings below)
+void delta_bp_stream(const uint8_t* p, size_t len, size_t* posp, int64_t count,
+                     std::vector<int64_t>* vals) {
+  size_t pos = *posp;
   auto uleb = [&]() -> uint64_t {
     uint64_t v = 0;
     int shift = 0;
@@ -348,14 +352,10 @@ void delta_bp_decode(const uint8_t* p, size_t len, int64_t count, int vw,
   uint64_t mini_size = block_size / mini_per_block;
   if (mini_size % 8 != 0) fail("delta: miniblock size not multiple of 8");
   if ((int64_t)total < count) fail("delta: fewer values than expected");
-  auto emit = [&](int64_t v) {
-    uint8_t b[8];
-    memcpy(b, &v, 8);
-    out->insert(out->end(), b, b + vw);
-  };
+  vals->reserve(vals->size() + (size_t)count);
   int64_t remaining = count;
   if (remaining > 0) {
-    emit(value);
+    vals->push_back(value);
     remaining--;
   }
   std::vector<uint8_t> widths(mini_per_block);
@@ -375,36 +375,43 @@ void delta_bp_decode(const uint8_t* p, size_t len, int64_t count, int vw,
       const uint64_t mask = bw == 64 ? ~0ull : ((1ull << bw) - 1);
       for (uint64_t i = 0; i < mini_size && remaining > 0; i++) {
         uint64_t d = 0;
-        if (bw > 0) {
+        if (bw > 56) {  // bit-addressed path: the 64-bit acc can't stage it
+          uint64_t bit = (uint64_t)i * bw;
+          for (int bbit = 0; bbit < bw; bbit++) {
+            uint64_t idx = bit + bbit;
+            d |= (uint64_t)((p[pos + idx / 8] >> (idx % 8)) & 1) << bbit;
+          }
+        } else if (bw > 0) {
           while (bits < bw) {
             acc |= (uint64_t)p[bp++] << bits;
             bits += 8;
-            if (bits > 64) {  // bw>56: assemble via shift-out first
-              break;
-            }
           }
-          if (bw > 56) {
-            // slow path: bit-address arithmetic, avoids the 64-bit acc limit
-            uint64_t bit = (uint64_t)i * bw;
-            d = 0;
-            for (int bbit = 0; bbit < bw; bbit++) {
-              uint64_t idx = bit + bbit;
-              d |= (uint64_t)((p[pos + idx / 8] >> (idx % 8)) & 1) << bbit;
-            }
-          } else {
-            d = acc & mask;
-            acc >>= bw;
-            bits -= bw;
-          }
+          d = acc & mask;
+          acc >>= bw;
+          bits -= bw;
         }
         value += min_delta + (int64_t)d;
-        emit(value);
+        vals->push_back(value);
         remaining--;
       }
       pos += bytes;
       acc = 0;
       bits = 0;
     }
+  }
+  *posp = pos;
+}
+
+void delta_bp_decode(const uint8_t* p, size_t len, int64_t count, int vw,
+                     std::vector<uint8_t>* out) {
+  std::vector<int64_t> vals;
+  size_t pos = 0;
+  delta_bp_stream(p, len, &pos, count, &vals);
+  out->reserve(out->size() + vals.size() * (size_t)vw);
+  for (int64_t v : vals) {
+    uint8_t b[8];
+    memcpy(b, &v, 8);
+    out->insert(out->end(), b, b + vw);
   }
 }
 
@@ -848,6 +855,42 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
         if (is_bytes) fail("parquet: delta encoding on byte arrays");
         flatten_dict();
         delta_bp_decode(data, (size_t)dlen, non_null, vw, &out.plain);
+        break;
+      }
+      case 6: {  // DELTA_LENGTH_BYTE_ARRAY: delta lens ++ concatenated bytes
+        if (!is_bytes) fail("parquet: DELTA_LENGTH on non-byte column");
+        std::vector<int64_t> lens;
+        size_t pos2 = 0;
+        delta_bp_stream(data, (size_t)dlen, &pos2, non_null, &lens);
+        for (int64_t l : lens) {
+          if (l < 0 || pos2 + (size_t)l > (size_t)dlen)
+            fail("parquet: DELTA_LENGTH overrun");
+          nn_lens.push_back((int32_t)l);
+          nn_data.insert(nn_data.end(), data + pos2, data + pos2 + l);
+          pos2 += (size_t)l;
+        }
+        break;
+      }
+      case 7: {  // DELTA_BYTE_ARRAY: prefix-len stream ++ suffix-len stream
+                 // ++ suffix bytes; value[i] = value[i-1][:prefix] + suffix
+        if (!is_bytes) fail("parquet: DELTA_BYTE_ARRAY on non-byte column");
+        std::vector<int64_t> plens, slens;
+        size_t pos2 = 0;
+        delta_bp_stream(data, (size_t)dlen, &pos2, non_null, &plens);
+        delta_bp_stream(data, (size_t)dlen, &pos2, non_null, &slens);
+        std::vector<uint8_t> prev;
+        for (int64_t i = 0; i < non_null; i++) {
+          int64_t pl = plens[(size_t)i], sl = slens[(size_t)i];
+          if (pl < 0 || sl < 0 || (size_t)pl > prev.size() ||
+              pos2 + (size_t)sl > (size_t)dlen)
+            fail("parquet: DELTA_BYTE_ARRAY overrun");
+          std::vector<uint8_t> cur(prev.begin(), prev.begin() + pl);
+          cur.insert(cur.end(), data + pos2, data + pos2 + sl);
+          pos2 += (size_t)sl;
+          nn_lens.push_back((int32_t)cur.size());
+          nn_data.insert(nn_data.end(), cur.begin(), cur.end());
+          prev = std::move(cur);
+        }
         break;
       }
       case 2:    // PLAIN_DICTIONARY
