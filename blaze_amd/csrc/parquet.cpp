@@ -857,6 +857,19 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
         delta_bp_decode(data, (size_t)dlen, non_null, vw, &out.plain);
         break;
       }
+      case 9: {  // BYTE_STREAM_SPLIT: w byte planes, de-interleave
+        if (is_bytes) fail("parquet: BYTE_STREAM_SPLIT on byte arrays");
+        if (dlen < non_null * vw) fail("parquet: short BYTE_STREAM_SPLIT");
+        flatten_dict();
+        size_t base = out.plain.size();
+        out.plain.resize(base + (size_t)non_null * vw);
+        for (int b = 0; b < vw; b++) {
+          const uint8_t* plane = data + (size_t)b * non_null;
+          uint8_t* dst = out.plain.data() + base + b;
+          for (int64_t i = 0; i < non_null; i++) dst[(size_t)i * vw] = plane[i];
+        }
+        break;
+      }
       case 6: {  // DELTA_LENGTH_BYTE_ARRAY: delta lens ++ concatenated bytes
         if (!is_bytes) fail("parquet: DELTA_LENGTH on non-byte column");
         std::vector<int64_t> lens;

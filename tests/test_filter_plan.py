@@ -298,3 +298,43 @@ def test_parquet_delta_byte_array_cpu():
                 exp_total += cs if cs < (1 << 63) else cs - (1 << 64)
             got = sum(int(m) for m in re.findall(r"c0\{[^}]*csum=(-?\d+)", s))
             assert got == exp_total, (got, exp_total)
+
+
+def test_parquet_byte_stream_split_cpu():
+    """BYTE_STREAM_SPLIT (encoding 9; float planes) — bit-exact via the
+    value checksum of the raw f64/f32 bit patterns."""
+    pa = pytest.importorskip("pyarrow")
+    import pyarrow.parquet as pq
+    import numpy as np
+    import re
+    import ctypes
+    import tempfile, os
+
+    lib = blaze_amd.lib()
+    lib.auron_debug_parquet_summary.restype = ctypes.c_int32
+    lib.auron_debug_parquet_summary.argtypes = [ctypes.c_char_p,
+                                               ctypes.c_char_p,
+                                               ctypes.c_size_t]
+    rng = np.random.default_rng(127)
+    n = 30_000
+    vals = rng.random(n) * 1e6 - 5e5
+    mask = rng.random(n) < 0.05
+    with tempfile.TemporaryDirectory() as td:
+        path = os.path.join(td, "b.parquet")
+        pq.write_table(
+            pa.table({"f": pa.array(vals, pa.float64(), mask=mask)}), path,
+            compression="zstd", use_dictionary=False,
+            use_byte_stream_split=["f"], version="2.6",
+            row_group_size=9_999)
+        encs = pq.ParquetFile(path).metadata.row_group(0).column(0).encodings
+        assert "BYTE_STREAM_SPLIT" in encs
+        out = ctypes.create_string_buffer(1 << 16)
+        rc = lib.auron_debug_parquet_summary(path.encode(), out, len(out))
+        assert rc > 0, out.value
+        s = out.value.decode()
+        bits = vals.view(np.int64)[~mask]
+        with np.errstate(over="ignore"):
+            exp = int(np.int64(bits.sum()))
+        got = sum(int(m) for m in re.findall(r"c0\{[^}]*csum=(-?\d+)", s))
+        # per-rg wrapping sums add without wrap at this scale
+        assert np.int64(got) == np.int64(exp), (got, exp)
