@@ -171,11 +171,27 @@ def test_parquet_decode_fuzz_cpu():
             null_counts.append(int(mask.sum()))
         with tempfile.TemporaryDirectory() as td:
             path = os.path.join(td, "f.parquet")
+            kw = {}
+            mode = int(rng.integers(0, 3))
+            if mode == 1:  # delta int columns (v2)
+                kw = dict(use_dictionary=False, version="2.6",
+                          data_page_version="2.0",
+                          column_encoding={k: "DELTA_BINARY_PACKED"
+                                           for k, a in cols.items()
+                                           if pa.types.is_integer(a.type)})
+            elif mode == 2:  # byte-stream-split float columns (v2)
+                bss = [k for k, a in cols.items()
+                       if pa.types.is_floating(a.type)]
+                if bss:
+                    kw = dict(use_dictionary=False, version="2.6",
+                              use_byte_stream_split=bss)
+            else:
+                kw = dict(use_dictionary=bool(rng.integers(0, 2)))
             pq.write_table(
                 pa.table(cols), path, compression=codec,
                 row_group_size=int(rng.integers(100, n + 1)),
                 data_page_size=int(rng.choice([1024, 64 * 1024, 1 << 20])),
-                use_dictionary=bool(rng.integers(0, 2)))
+                **kw)
             out = ctypes.create_string_buffer(1 << 16)
             rc = lib.auron_debug_parquet_summary(path.encode(), out, len(out))
             assert rc > 0, (trial, out.value)
