@@ -1644,6 +1644,9 @@ class ShuffleOp {
                node.partitioning.kind != Repartition::RoundRobin) {
       FAIL("unsupported partitioning kind");
     }
+    if (node.partitioning.partition_count <= 0 ||
+        node.partitioning.partition_count > (1 << 24))
+      FAIL("ShuffleWriter: partition_count out of range");
     P_ = (uint32_t)node.partitioning.partition_count;
     batch_size_ = conf.get_i("BATCH_SIZE", 10000);
     if (batch_size_ <= 0) FAIL("invalid BATCH_SIZE conf");
@@ -2319,24 +2322,25 @@ struct Runtime {
                            &sl, &err2) ||
         sl.is_null)
       return true;
-    double lo, hi, v;
-    int pt = pf.columns()[col->col_index].physical_type;
-    if (pt == 1 || pt == 2) {
-      lo = (double)s.min_i;
-      hi = (double)s.max_i;
-      v = (double)sl.i64;
-    } else if (pt == 4 || pt == 5) {
-      lo = s.min_f;
-      hi = s.max_f;
-      v = sl.f64;
-    } else {
-      return true;
-    }
     std::string op = e.op;
     if (flipped) {
       op = op == "Lt" ? "Gt" : op == "LtEq" ? "GtEq"
            : op == "Gt" ? "Lt" : op == "GtEq" ? "LtEq" : op;
     }
+    int pt = pf.columns()[col->col_index].physical_type;
+    if (pt == 1 || pt == 2) {
+      // compare in the int64 domain: a round-trip through double loses
+      // precision past 2^53 and could prune a row group that matches
+      int64_t lo = s.min_i, hi = s.max_i, v = sl.i64;
+      if (op == "Lt") return lo < v;
+      if (op == "LtEq") return lo <= v;
+      if (op == "Gt") return hi > v;
+      if (op == "GtEq") return hi >= v;
+      if (op == "Eq") return lo <= v && v <= hi;
+      return true;
+    }
+    if (pt != 4 && pt != 5) return true;
+    double lo = s.min_f, hi = s.max_f, v = sl.f64;
     if (op == "Lt") return lo < v;
     if (op == "LtEq") return lo <= v;
     if (op == "Gt") return hi > v;

@@ -447,6 +447,10 @@ bool snappy_uncompress(const uint8_t* src, size_t n, std::vector<uint8_t>* out,
       }
       if (len > 60) {
         int nb = (int)len - 60;
+        if (pos + (size_t)nb > n) {
+          *err = "snappy: truncated literal length";
+          return false;
+        }
         len = 0;
         for (int i = 0; i < nb; i++) len |= (uint64_t)src[pos++] << (8 * i);
         len += 1;
@@ -460,6 +464,11 @@ bool snappy_uncompress(const uint8_t* src, size_t n, std::vector<uint8_t>* out,
       pos += len;
     } else {
       uint64_t len, off;
+      const size_t offw = (type == 1) ? 1 : (type == 2) ? 2 : 4;
+      if (pos + offw > n) {
+        *err = "snappy: truncated copy offset";
+        return false;
+      }
       if (type == 1) {  // copy with 1-byte offset
         len = ((tag >> 2) & 7) + 4;
         off = ((uint64_t)(tag >> 5) << 8) | src[pos++];
@@ -711,6 +720,11 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
     TReader hr{file_.data() + pos, file_.data() + chunk_end};
     PageHeader ph = read_page_header(hr);
     const uint8_t* page = hr.p;
+    // every size below comes from the (untrusted) page header: validate
+    // against the chunk bounds before any read
+    if (ph.compressed_size < 0 || ph.uncompressed_size < 0 ||
+        page - file_.data() + ph.compressed_size > chunk_end)
+      fail("parquet: page overruns chunk");
     pos = (page - file_.data()) + ph.compressed_size;
 
     if (ph.type == 2) {  // dictionary page
@@ -731,10 +745,15 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
           uint32_t l;
           memcpy(&l, out.dict_values.data() + pos2, 4);
           pos2 += 4;
+          if (l > out.dict_values.size() - pos2)
+            fail("parquet: byte-array dict entry overruns dict");
           dict_lens.push_back((int32_t)pos2);  // start of bytes
           dict_lens.push_back((int32_t)l);     // length
           pos2 += l;
         }
+      } else if (out.dict_count > 0 &&
+                 out.dict_values.size() < (size_t)out.dict_count * vw) {
+        fail("parquet: dict page shorter than dict_count * width");
       }
       continue;
     }
@@ -751,13 +770,18 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
       if (ci.nullable) {
         if (ph.def_encoding != 3) fail("parquet: def levels must be RLE");
         uint32_t ll;
+        if (dlen < 4) fail("parquet: truncated def-level length");
         memcpy(&ll, data, 4);
+        if ((int64_t)ll > dlen - 4) fail("parquet: def levels overrun page");
         rle_bp_decode(data + 4, ll, 1, ph.num_values, &def_levels);
         data += 4 + ll;
         dlen -= 4 + ll;
       }
     } else {  // V2: def levels uncompressed before the (compressed) values
       const uint8_t* dp = page;
+      if (ph.def_len < 0 || ph.rep_len < 0 ||
+          ph.def_len + ph.rep_len > ph.compressed_size)
+        fail("parquet: V2 level lengths overrun page");
       if (ci.nullable) {
         if (ph.def_len > 0)
           rle_bp_decode(dp, ph.def_len, 1, ph.num_values, &def_levels);
@@ -765,6 +789,7 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
       dp += ph.def_len + ph.rep_len;
       int64_t comp = ph.compressed_size - ph.def_len - ph.rep_len;
       int64_t uncomp = ph.uncompressed_size - ph.def_len - ph.rep_len;
+      if (uncomp < 0) fail("parquet: V2 level lengths exceed uncompressed size");
       data = ph.v2_compressed ? decompress(dp, comp, uncomp, &buf) : dp;
       dlen = uncomp;
     }
@@ -909,7 +934,9 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
       case 2:    // PLAIN_DICTIONARY
       case 8: {  // RLE_DICTIONARY
         if (out.dict_values.empty()) fail("parquet: dict page missing");
+        if (dlen < 1) fail("parquet: empty dict-index page");
         int bw = data[0];
+        if (bw > 32) fail("parquet: dict index bit width > 32");
         std::vector<uint32_t> idx;
         rle_bp_decode(data + 1, dlen - 1, bw, non_null, &idx);
         if (is_bytes) {

@@ -189,6 +189,10 @@ bool ipc_decode_blocks(const uint8_t* p, size_t len,
     if (!lz4_decompress_frame(p + pos, block_len, payload, err)) return false;
     pos += block_len;
   }
+  if (pos != len) {  // 1-3 trailing bytes = a truncated length prefix
+    *err = "ipc: truncated block length prefix";
+    return false;
+  }
   return true;
 }
 
