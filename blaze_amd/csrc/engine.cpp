@@ -569,7 +569,7 @@ class AggOp {
       // holds {key,cnt,sum,first} only (perf note in DESIGN.md)
       if (!merge_mode_ && !has_mm_ && !has_first_ && !has_coll_ &&
           b.num_rows - done >= AGG2_MIN_CHUNK) {
-        int64_t chunk2 = std::min(b.num_rows - done, AGG2_MAX_CHUNK);
+        int64_t chunk2 = std::min(b.num_rows - done, agg2_chunk_max_);
         if (done + chunk2 < b.num_rows) chunk2 &= ~(int64_t)7;
         two_phase_chunk(b, done, chunk2);
         done += chunk2;
@@ -1005,7 +1005,10 @@ class AggOp {
   }
 
   static constexpr int64_t AGG2_MIN_CHUNK = 4 << 20;   // below: single-phase
-  static constexpr int64_t AGG2_MAX_CHUNK = 64 << 20;  // partition buffer size
+  // partition-buffer rows per two-phase chunk: bigger chunks amortize launch
+  // tails and per-chunk syncs (24 B/row + 24 B/row leftover scratch; 256M
+  // rows = 12 GB of the 288 GB HBM). Runtime-tunable for A/B sweeps.
+  int64_t agg2_chunk_max_ = 256 << 20;
   static constexpr int AGG2_NBUCK_LOG2 = 10;           // 1024 buckets: scatter write-line footprint ~L2-sized (512 buckets overflow the 2048-slot LDS window: 2.3% leftovers, 3x slower)
 
   // Two-phase aggregation of rows [done, done+chunk) of batch b:
@@ -1036,13 +1039,26 @@ class AggOp {
       if (g && g[0]) agg2_grid_log2_ = atoi(g);
       if (agg2_grid_log2_ < 6 || agg2_grid_log2_ > AGG2_GRID_LOG2_MAX)
         agg2_grid_log2_ = 9;
-      if (agg2_split_) {
-        d_partkv_.alloc(AGG2_MAX_CHUNK * sizeof(PartKV));
-        d_rowv_.alloc(AGG2_MAX_CHUNK * 4);
-      } else {
-        d_partkv_.alloc(AGG2_MAX_CHUNK * sizeof(PartRow));
+      // hist/scatter/bucket workgroup size: 1024 threads = 16 waves/block
+      // fill the chip's 32-wave/CU slots (the 256-thread launch left the
+      // bucket kernel at 12 waves/CU = 38% occupancy; microarch guide
+      // "Chip-level parameters"/"Register files")
+      const char* bl = getenv("AURON_AGG2_BLOCK");
+      if (bl && bl[0]) agg2_block_ = atoi(bl);
+      if (agg2_block_ != 256 && agg2_block_ != 512 && agg2_block_ != 1024)
+        agg2_block_ = 1024;
+      const char* cm = getenv("AURON_AGG2_CHUNK_M");  // millions of rows
+      if (cm && cm[0]) {
+        int64_t m = atoll(cm);
+        if (m >= 4 && m <= 1024) agg2_chunk_max_ = m << 20;
       }
-      d_leftover_.alloc(AGG2_MAX_CHUNK * sizeof(PartRow));
+      if (agg2_split_) {
+        d_partkv_.alloc(agg2_chunk_max_ * sizeof(PartKV));
+        d_rowv_.alloc(agg2_chunk_max_ * 4);
+      } else {
+        d_partkv_.alloc(agg2_chunk_max_ * sizeof(PartRow));
+      }
+      d_leftover_.alloc(agg2_chunk_max_ * sizeof(PartRow));
       int64_t mat = (int64_t)nbuck << AGG2_GRID_LOG2_MAX;
       d_counts_.alloc((mat + 1) * 4);   // +1: scan total slot
       d_scanned_.alloc((mat + 1) * 4);
@@ -1067,7 +1083,8 @@ class AggOp {
     AURON_HIP(hipMemsetAsync(d_counters_.get(), 0, 24, stream_));
     launch_agg2_hist(keys, kv, chunk, AGG2_NBUCK_LOG2, agg2_grid_log2_,
                      d_counts_.get<uint32_t>(),
-                     (uint32_t*)(d_counters_.get<uint8_t>() + 16), stream_);
+                     (uint32_t*)(d_counters_.get<uint8_t>() + 16), agg2_block_,
+                     stream_);
     size_t tb = d_scan_tmp_.size();
     scan_counts_matrix(d_counts_.get<uint32_t>(), d_scanned_.get<uint32_t>(),
                        mat + 1, d_scan_tmp_.get(), &tb, stream_);
@@ -1078,11 +1095,11 @@ class AggOp {
       launch_agg2_scatter(keys, kv, vals, vv, chunk, AGG2_NBUCK_LOG2,
                           agg2_grid_log2_, d_scanned_.get<uint32_t>(),
                           d_partkv_.get<PartKV>(), d_rowv_.get<uint32_t>(),
-                          stream_);
+                          agg2_block_, stream_);
     else
       launch_agg2_scatter24(keys, kv, vals, vv, chunk, AGG2_NBUCK_LOG2,
                             agg2_grid_log2_, d_scanned_.get<uint32_t>(),
-                            d_partkv_.get<PartRow>(), stream_);
+                            d_partkv_.get<PartRow>(), agg2_block_, stream_);
     // A: per-bucket LDS aggregation (counters zeroed above; staged_n at +0,
     // lo_n at +1, special count at byte offset 8 via the +2 uint32 slot)
     if (agg2_split_)
@@ -1093,7 +1110,7 @@ class AggOp {
                          (int64_t)nbuck * AGG2_LSLOTS,
                          d_leftover_.get<PartRow>(),
                          d_counters_.get<unsigned long long>() + 1,
-                         t_.error_flag, stream_);
+                         t_.error_flag, agg2_block_, stream_);
     else
       launch_agg2_bucket24(d_partkv_.get<PartRow>(),
                            d_offsets_.get<uint32_t>(), val_is_int_ ? 1 : 0,
@@ -1102,7 +1119,7 @@ class AggOp {
                            (int64_t)nbuck * AGG2_LSLOTS,
                            d_leftover_.get<PartRow>(),
                            d_counters_.get<unsigned long long>() + 1,
-                           t_.error_flag, stream_);
+                           t_.error_flag, agg2_block_, stream_);
     unsigned long long* h_ctr =
         (unsigned long long*)(pinned_agg2_.get<uint8_t>() +
                               2 * (size_t)(nbuck + 1) * 4);
@@ -1621,6 +1638,7 @@ class AggOp {
   // two-phase scratch (allocated on first large chunk)
   bool agg2_split_ = false;
   int agg2_grid_log2_ = 9;
+  int agg2_block_ = 1024;
   DevBuf d_partkv_, d_rowv_, d_leftover_, d_counts_, d_scanned_, d_scan_tmp_, d_offsets_,
       d_staged_, d_counters_;
   PinnedBuf pinned_agg2_;

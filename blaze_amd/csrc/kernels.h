@@ -302,7 +302,7 @@ constexpr int AGG2_GRID_LOG2_MAX = 10;  // matrix sizing bound for the
 // [nbuck << AGG2_GRID_LOG2]
 void launch_agg2_hist(const int64_t* keys, const uint8_t* key_valid, int64_t n,
                       int nbuck_log2, int grid_log2, uint32_t* counts_matrix,
-                      uint32_t* special_rows, hipStream_t s);
+                      uint32_t* special_rows, int block, hipStream_t s);
 // exclusive scan over the flat counts matrix -> per-(block,bucket) bases
 void scan_counts_matrix(const uint32_t* counts, uint32_t* scanned, int64_t n,
                         void* temp, size_t* temp_bytes, hipStream_t s);
@@ -312,17 +312,18 @@ void launch_agg2_scatter(const int64_t* keys, const uint8_t* key_valid,
                          const double* vals, const uint8_t* val_valid,
                          int64_t n, int nbuck_log2, int grid_log2,
                          const uint32_t* scanned,
-                         PartKV* out_kv, uint32_t* out_rowv, hipStream_t s);
+                         PartKV* out_kv, uint32_t* out_rowv, int block,
+                         hipStream_t s);
 void launch_agg2_scatter24(const int64_t* keys, const uint8_t* key_valid,
                            const double* vals, const uint8_t* val_valid,
                            int64_t n, int nbuck_log2, int grid_log2,
                            const uint32_t* scanned,
-                           PartRow* out, hipStream_t s);
+                           PartRow* out, int block, hipStream_t s);
 void launch_agg2_bucket24(const PartRow* part, const uint32_t* offsets,
                           int is_int, int nbuckets, StagedGroup* staged,
                           unsigned long long* staged_n, int64_t staged_cap,
                           PartRow* leftover, unsigned long long* lo_n,
-                          uint32_t* error_flag, hipStream_t s);
+                          uint32_t* error_flag, int block, hipStream_t s);
 void launch_agg2_specials(const AggTable& t, const int64_t* keys,
                           const uint8_t* key_valid, const double* vals,
                           const uint8_t* val_valid, int64_t n,
@@ -332,7 +333,7 @@ void launch_agg2_bucket(const PartKV* part_kv, const uint32_t* part_rowv,
                         int nbuckets, StagedGroup* staged,
                         unsigned long long* staged_n, int64_t staged_cap,
                         PartRow* leftover, unsigned long long* lo_n,
-                        uint32_t* error_flag, hipStream_t s);
+                        uint32_t* error_flag, int block, hipStream_t s);
 void launch_agg2_merge_groups(const AggTable& t, const StagedGroup* staged,
                               int64_t n, uint64_t row_offset, hipStream_t s);
 void launch_agg2_leftovers(const AggTable& t, const PartRow* rows, int64_t n,

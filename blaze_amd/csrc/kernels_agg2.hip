@@ -71,8 +71,12 @@ __device__ __forceinline__ uint32_t bucket_of(int64_t key, int nbuck_log2) {
 // counts_matrix is bucket-major: counts[(b << GRID_LOG2) | blockIdx]. An
 // exclusive scan over the flat matrix then gives every block a private,
 // contiguous output range per bucket — the scatter needs NO global atomics.
+// Launched at 256 or 1024 threads (AURON_AGG2_BLOCK): 1024-thread blocks
+// reach 16 waves per block, so a 512-block grid fills all 8192 wave slots
+// of the chip where 256-thread blocks reached only 8 waves/CU.
 
-__global__ void k_agg2_hist(const int64_t* __restrict__ keys,
+__global__ __launch_bounds__(1024) void k_agg2_hist(
+    const int64_t* __restrict__ keys,
                             const uint8_t* __restrict__ key_valid, int64_t n,
                             int nbuck_log2, int grid_log2,
                             uint32_t* __restrict__ counts_matrix,
@@ -113,7 +117,8 @@ __global__ void k_agg2_offsets(const uint32_t* __restrict__ scanned,
 // ---- phase P2: scatter into per-(block,bucket) reserved ranges -------------
 // Must use the SAME grid geometry and row traversal as k_agg2_hist.
 // rowv packs the chunk-local row (bit 0..30) + value-validity (bit 31).
-__global__ void k_agg2_scatter(const int64_t* __restrict__ keys,
+__global__ __launch_bounds__(1024) void k_agg2_scatter(
+    const int64_t* __restrict__ keys,
                                const uint8_t* __restrict__ key_valid,
                                const double* __restrict__ vals,
                                const uint8_t* __restrict__ val_valid, int64_t n,
@@ -172,7 +177,7 @@ __global__ void k_agg2_specials(const AggTable t,
 static constexpr int LSLOTS = AGG2_LSLOTS;  // 48 KB LDS -> 3 blocks/CU
 static constexpr int LPROBE = 64;
 
-__global__ void __launch_bounds__(256) k_agg2_bucket(
+__global__ void __launch_bounds__(1024) k_agg2_bucket(
     const PartKV* __restrict__ part, const uint32_t* __restrict__ part_rowv,
     const uint32_t* __restrict__ offsets, int is_int,
     int nbuckets, StagedGroup* __restrict__ staged,
@@ -355,9 +360,9 @@ void launch_agg2_leftovers(const AggTable& t, const PartRow* rows, int64_t n,
 
 void launch_agg2_hist(const int64_t* keys, const uint8_t* key_valid, int64_t n,
                       int nbuck_log2, int grid_log2, uint32_t* counts_matrix,
-                      uint32_t* special_rows, hipStream_t s) {
+                      uint32_t* special_rows, int block, hipStream_t s) {
   size_t lds = (size_t)(1u << nbuck_log2) * 4;
-  hipLaunchKernelGGL(k_agg2_hist, dim3(1 << grid_log2), dim3(BLOCK), lds, s,
+  hipLaunchKernelGGL(k_agg2_hist, dim3(1 << grid_log2), dim3(block), lds, s,
                      keys, key_valid, n, nbuck_log2, grid_log2, counts_matrix,
                      special_rows);
   check_launch2("k_agg2_hist");
@@ -382,9 +387,10 @@ void launch_agg2_scatter(const int64_t* keys, const uint8_t* key_valid,
                          const double* vals, const uint8_t* val_valid,
                          int64_t n, int nbuck_log2, int grid_log2,
                          const uint32_t* scanned,
-                         PartKV* out_kv, uint32_t* out_rowv, hipStream_t s) {
+                         PartKV* out_kv, uint32_t* out_rowv, int block,
+                         hipStream_t s) {
   size_t lds = (size_t)(1u << nbuck_log2) * 4;
-  hipLaunchKernelGGL(k_agg2_scatter, dim3(1 << grid_log2), dim3(BLOCK), lds, s,
+  hipLaunchKernelGGL(k_agg2_scatter, dim3(1 << grid_log2), dim3(block), lds, s,
                      keys, key_valid, vals, val_valid, n, nbuck_log2,
                      grid_log2, scanned, out_kv, out_rowv);
   check_launch2("k_agg2_scatter");
@@ -404,9 +410,9 @@ void launch_agg2_bucket(const PartKV* part_kv, const uint32_t* part_rowv,
                         int nbuckets, StagedGroup* staged,
                         unsigned long long* staged_n, int64_t staged_cap,
                         PartRow* leftover, unsigned long long* lo_n,
-                        uint32_t* error_flag, hipStream_t s) {
+                        uint32_t* error_flag, int block, hipStream_t s) {
   int blocks = nbuckets < (int)MAX_BLOCKS ? nbuckets : (int)MAX_BLOCKS;
-  hipLaunchKernelGGL(k_agg2_bucket, dim3(blocks), dim3(BLOCK), 0, s, part_kv,
+  hipLaunchKernelGGL(k_agg2_bucket, dim3(blocks), dim3(block), 0, s, part_kv,
                      part_rowv, offsets, is_int, nbuckets, staged, staged_n,
                      staged_cap, leftover, lo_n, error_flag);
   check_launch2("k_agg2_bucket");
@@ -414,7 +420,8 @@ void launch_agg2_bucket(const PartKV* part_kv, const uint32_t* part_rowv,
 
 
 // ---- 24B AoS variants (A/B comparison path, selected by the engine) --------
-__global__ void k_agg2_scatter24(const int64_t* __restrict__ keys,
+__global__ __launch_bounds__(1024) void k_agg2_scatter24(
+    const int64_t* __restrict__ keys,
                                  const uint8_t* __restrict__ key_valid,
                                  const double* __restrict__ vals,
                                  const uint8_t* __restrict__ val_valid,
@@ -447,15 +454,15 @@ void launch_agg2_scatter24(const int64_t* keys, const uint8_t* key_valid,
                            const double* vals, const uint8_t* val_valid,
                            int64_t n, int nbuck_log2, int grid_log2,
                            const uint32_t* scanned,
-                           PartRow* out, hipStream_t s) {
+                           PartRow* out, int block, hipStream_t s) {
   size_t lds = (size_t)(1u << nbuck_log2) * 4;
-  hipLaunchKernelGGL(k_agg2_scatter24, dim3(1 << grid_log2), dim3(BLOCK), lds,
+  hipLaunchKernelGGL(k_agg2_scatter24, dim3(1 << grid_log2), dim3(block), lds,
                      s, keys, key_valid, vals, val_valid, n, nbuck_log2,
                      grid_log2, scanned, out);
   check_launch2("k_agg2_scatter24");
 }
 
-__global__ void __launch_bounds__(256) k_agg2_bucket24(
+__global__ void __launch_bounds__(1024) k_agg2_bucket24(
     const PartRow* __restrict__ part, const uint32_t* __restrict__ offsets,
     int is_int, int nbuckets, StagedGroup* __restrict__ staged,
     unsigned long long* __restrict__ staged_n, int64_t staged_cap,
@@ -531,9 +538,9 @@ void launch_agg2_bucket24(const PartRow* part, const uint32_t* offsets,
                           int is_int, int nbuckets, StagedGroup* staged,
                           unsigned long long* staged_n, int64_t staged_cap,
                           PartRow* leftover, unsigned long long* lo_n,
-                          uint32_t* error_flag, hipStream_t s) {
+                          uint32_t* error_flag, int block, hipStream_t s) {
   int blocks = nbuckets < (int)MAX_BLOCKS ? nbuckets : (int)MAX_BLOCKS;
-  hipLaunchKernelGGL(k_agg2_bucket24, dim3(blocks), dim3(BLOCK), 0, s, part,
+  hipLaunchKernelGGL(k_agg2_bucket24, dim3(blocks), dim3(block), 0, s, part,
                      offsets, is_int, nbuckets, staged, staged_n, staged_cap,
                      leftover, lo_n, error_flag);
   check_launch2("k_agg2_bucket24");
