@@ -56,60 +56,6 @@ def make_device_input(torch, rows, rank, device):
     return keys, vals, bitmap, int(nulls.sum().item())
 
 
-def collect_partial(outputs):
-    """Flatten partial-agg output batches -> (keys, acc_lens, acc_data)."""
-    keys, lens, datas = [], [], []
-    for ob in outputs:
-        keys.append(ob[0]["values"])
-        off = ob[1]["offsets"]
-        lens.append((off[1:] - off[:-1]).astype(np.int32))
-        datas.append(ob[1]["data"])
-    if not keys:
-        return (np.empty(0, np.int64), np.empty(0, np.int32),
-                np.empty(0, np.uint8))
-    return np.concatenate(keys), np.concatenate(lens), np.concatenate(datas)
-
-
-def exchange(torch, dist, device, keys, lens, data, dest, world):
-    """RCCL all-to-all of (keys, acc lens, acc bytes) by destination rank."""
-    order = np.argsort(dest, kind="stable")
-    keys_s, lens_s = keys[order], lens[order]
-    row_splits = np.bincount(dest, minlength=world).astype(np.int64)
-    # bytes per row, permuted: vectorized ragged gather (a python per-row
-    # loop here costs seconds at 1M groups per step)
-    offs = np.concatenate([[0], np.cumsum(lens)]).astype(np.int64)
-    total = int(lens_s.sum())
-    if total:
-        out_base = np.concatenate([[0], np.cumsum(lens_s)])[:-1].astype(np.int64)
-        pos = np.repeat(offs[order] - out_base, lens_s) + \
-            np.arange(total, dtype=np.int64)
-        data_s = data[pos]
-    else:
-        data_s = np.empty(0, np.uint8)
-    byte_splits = np.zeros(world, dtype=np.int64)
-    np.add.at(byte_splits, dest, lens.astype(np.int64))
-
-    t_rows = torch.tensor(row_splits, device=device)
-    t_bytes = torch.tensor(byte_splits, device=device)
-    r_rows = torch.empty_like(t_rows)
-    r_bytes = torch.empty_like(t_bytes)
-    dist.all_to_all_single(r_rows, t_rows)
-    dist.all_to_all_single(r_bytes, t_bytes)
-    in_rows, in_bytes = r_rows.cpu().numpy(), r_bytes.cpu().numpy()
-
-    def a2a(src_np, splits_out, splits_in, dtype):
-        src = torch.from_numpy(np.ascontiguousarray(src_np)).to(device)
-        dst = torch.empty(int(splits_in.sum()), dtype=src.dtype, device=device)
-        dist.all_to_all_single(dst, src, splits_in.tolist(),
-                               splits_out.tolist())
-        return dst.cpu().numpy()
-
-    rk = a2a(keys_s, row_splits, in_rows, np.int64)
-    rl = a2a(lens_s, row_splits, in_rows, np.int32)
-    rd = a2a(data_s, byte_splits, in_bytes, np.uint8)
-    return rk, rl, rd
-
-
 VERBOSE = os.environ.get("BENCH_VERBOSE", "0") == "1"
 
 
@@ -119,14 +65,28 @@ def _log(msg):
               flush=True)
 
 
+class _CudaBuf:
+    """Minimal __cuda_array_interface__ wrapper so torch can view raw HBM
+    pointers (the engine's device outputs) zero-copy."""
+
+    def __init__(self, ptr, n, typestr):
+        self.__cuda_array_interface__ = {
+            "shape": (int(n),), "typestr": typestr,
+            "data": (int(ptr), False), "version": 2}
+
+
 def run_step(ba, plan, dev_input, world, rank, torch, dist, device, stats):
-    # stage 1: partial hash-agg on the HBM-resident slice.
-    # BATCH_SIZE is the reference's own conf knob (conf.rs:32); raised here so
-    # the ~1M-group output is emitted in few chunks instead of 100 × 10k.
-    # Table sized for the 1M-group workload (2^22 slots at <=3/4 load; the
-    # engine grows 4x on overflow, so this is a starting size, not a cap) —
-    # the default 2^23 doubles slot-scan and compact time.
-    conf = {"BATCH_SIZE": 1 << 20, "AURON_HIP_AGG_TABLE_SLOTS": 1 << 22}
+    """One step, fully device-resident: partial hash-agg -> device
+    repartition (murmur3 pids + stable sort + a11 gather into dest-rank-major
+    partition order, buffered_data.rs:284-351 analog) -> RCCL all-to-all of
+    the partial records (N>1) -> final merge agg. The partial output batch
+    never round-trips through host memory.
+
+    BATCH_SIZE is the reference's own conf knob (conf.rs:32), raised so the
+    ~1M-group partial output is ONE device batch. Table sized for the
+    1M-group workload (2^22 slots; the engine still grows 4x on overflow)."""
+    conf = {"BATCH_SIZE": 4 << 20, "AURON_HIP_AGG_TABLE_SLOTS": 1 << 22,
+            "AURON_HIP_DEVICE_OUTPUT": 1}
 
     def mark(name, t0):
         dt = time.perf_counter() - t0
@@ -136,80 +96,212 @@ def run_step(ba, plan, dev_input, world, rank, torch, dist, device, stats):
         return time.perf_counter()
 
     tm = time.perf_counter()
-    t = ba.Task(plan.plan_partial_only(), device_batches=[dev_input], conf=conf)
+    t = ba.Task(plan.plan_partial_only(), device_batches=[dev_input],
+                conf=conf)
     tm = mark("s1_create", tm)
-    outs = t.run()
+    t.run()
     tm = mark("s1_run", tm)
     stats["agg_update_ns"] += t.metric("agg_update_ns")
     stats["agg_update_rows"] += t.metric("agg_update_rows")
-    stats["num_groups"] = t.metric("num_groups")
-    t.finalize()
-    keys, lens, data = collect_partial(outs)
-    tm = mark("s1_collect", tm)
+    assert len(t.device_outputs) == 1, "partial output must be one batch"
+    ob = t.device_outputs[0]
+    kc, bc = ob["cols"]
+    n = ob["num_rows"]
 
-    # stage 2: 200-way murmur3 partition, owner rank = partition % N
-    pids = ba.partition_ids(keys, NUM_PARTITIONS)
-    tm = mark("s2_partition", tm)
+    # stage 2: device repartition — 200-way murmur3 partition ids + stable
+    # sort + gather into dest-rank-major partition order (the a11 gather)
+    rp = ba.DeviceRepartition(n, kc["ptr"], bc["offsets_ptr"], bc["ptr"],
+                              NUM_PARTITIONS, world,
+                              key_validity_ptr=kc["validity_ptr"])
+    t.finalize()  # repartition copied out of the partial task's buffers
+    tm = mark("s2_repartition", tm)
+
     if world > 1:
-        dest = (pids % world).astype(np.int64)
-        keys, lens, data = exchange(torch, dist, device, keys, lens, data,
-                                    dest, world)
-        tm = mark("s2_exchange", tm)
+        # RCCL all-to-all of (keys, validity bytes, lens, data bytes) by
+        # destination rank; splits from the repartition counters
+        send_rows = torch.tensor(rp.rank_rows, device=device)
+        send_bytes = torch.tensor(rp.rank_bytes, device=device)
+        recv_rows_t = torch.empty_like(send_rows)
+        recv_bytes_t = torch.empty_like(send_bytes)
+        dist.all_to_all_single(recv_rows_t, send_rows)
+        dist.all_to_all_single(recv_bytes_t, send_bytes)
+        recv_rows = recv_rows_t.cpu().tolist()
+        recv_bytes = recv_bytes_t.cpu().tolist()
 
-    # stage 3: final merge agg of (local + received) partial records
-    offs = np.concatenate([[0], np.cumsum(lens)]).astype(np.int32)
+        keys_s = torch.as_tensor(_CudaBuf(rp.keys_ptr, n, "<i8"),
+                                 device=device)
+        offs_s = torch.as_tensor(_CudaBuf(rp.offsets_ptr, n + 1, "<i4"),
+                                 device=device)
+        data_s = torch.as_tensor(
+            _CudaBuf(rp.data_ptr, max(int(offs_s[-1].item()), 1), "|u1"),
+            device=device)
+        lens_s = (offs_s[1:] - offs_s[:-1]).contiguous()
+        # validity bitmap -> byte-per-row for the ragged exchange
+        vbytes_s = None
+        if rp.key_validity_ptr:
+            bm = torch.as_tensor(
+                _CudaBuf(rp.key_validity_ptr, (n + 7) // 8, "|u1"),
+                device=device)
+            bits = torch.tensor([1, 2, 4, 8, 16, 32, 64, 128],
+                                dtype=torch.uint8, device=device)
+            vbytes_s = ((bm.unsqueeze(1) & bits) != 0).reshape(-1)[:n] \
+                .to(torch.uint8).contiguous()
+
+        def a2a(src_t, out_n, splits_out, splits_in):
+            dst = torch.empty(max(int(out_n), 1), dtype=src_t.dtype,
+                              device=device)
+            dist.all_to_all_single(dst[:int(out_n)], src_t,
+                                   [int(x) for x in splits_in],
+                                   [int(x) for x in splits_out])
+            return dst
+
+        nr = sum(recv_rows)
+        keys_r = a2a(keys_s, nr, rp.rank_rows, recv_rows)
+        lens_r = a2a(lens_s, nr, rp.rank_rows, recv_rows)
+        data_r = a2a(data_s[:int(lens_s.sum().item())], sum(recv_bytes),
+                     rp.rank_bytes, recv_bytes)
+        vb_r = a2a(vbytes_s, nr, rp.rank_rows, recv_rows) \
+            if vbytes_s is not None else None
+        offs_r = torch.zeros(nr + 1, dtype=torch.int32, device=device)
+        offs_r[1:] = torch.cumsum(lens_r.to(torch.int64), 0).to(torch.int32)
+        null_count = 0
+        valid_bm_r = None
+        if vb_r is not None:
+            null_count = int((vb_r == 0).sum().item())
+            if null_count:
+                pad = (-nr) % 8
+                vb_pad = torch.cat([vb_r, torch.ones(pad, dtype=torch.uint8,
+                                                     device=device)]) \
+                    if pad else vb_r
+                bits = torch.tensor([1, 2, 4, 8, 16, 32, 64, 128],
+                                    dtype=torch.uint8, device=device)
+                valid_bm_r = ((vb_pad.view(-1, 8) != 0).to(torch.uint8) *
+                              bits).sum(dim=1).to(torch.uint8).contiguous()
+        torch.cuda.synchronize()
+        rp.free()
+        keep = [keys_r, offs_r, data_r, valid_bm_r]
+        fin_cols = [
+            {"ptr": keys_r.data_ptr(), "len": nr,
+             "validity_ptr": valid_bm_r.data_ptr() if valid_bm_r is not None
+             else None, "null_count": null_count},
+            {"ptr": data_r.data_ptr(), "offsets_ptr": offs_r.data_ptr(),
+             "len": nr},
+        ]
+        tm = mark("s2_exchange", tm)
+    else:
+        keep = [rp]
+        fin_cols = [
+            {"ptr": rp.keys_ptr, "len": n,
+             "validity_ptr": rp.key_validity_ptr, "null_count": -1},
+            {"ptr": rp.data_ptr, "offsets_ptr": rp.offsets_ptr, "len": n},
+        ]
+
+    # stage 3: final merge agg of the partition-ordered records (device in)
+    fin_in = ba.DeviceBatch(fin_cols, device_id=int(device.split(":")[1])) \
+        .as_input()
     t2 = ba.Task(plan.plan_final_only(),
-                 conf={"BATCH_SIZE": 1 << 20,
+                 conf={"BATCH_SIZE": 4 << 20,
                        "AURON_HIP_AGG_TABLE_SLOTS": 1 << 22},
-                 batches=[[(keys, None), ("binary", data, offs, None)]])
-    out2 = t2.run()
+                 device_batches=[fin_in])
+    t2.run()
     tm = mark("s3_final", tm)
-    nfinal = sum(ob[0]["values"].shape[0] for ob in out2)
+    nfinal = t2.metric("num_groups")
+    stats["num_groups"] = nfinal
     t2.finalize()
+    if world == 1:
+        keep[0].free()
+    del keep
     return nfinal
 
 
-def cpu_baseline_leg(total_rows):
-    """Time the oracle (CPU restatement, single thread) on a bounded sample of
-    the same workload; reported beside the GPU number, never as `value`."""
+def _cpu_worker(args):
+    """One pinned worker: partial-agg + freeze + partition ids over its row
+    slice; returns (frozen keys, acc bytes, offsets) for the parent's final
+    merge, plus its row count."""
+    import os as _os
+
+    wid, rows, seed, core = args
+    try:
+        _os.sched_setaffinity(0, {core})
+    except OSError:
+        pass
     from oracle import pywrap as oracle
 
+    rng = np.random.default_rng(seed)
+    keys = rng.integers(0, NUM_KEYS, rows).astype(np.int64)
+    vals = rng.integers(0, 1_000_000, rows).astype(np.float64)
+    vv = rng.random(rows) >= NULL_FRAC
+    t0 = time.perf_counter()
+    a = oracle.Agg()
+    a.update(keys, vals, val_valid=vv)
+    data, offsets = a.freeze()
+    g = a.output()
+    h = oracle.hash_cols([(g["keys"], None)])
+    oracle.partition_ids(h, NUM_PARTITIONS)
+    dt = time.perf_counter() - t0
+    return g["keys"], data, offsets, rows, dt
+
+
+def cpu_baseline_leg(total_rows):
+    """The oracle (CPU restatement of the reference's operators) timed on ALL
+    host cores, pinned, 3 warm + 5 timed runs, median — BASELINE.md's CPU
+    protocol — on a bounded sample of the same workload. Each core runs the
+    partial agg + freeze + partition-ids slice; the parent merges the frozen
+    records (the final-stage merge, single-thread like the reference's
+    per-partition reader). Reported beside the GPU number, never as
+    `value`. A single-core figure is included for context."""
+    import multiprocessing as mp
+
+    from oracle import pywrap as oracle
+
+    ncores = os.cpu_count() or 1
+    # calibrate single-core rate on 2M rows
     rng = np.random.default_rng(SEED)
     cal_n = 2_000_000
-    keys = rng.integers(0, NUM_KEYS, cal_n).astype(np.int64)
-    vals = rng.integers(0, 1_000_000, cal_n).astype(np.float64)
-    vv = rng.random(cal_n) >= NULL_FRAC
-
-    def one_pass(k, v, m):
-        a = oracle.Agg()
-        a.update(k, v, val_valid=m)
-        data, offsets = a.freeze()
-        g = a.output()
-        h = oracle.hash_cols([(g["keys"], None)])
-        oracle.partition_ids(h, NUM_PARTITIONS)
-        b = oracle.Agg()
-        b.merge_frozen(g["keys"], data, offsets)
-        return len(k)
-
+    k = rng.integers(0, NUM_KEYS, cal_n).astype(np.int64)
+    v = rng.integers(0, 1_000_000, cal_n).astype(np.float64)
+    m = rng.random(cal_n) >= NULL_FRAC
     t0 = time.perf_counter()
-    one_pass(keys, vals, vv)
-    cal_rate = cal_n / (time.perf_counter() - t0)
-    # size the sample for ~10 s of CPU work, capped at 100M rows
-    sample = int(min(1e8, max(cal_n, cal_rate * 10)))
-    keys = rng.integers(0, NUM_KEYS, sample).astype(np.int64)
-    vals = rng.integers(0, 1_000_000, sample).astype(np.float64)
-    vv = rng.random(sample) >= NULL_FRAC
-    t0 = time.perf_counter()
-    one_pass(keys, vals, vv)
-    dt = time.perf_counter() - t0
+    a = oracle.Agg()
+    a.update(k, v, val_valid=m)
+    data, offsets = a.freeze()
+    g = a.output()
+    h = oracle.hash_cols([(g["keys"], None)])
+    oracle.partition_ids(h, NUM_PARTITIONS)
+    b = oracle.Agg()
+    b.merge_frozen(g["keys"], data, offsets)
+    single_rate = cal_n / (time.perf_counter() - t0)
+
+    # size each timed run for ~2 s of all-core work (5 timed + 3 warm runs
+    # keep the default bench within a couple of minutes), cap 200M rows/run
+    sample = int(min(2e8, max(cal_n * ncores, single_rate * ncores * 2.0)))
+    per_w = sample // ncores
+    args = [(w, per_w, SEED + 7919 * (w + 1), w) for w in range(ncores)]
+
+    times = []
+    with mp.get_context("fork").Pool(ncores) as pool:
+        for it in range(8):  # 3 warm + 5 timed
+            t0 = time.perf_counter()
+            outs = pool.map(_cpu_worker, args)
+            # final merge of every worker's frozen records, single thread
+            fin = oracle.Agg()
+            for gk, gd, go, _, _ in outs:
+                fin.merge_frozen(gk, gd, go)
+            dt = time.perf_counter() - t0
+            if it >= 3:
+                times.append(dt)
+    times.sort()
+    med = times[len(times) // 2]
     return {
-        "value": sample / dt,
+        "value": per_w * ncores / med,
         "unit": "rows/s",
-        "cores": 1,
+        "cores": ncores,
         "kind": "port",
-        "sample": f"{sample} rows of the {total_rows}-row workload "
-                  f"(partial agg + freeze + partition ids + final merge), "
-                  f"single thread",
+        "single_core_value": round(single_rate, 1),
+        "sample": f"{per_w * ncores} rows of the {total_rows}-row workload "
+                  f"(per-core partial agg + freeze + partition ids, "
+                  f"single-thread final merge), {ncores} pinned cores, "
+                  f"median of 5 timed runs after 3 warmups",
     }
 
 

@@ -119,6 +119,14 @@ def lib():
         _lib.auron_version.restype = c.c_char_p
         _lib.auron_get_metric.restype = c.c_int64
         _lib.auron_get_metric.argtypes = [c.c_int64, c.c_char_p]
+        _lib.auron_repartition_device.restype = c.c_int64
+        _lib.auron_repartition_device.argtypes = [
+            c.c_int64, c.c_void_p, c.c_void_p, c.c_void_p, c.c_void_p,
+            c.c_int32, c.c_int32, c.POINTER(c.c_void_p),
+            c.POINTER(c.c_void_p), c.POINTER(c.c_void_p),
+            c.POINTER(c.c_void_p), c.POINTER(c.c_int64),
+            c.POINTER(c.c_int64)]
+        _lib.auron_repartition_free.argtypes = [c.c_int64]
         _lib.auron_debug_decode_plan.restype = c.c_int32
         _lib.auron_debug_decode_plan.argtypes = [c.c_char_p, c.c_size_t,
                                                  c.c_char_p, c.c_size_t]
@@ -217,13 +225,20 @@ class DeviceBatch:
             rows = col["len"]
             n = rows if n is None else n
             assert rows == n
-            bufs = (c.c_void_p * 2)(
-                c.c_void_p(col.get("validity_ptr") or None),
-                c.c_void_p(col["ptr"]))
+            if "offsets_ptr" in col:  # binary column: validity/offsets/data
+                bufs = (c.c_void_p * 3)(
+                    c.c_void_p(col.get("validity_ptr") or None),
+                    c.c_void_p(col["offsets_ptr"]),
+                    c.c_void_p(col["ptr"]))
+                ch.n_buffers = 3
+            else:
+                bufs = (c.c_void_p * 2)(
+                    c.c_void_p(col.get("validity_ptr") or None),
+                    c.c_void_p(col["ptr"]))
+                ch.n_buffers = 2
             ch.length = rows
             ch.null_count = col.get("null_count", 0)
             ch.offset = 0
-            ch.n_buffers = 2
             ch.buffers = bufs
             ch.n_children = 0
             ch.release = None
@@ -258,6 +273,48 @@ def partition_ids(keys, num_partitions):
     if rc != 0:
         raise RuntimeError("auron_partition_ids failed (GPU required)")
     return out
+
+
+class DeviceRepartition:
+    """Device-side exchange prep (auron_repartition_device): partition-id +
+    stable sort + gather of (key, accbuf) records into dest-rank-major
+    partition order, all in HBM. Pointers stay valid until .free()."""
+
+    def __init__(self, n, keys_ptr, offsets_ptr, data_ptr, num_partitions,
+                 world, key_validity_ptr=None):
+        L = lib()
+        ok = c.c_void_p()
+        okv = c.c_void_p()
+        oo = c.c_void_p()
+        od = c.c_void_p()
+        rows = (c.c_int64 * world)()
+        bts = (c.c_int64 * world)()
+        self.handle = L.auron_repartition_device(
+            n, c.c_void_p(keys_ptr), c.c_void_p(key_validity_ptr or None),
+            c.c_void_p(offsets_ptr or None), c.c_void_p(data_ptr or None),
+            num_partitions, world, c.byref(ok), c.byref(okv), c.byref(oo),
+            c.byref(od), rows, bts)
+        if self.handle == 0:
+            raise RuntimeError("auron_repartition_device failed "
+                               "(GPU required)")
+        self.n = n
+        self.keys_ptr = ok.value
+        self.key_validity_ptr = okv.value
+        self.offsets_ptr = oo.value
+        self.data_ptr = od.value
+        self.rank_rows = list(rows)
+        self.rank_bytes = list(bts)
+
+    def free(self):
+        if self.handle:
+            lib().auron_repartition_free(self.handle)
+            self.handle = 0
+
+    def __del__(self):
+        try:
+            self.free()
+        except Exception:
+            pass
 
 
 def _read_bitmap(ptr, n):
@@ -392,13 +449,42 @@ class Task:
             len_out[0] = seg.size
             return 1
 
+        @IMPORT_DEV
+        def import_device_batch(user, devp, schp):
+            # pointers stay valid until finalize (engine contract); binary
+            # data length rides in child.private_data (Arrow C has no
+            # data-length field)
+            d = devp.contents
+            a = d.array
+            cols = []
+            for i in range(a.n_children):
+                ch = a.children[i].contents
+                fmt = self.schema_fields[i][1] if \
+                    i < len(self.schema_fields) else ""
+                col = {"len": int(ch.length), "fmt": fmt,
+                       "validity_ptr": ch.buffers[0]}
+                if ch.n_buffers == 3:
+                    col["offsets_ptr"] = ch.buffers[1]
+                    col["ptr"] = ch.buffers[2]
+                    col["data_len"] = int(
+                        c.cast(ch.private_data, c.c_void_p).value or 0)
+                else:
+                    col["ptr"] = ch.buffers[1]
+                cols.append(col)
+            self.device_outputs.append(
+                {"num_rows": int(a.length), "device_id": int(d.device_id),
+                 "cols": cols})
+
+        self.device_outputs = []
+        use_dev_out = str(self._conf.get("AURON_HIP_DEVICE_OUTPUT", "")) == "1"
         self._cb = AuronCallbacks(
             user=None, get_conf=get_conf, next_input_batch=next_input,
             import_schema=import_schema, import_batch=import_batch,
-            import_device_batch=c.cast(None, IMPORT_DEV), set_error=set_error,
+            import_device_batch=import_device_batch if use_dev_out
+            else c.cast(None, IMPORT_DEV), set_error=set_error,
             next_ipc_bytes=next_ipc)
         self._keep += [get_conf, next_input, import_schema, import_batch,
-                       set_error, next_ipc]
+                       set_error, next_ipc, import_device_batch]
         self.handle = lib().auron_call_native(task_bytes, len(task_bytes),
                                               c.byref(self._cb))
         if self.handle == 0:

@@ -288,6 +288,18 @@ struct HostOutCol {
   bool is_list = false;
 };
 
+// device-resident column staging: partial-agg output that stays in HBM and
+// is exported through import_device_batch (buffers owned by the runtime,
+// valid until auron_finalize — the documented contract for this harness).
+struct DevOutCol {
+  DType dt;
+  DevBuf values;    // prim values / binary data bytes
+  DevBuf offsets;   // binary only: (n+1) int32
+  DevBuf validity;  // LSB bitmap or empty
+  int64_t data_len = 0;
+};
+
+
 void export_batch(int64_t num_rows, std::vector<HostOutCol>&& cols,
                   ArrowArray* out) {
   auto* priv = new ExportPriv;
@@ -477,6 +489,10 @@ class AggOp {
     skip_enabled_ = node.supports_partial_skipping && !merge_mode_;
     skip_ratio_ = conf.get_d("PARTIAL_AGG_SKIPPING_RATIO", 0.999);
     skip_min_rows_ = conf.get_i("PARTIAL_AGG_SKIPPING_MIN_ROWS", 20000);
+    // device output: partial freeze output stays in HBM (ArrowDeviceArray
+    // export) so the bench/exchange chain never round-trips through host
+    device_out_ = conf.get_i("AURON_HIP_DEVICE_OUTPUT", 0) != 0 &&
+                  !final_output_;
     int64_t slots = conf.get_i("AURON_HIP_AGG_TABLE_SLOTS", 1 << 23);
     conf_coll_cap_ =
         std::max<int64_t>(1024, conf.get_i("AURON_HIP_COLLECT_POOL", 16 << 20));
@@ -939,6 +955,72 @@ class AggOp {
     skipped_.clear();
     held_.clear();
     return out;
+  }
+
+  // device-resident variant of finish(): partial (key + a8 Binary) batches
+  // stay in HBM. Spill / partial-skipping fall outside this mode (the bench
+  // and exchange paths that use it never enter them) and FAIL loudly.
+  std::vector<std::pair<int64_t, std::vector<DevOutCol>>> finish_dev() {
+    std::vector<std::pair<int64_t, std::vector<DevOutCol>>> out;
+    drain_timing();
+    prepare_collect();
+    if (!spill_.empty())
+      FAIL("AURON_HIP_DEVICE_OUTPUT with spill unsupported (unset the conf)");
+    if (!skipped_.empty())
+      FAIL("AURON_HIP_DEVICE_OUTPUT with partial skipping unsupported");
+    DevBuf order, first;
+    int64_t ng = table_order(&order, &first);
+    for (int64_t beg = 0; beg < ng; beg += batch_size_) {
+      int64_t len = std::min(batch_size_, ng - beg);
+      out.emplace_back(emit_groups_dev(order.get<uint32_t>() + beg, len));
+    }
+    AURON_HIP(hipStreamSynchronize(stream_));  // order/first return to pool
+    held_.clear();
+    return out;
+  }
+
+  std::pair<int64_t, std::vector<DevOutCol>> emit_groups_dev(
+      const uint32_t* order_slots, int64_t n) {
+    size_t bm = (n + 7) / 8;
+    std::vector<DevOutCol> cols(2);
+    DevOutCol& kc = cols[0];
+    DevOutCol& bc = cols[1];
+    kc.dt = key_dt_ == DType::Unsupported ? DType::Int64 : key_dt_;
+    DevBuf keys(n * 8), sums(n * 8), svalid(bm), cnts(n * 8);
+    kc.validity.alloc(bm);
+    launch_agg_gather_out(t_, order_slots, n, keys.get<int64_t>(),
+                          kc.validity.get<uint8_t>(), sums.get<double>(),
+                          svalid.get<uint8_t>(), cnts.get<long long>(),
+                          nullptr, nullptr, nullptr, nullptr, stream_);
+    if (kc.dt == DType::Int32) {
+      kc.values.alloc(n * 4);
+      launch_narrow_i64_i32(keys.get<int64_t>(), n, kc.values.get<int32_t>(),
+                            stream_);
+    } else {
+      kc.values = std::move(keys);
+    }
+    // freeze lens -> device exclusive scan -> offsets (+ total in slot n)
+    bc.dt = DType::Binary;
+    DevBuf lens((n + 1) * 4);
+    bc.offsets.alloc((n + 1) * 4);
+    launch_agg_freeze_len(t_, order_slots, n, lens.get<int32_t>(), layout_,
+                          stream_);
+    size_t tb = 0;
+    scan_counts_matrix(lens.get<uint32_t>(), bc.offsets.get<uint32_t>(),
+                       n + 1, nullptr, &tb, stream_);
+    DevBuf scan_tmp(tb);
+    scan_counts_matrix(lens.get<uint32_t>(), bc.offsets.get<uint32_t>(),
+                       n + 1, scan_tmp.get(), &tb, stream_);
+    if (pinned_meta_.size() < 16) pinned_meta_.alloc(16);
+    AURON_HIP(hipMemcpyAsync(pinned_meta_.get(),
+                             bc.offsets.get<uint32_t>() + n, 4,
+                             hipMemcpyDeviceToHost, stream_));
+    AURON_HIP(hipStreamSynchronize(stream_));
+    bc.data_len = (int64_t)*pinned_meta_.get<uint32_t>();
+    bc.values.alloc(bc.data_len ? bc.data_len : 1);
+    launch_agg_freeze_write(t_, order_slots, n, bc.offsets.get<int32_t>(),
+                            bc.values.get<uint8_t>(), layout_, stream_);
+    return {n, std::move(cols)};
   }
 
   uint64_t num_groups_host() {
@@ -1620,6 +1702,7 @@ class AggOp {
   std::string key_name_;
   int64_t batch_size_ = 10000;
   bool skip_enabled_ = false, skipping_ = false;
+  bool device_out_ = false;
   double skip_ratio_ = 0.999;
   int64_t skip_min_rows_ = 20000;
   uint64_t row_cursor_ = 0;
@@ -2161,6 +2244,10 @@ struct Runtime {
   std::vector<OutField> out_fields;
   std::vector<std::pair<int64_t, std::vector<HostOutCol>>> outputs;
   size_t emit_idx = 0;
+  // device-resident outputs (AURON_HIP_DEVICE_OUTPUT): exported through
+  // import_device_batch; DevBufs stay owned here until finalize
+  std::vector<std::pair<int64_t, std::vector<DevOutCol>>> outputs_dev;
+  size_t emit_dev_idx = 0;
   std::map<std::string, int64_t> metrics;
 
   ~Runtime() {
@@ -2272,17 +2359,27 @@ struct Runtime {
       if (st.kind == Stage::AggS) {
         if (!first_agg) first_agg = st.agg.get();
         last_agg = st.agg.get();
-        auto outs = st.agg->finish();
         bool terminal = i + 1 == stages_.size();
-        for (auto& ob : outs) {
-          if (terminal) {
-            outputs.push_back(std::move(ob));
-          } else {
-            DevBatch b = host_out_to_dev(ob, st.agg->output_fields());
-            feed(i + 1, std::move(b));
+        if (st.agg->device_out_) {
+          if (!terminal)
+            FAIL("AURON_HIP_DEVICE_OUTPUT on a non-terminal Agg unsupported");
+          if (!cb.import_device_batch)
+            FAIL("AURON_HIP_DEVICE_OUTPUT set but import_device_batch is "
+                 "NULL");
+          outputs_dev = st.agg->finish_dev();
+          out_fields = st.agg->output_fields();
+        } else {
+          auto outs = st.agg->finish();
+          for (auto& ob : outs) {
+            if (terminal) {
+              outputs.push_back(std::move(ob));
+            } else {
+              DevBatch b = host_out_to_dev(ob, st.agg->output_fields());
+              feed(i + 1, std::move(b));
+            }
           }
+          if (terminal) out_fields = st.agg->output_fields();
         }
-        if (terminal) out_fields = st.agg->output_fields();
       } else if (st.kind == Stage::ShuffleS) {
         st.shuffle->finish();
         has_shuffle = true;
@@ -2300,6 +2397,7 @@ struct Runtime {
     metrics["input_rows"] = input_rows;
     int64_t out_rows = 0;
     for (auto& o : outputs) out_rows += o.first;
+    for (auto& o : outputs_dev) out_rows += o.first;
     metrics["output_rows"] = out_rows;
     metrics["elapsed_compute_ns"] =
         std::chrono::duration_cast<std::chrono::nanoseconds>(
@@ -2681,6 +2779,46 @@ struct Runtime {
         cb.import_schema(cb.user, &s);
         schema_sent = true;
       }
+      if (emit_dev_idx < outputs_dev.size()) {
+        auto& ob = outputs_dev[emit_dev_idx++];
+        // per-call export scaffolding: child arrays point at the DevBufs
+        // (owned by outputs_dev until finalize); the struct itself is only
+        // valid during the callback — the consumer copies the pointers out.
+        std::vector<ArrowArray> children(ob.second.size());
+        std::vector<ArrowArray*> child_ptrs(ob.second.size());
+        std::vector<std::array<const void*, 3>> bufs(ob.second.size());
+        for (size_t ci = 0; ci < ob.second.size(); ci++) {
+          DevOutCol& oc = ob.second[ci];
+          ArrowArray& ch = children[ci];
+          memset(&ch, 0, sizeof(ch));
+          ch.length = ob.first;
+          ch.null_count = -1;  // unknown; validity bitmap always attached
+          bufs[ci][0] = oc.validity.size() ? oc.validity.get() : nullptr;
+          if (oc.dt == DType::Binary || oc.dt == DType::Utf8) {
+            bufs[ci][1] = oc.offsets.get();
+            bufs[ci][2] = oc.values.get();
+            ch.n_buffers = 3;
+            // Arrow C has no data-length field; stash it for the consumer
+            ch.private_data = (void*)(intptr_t)oc.data_len;
+          } else {
+            bufs[ci][1] = oc.values.get();
+            ch.n_buffers = 2;
+          }
+          ch.buffers = (const void**)bufs[ci].data();
+          child_ptrs[ci] = &ch;
+        }
+        ArrowDeviceArray dev;
+        memset(&dev, 0, sizeof(dev));
+        dev.array.length = ob.first;
+        dev.array.n_children = (int64_t)children.size();
+        dev.array.children = child_ptrs.data();
+        int did = 0;
+        (void)hipGetDevice(&did);
+        dev.device_id = did;
+        dev.device_type = 10;  // ARROW_DEVICE_ROCM
+        cb.import_device_batch(cb.user, &dev, nullptr);
+        return 1;
+      }
       if (emit_idx >= outputs.size()) return 0;
       auto& ob = outputs[emit_idx++];
       ArrowArray a;
@@ -2961,6 +3099,136 @@ int32_t auron_partition_ids(const int64_t* keys, int64_t n, int32_t P,
     return -1;
   }
 }
+
+// ---- device repartition (in-memory exchange prep) --------------------------
+// The multi-GPU exchange analog of sort_batches_by_partition_id +
+// create_batch_interleaver (buffered_data.rs:284-351, selection.rs:65-300):
+// murmur3(seed 42) partition ids -> stable sort by (dest rank = pid % world,
+// pid) -> gather the (key, a8 accbuf) records into dest-rank-major,
+// partition-ordered layout entirely in HBM, returning the per-rank row/byte
+// splits the RCCL all-to-all needs. Input/output pointers are device memory;
+// output buffers are owned by the returned handle until
+// auron_repartition_free.
+namespace auron {
+namespace {
+struct RepartOut {
+  DevBuf keys, kvalid, offsets, data;
+  hipStream_t stream = nullptr;
+};
+std::mutex g_rp_mu;
+std::map<int64_t, std::unique_ptr<RepartOut>> g_repart;
+int64_t g_rp_next = 1;
+}  // namespace
+}  // namespace auron
+
+extern "C" {
+
+int64_t auron_repartition_device(int64_t n, const void* keys,
+                                 const void* key_validity,
+                                 const void* offsets, const void* data,
+                                 int32_t num_partitions, int32_t world,
+                                 const void** out_keys,
+                                 const void** out_key_validity,
+                                 const void** out_offsets,
+                                 const void** out_data,
+                                 int64_t* rank_rows, int64_t* rank_bytes) {
+  try {
+    if (n <= 0 || num_partitions <= 0 || world <= 0 || world > 64) return 0;
+    auto rp = std::make_unique<RepartOut>();
+    AURON_HIP(hipStreamCreate(&rp->stream));
+    hipStream_t s = rp->stream;
+    DevBuf hash(n * 4), pids(n * 4), ord(n * 4), ord_sorted(n * 4);
+    DevBuf idx(n * 4), perm(n * 4), counts(world * 16);
+    launch_hash_init(hash.get<int32_t>(), 42, n, s);
+    launch_hash_fold_i64((const int64_t*)keys, nullptr, n, hash.get<int32_t>(),
+                         s);
+    launch_pmod(hash.get<int32_t>(), n, num_partitions, pids.get<uint32_t>(),
+                s);
+    launch_exchange_ord(pids.get<uint32_t>(), n, (uint32_t)num_partitions,
+                        (uint32_t)world, ord.get<uint32_t>(), s);
+    AURON_HIP(hipMemsetAsync(counts.get(), 0, world * 16, s));
+    launch_dest_counts(pids.get<uint32_t>(), (const int32_t*)offsets, n,
+                       (uint32_t)world, counts.get<unsigned long long>(),
+                       counts.get<unsigned long long>() + world, s);
+    launch_iota_u32(idx.get<uint32_t>(), n, s);
+    // ord < world * P: sort only the live bits
+    int end_bit = 1;
+    while ((1u << end_bit) < (uint32_t)world * (uint32_t)num_partitions &&
+           end_bit < 32)
+      end_bit++;
+    size_t tb = 0;
+    sort_pairs_u32_u32(ord.get<uint32_t>(), idx.get<uint32_t>(),
+                       ord_sorted.get<uint32_t>(), perm.get<uint32_t>(), n,
+                       end_bit, nullptr, &tb, s);
+    DevBuf tmp(tb);
+    sort_pairs_u32_u32(ord.get<uint32_t>(), idx.get<uint32_t>(),
+                       ord_sorted.get<uint32_t>(), perm.get<uint32_t>(), n,
+                       end_bit, tmp.get(), &tb, s);
+    rp->keys.alloc(n * 8);
+    launch_gather_8((const uint8_t*)keys, perm.get<uint32_t>(), n,
+                    rp->keys.get<uint8_t>(), s);
+    if (key_validity) {
+      rp->kvalid.alloc((n + 7) / 8);
+      launch_gather_bits((const uint8_t*)key_validity, perm.get<uint32_t>(), n,
+                         rp->kvalid.get<uint8_t>(), s);
+    }
+    if (offsets && data) {
+      DevBuf lens((n + 1) * 4);
+      rp->offsets.alloc((n + 1) * 4);
+      launch_gather_lens((const int32_t*)offsets, perm.get<uint32_t>(), n,
+                         lens.get<int32_t>(), s);
+      size_t stb = 0;
+      scan_counts_matrix(lens.get<uint32_t>(), rp->offsets.get<uint32_t>(),
+                         n + 1, nullptr, &stb, s);
+      DevBuf stmp(stb);
+      scan_counts_matrix(lens.get<uint32_t>(), rp->offsets.get<uint32_t>(),
+                         n + 1, stmp.get(), &stb, s);
+      PinnedBuf pin;
+      pin.alloc(16);
+      AURON_HIP(hipMemcpyAsync(pin.get(), rp->offsets.get<uint32_t>() + n, 4,
+                               hipMemcpyDeviceToHost, s));
+      AURON_HIP(hipStreamSynchronize(s));
+      int64_t total = (int64_t)*pin.get<uint32_t>();
+      rp->data.alloc(total ? total : 1);
+      launch_gather_bytes((const uint8_t*)data, (const int32_t*)offsets,
+                          perm.get<uint32_t>(), rp->offsets.get<int32_t>(), n,
+                          rp->data.get<uint8_t>(), s);
+    }
+    std::vector<unsigned long long> h_counts(world * 2);
+    AURON_HIP(hipMemcpyAsync(h_counts.data(), counts.get(), world * 16,
+                             hipMemcpyDeviceToHost, s));
+    AURON_HIP(hipStreamSynchronize(s));
+    for (int32_t w = 0; w < world; w++) {
+      if (rank_rows) rank_rows[w] = (int64_t)h_counts[w];
+      if (rank_bytes) rank_bytes[w] = (int64_t)h_counts[world + w];
+    }
+    if (out_keys) *out_keys = rp->keys.get();
+    if (out_key_validity)
+      *out_key_validity = key_validity ? rp->kvalid.get() : nullptr;
+    if (out_offsets) *out_offsets = rp->offsets.get();
+    if (out_data) *out_data = rp->data.get();
+    std::lock_guard<std::mutex> lk(g_rp_mu);
+    int64_t h = g_rp_next++;
+    g_repart[h] = std::move(rp);
+    return h;
+  } catch (const std::exception&) {
+    return 0;
+  }
+}
+
+void auron_repartition_free(int64_t handle) {
+  std::lock_guard<std::mutex> lk(g_rp_mu);
+  auto it = g_repart.find(handle);
+  if (it != g_repart.end()) {
+    if (it->second->stream) {
+      (void)hipStreamSynchronize(it->second->stream);
+      (void)hipStreamDestroy(it->second->stream);
+    }
+    g_repart.erase(it);
+  }
+}
+
+}  // extern "C"
 
 // test-only introspection: decode a TaskDefinition and render a one-line
 // summary (verifies the hand-rolled proto reader against encoders)

@@ -363,6 +363,14 @@ void launch_gather_bytes(const uint8_t* src_data, const int32_t* src_offsets,
                          const uint32_t* perm, const int32_t* dst_offsets,
                          int64_t n, uint8_t* dst_data, hipStream_t s);
 
+// device repartition (exchange prep): dest-rank-major partition order key
+// and per-rank all-to-all split counts
+void launch_exchange_ord(const uint32_t* pids, int64_t n, uint32_t P,
+                         uint32_t world, uint32_t* ord, hipStream_t s);
+void launch_dest_counts(const uint32_t* pids, const int32_t* offsets,
+                        int64_t n, uint32_t world, unsigned long long* rows,
+                        unsigned long long* bytes, hipStream_t s);
+
 // sort groups by first_row: rocprim radix sort pairs wrapper
 void sort_pairs_u64_u32(const unsigned long long* keys_in, const uint32_t* vals_in,
                         unsigned long long* keys_out, uint32_t* vals_out,

@@ -1550,4 +1550,62 @@ void launch_cast_i64_f64(const int64_t* in, int64_t n, double* out,
   check_launch("k_cast_i64_f64");
 }
 
+// ---- device repartition helpers (in-memory exchange prep) ------------------
+// ord = (pid % world) * P + pid: sorting by ord groups rows by destination
+// rank, partition-ordered within each rank (buffered_data.rs:284-351 analog
+// with the rank grouping the RCCL all-to-all needs).
+__global__ void k_exchange_ord(const uint32_t* __restrict__ pids, int64_t n,
+                               uint32_t P, uint32_t world,
+                               uint32_t* __restrict__ ord) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    uint32_t p = pids[i];
+    ord[i] = (p % world) * P + p;
+  }
+}
+
+void launch_exchange_ord(const uint32_t* pids, int64_t n, uint32_t P,
+                         uint32_t world, uint32_t* ord, hipStream_t s) {
+  hipLaunchKernelGGL(k_exchange_ord, dim3(grid_for(n)), dim3(BLOCK), 0, s,
+                     pids, n, P, world, ord);
+  check_launch("k_exchange_ord");
+}
+
+// per-destination-rank row and byte counts (the all-to-all splits); LDS
+// staged, world <= 8 on one node
+__global__ void k_dest_counts(const uint32_t* __restrict__ pids,
+                              const int32_t* __restrict__ offsets, int64_t n,
+                              uint32_t world,
+                              unsigned long long* __restrict__ rows,
+                              unsigned long long* __restrict__ bytes) {
+  __shared__ unsigned long long l_rows[64], l_bytes[64];
+  for (uint32_t w = threadIdx.x; w < world; w += blockDim.x) {
+    l_rows[w] = 0;
+    l_bytes[w] = 0;
+  }
+  __syncthreads();
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    uint32_t d = pids[i] % world;
+    atomicAdd(&l_rows[d], 1ull);
+    if (offsets)
+      atomicAdd(&l_bytes[d],
+                (unsigned long long)(offsets[i + 1] - offsets[i]));
+  }
+  __syncthreads();
+  for (uint32_t w = threadIdx.x; w < world; w += blockDim.x) {
+    if (l_rows[w]) atomicAdd(&rows[w], l_rows[w]);
+    if (l_bytes[w]) atomicAdd(&bytes[w], l_bytes[w]);
+  }
+}
+
+void launch_dest_counts(const uint32_t* pids, const int32_t* offsets,
+                        int64_t n, uint32_t world, unsigned long long* rows,
+                        unsigned long long* bytes, hipStream_t s) {
+  if (world > 64) throw std::runtime_error("dest_counts: world > 64");
+  hipLaunchKernelGGL(k_dest_counts, dim3(grid_for(n)), dim3(BLOCK), 0, s,
+                     pids, offsets, n, world, rows, bytes);
+  check_launch("k_dest_counts");
+}
+
 }  // namespace auron

@@ -140,6 +140,27 @@ void auron_finalize(int64_t handle);
 /* Global teardown. Mirrors exec.rs:138-143. */
 void auron_on_exit(void);
 
+/* ---- device exchange prep (multi-GPU shuffle leg) ----
+ * In-memory analog of sort_batches_by_partition_id + create_batch_interleaver
+ * (buffered_data.rs:284-351, selection.rs:65-300) for the RCCL all-to-all
+ * exchange: murmur3(seed 42) partition ids -> stable sort by
+ * (dest rank = pid % world, pid) -> gather (key, accbuf) records into
+ * dest-rank-major partition order, all in HBM. All data pointers are DEVICE
+ * memory; out_* buffers are owned by the returned handle (>0) until
+ * auron_repartition_free; rank_rows/rank_bytes are host arrays[world] = the
+ * all-to-all split sizes. offsets/data may be NULL (keys-only). Returns 0 on
+ * error. */
+int64_t auron_repartition_device(int64_t n, const void* keys,
+                                 const void* key_validity,
+                                 const void* offsets, const void* data,
+                                 int32_t num_partitions, int32_t world,
+                                 const void** out_keys,
+                                 const void** out_key_validity,
+                                 const void** out_offsets,
+                                 const void** out_data,
+                                 int64_t* rank_rows, int64_t* rank_bytes);
+void auron_repartition_free(int64_t handle);
+
 /* ---- introspection for tests/bench ---- */
 /* Version / build info, e.g. "auron-hip 0.1 gfx950". */
 const char* auron_version(void);
