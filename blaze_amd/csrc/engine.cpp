@@ -581,6 +581,7 @@ class AggOp {
       // two-phase path (update mode, large chunks): its table inserts are
       // bounded by counted staged/leftover lists, not by chunk rows, so it
       // chunks on partition-buffer size instead of table free slots.
+      init_agg2_conf();  // chunk bound must be read BEFORE sizing the chunk
       // MIN/MAX agg sets stay single-phase: the LDS bucket kernel's slot
       // holds {key,cnt,sum,first} only (perf note in DESIGN.md)
       if (!merge_mode_ && !has_mm_ && !has_first_ && !has_coll_ &&
@@ -1093,6 +1094,38 @@ class AggOp {
   int64_t agg2_chunk_max_ = 256 << 20;
   static constexpr int AGG2_NBUCK_LOG2 = 10;           // 1024 buckets: scatter write-line footprint ~L2-sized (512 buckets overflow the 2048-slot LDS window: 2.3% leftovers, 3x slower)
 
+  // Read the two-phase tuning knobs once, BEFORE the first chunk size is
+  // computed (the scratch buffers are sized from agg2_chunk_max_, so the
+  // bound must be final by then).
+  void init_agg2_conf() {
+    if (agg2_conf_read_) return;
+    agg2_conf_read_ = true;
+    // A/B toggle for the partition-record layout: the 24B AoS record
+    // (default) vs 16B kv + 4B rowv split streams — measured a wash within
+    // box noise; AURON_AGG2_SPLIT=1 selects the split for experiments.
+    const char* e = getenv("AURON_AGG2_SPLIT");
+    agg2_split_ = (e && e[0] == '1');
+    // hist/scatter grid (blocks = 1<<log2): 512 default (same-box sweeps:
+    // occupancy beats write locality — grid 256/128 measured slower)
+    const char* g = getenv("AURON_AGG2_GRID_LOG2");
+    if (g && g[0]) agg2_grid_log2_ = atoi(g);
+    if (agg2_grid_log2_ < 6 || agg2_grid_log2_ > AGG2_GRID_LOG2_MAX)
+      agg2_grid_log2_ = 9;
+    // hist/scatter/bucket workgroup size: 1024 threads = 16 waves/block
+    // fill the chip's 32-wave/CU slots (the 256-thread launch left the
+    // bucket kernel at 12 waves/CU = 38% occupancy; microarch guide
+    // "Chip-level parameters"/"Register files")
+    const char* bl = getenv("AURON_AGG2_BLOCK");
+    if (bl && bl[0]) agg2_block_ = atoi(bl);
+    if (agg2_block_ != 256 && agg2_block_ != 512 && agg2_block_ != 1024)
+      agg2_block_ = 1024;
+    const char* cm = getenv("AURON_AGG2_CHUNK_M");  // millions of rows
+    if (cm && cm[0]) {
+      int64_t m = atoll(cm);
+      if (m >= 4 && m <= 1024) agg2_chunk_max_ = m << 20;
+    }
+  }
+
   // Two-phase aggregation of rows [done, done+chunk) of batch b:
   // histogram -> scatter to bucket-major SoA -> per-bucket LDS aggregate ->
   // merge counted staged groups into the slot table (kernels_agg2.hip).
@@ -1106,34 +1139,6 @@ class AggOp {
     const uint8_t* vv = val.validity ? val.validity + done / 8 : nullptr;
 
     if (!d_partkv_) {
-      // A/B toggle for the partition-record layout: the 24B AoS record
-      // (default) vs 16B kv + 4B rowv split streams. Same-box A/B (4
-      // interleaved 1B-row runs) measured the split at 82.1/82.6 ms/step vs
-      // AoS at 78.4/88.9 — a wash within box noise, so the simpler AoS
-      // stays; AURON_AGG2_SPLIT=1 selects the split for experiments.
-      const char* e = getenv("AURON_AGG2_SPLIT");
-      agg2_split_ = (e && e[0] == '1');
-      // hist/scatter grid (blocks = 1<<log2): 512 default; runtime-tunable
-      // for write-line-footprint experiments (PMC measured ~1.9x write
-      // amplification at 512 blocks x 1024 buckets = one open line per L2
-      // line per XCD)
-      const char* g = getenv("AURON_AGG2_GRID_LOG2");
-      if (g && g[0]) agg2_grid_log2_ = atoi(g);
-      if (agg2_grid_log2_ < 6 || agg2_grid_log2_ > AGG2_GRID_LOG2_MAX)
-        agg2_grid_log2_ = 9;
-      // hist/scatter/bucket workgroup size: 1024 threads = 16 waves/block
-      // fill the chip's 32-wave/CU slots (the 256-thread launch left the
-      // bucket kernel at 12 waves/CU = 38% occupancy; microarch guide
-      // "Chip-level parameters"/"Register files")
-      const char* bl = getenv("AURON_AGG2_BLOCK");
-      if (bl && bl[0]) agg2_block_ = atoi(bl);
-      if (agg2_block_ != 256 && agg2_block_ != 512 && agg2_block_ != 1024)
-        agg2_block_ = 1024;
-      const char* cm = getenv("AURON_AGG2_CHUNK_M");  // millions of rows
-      if (cm && cm[0]) {
-        int64_t m = atoll(cm);
-        if (m >= 4 && m <= 1024) agg2_chunk_max_ = m << 20;
-      }
       if (agg2_split_) {
         d_partkv_.alloc(agg2_chunk_max_ * sizeof(PartKV));
         d_rowv_.alloc(agg2_chunk_max_ * 4);
@@ -1720,6 +1725,7 @@ class AggOp {
   PinnedBuf pinned_meta_, pinned_emit_;
   // two-phase scratch (allocated on first large chunk)
   bool agg2_split_ = false;
+  bool agg2_conf_read_ = false;
   int agg2_grid_log2_ = 9;
   int agg2_block_ = 1024;
   DevBuf d_partkv_, d_rowv_, d_leftover_, d_counts_, d_scanned_, d_scan_tmp_, d_offsets_,

@@ -1,7 +1,9 @@
-"""Multi-process (world_size=2, gloo on CPU) coverage of bench.py's RCCL
-exchange logic: partial records routed by partition%N, exchanged with
-all_to_all_single, merged — the result must equal the single-process oracle
-aggregate. This pins the N>1 control flow without a GPU."""
+"""Multi-process (world_size=2, gloo on CPU) coverage of the bench's RCCL
+exchange control flow: partial records sorted into dest-rank-major partition
+order (the numpy restatement of auron_repartition_device, which is itself
+GPU-parity-tested in test_gpu_device_output.py), exchanged with
+all_to_all_single using the per-rank splits, merged — the result must equal
+the single-process oracle aggregate. Pins the N>1 flow without a GPU."""
 import os
 
 import numpy as np
@@ -14,7 +16,6 @@ def _worker(rank, world, q):
     import torch
     import torch.distributed as dist
 
-    import bench
     from oracle import pywrap as oracle
 
     dist.init_process_group("gloo", rank=rank, world_size=world)
@@ -30,11 +31,42 @@ def _worker(rank, world, q):
         data, offs64 = part.freeze()
         lens = (offs64[1:] - offs64[:-1]).astype(np.int32)
 
+        # dest-rank-major partition-ordered sort + gather (the
+        # auron_repartition_device contract, restated on host)
         hashes = oracle.hash_cols([(g["keys"], None)])
         pids = oracle.partition_ids(hashes, 200)
         dest = (pids % world).astype(np.int64)
-        rk, rl, rd = bench.exchange(torch, dist, "cpu", g["keys"], lens,
-                                    np.asarray(data, np.uint8), dest, world)
+        order = np.argsort(dest * 200 + pids, kind="stable")
+        keys_s = g["keys"][order]
+        lens_s = lens[order]
+        offs = np.concatenate([[0], np.cumsum(lens)]).astype(np.int64)
+        out_base = np.concatenate([[0], np.cumsum(lens_s)])[:-1]
+        data_np = np.asarray(data, np.uint8)
+        pos = np.repeat(offs[order] - out_base, lens_s) + \
+            np.arange(int(lens_s.sum()), dtype=np.int64)
+        data_s = data_np[pos]
+        row_splits = np.bincount(dest, minlength=world).astype(np.int64)
+        byte_splits = np.zeros(world, dtype=np.int64)
+        np.add.at(byte_splits, dest, lens.astype(np.int64))
+
+        def a2a(src_np, splits_out, splits_in):
+            s = torch.from_numpy(np.ascontiguousarray(src_np))
+            dst = torch.empty(int(splits_in.sum()), dtype=s.dtype)
+            dist.all_to_all_single(dst, s, splits_in.tolist(),
+                                   splits_out.tolist())
+            return dst.numpy()
+
+        t_rows = torch.tensor(row_splits)
+        r_rows_t = torch.empty_like(t_rows)
+        dist.all_to_all_single(r_rows_t, t_rows)
+        in_rows = r_rows_t.numpy()
+        t_bytes = torch.tensor(byte_splits)
+        r_bytes_t = torch.empty_like(t_bytes)
+        dist.all_to_all_single(r_bytes_t, t_bytes)
+        in_bytes = r_bytes_t.numpy()
+        rk = a2a(keys_s, row_splits, in_rows)
+        rl = a2a(lens_s, row_splits, in_rows)
+        rd = a2a(data_s, byte_splits, in_bytes)
         # every received key must belong to this rank
         rh = oracle.hash_cols([(rk, None)])
         rp = oracle.partition_ids(rh, 200)
