@@ -1092,7 +1092,10 @@ class AggOp {
   // tails and per-chunk syncs (24 B/row + 24 B/row leftover scratch; 256M
   // rows = 12 GB of the 288 GB HBM). Runtime-tunable for A/B sweeps.
   int64_t agg2_chunk_max_ = 256 << 20;
-  static constexpr int AGG2_NBUCK_LOG2 = 10;           // 1024 buckets: scatter write-line footprint ~L2-sized (512 buckets overflow the 2048-slot LDS window: 2.3% leftovers, 3x slower)
+  static constexpr int AGG2_NBUCK_LOG2 = 10;
+  static constexpr int AGG3_NBUCK_LOG2 = 9;   // v3: 512 buckets
+  static constexpr int AGG3_NBUCK = 1 << AGG3_NBUCK_LOG2;
+  static constexpr int AGG3_GRID_LOG2 = 8;    // v3 scatter: 256 x 1024-thr           // 1024 buckets: scatter write-line footprint ~L2-sized (512 buckets overflow the 2048-slot LDS window: 2.3% leftovers, 3x slower)
 
   // Read the two-phase tuning knobs once, BEFORE the first chunk size is
   // computed (the scratch buffers are sized from agg2_chunk_max_, so the
@@ -1124,6 +1127,11 @@ class AggOp {
       int64_t m = atoll(cm);
       if (m >= 4 && m <= 1024) agg2_chunk_max_ = m << 20;
     }
+    // v3 pipeline (default): LDS-staged packet scatter, 512 buckets,
+    // 4096-slot bucket window — kernels_agg3.hip header comment has the
+    // evidence trail. AURON_AGG2_V3=0 falls back to the v2 kernels.
+    const char* v3 = getenv("AURON_AGG2_V3");
+    agg2_v3_ = !(v3 && v3[0] == '0');
   }
 
   // Two-phase aggregation of rows [done, done+chunk) of batch b:
@@ -1139,7 +1147,11 @@ class AggOp {
     const uint8_t* vv = val.validity ? val.validity + done / 8 : nullptr;
 
     if (!d_partkv_) {
-      if (agg2_split_) {
+      if (agg2_v3_) {
+        // 24B records + one line of alignment padding per (block,bucket)
+        d_partkv_.alloc(agg2_chunk_max_ * 24 +
+                        ((int64_t)AGG3_NBUCK << AGG3_GRID_LOG2) * 64);
+      } else if (agg2_split_) {
         d_partkv_.alloc(agg2_chunk_max_ * sizeof(PartKV));
         d_rowv_.alloc(agg2_chunk_max_ * 4);
       } else {
@@ -1150,11 +1162,14 @@ class AggOp {
       d_counts_.alloc((mat + 1) * 4);   // +1: scan total slot
       d_scanned_.alloc((mat + 1) * 4);
       d_offsets_.alloc((nbuck + 1) * 4);
+      if (agg2_v3_) d_linesz_.alloc((mat + 1) * 4);
       size_t tb = 0;
       scan_counts_matrix(d_counts_.get<uint32_t>(), d_scanned_.get<uint32_t>(),
                          mat + 1, nullptr, &tb, stream_);
       d_scan_tmp_.alloc(tb);
-      d_staged_.alloc((int64_t)nbuck * AGG2_LSLOTS * sizeof(StagedGroup));
+      d_staged_.alloc(std::max<int64_t>((int64_t)nbuck * AGG2_LSLOTS,
+                                        (int64_t)AGG3_NBUCK * 4096) *
+                      sizeof(StagedGroup));
       d_counters_.alloc(24);  // staged_n, lo_n, special_rows
       // layout: counts[nbuck+1] | offs[nbuck+1] | counters[2] (8-aligned)
       pinned_agg2_.alloc(2 * (size_t)(nbuck + 1) * 4 + 64);
@@ -1166,8 +1181,35 @@ class AggOp {
     // P1: per-block histogram matrix (+ special-row count). ONE host sync
     // per chunk: the scan total feeds the offsets kernel on-device, and the
     // special/staged/leftover counters are read back together after phase A.
-    int64_t mat = (int64_t)nbuck << agg2_grid_log2_;
     AURON_HIP(hipMemsetAsync(d_counters_.get(), 0, 24, stream_));
+    if (agg2_v3_) {
+      // v3: hist -> line-padded scan -> LDS-staged packet scatter -> 4096-
+      // slot bucket aggregation (kernels_agg3.hip)
+      int64_t mat3 = (int64_t)AGG3_NBUCK << AGG3_GRID_LOG2;
+      launch_agg2_hist(keys, kv, chunk, AGG3_NBUCK_LOG2, AGG3_GRID_LOG2,
+                       d_counts_.get<uint32_t>(),
+                       (uint32_t*)(d_counters_.get<uint8_t>() + 16), 1024,
+                       stream_);
+      launch_agg3_line_sizes(d_counts_.get<uint32_t>(), mat3 + 1,
+                             d_linesz_.get<uint32_t>(), stream_);
+      size_t tb3 = d_scan_tmp_.size();
+      scan_counts_matrix(d_linesz_.get<uint32_t>(),
+                         d_scanned_.get<uint32_t>(), mat3 + 1,
+                         d_scan_tmp_.get(), &tb3, stream_);
+      launch_agg3_scatter(keys, kv, vals, vv, chunk, AGG3_NBUCK_LOG2,
+                          AGG3_GRID_LOG2, d_scanned_.get<uint32_t>(),
+                          d_partkv_.get<uint8_t>(), stream_);
+      launch_agg3_bucket(d_partkv_.get<uint8_t>(), d_counts_.get<uint32_t>(),
+                         d_scanned_.get<uint32_t>(), AGG3_GRID_LOG2,
+                         val_is_int_ ? 1 : 0, AGG3_NBUCK,
+                         d_staged_.get<StagedGroup>(),
+                         d_counters_.get<unsigned long long>(),
+                         (int64_t)AGG3_NBUCK * 4096,
+                         d_leftover_.get<PartRow>(),
+                         d_counters_.get<unsigned long long>() + 1,
+                         t_.error_flag, stream_);
+    } else {
+    int64_t mat = (int64_t)nbuck << agg2_grid_log2_;
     launch_agg2_hist(keys, kv, chunk, AGG2_NBUCK_LOG2, agg2_grid_log2_,
                      d_counts_.get<uint32_t>(),
                      (uint32_t*)(d_counters_.get<uint8_t>() + 16), agg2_block_,
@@ -1207,6 +1249,7 @@ class AggOp {
                            d_leftover_.get<PartRow>(),
                            d_counters_.get<unsigned long long>() + 1,
                            t_.error_flag, agg2_block_, stream_);
+    }
     unsigned long long* h_ctr =
         (unsigned long long*)(pinned_agg2_.get<uint8_t>() +
                               2 * (size_t)(nbuck + 1) * 4);
@@ -1725,11 +1768,12 @@ class AggOp {
   PinnedBuf pinned_meta_, pinned_emit_;
   // two-phase scratch (allocated on first large chunk)
   bool agg2_split_ = false;
+  bool agg2_v3_ = true;
   bool agg2_conf_read_ = false;
   int agg2_grid_log2_ = 9;
   int agg2_block_ = 1024;
   DevBuf d_partkv_, d_rowv_, d_leftover_, d_counts_, d_scanned_, d_scan_tmp_, d_offsets_,
-      d_staged_, d_counters_;
+      d_staged_, d_counters_, d_linesz_;
   PinnedBuf pinned_agg2_;
   std::vector<DevBatch> held_, skipped_;
 };
@@ -2562,7 +2606,35 @@ struct Runtime {
           const int w = (int)dtype_width(fcols[ci].dtype());
           c.own_values.alloc((size_t)rows * w);
           DevBuf d_valid, d_mask, d_positions, d_packed, d_indices;
-          if (has_nulls) {
+          DevBuf d_runs, d_runbytes;
+          if (cd.gpu_def) {
+            // def levels expand to the validity bitmap ON DEVICE (the host
+            // only walked the run headers — parquet.cpp rle_bp_runs)
+            d_runs.alloc(cd.def_runs.size() * sizeof(PqRun));
+            PinnedUploader::inst().copy(d_runs.get(), cd.def_runs.data(),
+                                        cd.def_runs.size() * sizeof(PqRun),
+                                        stream);
+            d_runbytes.alloc(cd.def_bytes.empty() ? 8 : cd.def_bytes.size());
+            if (!cd.def_bytes.empty())
+              PinnedUploader::inst().copy(d_runbytes.get(),
+                                          cd.def_bytes.data(),
+                                          cd.def_bytes.size(), stream);
+            c.own_validity.alloc((rows + 7) / 8);
+            launch_def_expand_validity(d_runs.get(), (int)cd.def_runs.size(),
+                                       d_runbytes.get<uint8_t>(), rows,
+                                       c.own_validity.get<uint8_t>(), stream);
+            c.validity = c.own_validity.get<uint8_t>();
+            d_mask.alloc(rows);
+            launch_bits_to_mask(c.validity, rows, d_mask.get<uint8_t>(),
+                                stream);
+            d_positions.alloc((rows + 1) * 4);
+            size_t tb = 0;
+            scan_mask_u8(d_mask.get<uint8_t>(), d_positions.get<uint32_t>(),
+                         rows, nullptr, &tb, stream);
+            if (tb > scan_tmp.size()) scan_tmp.alloc(tb);
+            scan_mask_u8(d_mask.get<uint8_t>(), d_positions.get<uint32_t>(),
+                         rows, scan_tmp.get(), &tb, stream);
+          } else if (has_nulls) {
             c.own_validity.alloc(cd.validity.size());
             AURON_HIP(hipMemcpyAsync(c.own_validity.get(), cd.validity.data(),
                                      cd.validity.size(), hipMemcpyHostToDevice,
@@ -2579,7 +2651,48 @@ struct Runtime {
             scan_mask_u8(d_mask.get<uint8_t>(), d_positions.get<uint32_t>(),
                          rows, scan_tmp.get(), &tb, stream);
           }
-          if (!cd.uses_dict) {
+          if (cd.gpu_dict) {
+            // dictionary indices: host walked run headers only; expand the
+            // dense non-null index array on device, then scatter/gather as
+            // in the host-materialized path
+            DevBuf d_iruns(cd.idx_runs.size() * sizeof(PqRun));
+            PinnedUploader::inst().copy(d_iruns.get(), cd.idx_runs.data(),
+                                        cd.idx_runs.size() * sizeof(PqRun),
+                                        stream);
+            DevBuf d_ibytes(cd.idx_bytes.empty() ? 8 : cd.idx_bytes.size());
+            if (!cd.idx_bytes.empty())
+              PinnedUploader::inst().copy(d_ibytes.get(), cd.idx_bytes.data(),
+                                          cd.idx_bytes.size(), stream);
+            DevBuf d_idx_packed((cd.nn_count ? cd.nn_count : 1) * 4);
+            launch_runs_expand_u32(d_iruns.get(), (int)cd.idx_runs.size(),
+                                   d_ibytes.get<uint8_t>(), cd.nn_count,
+                                   d_idx_packed.get<uint32_t>(), stream);
+            d_indices.alloc((size_t)rows * 4);
+            if (has_nulls) {
+              launch_scatter_packed(4, d_idx_packed.get<uint8_t>(),
+                                    d_positions.get<uint32_t>(),
+                                    d_mask.get<uint8_t>(), rows,
+                                    d_indices.get<uint8_t>(), stream);
+            } else {
+              AURON_HIP(hipMemcpyAsync(d_indices.get(), d_idx_packed.get(),
+                                       (size_t)rows * 4,
+                                       hipMemcpyDeviceToDevice, stream));
+            }
+            DevBuf d_dict(cd.dict_values.empty() ? 1 : cd.dict_values.size());
+            if (!cd.dict_values.empty())
+              AURON_HIP(hipMemcpyAsync(d_dict.get(), cd.dict_values.data(),
+                                       cd.dict_values.size(),
+                                       hipMemcpyHostToDevice, stream));
+            if (w == 8)
+              launch_gather_8(d_dict.get<uint8_t>(),
+                              d_indices.get<uint32_t>(), rows,
+                              c.own_values.get<uint8_t>(), stream);
+            else
+              launch_gather_4(d_dict.get<uint8_t>(),
+                              d_indices.get<uint32_t>(), rows,
+                              c.own_values.get<uint8_t>(), stream);
+            AURON_HIP(hipStreamSynchronize(stream));  // temps die below
+          } else if (!cd.uses_dict) {
             // PLAIN: non-null values packed densely
             if (!has_nulls) {
               PinnedUploader::inst().copy(c.own_values.get(), cd.plain.data(),
@@ -2985,6 +3098,7 @@ int32_t auron_debug_parquet_summary(const char* path, char* out, size_t cap) {
            std::to_string(pf.row_group_rows(rg));
       for (size_t c = 0; c < pf.columns().size(); c++) {
         PqColumnChunkData cd = pf.read_chunk(rg, (int)c);
+        pq_materialize_gpu_staging(&cd);  // host-expand staged GPU runs
         // value checksum: wrapping i64 sum of the raw fixed-width values
         // (sign-extended for narrow ints) over the dense non-null stream —
         // lets CPU tests verify DECODED VALUES against numpy ground truth,

@@ -371,6 +371,29 @@ void launch_dest_counts(const uint32_t* pids, const int32_t* offsets,
                         int64_t n, uint32_t world, unsigned long long* rows,
                         unsigned long long* bytes, hipStream_t s);
 
+// parquet RLE/bit-packed run expansion (runs = PqRun[] from parquet.h)
+void launch_runs_expand_u32(const void* runs, int nruns, const uint8_t* bytes,
+                            int64_t n, uint32_t* out, hipStream_t s);
+void launch_def_expand_validity(const void* runs, int nruns,
+                                const uint8_t* bytes, int64_t n,
+                                uint8_t* bitmap, hipStream_t s);
+
+// v3 two-phase partition pipeline (kernels_agg3.hip): LDS-staged packet
+// scatter into 64B-aligned per-(block,bucket) ranges + 4096-slot bucket agg
+void launch_agg3_line_sizes(const uint32_t* counts, int64_t n,
+                            uint32_t* sizes, hipStream_t s);
+void launch_agg3_scatter(const int64_t* keys, const uint8_t* key_valid,
+                         const double* vals, const uint8_t* val_valid,
+                         int64_t n, int nbuck_log2, int grid_log2,
+                         const uint32_t* line_scan, uint8_t* out,
+                         hipStream_t s);
+void launch_agg3_bucket(const uint8_t* part, const uint32_t* counts,
+                        const uint32_t* line_scan, int grid_log2, int is_int,
+                        int nbuckets, StagedGroup* staged,
+                        unsigned long long* staged_n, int64_t staged_cap,
+                        PartRow* leftover, unsigned long long* lo_n,
+                        uint32_t* error_flag, hipStream_t s);
+
 // sort groups by first_row: rocprim radix sort pairs wrapper
 void sort_pairs_u64_u32(const unsigned long long* keys_in, const uint32_t* vals_in,
                         unsigned long long* keys_out, uint32_t* vals_out,

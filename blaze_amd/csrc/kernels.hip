@@ -1608,4 +1608,84 @@ void launch_dest_counts(const uint32_t* pids, const int32_t* offsets,
   check_launch("k_dest_counts");
 }
 
+// ---- parquet RLE/bit-packed run expansion ----------------------------------
+// The host walks run HEADERS only (parquet.cpp rle_bp_runs); these kernels do
+// the wide work: each output element binary-searches the run table by output
+// position and extracts its value — coalesced writes, ~log2(#runs) cheap
+// reads. Run layout mirrors struct PqRun (parquet.h, 16 B).
+struct RunView {
+  uint32_t out_pos, count, src_off;
+  uint8_t kind, bw;
+  uint16_t _pad;
+};
+
+__device__ __forceinline__ uint32_t run_value(const RunView& r,
+                                              const uint8_t* __restrict__ bytes,
+                                              uint32_t j) {
+  if (r.kind == 0) return r.src_off;  // RLE: the value itself
+  // bit-packed, LSB-first: extract bw bits at bit position j*bw
+  uint64_t bitpos = (uint64_t)j * r.bw;
+  uint64_t word;
+  memcpy(&word, bytes + r.src_off + (bitpos >> 3), 8);  // 8B tail pad (host)
+  uint32_t mask = r.bw >= 32 ? 0xFFFFFFFFu : ((1u << r.bw) - 1u);
+  return (uint32_t)(word >> (bitpos & 7)) & mask;
+}
+
+__device__ __forceinline__ int run_find(const RunView* __restrict__ runs,
+                                        int nruns, uint32_t i) {
+  int lo = 0, hi = nruns - 1;
+  while (lo < hi) {
+    int mid = (lo + hi + 1) >> 1;
+    if (runs[mid].out_pos <= i) lo = mid; else hi = mid - 1;
+  }
+  return lo;
+}
+
+__global__ void k_runs_expand_u32(const RunView* __restrict__ runs, int nruns,
+                                  const uint8_t* __restrict__ bytes, int64_t n,
+                                  uint32_t* __restrict__ out) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const RunView& r = runs[run_find(runs, nruns, (uint32_t)i)];
+    out[i] = run_value(r, bytes, (uint32_t)i - r.out_pos);
+  }
+}
+
+void launch_runs_expand_u32(const void* runs, int nruns, const uint8_t* bytes,
+                            int64_t n, uint32_t* out, hipStream_t s) {
+  hipLaunchKernelGGL(k_runs_expand_u32, dim3(grid_for(n)), dim3(BLOCK), 0, s,
+                     (const RunView*)runs, nruns, bytes, n, out);
+  check_launch("k_runs_expand_u32");
+}
+
+// def levels (bw=1) -> LSB validity bitmap; one thread per output BYTE
+__global__ void k_def_expand_validity(const RunView* __restrict__ runs,
+                                      int nruns,
+                                      const uint8_t* __restrict__ bytes,
+                                      int64_t n, uint8_t* __restrict__ bitmap) {
+  int64_t nbytes = (n + 7) / 8;
+  for (int64_t byte = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       byte < nbytes; byte += (int64_t)gridDim.x * blockDim.x) {
+    uint8_t out = 0;
+    int64_t base = byte * 8;
+    int ri = run_find(runs, nruns, (uint32_t)base);
+    for (int b = 0; b < 8 && base + b < n; b++) {
+      uint32_t i = (uint32_t)(base + b);
+      while (ri + 1 < nruns && runs[ri + 1].out_pos <= i) ri++;
+      const RunView& r = runs[ri];
+      if (run_value(r, bytes, i - r.out_pos)) out |= (uint8_t)(1u << b);
+    }
+    bitmap[byte] = out;
+  }
+}
+
+void launch_def_expand_validity(const void* runs, int nruns,
+                                const uint8_t* bytes, int64_t n,
+                                uint8_t* bitmap, hipStream_t s) {
+  hipLaunchKernelGGL(k_def_expand_validity, dim3(grid_for((n + 7) / 8)),
+                     dim3(BLOCK), 0, s, (const RunView*)runs, nruns, bytes, n,
+                     bitmap);
+  check_launch("k_def_expand_validity");
+}
+
 }  // namespace auron

@@ -1,0 +1,299 @@
+// kernels_agg3.hip — v3 partition scatter for the two-phase aggregation:
+// LDS-staged packet flushes into 64B-aligned per-(block,bucket) ranges.
+//
+// Round-1/2 PMC evidence (profiles/r01_final_pmc.md, r2chain stats): the
+// per-record scatter wrote partial cache lines across ~512K open regions and
+// paid ~2x write-back amplification — 30 ms of the 41 ms step at 1.35 TB/s
+// effective. Here every (block,bucket) range starts 64B-aligned (the hist
+// scan pads sizes to whole lines) and records are staged in LDS until a full
+// 8-record packet (8 x 24 B = 192 B = 3 exact lines) can be written, so the
+// L2 never holds a partially written line between flushes. Layout per block:
+//   staging[512 buckets][12 records][24 B]  = 147 KB LDS
+//   cnt/fl counters                          = 4 KB
+// (one 1024-thread workgroup per CU; the scatter is bandwidth-bound, not
+// latency-bound, so 16 waves suffice — measured round 2.)
+//
+// The bucket aggregation kernel widens its LDS hash window to 4096 slots
+// (512 buckets x ~2000 distinct keys/bucket at the 1M-group north star needs
+// >2048) at 24 B/slot SoA = 96 KB -> one 1024-thread workgroup per CU.
+#include <hip/hip_runtime.h>
+
+#include <stdexcept>
+#include <string>
+
+#include "kernels.h"
+
+namespace auron {
+
+namespace {
+inline void check_launch3(const char* name) {
+  hipError_t e = hipGetLastError();
+  if (e != hipSuccess)
+    throw std::runtime_error(std::string("kernel launch failed: ") + name +
+                             ": " + hipGetErrorString(e));
+}
+constexpr int64_t KEY_EMPTY3 = INT64_MIN;
+
+__device__ __forceinline__ uint64_t mix64_3(uint64_t x) {
+  x += 0x9E3779B97F4A7C15ull;
+  x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+  x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+  return x ^ (x >> 31);
+}
+
+__device__ __forceinline__ bool bit_get3(const uint8_t* bm, int64_t i) {
+  return (bm[i >> 3] >> (i & 7)) & 1;
+}
+
+__device__ __forceinline__ void sum_accum3(double* acc, double v,
+                                           bool is_int) {
+  if (is_int) {
+    uint64_t b;
+    memcpy(&b, &v, 8);
+    atomicAdd(reinterpret_cast<unsigned long long*>(acc),
+              (unsigned long long)b);
+  } else {
+    unsafeAtomicAdd(acc, v);
+  }
+}
+}  // namespace
+
+// ---- byte-line offsets: pad each (block,bucket) range to whole 64B lines --
+// line_sizes[i] = ceil(counts[i] * 24 / 64): the exclusive scan of THIS
+// matrix (in 64-byte line units, so 6 GB chunks still fit u32) gives every
+// range a line-aligned start.
+__global__ void k_agg3_line_sizes(const uint32_t* __restrict__ counts,
+                                  int64_t n, uint32_t* __restrict__ sizes) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    sizes[i] = (counts[i] * 24u + 63u) >> 6;
+}
+
+void launch_agg3_line_sizes(const uint32_t* counts, int64_t n,
+                            uint32_t* sizes, hipStream_t s) {
+  hipLaunchKernelGGL(k_agg3_line_sizes, dim3(256), dim3(256), 0, s, counts, n,
+                     sizes);
+  check_launch3("k_agg3_line_sizes");
+}
+
+// ---- v3 scatter ------------------------------------------------------------
+static constexpr int A3_CAP = 12;      // staged records per bucket
+static constexpr int A3_QUANT = 8;     // records per flushed packet (192 B)
+
+__global__ void __launch_bounds__(1024) k_agg3_scatter(
+    const int64_t* __restrict__ keys, const uint8_t* __restrict__ key_valid,
+    const double* __restrict__ vals, const uint8_t* __restrict__ val_valid,
+    int64_t n, int nbuck_log2, int grid_log2,
+    const uint32_t* __restrict__ line_scan, uint8_t* __restrict__ out) {
+  const uint32_t nbuck = 1u << nbuck_log2;
+  // staging[b][slot]: 24B records, SoA-of-packets layout kept simple as AoS
+  extern __shared__ uint8_t lds[];
+  // layout: records [nbuck][A3_CAP][24] | cnt[nbuck] u32 | fl[nbuck] u32 |
+  //         base_line[nbuck] u32
+  uint8_t* stage = lds;
+  uint32_t* cnt = (uint32_t*)(lds + (size_t)nbuck * A3_CAP * 24);
+  uint32_t* fl = cnt + nbuck;
+  uint32_t* base_line = fl + nbuck;
+
+  for (uint32_t b = threadIdx.x; b < nbuck; b += blockDim.x) {
+    cnt[b] = 0;
+    fl[b] = 0;
+    base_line[b] = line_scan[((size_t)b << grid_log2) | blockIdx.x];
+  }
+  __syncthreads();
+
+  int64_t my = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const int nwave = (int)(blockDim.x >> 6);
+  const int wave = (int)(threadIdx.x >> 6);
+  const int lane = (int)(threadIdx.x & 63);
+  const uint32_t bpw = (nbuck + nwave - 1) / nwave;  // buckets per wave
+
+  for (;;) {
+    // tile: each thread tries to stage its current row
+    bool pending = my < n;
+    if (pending) {
+      bool knull = key_valid && !bit_get3(key_valid, my);
+      int64_t k = keys[my];
+      if (knull || k == KEY_EMPTY3) {
+        my += stride;  // specials handled by k_agg2_specials
+        pending = my < n;
+      }
+    }
+    if (pending) {
+      int64_t k = keys[my];
+      uint32_t b = (uint32_t)(mix64_3((uint64_t)k) >> (64 - nbuck_log2));
+      uint32_t pos = atomicAdd(&cnt[b], 1u);
+      if (pos < A3_CAP) {
+        bool vvalid = !val_valid || bit_get3(val_valid, my);
+        uint8_t* r = stage + ((size_t)b * A3_CAP + pos) * 24;
+        *(int64_t*)r = k;
+        *(double*)(r + 8) = vals[my];
+        *(uint32_t*)(r + 16) = (uint32_t)my | (vvalid ? 0x80000000u : 0u);
+        *(uint32_t*)(r + 20) = 0;
+        my += stride;
+      } else {
+        atomicSub(&cnt[b], 1u);  // full: retry this row next tile
+      }
+    }
+    // block-wide termination check doubles as the pre-flush barrier
+    int live = __syncthreads_count((my < n) ? 1 : 0);
+    // cooperative flush: wave w drains buckets [w*bpw, (w+1)*bpw)
+    for (uint32_t b = wave * bpw; b < (wave + 1u) * bpw && b < nbuck; b++) {
+      uint32_t c = cnt[b];
+      uint32_t nfl = (live == 0) ? c : (c & ~(uint32_t)(A3_QUANT - 1));
+      if (nfl) {
+        // packet write: nfl*24 bytes, dword-per-lane from the 64-lane wave
+        uint8_t* dst = out + ((size_t)base_line[b] << 6) +
+                       (size_t)fl[b] * 24;
+        const uint8_t* src = stage + (size_t)b * A3_CAP * 24;
+        uint32_t ndw = nfl * 6;
+        for (uint32_t d = lane; d < ndw; d += 64)
+          ((uint32_t*)dst)[d] = ((const uint32_t*)src)[d];
+        // move the remainder to the staging front (lane-parallel)
+        uint32_t rem = c - nfl;
+        for (uint32_t d = lane; d < rem * 6; d += 64)
+          ((uint32_t*)src)[d] = ((const uint32_t*)(src + (size_t)nfl * 24))[d];
+        if (lane == 0) {
+          fl[b] += nfl;
+          cnt[b] = rem;
+        }
+      }
+    }
+    __syncthreads();
+    if (live == 0) break;
+  }
+}
+
+void launch_agg3_scatter(const int64_t* keys, const uint8_t* key_valid,
+                         const double* vals, const uint8_t* val_valid,
+                         int64_t n, int nbuck_log2, int grid_log2,
+                         const uint32_t* line_scan, uint8_t* out,
+                         hipStream_t s) {
+  const uint32_t nbuck = 1u << nbuck_log2;
+  size_t lds = (size_t)nbuck * A3_CAP * 24 + (size_t)nbuck * 12;
+  if (lds > 160 * 1024)
+    throw std::runtime_error("agg3 scatter LDS over 160KB");
+  hipError_t e = hipFuncSetAttribute(
+      (const void*)k_agg3_scatter,
+      hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
+  if (e != hipSuccess)
+    throw std::runtime_error("agg3 scatter LDS attribute failed");
+  hipLaunchKernelGGL(k_agg3_scatter, dim3(1 << grid_log2), dim3(1024), lds, s,
+                     keys, key_valid, vals, val_valid, n, nbuck_log2,
+                     grid_log2, line_scan, out);
+  check_launch3("k_agg3_scatter");
+}
+
+// ---- v3 bucket aggregation (4096-slot LDS window, 1024 threads) ------------
+static constexpr int L3SLOTS = 4096;
+static constexpr int L3PROBE = 128;
+
+__global__ void __launch_bounds__(1024) k_agg3_bucket(
+    const uint8_t* __restrict__ part, const uint32_t* __restrict__ counts,
+    const uint32_t* __restrict__ line_scan, int grid_log2, int is_int,
+    int nbuckets, StagedGroup* __restrict__ staged,
+    unsigned long long* __restrict__ staged_n, int64_t staged_cap,
+    PartRow* __restrict__ leftover, unsigned long long* __restrict__ lo_n,
+    uint32_t* __restrict__ error_flag) {
+  __shared__ int64_t ls_key[L3SLOTS];
+  __shared__ double ls_sum[L3SLOTS];
+  __shared__ uint32_t ls_cnt[L3SLOTS];
+  __shared__ uint32_t ls_first[L3SLOTS];
+  // per-range cumulative record counts for this bucket (grid of ranges)
+  __shared__ uint32_t cum[1 << AGG2_GRID_LOG2_MAX];
+  const int nrange = 1 << grid_log2;
+
+  for (int b = blockIdx.x; b < nbuckets; b += gridDim.x) {
+    for (int s = threadIdx.x; s < L3SLOTS; s += blockDim.x) {
+      ls_key[s] = KEY_EMPTY3;
+      ls_sum[s] = 0.0;
+      ls_cnt[s] = 0;
+      ls_first[s] = 0xFFFFFFFFu;
+    }
+    // serial scan of <=1024 range counts (one thread; ~us)
+    if (threadIdx.x == 0) {
+      uint32_t acc = 0;
+      for (int k = 0; k < nrange; k++) {
+        acc += counts[((size_t)b << grid_log2) | k];
+        cum[k] = acc;
+      }
+    }
+    __syncthreads();
+    uint32_t total = cum[nrange - 1];
+    for (uint32_t j = threadIdx.x; j < total; j += blockDim.x) {
+      // binary search the range holding record j
+      int lo = 0, hi = nrange - 1;
+      while (lo < hi) {
+        int mid = (lo + hi) >> 1;
+        if (j < cum[mid]) hi = mid; else lo = mid + 1;
+      }
+      uint32_t before = lo ? cum[lo - 1] : 0;
+      const uint8_t* r =
+          part + ((size_t)line_scan[((size_t)b << grid_log2) | lo] << 6) +
+          (size_t)(j - before) * 24;
+      int64_t k = *(const int64_t*)r;
+      double v = *(const double*)(r + 8);
+      uint32_t rowv = *(const uint32_t*)(r + 16);
+      uint32_t row = rowv & 0x7FFFFFFFu;
+      bool vvalid = (rowv & 0x80000000u) != 0;
+      uint32_t h = (uint32_t)mix64_3((uint64_t)k) & (L3SLOTS - 1);
+      int found = -1;
+      for (int p = 0; p < L3PROBE; p++) {
+        int64_t cur = ls_key[h];
+        if (cur == k) {
+          found = (int)h;
+          break;
+        }
+        if (cur == KEY_EMPTY3) {
+          long long prev = atomicCAS((unsigned long long*)&ls_key[h],
+                                     (unsigned long long)KEY_EMPTY3,
+                                     (unsigned long long)k);
+          if (prev == (long long)KEY_EMPTY3 || prev == (long long)k) {
+            found = (int)h;
+            break;
+          }
+        }
+        h = (h + 1) & (L3SLOTS - 1);
+      }
+      if (found >= 0) {
+        atomicMin(&ls_first[found], row);
+        if (vvalid) {
+          sum_accum3(&ls_sum[found], v, is_int);
+          atomicAdd(&ls_cnt[found], 1u);
+        }
+      } else {
+        unsigned long long p = atomicAdd(lo_n, 1ull);
+        leftover[p] = PartRow{k, v, rowv, 0};
+      }
+    }
+    __syncthreads();
+    for (int s = threadIdx.x; s < L3SLOTS; s += blockDim.x) {
+      if (ls_key[s] == KEY_EMPTY3) continue;
+      unsigned long long p = atomicAdd(staged_n, 1ull);
+      if ((int64_t)p >= staged_cap) {
+        atomicOr(error_flag, 2u);
+        continue;
+      }
+      staged[p].key = ls_key[s];
+      staged[p].sum = ls_sum[s];
+      staged[p].cnt_first =
+          ((unsigned long long)ls_cnt[s] << 32) | ls_first[s];
+    }
+    __syncthreads();
+  }
+}
+
+void launch_agg3_bucket(const uint8_t* part, const uint32_t* counts,
+                        const uint32_t* line_scan, int grid_log2, int is_int,
+                        int nbuckets, StagedGroup* staged,
+                        unsigned long long* staged_n, int64_t staged_cap,
+                        PartRow* leftover, unsigned long long* lo_n,
+                        uint32_t* error_flag, hipStream_t s) {
+  hipLaunchKernelGGL(k_agg3_bucket, dim3(nbuckets), dim3(1024), 0, s, part,
+                     counts, line_scan, grid_log2, is_int, nbuckets, staged,
+                     staged_n, staged_cap, leftover, lo_n, error_flag);
+  check_launch3("k_agg3_bucket");
+}
+
+}  // namespace auron

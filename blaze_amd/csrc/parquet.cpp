@@ -313,6 +313,99 @@ void rle_bp_decode(const uint8_t* p, size_t len, int bit_width, int64_t count,
   }
 }
 
+// walk an RLE/bit-packed hybrid stream's run HEADERS only (sequential
+// varints — ~#runs of host work instead of ~#values): append PqRun records
+// plus the raw bit-packed span bytes; the GPU expands values
+// (binary-search-over-output-position kernels in kernels.hip). When `ones`
+// is non-null (def levels, bit_width 1) it accumulates the 1-count via RLE
+// headers + span popcounts. Returns the number of values consumed.
+int64_t rle_bp_runs(const uint8_t* p, size_t len, int bit_width,
+                    int64_t count, uint32_t out_base, std::vector<PqRun>* runs,
+                    std::vector<uint8_t>* span_bytes, int64_t* ones) {
+  int64_t filled = 0;
+  size_t pos = 0;
+  const int byte_w = (bit_width + 7) / 8;
+  while (filled < count) {
+    if (pos >= len) fail("rle: truncated stream");
+    uint64_t header = 0;
+    int shift = 0;
+    while (true) {
+      if (pos >= len) fail("rle: truncated varint");
+      uint8_t b = p[pos++];
+      header |= (uint64_t)(b & 0x7f) << shift;
+      if (!(b & 0x80)) break;
+      shift += 7;
+    }
+    PqRun r;
+    r.out_pos = out_base + (uint32_t)filled;
+    r.bw = (uint8_t)bit_width;
+    if (header & 1) {
+      int64_t n = (int64_t)(header >> 1) * 8;
+      size_t bytes = (size_t)n * bit_width / 8;
+      if (pos + bytes > len) fail("rle: truncated bit-packed run");
+      int64_t take = n < count - filled ? n : count - filled;
+      r.kind = 1;
+      r.count = (uint32_t)take;
+      r.src_off = (uint32_t)span_bytes->size();
+      span_bytes->insert(span_bytes->end(), p + pos, p + pos + bytes);
+      if (ones) {  // def levels: popcount the span, bounded by `take` bits
+        int64_t full = take >> 3, rem = take & 7;
+        for (int64_t i = 0; i < full; i++)
+          *ones += __builtin_popcount(p[pos + (size_t)i]);
+        if (rem)
+          *ones += __builtin_popcount(p[pos + (size_t)full] &
+                                      ((1u << rem) - 1));
+      }
+      filled += take;
+      pos += bytes;
+    } else {
+      int64_t n = (int64_t)(header >> 1);
+      uint32_t v = 0;
+      if (pos + byte_w > len) fail("rle: truncated repeated run");
+      for (int b = 0; b < byte_w; b++) v |= (uint32_t)p[pos + b] << (8 * b);
+      pos += byte_w;
+      int64_t take = n < count - filled ? n : count - filled;
+      r.kind = 0;
+      r.count = (uint32_t)take;
+      r.src_off = v;
+      if (ones && v) *ones += take;
+      filled += take;
+    }
+    if (r.count) runs->push_back(r);
+  }
+  return filled;
+}
+
+// host expansion of staged runs (the fallback when a chunk turns out to mix
+// dict and PLAIN pages and must be materialized after all)
+void expand_runs_host(const std::vector<PqRun>& runs,
+                      const std::vector<uint8_t>& span_bytes,
+                      std::vector<uint32_t>* out) {
+  for (const PqRun& r : runs) {
+    size_t base = out->size();
+    out->resize(base + r.count);
+    uint32_t* dst = out->data() + base;
+    if (r.kind == 0) {
+      for (uint32_t i = 0; i < r.count; i++) dst[i] = r.src_off;
+    } else {
+      uint64_t acc = 0;
+      int bits = 0;
+      size_t bp = r.src_off;
+      const uint32_t mask =
+          r.bw >= 32 ? 0xFFFFFFFFu : (uint32_t)((1ull << r.bw) - 1);
+      for (uint32_t i = 0; i < r.count; i++) {
+        while (bits < r.bw) {
+          acc |= (uint64_t)span_bytes[bp++] << bits;
+          bits += 8;
+        }
+        dst[i] = (uint32_t)acc & mask;
+        acc >>= r.bw;
+        bits -= r.bw;
+      }
+    }
+  }
+}
+
 }  // namespace
 
 // ---- DELTA_BINARY_PACKED (encoding 5; parquet spec delta encoding) ---------
@@ -649,6 +742,26 @@ ParquetFile::ParquetFile(const std::string& path) : path_(path) {
       fail("parquet: row group column count mismatch");
 }
 
+void pq_materialize_gpu_staging(PqColumnChunkData* cd) {
+  if (cd->gpu_dict) {
+    expand_runs_host(cd->idx_runs, cd->idx_bytes, &cd->dict_indices);
+    cd->gpu_dict = false;
+    cd->uses_dict = true;
+    cd->idx_runs.clear();
+    cd->idx_bytes.clear();
+  }
+  if (cd->gpu_def) {
+    std::vector<uint32_t> bits;
+    expand_runs_host(cd->def_runs, cd->def_bytes, &bits);
+    cd->validity.assign((bits.size() + 7) / 8, 0);
+    for (size_t i = 0; i < bits.size(); i++)
+      if (bits[i]) cd->validity[i >> 3] |= (uint8_t)(1u << (i & 7));
+    cd->gpu_def = false;
+    cd->def_runs.clear();
+    cd->def_bytes.clear();
+  }
+}
+
 PqColStats ParquetFile::column_stats(int rg, int col) const {
   const ChunkMeta& cm = row_groups_.at(rg).chunks.at(col);
   const PqColumnInfo& ci = columns_.at(col);
@@ -676,6 +789,13 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
 
   PqColumnChunkData out;
   out.num_values = 0;
+  // GPU run-expansion mode (fixed-width columns): the host walks RLE/
+  // bit-packed run headers only and the GPU expands def levels and dict
+  // indices (kernels.hip k_runs_expand_u32 / k_def_expand_validity).
+  // AURON_PARQUET_GPU=0 forces the all-host decode for A/B comparison.
+  const char* pg = getenv("AURON_PARQUET_GPU");
+  const bool gpu_ok = ci.physical_type != 6 && !(pg && pg[0] == '0');
+  const bool gpu_def_mode = gpu_ok && ci.nullable;
   if (ci.physical_type != 6)
     out.plain.reserve((size_t)cm.num_values * vw);
   std::vector<uint8_t> valid_bits;  // byte per value (bit-packed at the end)
@@ -764,6 +884,8 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
     const uint8_t* data;
     int64_t dlen;
     std::vector<uint32_t> def_levels;
+    int64_t page_ones = 0;
+    bool page_def_staged = false;
     if (ph.type == 0) {  // V1: whole page compressed, def levels inside
       data = decompress(page, ph.compressed_size, ph.uncompressed_size, &buf);
       dlen = ph.uncompressed_size;
@@ -773,7 +895,15 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
         if (dlen < 4) fail("parquet: truncated def-level length");
         memcpy(&ll, data, 4);
         if ((int64_t)ll > dlen - 4) fail("parquet: def levels overrun page");
-        rle_bp_decode(data + 4, ll, 1, ph.num_values, &def_levels);
+        if (gpu_def_mode) {
+          page_ones = 0;
+          rle_bp_runs(data + 4, ll, 1, ph.num_values,
+                      (uint32_t)out.num_values, &out.def_runs, &out.def_bytes,
+                      &page_ones);
+          page_def_staged = true;
+        } else {
+          rle_bp_decode(data + 4, ll, 1, ph.num_values, &def_levels);
+        }
         data += 4 + ll;
         dlen -= 4 + ll;
       }
@@ -783,8 +913,22 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
           ph.def_len + ph.rep_len > ph.compressed_size)
         fail("parquet: V2 level lengths overrun page");
       if (ci.nullable) {
-        if (ph.def_len > 0)
+        if (gpu_def_mode) {
+          page_ones = 0;
+          if (ph.def_len > 0) {
+            rle_bp_runs(dp, ph.def_len, 1, ph.num_values,
+                        (uint32_t)out.num_values, &out.def_runs,
+                        &out.def_bytes, &page_ones);
+          } else {  // whole page valid: synthetic all-ones run
+            out.def_runs.push_back(PqRun{(uint32_t)out.num_values,
+                                         (uint32_t)ph.num_values, 1u, 0, 1,
+                                         0});
+            page_ones = ph.num_values;
+          }
+          page_def_staged = true;
+        } else if (ph.def_len > 0) {
           rle_bp_decode(dp, ph.def_len, 1, ph.num_values, &def_levels);
+        }
       }
       dp += ph.def_len + ph.rep_len;
       int64_t comp = ph.compressed_size - ph.def_len - ph.rep_len;
@@ -796,7 +940,10 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
 
     int64_t nvals = ph.num_values;
     int64_t non_null = nvals;
-    if (ci.nullable && !def_levels.empty()) {
+    if (page_def_staged) {
+      non_null = page_ones;
+      out.null_count += nvals - page_ones;
+    } else if (ci.nullable && !def_levels.empty()) {
       non_null = 0;
       size_t base = valid_bits.size();
       valid_bits.resize(base + (size_t)nvals);
@@ -806,7 +953,7 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
         non_null += v;
       }
       out.null_count += nvals - non_null;
-    } else {
+    } else if (!gpu_def_mode) {
       valid_bits.resize(valid_bits.size() + (size_t)nvals, 1);
     }
 
@@ -848,6 +995,16 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
       }
     };
     auto flatten_dict = [&]() {
+      if (out.gpu_dict) {
+        // staged GPU runs must materialize after all (dict/PLAIN mix)
+        std::vector<uint32_t> idx;
+        expand_runs_host(out.idx_runs, out.idx_bytes, &idx);
+        out.gpu_dict = false;
+        out.idx_runs.clear();
+        out.idx_bytes.clear();
+        out.nn_count = 0;
+        expand_dict(idx.data(), idx.size());
+      }
       if (!out.uses_dict) return;
       std::vector<uint32_t> idx = std::move(out.dict_indices);
       out.dict_indices.clear();
@@ -937,6 +1094,15 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
         if (dlen < 1) fail("parquet: empty dict-index page");
         int bw = data[0];
         if (bw > 32) fail("parquet: dict index bit width > 32");
+        if (gpu_ok && out.plain.empty() && !out.uses_dict) {
+          // stage run headers for GPU expansion (all pages so far dict)
+          out.gpu_dict = true;
+          rle_bp_runs(data + 1, dlen - 1, bw, non_null,
+                      (uint32_t)out.nn_count, &out.idx_runs, &out.idx_bytes,
+                      nullptr);
+          out.nn_count += non_null;
+          break;
+        }
         std::vector<uint32_t> idx;
         rle_bp_decode(data + 1, dlen - 1, bw, non_null, &idx);
         if (is_bytes) {
@@ -981,11 +1147,20 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
     }
     if (nn != nn_lens.size()) fail("parquet: byte-array count mismatch");
   }
-  if (out.null_count > 0) {
+  if (!out.def_runs.empty()) {
+    if (out.null_count > 0) {
+      out.gpu_def = true;
+      out.def_bytes.resize(out.def_bytes.size() + 8, 0);  // u64 tail pad
+    } else {  // no nulls after all: no validity needed
+      out.def_runs.clear();
+      out.def_bytes.clear();
+    }
+  } else if (out.null_count > 0) {
     out.validity.assign((valid_bits.size() + 7) / 8, 0);
     for (size_t i = 0; i < valid_bits.size(); i++)
       if (valid_bits[i]) out.validity[i >> 3] |= (uint8_t)(1u << (i & 7));
   }
+  if (out.gpu_dict) out.idx_bytes.resize(out.idx_bytes.size() + 8, 0);
   return out;
 }
 

@@ -28,6 +28,20 @@ struct RunList {
   std::vector<uint32_t> values;
 };
 
+// one decoded RLE/bit-packed hybrid run HEADER: the host walks headers only
+// (sequential varints, ~#runs work) and the GPU expands values — the wide
+// work of the decode (binary-search-over-output-position kernels)
+struct PqRun {
+  uint32_t out_pos;  // first output slot this run fills
+  uint32_t count;    // values in this run (clipped to the page's remainder)
+  uint32_t src_off;  // bit-packed: byte offset into the span bytes buffer;
+                     // RLE: the repeated VALUE itself (<= 32 bits)
+  uint8_t kind;      // 0 = RLE, 1 = bit-packed
+  uint8_t bw;        // bit width
+  uint16_t _pad = 0;
+};
+static_assert(sizeof(PqRun) == 16, "PqRun must be 16 bytes");
+
 struct PqColumnChunkData {
   // per row-group column results, host-resident, ready for GPU upload
   int64_t num_values = 0;              // value slots incl. nulls
@@ -41,7 +55,23 @@ struct PqColumnChunkData {
   // BYTE_ARRAY columns: row-aligned (null rows zero-length), host-assembled
   std::vector<int32_t> bin_offsets;    // num_values + 1
   std::vector<uint8_t> bin_data;
+  // GPU run-expansion staging (fixed-width dict chunks; parquet.cpp sets
+  // gpu_dict/gpu_def when every page qualified). Span byte buffers carry 8
+  // bytes of tail padding for unaligned u64 bit extraction on device.
+  bool gpu_dict = false;   // dict_indices replaced by idx_runs/idx_bytes
+  std::vector<PqRun> idx_runs;
+  std::vector<uint8_t> idx_bytes;
+  int64_t nn_count = 0;    // non-null value count (== dense index count)
+  bool gpu_def = false;    // validity replaced by def_runs/def_bytes
+  std::vector<PqRun> def_runs;
+  std::vector<uint8_t> def_bytes;
 };
+
+// host materialization of GPU-staged runs (debug / CPU-test path): expands
+// idx_runs -> dict_indices and def_runs -> validity bitmap exactly as the
+// device kernels (k_runs_expand_u32 / k_def_expand_validity) would, so CPU
+// tests pin the header walk without a GPU.
+void pq_materialize_gpu_staging(PqColumnChunkData* cd);
 
 struct PqColumnInfo {
   std::string name;
