@@ -80,88 +80,116 @@ void launch_agg3_line_sizes(const uint32_t* counts, int64_t n,
 static constexpr int A3_CAP = 12;      // staged records per bucket
 static constexpr int A3_QUANT = 8;     // records per flushed packet (192 B)
 
+// A3_RPT rows per thread per tile; a dirty QUEUE records buckets that
+// crossed a packet boundary so the flush phase touches only those (the v3.0
+// per-tile scan of every bucket made the kernel instruction-bound: PMC
+// showed 20x the dynamic instructions of v2 and 77% parked waves while HBM
+// writes ran at 1.0x algorithmic — gpurun_out/r2_pmc_scatter.json).
+static constexpr int A3_RPT = 2;
+
 __global__ void __launch_bounds__(1024) k_agg3_scatter(
     const int64_t* __restrict__ keys, const uint8_t* __restrict__ key_valid,
     const double* __restrict__ vals, const uint8_t* __restrict__ val_valid,
     int64_t n, int nbuck_log2, int grid_log2,
     const uint32_t* __restrict__ line_scan, uint8_t* __restrict__ out) {
   const uint32_t nbuck = 1u << nbuck_log2;
-  // staging[b][slot]: 24B records, SoA-of-packets layout kept simple as AoS
   extern __shared__ uint8_t lds[];
-  // layout: records [nbuck][A3_CAP][24] | cnt[nbuck] u32 | fl[nbuck] u32 |
-  //         base_line[nbuck] u32
+  // layout: records [nbuck][A3_CAP][24] | cnt[nbuck] | fl[nbuck] |
+  //         base_line[nbuck] | dirty queue [nbuck] u16-as-u32 | qn
   uint8_t* stage = lds;
   uint32_t* cnt = (uint32_t*)(lds + (size_t)nbuck * A3_CAP * 24);
   uint32_t* fl = cnt + nbuck;
   uint32_t* base_line = fl + nbuck;
+  uint16_t* queue = (uint16_t*)(base_line + nbuck);
+  uint32_t* qn = (uint32_t*)(queue + nbuck + 64);
 
   for (uint32_t b = threadIdx.x; b < nbuck; b += blockDim.x) {
     cnt[b] = 0;
     fl[b] = 0;
     base_line[b] = line_scan[((size_t)b << grid_log2) | blockIdx.x];
   }
+  if (threadIdx.x == 0) *qn = 0;
   __syncthreads();
 
-  int64_t my = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   const int nwave = (int)(blockDim.x >> 6);
   const int wave = (int)(threadIdx.x >> 6);
   const int lane = (int)(threadIdx.x & 63);
-  const uint32_t bpw = (nbuck + nwave - 1) / nwave;  // buckets per wave
+  int64_t my[A3_RPT];
+#pragma unroll
+  for (int r = 0; r < A3_RPT; r++)
+    my[r] = (int64_t)blockIdx.x * blockDim.x + threadIdx.x +
+            (int64_t)r * stride;
+  const int64_t step = (int64_t)A3_RPT * stride;
 
   for (;;) {
-    // tile: each thread tries to stage its current row
-    bool pending = my < n;
-    if (pending) {
-      bool knull = key_valid && !bit_get3(key_valid, my);
-      int64_t k = keys[my];
+    int mine = 0;
+#pragma unroll
+    for (int r = 0; r < A3_RPT; r++) {
+      if (my[r] >= n) continue;
+      mine = 1;
+      int64_t k = keys[my[r]];
+      bool knull = key_valid && !bit_get3(key_valid, my[r]);
       if (knull || k == KEY_EMPTY3) {
-        my += stride;  // specials handled by k_agg2_specials
-        pending = my < n;
+        my[r] += step;  // specials handled by k_agg2_specials
+        mine |= (my[r] < n);
+        continue;
       }
-    }
-    if (pending) {
-      int64_t k = keys[my];
       uint32_t b = (uint32_t)(mix64_3((uint64_t)k) >> (64 - nbuck_log2));
       uint32_t pos = atomicAdd(&cnt[b], 1u);
       if (pos < A3_CAP) {
-        bool vvalid = !val_valid || bit_get3(val_valid, my);
-        uint8_t* r = stage + ((size_t)b * A3_CAP + pos) * 24;
-        *(int64_t*)r = k;
-        *(double*)(r + 8) = vals[my];
-        *(uint32_t*)(r + 16) = (uint32_t)my | (vvalid ? 0x80000000u : 0u);
-        *(uint32_t*)(r + 20) = 0;
-        my += stride;
+        bool vvalid = !val_valid || bit_get3(val_valid, my[r]);
+        uint8_t* rec = stage + ((size_t)b * A3_CAP + pos) * 24;
+        *(int64_t*)rec = k;
+        *(double*)(rec + 8) = vals[my[r]];
+        *(uint32_t*)(rec + 16) =
+            (uint32_t)my[r] | (vvalid ? 0x80000000u : 0u);
+        *(uint32_t*)(rec + 20) = 0;
+        if ((pos & (A3_QUANT - 1)) == A3_QUANT - 1)  // crossed a packet
+          queue[atomicAdd(qn, 1u) & (nbuck + 63)] = (uint16_t)b;
+        my[r] += step;
       } else {
-        atomicSub(&cnt[b], 1u);  // full: retry this row next tile
+        atomicSub(&cnt[b], 1u);  // staging full: retry this row next tile
       }
     }
     // block-wide termination check doubles as the pre-flush barrier
-    int live = __syncthreads_count((my < n) ? 1 : 0);
-    // cooperative flush: wave w drains buckets [w*bpw, (w+1)*bpw)
-    for (uint32_t b = wave * bpw; b < (wave + 1u) * bpw && b < nbuck; b++) {
-      uint32_t c = cnt[b];
-      uint32_t nfl = (live == 0) ? c : (c & ~(uint32_t)(A3_QUANT - 1));
-      if (nfl) {
-        // packet write: nfl*24 bytes, dword-per-lane from the 64-lane wave
-        uint8_t* dst = out + ((size_t)base_line[b] << 6) +
-                       (size_t)fl[b] * 24;
+    int live = __syncthreads_count(mine);
+    uint32_t nq = *qn;
+    if (live == 0) {
+      // drain: flush every bucket's remainder as (possibly partial) packets
+      for (uint32_t b = wave; b < nbuck; b += nwave) {
+        uint32_t c = cnt[b];
+        if (!c) continue;
+        uint8_t* dst = out + ((size_t)base_line[b] << 6) + (size_t)fl[b] * 24;
         const uint8_t* src = stage + (size_t)b * A3_CAP * 24;
-        uint32_t ndw = nfl * 6;
-        for (uint32_t d = lane; d < ndw; d += 64)
+        for (uint32_t d = lane; d < c * 6; d += 64)
           ((uint32_t*)dst)[d] = ((const uint32_t*)src)[d];
-        // move the remainder to the staging front (lane-parallel)
-        uint32_t rem = c - nfl;
-        for (uint32_t d = lane; d < rem * 6; d += 64)
-          ((uint32_t*)src)[d] = ((const uint32_t*)(src + (size_t)nfl * 24))[d];
-        if (lane == 0) {
-          fl[b] += nfl;
-          cnt[b] = rem;
-        }
+      }
+      __syncthreads();
+      break;
+    }
+    // flush only queued (full-packet) buckets; duplicates flush to 0 extra
+    for (uint32_t i = wave; i < nq; i += nwave) {
+      uint32_t b = queue[i & (nbuck + 63)];
+      uint32_t c = cnt[b];
+      uint32_t nfl = c & ~(uint32_t)(A3_QUANT - 1);
+      if (!nfl) continue;
+      uint8_t* dst = out + ((size_t)base_line[b] << 6) + (size_t)fl[b] * 24;
+      const uint8_t* src = stage + (size_t)b * A3_CAP * 24;
+      for (uint32_t d = lane; d < nfl * 6; d += 64)
+        ((uint32_t*)dst)[d] = ((const uint32_t*)src)[d];
+      uint32_t rem = c - nfl;
+      for (uint32_t d = lane; d < rem * 6; d += 64)
+        ((uint32_t*)src)[d] = ((const uint32_t*)(src + (size_t)nfl * 24))[d];
+      if (lane == 0) {
+        fl[b] += nfl;
+        cnt[b] = rem;
       }
     }
+    // queue ENTRIES below nq are no longer needed; resetting the counter
+    // concurrently with other waves' reads of those entries is benign
+    if (threadIdx.x == 0) *qn = 0;
     __syncthreads();
-    if (live == 0) break;
   }
 }
 
@@ -221,23 +249,22 @@ __global__ void __launch_bounds__(1024) k_agg3_bucket(
     }
     __syncthreads();
     uint32_t total = cum[nrange - 1];
-    for (uint32_t j = threadIdx.x; j < total; j += blockDim.x) {
-      // binary search the range holding record j
+    // 2-way ILP: both record loads (random-ish global) and both first LDS
+    // probe reads go out before either resolution — halves exposed latency
+    // at the 16-wave occupancy this 100KB-LDS kernel gets
+    auto locate = [&](uint32_t j) -> const uint8_t* {
       int lo = 0, hi = nrange - 1;
       while (lo < hi) {
         int mid = (lo + hi) >> 1;
         if (j < cum[mid]) hi = mid; else lo = mid + 1;
       }
       uint32_t before = lo ? cum[lo - 1] : 0;
-      const uint8_t* r =
-          part + ((size_t)line_scan[((size_t)b << grid_log2) | lo] << 6) +
-          (size_t)(j - before) * 24;
-      int64_t k = *(const int64_t*)r;
-      double v = *(const double*)(r + 8);
-      uint32_t rowv = *(const uint32_t*)(r + 16);
+      return part + ((size_t)line_scan[((size_t)b << grid_log2) | lo] << 6) +
+             (size_t)(j - before) * 24;
+    };
+    auto resolve = [&](int64_t k, double v, uint32_t rowv, uint32_t h) {
       uint32_t row = rowv & 0x7FFFFFFFu;
       bool vvalid = (rowv & 0x80000000u) != 0;
-      uint32_t h = (uint32_t)mix64_3((uint64_t)k) & (L3SLOTS - 1);
       int found = -1;
       for (int p = 0; p < L3PROBE; p++) {
         int64_t cur = ls_key[h];
@@ -266,6 +293,27 @@ __global__ void __launch_bounds__(1024) k_agg3_bucket(
         unsigned long long p = atomicAdd(lo_n, 1ull);
         leftover[p] = PartRow{k, v, rowv, 0};
       }
+    };
+    uint32_t j = threadIdx.x;
+    for (; j + blockDim.x < total; j += 2 * blockDim.x) {
+      const uint8_t* r0 = locate(j);
+      const uint8_t* r1 = locate(j + blockDim.x);
+      int64_t k0 = *(const int64_t*)r0;        // both loads in flight
+      int64_t k1 = *(const int64_t*)r1;
+      double v0 = *(const double*)(r0 + 8);
+      double v1 = *(const double*)(r1 + 8);
+      uint32_t rv0 = *(const uint32_t*)(r0 + 16);
+      uint32_t rv1 = *(const uint32_t*)(r1 + 16);
+      uint32_t h0 = (uint32_t)mix64_3((uint64_t)k0) & (L3SLOTS - 1);
+      uint32_t h1 = (uint32_t)mix64_3((uint64_t)k1) & (L3SLOTS - 1);
+      resolve(k0, v0, rv0, h0);
+      resolve(k1, v1, rv1, h1);
+    }
+    for (; j < total; j += blockDim.x) {
+      const uint8_t* r = locate(j);
+      int64_t k = *(const int64_t*)r;
+      resolve(k, *(const double*)(r + 8), *(const uint32_t*)(r + 16),
+              (uint32_t)mix64_3((uint64_t)k) & (L3SLOTS - 1));
     }
     __syncthreads();
     for (int s = threadIdx.x; s < L3SLOTS; s += blockDim.x) {
