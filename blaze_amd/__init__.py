@@ -348,7 +348,7 @@ def _import_output(array_ptr, schema_fields):
             out.append(dict(dtype="list", offsets=offsets, values=vals,
                             valid=valid))
             continue
-        if dt == "z":
+        if dt in ("z", "u"):
             offsets = np.ctypeslib.as_array(
                 c.cast(ch.buffers[1], c.POINTER(c.c_int32)), shape=(n + 1,)).copy()
             nbytes = int(offsets[-1])

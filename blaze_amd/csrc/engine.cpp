@@ -399,12 +399,20 @@ class AggOp {
   AggOp(const AggNode& node, const Conf& conf, hipStream_t stream)
       : stream_(stream) {
     if (node.exec_mode != 0) FAIL("SORT_AGG unsupported (hot path is HASH_AGG)");
-    if (node.grouping_exprs.size() != 1 ||
-        node.grouping_exprs[0].kind != Expr::Column)
-      FAIL("AggExec: exactly one Column grouping expr supported");
-    key_col_ = node.grouping_exprs[0].col_index;
-    key_name_ = node.grouping_names.empty() ? std::string("key")
-                                            : node.grouping_names[0];
+    if (node.grouping_exprs.empty() ||
+        node.grouping_exprs.size() > (size_t)GKEY_MAX_COLS)
+      FAIL("AggExec: 1..4 Column grouping exprs supported");
+    for (size_t gi = 0; gi < node.grouping_exprs.size(); gi++) {
+      const Expr& ge = node.grouping_exprs[gi];
+      if (ge.kind != Expr::Column)
+        FAIL("AggExec: grouping exprs must be Columns");
+      key_cols_.push_back(ge.col_index);
+      key_names_.push_back(gi < node.grouping_names.size()
+                               ? node.grouping_names[gi]
+                               : "key" + std::to_string(gi));
+    }
+    key_col_ = key_cols_[0];
+    key_name_ = key_names_[0];
     // agg set: any list of SUM/COUNT/AVG/MIN/MAX over ONE shared argument
     // column — {sum, cnt} accumulators (sum.rs, count.rs, avg.rs; AVG freeze
     // = sum ++ count, avg.rs:208-217) plus the optional side {min, max} pair
@@ -524,6 +532,35 @@ class AggOp {
   void consume(DevBatch&& b) {
     if (b.num_rows == 0) return;
     DevColumn& key = b.cols.at(key_col_);
+    if (!key_mode_set_) {
+      // key mode: single numeric column keeps the i64 slot fast path
+      // (incl. the two-phase pipeline); anything else — Utf8, Float64, or
+      // 2..4-column tuples — runs the generalized-key path (a4 closure,
+      // agg_ctx.rs:219-231 substitution per SURVEY.md 8c(i))
+      key_mode_set_ = true;
+      gkey_ = !(key_cols_.size() == 1 &&
+                (key.dt == DType::Int64 || key.dt == DType::Int32));
+      if (gkey_) {
+        if (has_first_ || has_coll_)
+          FAIL("FIRST/COLLECT with Utf8/multi-column grouping keys "
+               "unsupported on this path");
+        skip_enabled_ = false;  // emit_skipped is single-key-column shaped
+        for (uint32_t ci : key_cols_) {
+          DType dt = b.cols.at(ci).dt;
+          if (dt != DType::Int64 && dt != DType::Int32 &&
+              dt != DType::Float64 && dt != DType::Utf8 &&
+              dt != DType::Binary)
+            FAIL("unsupported grouping key dtype");
+          key_dts_.push_back(dt);
+        }
+        init_gkey();
+      }
+    }
+    if (gkey_) {
+      for (size_t k = 0; k < key_cols_.size(); k++)
+        if (b.cols.at(key_cols_[k]).dt != key_dts_[k])
+          FAIL("grouping key dtype changed mid-stream");
+    } else {
     if (key.dt != DType::Int64 && key.dt != DType::Int32)
       FAIL("grouping key must be Int64/Int32");
     if (key_dt_ == DType::Unsupported) key_dt_ = key.dt;
@@ -537,6 +574,7 @@ class AggOp {
       key.own_values = std::move(wide);
       key.values = key.own_values.get();
       key.dt = DType::Int64;
+    }
     }
     if (!merge_mode_) {
       // sum.rs:78-88 prepare_partial_args: the argument is CAST to the
@@ -584,8 +622,8 @@ class AggOp {
       init_agg2_conf();  // chunk bound must be read BEFORE sizing the chunk
       // MIN/MAX agg sets stay single-phase: the LDS bucket kernel's slot
       // holds {key,cnt,sum,first} only (perf note in DESIGN.md)
-      if (!merge_mode_ && !has_mm_ && !has_first_ && !has_coll_ &&
-          b.num_rows - done >= AGG2_MIN_CHUNK) {
+      if (!gkey_ && !merge_mode_ && !has_mm_ && !has_first_ &&
+          !has_coll_ && b.num_rows - done >= AGG2_MIN_CHUNK) {
         int64_t chunk2 = std::min(b.num_rows - done, agg2_chunk_max_);
         if (done + chunk2 < b.num_rows) chunk2 &= ~(int64_t)7;
         two_phase_chunk(b, done, chunk2);
@@ -607,7 +645,9 @@ class AggOp {
       }
       int64_t chunk = std::min(b.num_rows - done, free_slots);
       if (done + chunk < b.num_rows) chunk &= ~(int64_t)7;  // bitmap-sliceable
-      if (merge_mode_) {
+      if (gkey_) {
+        gkey_chunk(b, done, chunk);
+      } else if (merge_mode_) {
         const DevColumn& buf = b.cols.at(1);
         if (buf.dt != DType::Binary) FAIL("agg-buf column must be Binary");
         // acc_offsets are absolute into buf.values, so only the index shifts
@@ -689,10 +729,19 @@ class AggOp {
     return true;
   }
 
+  std::vector<OutField> key_fields() const {
+    std::vector<OutField> f;
+    for (size_t k = 0; k < key_cols_.size(); k++) {
+      DType kd = k < key_dts_.size() ? key_dts_[k]
+                 : (key_dt_ == DType::Unsupported ? DType::Int64 : key_dt_);
+      f.push_back({key_names_[k], kd, true});
+    }
+    return f;
+  }
+
   std::vector<OutField> output_fields() const {
-    DType kd = key_dt_ == DType::Unsupported ? DType::Int64 : key_dt_;
     if (final_output_) {
-      std::vector<OutField> f = {{key_name_, kd, true}};
+      std::vector<OutField> f = key_fields();
       const DType vdt = val_is_int_ ? DType::Int64 : DType::Float64;
       for (size_t i = 0; i < agg_kinds_.size(); i++) {
         if (agg_kinds_[i] == AGGL_CNT)
@@ -708,8 +757,9 @@ class AggOp {
       return f;
     }
     // partial/partial-merge: grouping + AGG_BUF (agg/mod.rs:37)
-    return {{key_name_, kd, true},
-            {"#9223372036854775807", DType::Binary, false}};
+    std::vector<OutField> f = key_fields();
+    f.push_back({"#9223372036854775807", DType::Binary, false});
+    return f;
   }
 
   // drain: produce all output batches (host-staged)
@@ -965,6 +1015,9 @@ class AggOp {
     std::vector<std::pair<int64_t, std::vector<DevOutCol>>> out;
     drain_timing();
     prepare_collect();
+    if (gkey_)
+      FAIL("AURON_HIP_DEVICE_OUTPUT with generalized grouping keys "
+           "unsupported (unset the conf)");
     if (!spill_.empty())
       FAIL("AURON_HIP_DEVICE_OUTPUT with spill unsupported (unset the conf)");
     if (!skipped_.empty())
@@ -1070,6 +1123,11 @@ class AggOp {
       t_.f_st = d_fst_.get<uint8_t>();
       launch_first_init(t_.f_row, t_.f_val, t_.f_st, cap + 2, stream_);
     }
+    if (gkey_) {  // growth re-allocs the off array; pool survives
+      d_gkoff_.alloc(cap * 8);
+      AURON_HIP(hipMemsetAsync(d_gkoff_.get(), 0, cap * 8, stream_));
+      g_.off = d_gkoff_.get<unsigned long long>();
+    }
     if (has_coll_ && !d_ckey_) {  // pool survives grows (keys, not slots)
       coll_cap_ = conf_coll_cap_;
       d_ckey_.alloc(coll_cap_ * 8);
@@ -1096,6 +1154,135 @@ class AggOp {
   static constexpr int AGG3_NBUCK_LOG2 = 9;   // v3: 512 buckets
   static constexpr int AGG3_NBUCK = 1 << AGG3_NBUCK_LOG2;
   static constexpr int AGG3_GRID_LOG2 = 8;    // v3 scatter: 256 x 1024-thr           // 1024 buckets: scatter write-line footprint ~L2-sized (512 buckets overflow the 2048-slot LDS window: 2.3% leftovers, 3x slower)
+
+  // ---- generalized-key machinery (kernels_gkey.hip) ----------------------
+  void init_gkey() {
+    gk_pool_cap_ = 64 << 20;
+    d_gkpool_.alloc(gk_pool_cap_);
+    d_gkpn_.alloc(8);
+    AURON_HIP(hipMemsetAsync(d_gkpn_.get(), 0, 8, stream_));
+    d_gkoff_.alloc(t_.cap * 8);
+    AURON_HIP(hipMemsetAsync(d_gkoff_.get(), 0, t_.cap * 8, stream_));
+    g_.off = d_gkoff_.get<unsigned long long>();
+    g_.pool = d_gkpool_.get<uint8_t>();
+    g_.pool_n = d_gkpn_.get<unsigned long long>();
+    g_.pool_cap = gk_pool_cap_;
+    uint8_t dts[GKEY_MAX_COLS] = {};
+    for (size_t k = 0; k < key_dts_.size(); k++) {
+      switch (key_dts_[k]) {
+        case DType::Int64: dts[k] = GK_I64; break;
+        case DType::Int32: dts[k] = GK_I32; break;
+        case DType::Float64: dts[k] = GK_F64; break;
+        default: dts[k] = GK_UTF8; break;  // Utf8/Binary
+      }
+    }
+    d_gkdts_.alloc(GKEY_MAX_COLS);
+    AURON_HIP(hipMemcpyAsync(d_gkdts_.get(), dts, GKEY_MAX_COLS,
+                             hipMemcpyHostToDevice, stream_));
+    AURON_HIP(hipStreamSynchronize(stream_));
+  }
+
+  void gk_ensure_pool(int64_t need_more) {
+    if (gk_used_bound_ + need_more <= gk_pool_cap_) return;
+    // refresh the true cursor (the bound counts every row as a new group)
+    if (pinned_meta_.size() < 16) pinned_meta_.alloc(16);
+    AURON_HIP(hipMemcpyAsync(pinned_meta_.get(), d_gkpn_.get(), 8,
+                             hipMemcpyDeviceToHost, stream_));
+    AURON_HIP(hipStreamSynchronize(stream_));
+    gk_used_bound_ = (int64_t)*pinned_meta_.get<unsigned long long>();
+    if (gk_used_bound_ + need_more <= gk_pool_cap_) return;
+    int64_t new_cap = gk_pool_cap_;
+    while (new_cap < gk_used_bound_ + need_more) new_cap *= 2;
+    DevBuf np(new_cap);
+    AURON_HIP(hipMemcpyAsync(np.get(), d_gkpool_.get(), gk_pool_cap_,
+                             hipMemcpyDeviceToDevice, stream_));
+    AURON_HIP(hipStreamSynchronize(stream_));
+    d_gkpool_ = std::move(np);
+    gk_pool_cap_ = new_cap;
+    g_.pool = d_gkpool_.get<uint8_t>();
+    g_.pool_cap = gk_pool_cap_;
+  }
+
+  GKeyCols gk_cols(const DevBatch& b, int64_t done) const {
+    GKeyCols gc;
+    gc.ncols = (int)key_cols_.size();
+    for (size_t k = 0; k < key_cols_.size(); k++) {
+      const DevColumn& c = b.cols.at(key_cols_[k]);
+      switch (key_dts_[k]) {
+        case DType::Int64:
+        case DType::Float64:
+          gc.dt[k] = key_dts_[k] == DType::Int64 ? GK_I64 : GK_F64;
+          gc.values[k] = (const uint8_t*)c.values + done * 8;
+          break;
+        case DType::Int32:
+          gc.dt[k] = GK_I32;
+          gc.values[k] = (const uint8_t*)c.values + done * 4;
+          break;
+        default:  // Utf8/Binary: offsets are absolute into the data buffer
+          gc.dt[k] = GK_UTF8;
+          gc.values[k] = c.values;
+          gc.offsets[k] = c.offsets + done;
+          break;
+      }
+      gc.validity[k] = c.validity ? c.validity + done / 8 : nullptr;
+    }
+    return gc;
+  }
+
+  // one bounded-size chunk of the generalized-key path: encode the key
+  // tuples -> scan -> upsert (slot per row) -> slot-indexed accumulate or
+  // frozen merge
+  void gkey_chunk(const DevBatch& b, int64_t done, int64_t chunk) {
+    GKeyCols gc = gk_cols(b, done);
+    if (d_enclens_.size() < (size_t)(chunk + 1) * 4) {
+      d_enclens_.alloc((chunk + 1) * 4);
+      d_encoffs_.alloc((chunk + 1) * 4);
+      d_gkslots_.alloc(chunk * 4);
+    }
+    launch_gkey_enc_lens(gc, chunk, d_enclens_.get<uint32_t>(), stream_);
+    size_t tb = 0;
+    scan_counts_matrix(d_enclens_.get<uint32_t>(), d_encoffs_.get<uint32_t>(),
+                       chunk + 1, nullptr, &tb, stream_);
+    if (d_gkscan_tmp_.size() < tb) d_gkscan_tmp_.alloc(tb);
+    scan_counts_matrix(d_enclens_.get<uint32_t>(), d_encoffs_.get<uint32_t>(),
+                       chunk + 1, d_gkscan_tmp_.get(), &tb, stream_);
+    if (pinned_meta_.size() < 16) pinned_meta_.alloc(16);
+    AURON_HIP(hipMemcpyAsync(pinned_meta_.get(),
+                             d_encoffs_.get<uint32_t>() + chunk, 4,
+                             hipMemcpyDeviceToHost, stream_));
+    AURON_HIP(hipStreamSynchronize(stream_));
+    int64_t total = (int64_t)*pinned_meta_.get<uint32_t>();
+    if (d_encbytes_.size() < (size_t)total + 8)
+      d_encbytes_.alloc(total + 8);
+    launch_gkey_enc_write(gc, d_encoffs_.get<uint32_t>(),
+                          d_encbytes_.get<uint8_t>(), chunk, stream_);
+    gk_ensure_pool(total + 4 * chunk);  // worst case: every row a new group
+    hipEvent_t e0, e1;
+    AURON_HIP(hipEventCreate(&e0));
+    AURON_HIP(hipEventCreate(&e1));
+    AURON_HIP(hipEventRecord(e0, stream_));
+    launch_gkey_upsert(t_, g_, d_encoffs_.get<uint32_t>(),
+                       d_encbytes_.get<uint8_t>(), chunk, row_cursor_,
+                       d_gkslots_.get<uint32_t>(), stream_);
+    if (merge_mode_) {
+      const DevColumn& buf = b.cols.back();
+      if (buf.dt != DType::Binary) FAIL("agg-buf column must be Binary");
+      launch_gkey_merge_frozen_idx(t_, d_gkslots_.get<uint32_t>(),
+                                   (const uint8_t*)buf.values,
+                                   buf.offsets + done, chunk, layout_,
+                                   stream_);
+    } else {
+      const DevColumn& val = b.cols.at(val_col_);
+      launch_gkey_update_idx(t_, d_gkslots_.get<uint32_t>(),
+                             (const double*)val.values + done,
+                             val.validity ? val.validity + done / 8 : nullptr,
+                             chunk, stream_);
+      update_rows_ += chunk;
+    }
+    AURON_HIP(hipEventRecord(e1, stream_));
+    ev_pairs_.push_back({e0, e1});
+    gk_used_bound_ += total + 4 * chunk;
+  }
 
   // Read the two-phase tuning knobs once, BEFORE the first chunk size is
   // computed (the scratch buffers are sized from agg2_chunk_max_, so the
@@ -1461,6 +1648,9 @@ class AggOp {
     while (t_.cap * 3 / 4 < need && t_.cap < max_cap_) grow(t_.cap * 4);
     if (t_.cap * 3 / 4 < need &&
         (int64_t)num_groups_host() > specials_count_) {
+      if (gkey_)
+        FAIL("spill with Utf8/multi-column grouping keys unsupported "
+             "(raise AURON_HIP_MEM_BUDGET)");
       if (has_coll_)
         FAIL("COLLECT_LIST with table spill unsupported (round-2: drain the "
              "pool into the spill buckets)");
@@ -1470,12 +1660,17 @@ class AggOp {
 
   void grow(int64_t new_cap) {
     AggTable old = t_;
+    GKeyTable oldg = g_;
     DevBuf oslots = std::move(d_slots_), os = std::move(d_special_),
            ong = std::move(d_ng_), oerr = std::move(d_err_),
            omm = std::move(d_mm_), ofrow = std::move(d_frow_),
-           ofval = std::move(d_fval_), ofst = std::move(d_fst_);
+           ofval = std::move(d_fval_), ofst = std::move(d_fst_),
+           ogkoff = std::move(d_gkoff_);
     init_table(new_cap);
-    launch_agg_rebuild(t_, old, stream_);
+    if (gkey_)
+      launch_gkey_rebuild(t_, g_, old, oldg, stream_);
+    else
+      launch_agg_rebuild(t_, old, stream_);
     AURON_HIP(hipStreamSynchronize(stream_));
   }
 
@@ -1519,6 +1714,55 @@ class AggOp {
     AURON_HIP(hipStreamSynchronize(stream_));
     dst->resize(len);
     memcpy(dst->data(), pinned_emit_.get(), len);
+  }
+
+  // decode the generalized-key grouping columns for the ordered groups
+  std::vector<HostOutCol> emit_gkey_cols(const uint32_t* order_slots,
+                                         int64_t n) {
+    std::vector<HostOutCol> out;
+    size_t bm = (n + 7) / 8;
+    for (size_t k = 0; k < key_cols_.size(); k++) {
+      HostOutCol c;
+      c.dt = key_dts_[k];
+      DevBuf bmbuf(bm);
+      if (c.dt == DType::Utf8 || c.dt == DType::Binary) {
+        DevBuf lens(n * 4);
+        launch_gkey_out_lens(g_, order_slots, n, d_gkdts_.get<uint8_t>(),
+                             (int)key_cols_.size(), (int)k,
+                             lens.get<uint32_t>(), bmbuf.get<uint8_t>(),
+                             stream_);
+        std::vector<uint32_t> h_lens(n);
+        AURON_HIP(hipMemcpyAsync(h_lens.data(), lens.get(), n * 4,
+                                 hipMemcpyDeviceToHost, stream_));
+        AURON_HIP(hipStreamSynchronize(stream_));
+        c.offsets.assign(n + 1, 0);
+        for (int64_t i = 0; i < n; i++)
+          c.offsets[i + 1] = c.offsets[i] + (int32_t)h_lens[i];
+        DevBuf d_off((n + 1) * 4), data(c.offsets[n] ? c.offsets[n] : 1);
+        AURON_HIP(hipMemcpyAsync(d_off.get(), c.offsets.data(), (n + 1) * 4,
+                                 hipMemcpyHostToDevice, stream_));
+        launch_gkey_out_bytes(g_, order_slots, n, d_gkdts_.get<uint8_t>(),
+                              (int)key_cols_.size(), (int)k,
+                              d_off.get<int32_t>(), data.get<uint8_t>(),
+                              stream_);
+        d2h_pinned(data.get(), &c.values, c.offsets[n]);
+      } else {
+        size_t w = dtype_width(c.dt);
+        DevBuf vals(n * w);
+        launch_gkey_out_fixed(g_, order_slots, n, d_gkdts_.get<uint8_t>(),
+                              (int)key_cols_.size(), (int)k,
+                              vals.get<uint8_t>(), bmbuf.get<uint8_t>(),
+                              stream_);
+        d2h_pinned(vals.get(), &c.values, n * w);
+      }
+      std::vector<uint8_t> h_bm(bm);
+      AURON_HIP(hipMemcpyAsync(h_bm.data(), bmbuf.get(), bm,
+                               hipMemcpyDeviceToHost, stream_));
+      AURON_HIP(hipStreamSynchronize(stream_));
+      attach_validity(&c, h_bm, n);
+      out.push_back(std::move(c));
+    }
+    return out;
   }
 
   std::pair<int64_t, std::vector<HostOutCol>> emit_groups(
@@ -1613,8 +1857,13 @@ class AggOp {
       }
       AURON_HIP(hipStreamSynchronize(stream_));
       DBG("agg.emit final d2h done");
-      attach_validity(&key_col, kv, n);
-      cols.push_back(std::move(key_col));
+      if (gkey_) {
+        for (auto& kc2 : emit_gkey_cols(order_slots, n))
+          cols.push_back(std::move(kc2));
+      } else {
+        attach_validity(&key_col, kv, n);
+        cols.push_back(std::move(key_col));
+      }
       const DType vdt = val_is_int_ ? DType::Int64 : DType::Float64;
       for (uint32_t k : agg_kinds_) {
         HostOutCol ac;
@@ -1667,8 +1916,13 @@ class AggOp {
       buf_col.offsets = std::move(h_offs);
       d2h_pinned(data.get(), &buf_col.values, (size_t)buf_col.offsets[n]);
       DBG("agg.emit partial freeze d2h done");
-      attach_validity(&key_col, kv, n);
-      cols = {std::move(key_col), std::move(buf_col)};
+      if (gkey_) {
+        cols = emit_gkey_cols(order_slots, n);
+        cols.push_back(std::move(buf_col));
+      } else {
+        attach_validity(&key_col, kv, n);
+        cols = {std::move(key_col), std::move(buf_col)};
+      }
     }
     return {n, std::move(cols)};
   }
@@ -1751,6 +2005,15 @@ class AggOp {
   int64_t batch_size_ = 10000;
   bool skip_enabled_ = false, skipping_ = false;
   bool device_out_ = false;
+  // generalized grouping keys (Utf8 / multi-column; kernels_gkey.hip)
+  bool key_mode_set_ = false, gkey_ = false;
+  std::vector<uint32_t> key_cols_;
+  std::vector<std::string> key_names_;
+  std::vector<DType> key_dts_;
+  GKeyTable g_;
+  int64_t gk_pool_cap_ = 0, gk_used_bound_ = 0;
+  DevBuf d_gkpool_, d_gkpn_, d_gkoff_, d_gkdts_, d_enclens_, d_encoffs_,
+      d_encbytes_, d_gkslots_, d_gkscan_tmp_;
   double skip_ratio_ = 0.999;
   int64_t skip_min_rows_ = 20000;
   uint64_t row_cursor_ = 0;
@@ -1853,8 +2116,12 @@ class ShuffleOp {
         else if (c.dt == DType::Int32)
           launch_hash_fold_i32((const int32_t*)c.values, c.validity, n,
                                hashes.get<int32_t>(), stream_);
+        else if (c.dt == DType::Utf8 || c.dt == DType::Binary)
+          launch_hash_fold_bytes(c.offsets, (const uint8_t*)c.values,
+                                 c.validity, n, hashes.get<int32_t>(),
+                                 stream_);
         else
-          FAIL("hash expr column must be Int64/Int32 (hot-path scope)");
+          FAIL("hash expr column must be Int64/Int32/Utf8 (hot-path scope)");
       }
       launch_pmod(hashes.get<int32_t>(), n, (int32_t)P_,
                   part_ids.get<uint32_t>(), stream_);

@@ -371,6 +371,70 @@ void launch_dest_counts(const uint32_t* pids, const int32_t* offsets,
                         int64_t n, uint32_t world, unsigned long long* rows,
                         unsigned long long* bytes, hipStream_t s);
 
+// ---- generalized grouping keys (a4 closure: Utf8 / multi-column) ---------
+// The reference row-encodes key tuples (agg_ctx.rs:219-231 arrow-row); the
+// key ENCODING is substitutable as long as the grouping
+// column VALUES round-trip — this engine uses its own per-column
+// [valid u8][value bytes] encoding (utf8: u32 len + bytes), hashed into the
+// same 32-byte slot table (slot.key holds the 64-bit key-bytes hash; a side
+// `off` array points into a device byte pool holding each group's encoded
+// key for the byte-compare probe).
+constexpr int GKEY_MAX_COLS = 4;
+enum GKeyDt : uint8_t { GK_I64 = 0, GK_I32 = 1, GK_F64 = 2, GK_UTF8 = 3 };
+struct GKeyCols {
+  int ncols = 0;
+  const void* values[GKEY_MAX_COLS] = {};
+  const int32_t* offsets[GKEY_MAX_COLS] = {};   // utf8 only
+  const uint8_t* validity[GKEY_MAX_COLS] = {};
+  uint8_t dt[GKEY_MAX_COLS] = {};
+};
+struct GKeyTable {
+  unsigned long long* off = nullptr;  // [cap]: 0 = pending/absent, else
+                                      // pool offset + 1 (release-published)
+  uint8_t* pool = nullptr;            // [u32 len][encoded bytes] records
+  unsigned long long* pool_n = nullptr;  // [1] byte cursor
+  int64_t pool_cap = 0;
+};
+// row murmur3 fold over a Utf8/Binary column (mur.rs:19-30 byte path)
+void launch_hash_fold_bytes(const int32_t* offsets, const uint8_t* data,
+                            const uint8_t* valid, int64_t n, int32_t* hashes,
+                            hipStream_t s);
+// encoded key length per row (u32), then the byte write after a scan
+void launch_gkey_enc_lens(const GKeyCols& c, int64_t n, uint32_t* lens,
+                          hipStream_t s);
+void launch_gkey_enc_write(const GKeyCols& c, const uint32_t* enc_offsets,
+                           uint8_t* enc_bytes, int64_t n, hipStream_t s);
+// probe-or-insert each row's encoded key; writes the slot index per row and
+// takes the first_row atomicMin. Pool capacity must be host-ensured.
+void launch_gkey_upsert(const AggTable& t, const GKeyTable& g,
+                        const uint32_t* enc_offsets, const uint8_t* enc_bytes,
+                        int64_t n, uint64_t row_offset, uint32_t* slots,
+                        hipStream_t s);
+// slot-indexed accumulate / frozen-record merge (no per-row probing)
+void launch_gkey_update_idx(const AggTable& t, const uint32_t* slots,
+                            const double* vals, const uint8_t* val_valid,
+                            int64_t n, hipStream_t s);
+void launch_gkey_merge_frozen_idx(const AggTable& t, const uint32_t* slots,
+                                  const uint8_t* acc_data,
+                                  const int32_t* acc_offsets, int64_t n,
+                                  uint32_t layout, hipStream_t s);
+// table growth: re-probe every occupied slot of src into dst by stored hash
+void launch_gkey_rebuild(const AggTable& dst, const GKeyTable& dg,
+                         const AggTable& src, const GKeyTable& sg,
+                         hipStream_t s);
+// emit: decode grouping column `col` for the ordered groups
+void launch_gkey_out_fixed(const GKeyTable& g, const uint32_t* order_slots,
+                           int64_t n, const uint8_t* dts, int ncols, int col,
+                           uint8_t* values, uint8_t* valid_bitmap,
+                           hipStream_t s);
+void launch_gkey_out_lens(const GKeyTable& g, const uint32_t* order_slots,
+                          int64_t n, const uint8_t* dts, int ncols, int col,
+                          uint32_t* lens, uint8_t* valid_bitmap, hipStream_t s);
+void launch_gkey_out_bytes(const GKeyTable& g, const uint32_t* order_slots,
+                           int64_t n, const uint8_t* dts, int ncols, int col,
+                           const int32_t* out_offsets, uint8_t* out_data,
+                           hipStream_t s);
+
 // parquet RLE/bit-packed run expansion (runs = PqRun[] from parquet.h)
 void launch_runs_expand_u32(const void* runs, int nruns, const uint8_t* bytes,
                             int64_t n, uint32_t* out, hipStream_t s);

@@ -225,6 +225,52 @@ def plan_final_only(resource_id="input0"):
     return task_definition(final)
 
 
+def plan_partial_final_gkey(key_fields, agg_fns=("sum", "count"),
+                            val_dt=DT_FLOAT64, resource_id="input0"):
+    """FFIReader -> Agg(Partial) -> Agg(Final) over GENERALIZED grouping
+    keys: key_fields = [(name, dt_tag, nullable), ...] (1..4 columns, Utf8 /
+    Int64 / Int32 / Float64); the val column follows the key columns."""
+    fields = [field(n, dt, nul) for (n, dt, nul) in key_fields]
+    vi = len(fields)
+    fields.append(field("val", val_dt, True))
+    reader = ffi_reader(fields, resource_id)
+    fn_tags = {"sum": AGG_SUM, "count": AGG_COUNT, "avg": AGG_AVG,
+               "min": AGG_MIN, "max": AGG_MAX}
+    aggs = []
+    for fn in agg_fns:
+        tag = fn_tags[fn]
+        dt = DT_INT64 if tag == AGG_COUNT else             (DT_FLOAT64 if tag == AGG_AVG else val_dt)
+        aggs.append(agg_expr(tag, [column("val", vi)], dt))
+    grouping = [column(n, i) for i, (n, _, _) in enumerate(key_fields)]
+    gnames = [n for (n, _, _) in key_fields]
+    partial = agg(reader, grouping, aggs, [MODE_PARTIAL] * len(aggs),
+                  gnames, list(agg_fns))
+    final = agg(partial, [column(n, i) for i, (n, _, _)
+                          in enumerate(key_fields)], aggs,
+                [MODE_FINAL] * len(aggs), gnames, list(agg_fns))
+    return task_definition(final)
+
+
+def plan_partial_only_gkey(key_fields, agg_fns=("sum", "count"),
+                           val_dt=DT_FLOAT64, resource_id="input0"):
+    fields = [field(n, dt, nul) for (n, dt, nul) in key_fields]
+    vi = len(fields)
+    fields.append(field("val", val_dt, True))
+    reader = ffi_reader(fields, resource_id)
+    fn_tags = {"sum": AGG_SUM, "count": AGG_COUNT, "avg": AGG_AVG,
+               "min": AGG_MIN, "max": AGG_MAX}
+    aggs = []
+    for fn in agg_fns:
+        tag = fn_tags[fn]
+        dt = DT_INT64 if tag == AGG_COUNT else             (DT_FLOAT64 if tag == AGG_AVG else val_dt)
+        aggs.append(agg_expr(tag, [column("val", vi)], dt))
+    grouping = [column(n, i) for i, (n, _, _) in enumerate(key_fields)]
+    gnames = [n for (n, _, _) in key_fields]
+    partial = agg(reader, grouping, aggs, [MODE_PARTIAL] * len(aggs),
+                  gnames, list(agg_fns))
+    return task_definition(partial)
+
+
 def plan_agg_shuffle(data_file, index_file, num_partitions=200,
                      resource_id="input0", partition_id=0, skipping=False):
     """FFIReader -> Agg(Partial) -> ShuffleWriter(hash(key), P): config 4

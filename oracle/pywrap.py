@@ -413,3 +413,37 @@ def collect_freeze_rec(values, fmt="<d"):
     import struct
     raw = b"".join(struct.pack(fmt, v) for v in values)
     return write_len(len(raw)) + raw
+
+
+# ---- generalized grouping keys (Utf8 / multi-column tuples) ----------------
+# Reference semantics: GROUP BY over the key TUPLE with null values grouping
+# together per column (agg_ctx.rs grouping rows; arrow-row encoding makes
+# null a distinct point in the ordering). Insertion order = first occurrence,
+# like the single-key paths. Pure-python restatement for test sizes.
+def gkey_agg_groups(key_cols, vals, val_valid=None, key_valids=None):
+    """key_cols: list of per-column value sequences (str/int/float);
+    key_valids: optional list of per-column validity arrays. Returns
+    (ordered key tuples with None for nulls, sums, counts, mins, maxs)."""
+    n = len(vals)
+    groups = {}
+    for i in range(n):
+        kt = []
+        for c, col in enumerate(key_cols):
+            valid = key_valids is None or key_valids[c] is None or \
+                bool(key_valids[c][i])
+            kt.append(col[i] if valid else None)
+        kt = tuple(kt)
+        if kt not in groups:
+            groups[kt] = [0.0, 0, None, None]
+        if val_valid is None or val_valid[i]:
+            g = groups[kt]
+            v = vals[i]
+            g[0] += v
+            g[1] += 1
+            g[2] = v if g[2] is None or v < g[2] else g[2]
+            g[3] = v if g[3] is None or v > g[3] else g[3]
+    ordered = list(groups.keys())
+    return (ordered, [groups[k][0] for k in ordered],
+            [groups[k][1] for k in ordered],
+            [groups[k][2] for k in ordered],
+            [groups[k][3] for k in ordered])
