@@ -2067,7 +2067,10 @@ class ShuffleOp {
     target_block_ = (size_t)conf.get_i("SHUFFLE_COMPRESSION_TARGET_BUF_SIZE",
                                        4194304);
     std::string codec = conf.get("SPARK_IO_COMPRESSION_CODEC", "lz4");
-    if (codec != "lz4") FAIL("only lz4 shuffle codec implemented");
+    if (codec == "lz4") codec_ = 0;
+    else if (codec == "zstd") codec_ = 1;  // ipc_compression.rs:189-196
+    else FAIL("unsupported shuffle codec (lz4/zstd on this path)");
+    zstd_level_ = (int)conf.get_i("SPARK_IO_COMPRESSION_ZSTD_LEVEL", 1);
   }
 
   void consume(DevBatch&& b) {
@@ -2307,7 +2310,7 @@ class ShuffleOp {
     std::string err;
     if (!write_shuffle_files(cols, part_offsets, batch_size_,
                              node_.output_data_file, node_.output_index_file,
-                             &err))
+                             &err, codec_, zstd_level_))
       FAIL(err);
   }
 
@@ -2318,6 +2321,7 @@ class ShuffleOp {
   uint32_t P_ = 1;
   int64_t batch_size_ = 10000;
   size_t target_block_ = 4194304;
+  int codec_ = 0, zstd_level_ = 1;
   std::vector<DevBatch> staged_;
 };
 
@@ -2559,6 +2563,7 @@ struct Runtime {
   };
   std::vector<Stage> stages_;
   std::vector<OutField> out_fields;
+  int ipc_codec_ = 0;  // IpcReader block codec (SPARK_IO_COMPRESSION_CODEC)
   std::vector<std::pair<int64_t, std::vector<HostOutCol>>> outputs;
   size_t emit_idx = 0;
   // device-resident outputs (AURON_HIP_DEVICE_OUTPUT): exported through
@@ -2580,6 +2585,12 @@ struct Runtime {
   void run() {
     auto t0 = std::chrono::steady_clock::now();
     Conf conf{&cb};
+    {
+      std::string codec = conf.get("SPARK_IO_COMPRESSION_CODEC", "lz4");
+      if (codec == "zstd") ipc_codec_ = 1;
+      else if (codec != "lz4")
+        FAIL("unsupported shuffle codec (lz4/zstd on this path)");
+    }
     // collect plan chain root→leaf
     std::vector<const PlanNode*> chain;
     const PlanNode* p = td->plan.get();
@@ -3043,7 +3054,8 @@ struct Runtime {
       if (rc == 0) break;
       std::vector<uint8_t> payload;
       std::string err;
-      if (!ipc_decode_blocks(data, len, &payload, &err)) FAIL(err);
+      if (!ipc_decode_blocks(data, len, &payload, &err, ipc_codec_))
+        FAIL(err);
       size_t used = 0;
       while (used < payload.size()) {
         int64_t rows = 0;

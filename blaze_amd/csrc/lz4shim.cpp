@@ -95,4 +95,82 @@ bool lz4_decompress_frame(const uint8_t* src, size_t len,
   return true;
 }
 
+// ---- zstd frames -----------------------------------------------------------
+namespace {
+typedef size_t (*ZSTD_compressBound_t)(size_t);
+typedef size_t (*ZSTD_compress_t)(void*, size_t, const void*, size_t, int);
+typedef unsigned long long (*ZSTD_getFrameContentLength_t)(const void*,
+                                                           size_t);
+typedef size_t (*ZSTD_decompress_t)(void*, size_t, const void*, size_t);
+typedef unsigned (*ZSTD_isError_t)(size_t);
+
+struct ZstdApi {
+  void* handle = nullptr;
+  ZSTD_compressBound_t bound = nullptr;
+  ZSTD_compress_t compress = nullptr;
+  ZSTD_getFrameContentLength_t content_len = nullptr;
+  ZSTD_decompress_t decompress = nullptr;
+  ZSTD_isError_t iserr = nullptr;
+};
+
+ZstdApi& zstd_api() {
+  static ZstdApi z;
+  if (!z.handle) {
+    void* h = dlopen("libzstd.so.1", RTLD_NOW | RTLD_GLOBAL);
+    if (!h) h = dlopen("libzstd.so", RTLD_NOW | RTLD_GLOBAL);
+    if (h) {
+      z.handle = h;
+      z.bound = (ZSTD_compressBound_t)dlsym(h, "ZSTD_compressBound");
+      z.compress = (ZSTD_compress_t)dlsym(h, "ZSTD_compress");
+      z.content_len =
+          (ZSTD_getFrameContentLength_t)dlsym(h, "ZSTD_getDecompressedSize");
+      z.decompress = (ZSTD_decompress_t)dlsym(h, "ZSTD_decompress");
+      z.iserr = (ZSTD_isError_t)dlsym(h, "ZSTD_isError");
+    }
+  }
+  return z;
+}
+}  // namespace
+
+bool zstd_compress_frame(const uint8_t* src, size_t len, int level,
+                         std::vector<uint8_t>* out, std::string* err) {
+  ZstdApi& z = zstd_api();
+  if (!z.compress || !z.bound || !z.iserr) {
+    *err = "libzstd.so.1 not available";
+    return false;
+  }
+  size_t base = out->size();
+  size_t cap = z.bound(len);
+  out->resize(base + cap);
+  size_t rc = z.compress(out->data() + base, cap, src, len, level);
+  if (z.iserr(rc)) {
+    *err = "zstd compress failed";
+    return false;
+  }
+  out->resize(base + rc);
+  return true;
+}
+
+bool zstd_decompress_frame(const uint8_t* src, size_t len,
+                           std::vector<uint8_t>* out, std::string* err) {
+  ZstdApi& z = zstd_api();
+  if (!z.decompress || !z.content_len || !z.iserr) {
+    *err = "libzstd.so.1 not available";
+    return false;
+  }
+  unsigned long long need = z.content_len(src, len);
+  if ((need == 0 && len > 8) || need > (1ull << 33)) {
+    *err = "zstd frame without content length";
+    return false;
+  }
+  size_t base = out->size();
+  out->resize(base + (size_t)need);
+  size_t rc = z.decompress(out->data() + base, (size_t)need, src, len);
+  if (z.iserr(rc) || rc != (size_t)need) {
+    *err = "zstd decompress failed";
+    return false;
+  }
+  return true;
+}
+
 }  // namespace auron

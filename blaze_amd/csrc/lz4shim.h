@@ -20,4 +20,13 @@ bool lz4_compress_frame(const uint8_t* src, size_t len, std::vector<uint8_t>* ou
 bool lz4_decompress_frame(const uint8_t* src, size_t len,
                           std::vector<uint8_t>* out, std::string* err);
 
+// zstd frames (libzstd.so.1 via dlopen): the shuffle codec alternative the
+// reference selects through spark.io.compression.codec
+// (ipc_compression.rs:189-196; level conf SPARK_IO_COMPRESSION_ZSTD_LEVEL,
+// default 1)
+bool zstd_compress_frame(const uint8_t* src, size_t len, int level,
+                         std::vector<uint8_t>* out, std::string* err);
+bool zstd_decompress_frame(const uint8_t* src, size_t len,
+                           std::vector<uint8_t>* out, std::string* err);
+
 }  // namespace auron

@@ -176,7 +176,8 @@ bool serde_read_batch(const uint8_t* p, size_t len, size_t* used,
 }
 
 bool ipc_decode_blocks(const uint8_t* p, size_t len,
-                       std::vector<uint8_t>* payload, std::string* err) {
+                       std::vector<uint8_t>* payload, std::string* err,
+                       int codec) {
   size_t pos = 0;
   while (pos + 4 <= len) {
     uint32_t block_len;
@@ -186,7 +187,10 @@ bool ipc_decode_blocks(const uint8_t* p, size_t len,
       *err = "ipc: truncated block";
       return false;
     }
-    if (!lz4_decompress_frame(p + pos, block_len, payload, err)) return false;
+    bool ok = codec == 1
+                  ? zstd_decompress_frame(p + pos, block_len, payload, err)
+                  : lz4_decompress_frame(p + pos, block_len, payload, err);
+    if (!ok) return false;
     pos += block_len;
   }
   if (pos != len) {  // 1-3 trailing bytes = a truncated length prefix
@@ -208,8 +212,12 @@ bool IpcBlockWriter::finish_block(std::string* err) {
   if (staged_.empty()) return true;
   size_t base = out_.size();
   out_.resize(base + 4);
-  if (!lz4_compress_frame(staged_.data(), staged_.size(), &out_, err))
-    return false;
+  bool ok = codec_ == 1
+                ? zstd_compress_frame(staged_.data(), staged_.size(),
+                                      zstd_level_, &out_, err)
+                : lz4_compress_frame(staged_.data(), staged_.size(), &out_,
+                                     err);
+  if (!ok) return false;
   uint32_t block_len = (uint32_t)(out_.size() - base - 4);
   memcpy(out_.data() + base, &block_len, 4);  // u32-LE, ipc_compression.rs:87-92
   staged_.clear();
@@ -219,7 +227,8 @@ bool IpcBlockWriter::finish_block(std::string* err) {
 bool write_shuffle_files(const std::vector<HostCol>& sorted_cols,
                          const std::vector<int64_t>& part_offsets,
                          int64_t batch_size, const std::string& data_file,
-                         const std::string& index_file, std::string* err) {
+                         const std::string& index_file, std::string* err,
+                         int codec, int zstd_level) {
   size_t P = part_offsets.size() - 1;
   FILE* df = fopen(data_file.c_str(), "wb");
   if (!df) {
@@ -233,7 +242,7 @@ bool write_shuffle_files(const std::vector<HostCol>& sorted_cols,
     index[p] = pos;
     int64_t beg = part_offsets[p], end = part_offsets[p + 1];
     if (beg == end) continue;
-    IpcBlockWriter w;
+    IpcBlockWriter w(4194304, codec, zstd_level);
     for (int64_t b = beg; b < end; b += batch_size) {
       int64_t e = b + batch_size < end ? b + batch_size : end;
       payload.clear();

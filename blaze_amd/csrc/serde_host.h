@@ -26,7 +26,11 @@ void serde_write_batch(const std::vector<HostCol>& cols, int64_t row_beg,
 
 class IpcBlockWriter {
  public:
-  explicit IpcBlockWriter(size_t target = 4194304) : target_(target) {}
+  // codec 0 = lz4 frames, 1 = zstd frames (level per
+  // SPARK_IO_COMPRESSION_ZSTD_LEVEL; ipc_compression.rs:189-196)
+  explicit IpcBlockWriter(size_t target = 4194304, int codec = 0,
+                          int zstd_level = 1)
+      : target_(target), codec_(codec), zstd_level_(zstd_level) {}
   bool write_payload(const uint8_t* p, size_t len, std::string* err);
   bool finish_block(std::string* err);
   const std::vector<uint8_t>& bytes() const { return out_; }
@@ -34,6 +38,8 @@ class IpcBlockWriter {
 
  private:
   size_t target_;
+  int codec_ = 0;
+  int zstd_level_ = 1;
   std::vector<uint8_t> staged_;
   std::vector<uint8_t> out_;
 };
@@ -55,7 +61,8 @@ bool serde_read_batch(const uint8_t* p, size_t len, size_t* used,
 // decode a [u32-LE len][lz4 frame] block stream into the concatenated
 // uncompressed payload (ipc_compression.rs:64-112 read side)
 bool ipc_decode_blocks(const uint8_t* p, size_t len,
-                       std::vector<uint8_t>* payload, std::string* err);
+                       std::vector<uint8_t>* payload, std::string* err,
+                       int codec = 0);
 
 // write the shuffle data + index files: per partition, rows
 // [part_offsets[p], part_offsets[p+1]) of the partition-sorted cols, framed
@@ -63,6 +70,7 @@ bool ipc_decode_blocks(const uint8_t* p, size_t len,
 bool write_shuffle_files(const std::vector<HostCol>& sorted_cols,
                          const std::vector<int64_t>& part_offsets,
                          int64_t batch_size, const std::string& data_file,
-                         const std::string& index_file, std::string* err);
+                         const std::string& index_file, std::string* err,
+                         int codec = 0, int zstd_level = 1);
 
 }  // namespace auron

@@ -571,7 +571,40 @@ struct OracleIpcWriter {
     uint8_t* staged; /* uncompressed payload of the current block */
     size_t staged_len, staged_cap;
     size_t target;
+    int codec;      /* 0 = lz4 frames, 1 = zstd (ipc_compression.rs:189-196) */
+    int zstd_level; /* SPARK_IO_COMPRESSION_ZSTD_LEVEL, reference default 1 */
 };
+
+/* zstd simple API via dlopen (no headers in image) */
+typedef size_t (*ZSTD_compressBound_t)(size_t);
+typedef size_t (*ZSTD_compress_t)(void*, size_t, const void*, size_t, int);
+typedef unsigned long long (*ZSTD_getFrameContentLength_t)(const void*, size_t);
+typedef size_t (*ZSTD_decompress_t)(void*, size_t, const void*, size_t);
+typedef unsigned (*ZSTD_isError_t)(size_t);
+static struct {
+    void* handle;
+    ZSTD_compressBound_t bound;
+    ZSTD_compress_t compress;
+    ZSTD_getFrameContentLength_t clen;
+    ZSTD_decompress_t decompress;
+    ZSTD_isError_t iserr;
+} g_zstd;
+
+static int zstd_init(void) {
+    if (g_zstd.handle) return 1;
+    void* h = dlopen("libzstd.so.1", RTLD_NOW | RTLD_GLOBAL);
+    if (!h) h = dlopen("libzstd.so", RTLD_NOW | RTLD_GLOBAL);
+    if (!h) return 0;
+    g_zstd.handle = h;
+    g_zstd.bound = (ZSTD_compressBound_t)dlsym(h, "ZSTD_compressBound");
+    g_zstd.compress = (ZSTD_compress_t)dlsym(h, "ZSTD_compress");
+    g_zstd.clen = (ZSTD_getFrameContentLength_t)dlsym(h,
+                                                 "ZSTD_getDecompressedSize");
+    g_zstd.decompress = (ZSTD_decompress_t)dlsym(h, "ZSTD_decompress");
+    g_zstd.iserr = (ZSTD_isError_t)dlsym(h, "ZSTD_isError");
+    return g_zstd.bound && g_zstd.compress && g_zstd.clen &&
+           g_zstd.decompress && g_zstd.iserr;
+}
 
 static void buf_reserve(uint8_t** buf, size_t* cap, size_t need) {
     if (need <= *cap) return;
@@ -581,9 +614,19 @@ static void buf_reserve(uint8_t** buf, size_t* cap, size_t need) {
     *cap = ncap;
 }
 
+OracleIpcWriter* oracle_ipc_writer_new2(size_t target_block_size, int codec,
+                                        int zstd_level);
+
 OracleIpcWriter* oracle_ipc_writer_new(size_t target_block_size) {
+    return oracle_ipc_writer_new2(target_block_size, 0, 1);
+}
+
+OracleIpcWriter* oracle_ipc_writer_new2(size_t target_block_size, int codec,
+                                        int zstd_level) {
     OracleIpcWriter* w = calloc(1, sizeof(*w));
     w->target = target_block_size ? target_block_size : 4194304; /* conf default */
+    w->codec = codec;
+    w->zstd_level = zstd_level ? zstd_level : 1;
     return w;
 }
 
@@ -595,13 +638,24 @@ void oracle_ipc_writer_free(OracleIpcWriter* w) {
 }
 
 int oracle_ipc_finish_block(OracleIpcWriter* w) {
+    size_t clen, bound;
     if (w->staged_len == 0) return 0;
+    if (w->codec == 1) {
+        if (!zstd_init()) return -1;
+        bound = g_zstd.bound(w->staged_len) + 64;
+        buf_reserve(&w->out, &w->out_cap, w->out_len + 4 + bound);
+        clen = g_zstd.compress(w->out + w->out_len + 4, bound, w->staged,
+                               w->staged_len, w->zstd_level);
+        if (g_zstd.iserr(clen)) return -1;
+        goto framed;
+    }
     if (!lz4_init()) return -1;
-    size_t bound = g_lz4.compressBound(w->staged_len, NULL) + 64;
+    bound = g_lz4.compressBound(w->staged_len, NULL) + 64;
     buf_reserve(&w->out, &w->out_cap, w->out_len + 4 + bound);
-    size_t clen = g_lz4.compressFrame(w->out + w->out_len + 4, bound, w->staged,
-                                      w->staged_len, NULL);
+    clen = g_lz4.compressFrame(w->out + w->out_len + 4, bound, w->staged,
+                               w->staged_len, NULL);
     if (g_lz4.isError(clen)) return -1;
+framed:;
     uint32_t len32 = (uint32_t)clen;
     memcpy(w->out + w->out_len, &len32, 4); /* u32-LE, ipc_compression.rs:87-92 */
     w->out_len += 4 + clen;
@@ -625,6 +679,30 @@ int oracle_ipc_write_payload(OracleIpcWriter* w, const uint8_t* payload, size_t 
 size_t oracle_ipc_bytes(OracleIpcWriter* w, const uint8_t** data) {
     *data = w->out;
     return w->out_len;
+}
+
+/* zstd variant of the block-stream decoder */
+size_t oracle_ipc_decode_zstd(const uint8_t* in, size_t in_len, uint8_t* out,
+                              size_t out_cap) {
+    size_t pos = 0, out_len = 0;
+    if (!zstd_init()) return (size_t)-1;
+    while (pos + 4 <= in_len) {
+        uint32_t block_len;
+        unsigned long long need;
+        memcpy(&block_len, in + pos, 4);
+        pos += 4;
+        if (pos + block_len > in_len) return (size_t)-1;
+        need = g_zstd.clen(in + pos, block_len);
+        if (need == 0 && block_len > 0) return (size_t)-1; /* unknown size */
+        if (out && out_len + need <= out_cap) {
+            size_t rc = g_zstd.decompress(out + out_len, (size_t)need,
+                                          in + pos, block_len);
+            if (g_zstd.iserr(rc) || rc != (size_t)need) return (size_t)-1;
+        }
+        out_len += (size_t)need;
+        pos += block_len;
+    }
+    return out_len;
 }
 
 size_t oracle_ipc_decode(const uint8_t* in, size_t in_len, uint8_t* out,

@@ -240,9 +240,11 @@ def serde_batch(num_rows, cols):
 
 
 class IpcWriter:
-    def __init__(self, target=0):
-        _lib.oracle_ipc_writer_new.restype = ctypes.c_void_p
-        self._h = ctypes.c_void_p(_lib.oracle_ipc_writer_new(target))
+    def __init__(self, target=0, codec=0, zstd_level=1):
+        """codec 0 = lz4 frames, 1 = zstd (ipc_compression.rs:189-196)."""
+        _lib.oracle_ipc_writer_new2.restype = ctypes.c_void_p
+        self._h = ctypes.c_void_p(
+            _lib.oracle_ipc_writer_new2(target, codec, zstd_level))
 
     def __del__(self):
         if getattr(self, "_h", None):
@@ -263,13 +265,13 @@ class IpcWriter:
         return bytes(bytearray(ptr[i] for i in range(ln))) if ln else b""
 
 
-def ipc_decode(blob: bytes) -> bytes:
-    _lib.oracle_ipc_decode.restype = ctypes.c_size_t
-    need = _lib.oracle_ipc_decode(blob, len(blob), None, 0)
+def ipc_decode(blob: bytes, codec=0) -> bytes:
+    fn = _lib.oracle_ipc_decode_zstd if codec == 1 else _lib.oracle_ipc_decode
+    fn.restype = ctypes.c_size_t
+    need = fn(blob, len(blob), None, 0)
     assert need != ctypes.c_size_t(-1).value, "ipc decode failed"
     out = np.empty(need, dtype=np.uint8)
-    got = _lib.oracle_ipc_decode(blob, len(blob), _np_ptr(out, ctypes.c_uint8),
-                                 need)
+    got = fn(blob, len(blob), _np_ptr(out, ctypes.c_uint8), need)
     assert got == need
     return out.tobytes()
 
