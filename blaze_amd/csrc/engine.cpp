@@ -942,7 +942,6 @@ class AggOp {
   std::vector<std::pair<int64_t, std::vector<HostOutCol>>> finish() {
     std::vector<std::pair<int64_t, std::vector<HostOutCol>>> out;
     drain_timing();
-    prepare_collect();
     if (spill_.empty()) {
       emit_table(&out, false);
     } else {
@@ -995,7 +994,12 @@ class AggOp {
         }
         bool last = (b + 1 == spill_.size());
         emit_table(&out, /*exclude_specials=*/!last);
-        if (!last) reset_main();
+        if (!last) {
+          reset_main();
+          // the pool holds this bucket's merged items + the resident
+          // specials; keep only the specials for the next bucket
+          if (has_coll_) reset_collect_pool();
+        }
       }
       spill_.clear();
     }
@@ -1546,6 +1550,11 @@ class AggOp {
   // region (the two special groups stay resident across spills)
   void spill_table() {
     drain_timing();
+    // COLLECT pool drain (collect.rs spill analog): sort the pool so the
+    // freeze kernels can binary-search each group's item run — the frozen
+    // records then CARRY the items into the spill buckets; afterwards the
+    // pool is reset to the resident specials' items only
+    prepare_collect();
     DevBuf order, first;
     int64_t ng = table_order(&order, &first);
     if (ng == 0) return;
@@ -1610,7 +1619,49 @@ class AggOp {
       spill_count_++;
     }
     reset_main();
+    if (has_coll_) reset_collect_pool();
     DBG("agg.spill n=%lld buckets=%d", (long long)n, num_spill_buckets_);
+  }
+
+  // Rebuild the pool to hold exactly the two RESIDENT special groups' items
+  // (taken from the sorted views, order preserved; re-assigned prios 0..m
+  // stay below every future row/record priority) and restore the original
+  // pool arrays as the append target.
+  void reset_collect_pool() {
+    if (!has_coll_ || !coll_sorted_) return;
+    uint32_t spec[2] = {(uint32_t)t_.cap, (uint32_t)(t_.cap + 1)};
+    DevBuf d_spec(8), cnts(8), items;
+    AURON_HIP(hipMemcpyAsync(d_spec.get(), spec, 8, hipMemcpyHostToDevice,
+                             stream_));
+    launch_coll_counts(t_, d_spec.get<uint32_t>(), 2, cnts.get<int32_t>(),
+                       stream_);
+    int32_t h_cnts[2];
+    AURON_HIP(hipMemcpyAsync(h_cnts, cnts.get(), 8, hipMemcpyDeviceToHost,
+                             stream_));
+    AURON_HIP(hipStreamSynchronize(stream_));
+    int64_t m0 = h_cnts[0], m1 = h_cnts[1];
+    int32_t h_offs[3] = {0, (int32_t)m0, (int32_t)(m0 + m1)};
+    DevBuf d_offs(12);
+    AURON_HIP(hipMemcpyAsync(d_offs.get(), h_offs, 12, hipMemcpyHostToDevice,
+                             stream_));
+    items.alloc((m0 + m1) * 8 + 8);
+    launch_coll_gather(t_, d_spec.get<uint32_t>(), 2, d_offs.get<int32_t>(),
+                       items.get<unsigned long long>(), stream_);
+    // restore the ORIGINAL pool arrays as the live pool, then refill
+    t_.c_key = d_ckey_.get<long long>();
+    t_.c_prio = d_cprio_.get<unsigned long long>();
+    t_.c_val = d_cval_.get<unsigned long long>();
+    t_.c_cap = coll_cap_;
+    launch_coll_refill(t_, items.get<unsigned long long>(), m0, m1, stream_);
+    unsigned long long nn[2] = {(unsigned long long)m0,
+                                (unsigned long long)m1};
+    AURON_HIP(hipMemcpyAsync(d_cn_.get(), nn, 16, hipMemcpyHostToDevice,
+                             stream_));
+    AURON_HIP(hipStreamSynchronize(stream_));
+    coll_sorted_ = false;
+    coll_n0_ = coll_n1_ = 0;
+    d_cskey_ = DevBuf();
+    d_csval_ = DevBuf();
   }
 
   void reset_main() {
@@ -1651,9 +1702,6 @@ class AggOp {
       if (gkey_)
         FAIL("spill with Utf8/multi-column grouping keys unsupported "
              "(raise AURON_HIP_MEM_BUDGET)");
-      if (has_coll_)
-        FAIL("COLLECT_LIST with table spill unsupported (round-2: drain the "
-             "pool into the spill buckets)");
       spill_table();  // frees the main region; caller re-checks free space
     }
   }
@@ -1678,6 +1726,7 @@ class AggOp {
   // batch_size_-row output chunks
   void emit_table(std::vector<std::pair<int64_t, std::vector<HostOutCol>>>* out,
                   bool exclude_specials) {
+    prepare_collect();  // one-shot per pool state (coll_sorted_ guard)
     DevBuf order, first;
     int64_t ng = table_order(&order, &first);
     DBG("agg.finish ng=%lld excl=%d", (long long)ng, (int)exclude_specials);

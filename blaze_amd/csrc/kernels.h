@@ -140,6 +140,8 @@ void launch_agg_merge_frozen(const AggTable& t, const int64_t* keys,
 
 // merge spilled records back (a9 analog): like merge_frozen but each record
 // carries its preserved global first_row explicitly
+void launch_coll_refill(const AggTable& t, const unsigned long long* items,
+                        int64_t m0, int64_t m1, hipStream_t s);
 void launch_agg_merge_spill(const AggTable& t, const int64_t* keys,
                             const uint8_t* acc_data, const int32_t* acc_offsets,
                             const unsigned long long* first_rows, int64_t n,

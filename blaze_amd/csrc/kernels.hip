@@ -322,7 +322,44 @@ __global__ void k_agg_merge_spill(const AggTable t,
       if (acc.f_st) atomicMin(&t.f_row[2 * a], fr);
       if (acc.fn_st) atomicMin(&t.f_row[2 * a + 1], fr);
     }
+    if (t.c_key && acc.c_cnt) {  // spilled records carry the item runs;
+      for (uint32_t j = 0; j < acc.c_cnt; j++) {  // prio keeps record order
+        double v;
+        memcpy(&v, acc.c_raw + (size_t)j * 8, 8);
+        coll_append(t, key, false, (fr << 20) | j, v);
+      }
+    }
   }
+}
+
+// refill the pool with the resident special groups' gathered items
+// (engine.cpp reset_collect_pool): front = i64::MIN-key group, back =
+// null-key group, prios re-assigned 0..m (order preserved)
+__global__ void k_coll_refill(const AggTable t,
+                              const unsigned long long* __restrict__ items,
+                              int64_t m0, int64_t m1) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < m0 + m1; i += (int64_t)gridDim.x * blockDim.x) {
+    if (i < m0) {
+      t.c_key[i] = INT64_MIN;
+      t.c_prio[i] = (unsigned long long)i;
+      t.c_val[i] = items[i];
+    } else {
+      int64_t p = i - m0;
+      int64_t at = t.c_cap - 1 - p;
+      t.c_key[at] = 0;
+      t.c_prio[at] = (unsigned long long)p;
+      t.c_val[at] = items[i];
+    }
+  }
+}
+
+void launch_coll_refill(const AggTable& t, const unsigned long long* items,
+                        int64_t m0, int64_t m1, hipStream_t s) {
+  if (m0 + m1 == 0) return;
+  hipLaunchKernelGGL(k_coll_refill, dim3(grid_for(m0 + m1)), dim3(BLOCK), 0,
+                     s, t, items, m0, m1);
+  check_launch("k_coll_refill");
 }
 
 void launch_agg_merge_spill(const AggTable& t, const int64_t* keys,

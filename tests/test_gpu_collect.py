@@ -190,3 +190,74 @@ def test_collect_set_i64_null_keys_multistage():
     assert len(gl) == len(sets)
     for g, e in zip(gl, sets):
         np.testing.assert_array_equal(g, np.array(e, np.int64))
+
+
+def test_collect_spill_drain():
+    """Forced table spill with COLLECT_LIST: the pool drains into the frozen
+    spill records (chained a8 wire) and the bucket-merged output reproduces
+    every group's items in arrival order; null-key and i64::MIN-key groups
+    stay resident with their items intact."""
+    rng = np.random.default_rng(47)
+    n = 300_000
+    keys = rng.integers(0, 90_000, n).astype(np.int64)
+    keys[::5000] = -2**63
+    kv = np.ones(n, bool)
+    kv[::4999] = False
+    vals = (rng.random(n) * 100).round(3)
+    vv = rng.random(n) >= 0.1
+    conf = {"AURON_HIP_MEM_BUDGET": 1 << 20,
+            "AURON_HIP_AGG_TABLE_SLOTS": 1 << 14}
+    batches = [[(keys[i:i + 50_000], kv[i:i + 50_000]),
+                (vals[i:i + 50_000], vv[i:i + 50_000])]
+               for i in range(0, n, 50_000)]
+    t = blaze_amd.Task(plan.plan_partial_final_named(["collect_list",
+                                                      "count"]),
+                       batches=batches, conf=conf)
+    outs = t.run()
+    assert t.metric("spill_count") > 0
+    t.finalize()
+    got_keys = np.concatenate([o[0]["values"] for o in outs])
+    got_kv = np.concatenate(
+        [o[0].get("valid") if o[0].get("valid") is not None
+         else np.ones(len(o[0]["values"]), bool) for o in outs])
+    gl = _concat_lists([o[1] for o in outs])
+    got_cnt = np.concatenate([o[2]["values"] for o in outs])
+
+    ok, lists = oracle.collect_groups(keys, vals, vv, key_valid=kv)
+    assert len(got_keys) == len(ok)
+    # spill output order is per-bucket: compare as key-indexed dict, but
+    # each group's ITEM order must still be exact arrival order
+    ref = {k: lst for k, lst in zip(ok, lists)}
+    seen = set()
+    for i in range(len(got_keys)):
+        k = None if not got_kv[i] else int(got_keys[i])
+        assert k in ref and k not in seen
+        seen.add(k)
+        np.testing.assert_allclose(gl[i], np.array(ref[k]), rtol=1e-12)
+        assert got_cnt[i] == int(np.sum([1 for _ in ref[k]]))
+    assert seen == set(ref.keys())
+
+
+def test_collect_set_spill_drain():
+    """Same with COLLECT_SET: dedup happens per spill freeze AND at the
+    final merge; first-occurrence order survives the spill chain."""
+    rng = np.random.default_rng(48)
+    n = 250_000
+    keys = rng.integers(0, 80_000, n).astype(np.int64)
+    vals = rng.integers(0, 5, n).astype(np.float64)  # heavy duplication
+    conf = {"AURON_HIP_MEM_BUDGET": 1 << 20,
+            "AURON_HIP_AGG_TABLE_SLOTS": 1 << 14}
+    batches = [[(keys[i:i + 50_000], None), (vals[i:i + 50_000], None)]
+               for i in range(0, n, 50_000)]
+    t = blaze_amd.Task(plan.plan_partial_final_named(["collect_set"]),
+                       batches=batches, conf=conf)
+    outs = t.run()
+    assert t.metric("spill_count") > 0
+    t.finalize()
+    got_keys = np.concatenate([o[0]["values"] for o in outs])
+    gl = _concat_lists([o[1] for o in outs])
+    ok, sets_ = oracle.collect_groups(keys, vals, distinct=True)
+    ref = {k: lst for k, lst in zip(ok, sets_)}
+    assert len(got_keys) == len(ref)
+    for i, k in enumerate(got_keys):
+        np.testing.assert_allclose(gl[i], np.array(ref[int(k)]), rtol=1e-12)
