@@ -1389,7 +1389,7 @@ class AggOp {
                          d_scan_tmp_.get(), &tb3, stream_);
       launch_agg3_scatter(keys, kv, vals, vv, chunk, AGG3_NBUCK_LOG2,
                           AGG3_GRID_LOG2, d_scanned_.get<uint32_t>(),
-                          d_partkv_.get<uint8_t>(), stream_);
+                          d_partkv_.get<uint8_t>(), t_.error_flag, stream_);
       launch_agg3_bucket(d_partkv_.get<uint8_t>(), d_counts_.get<uint32_t>(),
                          d_scanned_.get<uint32_t>(), AGG3_GRID_LOG2,
                          val_is_int_ ? 1 : 0, AGG3_NBUCK,

@@ -91,7 +91,8 @@ __global__ void __launch_bounds__(1024) k_agg3_scatter(
     const int64_t* __restrict__ keys, const uint8_t* __restrict__ key_valid,
     const double* __restrict__ vals, const uint8_t* __restrict__ val_valid,
     int64_t n, int nbuck_log2, int grid_log2,
-    const uint32_t* __restrict__ line_scan, uint8_t* __restrict__ out) {
+    const uint32_t* __restrict__ line_scan, uint8_t* __restrict__ out,
+    uint32_t* __restrict__ err_flag) {
   const uint32_t nbuck = 1u << nbuck_log2;
   extern __shared__ uint8_t lds[];
   // layout: records [nbuck][A3_CAP][24] | cnt[nbuck] | fl[nbuck] |
@@ -121,8 +122,17 @@ __global__ void __launch_bounds__(1024) k_agg3_scatter(
     my[r] = (int64_t)blockIdx.x * blockDim.x + threadIdx.x +
             (int64_t)r * stride;
   const int64_t step = (int64_t)A3_RPT * stride;
+  // hard tile bound: rows/thread plus a generous retry allowance — a full
+  // staging bucket retries a row for a few tiles at most; exceeding this
+  // means a logic bug, and the kernel must FAIL, never hang the box
+  const int64_t max_tiles = 8 * ((n + step - 1) / step) + 256;
+  int64_t tile = 0;
 
   for (;;) {
+    if (++tile > max_tiles) {
+      atomicOr(err_flag, 32u);
+      break;
+    }
     int mine = 0;
 #pragma unroll
     for (int r = 0; r < A3_RPT; r++) {
@@ -145,8 +155,10 @@ __global__ void __launch_bounds__(1024) k_agg3_scatter(
         *(uint32_t*)(rec + 16) =
             (uint32_t)my[r] | (vvalid ? 0x80000000u : 0u);
         *(uint32_t*)(rec + 20) = 0;
-        if ((pos & (A3_QUANT - 1)) == A3_QUANT - 1)  // crossed a packet
-          queue[atomicAdd(qn, 1u) & (nbuck + 63)] = (uint16_t)b;
+        if ((pos & (A3_QUANT - 1)) == A3_QUANT - 1) {  // crossed a packet
+          uint32_t qi = atomicAdd(qn, 1u);
+          if (qi < nbuck + 64) queue[qi] = (uint16_t)b;  // <=1 push per
+        }                                                // bucket per tile
         my[r] += step;
       } else {
         atomicSub(&cnt[b], 1u);  // staging full: retry this row next tile
@@ -155,6 +167,7 @@ __global__ void __launch_bounds__(1024) k_agg3_scatter(
     // block-wide termination check doubles as the pre-flush barrier
     int live = __syncthreads_count(mine);
     uint32_t nq = *qn;
+    if (nq > nbuck + 64) nq = nbuck + 64;
     if (live == 0) {
       // drain: flush every bucket's remainder as (possibly partial) packets
       for (uint32_t b = wave; b < nbuck; b += nwave) {
@@ -170,7 +183,7 @@ __global__ void __launch_bounds__(1024) k_agg3_scatter(
     }
     // flush only queued (full-packet) buckets; duplicates flush to 0 extra
     for (uint32_t i = wave; i < nq; i += nwave) {
-      uint32_t b = queue[i & (nbuck + 63)];
+      uint32_t b = queue[i];
       uint32_t c = cnt[b];
       uint32_t nfl = c & ~(uint32_t)(A3_QUANT - 1);
       if (!nfl) continue;
@@ -197,9 +210,10 @@ void launch_agg3_scatter(const int64_t* keys, const uint8_t* key_valid,
                          const double* vals, const uint8_t* val_valid,
                          int64_t n, int nbuck_log2, int grid_log2,
                          const uint32_t* line_scan, uint8_t* out,
-                         hipStream_t s) {
+                         uint32_t* err_flag, hipStream_t s) {
   const uint32_t nbuck = 1u << nbuck_log2;
-  size_t lds = (size_t)nbuck * A3_CAP * 24 + (size_t)nbuck * 12;
+  size_t lds = (size_t)nbuck * A3_CAP * 24 + (size_t)nbuck * 12 +
+               ((size_t)nbuck + 64) * 2 + 8;  // + dirty queue + counter
   if (lds > 160 * 1024)
     throw std::runtime_error("agg3 scatter LDS over 160KB");
   hipError_t e = hipFuncSetAttribute(
@@ -209,7 +223,7 @@ void launch_agg3_scatter(const int64_t* keys, const uint8_t* key_valid,
     throw std::runtime_error("agg3 scatter LDS attribute failed");
   hipLaunchKernelGGL(k_agg3_scatter, dim3(1 << grid_log2), dim3(1024), lds, s,
                      keys, key_valid, vals, val_valid, n, nbuck_log2,
-                     grid_log2, line_scan, out);
+                     grid_log2, line_scan, out, err_flag);
   check_launch3("k_agg3_scatter");
 }
 
