@@ -77,8 +77,17 @@ void launch_agg3_line_sizes(const uint32_t* counts, int64_t n,
 }
 
 // ---- v3 scatter ------------------------------------------------------------
-static constexpr int A3_CAP = 12;      // staged records per bucket
+static constexpr int A3_CAP = 10;      // staged records per bucket
 static constexpr int A3_QUANT = 8;     // records per flushed packet (192 B)
+// staging slot stride: 28 B (7 dwords, odd) — the packed 24 B stride put
+// every record on one of FOUR of the 32 LDS banks (288 B bucket stride ==
+// 72 dwords == 8 mod 32); the pad restores full bank spread
+static constexpr int A3_SLOT = 28;
+// a thread that cannot stage its row after a few tiles (hot bucket under
+// key skew: drain is bounded by CAP per tile) bypasses to the leftover
+// list; the single-phase leftover kernel handles it — correct and bounded
+// for ANY distribution, perf-relevant only under adversarial skew
+static constexpr int A3_RETRY_BYPASS = 4;
 
 // A3_RPT rows per thread per tile; a dirty QUEUE records buckets that
 // crossed a packet boundary so the flush phase touches only those (the v3.0
@@ -92,13 +101,14 @@ __global__ void __launch_bounds__(1024) k_agg3_scatter(
     const double* __restrict__ vals, const uint8_t* __restrict__ val_valid,
     int64_t n, int nbuck_log2, int grid_log2,
     const uint32_t* __restrict__ line_scan, uint8_t* __restrict__ out,
+    PartRow* __restrict__ leftover, unsigned long long* __restrict__ lo_n,
     uint32_t* __restrict__ err_flag) {
   const uint32_t nbuck = 1u << nbuck_log2;
   extern __shared__ uint8_t lds[];
   // layout: records [nbuck][A3_CAP][24] | cnt[nbuck] | fl[nbuck] |
   //         base_line[nbuck] | dirty queue [nbuck] u16-as-u32 | qn
   uint8_t* stage = lds;
-  uint32_t* cnt = (uint32_t*)(lds + (size_t)nbuck * A3_CAP * 24);
+  uint32_t* cnt = (uint32_t*)(lds + (size_t)nbuck * A3_CAP * A3_SLOT);
   uint32_t* fl = cnt + nbuck;
   uint32_t* base_line = fl + nbuck;
   uint16_t* queue = (uint16_t*)(base_line + nbuck);
@@ -117,6 +127,7 @@ __global__ void __launch_bounds__(1024) k_agg3_scatter(
   const int wave = (int)(threadIdx.x >> 6);
   const int lane = (int)(threadIdx.x & 63);
   int64_t my[A3_RPT];
+  int tries[A3_RPT] = {};
 #pragma unroll
   for (int r = 0; r < A3_RPT; r++)
     my[r] = (int64_t)blockIdx.x * blockDim.x + threadIdx.x +
@@ -149,19 +160,29 @@ __global__ void __launch_bounds__(1024) k_agg3_scatter(
       uint32_t pos = atomicAdd(&cnt[b], 1u);
       if (pos < A3_CAP) {
         bool vvalid = !val_valid || bit_get3(val_valid, my[r]);
-        uint8_t* rec = stage + ((size_t)b * A3_CAP + pos) * 24;
+        uint8_t* rec = stage + ((size_t)b * A3_CAP + pos) * A3_SLOT;
         *(int64_t*)rec = k;
         *(double*)(rec + 8) = vals[my[r]];
         *(uint32_t*)(rec + 16) =
             (uint32_t)my[r] | (vvalid ? 0x80000000u : 0u);
-        *(uint32_t*)(rec + 20) = 0;
         if ((pos & (A3_QUANT - 1)) == A3_QUANT - 1) {  // crossed a packet
           uint32_t qi = atomicAdd(qn, 1u);
           if (qi < nbuck + 64) queue[qi] = (uint16_t)b;  // <=1 push per
         }                                                // bucket per tile
         my[r] += step;
+        tries[r] = 0;
       } else {
-        atomicSub(&cnt[b], 1u);  // staging full: retry this row next tile
+        atomicSub(&cnt[b], 1u);
+        if (++tries[r] >= A3_RETRY_BYPASS) {
+          // hot bucket (skew): route the row to the leftover list
+          bool vvalid = !val_valid || bit_get3(val_valid, my[r]);
+          unsigned long long p = atomicAdd(lo_n, 1ull);
+          leftover[p] = PartRow{
+              k, vals[my[r]],
+              (uint32_t)my[r] | (vvalid ? 0x80000000u : 0u), 0};
+          my[r] += step;
+          tries[r] = 0;
+        }  // else: retry next tile
       }
     }
     // block-wide termination check doubles as the pre-flush barrier
@@ -174,9 +195,10 @@ __global__ void __launch_bounds__(1024) k_agg3_scatter(
         uint32_t c = cnt[b];
         if (!c) continue;
         uint8_t* dst = out + ((size_t)base_line[b] << 6) + (size_t)fl[b] * 24;
-        const uint8_t* src = stage + (size_t)b * A3_CAP * 24;
+        const uint8_t* src = stage + (size_t)b * A3_CAP * A3_SLOT;
         for (uint32_t d = lane; d < c * 6; d += 64)
-          ((uint32_t*)dst)[d] = ((const uint32_t*)src)[d];
+          ((uint32_t*)dst)[d] =
+              *(const uint32_t*)(src + (d / 6) * A3_SLOT + (d % 6) * 4);
       }
       __syncthreads();
       break;
@@ -188,12 +210,14 @@ __global__ void __launch_bounds__(1024) k_agg3_scatter(
       uint32_t nfl = c & ~(uint32_t)(A3_QUANT - 1);
       if (!nfl) continue;
       uint8_t* dst = out + ((size_t)base_line[b] << 6) + (size_t)fl[b] * 24;
-      const uint8_t* src = stage + (size_t)b * A3_CAP * 24;
+      uint8_t* src = stage + (size_t)b * A3_CAP * A3_SLOT;
       for (uint32_t d = lane; d < nfl * 6; d += 64)
-        ((uint32_t*)dst)[d] = ((const uint32_t*)src)[d];
+        ((uint32_t*)dst)[d] =
+            *(const uint32_t*)(src + (d / 6) * A3_SLOT + (d % 6) * 4);
       uint32_t rem = c - nfl;
-      for (uint32_t d = lane; d < rem * 6; d += 64)
-        ((uint32_t*)src)[d] = ((const uint32_t*)(src + (size_t)nfl * 24))[d];
+      for (uint32_t d = lane; d < rem * 7; d += 64)
+        *(uint32_t*)(src + (d / 7) * A3_SLOT + (d % 7) * 4) =
+            *(const uint32_t*)(src + (nfl + d / 7) * A3_SLOT + (d % 7) * 4);
       if (lane == 0) {
         fl[b] += nfl;
         cnt[b] = rem;
@@ -210,9 +234,10 @@ void launch_agg3_scatter(const int64_t* keys, const uint8_t* key_valid,
                          const double* vals, const uint8_t* val_valid,
                          int64_t n, int nbuck_log2, int grid_log2,
                          const uint32_t* line_scan, uint8_t* out,
+                         PartRow* leftover, unsigned long long* lo_n,
                          uint32_t* err_flag, hipStream_t s) {
   const uint32_t nbuck = 1u << nbuck_log2;
-  size_t lds = (size_t)nbuck * A3_CAP * 24 + (size_t)nbuck * 12 +
+  size_t lds = (size_t)nbuck * A3_CAP * A3_SLOT + (size_t)nbuck * 12 +
                ((size_t)nbuck + 64) * 2 + 8;  // + dirty queue + counter
   if (lds > 160 * 1024)
     throw std::runtime_error("agg3 scatter LDS over 160KB");
@@ -223,7 +248,7 @@ void launch_agg3_scatter(const int64_t* keys, const uint8_t* key_valid,
     throw std::runtime_error("agg3 scatter LDS attribute failed");
   hipLaunchKernelGGL(k_agg3_scatter, dim3(1 << grid_log2), dim3(1024), lds, s,
                      keys, key_valid, vals, val_valid, n, nbuck_log2,
-                     grid_log2, line_scan, out, err_flag);
+                     grid_log2, line_scan, out, leftover, lo_n, err_flag);
   check_launch3("k_agg3_scatter");
 }
 
