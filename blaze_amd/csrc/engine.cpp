@@ -1412,6 +1412,7 @@ class AggOp {
                             d_ppart_[pb].get<uint8_t>(),
                             d_pleft_[pb].get<PartRow>(),
                             d_pctr_[pb].get<unsigned long long>() + 1,
+                            d_plinesz_[pb].get<uint32_t>(),
                             t_.error_flag, s_aux_);
         AURON_HIP(hipEventRecord(ev_scatter_[pb], s_aux_));
       }
@@ -1458,6 +1459,7 @@ class AggOp {
         AURON_HIP(hipStreamWaitEvent(stream_, ev_scatter_[pb], 0));
         launch_agg3_bucket(d_ppart_[pb].get<uint8_t>(),
                            d_pcounts_[pb].get<uint32_t>(),
+                           d_plinesz_[pb].get<uint32_t>(),
                            d_pscanned_[pb].get<uint32_t>(), AGG3_GRID_LOG2,
                            val_is_int_ ? 1 : 0, nbuck,
                            d_staged_.get<StagedGroup>(),
@@ -1575,8 +1577,10 @@ class AggOp {
                           AGG3_GRID_LOG2, d_scanned_.get<uint32_t>(),
                           d_partkv_.get<uint8_t>(), d_leftover_.get<PartRow>(),
                           d_counters_.get<unsigned long long>() + 1,
+                          d_linesz_.get<uint32_t>(),  // free after the scan
                           t_.error_flag, stream_);
       launch_agg3_bucket(d_partkv_.get<uint8_t>(), d_counts_.get<uint32_t>(),
+                         d_linesz_.get<uint32_t>(),
                          d_scanned_.get<uint32_t>(), AGG3_GRID_LOG2,
                          val_is_int_ ? 1 : 0, AGG3_NBUCK,
                          d_staged_.get<StagedGroup>(),

@@ -102,7 +102,7 @@ __global__ void __launch_bounds__(1024) k_agg3_scatter(
     int64_t n, int nbuck_log2, int grid_log2,
     const uint32_t* __restrict__ line_scan, uint8_t* __restrict__ out,
     PartRow* __restrict__ leftover, unsigned long long* __restrict__ lo_n,
-    uint32_t* __restrict__ err_flag) {
+    uint32_t* __restrict__ bypass_matrix, uint32_t* __restrict__ err_flag) {
   const uint32_t nbuck = 1u << nbuck_log2;
   extern __shared__ uint8_t lds[];
   // layout: records [nbuck][A3_CAP][24] | cnt[nbuck] | fl[nbuck] |
@@ -113,10 +113,12 @@ __global__ void __launch_bounds__(1024) k_agg3_scatter(
   uint32_t* base_line = fl + nbuck;
   uint16_t* queue = (uint16_t*)(base_line + nbuck);
   uint32_t* qn = (uint32_t*)(queue + nbuck + 64);
+  uint32_t* byp = qn + 1;  // rows bypassed to the leftover list, per bucket
 
   for (uint32_t b = threadIdx.x; b < nbuck; b += blockDim.x) {
     cnt[b] = 0;
     fl[b] = 0;
+    byp[b] = 0;
     base_line[b] = line_scan[((size_t)b << grid_log2) | blockIdx.x];
   }
   if (threadIdx.x == 0) *qn = 0;
@@ -174,12 +176,15 @@ __global__ void __launch_bounds__(1024) k_agg3_scatter(
       } else {
         atomicSub(&cnt[b], 1u);
         if (++tries[r] >= A3_RETRY_BYPASS) {
-          // hot bucket (skew): route the row to the leftover list
+          // hot bucket (skew): route the row to the leftover list; the
+          // histogram already counted it, so record the correction the
+          // bucket kernel must subtract from this (block,bucket) range
           bool vvalid = !val_valid || bit_get3(val_valid, my[r]);
           unsigned long long p = atomicAdd(lo_n, 1ull);
           leftover[p] = PartRow{
               k, vals[my[r]],
               (uint32_t)my[r] | (vvalid ? 0x80000000u : 0u), 0};
+          atomicAdd(&byp[b], 1u);
           my[r] += step;
           tries[r] = 0;
         }  // else: retry next tile
@@ -228,6 +233,11 @@ __global__ void __launch_bounds__(1024) k_agg3_scatter(
     if (threadIdx.x == 0) *qn = 0;
     __syncthreads();
   }
+  // every (bucket, block) entry is written by its own block — on EVERY exit
+  // path (incl. the tile-bound failure), so the bucket kernel never reads a
+  // stale matrix
+  for (uint32_t b = threadIdx.x; b < nbuck; b += blockDim.x)
+    bypass_matrix[((size_t)b << grid_log2) | blockIdx.x] = byp[b];
 }
 
 void launch_agg3_scatter(const int64_t* keys, const uint8_t* key_valid,
@@ -235,10 +245,11 @@ void launch_agg3_scatter(const int64_t* keys, const uint8_t* key_valid,
                          int64_t n, int nbuck_log2, int grid_log2,
                          const uint32_t* line_scan, uint8_t* out,
                          PartRow* leftover, unsigned long long* lo_n,
-                         uint32_t* err_flag, hipStream_t s) {
+                         uint32_t* bypass_matrix, uint32_t* err_flag,
+                         hipStream_t s) {
   const uint32_t nbuck = 1u << nbuck_log2;
-  size_t lds = (size_t)nbuck * A3_CAP * A3_SLOT + (size_t)nbuck * 12 +
-               ((size_t)nbuck + 64) * 2 + 8;  // + dirty queue + counter
+  size_t lds = (size_t)nbuck * A3_CAP * A3_SLOT + (size_t)nbuck * 16 +
+               ((size_t)nbuck + 64) * 2 + 8;  // + queue + counters + bypass
   if (lds > 160 * 1024)
     throw std::runtime_error("agg3 scatter LDS over 160KB");
   hipError_t e = hipFuncSetAttribute(
@@ -248,7 +259,8 @@ void launch_agg3_scatter(const int64_t* keys, const uint8_t* key_valid,
     throw std::runtime_error("agg3 scatter LDS attribute failed");
   hipLaunchKernelGGL(k_agg3_scatter, dim3(1 << grid_log2), dim3(1024), lds, s,
                      keys, key_valid, vals, val_valid, n, nbuck_log2,
-                     grid_log2, line_scan, out, leftover, lo_n, err_flag);
+                     grid_log2, line_scan, out, leftover, lo_n,
+                     bypass_matrix, err_flag);
   check_launch3("k_agg3_scatter");
 }
 
@@ -258,6 +270,7 @@ static constexpr int L3PROBE = 128;
 
 __global__ void __launch_bounds__(1024) k_agg3_bucket(
     const uint8_t* __restrict__ part, const uint32_t* __restrict__ counts,
+    const uint32_t* __restrict__ bypass,
     const uint32_t* __restrict__ line_scan, int grid_log2, int is_int,
     int nbuckets, StagedGroup* __restrict__ staged,
     unsigned long long* __restrict__ staged_n, int64_t staged_cap,
@@ -282,7 +295,8 @@ __global__ void __launch_bounds__(1024) k_agg3_bucket(
     if (threadIdx.x == 0) {
       uint32_t acc = 0;
       for (int k = 0; k < nrange; k++) {
-        acc += counts[((size_t)b << grid_log2) | k];
+        size_t e = ((size_t)b << grid_log2) | k;
+        acc += counts[e] - bypass[e];  // bypassed rows were never written
         cum[k] = acc;
       }
     }
@@ -372,13 +386,15 @@ __global__ void __launch_bounds__(1024) k_agg3_bucket(
 }
 
 void launch_agg3_bucket(const uint8_t* part, const uint32_t* counts,
+                        const uint32_t* bypass,
                         const uint32_t* line_scan, int grid_log2, int is_int,
                         int nbuckets, StagedGroup* staged,
                         unsigned long long* staged_n, int64_t staged_cap,
                         PartRow* leftover, unsigned long long* lo_n,
                         uint32_t* error_flag, hipStream_t s) {
   hipLaunchKernelGGL(k_agg3_bucket, dim3(nbuckets), dim3(1024), 0, s, part,
-                     counts, line_scan, grid_log2, is_int, nbuckets, staged,
+                     counts, bypass, line_scan, grid_log2, is_int, nbuckets,
+                     staged,
                      staged_n, staged_cap, leftover, lo_n, error_flag);
   check_launch3("k_agg3_bucket");
 }
