@@ -635,7 +635,7 @@ class AggOp {
       if (!gkey_ && !merge_mode_ && !has_mm_ && !has_first_ &&
           !has_coll_ && b.num_rows - done >= AGG2_MIN_CHUNK) {
         init_agg2_conf();
-        if (agg2_v3_ && !skip_enabled_ &&
+        if (agg2_v3_ && agg2_pipe_ && !skip_enabled_ &&
             b.num_rows - done >= 2 * agg2_chunk_max_) {
           // multi-chunk batch: cross-chunk pipeline — scatter(k+1) on the
           // aux stream overlaps bucket/merge(k) on the engine stream
@@ -1341,6 +1341,12 @@ class AggOp {
     // evidence trail. AURON_AGG2_V3=0 falls back to the v2 kernels.
     const char* v3 = getenv("AURON_AGG2_V3");
     agg2_v3_ = !(v3 && v3[0] == '0');
+    // cross-chunk pipeline: OFF by default — the scatter (151 KB LDS/WG)
+    // and bucket (100 KB) kernels cannot co-reside on a CU, so overlapping
+    // them only splits the chip and adds sync bubbles (measured 60.5 vs
+    // 51 ms/step); kept behind AURON_AGG2_PIPE=1 for experiments
+    const char* pe = getenv("AURON_AGG2_PIPE");
+    agg2_pipe_ = (pe && pe[0] == '1');
   }
 
   // Cross-chunk pipelined two-phase aggregation (v3 kernels): the scatter
@@ -2271,6 +2277,7 @@ class AggOp {
   // two-phase scratch (allocated on first large chunk)
   bool agg2_split_ = false;
   bool agg2_v3_ = true;
+  bool agg2_pipe_ = false;
   bool agg2_conf_read_ = false;
   int agg2_grid_log2_ = 9;
   int agg2_block_ = 1024;

@@ -437,6 +437,93 @@ void launch_gkey_out_bytes(const GKeyTable& g, const uint32_t* order_slots,
                            const int32_t* out_offsets, uint8_t* out_data,
                            hipStream_t s);
 
+// ---- multi-argument aggregation (per-agg arg columns / acc types) --------
+// The reference evaluates INDEPENDENT argument expressions per aggregate
+// (agg.rs:73-169 prepare_partial_args) with per-agg accumulator columns of
+// the declared type (sum.rs:78-88, maxmin.rs:81-83). This mode drops the
+// shared-argument restriction: each agg owns an accumulator bank column
+// (acc/meta/st arrays indexed by slot) and reads its own argument column
+// (or a NULL literal, collect.rs-style empty behavior). Single-phase only.
+constexpr int MA_MAX_AGGS = 12;
+constexpr int MA_MAX_POOLS = 4;
+struct MaAgg {
+  uint8_t kind = 0;    // AGGL_* code
+  uint8_t acc_t = 0;   // 0 = f64, 1 = i64, 2 = i32 (stored widened to i64;
+                       //     freeze/emit narrow to 4 bytes)
+  uint8_t arg_dt = 0;  // 0 = f64, 1 = i64, 2 = i32, 3 = NULL literal
+  uint8_t pool = 0xFF; // collect pool index (0..MA_MAX_POOLS) or 0xFF
+};
+struct MaDesc {
+  int n = 0;
+  MaAgg a[MA_MAX_AGGS];
+};
+struct MaArgs {  // per-chunk argument column pointers (null = NULL literal)
+  const void* vals[MA_MAX_AGGS] = {};
+  const uint8_t* valid[MA_MAX_AGGS] = {};
+};
+struct MaAcc {   // accumulator bank: [agg][cap + 2] each
+  unsigned long long* acc = nullptr;   // sum bits / omap value / first value
+  unsigned long long* meta = nullptr;  // count / first-priority
+  uint8_t* st = nullptr;               // FIRST state 0/1/2
+  int64_t stride = 0;                  // cap + 2
+};
+struct MaPools {  // per-collect-agg pools (single-key i64 mode only)
+  long long* key[MA_MAX_POOLS] = {};
+  unsigned long long* prio[MA_MAX_POOLS] = {};
+  unsigned long long* val[MA_MAX_POOLS] = {};
+  unsigned long long* n = nullptr;  // [MA_MAX_POOLS][2] counters
+  int64_t cap = 0;                  // per pool
+};
+// probe-or-insert only: slot index per row + first_row min (the accumulate
+// runs slot-indexed afterwards)
+void launch_slots_upsert(const AggTable& t, const int64_t* keys,
+                         const uint8_t* key_valid, int64_t n,
+                         uint64_t row_offset, uint32_t* slots, hipStream_t s);
+void launch_ma_init(const MaDesc& d, const MaAcc& m, int64_t cap2,
+                    hipStream_t s);
+void launch_ma_update(const AggTable& t, const MaDesc& d, const MaAcc& m,
+                      const MaPools& p, const MaArgs& args,
+                      const int64_t* keys, const uint8_t* key_valid,
+                      const uint32_t* slots, int64_t n, uint64_t row_offset,
+                      hipStream_t s);
+void launch_ma_first_capture(const AggTable& t, const MaDesc& d,
+                             const MaAcc& m, const MaArgs& args,
+                             const uint32_t* slots, int64_t n,
+                             uint64_t row_offset, hipStream_t s);
+void launch_ma_merge_frozen(const AggTable& t, const MaDesc& d,
+                            const MaAcc& m, const MaPools& p,
+                            const int64_t* keys, const uint8_t* key_valid,
+                            const uint32_t* slots,
+                            const uint8_t* acc_data,
+                            const int32_t* acc_offsets, int64_t n,
+                            uint64_t row_offset, hipStream_t s);
+void launch_ma_first_capture_frozen(const AggTable& t, const MaDesc& d,
+                                    const MaAcc& m,
+                                    const uint32_t* slots,
+                                    const uint8_t* acc_data,
+                                    const int32_t* acc_offsets, int64_t n,
+                                    uint64_t row_offset, hipStream_t s);
+// freeze: per-record lens then bytes (sorted pool views per collect agg are
+// installed into `p` by the engine before freezing)
+void launch_ma_freeze_len(const AggTable& t, const MaDesc& d, const MaAcc& m,
+                          const MaPools& p, const uint32_t* order_slots,
+                          int64_t n, int32_t* lens, hipStream_t s);
+void launch_ma_freeze_write(const AggTable& t, const MaDesc& d,
+                            const MaAcc& m, const MaPools& p,
+                            const uint32_t* order_slots, int64_t n,
+                            const int32_t* offsets, uint8_t* data,
+                            hipStream_t s);
+// final-output gather for agg j: values (width 8, or 4 for acc_t i32) +
+// validity bitmap; collect aggs use launch_coll_counts/gather with the pool
+void launch_ma_gather_out(const MaDesc& d, const MaAcc& m, int agg,
+                          const uint32_t* order_slots, int64_t n,
+                          uint8_t* values, uint8_t* valid_bitmap,
+                          hipStream_t s);
+// growth: re-probe occupied src slots into dst, carrying the bank columns
+void launch_ma_rebuild(const AggTable& dst, const MaAcc& dm,
+                       const AggTable& src, const MaAcc& sm, int naggs,
+                       hipStream_t s);
+
 // parquet RLE/bit-packed run expansion (runs = PqRun[] from parquet.h)
 void launch_runs_expand_u32(const void* runs, int nruns, const uint8_t* bytes,
                             int64_t n, uint32_t* out, hipStream_t s);
