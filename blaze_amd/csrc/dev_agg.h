@@ -116,6 +116,17 @@ __device__ __forceinline__ uint64_t val_omap(double v, bool is_int) {
   if (is_int) return b ^ 0x8000000000000000ull;
   return (b >> 63) ? ~b : (b | 0x8000000000000000ull);
 }
+// bit-domain forms (value already as raw 8-byte bits)
+__device__ __forceinline__ uint64_t val_omap_bits(uint64_t b, bool is_int) {
+  if (is_int) return b ^ 0x8000000000000000ull;
+  return (b >> 63) ? ~b : (b | 0x8000000000000000ull);
+}
+__device__ __forceinline__ uint64_t val_omap_inv_bits(uint64_t u,
+                                                      bool is_int) {
+  return is_int ? (u ^ 0x8000000000000000ull)
+                : ((u >> 63) ? (u & 0x7FFFFFFFFFFFFFFFull) : ~u);
+}
+
 __device__ __forceinline__ double val_omap_inv(uint64_t u, bool is_int) {
   uint64_t b = is_int ? (u ^ 0x8000000000000000ull)
                       : ((u >> 63) ? (u & 0x7FFFFFFFFFFFFFFFull) : ~u);

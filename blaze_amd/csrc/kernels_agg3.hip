@@ -77,12 +77,12 @@ void launch_agg3_line_sizes(const uint32_t* counts, int64_t n,
 }
 
 // ---- v3 scatter ------------------------------------------------------------
-static constexpr int A3_CAP = 10;      // staged records per bucket
+static constexpr int A3_CAP = 12;      // staged records per bucket
 static constexpr int A3_QUANT = 8;     // records per flushed packet (192 B)
-// staging slot stride: 28 B (7 dwords, odd) — the packed 24 B stride put
-// every record on one of FOUR of the 32 LDS banks (288 B bucket stride ==
-// 72 dwords == 8 mod 32); the pad restores full bank spread
-static constexpr int A3_SLOT = 28;
+// staging slot stride: 24 B packed (a 28 B bank-spread pad was measured
+// SLOWER — 7.71 vs 6.58 ms/chunk — the kernel is flush-latency-bound, not
+// LDS-bank-bound)
+static constexpr int A3_SLOT = 24;
 // a thread that cannot stage its row after a few tiles (hot bucket under
 // key skew: drain is bounded by CAP per tile) bypasses to the leftover
 // list; the single-phase leftover kernel handles it — correct and bounded
@@ -202,30 +202,35 @@ __global__ void __launch_bounds__(1024) k_agg3_scatter(
         uint8_t* dst = out + ((size_t)base_line[b] << 6) + (size_t)fl[b] * 24;
         const uint8_t* src = stage + (size_t)b * A3_CAP * A3_SLOT;
         for (uint32_t d = lane; d < c * 6; d += 64)
-          ((uint32_t*)dst)[d] =
-              *(const uint32_t*)(src + (d / 6) * A3_SLOT + (d % 6) * 4);
+          ((uint32_t*)dst)[d] = ((const uint32_t*)src)[d];
       }
       __syncthreads();
       break;
     }
-    // flush only queued (full-packet) buckets; duplicates flush to 0 extra
-    for (uint32_t i = wave; i < nq; i += nwave) {
-      uint32_t b = queue[i];
-      uint32_t c = cnt[b];
-      uint32_t nfl = c & ~(uint32_t)(A3_QUANT - 1);
-      if (!nfl) continue;
-      uint8_t* dst = out + ((size_t)base_line[b] << 6) + (size_t)fl[b] * 24;
-      uint8_t* src = stage + (size_t)b * A3_CAP * A3_SLOT;
-      for (uint32_t d = lane; d < nfl * 6; d += 64)
-        ((uint32_t*)dst)[d] =
-            *(const uint32_t*)(src + (d / 6) * A3_SLOT + (d % 6) * 4);
-      uint32_t rem = c - nfl;
-      for (uint32_t d = lane; d < rem * 7; d += 64)
-        *(uint32_t*)(src + (d / 7) * A3_SLOT + (d % 7) * 4) =
-            *(const uint32_t*)(src + (nfl + d / 7) * A3_SLOT + (d % 7) * 4);
-      if (lane == 0) {
-        fl[b] += nfl;
-        cnt[b] = rem;
+    // flush only queued (full-packet) buckets. Each entry is handled by an
+    // 8-LANE subgroup (8 independent entries in flight per wave) — a
+    // whole-wave-per-entry loop was a serial ~10-LDS-op latency chain per
+    // entry and dominated the kernel.
+    {
+      const int sg = lane >> 3;        // subgroup 0..7 within the wave
+      const int sl = lane & 7;         // lane within subgroup
+      for (uint32_t i = wave * 8 + sg; i < nq; i += nwave * 8) {
+        uint32_t b = queue[i];
+        uint32_t c = cnt[b];
+        uint32_t nfl = c & ~(uint32_t)(A3_QUANT - 1);
+        if (!nfl) continue;
+        uint8_t* dst = out + ((size_t)base_line[b] << 6) + (size_t)fl[b] * 24;
+        uint8_t* src = stage + (size_t)b * A3_CAP * A3_SLOT;
+        for (uint32_t d = sl; d < nfl * 6; d += 8)
+          ((uint32_t*)dst)[d] = ((const uint32_t*)src)[d];
+        uint32_t rem = c - nfl;
+        for (uint32_t d = sl; d < rem * 6; d += 8)
+          ((uint32_t*)src)[d] =
+              ((const uint32_t*)(src + (size_t)nfl * A3_SLOT))[d];
+        if (sl == 0) {
+          fl[b] += nfl;
+          cnt[b] = rem;
+        }
       }
     }
     // queue ENTRIES below nq are no longer needed; resetting the counter

@@ -265,6 +265,30 @@ __device__ __forceinline__ int ma_prim_w(const MaAgg& a) {
   return a.acc_t == 2 ? 4 : 8;
 }
 
+// group's item run in a SORTED pool view (installed by the engine before
+// freezing: front = {key asc, prio asc} over [0,n0), null-key group at
+// [n0, n0+n1) — the prepare_collect convention)
+__device__ __forceinline__ uint32_t ma_pool_run(
+    const long long* __restrict__ keys, unsigned long long n0,
+    unsigned long long n1, long long key, bool knull, int64_t* beg) {
+  if (knull) {
+    *beg = (int64_t)n0;
+    return (uint32_t)n1;
+  }
+  int64_t lo = 0, hi = (int64_t)n0;
+  while (lo < hi) {
+    int64_t mid = (lo + hi) >> 1;
+    if (keys[mid] < key) lo = mid + 1; else hi = mid;
+  }
+  int64_t lo2 = lo, hi2 = (int64_t)n0;
+  while (lo2 < hi2) {
+    int64_t mid = (lo2 + hi2) >> 1;
+    if (keys[mid] <= key) lo2 = mid + 1; else hi2 = mid;
+  }
+  *beg = lo;
+  return (uint32_t)(lo2 - lo);
+}
+
 // per-agg frozen-part length for group slot `a`
 __device__ __forceinline__ int ma_part_len(const MaDesc& d, const MaAcc& m,
                                            const MaPools& p, int j,
@@ -291,9 +315,10 @@ __device__ __forceinline__ int ma_part_len(const MaDesc& d, const MaAcc& m,
     case AGGL_CLIST:
     case AGGL_CSET: {
       if (ag.pool >= MA_MAX_POOLS) return varint_len_dev(0);
-      uint32_t cnt = coll_run_count(p.key[ag.pool], p.n[2 * ag.pool],
-                                    p.n[2 * ag.pool + 1], p.cap,
-                                    key_for_pool, knull);
+      int64_t beg;
+      uint32_t cnt = ma_pool_run(p.key[ag.pool], p.n[2 * ag.pool],
+                                 p.n[2 * ag.pool + 1], key_for_pool, knull,
+                                 &beg);
       uint64_t raw = (uint64_t)cnt * ma_prim_w(ag);
       return varint_len_dev(raw) + (int)raw;
     }
@@ -308,8 +333,8 @@ __global__ void k_ma_freeze_len(const AggTable t, const MaDesc d,
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     int64_t a = order_slots[i];
-    int64_t key = t.slots[a].key;
     bool knull = a == t.cap + 1;
+    int64_t key = (a == t.cap) ? INT64_MIN : t.slots[a].key;
     int len = 0;
     for (int j = 0; j < d.n; j++)
       len += ma_part_len(d, m, p, j, a, key, knull);
@@ -355,8 +380,8 @@ __global__ void k_ma_freeze_write(const AggTable t, const MaDesc d,
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     int64_t a = order_slots[i];
-    int64_t key = t.slots[a].key;
     bool knull = a == t.cap + 1;
+    int64_t key = (a == t.cap) ? INT64_MIN : t.slots[a].key;
     uint8_t* q = data + offsets[i];
     for (int j = 0; j < d.n; j++) {
       const MaAgg& ag = d.a[j];
@@ -392,16 +417,12 @@ __global__ void k_ma_freeze_write(const AggTable t, const MaDesc d,
         case AGGL_CSET: {
           uint32_t cnt = 0;
           int64_t beg = 0;
-          bool back = false;
           if (ag.pool < MA_MAX_POOLS)
-            cnt = coll_run_find(p.key[ag.pool], p.n[2 * ag.pool],
-                                p.n[2 * ag.pool + 1], p.cap, key, knull,
-                                &beg, &back);
+            cnt = ma_pool_run(p.key[ag.pool], p.n[2 * ag.pool],
+                              p.n[2 * ag.pool + 1], key, knull, &beg);
           q = ma_write_varint(q, (uint64_t)cnt * w);
           for (uint32_t c = 0; c < cnt; c++) {
-            unsigned long long vb =
-                back ? p.val[ag.pool][p.cap - 1 - (beg + c)]
-                     : p.val[ag.pool][beg + c];
+            unsigned long long vb = p.val[ag.pool][beg + c];
             for (int b = 0; b < w; b++) *q++ = (uint8_t)(vb >> (8 * b));
           }
           break;
