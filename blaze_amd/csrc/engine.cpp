@@ -418,8 +418,9 @@ class AggOp {
     // = sum ++ count, avg.rs:208-217) plus the optional side {min, max} pair
     // (maxmin.rs:104-119; single-phase path only). Anything else fails
     // loudly at plan build.
-    if (node.agg_exprs.empty() || node.agg_exprs.size() > 8)
-      FAIL("AggExec: 1..8 aggregates supported");
+    if (node.agg_exprs.empty() ||
+        node.agg_exprs.size() > (size_t)MA_MAX_AGGS)
+      FAIL("AggExec: 1..12 aggregates supported");
     if (node.modes.size() != node.agg_exprs.size())
       FAIL("AggExec: modes/aggs length mismatch");
     for (const AggMode m : node.modes)
@@ -428,21 +429,33 @@ class AggOp {
     merge_mode_ = (mode_ != AggMode::Partial);
     final_output_ = (mode_ == AggMode::Final);
     layout_ = 0;
-    bool have_col = false;
+    bool have_col = false, have_distinct_cols = false, have_null_lit = false;
+    int fam_sum = 0, fam_cnt = 0, fam_min = 0, fam_max = 0, fam_first = 0,
+        fam_firstin = 0, npools = 0;
+    bool any_i32 = false, mixed_acc = false;
     for (size_t i = 0; i < node.agg_exprs.size(); i++) {
       const Expr& a = node.agg_exprs[i];
       uint32_t k;
       switch (a.agg_function) {
-        case AGG_SUM: k = AGGL_SUM; break;
-        case AGG_COUNT: k = AGGL_CNT; break;
-        case AGG_AVG: k = AGGL_AVG; break;
-        case AGG_MIN: k = AGGL_MIN; has_mm_ = true; break;
-        case AGG_MAX: k = AGGL_MAX; has_mm_ = true; break;
-        case AGG_FIRST: k = AGGL_FIRST; has_first_ = true; break;
-        case AGG_FIRST_IGNORES_NULL: k = AGGL_FIRSTIN; has_first_ = true; break;
-        case AGG_COLLECT_LIST: k = AGGL_CLIST; has_coll_ = true; break;
+        case AGG_SUM: k = AGGL_SUM; fam_sum++; break;
+        case AGG_COUNT: k = AGGL_CNT; fam_cnt++; break;
+        case AGG_AVG: k = AGGL_AVG; fam_sum++; fam_cnt++; break;
+        case AGG_MIN: k = AGGL_MIN; fam_min++; has_mm_ = true; break;
+        case AGG_MAX: k = AGGL_MAX; fam_max++; has_mm_ = true; break;
+        case AGG_FIRST: k = AGGL_FIRST; fam_first++; has_first_ = true; break;
+        case AGG_FIRST_IGNORES_NULL:
+          k = AGGL_FIRSTIN;
+          fam_firstin++;
+          has_first_ = true;
+          break;
+        case AGG_COLLECT_LIST:
+          k = AGGL_CLIST;
+          npools++;
+          has_coll_ = true;
+          break;
         case AGG_COLLECT_SET:
           k = AGGL_CSET;
+          npools++;
           has_coll_ = true;
           coll_set_ = true;
           break;
@@ -450,51 +463,99 @@ class AggOp {
           FAIL("AggExec: only SUM/COUNT/AVG/MIN/MAX/FIRST[_IGNORES_NULL]/"
                "COLLECT_LIST aggregates on this path");
       }
-      layout_ |= k << (4 * i);
+      if (i < 8) layout_ |= k << (4 * i);
       // accumulator arithmetic type follows the agg's declared data type
       // (sum.rs:78-88 casts inputs to it; maxmin.rs:81-83 preserves it)
+      MaAgg ma;
+      ma.kind = (uint8_t)k;
       if (k != AGGL_CNT) {
         bool is_int = (a.return_type == DType::Int64);
+        bool is_i32 = (a.return_type == DType::Int32);
+        ma.acc_t = is_i32 ? 2 : (is_int ? 1 : 0);
+        any_i32 |= is_i32;
         if (!val_typed_seen_) {
           val_is_int_ = is_int;
           val_typed_seen_ = true;
-        } else if (val_is_int_ != is_int) {
-          FAIL("AggExec: mixed i64/f64 accumulator types unsupported "
-               "(shared-argument agg set)");
+        } else if (val_is_int_ != (is_int || is_i32)) {
+          mixed_acc = true;
         }
-        if (k == AGGL_AVG && is_int)
+        if (k == AGGL_AVG && (is_int || is_i32))
           FAIL("AggExec: AVG with integer output unsupported (avg.rs "
                "declares a floating result)");
       }
+      ma_desc_.a[ma_desc_.n++] = ma;
       agg_kinds_.push_back(k);
       agg_names_.push_back(i < node.agg_names.size() ? node.agg_names[i]
                                                      : "agg" + std::to_string(i));
+      ma_arg_cols_.push_back(-1);
       if (!merge_mode_) {
         const Expr& child = a.children.at(0);
-        if (child.kind != Expr::Column)
-          FAIL("aggregate arg must be a Column");
-        if (!have_col) {
-          val_col_ = child.col_index;
-          have_col = true;
-        } else if (child.col_index != val_col_) {
-          // shared {sum,cnt} accumulators require one argument column
-          FAIL("all aggregates must share one argument column on this path");
+        if (child.kind == Expr::Literal) {
+          ScalarLit sl;
+          std::string e2;
+          if (!decode_ipc_scalar(child.literal_ipc.data(),
+                                 child.literal_ipc.size(), &sl, &e2) ||
+              !sl.is_null)
+            FAIL("aggregate Literal arg must be NULL on this path");
+          have_null_lit = true;  // collect.rs-style empty behavior
+        } else if (child.kind == Expr::Column) {
+          ma_arg_cols_.back() = (int)child.col_index;
+          if (!have_col) {
+            val_col_ = child.col_index;
+            have_col = true;
+          } else if (child.col_index != val_col_) {
+            have_distinct_cols = true;
+          }
+        } else {
+          FAIL("aggregate arg must be a Column or NULL literal");
         }
       }
     }
-    {
+    // Multi-argument mode (agg.rs:73-169: independent arg expressions per
+    // agg; per-agg accumulator columns at the declared types). The trigger
+    // rule is computable from the PLAN alone, so a Partial task and its
+    // Final consumer always agree on the a8 wire layout: family duplicates
+    // make the legacy shared-accumulator wire ambiguous; Int32/mixed
+    // accumulator types and multiple collect pools are not representable in
+    // it at all. Distinct arg columns / NULL-literal args additionally
+    // trigger ma on the PARTIAL side (the wire stays legacy-parseable when
+    // families do not repeat, so the Final side may run either mode).
+    ma_mode_ = fam_sum > 1 || fam_cnt > 1 || fam_min > 1 || fam_max > 1 ||
+               fam_first > 1 || fam_firstin > 1 || npools > 1 || any_i32 ||
+               mixed_acc ||
+               (!merge_mode_ && (have_distinct_cols || have_null_lit));
+    if (!ma_mode_ && node.agg_exprs.size() > 8)
+      FAIL("AggExec: >8 aggregates require the multi-arg layout");
+    if (ma_mode_) {
+      if (npools > MA_MAX_POOLS)
+        FAIL("AggExec: at most 4 COLLECT aggregates");
+      int pool = 0;
+      for (int j = 0; j < ma_desc_.n; j++)
+        if (ma_desc_.a[j].kind == AGGL_CLIST ||
+            ma_desc_.a[j].kind == AGGL_CSET)
+          ma_desc_.a[j].pool = (uint8_t)pool++;
+      // the legacy side-array machinery is replaced by the bank
+      has_mm_ = has_first_ = has_coll_ = false;
+      skip_enabled_ = false;
+    } else {
       bool has_l = false, has_s = false;
-      for (uint32_t k : agg_kinds_) {
-        has_l |= (k == AGGL_CLIST);
-        has_s |= (k == AGGL_CSET);
+      for (uint32_t kk : agg_kinds_) {
+        has_l |= (kk == AGGL_CLIST);
+        has_s |= (kk == AGGL_CSET);
       }
       if (has_l && has_s)
         FAIL("COLLECT_LIST and COLLECT_SET together share one argument pool "
-             "— unsupported on this path");
+             "— unsupported on this path (single-pool legacy layout)");
+      if (!merge_mode_ && !have_col && !node.agg_exprs.empty()) {
+        bool only_counts = true;
+        for (uint32_t kk : agg_kinds_) only_counts &= (kk == AGGL_CNT);
+        if (!only_counts) FAIL("aggregate arg must be a Column");
+      }
     }
     batch_size_ = conf.get_i("BATCH_SIZE", 10000);
     if (batch_size_ <= 0) FAIL("invalid BATCH_SIZE conf");
-    skip_enabled_ = node.supports_partial_skipping && !merge_mode_;
+    skip_enabled_ =
+        node.supports_partial_skipping && !merge_mode_ && !ma_mode_;
     skip_ratio_ = conf.get_d("PARTIAL_AGG_SKIPPING_RATIO", 0.999);
     skip_min_rows_ = conf.get_i("PARTIAL_AGG_SKIPPING_MIN_ROWS", 20000);
     // device output: partial freeze output stays in HBM (ArrowDeviceArray
@@ -586,7 +647,28 @@ class AggOp {
       key.dt = DType::Int64;
     }
     }
-    if (!merge_mode_) {
+    if (ma_mode_ && gkey_)
+      FAIL("multi-argument aggregates with Utf8/multi-column grouping keys "
+           "unsupported on this path");
+    if (ma_mode_ && !merge_mode_ && !ma_args_checked_) {
+      ma_args_checked_ = true;
+      for (int j = 0; j < ma_desc_.n; j++) {
+        if (ma_arg_cols_[j] < 0) {
+          ma_desc_.a[j].arg_dt = 3;  // NULL literal
+          continue;
+        }
+        DType dt = b.cols.at(ma_arg_cols_[j]).dt;
+        uint8_t code;
+        if (dt == DType::Float64) code = 0;
+        else if (dt == DType::Int64) code = 1;
+        else if (dt == DType::Int32) code = 2;
+        else FAIL("unsupported aggregate argument dtype");
+        if (ma_desc_.a[j].acc_t != 0 && code == 0)
+          FAIL("float argument with integer accumulator unsupported");
+        ma_desc_.a[j].arg_dt = code;
+      }
+    }
+    if (!merge_mode_ && !ma_mode_) {
       // sum.rs:78-88 prepare_partial_args: the argument is CAST to the
       // accumulator type before update. Narrower numeric args widen on
       // device; anything else fails loudly.
@@ -632,7 +714,7 @@ class AggOp {
       init_agg2_conf();  // chunk bound must be read BEFORE sizing the chunk
       // MIN/MAX agg sets stay single-phase: the LDS bucket kernel's slot
       // holds {key,cnt,sum,first} only (perf note in DESIGN.md)
-      if (!gkey_ && !merge_mode_ && !has_mm_ && !has_first_ &&
+      if (!gkey_ && !ma_mode_ && !merge_mode_ && !has_mm_ && !has_first_ &&
           !has_coll_ && b.num_rows - done >= AGG2_MIN_CHUNK) {
         init_agg2_conf();
         if (agg2_v3_ && agg2_pipe_ && !skip_enabled_ &&
@@ -663,7 +745,9 @@ class AggOp {
       }
       int64_t chunk = std::min(b.num_rows - done, free_slots);
       if (done + chunk < b.num_rows) chunk &= ~(int64_t)7;  // bitmap-sliceable
-      if (gkey_) {
+      if (ma_mode_) {
+        ma_chunk(b, done, chunk);
+      } else if (gkey_) {
         gkey_chunk(b, done, chunk);
       } else if (merge_mode_) {
         const DevColumn& buf = b.cols.at(1);
@@ -757,11 +841,20 @@ class AggOp {
     return f;
   }
 
+  DType ma_out_dt(size_t i) const {  // per-agg declared type (ma mode)
+    switch (ma_desc_.a[i].acc_t) {
+      case 1: return DType::Int64;
+      case 2: return DType::Int32;
+      default: return DType::Float64;
+    }
+  }
+
   std::vector<OutField> output_fields() const {
     if (final_output_) {
       std::vector<OutField> f = key_fields();
-      const DType vdt = val_is_int_ ? DType::Int64 : DType::Float64;
+      const DType svdt = val_is_int_ ? DType::Int64 : DType::Float64;
       for (size_t i = 0; i < agg_kinds_.size(); i++) {
+        DType vdt = ma_mode_ ? ma_out_dt(i) : svdt;
         if (agg_kinds_[i] == AGGL_CNT)
           f.push_back({agg_names_[i], DType::Int64, false});
         else if (agg_kinds_[i] == AGGL_AVG)
@@ -957,6 +1050,172 @@ class AggOp {
     t_.c_cap = coll_n0_ + coll_n1_;  // null segment ends the sorted view
   }
 
+  // Sort each multi-arg collect pool exactly like prepare_collect sorts the
+  // legacy pool: {key asc (signed), prio asc within key} + null-key back
+  // segment in prio order appended at [n0, n0+n1); COLLECT_SET pools dedup
+  // by (key, value) keeping first occurrence. The sorted views replace the
+  // raw pool pointers in mp_ (no further appends happen after finish) and
+  // the device counters are updated to the post-dedup counts.
+  void prepare_ma_pools() {
+    if (!ma_mode_ || ma_pools_sorted_) return;
+    ma_pools_sorted_ = true;
+    if (pinned_meta_.size() < MA_MAX_POOLS * 16 + 16)
+      pinned_meta_.alloc(MA_MAX_POOLS * 16 + 16);
+    unsigned long long* h_n = pinned_meta_.get<unsigned long long>();
+    AURON_HIP(hipMemcpyAsync(h_n, d_mp_n_.get(), MA_MAX_POOLS * 16,
+                             hipMemcpyDeviceToHost, stream_));
+    AURON_HIP(hipStreamSynchronize(stream_));
+    for (int j = 0; j < ma_desc_.n; j++) {
+      uint8_t p = ma_desc_.a[j].pool;
+      if (p >= MA_MAX_POOLS) continue;
+      bool dedup = ma_desc_.a[j].kind == AGGL_CSET;
+      int64_t n0 = (int64_t)h_n[2 * p], n1 = (int64_t)h_n[2 * p + 1];
+      if (n0 + n1 > mp_.cap) FAIL("collect pool overflow (ma mode)");
+      int64_t nmax = std::max<int64_t>(std::max(n0, n1), 1);
+      size_t tb0 = 0;
+      sort_pairs_u64_u32(nullptr, nullptr, nullptr, nullptr, nmax, nullptr,
+                         &tb0, stream_);
+      DevBuf tmp(tb0), idx(nmax * 4), idxo(nmax * 4), scr(nmax * 8);
+      auto sort_pass = [&](unsigned long long* by, int64_t n,
+                           unsigned long long* key, unsigned long long* prio,
+                           unsigned long long* val, DevBuf& ak, DevBuf& ap,
+                           DevBuf& av) {
+        launch_iota_u32(idx.get<uint32_t>(), n, stream_);
+        size_t tb = tmp.size();
+        sort_pairs_u64_u32(by, idx.get<uint32_t>(),
+                           scr.get<unsigned long long>(),
+                           idxo.get<uint32_t>(), n, tmp.get(), &tb, stream_);
+        if (key)
+          launch_gather_u64_idx(key, idxo.get<uint32_t>(), n,
+                                ak.get<unsigned long long>(), stream_);
+        launch_gather_u64_idx(prio, idxo.get<uint32_t>(), n,
+                              ap.get<unsigned long long>(), stream_);
+        launch_gather_u64_idx(val, idxo.get<uint32_t>(), n,
+                              av.get<unsigned long long>(), stream_);
+      };
+      d_mp_skey_[p].alloc(std::max<int64_t>(n0 + n1, 1) * 8);
+      d_mp_sval_[p].alloc(std::max<int64_t>(n0 + n1, 1) * 8);
+      auto K = [&](DevBuf& bb) { return bb.get<unsigned long long>(); };
+      if (n0 > 0) {
+        DevBuf ka(n0 * 8), kb(n0 * 8), pa(n0 * 8), pb(n0 * 8), va(n0 * 8),
+            vb(n0 * 8), biased(n0 * 8);
+        AURON_HIP(hipMemcpyAsync(ka.get(), mp_.key[p], n0 * 8,
+                                 hipMemcpyDeviceToDevice, stream_));
+        AURON_HIP(hipMemcpyAsync(pa.get(), mp_.prio[p], n0 * 8,
+                                 hipMemcpyDeviceToDevice, stream_));
+        AURON_HIP(hipMemcpyAsync(va.get(), mp_.val[p], n0 * 8,
+                                 hipMemcpyDeviceToDevice, stream_));
+        sort_pass(K(pa), n0, K(ka), K(pa), K(va), kb, pb, vb);
+        std::swap(ka, kb); std::swap(pa, pb); std::swap(va, vb);
+        if (dedup) {
+          sort_pass(K(va), n0, K(ka), K(pa), K(va), kb, pb, vb);
+          std::swap(ka, kb); std::swap(pa, pb); std::swap(va, vb);
+          launch_bias_i64(K(ka), n0, biased.get<unsigned long long>(),
+                          stream_);
+          sort_pass(biased.get<unsigned long long>(), n0, K(ka), K(pa),
+                    K(va), kb, pb, vb);
+          std::swap(ka, kb); std::swap(pa, pb); std::swap(va, vb);
+          DevBuf mark(n0), pos2((n0 + 1) * 4);
+          launch_coll_mark_heads((const long long*)K(ka), K(va), n0, 1,
+                                 mark.get<uint8_t>(), stream_);
+          size_t stb = 0;
+          scan_mask_u8(mark.get<uint8_t>(), pos2.get<uint32_t>(), n0,
+                       nullptr, &stb, stream_);
+          DevBuf stmp(stb);
+          scan_mask_u8(mark.get<uint8_t>(), pos2.get<uint32_t>(), n0,
+                       stmp.get(), &stb, stream_);
+          for (auto arr : {&ka, &pa, &va}) {
+            launch_compact_u64(K(*arr), mark.get<uint8_t>(),
+                               pos2.get<uint32_t>(), n0,
+                               scr.get<unsigned long long>(), stream_);
+            AURON_HIP(hipMemcpyAsync(arr->get(), scr.get(), n0 * 8,
+                                     hipMemcpyDeviceToDevice, stream_));
+          }
+          uint32_t* hc = (uint32_t*)(h_n + MA_MAX_POOLS * 2);
+          AURON_HIP(hipMemcpyAsync(hc, pos2.get<uint32_t>() + n0, 4,
+                                   hipMemcpyDeviceToHost, stream_));
+          AURON_HIP(hipStreamSynchronize(stream_));
+          n0 = (int64_t)*hc;
+          if (n0 > 0) {  // restore prio order for the final key pass
+            sort_pass(K(pa), n0, K(ka), K(pa), K(va), kb, pb, vb);
+            std::swap(ka, kb); std::swap(pa, pb); std::swap(va, vb);
+          }
+        }
+        if (n0 > 0) {
+          launch_bias_i64(K(ka), n0, biased.get<unsigned long long>(),
+                          stream_);
+          launch_iota_u32(idx.get<uint32_t>(), n0, stream_);
+          size_t tb = tmp.size();
+          sort_pairs_u64_u32(biased.get<unsigned long long>(),
+                             idx.get<uint32_t>(),
+                             scr.get<unsigned long long>(),
+                             idxo.get<uint32_t>(), n0, tmp.get(), &tb,
+                             stream_);
+          launch_gather_u64_idx(K(ka), idxo.get<uint32_t>(), n0,
+                                d_mp_skey_[p].get<unsigned long long>(),
+                                stream_);
+          launch_gather_u64_idx(K(va), idxo.get<uint32_t>(), n0,
+                                d_mp_sval_[p].get<unsigned long long>(),
+                                stream_);
+        }
+      }
+      if (n1 > 0) {  // null-key back segment -> [n0, n0+n1) of the view
+        DevBuf pa(n1 * 8), pb(n1 * 8), va(n1 * 8), vb(n1 * 8);
+        // back segment occupies [cap-n1, cap); arrival order is reversed in
+        // memory but the prio sort below restores it regardless
+        AURON_HIP(hipMemcpyAsync(pa.get(), mp_.prio[p] + (mp_.cap - n1),
+                                 n1 * 8, hipMemcpyDeviceToDevice, stream_));
+        AURON_HIP(hipMemcpyAsync(va.get(), mp_.val[p] + (mp_.cap - n1),
+                                 n1 * 8, hipMemcpyDeviceToDevice, stream_));
+        sort_pass(K(pa), n1, nullptr, K(pa), K(va), pb, pb, vb);
+        std::swap(pa, pb); std::swap(va, vb);
+        if (dedup) {
+          sort_pass(K(va), n1, nullptr, K(pa), K(va), pb, pb, vb);
+          std::swap(pa, pb); std::swap(va, vb);
+          DevBuf mark(n1), pos2((n1 + 1) * 4);
+          launch_coll_mark_heads(nullptr, K(va), n1, 0, mark.get<uint8_t>(),
+                                 stream_);
+          size_t stb = 0;
+          scan_mask_u8(mark.get<uint8_t>(), pos2.get<uint32_t>(), n1,
+                       nullptr, &stb, stream_);
+          DevBuf stmp(stb);
+          scan_mask_u8(mark.get<uint8_t>(), pos2.get<uint32_t>(), n1,
+                       stmp.get(), &stb, stream_);
+          for (auto arr : {&pa, &va}) {
+            launch_compact_u64(K(*arr), mark.get<uint8_t>(),
+                               pos2.get<uint32_t>(), n1,
+                               scr.get<unsigned long long>(), stream_);
+            AURON_HIP(hipMemcpyAsync(arr->get(), scr.get(), n1 * 8,
+                                     hipMemcpyDeviceToDevice, stream_));
+          }
+          uint32_t* hc = (uint32_t*)(h_n + MA_MAX_POOLS * 2);
+          AURON_HIP(hipMemcpyAsync(hc, pos2.get<uint32_t>() + n1, 4,
+                                   hipMemcpyDeviceToHost, stream_));
+          AURON_HIP(hipStreamSynchronize(stream_));
+          n1 = (int64_t)*hc;
+          if (n1 > 0) {
+            sort_pass(K(pa), n1, nullptr, K(pa), K(va), pb, pb, vb);
+            std::swap(pa, pb); std::swap(va, vb);
+          }
+        }
+        if (n1 > 0)
+          AURON_HIP(hipMemcpyAsync(
+              d_mp_sval_[p].get<unsigned long long>() + n0, va.get(), n1 * 8,
+              hipMemcpyDeviceToDevice, stream_));
+      }
+      AURON_HIP(hipStreamSynchronize(stream_));  // temps return to the pool
+      mp_.key[p] = d_mp_skey_[p].get<long long>();
+      mp_.val[p] = d_mp_sval_[p].get<unsigned long long>();
+      mp_counts_[p][0] = n0;
+      mp_counts_[p][1] = n1;
+      unsigned long long nn[2] = {(unsigned long long)n0,
+                                  (unsigned long long)n1};
+      AURON_HIP(hipMemcpyAsync(d_mp_n_.get<unsigned long long>() + 2 * p, nn,
+                               16, hipMemcpyHostToDevice, stream_));
+    }
+    AURON_HIP(hipStreamSynchronize(stream_));
+  }
+
   std::vector<std::pair<int64_t, std::vector<HostOutCol>>> finish() {
     std::vector<std::pair<int64_t, std::vector<HostOutCol>>> out;
     drain_timing();
@@ -1037,9 +1296,9 @@ class AggOp {
     std::vector<std::pair<int64_t, std::vector<DevOutCol>>> out;
     drain_timing();
     prepare_collect();
-    if (gkey_)
-      FAIL("AURON_HIP_DEVICE_OUTPUT with generalized grouping keys "
-           "unsupported (unset the conf)");
+    if (gkey_ || ma_mode_)
+      FAIL("AURON_HIP_DEVICE_OUTPUT with generalized keys / multi-argument "
+           "aggregates unsupported (unset the conf)");
     if (!spill_.empty())
       FAIL("AURON_HIP_DEVICE_OUTPUT with spill unsupported (unset the conf)");
     if (!skipped_.empty())
@@ -1145,6 +1404,38 @@ class AggOp {
       t_.f_st = d_fst_.get<uint8_t>();
       launch_first_init(t_.f_row, t_.f_val, t_.f_st, cap + 2, stream_);
     }
+    if (ma_mode_) {  // per-agg accumulator bank (growth re-allocs)
+      int64_t cap2 = cap + 2;
+      d_ma_acc_.alloc((int64_t)ma_desc_.n * cap2 * 8);
+      d_ma_meta_.alloc((int64_t)ma_desc_.n * cap2 * 8);
+      d_ma_st_.alloc((int64_t)ma_desc_.n * cap2);
+      ma_.acc = d_ma_acc_.get<unsigned long long>();
+      ma_.meta = d_ma_meta_.get<unsigned long long>();
+      ma_.st = d_ma_st_.get<uint8_t>();
+      ma_.stride = cap2;
+      launch_ma_init(ma_desc_, ma_, cap2, stream_);
+      if (!d_mp_n_) {  // pools survive growth (keyed by VALUE not slot)
+        int64_t pc = std::max<int64_t>(1024, conf_coll_cap_ / 4);
+        for (int j = 0; j < ma_desc_.n; j++) {
+          uint8_t p = ma_desc_.a[j].pool;
+          if (p >= MA_MAX_POOLS) continue;
+          d_mp_key_[p].alloc(pc * 8);
+          d_mp_prio_[p].alloc(pc * 8);
+          d_mp_val_[p].alloc(pc * 8);
+          mp_.cap = pc;
+        }
+        d_mp_n_.alloc(MA_MAX_POOLS * 16);
+        AURON_HIP(hipMemsetAsync(d_mp_n_.get(), 0, MA_MAX_POOLS * 16,
+                                 stream_));
+      }
+      for (int p = 0; p < MA_MAX_POOLS; p++) {
+        if (!d_mp_key_[p]) continue;
+        mp_.key[p] = d_mp_key_[p].get<long long>();
+        mp_.prio[p] = d_mp_prio_[p].get<unsigned long long>();
+        mp_.val[p] = d_mp_val_[p].get<unsigned long long>();
+      }
+      mp_.n = d_mp_n_.get<unsigned long long>();
+    }
     if (gkey_) {  // growth re-allocs the off array; pool survives
       d_gkoff_.alloc(cap * 8);
       AURON_HIP(hipMemsetAsync(d_gkoff_.get(), 0, cap * 8, stream_));
@@ -1176,6 +1467,59 @@ class AggOp {
   static constexpr int AGG3_NBUCK_LOG2 = 9;   // v3: 512 buckets
   static constexpr int AGG3_NBUCK = 1 << AGG3_NBUCK_LOG2;
   static constexpr int AGG3_GRID_LOG2 = 8;    // v3 scatter: 256 x 1024-thr           // 1024 buckets: scatter write-line footprint ~L2-sized (512 buckets overflow the 2048-slot LDS window: 2.3% leftovers, 3x slower)
+
+  // ---- multi-argument chunk (kernels_maarg.hip) --------------------------
+  void ma_chunk(const DevBatch& b, int64_t done, int64_t chunk) {
+    const DevColumn& key = b.cols.at(key_col_);
+    const int64_t* keys = (const int64_t*)key.values + done;
+    const uint8_t* kv = key.validity ? key.validity + done / 8 : nullptr;
+    if (d_ma_slots_.size() < (size_t)chunk * 4) d_ma_slots_.alloc(chunk * 4);
+    MaArgs args;
+    if (!merge_mode_) {
+      for (int j = 0; j < ma_desc_.n; j++) {
+        if (ma_arg_cols_[j] < 0) continue;
+        const DevColumn& c = b.cols.at(ma_arg_cols_[j]);
+        size_t w = dtype_width(c.dt);
+        args.vals[j] = (const uint8_t*)c.values + (size_t)done * w;
+        args.valid[j] = c.validity ? c.validity + done / 8 : nullptr;
+      }
+    }
+    hipEvent_t e0, e1;
+    AURON_HIP(hipEventCreate(&e0));
+    AURON_HIP(hipEventCreate(&e1));
+    AURON_HIP(hipEventRecord(e0, stream_));
+    launch_slots_upsert(t_, keys, kv, chunk, row_cursor_,
+                        d_ma_slots_.get<uint32_t>(), stream_);
+    bool any_first = false;
+    for (int j = 0; j < ma_desc_.n; j++)
+      any_first |= (ma_desc_.a[j].kind == AGGL_FIRST ||
+                    ma_desc_.a[j].kind == AGGL_FIRSTIN);
+    if (merge_mode_) {
+      const DevColumn& buf = b.cols.at(1);
+      if (buf.dt != DType::Binary) FAIL("agg-buf column must be Binary");
+      launch_ma_merge_frozen(t_, ma_desc_, ma_, mp_, keys, kv,
+                             d_ma_slots_.get<uint32_t>(),
+                             (const uint8_t*)buf.values, buf.offsets + done,
+                             chunk, row_cursor_, stream_);
+      if (any_first)
+        launch_ma_first_capture_frozen(t_, ma_desc_, ma_,
+                                       d_ma_slots_.get<uint32_t>(),
+                                       (const uint8_t*)buf.values,
+                                       buf.offsets + done, chunk,
+                                       row_cursor_, stream_);
+    } else {
+      launch_ma_update(t_, ma_desc_, ma_, mp_, args, keys, kv,
+                       d_ma_slots_.get<uint32_t>(), chunk, row_cursor_,
+                       stream_);
+      if (any_first)
+        launch_ma_first_capture(t_, ma_desc_, ma_, args,
+                                d_ma_slots_.get<uint32_t>(), chunk,
+                                row_cursor_, stream_);
+      update_rows_ += chunk;
+    }
+    AURON_HIP(hipEventRecord(e1, stream_));
+    ev_pairs_.push_back({e0, e1});
+  }
 
   // ---- generalized-key machinery (kernels_gkey.hip) ----------------------
   void init_gkey() {
@@ -1898,6 +2242,9 @@ class AggOp {
       if (gkey_)
         FAIL("spill with Utf8/multi-column grouping keys unsupported "
              "(raise AURON_HIP_MEM_BUDGET)");
+      if (ma_mode_)
+        FAIL("spill with multi-argument aggregates unsupported "
+             "(raise AURON_HIP_MEM_BUDGET)");
       spill_table();  // frees the main region; caller re-checks free space
     }
   }
@@ -1910,8 +2257,13 @@ class AggOp {
            omm = std::move(d_mm_), ofrow = std::move(d_frow_),
            ofval = std::move(d_fval_), ofst = std::move(d_fst_),
            ogkoff = std::move(d_gkoff_);
+    MaAcc oldma = ma_;
+    DevBuf oma = std::move(d_ma_acc_), omam = std::move(d_ma_meta_),
+           omas = std::move(d_ma_st_);
     init_table(new_cap);
-    if (gkey_)
+    if (ma_mode_)
+      launch_ma_rebuild(t_, ma_, old, oldma, ma_desc_.n, stream_);
+    else if (gkey_)
       launch_gkey_rebuild(t_, g_, old, oldg, stream_);
     else
       launch_agg_rebuild(t_, old, stream_);
@@ -1923,6 +2275,7 @@ class AggOp {
   void emit_table(std::vector<std::pair<int64_t, std::vector<HostOutCol>>>* out,
                   bool exclude_specials) {
     prepare_collect();  // one-shot per pool state (coll_sorted_ guard)
+    prepare_ma_pools();
     DevBuf order, first;
     int64_t ng = table_order(&order, &first);
     DBG("agg.finish ng=%lld excl=%d", (long long)ng, (int)exclude_specials);
@@ -2010,6 +2363,95 @@ class AggOp {
     return out;
   }
 
+  // multi-arg emit: per-agg columns from the accumulator bank; partial mode
+  // freezes the per-agg wire instead (ma_freeze_*). Collect aggs read their
+  // sorted pool views through the legacy coll kernels (the AggTable's c_*
+  // pointers are temporarily re-aimed per pool).
+  void emit_ma_aggs(const uint32_t* order_slots, int64_t n,
+                    std::vector<HostOutCol>* cols) {
+    size_t bm = (n + 7) / 8;
+    if (!final_output_) {
+      // partial: ONE Binary agg-buf column with the per-agg wire
+      DevBuf lens(n * 4), offs((n + 1) * 4);
+      launch_ma_freeze_len(t_, ma_desc_, ma_, mp_, order_slots, n,
+                           lens.get<int32_t>(), stream_);
+      std::vector<int32_t> h_lens(n);
+      AURON_HIP(hipMemcpyAsync(h_lens.data(), lens.get(), n * 4,
+                               hipMemcpyDeviceToHost, stream_));
+      AURON_HIP(hipStreamSynchronize(stream_));
+      std::vector<int32_t> h_offs(n + 1, 0);
+      for (int64_t i = 0; i < n; i++) h_offs[i + 1] = h_offs[i] + h_lens[i];
+      AURON_HIP(hipMemcpyAsync(offs.get(), h_offs.data(), (n + 1) * 4,
+                               hipMemcpyHostToDevice, stream_));
+      DevBuf data(h_offs[n] ? h_offs[n] : 1);
+      launch_ma_freeze_write(t_, ma_desc_, ma_, mp_, order_slots, n,
+                             offs.get<int32_t>(), data.get<uint8_t>(),
+                             stream_);
+      HostOutCol buf_col;
+      buf_col.dt = DType::Binary;
+      buf_col.offsets = std::move(h_offs);
+      d2h_pinned(data.get(), &buf_col.values, (size_t)buf_col.offsets[n]);
+      cols->push_back(std::move(buf_col));
+      return;
+    }
+    for (int j = 0; j < ma_desc_.n; j++) {
+      const MaAgg& ag = ma_desc_.a[j];
+      HostOutCol ac;
+      if (ag.kind == AGGL_CLIST || ag.kind == AGGL_CSET) {
+        // re-aim the legacy collect view at this pool's sorted arrays
+        AggTable tv = t_;
+        uint8_t p = ag.pool;
+        tv.c_key = mp_.key[p];
+        tv.c_val = mp_.val[p];
+        tv.c_n = d_mp_n_.get<unsigned long long>() + 2 * p;
+        tv.c_cap = mp_counts_[p][0] + mp_counts_[p][1];
+        DevBuf ccnt(n * 4);
+        launch_coll_counts(tv, order_slots, n, ccnt.get<int32_t>(), stream_);
+        std::vector<int32_t> h_cnts(n);
+        AURON_HIP(hipMemcpyAsync(h_cnts.data(), ccnt.get(), n * 4,
+                                 hipMemcpyDeviceToHost, stream_));
+        AURON_HIP(hipStreamSynchronize(stream_));
+        ac.offsets.assign(n + 1, 0);
+        for (int64_t i = 0; i < n; i++)
+          ac.offsets[i + 1] = ac.offsets[i] + h_cnts[i];
+        DevBuf d_off((n + 1) * 4),
+            items((int64_t)ac.offsets[n] * 8 + 8);
+        AURON_HIP(hipMemcpyAsync(d_off.get(), ac.offsets.data(),
+                                 (n + 1) * 4, hipMemcpyHostToDevice,
+                                 stream_));
+        launch_coll_gather(tv, order_slots, n, d_off.get<int32_t>(),
+                           items.get<unsigned long long>(), stream_);
+        ac.dt = ma_out_dt(j);
+        ac.is_list = true;
+        if (ac.dt == DType::Int32) {
+          DevBuf narrow((int64_t)ac.offsets[n] * 4 + 4);
+          launch_narrow_i64_i32(items.get<int64_t>(), ac.offsets[n],
+                                narrow.get<int32_t>(), stream_);
+          d2h_pinned(narrow.get(), &ac.values, (size_t)ac.offsets[n] * 4);
+        } else {
+          d2h_pinned(items.get(), &ac.values, (size_t)ac.offsets[n] * 8);
+        }
+        cols->push_back(std::move(ac));
+        continue;
+      }
+      int w = (ag.kind == AGGL_CNT) ? 8 : (ag.acc_t == 2 ? 4 : 8);
+      DevBuf vals(n * w), bmbuf(bm);
+      launch_ma_gather_out(ma_desc_, ma_, j, order_slots, n,
+                           vals.get<uint8_t>(), bmbuf.get<uint8_t>(),
+                           stream_);
+      ac.dt = (ag.kind == AGGL_CNT) ? DType::Int64
+              : (ag.kind == AGGL_AVG) ? DType::Float64
+                                      : ma_out_dt(j);
+      d2h_pinned(vals.get(), &ac.values, (size_t)n * w);
+      std::vector<uint8_t> h_bm(bm);
+      AURON_HIP(hipMemcpyAsync(h_bm.data(), bmbuf.get(), bm,
+                               hipMemcpyDeviceToHost, stream_));
+      AURON_HIP(hipStreamSynchronize(stream_));
+      if (ag.kind != AGGL_CNT) attach_validity(&ac, h_bm, n);
+      cols->push_back(std::move(ac));
+    }
+  }
+
   std::pair<int64_t, std::vector<HostOutCol>> emit_groups(
       const uint32_t* order_slots, int64_t n) {
     DBG("agg.emit n=%lld final=%d", (long long)n, (int)final_output_);
@@ -2037,6 +2479,16 @@ class AggOp {
       d2h_pinned(k32.get(), &key_col.values, n * 4);
     } else {
       d2h_pinned(keys.get(), &key_col.values, n * 8);
+    }
+    if (ma_mode_) {
+      std::vector<uint8_t> kvh(bm);
+      AURON_HIP(hipMemcpyAsync(kvh.data(), kvalid.get(), bm,
+                               hipMemcpyDeviceToHost, stream_));
+      AURON_HIP(hipStreamSynchronize(stream_));
+      attach_validity(&key_col, kvh, n);
+      cols.push_back(std::move(key_col));
+      emit_ma_aggs(order_slots, n, &cols);
+      return {n, std::move(cols)};
     }
     std::vector<uint8_t> kv(bm), sv(bm);
     AURON_HIP(hipMemcpyAsync(kv.data(), kvalid.get(), bm, hipMemcpyDeviceToHost,
@@ -2252,6 +2704,18 @@ class AggOp {
   bool device_out_ = false;
   // generalized grouping keys (Utf8 / multi-column; kernels_gkey.hip)
   bool key_mode_set_ = false, gkey_ = false;
+  // multi-argument aggregation (kernels_maarg.hip)
+  bool ma_mode_ = false, ma_args_checked_ = false;
+  MaDesc ma_desc_;
+  std::vector<int> ma_arg_cols_;   // per agg: input column or -1 (NULL lit)
+  MaAcc ma_;
+  MaPools mp_;
+  DevBuf d_ma_acc_, d_ma_meta_, d_ma_st_, d_ma_slots_, d_mp_n_;
+  DevBuf d_mp_key_[MA_MAX_POOLS], d_mp_prio_[MA_MAX_POOLS],
+      d_mp_val_[MA_MAX_POOLS], d_mp_skey_[MA_MAX_POOLS],
+      d_mp_sval_[MA_MAX_POOLS];
+  bool ma_pools_sorted_ = false;
+  int64_t mp_counts_[MA_MAX_POOLS][2] = {};
   std::vector<uint32_t> key_cols_;
   std::vector<std::string> key_names_;
   std::vector<DType> key_dts_;
