@@ -130,10 +130,17 @@ __global__ void __launch_bounds__(1024) k_agg3_scatter(
   const int lane = (int)(threadIdx.x & 63);
   int64_t my[A3_RPT];
   int tries[A3_RPT] = {};
+  int64_t k_cur[A3_RPT];
+  double v_cur[A3_RPT];
 #pragma unroll
-  for (int r = 0; r < A3_RPT; r++)
+  for (int r = 0; r < A3_RPT; r++) {
     my[r] = (int64_t)blockIdx.x * blockDim.x + threadIdx.x +
             (int64_t)r * stride;
+    if (my[r] < n) {  // preload; refreshed right after each advance so the
+      k_cur[r] = keys[my[r]];     // next tile's loads hide behind the flush
+      v_cur[r] = vals[my[r]];
+    }
+  }
   const int64_t step = (int64_t)A3_RPT * stride;
   // hard tile bound: rows/thread plus a generous retry allowance — a full
   // staging bucket retries a row for a few tiles at most; exceeding this
@@ -151,43 +158,51 @@ __global__ void __launch_bounds__(1024) k_agg3_scatter(
     for (int r = 0; r < A3_RPT; r++) {
       if (my[r] >= n) continue;
       mine = 1;
-      int64_t k = keys[my[r]];
+      int64_t k = k_cur[r];
+      bool advanced = false;
       bool knull = key_valid && !bit_get3(key_valid, my[r]);
       if (knull || k == KEY_EMPTY3) {
         my[r] += step;  // specials handled by k_agg2_specials
         mine |= (my[r] < n);
-        continue;
-      }
-      uint32_t b = (uint32_t)(mix64_3((uint64_t)k) >> (64 - nbuck_log2));
-      uint32_t pos = atomicAdd(&cnt[b], 1u);
-      if (pos < A3_CAP) {
-        bool vvalid = !val_valid || bit_get3(val_valid, my[r]);
-        uint8_t* rec = stage + ((size_t)b * A3_CAP + pos) * A3_SLOT;
-        *(int64_t*)rec = k;
-        *(double*)(rec + 8) = vals[my[r]];
-        *(uint32_t*)(rec + 16) =
-            (uint32_t)my[r] | (vvalid ? 0x80000000u : 0u);
-        if ((pos & (A3_QUANT - 1)) == A3_QUANT - 1) {  // crossed a packet
-          uint32_t qi = atomicAdd(qn, 1u);
-          if (qi < nbuck + 64) queue[qi] = (uint16_t)b;  // <=1 push per
-        }                                                // bucket per tile
-        my[r] += step;
-        tries[r] = 0;
+        advanced = true;
       } else {
-        atomicSub(&cnt[b], 1u);
-        if (++tries[r] >= A3_RETRY_BYPASS) {
-          // hot bucket (skew): route the row to the leftover list; the
-          // histogram already counted it, so record the correction the
-          // bucket kernel must subtract from this (block,bucket) range
+        uint32_t b = (uint32_t)(mix64_3((uint64_t)k) >> (64 - nbuck_log2));
+        uint32_t pos = atomicAdd(&cnt[b], 1u);
+        if (pos < A3_CAP) {
           bool vvalid = !val_valid || bit_get3(val_valid, my[r]);
-          unsigned long long p = atomicAdd(lo_n, 1ull);
-          leftover[p] = PartRow{
-              k, vals[my[r]],
-              (uint32_t)my[r] | (vvalid ? 0x80000000u : 0u), 0};
-          atomicAdd(&byp[b], 1u);
+          uint8_t* rec = stage + ((size_t)b * A3_CAP + pos) * A3_SLOT;
+          *(int64_t*)rec = k;
+          *(double*)(rec + 8) = v_cur[r];
+          *(uint32_t*)(rec + 16) =
+              (uint32_t)my[r] | (vvalid ? 0x80000000u : 0u);
+          if ((pos & (A3_QUANT - 1)) == A3_QUANT - 1) {  // crossed a packet
+            uint32_t qi = atomicAdd(qn, 1u);
+            if (qi < nbuck + 64) queue[qi] = (uint16_t)b;  // <=1 push per
+          }                                                // bucket per tile
           my[r] += step;
           tries[r] = 0;
-        }  // else: retry next tile
+          advanced = true;
+        } else {
+          atomicSub(&cnt[b], 1u);
+          if (++tries[r] >= A3_RETRY_BYPASS) {
+            // hot bucket (skew): route the row to the leftover list; the
+            // histogram already counted it, so record the correction the
+            // bucket kernel must subtract from this (block,bucket) range
+            bool vvalid = !val_valid || bit_get3(val_valid, my[r]);
+            unsigned long long p = atomicAdd(lo_n, 1ull);
+            leftover[p] = PartRow{
+                k, v_cur[r],
+                (uint32_t)my[r] | (vvalid ? 0x80000000u : 0u), 0};
+            atomicAdd(&byp[b], 1u);
+            my[r] += step;
+            tries[r] = 0;
+            advanced = true;
+          }  // else: retry next tile (k_cur/v_cur stay valid)
+        }
+      }
+      if (advanced && my[r] < n) {
+        k_cur[r] = keys[my[r]];
+        v_cur[r] = vals[my[r]];
       }
     }
     // block-wide termination check doubles as the pre-flush barrier
