@@ -340,11 +340,13 @@ def _import_output(array_ptr, schema_fields):
                 c.cast(ch.buffers[1], c.POINTER(c.c_int32)),
                 shape=(n + 1,)).copy()
             item = ch.children[0].contents
-            npdt = {"l": np.int64, "g": np.float64}[dt[3:]]
+            npdt = {"l": np.int64, "g": np.float64,
+                    "i": np.int32}[dt[3:]]
+            w = np.dtype(npdt).itemsize
             ni = int(item.length)
             vals = np.ctypeslib.as_array(
                 c.cast(item.buffers[1], c.POINTER(c.c_uint8)),
-                shape=(max(ni, 1) * 8,))[:ni * 8].view(npdt).copy()
+                shape=(max(ni, 1) * w,))[:ni * w].view(npdt).copy()
             out.append(dict(dtype="list", offsets=offsets, values=vals,
                             valid=valid))
             continue
