@@ -94,8 +94,10 @@ static constexpr int A3_RETRY_BYPASS = 4;
 // per-tile scan of every bucket made the kernel instruction-bound: PMC
 // showed 20x the dynamic instructions of v2 and 77% parked waves while HBM
 // writes ran at 1.0x algorithmic — gpurun_out/r2_pmc_scatter.json).
-static constexpr int A3_RPT = 2;
+static constexpr int A3_RPT = 2;  // default; AURON_AGG2_RPT=4 selects the
+                                  // wider tile instantiation
 
+template <int A3_RPT>
 __global__ void __launch_bounds__(1024) k_agg3_scatter(
     const int64_t* __restrict__ keys, const uint8_t* __restrict__ key_valid,
     const double* __restrict__ vals, const uint8_t* __restrict__ val_valid,
@@ -272,15 +274,26 @@ void launch_agg3_scatter(const int64_t* keys, const uint8_t* key_valid,
                ((size_t)nbuck + 64) * 2 + 8;  // + queue + counters + bypass
   if (lds > 160 * 1024)
     throw std::runtime_error("agg3 scatter LDS over 160KB");
+  static int rpt = [] {
+    const char* e = getenv("AURON_AGG2_RPT");
+    return (e && e[0] == '4') ? 4 : 2;
+  }();
+  const void* fn = rpt == 4 ? (const void*)k_agg3_scatter<4>
+                            : (const void*)k_agg3_scatter<2>;
   hipError_t e = hipFuncSetAttribute(
-      (const void*)k_agg3_scatter,
-      hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
+      fn, hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
   if (e != hipSuccess)
     throw std::runtime_error("agg3 scatter LDS attribute failed");
-  hipLaunchKernelGGL(k_agg3_scatter, dim3(1 << grid_log2), dim3(1024), lds, s,
-                     keys, key_valid, vals, val_valid, n, nbuck_log2,
-                     grid_log2, line_scan, out, leftover, lo_n,
-                     bypass_matrix, err_flag);
+  if (rpt == 4)
+    hipLaunchKernelGGL(k_agg3_scatter<4>, dim3(1 << grid_log2), dim3(1024),
+                       lds, s, keys, key_valid, vals, val_valid, n,
+                       nbuck_log2, grid_log2, line_scan, out, leftover, lo_n,
+                       bypass_matrix, err_flag);
+  else
+    hipLaunchKernelGGL(k_agg3_scatter<2>, dim3(1 << grid_log2), dim3(1024),
+                       lds, s, keys, key_valid, vals, val_valid, n,
+                       nbuck_log2, grid_log2, line_scan, out, leftover, lo_n,
+                       bypass_matrix, err_flag);
   check_launch3("k_agg3_scatter");
 }
 
@@ -368,19 +381,31 @@ __global__ void __launch_bounds__(1024) k_agg3_bucket(
       }
     };
     uint32_t j = threadIdx.x;
-    for (; j + blockDim.x < total; j += 2 * blockDim.x) {
+    for (; j + 3 * blockDim.x < total; j += 4 * blockDim.x) {
       const uint8_t* r0 = locate(j);
       const uint8_t* r1 = locate(j + blockDim.x);
-      int64_t k0 = *(const int64_t*)r0;        // both loads in flight
+      const uint8_t* r2 = locate(j + 2 * blockDim.x);
+      const uint8_t* r3 = locate(j + 3 * blockDim.x);
+      int64_t k0 = *(const int64_t*)r0;        // all four loads in flight
       int64_t k1 = *(const int64_t*)r1;
+      int64_t k2 = *(const int64_t*)r2;
+      int64_t k3 = *(const int64_t*)r3;
       double v0 = *(const double*)(r0 + 8);
       double v1 = *(const double*)(r1 + 8);
+      double v2 = *(const double*)(r2 + 8);
+      double v3 = *(const double*)(r3 + 8);
       uint32_t rv0 = *(const uint32_t*)(r0 + 16);
       uint32_t rv1 = *(const uint32_t*)(r1 + 16);
+      uint32_t rv2 = *(const uint32_t*)(r2 + 16);
+      uint32_t rv3 = *(const uint32_t*)(r3 + 16);
       uint32_t h0 = (uint32_t)mix64_3((uint64_t)k0) & (L3SLOTS - 1);
       uint32_t h1 = (uint32_t)mix64_3((uint64_t)k1) & (L3SLOTS - 1);
+      uint32_t h2 = (uint32_t)mix64_3((uint64_t)k2) & (L3SLOTS - 1);
+      uint32_t h3 = (uint32_t)mix64_3((uint64_t)k3) & (L3SLOTS - 1);
       resolve(k0, v0, rv0, h0);
       resolve(k1, v1, rv1, h1);
+      resolve(k2, v2, rv2, h2);
+      resolve(k3, v3, rv3, h3);
     }
     for (; j < total; j += blockDim.x) {
       const uint8_t* r = locate(j);
