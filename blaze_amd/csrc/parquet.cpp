@@ -1,6 +1,10 @@
 #include "parquet.h"
 
 #include <dlfcn.h>
+#include <fcntl.h>
+#include <sys/mman.h>
+#include <sys/stat.h>
+#include <unistd.h>
 
 #include <cstdio>
 #include <cstring>
@@ -621,18 +625,25 @@ bool snappy_uncompress(const uint8_t* src, size_t n, std::vector<uint8_t>* out,
 }
 
 // ---- ParquetFile ------------------------------------------------------------
+ParquetFile::~ParquetFile() {
+  if (map_.data) munmap((void*)map_.data, map_.size);
+}
+
 ParquetFile::ParquetFile(const std::string& path) : path_(path) {
-  FILE* f = fopen(path.c_str(), "rb");
-  if (!f) fail("parquet: cannot open " + path);
-  fseek(f, 0, SEEK_END);
-  long sz = ftell(f);
-  fseek(f, 0, SEEK_SET);
-  file_.resize(sz);
-  if (fread(file_.data(), 1, sz, f) != (size_t)sz) {
-    fclose(f);
-    fail("parquet: short read " + path);
+  int fd = open(path.c_str(), O_RDONLY);
+  if (fd < 0) fail("parquet: cannot open " + path);
+  struct stat st;
+  if (fstat(fd, &st) != 0) {
+    close(fd);
+    fail("parquet: cannot stat " + path);
   }
-  fclose(f);
+  long sz = (long)st.st_size;
+  void* p = mmap(nullptr, sz, PROT_READ, MAP_PRIVATE, fd, 0);
+  close(fd);
+  if (p == MAP_FAILED) fail("parquet: mmap failed for " + path);
+  (void)madvise(p, sz, MADV_WILLNEED);
+  map_.data = (const uint8_t*)p;
+  map_.size = (size_t)sz;
   if (sz < 12 || memcmp(file_.data() + sz - 4, "PAR1", 4) != 0)
     fail("parquet: bad magic in " + path);
   uint32_t meta_len;
