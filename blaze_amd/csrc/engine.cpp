@@ -3532,8 +3532,10 @@ struct Runtime {
       std::string decode_err;
       std::mutex err_mu;
       {
+        // cap the worker count: 256 hyperthreads each faulting a ~50 MB
+        // chunk working set thrash; ~48 workers keep the memory system busy
         unsigned nw = std::min<unsigned>(
-            std::max(1u, std::thread::hardware_concurrency()),
+            std::min(48u, std::max(1u, std::thread::hardware_concurrency())),
             (unsigned)((size_t)nrg * width));
         std::atomic<size_t> next{0};
         std::vector<std::thread> ws;
