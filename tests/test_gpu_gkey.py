@@ -214,3 +214,51 @@ def test_utf8_shuffle_partition_ids(tmp_path):
     sizes = np.diff(index.astype(np.int64))
     for p in range(P):
         assert (sizes[p] > 0) == (counts[p] > 0), (p, sizes[p], counts[p])
+
+
+def test_utf8_keys_at_scale():
+    """1M rows / 120K distinct utf8 keys: pool growth + probe behavior at
+    load, full partial->final chain vs the oracle reference."""
+    rng = np.random.default_rng(36)
+    n = 1_000_000
+    nk = 120_000
+    idx = rng.integers(0, nk, n)
+    # varying lengths incl. empty and long keys
+    vocab = [("k%06x" % i) * (1 + i % 3) for i in range(nk)]
+    vocab[0] = ""
+    strs = [vocab[i] for i in idx]
+    vals = rng.integers(0, 1000, n).astype(np.float64)
+    data, offs = _utf8_col(strs)
+
+    td = plan.plan_partial_final_gkey([("k", plan.DT_UTF8, False)],
+                                      ("sum", "count"))
+    t = blaze_amd.Task(td, batches=[[("binary", data, offs, None),
+                                     (vals, None)]],
+                       conf={"BATCH_SIZE": 1 << 20,
+                             "AURON_HIP_AGG_TABLE_SLOTS": 1 << 14})
+    outs = t.run()
+    t.finalize()
+    koffs = outs[0][0]["offsets"]
+    kdata = np.concatenate([o[0]["data"] for o in outs]).tobytes()
+    got_sum = np.concatenate([o[1]["values"] for o in outs])
+    got_cnt = np.concatenate([o[2]["values"] for o in outs])
+
+    import collections
+    rs = collections.defaultdict(float)
+    rc = collections.Counter()
+    order = []
+    seen = set()
+    for i in range(n):
+        s = strs[i]
+        if s not in seen:
+            seen.add(s)
+            order.append(s)
+        rs[s] += vals[i]
+        rc[s] += 1
+    assert len(got_cnt) == len(order)
+    got_keys = [kdata[koffs[i]:koffs[i + 1]].decode()
+                for i in range(len(koffs) - 1)]
+    assert got_keys == order[:len(got_keys)]
+    for i, s in enumerate(got_keys):
+        assert got_cnt[i] == rc[s]
+        np.testing.assert_allclose(got_sum[i], rs[s], rtol=1e-9)
