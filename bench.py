@@ -214,48 +214,22 @@ def run_step(ba, plan, dev_input, world, rank, torch, dist, device, stats):
     return nfinal
 
 
-def _cpu_worker(args):
-    """One pinned worker: partial-agg + freeze + partition ids over its row
-    slice; returns (frozen keys, acc bytes, offsets) for the parent's final
-    merge, plus its row count."""
-    import os as _os
-
-    wid, rows, seed, core = args
-    try:
-        _os.sched_setaffinity(0, {core})
-    except OSError:
-        pass
-    from oracle import pywrap as oracle
-
-    rng = np.random.default_rng(seed)
-    keys = rng.integers(0, NUM_KEYS, rows).astype(np.int64)
-    vals = rng.integers(0, 1_000_000, rows).astype(np.float64)
-    vv = rng.random(rows) >= NULL_FRAC
-    t0 = time.perf_counter()
-    a = oracle.Agg()
-    a.update(keys, vals, val_valid=vv)
-    data, offsets = a.freeze()
-    g = a.output()
-    h = oracle.hash_cols([(g["keys"], None)])
-    oracle.partition_ids(h, NUM_PARTITIONS)
-    dt = time.perf_counter() - t0
-    return g["keys"], data, offsets, rows, dt
-
-
 def cpu_baseline_leg(total_rows):
-    """The oracle (CPU restatement of the reference's operators) timed on ALL
-    host cores, pinned, 3 warm + 5 timed runs, median — BASELINE.md's CPU
-    protocol — on a bounded sample of the same workload. Each core runs the
-    partial agg + freeze + partition-ids slice; the parent merges the frozen
-    records (the final-stage merge, single-thread like the reference's
-    per-partition reader). Reported beside the GPU number, never as
-    `value`. A single-core figure is included for context."""
-    import multiprocessing as mp
+    """The oracle (CPU restatement of the reference's operators) on ALL host
+    cores in the reference's own map/reduce topology: per-core map tasks run
+    partial agg + freeze + partition ids over a row slice and sort their
+    frozen records into partition order (the shuffle write side); per-
+    PARTITION reduce tasks then merge every map's slice of that partition
+    (the shuffle read + final agg side). Threads, not processes: the
+    oracle's ctypes calls and numpy release the GIL, and the frozen records
+    move by slicing shared arrays — like the reference's in-memory shuffle
+    on one box. 3 warm + 5 timed runs, median (BASELINE.md protocol);
+    single-core figure reported alongside."""
+    from concurrent.futures import ThreadPoolExecutor
 
     from oracle import pywrap as oracle
 
     ncores = os.cpu_count() or 1
-    # calibrate single-core rate on 2M rows
     rng = np.random.default_rng(SEED)
     cal_n = 2_000_000
     k = rng.integers(0, NUM_KEYS, cal_n).astype(np.int64)
@@ -272,36 +246,69 @@ def cpu_baseline_leg(total_rows):
     b.merge_frozen(g["keys"], data, offsets)
     single_rate = cal_n / (time.perf_counter() - t0)
 
-    # size each timed run for ~2 s of all-core work (5 timed + 3 warm runs
-    # keep the default bench within a couple of minutes), cap 200M rows/run
-    sample = int(min(2e8, max(cal_n * ncores, single_rate * ncores * 2.0)))
-    per_w = sample // ncores
-    args = [(w, per_w, SEED + 7919 * (w + 1), w) for w in range(ncores)]
+    # sample: bounded to keep the default bench within minutes
+    sample = int(min(2e8, max(cal_n * ncores, single_rate * ncores * 1.0)))
+    per_w = max(1, sample // ncores)
+    sample = per_w * ncores
+    # pregenerate ONCE (generation is not part of the measured pass)
+    keys = rng.integers(0, NUM_KEYS, sample).astype(np.int64)
+    vals = rng.integers(0, 1_000_000, sample).astype(np.float64)
+    vv = rng.random(sample) >= NULL_FRAC
+
+    P = NUM_PARTITIONS
+
+    def map_task(w):
+        lo, hi = w * per_w, (w + 1) * per_w
+        a = oracle.Agg()
+        a.update(keys[lo:hi], vals[lo:hi], val_valid=vv[lo:hi])
+        data, offs = a.freeze()
+        g = a.output()
+        h = oracle.hash_cols([(g["keys"], None)])
+        pids = oracle.partition_ids(h, P)
+        order = np.argsort(pids, kind="stable")
+        keys_s = g["keys"][order]
+        lens = (offs[1:] - offs[:-1]).astype(np.int64)
+        lens_s = lens[order]
+        offs_s = np.concatenate([[0], np.cumsum(lens_s)]).astype(np.int64)
+        total = int(offs_s[-1])
+        pos = np.repeat(offs[order][: len(lens_s)] - offs_s[:-1], lens_s) + \
+            np.arange(total, dtype=np.int64)
+        data_np = np.asarray(data, np.uint8)
+        data_s = data_np[pos] if total else np.empty(0, np.uint8)
+        bounds = np.searchsorted(pids[order], np.arange(P + 1))
+        return keys_s, offs_s, data_s, bounds
+
+    def reduce_task(p, maps):
+        fin = oracle.Agg()
+        for keys_s, offs_s, data_s, bounds in maps:
+            b0, b1 = int(bounds[p]), int(bounds[p + 1])
+            if b0 == b1:
+                continue
+            fin.merge_frozen(keys_s[b0:b1], data_s, offs_s[b0:b1 + 1])
+        return fin.num_groups
 
     times = []
-    with mp.get_context("fork").Pool(ncores) as pool:
+    with ThreadPoolExecutor(ncores) as ex:
         for it in range(8):  # 3 warm + 5 timed
             t0 = time.perf_counter()
-            outs = pool.map(_cpu_worker, args)
-            # final merge of every worker's frozen records, single thread
-            fin = oracle.Agg()
-            for gk, gd, go, _, _ in outs:
-                fin.merge_frozen(gk, gd, go)
+            maps = list(ex.map(map_task, range(ncores)))
+            list(ex.map(lambda p: reduce_task(p, maps), range(P)))
             dt = time.perf_counter() - t0
             if it >= 3:
                 times.append(dt)
     times.sort()
     med = times[len(times) // 2]
     return {
-        "value": per_w * ncores / med,
+        "value": sample / med,
         "unit": "rows/s",
         "cores": ncores,
         "kind": "port",
         "single_core_value": round(single_rate, 1),
-        "sample": f"{per_w * ncores} rows of the {total_rows}-row workload "
-                  f"(per-core partial agg + freeze + partition ids, "
-                  f"single-thread final merge), {ncores} pinned cores, "
-                  f"median of 5 timed runs after 3 warmups",
+        "sample": f"{sample} rows of the {total_rows}-row workload "
+                  f"(map: per-core partial agg + freeze + partition sort; "
+                  f"reduce: per-partition frozen merges across maps), "
+                  f"{ncores} worker threads, median of 5 timed runs after "
+                  f"3 warmups",
     }
 
 
