@@ -1691,6 +1691,10 @@ class AggOp {
     // 51 ms/step); kept behind AURON_AGG2_PIPE=1 for experiments
     const char* pe = getenv("AURON_AGG2_PIPE");
     agg2_pipe_ = (pe && pe[0] == '1');
+    // v4 scatter: barrier-free producer/flusher rings (kernels_agg3.hip) —
+    // experimental until measured; AURON_AGG2_V4=1 selects it
+    const char* v4 = getenv("AURON_AGG2_V4");
+    agg2_v4_ = (v4 && v4[0] == '1');
   }
 
   // Cross-chunk pipelined two-phase aggregation (v3 kernels): the scatter
@@ -1923,12 +1927,21 @@ class AggOp {
       scan_counts_matrix(d_linesz_.get<uint32_t>(),
                          d_scanned_.get<uint32_t>(), mat3 + 1,
                          d_scan_tmp_.get(), &tb3, stream_);
-      launch_agg3_scatter(keys, kv, vals, vv, chunk, AGG3_NBUCK_LOG2,
-                          AGG3_GRID_LOG2, d_scanned_.get<uint32_t>(),
-                          d_partkv_.get<uint8_t>(), d_leftover_.get<PartRow>(),
-                          d_counters_.get<unsigned long long>() + 1,
-                          d_linesz_.get<uint32_t>(),  // free after the scan
-                          t_.error_flag, stream_);
+      if (agg2_v4_)
+        launch_agg4_scatter(keys, kv, vals, vv, chunk, AGG3_NBUCK_LOG2,
+                            AGG3_GRID_LOG2, d_scanned_.get<uint32_t>(),
+                            d_partkv_.get<uint8_t>(),
+                            d_leftover_.get<PartRow>(),
+                            d_counters_.get<unsigned long long>() + 1,
+                            d_linesz_.get<uint32_t>(), t_.error_flag, stream_);
+      else
+        launch_agg3_scatter(keys, kv, vals, vv, chunk, AGG3_NBUCK_LOG2,
+                            AGG3_GRID_LOG2, d_scanned_.get<uint32_t>(),
+                            d_partkv_.get<uint8_t>(),
+                            d_leftover_.get<PartRow>(),
+                            d_counters_.get<unsigned long long>() + 1,
+                            d_linesz_.get<uint32_t>(),  // free after the scan
+                            t_.error_flag, stream_);
       launch_agg3_bucket(d_partkv_.get<uint8_t>(), d_counts_.get<uint32_t>(),
                          d_linesz_.get<uint32_t>(),
                          d_scanned_.get<uint32_t>(), AGG3_GRID_LOG2,
@@ -2741,6 +2754,7 @@ class AggOp {
   // two-phase scratch (allocated on first large chunk)
   bool agg2_split_ = false;
   bool agg2_v3_ = true;
+  bool agg2_v4_ = false;
   bool agg2_pipe_ = false;
   bool agg2_conf_read_ = false;
   int agg2_grid_log2_ = 9;

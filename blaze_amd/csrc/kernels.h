@@ -542,6 +542,15 @@ void launch_agg3_scatter(const int64_t* keys, const uint8_t* key_valid,
                          PartRow* leftover, unsigned long long* lo_n,
                          uint32_t* bypass_matrix, uint32_t* err_flag,
                          hipStream_t s);
+// v4 scatter: barrier-free per-bucket LDS rings drained by dedicated
+// flusher waves (same in/out contract as launch_agg3_scatter)
+void launch_agg4_scatter(const int64_t* keys, const uint8_t* key_valid,
+                         const double* vals, const uint8_t* val_valid,
+                         int64_t n, int nbuck_log2, int grid_log2,
+                         const uint32_t* line_scan, uint8_t* out,
+                         PartRow* leftover, unsigned long long* lo_n,
+                         uint32_t* bypass_matrix, uint32_t* err_flag,
+                         hipStream_t s);
 void launch_agg3_bucket(const uint8_t* part, const uint32_t* counts,
                         const uint32_t* bypass,
                         const uint32_t* line_scan, int grid_log2, int is_int,

@@ -297,6 +297,199 @@ void launch_agg3_scatter(const int64_t* keys, const uint8_t* key_valid,
   check_launch3("k_agg3_scatter");
 }
 
+// ---- v4 scatter: barrier-free producer/flusher rings -----------------------
+// The v3 tile loop parks waves 73% of the time on its per-tile barriers
+// (SQ_WAIT_ANY ~7x ACTIVE, gpurun_out sq pass). v4 removes every barrier
+// from the hot loop: 14 WORKER waves append records into per-bucket 8-slot
+// LDS rings; 2 FLUSHER waves (8-lane subgroups, 8 buckets in flight each)
+// continuously drain 4-record 96 B quanta into the same 64B-aligned
+// per-(block,bucket) ranges. Commit protocol: each ring slot carries a
+// sequence TAG (= the absolute ring position) written AFTER the record by
+// the same lane — LDS executes one wave's DS ops in order — so the flusher
+// flushes exactly the contiguous committed prefix, tolerating out-of-order
+// commits across workers. Hot buckets bypass to the leftover list BEFORE
+// reserving (the reservation sequence must stay dense), with the same
+// per-(block,bucket) correction matrix as v3.
+static constexpr int A4_RING = 8;   // ring slots per bucket (pow2)
+static constexpr int A4_QUANT = 4;  // records per flush quantum (96 B)
+
+#define A4_LD_RLX(p) \
+  __hip_atomic_load((p), __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_WORKGROUP)
+#define A4_LD_ACQ(p) \
+  __hip_atomic_load((p), __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_WORKGROUP)
+#define A4_ST_REL(p, v) \
+  __hip_atomic_store((p), (v), __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_WORKGROUP)
+
+__global__ void __launch_bounds__(1024) k_agg4_scatter(
+    const int64_t* __restrict__ keys, const uint8_t* __restrict__ key_valid,
+    const double* __restrict__ vals, const uint8_t* __restrict__ val_valid,
+    int64_t n, int nbuck_log2, int grid_log2, int worker_waves,
+    const uint32_t* __restrict__ line_scan, uint8_t* __restrict__ out,
+    PartRow* __restrict__ leftover, unsigned long long* __restrict__ lo_n,
+    uint32_t* __restrict__ bypass_matrix, uint32_t* __restrict__ err_flag) {
+  const uint32_t nbuck = 1u << nbuck_log2;
+  extern __shared__ uint8_t lds[];
+  // layout: ring[nbuck][A4_RING][24] | tag[nbuck][A4_RING] u32 |
+  //         cnt[nbuck] | fl[nbuck] | base_line[nbuck] | byp[nbuck] | done
+  uint8_t* ring = lds;
+  uint32_t* tag = (uint32_t*)(lds + (size_t)nbuck * A4_RING * 24);
+  uint32_t* cnt = tag + (size_t)nbuck * A4_RING;
+  uint32_t* fl = cnt + nbuck;
+  uint32_t* base_line = fl + nbuck;
+  uint32_t* byp = base_line + nbuck;
+  uint32_t* done = byp + nbuck;
+
+  for (uint32_t b = threadIdx.x; b < nbuck; b += blockDim.x) {
+    cnt[b] = 0;
+    fl[b] = 0;
+    byp[b] = 0;
+    base_line[b] = line_scan[((size_t)b << grid_log2) | blockIdx.x];
+  }
+  for (uint32_t i = threadIdx.x; i < nbuck * A4_RING; i += blockDim.x)
+    tag[i] = 0xFFFFFFFFu;  // != any pos; never reset (successive uses of a
+                           // slot carry distinct pos values: pos = slot mod 8)
+  if (threadIdx.x == 0) *done = 0;
+  __syncthreads();
+
+  const int wave = (int)(threadIdx.x >> 6);
+  const int lane = (int)(threadIdx.x & 63);
+  const int nwave = (int)(blockDim.x >> 6);
+
+  if (wave < worker_waves) {
+    // ---- worker: append rows; no barriers, next-row loads prefetched ----
+    const int64_t wtid =
+        (int64_t)blockIdx.x * (worker_waves * 64) + wave * 64 + lane;
+    const int64_t stride = (int64_t)gridDim.x * (worker_waves * 64);
+    int64_t i = wtid;
+    int64_t k = 0;
+    double v = 0;
+    if (i < n) {
+      k = keys[i];
+      v = vals[i];
+    }
+    while (i < n) {
+      int64_t nx = i + stride;
+      int64_t k2 = 0;
+      double v2 = 0;
+      if (nx < n) {  // prefetch hides behind the LDS append chain
+        k2 = keys[nx];
+        v2 = vals[nx];
+      }
+      bool knull = key_valid && !bit_get3(key_valid, i);
+      if (!knull && k != KEY_EMPTY3) {  // specials kernel handles the rest
+        uint32_t b = (uint32_t)(mix64_3((uint64_t)k) >> (64 - nbuck_log2));
+        bool vvalid = !val_valid || bit_get3(val_valid, i);
+        uint32_t rowv = (uint32_t)i | (vvalid ? 0x80000000u : 0u);
+        // hot-bucket bypass BEFORE reserving (the reservation sequence must
+        // stay dense — a reserved slot can never be abandoned)
+        if (A4_LD_RLX(&cnt[b]) - A4_LD_RLX(&fl[b]) >= A4_RING + A4_RING / 2) {
+          unsigned long long p = atomicAdd(lo_n, 1ull);
+          leftover[p] = PartRow{k, v, rowv, 0};
+          atomicAdd(&byp[b], 1u);
+        } else {
+          uint32_t pos = atomicAdd(&cnt[b], 1u);
+          // wait for ring space (the flusher drains continuously; bounded
+          // backstop so a logic bug fails the chunk instead of the box)
+          int spin = 0;
+          while (pos - A4_LD_ACQ(&fl[b]) >= A4_RING) {
+            __builtin_amdgcn_s_sleep(2);
+            if (++spin > (1 << 22)) {
+              atomicOr(err_flag, 64u);
+              break;
+            }
+          }
+          uint32_t slot = pos & (A4_RING - 1);
+          uint8_t* rec = ring + ((size_t)b * A4_RING + slot) * 24;
+          *(int64_t*)rec = k;
+          *(double*)(rec + 8) = v;
+          *(uint32_t*)(rec + 16) = rowv;
+          // commit: RELEASE orders the record stores before the tag becomes
+          // visible (lgkmcnt is per-wave, so this fences all 64 lanes)
+          A4_ST_REL(&tag[b * A4_RING + slot], pos);
+        }
+      }
+      i = nx;
+      k = k2;
+      v = v2;
+    }
+    // completion signal: RELEASE pairs with the flusher's ACQUIRE so every
+    // tag committed by this wave is visible once done == worker_waves
+    if (lane == 0)
+      __hip_atomic_fetch_add(done, 1u, __ATOMIC_RELEASE,
+                             __HIP_MEMORY_SCOPE_WORKGROUP);
+  } else {
+    // ---- flusher: 8-lane subgroups drain committed prefixes ----
+    const int nfw = nwave - worker_waves;
+    const int sg_global = (wave - worker_waves) * 8 + (lane >> 3);
+    const int sl = lane & 7;
+    const uint32_t sg_stride = (uint32_t)nfw * 8;
+    bool draining = false;
+    for (;;) {
+      bool all_done = (A4_LD_ACQ(done) == (uint32_t)worker_waves);
+      uint32_t moved = 0;
+      for (uint32_t b = (uint32_t)sg_global; b < nbuck; b += sg_stride) {
+        uint32_t f = fl[b];  // this sg is the only writer of fl[b]
+        uint32_t ready = 0;
+        // all 8 lanes read the same tag address in lockstep -> uniform k
+        while (ready < A4_RING &&
+               A4_LD_ACQ(&tag[b * A4_RING + ((f + ready) & (A4_RING - 1))]) ==
+                   f + ready)
+          ready++;
+        uint32_t k = draining ? ready : ready & ~(uint32_t)(A4_QUANT - 1);
+        if (!k) continue;
+        uint64_t* dst = (uint64_t*)(out + ((size_t)base_line[b] << 6) +
+                                    (size_t)f * 24);
+        const uint64_t* src = (const uint64_t*)(ring + (size_t)b * A4_RING * 24);
+        for (uint32_t d = sl; d < k * 3; d += 8) {
+          uint32_t r = d / 3;
+          uint32_t slot = (f + r) & (A4_RING - 1);
+          dst[d] = src[(size_t)slot * 3 + d % 3];
+        }
+        // RELEASE waits the wave's outstanding DS reads (all lanes) before
+        // the advance lets workers overwrite the drained slots
+        if (sl == 0) A4_ST_REL(&fl[b], f + k);
+        moved += k;
+      }
+      if (draining && moved == 0) break;
+      if (all_done) draining = true;  // final sweeps flush partial quanta
+    }
+  }
+  // every (bucket, block) entry is written by its own block — on EVERY exit
+  // path — so the bucket kernel never reads a stale correction matrix
+  __syncthreads();
+  for (uint32_t b = threadIdx.x; b < nbuck; b += blockDim.x)
+    bypass_matrix[((size_t)b << grid_log2) | blockIdx.x] = byp[b];
+}
+
+void launch_agg4_scatter(const int64_t* keys, const uint8_t* key_valid,
+                         const double* vals, const uint8_t* val_valid,
+                         int64_t n, int nbuck_log2, int grid_log2,
+                         const uint32_t* line_scan, uint8_t* out,
+                         PartRow* leftover, unsigned long long* lo_n,
+                         uint32_t* bypass_matrix, uint32_t* err_flag,
+                         hipStream_t s) {
+  const uint32_t nbuck = 1u << nbuck_log2;
+  size_t lds = (size_t)nbuck * A4_RING * 24 + (size_t)nbuck * A4_RING * 4 +
+               (size_t)nbuck * 16 + 64;
+  if (lds > 160 * 1024)
+    throw std::runtime_error("agg4 scatter LDS over 160KB");
+  static int ww = [] {
+    const char* e = getenv("AURON_AGG2_V4_WW");
+    int v = e ? atoi(e) : 12;
+    return (v >= 8 && v <= 15) ? v : 12;
+  }();
+  hipError_t e = hipFuncSetAttribute(
+      (const void*)k_agg4_scatter,
+      hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
+  if (e != hipSuccess)
+    throw std::runtime_error("agg4 scatter LDS attribute failed");
+  hipLaunchKernelGGL(k_agg4_scatter, dim3(1 << grid_log2), dim3(1024), lds, s,
+                     keys, key_valid, vals, val_valid, n, nbuck_log2,
+                     grid_log2, ww, line_scan, out, leftover, lo_n,
+                     bypass_matrix, err_flag);
+  check_launch3("k_agg4_scatter");
+}
+
 // ---- v3 bucket aggregation (4096-slot LDS window, 1024 threads) ------------
 static constexpr int L3SLOTS = 4096;
 static constexpr int L3PROBE = 128;
