@@ -543,7 +543,10 @@ void launch_agg3_scatter(const int64_t* keys, const uint8_t* key_valid,
                          uint32_t* bypass_matrix, uint32_t* err_flag,
                          hipStream_t s);
 // v4 scatter: barrier-free per-bucket LDS rings drained by dedicated
-// flusher waves (same in/out contract as launch_agg3_scatter)
+// flusher waves (same in/out contract as launch_agg3_scatter). The hist
+// feeding it must run with agg4_worker_waves()*64 threads per block so the
+// row->block mapping matches the v4 worker traversal.
+int agg4_worker_waves();
 void launch_agg4_scatter(const int64_t* keys, const uint8_t* key_valid,
                          const double* vals, const uint8_t* val_valid,
                          int64_t n, int nbuck_log2, int grid_log2,

@@ -461,6 +461,19 @@ __global__ void __launch_bounds__(1024) k_agg4_scatter(
     bypass_matrix[((size_t)b << grid_log2) | blockIdx.x] = byp[b];
 }
 
+// The histogram MUST be launched with agg4_worker_waves()*64 threads per
+// block when the v4 scatter consumes its counts: the per-(block,bucket)
+// ranges are credited by the hist's row->block grid-stride mapping, and the
+// v4 workers traverse rows with exactly that stride.
+int agg4_worker_waves() {
+  static int ww = [] {
+    const char* e = getenv("AURON_AGG2_V4_WW");
+    int v = e ? atoi(e) : 12;
+    return (v >= 8 && v <= 15) ? v : 12;
+  }();
+  return ww;
+}
+
 void launch_agg4_scatter(const int64_t* keys, const uint8_t* key_valid,
                          const double* vals, const uint8_t* val_valid,
                          int64_t n, int nbuck_log2, int grid_log2,
@@ -473,11 +486,7 @@ void launch_agg4_scatter(const int64_t* keys, const uint8_t* key_valid,
                (size_t)nbuck * 16 + 64;
   if (lds > 160 * 1024)
     throw std::runtime_error("agg4 scatter LDS over 160KB");
-  static int ww = [] {
-    const char* e = getenv("AURON_AGG2_V4_WW");
-    int v = e ? atoi(e) : 12;
-    return (v >= 8 && v <= 15) ? v : 12;
-  }();
+  int ww = agg4_worker_waves();
   hipError_t e = hipFuncSetAttribute(
       (const void*)k_agg4_scatter,
       hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
