@@ -1367,7 +1367,18 @@ class AggOp {
     AURON_HIP(hipMemcpyAsync(err, t_.error_flag, 4, hipMemcpyDeviceToHost,
                              stream_));
     AURON_HIP(hipStreamSynchronize(stream_));
-    if (*err) FAIL("agg hash table probe exhausted (table full/corrupt)");
+    if (*err) {
+      // bit 1: table probe exhausted; 16: gkey pool publish spin; 32: v3
+      // scatter tile bound; 64: v4 scatter ring-space retry bound
+      if (*err & ~1u)
+        FAIL("agg device error flags 0x" +
+             ([](uint32_t v) {
+               char b[16];
+               snprintf(b, sizeof b, "%x", v);
+               return std::string(b);
+             })(*err));
+      FAIL("agg hash table probe exhausted (table full/corrupt)");
+    }
     return *ng;
   }
 

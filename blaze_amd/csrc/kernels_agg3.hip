@@ -356,7 +356,14 @@ __global__ void __launch_bounds__(1024) k_agg4_scatter(
   const int nwave = (int)(blockDim.x >> 6);
 
   if (wave < worker_waves) {
-    // ---- worker: append rows; no barriers, next-row loads prefetched ----
+    // ---- worker: append rows; no barriers, next-row loads prefetched.
+    // A lane that reserved a ring position it cannot yet write (backlog)
+    // must NOT spin in an inner loop: divergent serialization would park
+    // its sibling lanes, including ones holding the very positions the
+    // flusher needs next — an intra-wave deadlock (measured: spin-bound
+    // trips under skew/bursts). Instead the reservation becomes per-lane
+    // PENDING state retried once per outer iteration, so every lane makes
+    // one bounded attempt per pass and siblings always progress.
     const int64_t wtid =
         (int64_t)blockIdx.x * (worker_waves * 64) + wave * 64 + lane;
     const int64_t stride = (int64_t)gridDim.x * (worker_waves * 64);
@@ -367,50 +374,76 @@ __global__ void __launch_bounds__(1024) k_agg4_scatter(
       k = keys[i];
       v = vals[i];
     }
-    while (i < n) {
-      int64_t nx = i + stride;
-      int64_t k2 = 0;
-      double v2 = 0;
-      if (nx < n) {  // prefetch hides behind the LDS append chain
-        k2 = keys[nx];
-        v2 = vals[nx];
-      }
-      bool knull = key_valid && !bit_get3(key_valid, i);
-      if (!knull && k != KEY_EMPTY3) {  // specials kernel handles the rest
-        uint32_t b = (uint32_t)(mix64_3((uint64_t)k) >> (64 - nbuck_log2));
-        bool vvalid = !val_valid || bit_get3(val_valid, i);
-        uint32_t rowv = (uint32_t)i | (vvalid ? 0x80000000u : 0u);
-        // hot-bucket bypass BEFORE reserving (the reservation sequence must
-        // stay dense — a reserved slot can never be abandoned)
-        if (A4_LD_RLX(&cnt[b]) - A4_LD_RLX(&fl[b]) >= A4_RING + A4_RING / 2) {
-          unsigned long long p = atomicAdd(lo_n, 1ull);
-          leftover[p] = PartRow{k, v, rowv, 0};
-          atomicAdd(&byp[b], 1u);
-        } else {
-          uint32_t pos = atomicAdd(&cnt[b], 1u);
-          // wait for ring space (the flusher drains continuously; bounded
-          // backstop so a logic bug fails the chunk instead of the box)
-          int spin = 0;
-          while (pos - A4_LD_ACQ(&fl[b]) >= A4_RING) {
-            __builtin_amdgcn_s_sleep(2);
-            if (++spin > (1 << 22)) {
-              atomicOr(err_flag, 64u);
-              break;
+    bool pending = false;
+    uint32_t p_b = 0, p_pos = 0, p_rowv = 0;
+    int64_t p_k = 0;
+    double p_v = 0;
+    int p_spin = 0;
+    while (i < n || pending) {
+      bool progress = false;
+      if (pending) {
+        if (p_pos - A4_LD_ACQ(&fl[p_b]) < A4_RING) {
+          uint32_t slot = p_pos & (A4_RING - 1);
+          uint8_t* rec = ring + ((size_t)p_b * A4_RING + slot) * 24;
+          *(int64_t*)rec = p_k;
+          *(double*)(rec + 8) = p_v;
+          *(uint32_t*)(rec + 16) = p_rowv;
+          A4_ST_REL(&tag[p_b * A4_RING + slot], p_pos);
+          pending = false;
+          progress = true;
+        } else if (++p_spin > (1 << 22)) {
+          atomicOr(err_flag, 64u);  // fails the chunk, never hangs the box
+          pending = false;
+        }
+      } else if (i < n) {
+        int64_t nx = i + stride;
+        int64_t k2 = 0;
+        double v2 = 0;
+        if (nx < n) {  // prefetch hides behind the LDS append chain
+          k2 = keys[nx];
+          v2 = vals[nx];
+        }
+        bool knull = key_valid && !bit_get3(key_valid, i);
+        if (!knull && k != KEY_EMPTY3) {  // specials kernel handles the rest
+          uint32_t b = (uint32_t)(mix64_3((uint64_t)k) >> (64 - nbuck_log2));
+          bool vvalid = !val_valid || bit_get3(val_valid, i);
+          uint32_t rowv = (uint32_t)i | (vvalid ? 0x80000000u : 0u);
+          // hot-bucket bypass BEFORE reserving (the reservation sequence
+          // must stay dense — a reserved slot can never be abandoned)
+          if (A4_LD_RLX(&cnt[b]) - A4_LD_RLX(&fl[b]) >=
+              A4_RING + A4_RING / 2) {
+            unsigned long long p = atomicAdd(lo_n, 1ull);
+            leftover[p] = PartRow{k, v, rowv, 0};
+            atomicAdd(&byp[b], 1u);
+          } else {
+            uint32_t pos = atomicAdd(&cnt[b], 1u);
+            if (pos - A4_LD_ACQ(&fl[b]) < A4_RING) {
+              uint32_t slot = pos & (A4_RING - 1);
+              uint8_t* rec = ring + ((size_t)b * A4_RING + slot) * 24;
+              *(int64_t*)rec = k;
+              *(double*)(rec + 8) = v;
+              *(uint32_t*)(rec + 16) = rowv;
+              // commit: RELEASE orders the record stores before the tag
+              // becomes visible (lgkmcnt fences the whole wave)
+              A4_ST_REL(&tag[b * A4_RING + slot], pos);
+            } else {  // hold as pending; retried next pass
+              pending = true;
+              p_b = b;
+              p_pos = pos;
+              p_rowv = rowv;
+              p_k = k;
+              p_v = v;
+              p_spin = 0;
             }
           }
-          uint32_t slot = pos & (A4_RING - 1);
-          uint8_t* rec = ring + ((size_t)b * A4_RING + slot) * 24;
-          *(int64_t*)rec = k;
-          *(double*)(rec + 8) = v;
-          *(uint32_t*)(rec + 16) = rowv;
-          // commit: RELEASE orders the record stores before the tag becomes
-          // visible (lgkmcnt is per-wave, so this fences all 64 lanes)
-          A4_ST_REL(&tag[b * A4_RING + slot], pos);
         }
+        i = nx;
+        k = k2;
+        v = v2;
+        progress = true;
       }
-      i = nx;
-      k = k2;
-      v = v2;
+      // back off only when the WHOLE wave is blocked on ring space
+      if (__ballot(progress) == 0) __builtin_amdgcn_s_sleep(2);
     }
     // completion signal: RELEASE pairs with the flusher's ACQUIRE so every
     // tag committed by this wave is visible once done == worker_waves
