@@ -452,8 +452,10 @@ __global__ void __launch_bounds__(1024) k_agg4_scatter(
                              __HIP_MEMORY_SCOPE_WORKGROUP);
   } else {
     // ---- flusher: 8-lane subgroups drain committed prefixes ----
+    static_assert(A4_RING == 8, "ballot tag check assumes 8-slot rings");
     const int nfw = nwave - worker_waves;
     const int sg_global = (wave - worker_waves) * 8 + (lane >> 3);
+    const int sg_w = lane >> 3;  // subgroup within this wave
     const int sl = lane & 7;
     const uint32_t sg_stride = (uint32_t)nfw * 8;
     bool draining = false;
@@ -462,12 +464,14 @@ __global__ void __launch_bounds__(1024) k_agg4_scatter(
       uint32_t moved = 0;
       for (uint32_t b = (uint32_t)sg_global; b < nbuck; b += sg_stride) {
         uint32_t f = fl[b];  // this sg is the only writer of fl[b]
-        uint32_t ready = 0;
-        // all 8 lanes read the same tag address in lockstep -> uniform k
-        while (ready < A4_RING &&
-               A4_LD_ACQ(&tag[b * A4_RING + ((f + ready) & (A4_RING - 1))]) ==
-                   f + ready)
-          ready++;
+        // read all 8 tags in ONE wave instruction (lane sl covers slot
+        // (f+sl)&7) and ballot-match: a serialized per-tag acquire loop was
+        // ~60 cycles PER TAG and made the flusher the pipeline bottleneck
+        uint32_t t = A4_LD_ACQ(&tag[b * A4_RING +
+                                    ((f + (uint32_t)sl) & (A4_RING - 1))]);
+        uint64_t m = __ballot(t == f + (uint32_t)sl);
+        uint32_t bits = (uint32_t)(m >> (sg_w * 8)) & 0xFFu;
+        uint32_t ready = __builtin_ctz(~bits | 0x100u);  // committed prefix
         uint32_t k = draining ? ready : ready & ~(uint32_t)(A4_QUANT - 1);
         if (!k) continue;
         uint64_t* dst = (uint64_t*)(out + ((size_t)base_line[b] << 6) +
