@@ -319,6 +319,16 @@ static constexpr int A4_QUANT = 4;  // records per flush quantum (96 B)
   __hip_atomic_load((p), __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_WORKGROUP)
 #define A4_ST_REL(p, v) \
   __hip_atomic_store((p), (v), __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_WORKGROUP)
+// LDS-only release: wait the wave's outstanding DS ops, then a relaxed
+// store. A generic workgroup-scope RELEASE also emits s_waitcnt vmcnt(0),
+// which stalls ~800 cycles for in-flight HBM stores/prefetches that are
+// irrelevant to LDS ring reuse — measured as the v4 flusher bottleneck.
+#define A4_ST_REL_LDS(p, v)                        \
+  do {                                             \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory"); \
+    __hip_atomic_store((p), (v), __ATOMIC_RELAXED, \
+                       __HIP_MEMORY_SCOPE_WORKGROUP); \
+  } while (0)
 
 __global__ void __launch_bounds__(1024) k_agg4_scatter(
     const int64_t* __restrict__ keys, const uint8_t* __restrict__ key_valid,
@@ -388,7 +398,7 @@ __global__ void __launch_bounds__(1024) k_agg4_scatter(
           *(int64_t*)rec = p_k;
           *(double*)(rec + 8) = p_v;
           *(uint32_t*)(rec + 16) = p_rowv;
-          A4_ST_REL(&tag[p_b * A4_RING + slot], p_pos);
+          A4_ST_REL_LDS(&tag[p_b * A4_RING + slot], p_pos);
           pending = false;
           progress = true;
         } else if (++p_spin > (1 << 22)) {
@@ -397,12 +407,6 @@ __global__ void __launch_bounds__(1024) k_agg4_scatter(
         }
       } else if (i < n) {
         int64_t nx = i + stride;
-        int64_t k2 = 0;
-        double v2 = 0;
-        if (nx < n) {  // prefetch hides behind the LDS append chain
-          k2 = keys[nx];
-          v2 = vals[nx];
-        }
         bool knull = key_valid && !bit_get3(key_valid, i);
         if (!knull && k != KEY_EMPTY3) {  // specials kernel handles the rest
           uint32_t b = (uint32_t)(mix64_3((uint64_t)k) >> (64 - nbuck_log2));
@@ -423,9 +427,9 @@ __global__ void __launch_bounds__(1024) k_agg4_scatter(
               *(int64_t*)rec = k;
               *(double*)(rec + 8) = v;
               *(uint32_t*)(rec + 16) = rowv;
-              // commit: RELEASE orders the record stores before the tag
-              // becomes visible (lgkmcnt fences the whole wave)
-              A4_ST_REL(&tag[b * A4_RING + slot], pos);
+              // commit: LDS-only release — the record stores are DS ops,
+              // lgkmcnt(0) fences all 64 lanes of the wave
+              A4_ST_REL_LDS(&tag[b * A4_RING + slot], pos);
             } else {  // hold as pending; retried next pass
               pending = true;
               p_b = b;
@@ -438,8 +442,12 @@ __global__ void __launch_bounds__(1024) k_agg4_scatter(
           }
         }
         i = nx;
-        k = k2;
-        v = v2;
+        // prefetch AFTER the append: issuing these loads before the record
+        // writes would drag them into the append's vmcnt wait
+        if (nx < n) {
+          k = keys[nx];
+          v = vals[nx];
+        }
         progress = true;
       }
       // back off only when the WHOLE wave is blocked on ring space
@@ -482,9 +490,9 @@ __global__ void __launch_bounds__(1024) k_agg4_scatter(
           uint32_t slot = (f + r) & (A4_RING - 1);
           dst[d] = src[(size_t)slot * 3 + d % 3];
         }
-        // RELEASE waits the wave's outstanding DS reads (all lanes) before
-        // the advance lets workers overwrite the drained slots
-        if (sl == 0) A4_ST_REL(&fl[b], f + k);
+        // LDS-only release: the slot reads must complete before workers may
+        // overwrite them; the in-flight HBM stores are irrelevant to reuse
+        if (sl == 0) A4_ST_REL_LDS(&fl[b], f + k);
         moved += k;
       }
       if (draining && moved == 0) break;
@@ -505,8 +513,8 @@ __global__ void __launch_bounds__(1024) k_agg4_scatter(
 int agg4_worker_waves() {
   static int ww = [] {
     const char* e = getenv("AURON_AGG2_V4_WW");
-    int v = e ? atoi(e) : 12;
-    return (v >= 8 && v <= 15) ? v : 12;
+    int v = e ? atoi(e) : 8;  // measured optimum: flusher-throughput-bound
+    return (v >= 6 && v <= 15) ? v : 8;
   }();
   return ww;
 }
