@@ -467,33 +467,49 @@ __global__ void __launch_bounds__(1024) k_agg4_scatter(
     const int sl = lane & 7;
     const uint32_t sg_stride = (uint32_t)nfw * 8;
     bool draining = false;
+    // software-pipelined sweep: the (fl, tag) reads for bucket b+stride are
+    // issued while bucket b's copy is in flight, hiding the two dependent
+    // LDS round-trips (~120 cycles) that otherwise bound every visit
     for (;;) {
       bool all_done = (A4_LD_ACQ(done) == (uint32_t)worker_waves);
       uint32_t moved = 0;
-      for (uint32_t b = (uint32_t)sg_global; b < nbuck; b += sg_stride) {
-        uint32_t f = fl[b];  // this sg is the only writer of fl[b]
-        // read all 8 tags in ONE wave instruction (lane sl covers slot
-        // (f+sl)&7) and ballot-match: a serialized per-tag acquire loop was
-        // ~60 cycles PER TAG and made the flusher the pipeline bottleneck
-        uint32_t t = A4_LD_ACQ(&tag[b * A4_RING +
-                                    ((f + (uint32_t)sl) & (A4_RING - 1))]);
+      uint32_t b = (uint32_t)sg_global;
+      uint32_t f = fl[b];  // this sg is the only writer of fl[b]
+      uint32_t t = A4_LD_ACQ(&tag[b * A4_RING +
+                                  ((f + (uint32_t)sl) & (A4_RING - 1))]);
+      while (b < nbuck) {
+        uint32_t b2 = b + sg_stride;
+        // ballot-match all 8 tags read in ONE wave instruction (lane sl
+        // covered slot (f+sl)&7): committed-prefix length for this bucket
         uint64_t m = __ballot(t == f + (uint32_t)sl);
         uint32_t bits = (uint32_t)(m >> (sg_w * 8)) & 0xFFu;
-        uint32_t ready = __builtin_ctz(~bits | 0x100u);  // committed prefix
+        uint32_t ready = __builtin_ctz(~bits | 0x100u);
         uint32_t k = draining ? ready : ready & ~(uint32_t)(A4_QUANT - 1);
-        if (!k) continue;
-        uint64_t* dst = (uint64_t*)(out + ((size_t)base_line[b] << 6) +
-                                    (size_t)f * 24);
-        const uint64_t* src = (const uint64_t*)(ring + (size_t)b * A4_RING * 24);
-        for (uint32_t d = sl; d < k * 3; d += 8) {
-          uint32_t r = d / 3;
-          uint32_t slot = (f + r) & (A4_RING - 1);
-          dst[d] = src[(size_t)slot * 3 + d % 3];
+        // prefetch the NEXT bucket's fl+tag before this bucket's copy
+        uint32_t f2 = 0, t2 = 0;
+        if (b2 < nbuck) {
+          f2 = fl[b2];
+          t2 = A4_LD_ACQ(&tag[b2 * A4_RING +
+                              ((f2 + (uint32_t)sl) & (A4_RING - 1))]);
         }
-        // LDS-only release: the slot reads must complete before workers may
-        // overwrite them; the in-flight HBM stores are irrelevant to reuse
-        if (sl == 0) A4_ST_REL_LDS(&fl[b], f + k);
-        moved += k;
+        if (k) {
+          uint64_t* dst = (uint64_t*)(out + ((size_t)base_line[b] << 6) +
+                                      (size_t)f * 24);
+          const uint64_t* src =
+              (const uint64_t*)(ring + (size_t)b * A4_RING * 24);
+          for (uint32_t d = sl; d < k * 3; d += 8) {
+            uint32_t r = d / 3;
+            uint32_t slot = (f + r) & (A4_RING - 1);
+            dst[d] = src[(size_t)slot * 3 + d % 3];
+          }
+          // LDS-only release: the slot reads must complete before workers
+          // may overwrite them; in-flight HBM stores are irrelevant
+          if (sl == 0) A4_ST_REL_LDS(&fl[b], f + k);
+          moved += k;
+        }
+        b = b2;
+        f = f2;
+        t = t2;
       }
       if (draining && moved == 0) break;
       if (all_done) draining = true;  // final sweeps flush partial quanta
