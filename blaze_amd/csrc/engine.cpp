@@ -1928,12 +1928,11 @@ class AggOp {
       // v3: hist -> line-padded scan -> LDS-staged packet scatter -> 4096-
       // slot bucket aggregation (kernels_agg3.hip)
       int64_t mat3 = (int64_t)AGG3_NBUCK << AGG3_GRID_LOG2;
-      // v4 workers traverse rows with a (worker_waves*64)-thread stride, so
-      // the hist must credit rows to blocks with that same mapping
+      // the v4 workers cover the same 1024-wide virtual-lane tiles as the
+      // 1024-thread hist, so the hist launch is identical for v3 and v4
       launch_agg2_hist(keys, kv, chunk, AGG3_NBUCK_LOG2, AGG3_GRID_LOG2,
                        d_counts_.get<uint32_t>(),
-                       (uint32_t*)(d_counters_.get<uint8_t>() + 16),
-                       agg2_v4_ ? agg4_worker_waves() * 64 : 1024,
+                       (uint32_t*)(d_counters_.get<uint8_t>() + 16), 1024,
                        stream_);
       launch_agg3_line_sizes(d_counts_.get<uint32_t>(), mat3 + 1,
                              d_linesz_.get<uint32_t>(), stream_);

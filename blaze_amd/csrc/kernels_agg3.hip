@@ -311,7 +311,9 @@ void launch_agg3_scatter(const int64_t* keys, const uint8_t* key_valid,
 // reserving (the reservation sequence must stay dense), with the same
 // per-(block,bucket) correction matrix as v3.
 static constexpr int A4_RING = 8;   // ring slots per bucket (pow2)
-static constexpr int A4_QUANT = 4;  // records per flush quantum (96 B)
+static constexpr int A4_QUANT = 8;  // records per flush quantum (192 B —
+                                    // one full ring; halves flusher visits
+                                    // per record vs the 96 B quantum)
 
 #define A4_LD_RLX(p) \
   __hip_atomic_load((p), __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_WORKGROUP)
@@ -366,89 +368,120 @@ __global__ void __launch_bounds__(1024) k_agg4_scatter(
   const int nwave = (int)(blockDim.x >> 6);
 
   if (wave < worker_waves) {
-    // ---- worker: append rows; no barriers, next-row loads prefetched.
+    // ---- worker: append rows; no barriers. VIRTUAL-LANE row mapping: rows
+    // are credited to blocks by the 1024-thread histogram stride, so worker
+    // thread t covers virtual lanes {t, t+NW, ...} < 1024 of each tile —
+    // the hist keeps its full 1024-thread launch (384-thread hist measured
+    // 2.2x slower), and each worker carries 2-3 rows per tile of ILP.
+    //
     // A lane that reserved a ring position it cannot yet write (backlog)
     // must NOT spin in an inner loop: divergent serialization would park
     // its sibling lanes, including ones holding the very positions the
     // flusher needs next — an intra-wave deadlock (measured: spin-bound
-    // trips under skew/bursts). Instead the reservation becomes per-lane
-    // PENDING state retried once per outer iteration, so every lane makes
-    // one bounded attempt per pass and siblings always progress.
-    const int64_t wtid =
-        (int64_t)blockIdx.x * (worker_waves * 64) + wave * 64 + lane;
-    const int64_t stride = (int64_t)gridDim.x * (worker_waves * 64);
-    int64_t i = wtid;
-    int64_t k = 0;
-    double v = 0;
-    if (i < n) {
-      k = keys[i];
-      v = vals[i];
-    }
-    bool pending = false;
-    uint32_t p_b = 0, p_pos = 0, p_rowv = 0;
-    int64_t p_k = 0;
-    double p_v = 0;
+    // trips under skew/bursts). Reservations it cannot complete go into a
+    // small per-lane pending FIFO retried once per pass (leftover-bypass
+    // when full), so every lane makes bounded attempts and siblings always
+    // progress.
+    const int NW = worker_waves * 64;
+    const int tid = (int)threadIdx.x;
+    const int64_t tile_stride = (int64_t)gridDim.x << 10;
+    constexpr int MAXSUB = 3;   // ceil(1024 / (6 waves * 64))
+    constexpr int PCAP = 4;     // pending FIFO slots (pow2)
+    int64_t rk[MAXSUB];
+    double rv[MAXSUB];
+    uint32_t rvalid[MAXSUB];    // bit0: in range+normal key; bit1: val valid
+    int64_t rrow[MAXSUB];
+    uint32_t p_b[PCAP], p_pos[PCAP], p_rowv[PCAP];
+    int64_t p_k[PCAP];
+    double p_v[PCAP];
+    uint32_t p_hd = 0, p_tl = 0;
     int p_spin = 0;
-    while (i < n || pending) {
+    int64_t tb = (int64_t)blockIdx.x << 10;
+    int nr = 0;
+    auto load_tile = [&](int64_t base) {
+      nr = 0;
+      if (base >= n) return;
+#pragma unroll
+      for (int s = 0; s < MAXSUB; s++) {
+        int vt = tid + s * NW;
+        int64_t i2 = base + vt;
+        if (vt < 1024 && i2 < n) {
+          rk[nr] = keys[i2];
+          rv[nr] = vals[i2];
+          bool knull = key_valid && !bit_get3(key_valid, i2);
+          bool vvalid = !val_valid || bit_get3(val_valid, i2);
+          rvalid[nr] = (!knull && rk[nr] != KEY_EMPTY3 ? 1u : 0u) |
+                       (vvalid ? 2u : 0u);
+          rrow[nr] = i2;
+          nr++;
+        }
+      }
+    };
+    auto append = [&](uint32_t b, uint32_t pos, int64_t k, double v,
+                      uint32_t rowv) {
+      uint32_t slot = pos & (A4_RING - 1);
+      uint8_t* rec = ring + ((size_t)b * A4_RING + slot) * 24;
+      *(int64_t*)rec = k;
+      *(double*)(rec + 8) = v;
+      *(uint32_t*)(rec + 16) = rowv;
+      // commit: LDS-only release — the record stores are DS ops,
+      // lgkmcnt(0) fences all 64 lanes of the wave
+      A4_ST_REL_LDS(&tag[b * A4_RING + slot], pos);
+    };
+    load_tile(tb);
+    while (nr > 0 || p_hd != p_tl) {
       bool progress = false;
-      if (pending) {
-        if (p_pos - A4_LD_ACQ(&fl[p_b]) < A4_RING) {
-          uint32_t slot = p_pos & (A4_RING - 1);
-          uint8_t* rec = ring + ((size_t)p_b * A4_RING + slot) * 24;
-          *(int64_t*)rec = p_k;
-          *(double*)(rec + 8) = p_v;
-          *(uint32_t*)(rec + 16) = p_rowv;
-          A4_ST_REL_LDS(&tag[p_b * A4_RING + slot], p_pos);
-          pending = false;
+      if (p_hd != p_tl) {  // retry the oldest pending reservation
+        uint32_t q = p_tl & (PCAP - 1);
+        if (p_pos[q] - A4_LD_ACQ(&fl[p_b[q]]) < A4_RING) {
+          append(p_b[q], p_pos[q], p_k[q], p_v[q], p_rowv[q]);
+          p_tl++;
+          p_spin = 0;
           progress = true;
         } else if (++p_spin > (1 << 22)) {
           atomicOr(err_flag, 64u);  // fails the chunk, never hangs the box
-          pending = false;
+          p_tl = p_hd;
         }
-      } else if (i < n) {
-        int64_t nx = i + stride;
-        bool knull = key_valid && !bit_get3(key_valid, i);
-        if (!knull && k != KEY_EMPTY3) {  // specials kernel handles the rest
-          uint32_t b = (uint32_t)(mix64_3((uint64_t)k) >> (64 - nbuck_log2));
-          bool vvalid = !val_valid || bit_get3(val_valid, i);
-          uint32_t rowv = (uint32_t)i | (vvalid ? 0x80000000u : 0u);
-          // hot-bucket bypass BEFORE reserving (the reservation sequence
-          // must stay dense — a reserved slot can never be abandoned)
-          if (A4_LD_RLX(&cnt[b]) - A4_LD_RLX(&fl[b]) >=
-              A4_RING + A4_RING / 2) {
-            unsigned long long p = atomicAdd(lo_n, 1ull);
-            leftover[p] = PartRow{k, v, rowv, 0};
-            atomicAdd(&byp[b], 1u);
-          } else {
-            uint32_t pos = atomicAdd(&cnt[b], 1u);
-            if (pos - A4_LD_ACQ(&fl[b]) < A4_RING) {
-              uint32_t slot = pos & (A4_RING - 1);
-              uint8_t* rec = ring + ((size_t)b * A4_RING + slot) * 24;
-              *(int64_t*)rec = k;
-              *(double*)(rec + 8) = v;
-              *(uint32_t*)(rec + 16) = rowv;
-              // commit: LDS-only release — the record stores are DS ops,
-              // lgkmcnt(0) fences all 64 lanes of the wave
-              A4_ST_REL_LDS(&tag[b * A4_RING + slot], pos);
-            } else {  // hold as pending; retried next pass
-              pending = true;
-              p_b = b;
-              p_pos = pos;
-              p_rowv = rowv;
-              p_k = k;
-              p_v = v;
-              p_spin = 0;
-            }
-          }
+      }
+      // process the tile only when the FIFO can absorb every row going
+      // pending — this keeps the reserved-but-unwritable inline spin below
+      // unreachable (an inline spin can deadlock the wave: siblings parked
+      // by divergence may hold the window positions the flusher needs)
+      bool take = (p_hd - p_tl) + (uint32_t)nr <= PCAP;
+      for (int r = 0; take && r < nr; r++) {
+        if (!(rvalid[r] & 1u)) continue;  // special: handled elsewhere
+        int64_t k = rk[r];
+        uint32_t b = (uint32_t)(mix64_3((uint64_t)k) >> (64 - nbuck_log2));
+        uint32_t rowv =
+            (uint32_t)rrow[r] | ((rvalid[r] & 2u) ? 0x80000000u : 0u);
+        // hot-bucket bypass BEFORE reserving (the reservation sequence
+        // must stay dense — a reserved slot can never be abandoned)
+        if (A4_LD_RLX(&cnt[b]) - A4_LD_RLX(&fl[b]) >=
+            A4_RING + A4_RING / 2) {
+          unsigned long long p = atomicAdd(lo_n, 1ull);
+          leftover[p] = PartRow{k, rv[r], rowv, 0};
+          atomicAdd(&byp[b], 1u);
+          continue;
         }
-        i = nx;
-        // prefetch AFTER the append: issuing these loads before the record
-        // writes would drag them into the append's vmcnt wait
-        if (nx < n) {
-          k = keys[nx];
-          v = vals[nx];
+        uint32_t pos = atomicAdd(&cnt[b], 1u);
+        if (pos - A4_LD_ACQ(&fl[b]) < A4_RING) {
+          append(b, pos, k, rv[r], rowv);
+        } else {  // backlogged: park in the FIFO (room guaranteed above)
+          uint32_t q = p_hd & (PCAP - 1);
+          p_b[q] = b;
+          p_pos[q] = pos;
+          p_rowv[q] = rowv;
+          p_k[q] = k;
+          p_v[q] = rv[r];
+          p_hd++;
         }
-        progress = true;
+      }
+      if (take) {
+        if (nr) progress = true;
+        tb += tile_stride;
+        // prefetch the next tile AFTER the appends: issuing these loads
+        // earlier would drag them into the appends' vmcnt waits
+        load_tile(tb);
       }
       // back off only when the WHOLE wave is blocked on ring space
       if (__ballot(progress) == 0) __builtin_amdgcn_s_sleep(2);
