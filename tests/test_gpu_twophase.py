@@ -53,6 +53,42 @@ def test_two_phase_parity_6m(v3):
         os.environ.pop("AURON_AGG2_V3", None)
 
 
+@pytest.mark.parametrize("skew", [False, True])
+def test_two_phase_v4_ring_scatter(skew):
+    """The experimental barrier-free ring scatter (AURON_AGG2_V4=1,
+    kernels_agg3.hip k_agg4_scatter) must match the oracle, uniform and
+    under adversarial skew (pending FIFO + pre-reserve bypass paths)."""
+    os.environ["AURON_AGG2_V4"] = "1"
+    try:
+        n = 6_000_000
+        if skew:
+            rng = np.random.default_rng(53)
+            keys = np.full(n, 4242, dtype=np.int64)
+            spread = rng.integers(0, 50, n)
+            keys[spread > 10] = rng.integers(0, 100, int((spread > 10).sum()))
+            kv = vv = None
+            vals = np.ones(n, dtype=np.float64)
+        else:
+            keys, kv, vals, vv = _gen(n, 300_000, 54)
+        t = blaze_amd.Task(plan.plan_partial_final(),
+                           batches=[[(keys, kv), (vals, vv)]],
+                           conf={"BATCH_SIZE": 1 << 20,
+                                 "AURON_HIP_AGG_TABLE_SLOTS": 1 << 20})
+        outs = t.run()
+        t.finalize()
+        got_k = np.concatenate([o[0]["values"] for o in outs])
+        got_s = np.concatenate([o[1]["values"] for o in outs])
+        got_c = np.concatenate([o[2]["values"] for o in outs])
+        orc = oracle.Agg()
+        orc.update(keys, vals, key_valid=kv, val_valid=vv)
+        ref = orc.output()
+        np.testing.assert_array_equal(got_k, ref["keys"])
+        np.testing.assert_array_equal(got_c, ref["counts"])
+        np.testing.assert_array_equal(got_s, ref["sums"])
+    finally:
+        os.environ.pop("AURON_AGG2_V4", None)
+
+
 def test_two_phase_skewed_buckets():
     """Adversarial skew: one dominant key floods one partition bucket (the
     leftover path and packet flushes must stay correct)."""
