@@ -151,5 +151,10 @@ class ParquetFile {
 // literal/copy tags) — no libsnappy in this image
 bool snappy_uncompress(const uint8_t* src, size_t n, std::vector<uint8_t>* out,
                        std::string* err);
+// core over caller storage: dst needs out_len + 32 writable bytes (the hot
+// paths overshoot with fixed-size copies); lets the page loop reuse one
+// grow-only buffer instead of an alloc + zero-fill per page
+bool snappy_uncompress_raw(const uint8_t* src, size_t n, uint8_t* dst,
+                           size_t out_len, std::string* err);
 
 }  // namespace auron
