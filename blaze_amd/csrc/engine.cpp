@@ -3584,6 +3584,150 @@ struct Runtime {
         for (auto& w : ws) w.join();
       }
       if (!decode_err.empty()) FAIL(decode_err);
+      // ---- GPU page decompression batch (kernels_pq.hip) ----
+      // every gpu_comp chunk's PLAIN-suffix pages across ALL row groups go
+      // into ONE wave-per-page decompress+decode launch set (per-chunk
+      // launches would leave the chip nearly idle: ~30 pages per chunk).
+      // Results live in two blobs (validity bitmaps / dense values) that
+      // stay resident for the row-group assembly loop below.
+      struct GcResult {
+        uint64_t valid_base = 0, dense_base = 0;
+        int64_t total_nn = 0, null_total = 0;
+        bool ready = false;
+      };
+      std::vector<GcResult> gc_res((size_t)nrg * width);
+      DevBuf d_gc_valid, d_gc_dense;
+      {
+        std::vector<PqGpuPage> pages;
+        std::vector<PqGpuChunk> chmeta;
+        std::vector<size_t> ch_di;
+        uint64_t comp_total = 0, scratch_total = 0, valid_total = 0,
+                 dense_total = 0;
+        auto al8 = [](uint64_t v) { return (v + 7) & ~7ull; };
+        for (size_t di = 0; di < decoded.size(); di++) {
+          PqColumnChunkData& cd = decoded[di];
+          if (!cd.gpu_comp) continue;
+          const int vw = cd.value_width;
+          int64_t values = cd.prefix_values + cd.suffix_values;
+          PqGpuChunk ch;
+          ch.page0 = (uint32_t)pages.size();
+          ch.npages = (uint32_t)cd.comp_pages.size();
+          ch.valid_base = valid_total;
+          ch.dense_base = dense_total;
+          ch.prefix_nn = (uint32_t)(cd.plain.size() / (size_t)vw);
+          ch.vw = (uint32_t)vw;
+          uint32_t vbase = (uint32_t)cd.prefix_values;
+          for (const auto& pg : cd.comp_pages) {
+            PqGpuPage p;
+            p.comp_off = comp_total + (uint64_t)(pg.src - cd.comp_pages[0].src);
+            p.uncomp_off = scratch_total;
+            p.comp_len = pg.comp_len;
+            p.uncomp_len = pg.uncomp_len;
+            p.num_values = pg.num_values;
+            p.value_base = vbase;
+            p.has_def = cd.nullable ? 1u : 0u;
+            p.chunk_id = (uint32_t)chmeta.size();
+            pages.push_back(p);
+            vbase += pg.num_values;
+            scratch_total += al8(pg.uncomp_len);
+          }
+          const auto& lastp = cd.comp_pages.back();
+          comp_total += al8((uint64_t)(lastp.src - cd.comp_pages[0].src) +
+                            lastp.comp_len);
+          valid_total += al8(((uint64_t)values + 7) / 8);
+          dense_total += al8((uint64_t)values * vw);
+          chmeta.push_back(ch);
+          ch_di.push_back(di);
+        }
+        if (!pages.empty()) {
+          DevBuf d_comp(comp_total), d_scratch(scratch_total);
+          d_gc_valid.alloc(valid_total ? valid_total : 8);
+          d_gc_dense.alloc(dense_total);
+          DevBuf d_pages(pages.size() * sizeof(PqGpuPage));
+          DevBuf d_chunks(chmeta.size() * sizeof(PqGpuChunk));
+          DevBuf d_nn(pages.size() * 4), d_voff(pages.size() * 4);
+          DevBuf d_pdense(pages.size() * 4), d_chunknn(chmeta.size() * 4);
+          DevBuf d_gcerr(4);
+          AURON_HIP(hipMemsetAsync(d_gc_valid.get(), 0, valid_total, stream));
+          AURON_HIP(hipMemsetAsync(d_gcerr.get(), 0, 4, stream));
+          PinnedUploader::inst().copy(d_pages.get(), pages.data(),
+                                      pages.size() * sizeof(PqGpuPage),
+                                      stream);
+          PinnedUploader::inst().copy(d_chunks.get(), chmeta.data(),
+                                      chmeta.size() * sizeof(PqGpuChunk),
+                                      stream);
+          std::vector<std::vector<uint8_t>> pvbits(chmeta.size());
+          for (size_t c = 0; c < chmeta.size(); c++) {
+            PqColumnChunkData& cd = decoded[ch_di[c]];
+            // compressed span (pages of a chunk are contiguous in the file)
+            const auto& p0 = cd.comp_pages[0];
+            const auto& pl = cd.comp_pages.back();
+            size_t span = (size_t)(pl.src - p0.src) + pl.comp_len;
+            PinnedUploader::inst().copy(
+                d_comp.get<uint8_t>() + pages[chmeta[c].page0].comp_off -
+                    (uint64_t)(p0.src - cd.comp_pages[0].src),
+                p0.src, span, stream);
+            // host-decoded prefix: dense values + validity bits
+            if (!cd.plain.empty())
+              PinnedUploader::inst().copy(
+                  d_gc_dense.get<uint8_t>() + chmeta[c].dense_base,
+                  cd.plain.data(), cd.plain.size(), stream);
+            if (cd.nullable && cd.prefix_values > 0) {
+              size_t pb = ((size_t)cd.prefix_values + 7) / 8;
+              auto& vb = pvbits[c];
+              if (!cd.validity.empty()) {
+                vb.assign(cd.validity.begin(), cd.validity.begin() + pb);
+              } else {
+                vb.assign(pb, 0xFF);
+              }
+              if (cd.prefix_values & 7)  // clear bits beyond the prefix:
+                vb[pb - 1] &= (uint8_t)((1u << (cd.prefix_values & 7)) - 1);
+              PinnedUploader::inst().copy(
+                  d_gc_valid.get<uint8_t>() + chmeta[c].valid_base, vb.data(),
+                  pb, stream);
+            }
+          }
+          launch_pq_pages_decode(d_comp.get<uint8_t>(),
+                                 d_pages.get<PqGpuPage>(), (int)pages.size(),
+                                 d_scratch.get<uint8_t>(),
+                                 d_gc_valid.get<uint8_t>(),
+                                 d_chunks.get<PqGpuChunk>(),
+                                 d_nn.get<uint32_t>(), d_voff.get<uint32_t>(),
+                                 d_gcerr.get<uint32_t>(), stream);
+          launch_pq_page_offsets(d_chunks.get<PqGpuChunk>(),
+                                 (int)chmeta.size(), d_nn.get<uint32_t>(),
+                                 d_pdense.get<uint32_t>(),
+                                 d_chunknn.get<uint32_t>(), stream);
+          launch_pq_pages_compact(d_pages.get<PqGpuPage>(),
+                                  (int)pages.size(),
+                                  d_scratch.get<uint8_t>(),
+                                  d_chunks.get<PqGpuChunk>(),
+                                  d_nn.get<uint32_t>(),
+                                  d_voff.get<uint32_t>(),
+                                  d_pdense.get<uint32_t>(),
+                                  d_gc_dense.get<uint8_t>(), stream);
+          std::vector<uint32_t> h_chunknn(chmeta.size());
+          uint32_t h_err = 0;
+          AURON_HIP(hipMemcpyAsync(h_chunknn.data(), d_chunknn.get(),
+                                   chmeta.size() * 4, hipMemcpyDeviceToHost,
+                                   stream));
+          AURON_HIP(hipMemcpyAsync(&h_err, d_gcerr.get(), 4,
+                                   hipMemcpyDeviceToHost, stream));
+          AURON_HIP(hipStreamSynchronize(stream));
+          if (h_err) FAIL("parquet: GPU page decode failed (flags " +
+                          std::to_string(h_err) + ")");
+          for (size_t c = 0; c < chmeta.size(); c++) {
+            PqColumnChunkData& cd = decoded[ch_di[c]];
+            GcResult& r = gc_res[ch_di[c]];
+            r.valid_base = chmeta[c].valid_base;
+            r.dense_base = chmeta[c].dense_base;
+            r.total_nn = (int64_t)chmeta[c].prefix_nn + h_chunknn[c];
+            r.null_total =
+                cd.prefix_values + cd.suffix_values - r.total_nn;
+            r.ready = true;
+          }
+        }
+      }
       for (int rg = 0; rg < nrg; rg++) {
         bool keep = true;
         for (const Expr& pr : node.pruning)
@@ -3602,6 +3746,46 @@ struct Runtime {
           c.dt = fcols[ci].dtype();
           c.len = rows;
           bool has_nulls = cd.null_count > 0;
+          if (cd.gpu_comp) {
+            // device-decompressed chunk: validity bitmap + dense values are
+            // already resident in the batch blobs (pre-pass above)
+            const GcResult& r = gc_res[(size_t)rg * width + pi];
+            if (!r.ready) FAIL("parquet: gpu page results missing");
+            if (cd.prefix_values + cd.suffix_values != rows)
+              FAIL("parquet: gpu chunk row count mismatch");
+            const int w2 = (int)dtype_width(fcols[ci].dtype());
+            c.own_values.alloc((size_t)rows * w2);
+            const uint8_t* dense = d_gc_dense.get<uint8_t>() + r.dense_base;
+            if (r.null_total == 0) {
+              AURON_HIP(hipMemcpyAsync(c.own_values.get(), dense,
+                                       (size_t)rows * w2,
+                                       hipMemcpyDeviceToDevice, stream));
+            } else {
+              c.own_validity.alloc((rows + 7) / 8);
+              AURON_HIP(hipMemcpyAsync(c.own_validity.get(),
+                                       d_gc_valid.get<uint8_t>() +
+                                           r.valid_base,
+                                       (size_t)(rows + 7) / 8,
+                                       hipMemcpyDeviceToDevice, stream));
+              c.validity = c.own_validity.get<uint8_t>();
+              DevBuf d_mask2(rows), d_pos2((rows + 1) * 4);
+              launch_bits_to_mask(c.validity, rows, d_mask2.get<uint8_t>(),
+                                  stream);
+              size_t tb = 0;
+              scan_mask_u8(d_mask2.get<uint8_t>(), d_pos2.get<uint32_t>(),
+                           rows, nullptr, &tb, stream);
+              if (tb > scan_tmp.size()) scan_tmp.alloc(tb);
+              scan_mask_u8(d_mask2.get<uint8_t>(), d_pos2.get<uint32_t>(),
+                           rows, scan_tmp.get(), &tb, stream);
+              launch_scatter_packed(w2, dense, d_pos2.get<uint32_t>(),
+                                    d_mask2.get<uint8_t>(), rows,
+                                    c.own_values.get<uint8_t>(), stream);
+              AURON_HIP(hipStreamSynchronize(stream));  // temps die here
+            }
+            c.values = c.own_values.get();
+            b.cols.push_back(std::move(c));
+            continue;
+          }
           if (c.dt == DType::Utf8 || c.dt == DType::Binary) {
             // row-aligned offsets/data assembled on host (parquet.cpp)
             c.own_offsets.alloc((rows + 1) * 4);

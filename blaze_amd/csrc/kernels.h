@@ -562,6 +562,41 @@ void launch_agg3_bucket(const uint8_t* part, const uint32_t* counts,
                         PartRow* leftover, unsigned long long* lo_n,
                         uint32_t* error_flag, hipStream_t s);
 
+// GPU Parquet page staging (kernels_pq.hip): wave-per-page snappy decompress
+// + on-device def-level parse + dense compaction. Built by the host pre-scan
+// in parquet.cpp; consumed by engine pump_parquet.
+struct PqGpuPage {
+  uint64_t comp_off;    // into the uploaded compressed blob
+  uint64_t uncomp_off;  // into the scratch blob (8-aligned)
+  uint32_t comp_len;
+  uint32_t uncomp_len;
+  uint32_t num_values;  // value slots this page covers (incl nulls)
+  uint32_t value_base;  // first value slot within the chunk (incl prefix)
+  uint32_t has_def;     // 1 = page begins with [u32 ll][RLE def levels]
+  uint32_t chunk_id;
+};
+struct PqGpuChunk {
+  uint64_t valid_base;  // byte offset of this chunk's bitmap (4-aligned)
+  uint64_t dense_base;  // byte offset of this chunk's dense values blob
+  uint32_t page0;       // first page index
+  uint32_t npages;
+  uint32_t prefix_nn;   // host-decoded dense prefix length (values)
+  uint32_t vw;          // fixed value width (4 or 8)
+};
+void launch_pq_pages_decode(const uint8_t* comp, const PqGpuPage* pages,
+                            int npages, uint8_t* scratch, uint8_t* valid_blob,
+                            const PqGpuChunk* chunks, uint32_t* nn_counts,
+                            uint32_t* val_offs, uint32_t* err, hipStream_t s);
+void launch_pq_page_offsets(const PqGpuChunk* chunks, int nchunks,
+                            const uint32_t* nn_counts, uint32_t* page_dense,
+                            uint32_t* chunk_nn, hipStream_t s);
+void launch_pq_pages_compact(const PqGpuPage* pages, int npages,
+                             const uint8_t* scratch, const PqGpuChunk* chunks,
+                             const uint32_t* nn_counts,
+                             const uint32_t* val_offs,
+                             const uint32_t* page_dense,
+                             uint8_t* dense_blob, hipStream_t s);
+
 // sort groups by first_row: rocprim radix sort pairs wrapper
 void sort_pairs_u64_u32(const unsigned long long* keys_in, const uint32_t* vals_in,
                         unsigned long long* keys_out, uint32_t* vals_out,

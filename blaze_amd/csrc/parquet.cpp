@@ -901,6 +901,59 @@ void pq_materialize_gpu_staging(PqColumnChunkData* cd) {
     cd->def_runs.clear();
     cd->def_bytes.clear();
   }
+  if (cd->gpu_comp) {
+    // host re-statement of kernels_pq.hip: decompress each recorded page,
+    // parse its def levels, append dense values — CPU tests pin the
+    // pre-scan/split logic without a GPU
+    const int w = cd->value_width;
+    if (w <= 0) fail("parquet: gpu_comp on non-fixed-width column");
+    std::vector<uint8_t> vb;  // byte per value
+    if (cd->nullable) {
+      vb.assign((size_t)cd->prefix_values, 1);
+      if (!cd->validity.empty())
+        for (int64_t i = 0; i < cd->prefix_values; i++)
+          vb[(size_t)i] = (cd->validity[i >> 3] >> (i & 7)) & 1;
+    }
+    std::vector<uint8_t> buf;
+    std::string err;
+    for (const auto& pg : cd->comp_pages) {
+      if (!snappy_uncompress(pg.src, pg.comp_len, &buf, &err))
+        fail("parquet: " + err);
+      if (buf.size() != pg.uncomp_len) fail("parquet: page size mismatch");
+      const uint8_t* data = buf.data();
+      int64_t dlen = (int64_t)pg.uncomp_len;
+      int64_t nn = pg.num_values;
+      if (cd->nullable) {
+        uint32_t ll;
+        if (dlen < 4) fail("parquet: truncated def-level length");
+        memcpy(&ll, data, 4);
+        if ((int64_t)ll > dlen - 4) fail("parquet: def levels overrun page");
+        std::vector<uint32_t> def;
+        rle_bp_decode(data + 4, ll, 1, pg.num_values, &def);
+        nn = 0;
+        for (uint32_t d : def) {
+          vb.push_back((uint8_t)(d != 0));
+          nn += d != 0;
+        }
+        cd->null_count += pg.num_values - nn;
+        data += 4 + ll;
+        dlen -= 4 + ll;
+      }
+      if (dlen < nn * w) fail("parquet: short PLAIN data in gpu page");
+      cd->plain.insert(cd->plain.end(), data, data + nn * w);
+      cd->num_values += pg.num_values;
+    }
+    if (cd->nullable && cd->null_count > 0) {
+      cd->validity.assign((vb.size() + 7) / 8, 0);
+      for (size_t i = 0; i < vb.size(); i++)
+        if (vb[i]) cd->validity[i >> 3] |= (uint8_t)(1u << (i & 7));
+    } else {
+      cd->validity.clear();
+    }
+    cd->gpu_comp = false;
+    cd->comp_pages.clear();
+    cd->suffix_values = 0;
+  }
 }
 
 PqColStats ParquetFile::column_stats(int rg, int col) const {
@@ -930,13 +983,74 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
 
   PqColumnChunkData out;
   out.num_values = 0;
+  out.value_width = ci.physical_type == 6 ? 0 : (int)dtype_width(ci.dtype());
   // GPU run-expansion mode (fixed-width columns): the host walks RLE/
   // bit-packed run headers only and the GPU expands def levels and dict
   // indices (kernels.hip k_runs_expand_u32 / k_def_expand_validity).
   // AURON_PARQUET_GPU=0 forces the all-host decode for A/B comparison.
   const char* pg = getenv("AURON_PARQUET_GPU");
   const bool gpu_ok = ci.physical_type != 6 && !(pg && pg[0] == '0');
-  const bool gpu_def_mode = gpu_ok && ci.nullable;
+  // GPU page-decompression mode: pre-scan the page headers (cheap — no
+  // decompression); if every data page from the first v1 PLAIN page onward
+  // qualifies, the host decodes only the dictionary-encoded prefix and the
+  // device decompresses + decodes the PLAIN suffix (kernels_pq.hip)
+  const char* pc = getenv("AURON_PARQUET_GPUCOMP");
+  bool gpu_comp_mode = gpu_ok && cm.codec == 1 && !(pc && pc[0] == '0');
+  int64_t stop_pos = -1;  // file offset of the first GPU-handled page
+  if (gpu_comp_mode) {
+    gpu_comp_mode = false;
+    int64_t p0 = cm.dict_page_offset >= 0 ? std::min(cm.dict_page_offset,
+                                                     cm.data_page_offset)
+                                          : cm.data_page_offset;
+    int64_t pend = p0 + cm.total_compressed_size;
+    int64_t split = -1;
+    int64_t sfx_values = 0;
+    std::vector<PqColumnChunkData::GpuPageRef> refs;
+    bool ok = pend <= (int64_t)file_.size();
+    int64_t sp = p0;
+    while (ok && sp < pend) {
+      TReader hr{file_.data() + sp, file_.data() + pend};
+      PageHeader ph;
+      try {
+        ph = read_page_header(hr);
+      } catch (...) {
+        ok = false;
+        break;
+      }
+      const uint8_t* body = hr.p;
+      if (ph.compressed_size < 0 || ph.uncompressed_size < 0 ||
+          body - file_.data() + ph.compressed_size > pend) {
+        ok = false;
+        break;
+      }
+      bool suffix_ok = ph.type == 0 && ph.encoding == 0 &&
+                       (!ci.nullable || ph.def_encoding == 3) &&
+                       ph.num_values > 0 && ph.num_values <= (1 << 18) &&
+                       ph.uncompressed_size <= (1 << 20) + 65536;
+      if (split < 0) {
+        if (suffix_ok) split = sp;  // first qualifying PLAIN page
+      } else if (!suffix_ok) {
+        ok = false;  // non-qualifying page AFTER the split: give up
+        break;
+      }
+      if (split >= 0 && suffix_ok) {
+        refs.push_back({body, (uint32_t)ph.compressed_size,
+                        (uint32_t)ph.uncompressed_size,
+                        (uint32_t)ph.num_values});
+        sfx_values += ph.num_values;
+      }
+      sp = (body - file_.data()) + ph.compressed_size;
+    }
+    if (ok && split >= 0 && sfx_values > 0) {
+      gpu_comp_mode = true;
+      stop_pos = split;
+      out.gpu_comp = true;
+      out.comp_pages = std::move(refs);
+      out.suffix_values = sfx_values;
+      out.nullable = ci.nullable;
+    }
+  }
+  const bool gpu_def_mode = gpu_ok && ci.nullable && !gpu_comp_mode;
   if (ci.physical_type != 6)
     out.plain.reserve((size_t)cm.num_values * vw);
   std::vector<uint8_t> valid_bits;  // byte per value (bit-packed at the end)
@@ -981,8 +1095,47 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
     }
   };
 
+    // writers may FALL BACK from dictionary to PLAIN mid-chunk (e.g. pyarrow
+  // once the dictionary page hits its size limit); flatten the accumulated
+  // dict-encoded prefix to PLAIN on host when a mix appears (the dict
+  // portion is small by construction in that case)
+  // expand dict indices to PLAIN values appended to out.plain (typed
+  // copies: the per-value vector::insert this replaces dominated the
+  // whole chunk decode)
+  auto expand_dict = [&](const uint32_t* idx, size_t cnt) {
+    size_t base = out.plain.size();
+    out.plain.resize(base + cnt * (size_t)vw);
+    if (vw == 8) {
+      uint64_t* d = reinterpret_cast<uint64_t*>(out.plain.data() + base);
+      const uint64_t* dict =
+          reinterpret_cast<const uint64_t*>(out.dict_values.data());
+      for (size_t i = 0; i < cnt; i++) {
+        if ((int64_t)idx[i] >= out.dict_count)
+          fail("parquet: dict index range");
+        d[i] = dict[idx[i]];
+      }
+    } else if (vw == 4) {
+      uint32_t* d = reinterpret_cast<uint32_t*>(out.plain.data() + base);
+      const uint32_t* dict =
+          reinterpret_cast<const uint32_t*>(out.dict_values.data());
+      for (size_t i = 0; i < cnt; i++) {
+        if ((int64_t)idx[i] >= out.dict_count)
+          fail("parquet: dict index range");
+        d[i] = dict[idx[i]];
+      }
+    } else {
+      for (size_t i = 0; i < cnt; i++) {
+        if ((int64_t)idx[i] >= out.dict_count)
+          fail("parquet: dict index range");
+        memcpy(out.plain.data() + base + i * (size_t)vw,
+               out.dict_values.data() + (size_t)idx[i] * vw, (size_t)vw);
+      }
+    }
+  };
+
   std::vector<uint8_t> page_buf;  // reused across pages (grow-only)
-  while (pos < chunk_end && out.num_values < cm.num_values) {
+  const int64_t loop_end = gpu_comp_mode ? stop_pos : chunk_end;
+  while (pos < loop_end && out.num_values < cm.num_values) {
     TReader hr{file_.data() + pos, file_.data() + chunk_end};
     PageHeader ph = read_page_header(hr);
     const uint8_t* page = hr.p;
@@ -1104,44 +1257,7 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
       valid_bits.resize(valid_bits.size() + (size_t)nvals, 1);
     }
 
-    // writers may FALL BACK from dictionary to PLAIN mid-chunk (e.g. pyarrow
-    // once the dictionary page hits its size limit); flatten the accumulated
-    // dict-encoded prefix to PLAIN on host when a mix appears (the dict
-    // portion is small by construction in that case)
-    // expand dict indices to PLAIN values appended to out.plain (typed
-    // copies: the per-value vector::insert this replaces dominated the
-    // whole chunk decode)
-    auto expand_dict = [&](const uint32_t* idx, size_t cnt) {
-      size_t base = out.plain.size();
-      out.plain.resize(base + cnt * (size_t)vw);
-      if (vw == 8) {
-        uint64_t* d = reinterpret_cast<uint64_t*>(out.plain.data() + base);
-        const uint64_t* dict =
-            reinterpret_cast<const uint64_t*>(out.dict_values.data());
-        for (size_t i = 0; i < cnt; i++) {
-          if ((int64_t)idx[i] >= out.dict_count)
-            fail("parquet: dict index range");
-          d[i] = dict[idx[i]];
-        }
-      } else if (vw == 4) {
-        uint32_t* d = reinterpret_cast<uint32_t*>(out.plain.data() + base);
-        const uint32_t* dict =
-            reinterpret_cast<const uint32_t*>(out.dict_values.data());
-        for (size_t i = 0; i < cnt; i++) {
-          if ((int64_t)idx[i] >= out.dict_count)
-            fail("parquet: dict index range");
-          d[i] = dict[idx[i]];
-        }
-      } else {
-        for (size_t i = 0; i < cnt; i++) {
-          if ((int64_t)idx[i] >= out.dict_count)
-            fail("parquet: dict index range");
-          memcpy(out.plain.data() + base + i * (size_t)vw,
-                 out.dict_values.data() + (size_t)idx[i] * vw, (size_t)vw);
-        }
-      }
-    };
-    auto flatten_dict = [&]() {
+        auto flatten_dict = [&]() {
       if (out.gpu_dict) {
         // staged GPU runs must materialize after all (dict/PLAIN mix)
         std::vector<uint32_t> idx;
@@ -1282,7 +1398,29 @@ PqColumnChunkData ParquetFile::read_chunk(int rg, int col) const {
     out.num_values += nvals;
   }
 
-  if (out.num_values != cm.num_values)
+  if (out.gpu_comp) {
+    out.prefix_values = out.num_values;
+    // the engine needs the prefix as DENSE PLAIN values + packed validity;
+    // flatten any dict-encoded prefix on host (it is small by construction)
+    if (out.gpu_dict) {
+      std::vector<uint32_t> idx;
+      expand_runs_host(out.idx_runs, out.idx_bytes, &idx);
+      out.gpu_dict = false;
+      out.idx_runs.clear();
+      out.idx_bytes.clear();
+      out.nn_count = 0;
+      expand_dict(idx.data(), idx.size());
+    }
+    if (out.uses_dict) {
+      std::vector<uint32_t> idx = std::move(out.dict_indices);
+      out.dict_indices.clear();
+      out.uses_dict = false;
+      expand_dict(idx.data(), idx.size());
+    }
+    if (!out.def_runs.empty())
+      fail("parquet: gpu_comp prefix must not stage def runs");
+  }
+  if (out.num_values + out.suffix_values != cm.num_values)
     fail("parquet: value count mismatch in chunk");
   if (is_bytes) {
     // assemble row-aligned offsets (null rows zero-length)

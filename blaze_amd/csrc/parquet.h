@@ -65,6 +65,23 @@ struct PqColumnChunkData {
   bool gpu_def = false;    // validity replaced by def_runs/def_bytes
   std::vector<PqRun> def_runs;
   std::vector<uint8_t> def_bytes;
+  // GPU page decompression (kernels_pq.hip): SNAPPY fixed-width chunks
+  // whose data pages after a dictionary-encoded prefix are all v1 PLAIN.
+  // The prefix is host-decoded into plain/validity as usual; the suffix
+  // pages are recorded RAW (pointers into the file mapping — valid only
+  // while the ParquetFile is alive) and decompressed+decoded on device.
+  bool gpu_comp = false;
+  struct GpuPageRef {
+    const uint8_t* src;      // compressed page bytes (mmap)
+    uint32_t comp_len;
+    uint32_t uncomp_len;
+    uint32_t num_values;     // value slots (incl nulls)
+  };
+  std::vector<GpuPageRef> comp_pages;
+  int64_t prefix_values = 0;  // host-decoded value slots (incl nulls)
+  int64_t suffix_values = 0;  // sum over comp_pages
+  bool nullable = false;      // pages carry a def-level section
+  int value_width = 0;        // fixed-width byte size (0 for BYTE_ARRAY)
 };
 
 // host materialization of GPU-staged runs (debug / CPU-test path): expands

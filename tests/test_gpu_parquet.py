@@ -197,3 +197,25 @@ def test_parquet_row_group_pruning(tmp_path):
     # groups 0 (0..25k) and 1 (25k..50k) survive; 2,3 pruned
     assert len(got) == 50_000
     assert got.max() == 49_999
+
+
+def test_parquet_gpu_page_decompress_ab(tmp_path):
+    """The device page-decompression path (kernels_pq.hip, default-on for
+    snappy PLAIN suffixes) must produce byte-identical aggregation results
+    to the forced host decode (AURON_PARQUET_GPUCOMP=0) — dict prefix +
+    PLAIN fallback, 3 row groups, 10% nulls to stress the on-device
+    def-level walk across unaligned page bit boundaries."""
+    import os
+    rng = np.random.default_rng(33)
+    n = 2_000_000
+    keys = rng.integers(0, 200_000, n).astype(np.int64)
+    vals = rng.integers(0, 1000, n).astype(np.float64)
+    vv = rng.random(n) >= 0.1
+    ps = _write(tmp_path, "ab.parquet", keys, vals, vv, "snappy", True,
+                700_000)
+    _run_and_check(ps, keys, vals, vv)          # device path (default)
+    os.environ["AURON_PARQUET_GPUCOMP"] = "0"   # forced host decode
+    try:
+        _run_and_check(ps, keys, vals, vv)
+    finally:
+        os.environ.pop("AURON_PARQUET_GPUCOMP", None)
