@@ -104,9 +104,9 @@ def main():
             "compressed_bytes_per_s": round(size / dt, 0),
             "groups_out": groups,
         },
-        "notes": "scan decode: footer/pages+decompress on host (parallel "
-                 "across cores), validity/dictionary expansion + filter + agg "
-                 "on GPU; host decode still bounds this config — DESIGN.md",
+        "notes": "scan decode: host walks footers/page headers + decodes the "
+                 "dict-encoded prefix; snappy PLAIN pages decompress+decode ON "
+                 "DEVICE (kernels_pq.hip wave-per-page), filter + agg on GPU",
     }
     print(json.dumps(out))
 
