@@ -34,49 +34,19 @@ inline void check_launch_pq(const char* name) {
 }  // namespace
 
 // ---- snappy block decode, one wave per page --------------------------------
-// The token walk is a SEQUENTIAL dependent chain (each token's position
-// depends on the previous token's length); reading the stream straight from
-// global memory put a cache-miss round trip on every link (~2000 cyc/token
-// measured: SQ_WAIT 5x ACTIVE, decode 430 ms for 1B rows). The wave
-// therefore stages the compressed stream through a per-wave LDS WINDOW:
-// coalesced 64-lane refills stream from HBM at full bandwidth, and the
-// dependent token reads become ~50-cycle ds_reads.
-static constexpr uint32_t PQ_WIN = 4096;      // window payload bytes
-static constexpr uint32_t PQ_WIN_AL = 64;     // refill alignment slack
+// Tag walk is wave-uniform (every lane loads the same bytes — scalarized by
+// the compiler); copies are lane-parallel. Returns false on malformed input.
 __device__ static bool dev_snappy_page(const uint8_t* __restrict__ src,
                                        uint32_t n, uint8_t* __restrict__ dst,
-                                       uint32_t out_len, uint8_t* lds_win) {
+                                       uint32_t out_len) {
   const int lane = (int)(threadIdx.x & 63);
-  uint32_t wbase = 0, wlen = 0;  // window covers src[wbase, wbase+wlen)
-  auto refill = [&](uint32_t at) {
-    wbase = at & ~(PQ_WIN_AL - 1);
-    uint32_t avail = n - wbase;
-    wlen = avail < PQ_WIN + PQ_WIN_AL ? avail : PQ_WIN + PQ_WIN_AL;
-    // 64 lanes x u32 strided: wbase is 64-aligned so loads are aligned
-    for (uint32_t k = (uint32_t)lane * 4; k < wlen; k += 64 * 4) {
-      uint32_t v = 0;
-      uint32_t take = wlen - k < 4 ? wlen - k : 4;
-      __builtin_memcpy(&v, src + wbase + k, take);
-      *(uint32_t*)(lds_win + k) = v;
-    }
-    // drain the DS writes of all 64 lanes before any lane reads the
-    // window; the memory clobber stops the compiler reordering across it
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    __builtin_amdgcn_wave_barrier();
-  };
-  // ensure [at, at+need) is inside the window; need <= PQ_WIN
-  auto ensure = [&](uint32_t at, uint32_t need) {
-    if (at < wbase || at + need > wbase + wlen) refill(at);
-  };
-  refill(0);
   uint32_t pos = 0;
   {
     uint64_t hdr = 0;
     int shift = 0;
     bool got = false;
     while (pos < n) {
-      ensure(pos, 1);
-      uint8_t b = lds_win[pos++ - wbase];
+      uint8_t b = src[pos++];
       hdr |= (uint64_t)(b & 0x7f) << shift;
       if (!(b & 0x80)) {
         got = true;
@@ -88,11 +58,7 @@ __device__ static bool dev_snappy_page(const uint8_t* __restrict__ src,
   }
   uint32_t op = 0;
   while (pos < n && op < out_len) {
-    ensure(pos, 5);  // tag + up to 4 trailer bytes
-    const uint8_t* w = lds_win + (pos - wbase);
-    uint8_t tag = w[0];
-    pos++;
-    w++;
+    uint8_t tag = src[pos++];
     int type = tag & 3;
     if (type == 0) {  // literal
       uint32_t len = (tag >> 2) + 1;
@@ -100,29 +66,19 @@ __device__ static bool dev_snappy_page(const uint8_t* __restrict__ src,
         int nb = (int)len - 60;
         if (pos + nb > n) return false;
         len = 0;
-        for (int i = 0; i < nb; i++) len |= (uint32_t)w[i] << (8 * i);
+        for (int i = 0; i < nb; i++) len |= (uint32_t)src[pos++] << (8 * i);
         len += 1;
-        pos += nb;
       }
       if (pos + len > n || op + len > out_len) return false;
-      // copy through the window in segments (long literals span refills)
-      uint32_t done = 0;
-      while (done < len) {
-        ensure(pos + done, 1);
-        uint32_t seg = wbase + wlen - (pos + done);
-        if (seg > len - done) seg = len - done;
-        const uint8_t* ws = lds_win + (pos + done - wbase);
-        if (seg >= 16) {
-          uint32_t n8 = seg >> 3;
-          for (uint32_t d = lane; d < n8; d += 64)
-            __builtin_memcpy(dst + op + done + (size_t)d * 8,
-                             ws + (size_t)d * 8, 8);
-          for (uint32_t d = (n8 << 3) + lane; d < seg; d += 64)
-            dst[op + done + d] = ws[d];
-        } else if (lane < (int)seg) {
-          dst[op + done + lane] = ws[lane];
-        }
-        done += seg;
+      if (len >= 16) {  // u64 chunks (unaligned-capable), byte tail
+        uint32_t n8 = len >> 3;
+        for (uint32_t d = lane; d < n8; d += 64)
+          __builtin_memcpy(dst + op + (size_t)d * 8, src + pos + (size_t)d * 8,
+                           8);
+        for (uint32_t d = (n8 << 3) + lane; d < len; d += 64)
+          dst[op + d] = src[pos + d];
+      } else if (lane < (int)len) {  // short: one masked round, no loop
+        dst[op + lane] = src[pos + lane];
       }
       pos += len;
       op += len;
@@ -131,18 +87,18 @@ __device__ static bool dev_snappy_page(const uint8_t* __restrict__ src,
       if (type == 1) {
         if (pos + 1 > n) return false;
         len = ((tag >> 2) & 7) + 4;
-        off = ((uint32_t)(tag >> 5) << 8) | w[0];
+        off = ((uint32_t)(tag >> 5) << 8) | src[pos];
         pos += 1;
       } else if (type == 2) {
         if (pos + 2 > n) return false;
         len = (tag >> 2) + 1;
-        off = (uint32_t)w[0] | ((uint32_t)w[1] << 8);
+        off = (uint32_t)src[pos] | ((uint32_t)src[pos + 1] << 8);
         pos += 2;
       } else {
         if (pos + 4 > n) return false;
         len = (tag >> 2) + 1;
         off = 0;
-        for (int i = 0; i < 4; i++) off |= (uint32_t)w[i] << (8 * i);
+        for (int i = 0; i < 4; i++) off |= (uint32_t)src[pos + i] << (8 * i);
         pos += 4;
       }
       if (off == 0 || off > op || op + len > out_len) return false;
@@ -278,15 +234,12 @@ __global__ void __launch_bounds__(256) k_pq_pages_decode(
     uint32_t* __restrict__ val_offs, uint32_t* __restrict__ err) {
   // 4 independent waves per block: one PAGE per wave (64-thread blocks
   // capped resident waves per CU at the workgroup-slot limit)
-  extern __shared__ uint8_t pq_lds[];
   int pg = (int)blockIdx.x * 4 + (int)(threadIdx.x >> 6);
   if (pg >= npages) return;
-  uint8_t* lds_win =
-      pq_lds + (size_t)(threadIdx.x >> 6) * (PQ_WIN + PQ_WIN_AL);
   const PqGpuPage p = pages[pg];
   const uint8_t* src = comp + p.comp_off;
   uint8_t* dst = scratch + p.uncomp_off;
-  if (!dev_snappy_page(src, p.comp_len, dst, p.uncomp_len, lds_win)) {
+  if (!dev_snappy_page(src, p.comp_len, dst, p.uncomp_len)) {
     if ((threadIdx.x & 63) == 0) atomicOr(err, 1u);
     return;
   }
@@ -365,8 +318,8 @@ void launch_pq_pages_decode(const uint8_t* comp, const PqGpuPage* pages,
                             const PqGpuChunk* chunks, uint32_t* nn_counts,
                             uint32_t* val_offs, uint32_t* err, hipStream_t s) {
   hipLaunchKernelGGL(k_pq_pages_decode, dim3((npages + 3) / 4), dim3(256),
-                     4 * (PQ_WIN + PQ_WIN_AL), s, comp, pages, npages,
-                     scratch, valid_blob, chunks, nn_counts, val_offs, err);
+                     0, s, comp, pages, npages, scratch, valid_blob, chunks,
+                     nn_counts, val_offs, err);
   check_launch_pq("k_pq_pages_decode");
 }
 
