@@ -70,16 +70,7 @@ __device__ static bool dev_snappy_page(const uint8_t* __restrict__ src,
         len += 1;
       }
       if (pos + len > n || op + len > out_len) return false;
-      if (len >= 16) {  // u64 chunks (unaligned-capable), byte tail
-        uint32_t n8 = len >> 3;
-        for (uint32_t d = lane; d < n8; d += 64)
-          __builtin_memcpy(dst + op + (size_t)d * 8, src + pos + (size_t)d * 8,
-                           8);
-        for (uint32_t d = (n8 << 3) + lane; d < len; d += 64)
-          dst[op + d] = src[pos + d];
-      } else if (lane < (int)len) {  // short: one masked round, no loop
-        dst[op + lane] = src[pos + lane];
-      }
+      for (uint32_t d = lane; d < len; d += 64) dst[op + d] = src[pos + d];
       pos += len;
       op += len;
     } else {
@@ -108,16 +99,8 @@ __device__ static bool dev_snappy_page(const uint8_t* __restrict__ src,
       uint32_t done = 0, avail = off;
       while (done < len) {
         uint32_t take = len - done < avail ? len - done : avail;
-        if (take >= 16) {
-          uint32_t n8 = take >> 3;
-          for (uint32_t d = lane; d < n8; d += 64)
-            __builtin_memcpy(dst + op + done + (size_t)d * 8,
-                             dst + start + (size_t)d * 8, 8);
-          for (uint32_t d = (n8 << 3) + lane; d < take; d += 64)
-            dst[op + done + d] = dst[start + d];
-        } else if (lane < (int)take) {
-          dst[op + done + lane] = dst[start + lane];
-        }
+        for (uint32_t d = lane; d < take; d += 64)
+          dst[op + done + d] = dst[start + d];
         // dst[start + take .. ) stays the same pattern: extending the window
         // keeps sources valid because take bytes were just appended
         done += take;
@@ -226,33 +209,31 @@ __device__ static uint32_t dev_def_walk(const uint8_t* __restrict__ p,
   return ones;
 }
 
-__global__ void __launch_bounds__(256) k_pq_pages_decode(
+__global__ void __launch_bounds__(64) k_pq_pages_decode(
     const uint8_t* __restrict__ comp, const PqGpuPage* __restrict__ pages,
     int npages, uint8_t* __restrict__ scratch,
     uint8_t* __restrict__ valid_blob,
     const PqGpuChunk* __restrict__ chunks, uint32_t* __restrict__ nn_counts,
     uint32_t* __restrict__ val_offs, uint32_t* __restrict__ err) {
-  // 4 independent waves per block: one PAGE per wave (64-thread blocks
-  // capped resident waves per CU at the workgroup-slot limit)
-  int pg = (int)blockIdx.x * 4 + (int)(threadIdx.x >> 6);
+  int pg = (int)blockIdx.x;
   if (pg >= npages) return;
   const PqGpuPage p = pages[pg];
   const uint8_t* src = comp + p.comp_off;
   uint8_t* dst = scratch + p.uncomp_off;
   if (!dev_snappy_page(src, p.comp_len, dst, p.uncomp_len)) {
-    if ((threadIdx.x & 63) == 0) atomicOr(err, 1u);
+    if (threadIdx.x == 0) atomicOr(err, 1u);
     return;
   }
   uint32_t voff = 0;
   uint32_t nn = p.num_values;
   if (p.has_def) {
     if (p.uncomp_len < 4) {
-      if ((threadIdx.x & 63) == 0) atomicOr(err, 2u);
+      if (threadIdx.x == 0) atomicOr(err, 2u);
       return;
     }
     uint32_t ll = *(const uint32_t*)dst;
     if (ll > p.uncomp_len - 4) {
-      if ((threadIdx.x & 63) == 0) atomicOr(err, 2u);
+      if (threadIdx.x == 0) atomicOr(err, 2u);
       return;
     }
     bool bad = false;
@@ -261,12 +242,12 @@ __global__ void __launch_bounds__(256) k_pq_pages_decode(
                       valid_blob + c.valid_base,
                       (uint64_t)p.value_base, &bad);
     if (bad) {
-      if ((threadIdx.x & 63) == 0) atomicOr(err, 2u);
+      if (threadIdx.x == 0) atomicOr(err, 2u);
       return;
     }
     voff = 4 + ll;
   }
-  if ((threadIdx.x & 63) == 0) {
+  if (threadIdx.x == 0) {
     nn_counts[pg] = nn;
     val_offs[pg] = voff;
   }
@@ -317,9 +298,9 @@ void launch_pq_pages_decode(const uint8_t* comp, const PqGpuPage* pages,
                             int npages, uint8_t* scratch, uint8_t* valid_blob,
                             const PqGpuChunk* chunks, uint32_t* nn_counts,
                             uint32_t* val_offs, uint32_t* err, hipStream_t s) {
-  hipLaunchKernelGGL(k_pq_pages_decode, dim3((npages + 3) / 4), dim3(256),
-                     0, s, comp, pages, npages, scratch, valid_blob, chunks,
-                     nn_counts, val_offs, err);
+  hipLaunchKernelGGL(k_pq_pages_decode, dim3(npages), dim3(64), 0, s, comp,
+                     pages, npages, scratch, valid_blob, chunks, nn_counts,
+                     val_offs, err);
   check_launch_pq("k_pq_pages_decode");
 }
 
