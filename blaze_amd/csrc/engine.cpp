@@ -3561,9 +3561,14 @@ struct Runtime {
       std::vector<PqColumnChunkData> decoded((size_t)nrg * width);
       std::string decode_err;
       std::mutex err_mu;
-      const int RG_WIN =
-          std::max(1, (int)((256u << 20) /
-                            std::max<int64_t>(1, pf.row_group_rows(0) * 8)));
+      const int RG_WIN = [&] {
+        const char* e = getenv("AURON_PARQUET_WIN");
+        if (e && e[0] == '0') return nrg > 0 ? nrg : 1;  // one window
+        if (e && atoi(e) > 0) return atoi(e);
+        return std::max(1, (int)((256u << 20) /
+                                 std::max<int64_t>(
+                                     1, pf.row_group_rows(0) * 8)));
+      }();
       auto decode_window = [&](int rg_lo, int rg_hi) {
         size_t jlo = (size_t)rg_lo * width, jhi = (size_t)rg_hi * width;
         unsigned nw = std::min<unsigned>(
