@@ -1707,6 +1707,8 @@ class AggOp {
     // experimental until measured; AURON_AGG2_V4=1 selects it
     const char* v4 = getenv("AURON_AGG2_V4");
     agg2_v4_ = (v4 && v4[0] == '1');
+    const char* p16 = getenv("AURON_AGG2_P16");
+    agg2_p16_en_ = !(p16 && p16[0] == '0');
   }
 
   // Cross-chunk pipelined two-phase aggregation (v3 kernels): the scatter
@@ -1768,7 +1770,7 @@ class AggOp {
                          (uint32_t*)(d_pctr_[pb].get<uint8_t>() + 16), 1024,
                          s_aux_);
         launch_agg3_line_sizes(d_pcounts_[pb].get<uint32_t>(), mat3 + 1,
-                               d_plinesz_[pb].get<uint32_t>(), s_aux_);
+                               d_plinesz_[pb].get<uint32_t>(), 24, s_aux_);
         size_t tb = d_pscan_tmp_.size();
         scan_counts_matrix(d_plinesz_[pb].get<uint32_t>(),
                            d_pscanned_[pb].get<uint32_t>(), mat3 + 1,
@@ -1779,7 +1781,7 @@ class AggOp {
                             d_pleft_[pb].get<PartRow>(),
                             d_pctr_[pb].get<unsigned long long>() + 1,
                             d_plinesz_[pb].get<uint32_t>(),
-                            t_.error_flag, s_aux_);
+                            t_.error_flag, 24, 0, s_aux_);
         AURON_HIP(hipEventRecord(ev_scatter_[pb], s_aux_));
       }
       if (k > 0) {
@@ -1832,7 +1834,7 @@ class AggOp {
                            d_pctr_[pb].get<unsigned long long>(),
                            (int64_t)nbuck * 4096, d_pleft_[pb].get<PartRow>(),
                            d_pctr_[pb].get<unsigned long long>() + 1,
-                           t_.error_flag, stream_);
+                           t_.error_flag, 24, 0, stream_);
         AURON_HIP(hipMemcpyAsync(pin_pctr_[pb].get(), d_pctr_[pb].get(), 24,
                                  hipMemcpyDeviceToHost, stream_));
         AURON_HIP(hipMemcpyAsync(pin_pctr_[pb].get<uint8_t>() + 32,
@@ -1929,14 +1931,29 @@ class AggOp {
       // v3: hist -> line-padded scan -> LDS-staged packet scatter -> 4096-
       // slot bucket aggregation (kernels_agg3.hip)
       int64_t mat3 = (int64_t)AGG3_NBUCK << AGG3_GRID_LOG2;
+      // adaptive packed-16B records: chunk 1 runs the 24B layout while the
+      // hist probes the normal-key min/max; if the observed range fits u32,
+      // later chunks store (key - base) u32 — 33% less partition traffic.
+      // Out-of-range keys in later chunks take the counted leftover bypass,
+      // so the choice is a pure optimization, never a correctness bet.
+      const int rec3 = (packed16_ && !agg2_v4_) ? 16 : 24;
+      if (agg2_p16_en_ && !p16_decided_ && !d_kminmax_) {
+        d_kminmax_.alloc(16);
+        AURON_HIP(hipMemsetAsync(d_kminmax_.get(), 0xFF, 8, stream_));
+        AURON_HIP(hipMemsetAsync(d_kminmax_.get<uint8_t>() + 8, 0, 8,
+                                 stream_));
+      }
       // the v4 workers cover the same 1024-wide virtual-lane tiles as the
       // 1024-thread hist, so the hist launch is identical for v3 and v4
       launch_agg2_hist(keys, kv, chunk, AGG3_NBUCK_LOG2, AGG3_GRID_LOG2,
                        d_counts_.get<uint32_t>(),
                        (uint32_t*)(d_counters_.get<uint8_t>() + 16), 1024,
-                       stream_);
+                       stream_,
+                       (agg2_p16_en_ && !p16_decided_)
+                           ? d_kminmax_.get<unsigned long long>()
+                           : nullptr);
       launch_agg3_line_sizes(d_counts_.get<uint32_t>(), mat3 + 1,
-                             d_linesz_.get<uint32_t>(), stream_);
+                             d_linesz_.get<uint32_t>(), rec3, stream_);
       size_t tb3 = d_scan_tmp_.size();
       scan_counts_matrix(d_linesz_.get<uint32_t>(),
                          d_scanned_.get<uint32_t>(), mat3 + 1,
@@ -1955,7 +1972,7 @@ class AggOp {
                             d_leftover_.get<PartRow>(),
                             d_counters_.get<unsigned long long>() + 1,
                             d_linesz_.get<uint32_t>(),  // free after the scan
-                            t_.error_flag, stream_);
+                            t_.error_flag, rec3, key_base16_, stream_);
       launch_agg3_bucket(d_partkv_.get<uint8_t>(), d_counts_.get<uint32_t>(),
                          d_linesz_.get<uint32_t>(),
                          d_scanned_.get<uint32_t>(), AGG3_GRID_LOG2,
@@ -1965,7 +1982,7 @@ class AggOp {
                          (int64_t)AGG3_NBUCK * 4096,
                          d_leftover_.get<PartRow>(),
                          d_counters_.get<unsigned long long>() + 1,
-                         t_.error_flag, stream_);
+                         t_.error_flag, rec3, key_base16_, stream_);
     } else {
     int64_t mat = (int64_t)nbuck << agg2_grid_log2_;
     launch_agg2_hist(keys, kv, chunk, AGG2_NBUCK_LOG2, agg2_grid_log2_,
@@ -2016,7 +2033,22 @@ class AggOp {
     uint64_t* h_ng = (uint64_t*)(h_ctr + 3);
     AURON_HIP(hipMemcpyAsync(h_ng, t_.num_groups, 8, hipMemcpyDeviceToHost,
                              stream_));
+    uint64_t* h_kmm = (uint64_t*)(h_ctr + 4);
+    bool want_p16 = agg2_v3_ && !agg2_v4_ && agg2_p16_en_ &&
+                    !p16_decided_ && d_kminmax_;
+    if (want_p16)
+      AURON_HIP(hipMemcpyAsync(h_kmm, d_kminmax_.get(), 16,
+                               hipMemcpyDeviceToHost, stream_));
     AURON_HIP(hipStreamSynchronize(stream_));
+    if (want_p16) {
+      p16_decided_ = true;  // one probe chunk is enough
+      uint64_t mn = h_kmm[0], mx = h_kmm[1];
+      if (mn <= mx && mx - mn <= 0xFFFFFFFFull) {
+        packed16_ = true;
+        key_base16_ = (int64_t)(mn ^ 0x8000000000000000ull);
+        DBG("agg.p16 enabled: key range %llu", (unsigned long long)(mx - mn));
+      }
+    }
     int64_t staged_n = (int64_t)h_ctr[0];
     int64_t lo_n = (int64_t)h_ctr[1];
     uint32_t special_rows = (uint32_t)h_ctr[2];
@@ -2769,6 +2801,11 @@ class AggOp {
   bool agg2_split_ = false;
   bool agg2_v3_ = true;
   bool agg2_v4_ = false;
+  bool agg2_p16_en_ = true;   // AURON_AGG2_P16=0 disables the adaptive path
+  bool p16_decided_ = false;  // one hist range probe (chunk 1)
+  bool packed16_ = false;     // later chunks use 16B partition records
+  int64_t key_base16_ = 0;
+  DevBuf d_kminmax_;
   bool agg2_pipe_ = false;
   bool agg2_conf_read_ = false;
   int agg2_grid_log2_ = 9;

@@ -304,7 +304,8 @@ constexpr int AGG2_GRID_LOG2_MAX = 10;  // matrix sizing bound for the
 // [nbuck << AGG2_GRID_LOG2]
 void launch_agg2_hist(const int64_t* keys, const uint8_t* key_valid, int64_t n,
                       int nbuck_log2, int grid_log2, uint32_t* counts_matrix,
-                      uint32_t* special_rows, int block, hipStream_t s);
+                      uint32_t* special_rows, int block, hipStream_t s,
+                      unsigned long long* kminmax = nullptr);
 // exclusive scan over the flat counts matrix -> per-(block,bucket) bases
 void scan_counts_matrix(const uint32_t* counts, uint32_t* scanned, int64_t n,
                         void* temp, size_t* temp_bytes, hipStream_t s);
@@ -534,14 +535,16 @@ void launch_def_expand_validity(const void* runs, int nruns,
 // v3 two-phase partition pipeline (kernels_agg3.hip): LDS-staged packet
 // scatter into 64B-aligned per-(block,bucket) ranges + 4096-slot bucket agg
 void launch_agg3_line_sizes(const uint32_t* counts, int64_t n,
-                            uint32_t* sizes, hipStream_t s);
+                            uint32_t* sizes, int rec, hipStream_t s);
+// rec = 24 (i64 key) or 16 (u32 key offset from key_base — rows whose key
+// falls outside [key_base, key_base+2^32) bypass to the leftover list)
 void launch_agg3_scatter(const int64_t* keys, const uint8_t* key_valid,
                          const double* vals, const uint8_t* val_valid,
                          int64_t n, int nbuck_log2, int grid_log2,
                          const uint32_t* line_scan, uint8_t* out,
                          PartRow* leftover, unsigned long long* lo_n,
                          uint32_t* bypass_matrix, uint32_t* err_flag,
-                         hipStream_t s);
+                         int rec, int64_t key_base, hipStream_t s);
 // v4 scatter: barrier-free per-bucket LDS rings drained by dedicated
 // flusher waves (same in/out contract as launch_agg3_scatter). The hist
 // feeding it must run with agg4_worker_waves()*64 threads per block so the
@@ -560,7 +563,8 @@ void launch_agg3_bucket(const uint8_t* part, const uint32_t* counts,
                         int nbuckets, StagedGroup* staged,
                         unsigned long long* staged_n, int64_t staged_cap,
                         PartRow* leftover, unsigned long long* lo_n,
-                        uint32_t* error_flag, hipStream_t s);
+                        uint32_t* error_flag, int rec, int64_t key_base,
+                        hipStream_t s);
 
 // GPU Parquet page staging (kernels_pq.hip): wave-per-page snappy decompress
 // + on-device def-level parse + dense compaction. Built by the host pre-scan

@@ -80,12 +80,17 @@ __global__ __launch_bounds__(1024) void k_agg2_hist(
                             const uint8_t* __restrict__ key_valid, int64_t n,
                             int nbuck_log2, int grid_log2,
                             uint32_t* __restrict__ counts_matrix,
-                            uint32_t* __restrict__ special_rows) {
+                            uint32_t* __restrict__ special_rows,
+                            unsigned long long* __restrict__ kminmax) {
   extern __shared__ uint32_t lds_hist[];
   const uint32_t nbuck = 1u << nbuck_log2;
   for (uint32_t b = threadIdx.x; b < nbuck; b += blockDim.x) lds_hist[b] = 0;
   __syncthreads();
   uint32_t special = 0;
+  // optional NORMAL-key range probe (order-mapped u64 so unsigned
+  // atomicMin/Max implement signed i64 min/max): feeds the packed-16B
+  // partition record fast path (kernels_agg3.hip)
+  unsigned long long kmin = ~0ull, kmax = 0ull;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     bool knull = key_valid && !bit_get2(key_valid, i);
@@ -94,12 +99,21 @@ __global__ __launch_bounds__(1024) void k_agg2_hist(
       special++;
       continue;
     }
+    if (kminmax) {
+      unsigned long long m = (unsigned long long)k ^ 0x8000000000000000ull;
+      kmin = m < kmin ? m : kmin;
+      kmax = m > kmax ? m : kmax;
+    }
     atomicAdd(&lds_hist[bucket_of(k, nbuck_log2)], 1u);
   }
   __syncthreads();
   for (uint32_t b = threadIdx.x; b < nbuck; b += blockDim.x)
     counts_matrix[((size_t)b << grid_log2) | blockIdx.x] = lds_hist[b];
   if (special) atomicAdd(special_rows, special);
+  if (kminmax && kmin != ~0ull) {
+    atomicMin(&kminmax[0], kmin);
+    atomicMax(&kminmax[1], kmax);
+  }
 }
 
 // offsets[b] = scanned[(b << GRID_LOG2) | 0]; offsets[nbuck] = the scan
@@ -360,11 +374,12 @@ void launch_agg2_leftovers(const AggTable& t, const PartRow* rows, int64_t n,
 
 void launch_agg2_hist(const int64_t* keys, const uint8_t* key_valid, int64_t n,
                       int nbuck_log2, int grid_log2, uint32_t* counts_matrix,
-                      uint32_t* special_rows, int block, hipStream_t s) {
+                      uint32_t* special_rows, int block, hipStream_t s,
+                      unsigned long long* kminmax) {
   size_t lds = (size_t)(1u << nbuck_log2) * 4;
   hipLaunchKernelGGL(k_agg2_hist, dim3(1 << grid_log2), dim3(block), lds, s,
                      keys, key_valid, n, nbuck_log2, grid_log2, counts_matrix,
-                     special_rows);
+                     special_rows, kminmax);
   check_launch2("k_agg2_hist");
 }
 
