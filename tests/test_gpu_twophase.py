@@ -114,3 +114,44 @@ def test_two_phase_skewed_buckets():
         np.testing.assert_array_equal(got_c, ref["counts"])
     finally:
         os.environ.pop("AURON_AGG2_V3", None)
+
+
+def test_two_phase_packed16_adaptive():
+    """Adaptive 16B partition records: chunk 1 probes the key range (hist
+    min/max) and later chunks pack keys as u32 offsets. Three 4M-row chunks
+    (AURON_AGG2_CHUNK_M=4): chunk 1 narrow (decides packed), chunk 2 mixes
+    in keys FAR outside the observed range (must take the counted leftover
+    bypass), chunk 3 narrow again — parity against the oracle."""
+    os.environ["AURON_AGG2_CHUNK_M"] = "4"
+    try:
+        rng = np.random.default_rng(63)
+        n = 4_000_000
+        base = 7_000_000_000  # away from zero: exercises non-zero key_base
+        k1 = base + rng.integers(0, 200_000, n)
+        k2 = base + rng.integers(0, 200_000, n)
+        k2[::5000] = rng.integers(0, 50, (k2[::5000]).shape[0]) * 2**40
+        k3 = base + rng.integers(0, 200_000, n)
+        batches = []
+        vals_all, keys_all = [], []
+        for k in (k1, k2, k3):
+            v = rng.integers(0, 1000, n).astype(np.float64)
+            batches.append([(k.astype(np.int64), None), (v, None)])
+            keys_all.append(k.astype(np.int64))
+            vals_all.append(v)
+        t = blaze_amd.Task(plan.plan_partial_final(), batches=batches,
+                           conf={"BATCH_SIZE": 1 << 22,
+                                 "AURON_HIP_AGG_TABLE_SLOTS": 1 << 20})
+        outs = t.run()
+        t.finalize()
+        got_k = np.concatenate([o[0]["values"] for o in outs])
+        got_s = np.concatenate([o[1]["values"] for o in outs])
+        got_c = np.concatenate([o[2]["values"] for o in outs])
+        orc = oracle.Agg()
+        for k, v in zip(keys_all, vals_all):
+            orc.update(k, v)
+        ref = orc.output()
+        np.testing.assert_array_equal(got_k, ref["keys"])
+        np.testing.assert_array_equal(got_c, ref["counts"])
+        np.testing.assert_array_equal(got_s, ref["sums"])
+    finally:
+        os.environ.pop("AURON_AGG2_CHUNK_M", None)
