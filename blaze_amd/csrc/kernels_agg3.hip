@@ -78,7 +78,11 @@ void launch_agg3_line_sizes(const uint32_t* counts, int64_t n,
 }
 
 // ---- v3 scatter ------------------------------------------------------------
-static constexpr int A3_CAP = 12;      // staged records per bucket
+static constexpr int A3_CAP = 12;      // staged records per bucket (24B)
+static constexpr int A3_CAP16 = 18;    // 16B records: same LDS, more slots
+                                       // (512*18*16 = 147KB) — fewer flush
+                                       // retries lets RPT=4 halve the
+                                       // per-tile barrier count
 static constexpr int A3_QUANT = 8;     // records per flushed packet (192 B)
 // staging slot stride: 24 B packed (a 28 B bank-spread pad was measured
 // SLOWER — 7.71 vs 6.58 ms/chunk — the kernel is flush-latency-bound, not
@@ -118,7 +122,8 @@ __global__ void __launch_bounds__(1024) k_agg3_scatter(
   // layout: records [nbuck][A3_CAP][24] | cnt[nbuck] | fl[nbuck] |
   //         base_line[nbuck] | dirty queue [nbuck] u16-as-u32 | qn
   uint8_t* stage = lds;
-  uint32_t* cnt = (uint32_t*)(lds + (size_t)nbuck * A3_CAP * REC);
+  constexpr int CAP = REC == 16 ? A3_CAP16 : A3_CAP;
+  uint32_t* cnt = (uint32_t*)(lds + (size_t)nbuck * CAP * REC);
   uint32_t* fl = cnt + nbuck;
   uint32_t* base_line = fl + nbuck;
   uint16_t* queue = (uint16_t*)(base_line + nbuck);
@@ -201,9 +206,9 @@ __global__ void __launch_bounds__(1024) k_agg3_scatter(
           continue;
         }
         uint32_t pos = atomicAdd(&cnt[b], 1u);
-        if (pos < A3_CAP) {
+        if (pos < CAP) {
           bool vvalid = !val_valid || bit_get3(val_valid, my[r]);
-          uint8_t* rec = stage + ((size_t)b * A3_CAP + pos) * REC;
+          uint8_t* rec = stage + ((size_t)b * CAP + pos) * REC;
           uint32_t rowv = (uint32_t)my[r] | (vvalid ? 0x80000000u : 0u);
           if (REC == 16) {
             *(uint64_t*)rec = koff | ((uint64_t)rowv << 32);
@@ -253,7 +258,7 @@ __global__ void __launch_bounds__(1024) k_agg3_scatter(
         uint32_t c = cnt[b];
         if (!c) continue;
         uint8_t* dst = out + ((size_t)base_line[b] << 6) + (size_t)fl[b] * REC;
-        const uint8_t* src = stage + (size_t)b * A3_CAP * REC;
+        const uint8_t* src = stage + (size_t)b * CAP * REC;
         for (uint32_t d = lane; d < c * (REC / 4); d += 64)
           ((uint32_t*)dst)[d] = ((const uint32_t*)src)[d];
       }
@@ -273,7 +278,7 @@ __global__ void __launch_bounds__(1024) k_agg3_scatter(
         uint32_t nfl = c & ~(uint32_t)(A3_QUANT - 1);
         if (!nfl) continue;
         uint8_t* dst = out + ((size_t)base_line[b] << 6) + (size_t)fl[b] * REC;
-        uint8_t* src = stage + (size_t)b * A3_CAP * REC;
+        uint8_t* src = stage + (size_t)b * CAP * REC;
         for (uint32_t d = sl; d < nfl * (REC / 4); d += 8)
           ((uint32_t*)dst)[d] = ((const uint32_t*)src)[d];
         uint32_t rem = c - nfl;
@@ -306,7 +311,8 @@ void launch_agg3_scatter(const int64_t* keys, const uint8_t* key_valid,
                          uint32_t* bypass_matrix, uint32_t* err_flag,
                          int rec, int64_t key_base, hipStream_t s) {
   const uint32_t nbuck = 1u << nbuck_log2;
-  size_t lds = (size_t)nbuck * A3_CAP * (size_t)rec + (size_t)nbuck * 16 +
+  const int cap = rec == 16 ? A3_CAP16 : A3_CAP;
+  size_t lds = (size_t)nbuck * cap * (size_t)rec + (size_t)nbuck * 16 +
                ((size_t)nbuck + 64) * 2 + 8;  // + queue + counters + bypass
   if (lds > 160 * 1024)
     throw std::runtime_error("agg3 scatter LDS over 160KB");
