@@ -318,15 +318,18 @@ void launch_agg3_scatter(const int64_t* keys, const uint8_t* key_valid,
     throw std::runtime_error("agg3 scatter LDS over 160KB");
   static int rpt = [] {
     const char* e = getenv("AURON_AGG2_RPT");
-    return (e && e[0] == '4') ? 4 : 2;
+    int v = e ? atoi(e) : 2;
+    return (v >= 2 && v <= 4) ? v : 2;
   }();
   const void* fn;
   if (rec == 16)
-    fn = rpt == 4 ? (const void*)k_agg3_scatter<4, 16>
-                  : (const void*)k_agg3_scatter<2, 16>;
+    fn = rpt == 4   ? (const void*)k_agg3_scatter<4, 16>
+         : rpt == 3 ? (const void*)k_agg3_scatter<3, 16>
+                    : (const void*)k_agg3_scatter<2, 16>;
   else
-    fn = rpt == 4 ? (const void*)k_agg3_scatter<4, 24>
-                  : (const void*)k_agg3_scatter<2, 24>;
+    fn = rpt == 4   ? (const void*)k_agg3_scatter<4, 24>
+         : rpt == 3 ? (const void*)k_agg3_scatter<3, 24>
+                    : (const void*)k_agg3_scatter<2, 24>;
   hipError_t e = hipFuncSetAttribute(
       fn, hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
   if (e != hipSuccess)
@@ -337,9 +340,13 @@ void launch_agg3_scatter(const int64_t* keys, const uint8_t* key_valid,
                      n, nbuck_log2, grid_log2, key_base, line_scan, out,    \
                      leftover, lo_n, bypass_matrix, err_flag)
   if (rec == 16) {
-    if (rpt == 4) A3_LAUNCH(4, 16); else A3_LAUNCH(2, 16);
+    if (rpt == 4) A3_LAUNCH(4, 16);
+    else if (rpt == 3) A3_LAUNCH(3, 16);
+    else A3_LAUNCH(2, 16);
   } else {
-    if (rpt == 4) A3_LAUNCH(4, 24); else A3_LAUNCH(2, 24);
+    if (rpt == 4) A3_LAUNCH(4, 24);
+    else if (rpt == 3) A3_LAUNCH(3, 24);
+    else A3_LAUNCH(2, 24);
   }
 #undef A3_LAUNCH
   check_launch3("k_agg3_scatter");
