@@ -536,6 +536,9 @@ void launch_def_expand_validity(const void* runs, int nruns,
 // scatter into 64B-aligned per-(block,bucket) ranges + 4096-slot bucket agg
 void launch_agg3_line_sizes(const uint32_t* counts, int64_t n,
                             uint32_t* sizes, int rec, hipStream_t s);
+void launch_keys_minmax(const int64_t* keys, const uint8_t* key_valid,
+                        int64_t n, unsigned long long* kminmax,
+                        hipStream_t s);
 // rec = 24 (i64 key) or 16 (u32 key offset from key_base — rows whose key
 // falls outside [key_base, key_base+2^32) bypass to the leftover list)
 void launch_agg3_scatter(const int64_t* keys, const uint8_t* key_valid,
