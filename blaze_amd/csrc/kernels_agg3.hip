@@ -79,10 +79,15 @@ void launch_agg3_line_sizes(const uint32_t* counts, int64_t n,
 
 // ---- v3 scatter ------------------------------------------------------------
 static constexpr int A3_CAP = 12;      // staged records per bucket (24B)
-static constexpr int A3_CAP16 = 18;    // 16B records: same LDS, more slots
-                                       // (512*18*16 = 147KB) — fewer flush
-                                       // retries lets RPT=4 halve the
-                                       // per-tile barrier count
+static constexpr int A3_CAP16 = 15;    // 16B records: more slots in less
+                                       // LDS (512*15*16 = 123KB). HARD CAP:
+                                       // pos must stay < 16 so a bucket
+                                       // crosses the 8-record packet
+                                       // boundary at most ONCE per tile —
+                                       // two queue entries for one bucket
+                                       // would race two flush subgroups
+                                       // (observed as corrupted group keys
+                                       // at CAP 18)
 static constexpr int A3_QUANT = 8;     // records per flushed packet (192 B)
 // staging slot stride: 24 B packed (a 28 B bank-spread pad was measured
 // SLOWER — 7.71 vs 6.58 ms/chunk — the kernel is flush-latency-bound, not
