@@ -154,3 +154,39 @@ def test_repartition_partition_order():
     rows = np.bincount(pid_in % world, minlength=world)
     np.testing.assert_array_equal(np.array(rp.rank_rows), rows)
     rp.free()
+
+
+def test_cuda_buf_wrap_and_exchange_ops():
+    """bench.py's N>1 exchange helpers on ONE GPU: the _CudaBuf
+    __cuda_array_interface__ wrapper must view raw HBM pointers zero-copy
+    (it only runs at world>1, so no other GPU test covers it), and the
+    validity-bitmap <-> byte-per-row round trip used around the all-to-all
+    must be exact."""
+    import torch
+
+    import bench
+
+    t = torch.arange(1000, dtype=torch.int64, device="cuda:0") * 3
+    w = torch.as_tensor(bench._CudaBuf(t.data_ptr(), 1000, "<i8"),
+                        device="cuda:0")
+    assert w.data_ptr() == t.data_ptr()  # zero-copy view
+    torch.testing.assert_close(w, t)
+    t[5] = -7
+    assert int(w[5].item()) == -7  # same memory
+
+    b = torch.randint(0, 256, (125,), dtype=torch.uint8, device="cuda:0")
+    wb = torch.as_tensor(bench._CudaBuf(b.data_ptr(), 125, "|u1"),
+                         device="cuda:0")
+    n = 1000
+    bits = torch.tensor([1, 2, 4, 8, 16, 32, 64, 128], dtype=torch.uint8,
+                        device="cuda:0")
+    vbytes = ((wb.unsqueeze(1) & bits) != 0).reshape(-1)[:n] \
+        .to(torch.uint8).contiguous()
+    # repack (the receive side of the exchange) and compare
+    pad = (-n) % 8
+    vb_pad = torch.cat([vbytes, torch.ones(pad, dtype=torch.uint8,
+                                           device="cuda:0")]) if pad else \
+        vbytes
+    repacked = ((vb_pad.view(-1, 8) != 0).to(torch.uint8) *
+                bits).sum(dim=1).to(torch.uint8)
+    torch.testing.assert_close(repacked, b)
