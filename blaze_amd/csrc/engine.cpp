@@ -1834,7 +1834,7 @@ class AggOp {
                            d_pctr_[pb].get<unsigned long long>(),
                            (int64_t)nbuck * 4096, d_pleft_[pb].get<PartRow>(),
                            d_pctr_[pb].get<unsigned long long>() + 1,
-                           t_.error_flag, 24, 0, nullptr, stream_);
+                           t_.error_flag, 24, 0, stream_);
         AURON_HIP(hipMemcpyAsync(pin_pctr_[pb].get(), d_pctr_[pb].get(), 24,
                                  hipMemcpyDeviceToHost, stream_));
         AURON_HIP(hipMemcpyAsync(pin_pctr_[pb].get<uint8_t>() + 32,
@@ -1908,8 +1908,6 @@ class AggOp {
       d_scanned_.alloc((mat + 1) * 4);
       d_offsets_.alloc((nbuck + 1) * 4);
       if (agg2_v3_) d_linesz_.alloc((mat + 1) * 4);
-      if (agg2_v3_)
-        d_cumg_.alloc((size_t)AGG3_NBUCK * (1u << AGG3_GRID_LOG2) * 4);
       size_t tb = 0;
       scan_counts_matrix(d_counts_.get<uint32_t>(), d_scanned_.get<uint32_t>(),
                          mat + 1, nullptr, &tb, stream_);
@@ -1996,8 +1994,7 @@ class AggOp {
                          (int64_t)AGG3_NBUCK * 4096,
                          d_leftover_.get<PartRow>(),
                          d_counters_.get<unsigned long long>() + 1,
-                         t_.error_flag, rec3, key_base16_,
-                         d_cumg_.get<uint32_t>(), stream_);
+                         t_.error_flag, rec3, key_base16_, stream_);
     } else {
     int64_t mat = (int64_t)nbuck << agg2_grid_log2_;
     launch_agg2_hist(keys, kv, chunk, AGG2_NBUCK_LOG2, agg2_grid_log2_,
@@ -2806,7 +2803,6 @@ class AggOp {
   bool packed16_ = false;     // later chunks use 16B partition records
   int64_t key_base16_ = 0;
   DevBuf d_kminmax_;
-  DevBuf d_cumg_;  // p16 bucket per-range cumsums (global, 512KB)
   bool agg2_pipe_ = false;
   bool agg2_conf_read_ = false;
   int agg2_grid_log2_ = 9;

@@ -567,7 +567,7 @@ void launch_agg3_bucket(const uint8_t* part, const uint32_t* counts,
                         unsigned long long* staged_n, int64_t staged_cap,
                         PartRow* leftover, unsigned long long* lo_n,
                         uint32_t* error_flag, int rec, int64_t key_base,
-                        uint32_t* cum_scratch, hipStream_t s);
+                        hipStream_t s);
 
 // GPU Parquet page staging (kernels_pq.hip): wave-per-page snappy decompress
 // + on-device def-level parse + dense compaction. Built by the host pre-scan

@@ -694,40 +694,25 @@ __global__ void __launch_bounds__(1024) k_agg3_bucket(
     const uint8_t* __restrict__ part, const uint32_t* __restrict__ counts,
     const uint32_t* __restrict__ bypass,
     const uint32_t* __restrict__ line_scan, int grid_log2, int is_int,
-    int nbuckets, int64_t key_base, uint32_t* __restrict__ cum_g,
-    StagedGroup* __restrict__ staged,
+    int nbuckets, int64_t key_base, StagedGroup* __restrict__ staged,
     unsigned long long* __restrict__ staged_n, int64_t staged_cap,
     PartRow* __restrict__ leftover, unsigned long long* __restrict__ lo_n,
     uint32_t* __restrict__ error_flag) {
-  // packed records carry u32 key OFFSETS, so the LDS window key array is
-  // u32 too: 96 KB -> 80 KB static LDS, which fits TWO workgroups per CU —
-  // double the occupancy of the latency-chained probe loop. The u32 empty
-  // sentinel 0xFFFFFFFF collides with koff 0xFFFFFFFF; that one offset is
-  // routed to the leftover list (exact, ~1-in-4B keys).
-  using lkey_t = std::conditional_t<REC == 16, uint32_t, int64_t>;
-  __shared__ lkey_t ls_key[L3SLOTS];
+  __shared__ int64_t ls_key[L3SLOTS];
   __shared__ double ls_sum[L3SLOTS];
   __shared__ uint32_t ls_cnt[L3SLOTS];
   __shared__ uint32_t ls_first[L3SLOTS];
-  constexpr uint32_t LKEY_EMPTY32 = 0xFFFFFFFFu;
-  // per-range cumulative record counts for this bucket (grid of ranges).
-  // REC 16 keeps them in GLOBAL scratch (cum_g, L1-hot after first touch):
-  // that makes the static LDS exactly 80 KB = TWO workgroups per CU for
-  // the latency-chained probe loop (84 KB with an LDS cum = one)
-  __shared__ uint32_t cum_s[REC == 16 ? 1 : (1 << AGG2_GRID_LOG2_MAX)];
+  // per-range cumulative record counts for this bucket (grid of ranges)
+  __shared__ uint32_t cum[1 << AGG2_GRID_LOG2_MAX];
   const int nrange = 1 << grid_log2;
 
   for (int b = blockIdx.x; b < nbuckets; b += gridDim.x) {
     for (int s = threadIdx.x; s < L3SLOTS; s += blockDim.x) {
-      if constexpr (REC == 16)
-        ls_key[s] = LKEY_EMPTY32;
-      else
-        ls_key[s] = KEY_EMPTY3;
+      ls_key[s] = KEY_EMPTY3;
       ls_sum[s] = 0.0;
       ls_cnt[s] = 0;
       ls_first[s] = 0xFFFFFFFFu;
     }
-    uint32_t* cum = REC == 16 ? cum_g + (size_t)b * nrange : cum_s;
     // serial scan of <=1024 range counts (one thread; ~us)
     if (threadIdx.x == 0) {
       uint32_t acc = 0;
@@ -736,7 +721,6 @@ __global__ void __launch_bounds__(1024) k_agg3_bucket(
         acc += counts[e] - bypass[e];  // bypassed rows were never written
         cum[k] = acc;
       }
-      if (REC == 16) __threadfence();  // global cum visible to the block
     }
     __syncthreads();
     uint32_t total = cum[nrange - 1];
@@ -770,29 +754,8 @@ __global__ void __launch_bounds__(1024) k_agg3_bucket(
       uint32_t row = rowv & 0x7FFFFFFFu;
       bool vvalid = (rowv & 0x80000000u) != 0;
       int found = -1;
-      if constexpr (REC == 16) {
-        uint32_t ko = (uint32_t)(uint64_t)(k - key_base);
-        if (ko != LKEY_EMPTY32) {  // sentinel collision -> leftover below
-          for (int p = 0; p < L3PROBE; p++) {
-            uint32_t cur = (uint32_t)ls_key[h];
-            if (cur == ko) {
-              found = (int)h;
-              break;
-            }
-            if (cur == LKEY_EMPTY32) {
-              uint32_t prev = atomicCAS((unsigned int*)&ls_key[h],
-                                        LKEY_EMPTY32, ko);
-              if (prev == LKEY_EMPTY32 || prev == ko) {
-                found = (int)h;
-                break;
-              }
-            }
-            h = (h + 1) & (L3SLOTS - 1);
-          }
-        }
-      } else {
       for (int p = 0; p < L3PROBE; p++) {
-        int64_t cur = (int64_t)ls_key[h];
+        int64_t cur = ls_key[h];
         if (cur == k) {
           found = (int)h;
           break;
@@ -807,7 +770,6 @@ __global__ void __launch_bounds__(1024) k_agg3_bucket(
           }
         }
         h = (h + 1) & (L3SLOTS - 1);
-      }
       }
       if (found >= 0) {
         atomicMin(&ls_first[found], row);
@@ -852,19 +814,13 @@ __global__ void __launch_bounds__(1024) k_agg3_bucket(
     }
     __syncthreads();
     for (int s = threadIdx.x; s < L3SLOTS; s += blockDim.x) {
-      if constexpr (REC == 16) {
-        if ((uint32_t)ls_key[s] == LKEY_EMPTY32) continue;
-      } else {
-        if ((int64_t)ls_key[s] == KEY_EMPTY3) continue;
-      }
+      if (ls_key[s] == KEY_EMPTY3) continue;
       unsigned long long p = atomicAdd(staged_n, 1ull);
       if ((int64_t)p >= staged_cap) {
         atomicOr(error_flag, 2u);
         continue;
       }
-      staged[p].key = REC == 16
-                          ? key_base + (int64_t)(uint32_t)ls_key[s]
-                          : (int64_t)ls_key[s];
+      staged[p].key = ls_key[s];
       staged[p].sum = ls_sum[s];
       staged[p].cnt_first =
           ((unsigned long long)ls_cnt[s] << 32) | ls_first[s];
@@ -880,20 +836,17 @@ void launch_agg3_bucket(const uint8_t* part, const uint32_t* counts,
                         unsigned long long* staged_n, int64_t staged_cap,
                         PartRow* leftover, unsigned long long* lo_n,
                         uint32_t* error_flag, int rec, int64_t key_base,
-                        uint32_t* cum_scratch, hipStream_t s) {
-  if (rec == 16) {
-    if (!cum_scratch)
-      throw std::runtime_error("agg3 bucket p16 needs cum scratch");
+                        hipStream_t s) {
+  if (rec == 16)
     hipLaunchKernelGGL((k_agg3_bucket<16>), dim3(nbuckets), dim3(1024), 0, s,
                        part, counts, bypass, line_scan, grid_log2, is_int,
-                       nbuckets, key_base, cum_scratch, staged, staged_n,
-                       staged_cap, leftover, lo_n, error_flag);
-  } else {
+                       nbuckets, key_base, staged, staged_n, staged_cap,
+                       leftover, lo_n, error_flag);
+  else
     hipLaunchKernelGGL((k_agg3_bucket<24>), dim3(nbuckets), dim3(1024), 0, s,
                        part, counts, bypass, line_scan, grid_log2, is_int,
-                       nbuckets, key_base, nullptr, staged, staged_n,
-                       staged_cap, leftover, lo_n, error_flag);
-  }
+                       nbuckets, key_base, staged, staged_n, staged_cap,
+                       leftover, lo_n, error_flag);
   check_launch3("k_agg3_bucket");
 }
 
