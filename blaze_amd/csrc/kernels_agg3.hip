@@ -392,8 +392,9 @@ void launch_agg3_scatter(const int64_t* keys, const uint8_t* key_valid,
 // ---- v4 scatter: barrier-free producer/flusher rings -----------------------
 // The v3 tile loop parks waves 73% of the time on its per-tile barriers
 // (SQ_WAIT_ANY ~7x ACTIVE, gpurun_out sq pass). v4 removes every barrier
-// from the hot loop: 14 WORKER waves append records into per-bucket 8-slot
-// LDS rings; 2 FLUSHER waves (8-lane subgroups, 8 buckets in flight each)
+// from the hot loop: WORKER waves (AURON_AGG2_V4_WW, default 8 of 16)
+// append records into per-bucket 8-slot LDS rings; the remaining FLUSHER
+// waves (8-lane subgroups, 8 buckets in flight each)
 // continuously drain 4-record 96 B quanta into the same 64B-aligned
 // per-(block,bucket) ranges. Commit protocol: each ring slot carries a
 // sequence TAG (= the absolute ring position) written AFTER the record by
